@@ -1,0 +1,117 @@
+"""Table: ordered collection of named Columns (reference: ``table_info``,
+bodo/libs/_bodo_common.h)."""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence
+
+import pandas as pd
+import pyarrow as pa
+import torch
+
+from .column import Column
+
+
+class Table:
+    __slots__ = ("names", "columns", "_length")
+
+    def __init__(self, names: Sequence[str], columns: Sequence[Column],
+                 length: Optional[int] = None):
+        assert len(names) == len(columns)
+        self.names: List[str] = list(names)
+        self.columns: List[Column] = list(columns)
+        if length is None:
+            length = len(columns[0]) if columns else 0
+        self._length = length
+
+    def __len__(self) -> int:
+        return self._length
+
+    @property
+    def num_columns(self) -> int:
+        return len(self.columns)
+
+    def column(self, name: str) -> Column:
+        return self.columns[self.names.index(name)]
+
+    def has_column(self, name: str) -> bool:
+        return name in self.names
+
+    def as_dict(self) -> Dict[str, Column]:
+        return dict(zip(self.names, self.columns))
+
+    def select(self, names: Sequence[str]) -> "Table":
+        return Table(list(names), [self.column(n) for n in names], self._length)
+
+    def with_column(self, name: str, col: Column) -> "Table":
+        names, cols = list(self.names), list(self.columns)
+        if name in names:
+            cols[names.index(name)] = col
+        else:
+            names.append(name)
+            cols.append(col)
+        return Table(names, cols, self._length)
+
+    def rename(self, mapping: Dict[str, str]) -> "Table":
+        return Table([mapping.get(n, n) for n in self.names], self.columns, self._length)
+
+    @property
+    def device(self) -> torch.device:
+        if not self.columns:
+            return torch.device("cpu")
+        return self.columns[0].device
+
+    def to_device(self, device) -> "Table":
+        return Table(self.names, [c.to_device(device) for c in self.columns], self._length)
+
+    def nbytes(self) -> int:
+        return sum(c.nbytes() for c in self.columns)
+
+    # ------------------------------------------------------------------
+    @staticmethod
+    def from_arrow(tbl: pa.Table, device="cpu") -> "Table":
+        cols = [Column.from_arrow(tbl.column(i).combine_chunks(), device)
+                for i in range(tbl.num_columns)]
+        return Table(tbl.column_names, cols, tbl.num_rows)
+
+    def to_arrow(self) -> pa.Table:
+        return pa.table(
+            {n: c.to_arrow() for n, c in zip(self.names, self.columns)}
+        ) if self.columns else pa.table({})
+
+    @staticmethod
+    def from_pandas(df: pd.DataFrame, device="cpu") -> "Table":
+        tbl = pa.Table.from_pandas(df, preserve_index=False)
+        return Table.from_arrow(tbl, device)
+
+    def to_pandas(self) -> pd.DataFrame:
+        if not self.columns:
+            return pd.DataFrame(index=range(self._length))
+        out = self.to_arrow().to_pandas()
+        # arrow gives large_string -> object; fine for pandas parity
+        return out
+
+    @staticmethod
+    def empty_like(other: "Table") -> "Table":
+        import numpy as np
+
+        from . import types as bt
+        from .types import TypeKind
+
+        cols = []
+        dev = other.device
+        for c in other.columns:
+            if c.dtype.kind == TypeKind.STRING:
+                cols.append(Column(
+                    bt.string,
+                    torch.zeros(0, dtype=torch.uint8, device=dev), None,
+                    offsets=torch.zeros(1, dtype=torch.int64, device=dev), length=0))
+            else:
+                cols.append(Column(
+                    c.dtype,
+                    torch.zeros(0, dtype=bt.torch_storage_dtype(c.dtype), device=dev),
+                    None, dictionary=c.dictionary, length=0))
+        return Table(list(other.names), cols, 0)
+
+    def __repr__(self) -> str:  # pragma: no cover
+        return f"Table({len(self)} rows x {self.num_columns} cols, device={self.device})"

@@ -1,0 +1,438 @@
+"""Plan executor: interprets an optimized logical plan against the rank-local
+shard, inserting distributed exchanges (RCCL) where operators need them.
+
+Reference role: bodo/pandas/_executor.cpp + _physical_conv.cpp + the streaming
+operator states — redesigned for MI355X as a materialized (whole-shard)
+columnar executor first: with 288 GB HBM3E per GPU the SF100-class working
+sets fit resident, so operators run one fused device pass per shard instead
+of 32k-row host morsels; out-of-core streaming is layered on top for
+larger-than-HBM inputs (see engine/streaming notes).
+"""
+
+from __future__ import annotations
+
+import uuid
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import pandas as pd
+import torch
+
+from .. import ops
+from ..core import types as bt
+from ..core.column import Column
+from ..core.table import Table
+from ..core.types import TypeKind
+from ..ops import evaluate as ev
+from ..ops import relational as rel
+from ..parallel import comm
+from ..plan import nodes as pn
+from ..plan.expr import BinOp, ColRef, Const, Expr
+
+# registry of in-memory source objects (host DataFrames or distributed shards)
+_OBJECT_REGISTRY: Dict[str, object] = {}
+
+
+def register_object(obj, data_id: Optional[str] = None) -> str:
+    key = data_id or uuid.uuid4().hex
+    _OBJECT_REGISTRY[key] = obj
+    return key
+
+
+def get_object(data_id: str):
+    return _OBJECT_REGISTRY[data_id]
+
+
+def delete_object(data_id: str):
+    _OBJECT_REGISTRY.pop(data_id, None)
+
+
+class ExecutionContext:
+    def __init__(self, device: Optional[str] = None):
+        from .. import config
+
+        if device is None:
+            device = config.default_device()
+        self.device = torch.device(device)
+        self.rank = comm.get_rank()
+        self.world = comm.get_world_size()
+
+
+def execute(plan: pn.PlanNode, ctx: Optional[ExecutionContext] = None) -> Table:
+    """Execute the plan; returns the rank-local shard of the result."""
+    from .optimizer import optimize
+
+    ctx = ctx or ExecutionContext()
+    plan = optimize(plan)
+    return _exec(plan, ctx)
+
+
+def _exec(node: pn.PlanNode, ctx: ExecutionContext) -> Table:
+    h = _HANDLERS.get(type(node))
+    if h is None:
+        raise NotImplementedError(f"no executor for {type(node).__name__}")
+    return h(node, ctx)
+
+
+# ---------------------------------------------------------------- sources
+
+def _exec_parquet(node: pn.ParquetScan, ctx) -> Table:
+    from ..io import parquet as pio
+
+    return pio.read_shard(node.path, node.columns, node.filters, ctx)
+
+
+def _exec_csv(node: pn.CsvScan, ctx) -> Table:
+    from ..io import csv as cio
+
+    return cio.read_shard(node.path, dict(node.options), node.columns, ctx)
+
+
+def _exec_pandas_scan(node: pn.PandasScan, ctx) -> Table:
+    obj = get_object(node.data_id)
+    if node.distributed:
+        assert isinstance(obj, Table)
+        return obj.to_device(ctx.device)
+    if isinstance(obj, Table):
+        tbl = obj
+    else:
+        tbl = Table.from_pandas(obj)
+    # replicated host data: slice this rank's block (1D distribution)
+    n = len(tbl)
+    w, r = ctx.world, ctx.rank
+    start, stop = _block_bounds(n, w, r)
+    local = ops.slice_table(tbl, start, stop) if w > 1 else tbl
+    return local.to_device(ctx.device)
+
+
+def _block_bounds(n: int, w: int, r: int) -> Tuple[int, int]:
+    base, rem = divmod(n, w)
+    start = r * base + min(r, rem)
+    stop = start + base + (1 if r < rem else 0)
+    return start, stop
+
+
+# ---------------------------------------------------------------- unary
+
+def _exec_projection(node: pn.Projection, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    return ev.project(child, node.names, node.exprs)
+
+
+def _exec_filter(node: pn.Filter, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    return ev.eval_filter(node.cond, child)
+
+
+SINGLE_PHASE_AGGS = {"median", "nunique", "var", "std", "quantile"}
+
+
+def _exec_aggregate(node: pn.Aggregate, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    keys = list(node.keys)
+    aggs = list(node.aggs)
+    if ctx.world == 1:
+        return rel.groupby_local(child, keys, aggs, node.dropna)
+    if any(a[2] in SINGLE_PHASE_AGGS for a in aggs):
+        # shuffle raw rows by key hash, then single local groupby
+        shuffled = _shuffle_by_keys(child, keys, ctx)
+        return rel.groupby_local(shuffled, keys, aggs, node.dropna)
+    # two-phase: local pre-agg -> shuffle groups -> combine
+    partial_aggs, final_map = _decompose_aggs(aggs)
+    local = rel.groupby_local(child, keys, partial_aggs, node.dropna)
+    shuffled = _shuffle_by_keys(local, keys, ctx)
+    combine_aggs = [(name, name, rel.COMBINE_FUNC[f]) for name, _, f in partial_aggs]
+    combined = rel.groupby_local(shuffled, keys, combine_aggs, node.dropna)
+    # final projection (mean = sum/count etc.)
+    names, exprs = list(keys), [ColRef(k) for k in keys]
+    for out_name, expr in final_map:
+        names.append(out_name)
+        exprs.append(expr)
+    return ev.project(combined, names, exprs)
+
+
+def _decompose_aggs(aggs):
+    """Split each agg into partial physical aggs + a final expression."""
+    partials: List[Tuple[str, str, str]] = []
+    finals: List[Tuple[str, Expr]] = []
+    seen = {}
+
+    def add_partial(in_name, func):
+        key = (in_name, func)
+        if key not in seen:
+            nm = f"__p{len(partials)}_{func}"
+            seen[key] = nm
+            partials.append((nm, in_name, func))
+        return seen[key]
+
+    for out_name, in_name, func in aggs:
+        if func == "mean":
+            s = add_partial(in_name, "sum")
+            c = add_partial(in_name, "count")
+            finals.append((out_name, BinOp("div", ColRef(s), ColRef(c))))
+        elif func in ("sum", "min", "max", "count", "size", "first", "last", "prod"):
+            p = add_partial(in_name, func)
+            finals.append((out_name, ColRef(p)))
+        else:
+            raise NotImplementedError(f"two-phase agg {func}")
+    return partials, finals
+
+
+def _shuffle_by_keys(tbl: Table, keys: Sequence[str], ctx) -> Table:
+    cols = [tbl.column(k) for k in keys]
+    h = ops.hash_columns(cols)
+    part = torch.remainder(h, ctx.world)
+    part = torch.where(part < 0, part + ctx.world, part)
+    return comm.shuffle_table(tbl, part)
+
+
+def _exec_sort(node: pn.Sort, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    asc = list(node.ascending) or [True] * len(node.keys)
+    if ctx.world > 1:
+        child = _range_partition(child, list(node.keys), asc, node.na_position, ctx)
+    cols = [child.column(k) for k in node.keys]
+    idx = ops.sort_indices(cols, asc, node.na_position)
+    return ops.take_table(child, idx)
+
+
+def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:
+    """Sample-based range partitioning so rank r holds globally-contiguous key
+    range r (PSRS step 1-3; reference: streaming/_sort.h reservoir sampling +
+    bounds)."""
+    w = ctx.world
+    n = len(tbl)
+    # sample up to 64*w rows of the key columns
+    k = min(n, 64 * w)
+    if k > 0:
+        pos = torch.randint(0, n, (k,), device=tbl.device) if n > k else \
+            torch.arange(n, device=tbl.device)
+        sample = ops.take_table(tbl.select(list(keys)), pos).to_device("cpu").to_pandas()
+    else:
+        sample = tbl.select(list(keys)).to_device("cpu").to_pandas()
+    all_samples = comm.allgather_obj(sample)
+    merged = pd.concat(all_samples, ignore_index=True)
+    merged = merged.sort_values(list(keys), ascending=asc, na_position=na_position)
+    if len(merged) == 0:
+        return tbl
+    # pick w-1 splitters
+    q = [int(len(merged) * (i + 1) / w) for i in range(w - 1)]
+    splitters = merged.iloc[[min(x, len(merged) - 1) for x in q]]
+    # assign each local row a partition by comparing against splitters.
+    # host comparison on the key sample of local rows (vectorized via
+    # pandas merge_sorted rank); for performance the GPU path will move to a
+    # device searchsorted on normalized keys.
+    local_keys = tbl.select(list(keys)).to_device("cpu").to_pandas()
+    part = np.zeros(n, dtype=np.int64)
+    for i in range(w - 1):
+        row = splitters.iloc[i]
+        gt = _row_greater(local_keys, row, keys, asc, na_position)
+        part = np.where(gt, i + 1, part)
+    return comm.shuffle_table(tbl, torch.from_numpy(part).to(tbl.device))
+
+
+def _row_greater(df: pd.DataFrame, row, keys, asc, na_position) -> np.ndarray:
+    """True where df-row sorts strictly after `row` under keys/asc."""
+    n = len(df)
+    gt = np.zeros(n, dtype=bool)
+    eq = np.ones(n, dtype=bool)
+    for k, a in zip(keys, asc):
+        col = df[k]
+        rv = row[k]
+        colna = pd.isna(col).to_numpy()
+        rvna = pd.isna(rv)
+        if rvna:
+            c_gt = np.zeros(n, dtype=bool) if na_position == "last" else ~colna
+            c_eq = colna
+        else:
+            with np.errstate(invalid="ignore"):
+                raw_gt = (col > rv).to_numpy(dtype=bool, na_value=False) if a else (col < rv).to_numpy(dtype=bool, na_value=False)
+                c_eq = (col == rv).to_numpy(dtype=bool, na_value=False)
+            c_gt = np.where(colna, na_position == "last", raw_gt)
+        gt |= eq & c_gt
+        eq &= c_eq
+    return gt
+
+
+def _exec_limit(node: pn.Limit, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    n = node.n
+    if ctx.world == 1:
+        if node.tail:
+            return ops.slice_table(child, max(0, len(child) - n), len(child))
+        return ops.slice_table(child, node.offset, node.offset + n)
+    lengths = comm.allgather_obj(len(child))
+    if node.tail:
+        suffix_after = sum(lengths[ctx.rank + 1:])
+        take = max(0, min(len(child), n - suffix_after))
+        return ops.slice_table(child, len(child) - take, len(child))
+    prefix = sum(lengths[:ctx.rank])
+    start = max(0, node.offset - prefix)
+    stop = max(0, node.offset + n - prefix)
+    return ops.slice_table(child, start, min(stop, len(child)))
+
+
+def _exec_distinct(node: pn.Distinct, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    local = rel.distinct_local(child, node.subset, node.keep)
+    if ctx.world > 1:
+        keys = list(node.subset) if node.subset else list(local.names)
+        shuffled = _shuffle_by_keys(local, keys, ctx)
+        local = rel.distinct_local(shuffled, node.subset, node.keep)
+    return local
+
+
+def _exec_sample(node: pn.Sample, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    n_local = len(child)
+    if node.frac is not None:
+        k = int(round(n_local * node.frac))
+    else:
+        total = sum(comm.allgather_obj(n_local))
+        share = (node.n or 0) * n_local / max(total, 1)
+        k = int(round(share))
+    g = torch.Generator(device="cpu")
+    if node.seed is not None:
+        g.manual_seed(node.seed + ctx.rank)
+    pos = torch.randperm(n_local, generator=g)[:k].to(child.device)
+    return ops.take_table(child, torch.sort(pos).values)
+
+
+def _exec_shuffle_by_key(node: pn.ShuffleByKey, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    if ctx.world == 1:
+        return child
+    return _shuffle_by_keys(child, list(node.keys), ctx)
+
+
+def _exec_map_partitions(node: pn.MapPartitions, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    pdf = child.to_pandas()
+    res = node.func(pdf, *node.args)
+    if isinstance(res, pd.Series):
+        res = res.to_frame(name=res.name if res.name is not None else "0")
+    return Table.from_pandas(res.reset_index(drop=True), ctx.device)
+
+
+# ---------------------------------------------------------------- binary
+
+BROADCAST_JOIN_THRESHOLD = 256 * 1024 * 1024  # bytes (reference: gpu_join.h:92)
+
+
+def _exec_join(node: pn.Join, ctx) -> Table:
+    left = _exec(node.left, ctx)
+    right = _exec(node.right, ctx)
+    if ctx.world == 1 or node.how == "cross":
+        if ctx.world > 1 and node.how == "cross":
+            right = comm.allgather_table(right)
+        return rel.join_local(left, right, node.left_on, node.right_on,
+                              node.how, node.suffixes)
+    # broadcast the smaller side when cheap and semantics allow
+    lsize = sum(comm.allgather_obj(left.nbytes()))
+    rsize = sum(comm.allgather_obj(right.nbytes()))
+    if rsize <= BROADCAST_JOIN_THRESHOLD and node.how in ("inner", "left", "semi", "anti"):
+        right_full = comm.allgather_table(right)
+        return rel.join_local(left, right_full, node.left_on, node.right_on,
+                              node.how, node.suffixes)
+    if lsize <= BROADCAST_JOIN_THRESHOLD and node.how in ("inner", "right"):
+        left_full = comm.allgather_table(left)
+        out = rel.join_local(left_full, right, node.left_on, node.right_on,
+                             node.how, node.suffixes)
+        return out
+    # hash-shuffle both sides by join keys
+    lh = ops.hash_columns([left.column(k) for k in node.left_on])
+    rh = ops.hash_columns([right.column(k) for k in node.right_on])
+    lp = torch.remainder(lh, ctx.world)
+    lp = torch.where(lp < 0, lp + ctx.world, lp)
+    rp = torch.remainder(rh, ctx.world)
+    rp = torch.where(rp < 0, rp + ctx.world, rp)
+    lshuf = comm.shuffle_table(left, lp)
+    rshuf = comm.shuffle_table(right, rp)
+    return rel.join_local(lshuf, rshuf, node.left_on, node.right_on,
+                          node.how, node.suffixes)
+
+
+def _exec_union(node: pn.Union, ctx) -> Table:
+    parts = [_exec(c, ctx) for c in node.inputs]
+    # align columns by name of the first input
+    names = parts[0].names
+    aligned = [p.select([n for n in names]) if p.names != names else p for p in parts]
+    out = ops.concat_tables(aligned)
+    if node.distinct:
+        out = rel.distinct_local(out)
+        if ctx.world > 1:
+            out = _shuffle_by_keys(out, list(out.names), ctx)
+            out = rel.distinct_local(out)
+    return out
+
+
+# ---------------------------------------------------------------- sinks
+
+def _exec_parquet_write(node: pn.ParquetWrite, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    from ..io import parquet as pio
+
+    pio.write_shard(child, node.path, node.compression, ctx)
+    return Table([], [], 0)
+
+
+def _exec_reduce(node: pn.Reduce, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    out_names, out_cols = [], []
+    for out_name, in_name, func in node.aggs:
+        partial = ops.reduce_column(child.column(in_name), func)
+        partials = comm.allgather_obj(partial)
+        val = _combine_reduce(partials, func)
+        out_names.append(out_name)
+        arr = pd.Series([val])
+        out_cols.append(Column.from_numpy(arr.to_numpy(), ctx.device))
+    return Table(out_names, out_cols, 1)
+
+
+def _combine_reduce(partials: List[dict], func: str):
+    if func in ("sum",):
+        return sum(p["sum"] for p in partials)
+    if func in ("count", "size"):
+        return sum(p["count"] for p in partials)
+    if func == "mean":
+        c = sum(p["count"] for p in partials)
+        return (sum(p["sum"] for p in partials) / c) if c else float("nan")
+    if func == "min":
+        vals = [p["min"] for p in partials if p["min"] is not None]
+        return min(vals) if vals else None
+    if func == "max":
+        vals = [p["max"] for p in partials if p["max"] is not None]
+        return max(vals) if vals else None
+    if func in ("var", "std"):
+        c = sum(p["count"] for p in partials)
+        s = sum(p["sum"] for p in partials)
+        ss = sum(p["sumsq"] for p in partials)
+        if c < 2:
+            return float("nan")
+        var = (ss - s * s / c) / (c - 1)
+        return var if func == "var" else var ** 0.5
+    if func == "any":
+        return any(p["any"] for p in partials)
+    if func == "all":
+        return all(p["all"] for p in partials)
+    raise NotImplementedError(func)
+
+
+_HANDLERS = {
+    pn.ParquetScan: _exec_parquet,
+    pn.CsvScan: _exec_csv,
+    pn.PandasScan: _exec_pandas_scan,
+    pn.Projection: _exec_projection,
+    pn.Filter: _exec_filter,
+    pn.Aggregate: _exec_aggregate,
+    pn.Sort: _exec_sort,
+    pn.Limit: _exec_limit,
+    pn.Distinct: _exec_distinct,
+    pn.Sample: _exec_sample,
+    pn.MapPartitions: _exec_map_partitions,
+    pn.ShuffleByKey: _exec_shuffle_by_key,
+    pn.Join: _exec_join,
+    pn.Union: _exec_union,
+    pn.ParquetWrite: _exec_parquet_write,
+    pn.Reduce: _exec_reduce,
+}

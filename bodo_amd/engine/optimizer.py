@@ -1,0 +1,192 @@
+"""Logical plan optimizer.
+
+Own rule set instead of vendoring DuckDB (reference vendors ~115 kLoC of
+DuckDB optimizer at bodo/pandas/vendor/duckdb; SURVEY §7 step 6 calls for a
+native rule set): column pruning into scans, filter pushdown into parquet
+scans (row-group stats + row-level via Arrow dataset), projection fusion,
+filter fusion.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Set
+
+from ..plan import nodes as pn
+from ..plan.expr import (
+    BinOp, BoolOp, ColRef, Cmp, Const, Expr, expr_columns, substitute_cols,
+)
+
+
+def optimize(plan: pn.PlanNode) -> pn.PlanNode:
+    plan = fuse_projections(plan)
+    plan = push_filters(plan)
+    plan = prune_columns(plan, None)
+    return plan
+
+
+# ---------------------------------------------------------------- fusion
+
+def fuse_projections(node: pn.PlanNode) -> pn.PlanNode:
+    node = node.with_children(*[fuse_projections(c) for c in node.children()]) \
+        if node.children() else node
+    if isinstance(node, pn.Projection) and isinstance(node.child, pn.Projection):
+        inner = node.child
+        mapping = dict(zip(inner.names, inner.exprs))
+        new_exprs = tuple(substitute_cols(e, mapping) for e in node.exprs)
+        return pn.Projection(inner.child, node.names, new_exprs)
+    return node
+
+
+# ---------------------------------------------------------------- filters
+
+def push_filters(node: pn.PlanNode) -> pn.PlanNode:
+    node = node.with_children(*[push_filters(c) for c in node.children()]) \
+        if node.children() else node
+    if isinstance(node, pn.Filter):
+        child = node.child
+        # split conjunctions
+        conjuncts = _split_and(node.cond)
+        if isinstance(child, pn.ParquetScan):
+            pushable = [c for c in conjuncts if _arrow_convertible(c)]
+            rest = [c for c in conjuncts if not _arrow_convertible(c)]
+            if pushable:
+                new_scan = pn.ParquetScan(
+                    child.path, child.columns,
+                    child.filters + tuple(pushable), child.schema_names)
+                if rest:
+                    return pn.Filter(new_scan, _join_and(rest))
+                return new_scan
+        if isinstance(child, pn.Filter):
+            return pn.Filter(child.child, _join_and(
+                _split_and(child.cond) + conjuncts))
+        if isinstance(child, pn.Projection):
+            # push through projection when every referenced column is a
+            # plain passthrough ColRef
+            mapping = dict(zip(child.names, child.exprs))
+            refs = expr_columns(node.cond)
+            if all(isinstance(mapping.get(r), ColRef) for r in refs):
+                newcond = substitute_cols(node.cond, mapping)
+                return pn.Projection(
+                    push_filters(pn.Filter(child.child, newcond)),
+                    child.names, child.exprs)
+    return node
+
+
+def _split_and(e: Expr) -> List[Expr]:
+    if isinstance(e, BoolOp) and e.op == "and":
+        return _split_and(e.left) + _split_and(e.right)
+    return [e]
+
+
+def _join_and(es: Sequence[Expr]) -> Expr:
+    out = es[0]
+    for e in es[1:]:
+        out = BoolOp("and", out, e)
+    return out
+
+
+def _arrow_convertible(e: Expr) -> bool:
+    """Filters we can hand to the Arrow dataset scanner exactly."""
+    if isinstance(e, Cmp):
+        lc = isinstance(e.left, ColRef) and isinstance(e.right, Const)
+        rc = isinstance(e.right, ColRef) and isinstance(e.left, Const)
+        return lc or rc
+    from ..plan.expr import IsIn, IsNull
+
+    if isinstance(e, IsIn):
+        return isinstance(e.operand, ColRef)
+    if isinstance(e, IsNull):
+        return isinstance(e.operand, ColRef)
+    if isinstance(e, BoolOp):
+        return _arrow_convertible(e.left) and _arrow_convertible(e.right)
+    return False
+
+
+# ---------------------------------------------------------------- pruning
+
+def prune_columns(node: pn.PlanNode, required: Optional[Set[str]]) -> pn.PlanNode:
+    """Rewrite the tree so each subtree produces at least `required` columns
+    (None = all).  Narrows ParquetScan.columns for IO pruning."""
+    if isinstance(node, pn.ParquetScan):
+        if required is None:
+            return node
+        schema = list(node.schema_names) or None
+        filt_cols = set()
+        for f in node.filters:
+            filt_cols |= expr_columns(f)
+        want = set(required) | filt_cols
+        if schema:
+            cols = tuple(c for c in schema if c in want)
+        else:
+            cols = tuple(sorted(want))
+        return pn.ParquetScan(node.path, cols, node.filters, node.schema_names)
+    if isinstance(node, pn.CsvScan):
+        if required is None:
+            return node
+        schema = list(node.schema_names) or None
+        if schema:
+            cols = tuple(c for c in schema if c in required)
+        else:
+            cols = tuple(sorted(required))
+        return pn.CsvScan(node.path, node.options, cols, node.schema_names)
+    if isinstance(node, pn.Projection):
+        if required is not None:
+            keep = [(n, e) for n, e in zip(node.names, node.exprs) if n in required]
+            if not keep:  # keep at least one column for row count
+                keep = [(node.names[0], node.exprs[0])] if node.names else []
+        else:
+            keep = list(zip(node.names, node.exprs))
+        child_req = set()
+        for _, e in keep:
+            child_req |= expr_columns(e)
+        child = prune_columns(node.child, child_req)
+        return pn.Projection(child, tuple(n for n, _ in keep),
+                             tuple(e for _, e in keep))
+    if isinstance(node, pn.Filter):
+        child_req = None if required is None else set(required) | expr_columns(node.cond)
+        return pn.Filter(prune_columns(node.child, child_req), node.cond)
+    if isinstance(node, pn.Aggregate):
+        child_req = set(node.keys) | {a[1] for a in node.aggs if a[1]}
+        return node.with_children(prune_columns(node.child, child_req))
+    if isinstance(node, pn.Reduce):
+        child_req = {a[1] for a in node.aggs if a[1]}
+        return node.with_children(prune_columns(node.child, child_req))
+    if isinstance(node, (pn.Sort, pn.Limit, pn.Distinct, pn.Sample)):
+        req2 = None
+        if required is not None:
+            req2 = set(required)
+            if isinstance(node, pn.Sort):
+                req2 |= set(node.keys)
+            if isinstance(node, pn.Distinct) and node.subset:
+                req2 |= set(node.subset)
+        return node.with_children(prune_columns(node.children()[0], req2))
+    if isinstance(node, pn.Join):
+        lcols = node.left.out_columns()
+        rcols = node.right.out_columns()
+        if required is None or lcols is None or rcols is None:
+            lreq = rreq = None
+        else:
+            sl, sr = set(lcols), set(rcols)
+            sfx_l, sfx_r = node.suffixes
+            lreq, rreq = set(node.left_on), set(node.right_on)
+            for r in required:
+                base_l = r[:-len(sfx_l)] if sfx_l and r.endswith(sfx_l) else None
+                base_r = r[:-len(sfx_r)] if sfx_r and r.endswith(sfx_r) else None
+                if r in sl:
+                    lreq.add(r)
+                if r in sr:
+                    rreq.add(r)
+                if base_l and base_l in sl and base_l in sr:
+                    lreq.add(base_l)
+                if base_r and base_r in sr and base_r in sl:
+                    rreq.add(base_r)
+        return pn.Join(
+            prune_columns(node.left, lreq), prune_columns(node.right, rreq),
+            node.left_on, node.right_on, node.how, node.suffixes)
+    if isinstance(node, pn.Union):
+        return node.with_children(*[prune_columns(c, required) for c in node.inputs])
+    # unknown shape: don't prune below
+    ch = node.children()
+    if not ch:
+        return node
+    return node.with_children(*[prune_columns(c, None) for c in ch])

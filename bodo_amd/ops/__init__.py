@@ -1,0 +1,357 @@
+"""Physical kernel layer: dispatches each op to the HIP extension on HBM-resident
+columns or to host (numpy/pandas/pyarrow) kernels on CPU columns.
+
+The GPU implementations live in csrc/ (hand-written gfx950 HIP) and are
+mandatory on a GPU box: if a column is on CUDA and the extension is missing
+we raise rather than silently falling back (round-end native check).
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from ..core import types as bt
+from ..core.column import Column
+from ..core.table import Table
+from ..core.types import TypeKind
+
+from .evaluate import eval_expr, eval_filter, project  # noqa: F401
+
+
+# ----------------------------------------------------------------------
+# gather / take
+# ----------------------------------------------------------------------
+
+def gather(col: Column, idx: torch.Tensor) -> Column:
+    n = int(idx.shape[0])
+    if col.dtype.kind == TypeKind.STRING:
+        if col.is_cuda:
+            from . import gpu
+
+            return gpu.gather_string(col, idx)
+        return _gather_string_torch(col, idx)
+    data = col.data[idx]
+    mask = col.mask[idx] if col.mask is not None else None
+    return Column(col.dtype, data, mask, dictionary=col.dictionary, length=n)
+
+
+def _gather_string_torch(col: Column, idx: torch.Tensor) -> Column:
+    offsets = col.offsets
+    idx = idx.long()
+    starts = offsets[idx]
+    lens = offsets[idx + 1] - starts
+    total = int(lens.sum().item())
+    new_off = torch.zeros(len(idx) + 1, dtype=torch.int64, device=idx.device)
+    torch.cumsum(lens, 0, out=new_off[1:])
+    if total > 0:
+        rows = torch.repeat_interleave(
+            torch.arange(len(idx), device=idx.device), lens)
+        within = torch.arange(total, device=idx.device) - new_off[rows]
+        src_pos = starts[rows] + within
+        data = col.data[src_pos]
+    else:
+        data = torch.zeros(0, dtype=torch.uint8, device=idx.device)
+    mask = col.mask[idx] if col.mask is not None else None
+    return Column(bt.string, data, mask, offsets=new_off, length=len(idx))
+
+
+def take_table(tbl: Table, idx: torch.Tensor) -> Table:
+    return Table(tbl.names, [gather(c, idx) for c in tbl.columns], int(idx.shape[0]))
+
+
+def slice_table(tbl: Table, start: int, stop: int) -> Table:
+    n = len(tbl)
+    start = max(0, min(start, n))
+    stop = max(start, min(stop, n))
+    dev = tbl.device
+    idx = torch.arange(start, stop, dtype=torch.int64, device=dev)
+    return take_table(tbl, idx)
+
+
+# ----------------------------------------------------------------------
+# concat
+# ----------------------------------------------------------------------
+
+def concat_columns(cols: Sequence[Column]) -> Column:
+    cols = [c for c in cols]
+    first = cols[0]
+    if len(cols) == 1:
+        return first
+    n = sum(len(c) for c in cols)
+    masks = None
+    if any(c.mask is not None for c in cols):
+        masks = torch.cat([
+            c.mask if c.mask is not None
+            else torch.ones(len(c), dtype=torch.bool, device=first.device)
+            for c in cols])
+    if first.dtype.kind == TypeKind.STRING:
+        datas, offs, base = [], [torch.zeros(1, dtype=torch.int64, device=first.device)], 0
+        for c in cols:
+            datas.append(c.data)
+            offs.append(c.offsets[1:] + base)
+            base += int(c.offsets[-1].item())
+        return Column(bt.string, torch.cat(datas), masks,
+                      offsets=torch.cat(offs), length=n)
+    if first.dtype.kind == TypeKind.DICT:
+        # unify dictionaries
+        import pyarrow as pa
+
+        dicts = [c.dictionary for c in cols]
+        if all(d.equals(dicts[0]) for d in dicts[1:]):
+            return Column(bt.dictionary, torch.cat([c.data for c in cols]), masks,
+                          dictionary=dicts[0], length=n)
+        merged = pa.concat_arrays([d.cast(pa.large_string()) for d in dicts])
+        uniq = merged.unique()
+        lut = {v: i for i, v in enumerate(uniq.to_pylist())}
+        remapped = []
+        for c in cols:
+            rm = np.array([lut[v] for v in c.dictionary.to_pylist()], dtype=np.int32)
+            rm_t = torch.from_numpy(rm).to(c.device)
+            remapped.append(rm_t[c.data.long()])
+        return Column(bt.dictionary, torch.cat(remapped).to(torch.int32), masks,
+                      dictionary=uniq, length=n)
+    datas = [c.data for c in cols]
+    if len({d.dtype for d in datas}) > 1:
+        t = datas[0].dtype
+        for d in datas[1:]:
+            t = torch.promote_types(t, d.dtype)
+        datas = [d.to(t) for d in datas]
+    return Column(first.dtype, torch.cat(datas), masks,
+                  dictionary=first.dictionary, length=n)
+
+
+def concat_tables(tables: Sequence[Table]) -> Table:
+    tables = [t for t in tables if t is not None]
+    first = tables[0]
+    if len(tables) == 1:
+        return first
+    cols = []
+    for i, name in enumerate(first.names):
+        cols.append(concat_columns([t.columns[t.names.index(name)] for t in tables]))
+    return Table(first.names, cols, sum(len(t) for t in tables))
+
+
+# ----------------------------------------------------------------------
+# hashing / partitioning
+# ----------------------------------------------------------------------
+
+def hash_columns(cols: Sequence[Column], seed: int = 0) -> torch.Tensor:
+    """64-bit row hash over multiple key columns (int64 tensor, bit-identical
+    across CPU and GPU so shuffles agree).  Reference role:
+    bodo/libs/_array_hash.cpp."""
+    if cols and cols[0].is_cuda:
+        from . import gpu
+
+        return gpu.hash_columns(cols, seed)
+    return _hash_columns_cpu(cols, seed)
+
+
+_M1 = np.uint64(0xff51afd7ed558ccd)
+_M2 = np.uint64(0xc4ceb9fe1a85ec53)
+
+
+def _mix64_np(x: np.ndarray) -> np.ndarray:
+    # splitmix64-style finalizer; must match csrc/common.h mix64()
+    with np.errstate(over="ignore"):
+        x = x.astype(np.uint64, copy=True)
+        x ^= x >> np.uint64(33)
+        x *= _M1
+        x ^= x >> np.uint64(33)
+        x *= _M2
+        x ^= x >> np.uint64(33)
+    return x
+
+
+def _col_hash_cpu(c: Column, seed: int) -> np.ndarray:
+    k = c.dtype.kind
+    if k == TypeKind.STRING:
+        # hash the bytes per row with FNV-1a (matches csrc string hash)
+        off = c.offsets.numpy()
+        data = c.data.numpy()
+        out = np.empty(len(c), dtype=np.uint64)
+        _fnv_rows(data, off, out)
+        h = out
+    elif k == TypeKind.DICT:
+        # hash the dictionary VALUES (not codes) so hashes agree across ranks
+        # with different dictionaries
+        dvals = c.dictionary.to_pylist()
+        lut = np.array([_fnv_bytes(v.encode() if v is not None else b"")
+                        for v in dvals], dtype=np.uint64)
+        h = lut[c.data.numpy()]
+    elif k == TypeKind.BOOL:
+        h = _mix64_np(c.data.numpy().astype(np.uint64))
+    elif c.dtype.is_float:
+        f = c.data.numpy()
+        if f.dtype == np.float32:
+            f = f.astype(np.float64)
+        bits = f.view(np.uint64).copy()
+        bits[f == 0.0] = 0  # -0.0 == 0.0
+        bits[np.isnan(f)] = np.uint64(0x7FF8000000000000)
+        h = _mix64_np(bits)
+    else:
+        h = _mix64_np(c.data.numpy().astype(np.int64).view(np.uint64))
+    if c.mask is not None:
+        h = h.copy()
+        h[~c.mask.numpy()] = np.uint64(0x9E3779B97F4A7C15)
+    if seed:
+        h = _mix64_np(h ^ np.uint64(seed))
+    return h
+
+
+def _fnv_bytes(b: bytes) -> int:
+    h = 0xcbf29ce484222325
+    for ch in b:
+        h = ((h ^ ch) * 0x100000001b3) & 0xFFFFFFFFFFFFFFFF
+    return h
+
+
+def _fnv_rows(data: np.ndarray, off: np.ndarray, out: np.ndarray):
+    FNV = np.uint64(0x100000001b3)
+    with np.errstate(over="ignore"):
+        for i in range(len(out)):
+            h = np.uint64(0xcbf29ce484222325)
+            for j in range(off[i], off[i + 1]):
+                h = (h ^ np.uint64(data[j])) * FNV
+            out[i] = h
+
+
+def _hash_columns_cpu(cols, seed):
+    acc = None
+    with np.errstate(over="ignore"):
+        for c in cols:
+            h = _col_hash_cpu(c, seed)
+            if acc is None:
+                acc = h.copy()
+            else:
+                # boost-style hash combine, must match csrc combine
+                acc = _mix64_np(acc * np.uint64(0x9E3779B97F4A7C15) + h)
+    if acc is None:
+        acc = np.zeros(0, dtype=np.uint64)
+    return torch.from_numpy(acc.view(np.int64).copy())
+
+
+def partition_indices(hashes: torch.Tensor, nparts: int) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Stable partition permutation: returns (perm, counts) such that rows
+    perm[counts[:p].sum():counts[:p+1].sum()] belong to partition p."""
+    p = torch.remainder(hashes, nparts)
+    p = torch.where(p < 0, p + nparts, p)
+    perm = torch.argsort(p, stable=True)
+    counts = torch.bincount(p, minlength=nparts)
+    return perm, counts
+
+
+# ----------------------------------------------------------------------
+# sort
+# ----------------------------------------------------------------------
+
+def sort_indices(cols: Sequence[Column], ascending: Sequence[bool],
+                 na_position: str = "last") -> torch.Tensor:
+    """Stable lexicographic argsort (last key first, stable iterations)."""
+    device = cols[0].device if cols else torch.device("cpu")
+    n = len(cols[0]) if cols else 0
+    idx = torch.arange(n, dtype=torch.int64, device=device)
+    for c, asc in reversed(list(zip(cols, ascending))):
+        keys = _sort_key_tensor(c, asc, na_position)
+        k = keys[idx]
+        order = torch.argsort(k, stable=True)
+        idx = idx[order]
+    return idx
+
+
+def _sort_key_tensor(c: Column, asc: bool, na_position: str) -> torch.Tensor:
+    k = c.dtype.kind
+    if k == TypeKind.DICT:
+        # dense rank: order codes by dictionary value order (equal values
+        # MUST get equal ranks for multi-key lexsort stability)
+        import pyarrow.compute as pc
+
+        order = pc.array_sort_indices(c.dictionary).to_numpy()
+        rank = np.empty(len(order), dtype=np.int64)
+        rank[order] = np.arange(len(order))
+        data = torch.from_numpy(rank).to(c.device)[c.data.long()]
+    elif k == TypeKind.STRING:
+        import pyarrow as pa
+        import pyarrow.compute as pc
+
+        arr = c.to_device("cpu").to_arrow()
+        denc = pc.dictionary_encode(arr)
+        order = pc.array_sort_indices(denc.dictionary).to_numpy()
+        rank = np.empty(len(order), dtype=np.int64)
+        rank[order] = np.arange(len(order))
+        codes = denc.indices.to_numpy(zero_copy_only=False)
+        codes = np.where(np.isnan(codes.astype(np.float64)), 0, codes).astype(np.int64) \
+            if codes.dtype == object else codes.astype(np.int64)
+        data = torch.from_numpy(rank[codes]).to(c.device)
+        if c.mask is None and arr.null_count:
+            nullmask = torch.from_numpy(
+                pc.is_valid(arr).to_numpy(zero_copy_only=False).astype(bool)
+            ).to(c.device)
+            sent = np.iinfo(np.int64).max if na_position == "last" else np.iinfo(np.int64).min
+            data = torch.where(nullmask, data, torch.full_like(data, sent))
+    elif c.dtype.is_float:
+        data = c.data.clone()
+        nan = torch.isnan(data)
+        big = torch.finfo(data.dtype).max if na_position == "last" else torch.finfo(data.dtype).min
+        if not asc:
+            big = -big
+        data = torch.where(nan, torch.full_like(data, big), data)
+    elif c.data.dtype == torch.bool:
+        data = c.data.to(torch.int8)
+    else:
+        data = c.data
+    if not asc:
+        if data.dtype == torch.bool:
+            data = data.to(torch.int8)
+        data = -data.to(torch.float64) if data.dtype.is_floating_point else -data
+    if c.mask is not None:
+        data = data.clone()
+        if data.dtype.is_floating_point:
+            sentinel = torch.finfo(data.dtype).max if na_position == "last" else torch.finfo(data.dtype).min
+        else:
+            sentinel = torch.iinfo(data.dtype).max if na_position == "last" else torch.iinfo(data.dtype).min
+        data[~c.mask] = sentinel
+    return data
+
+
+# ----------------------------------------------------------------------
+# reductions
+# ----------------------------------------------------------------------
+
+def reduce_column(col: Column, func: str):
+    """Local partial reduction -> dict of partials (combined across ranks by
+    the executor)."""
+    data, mask = col.data, col.mask
+    if col.dtype.is_float:
+        valid = ~torch.isnan(data)
+        if mask is not None:
+            valid &= mask
+    elif mask is not None:
+        valid = mask
+    else:
+        valid = None
+    if valid is not None:
+        data = data[valid]
+    n = int(data.numel())
+    if func in ("sum", "mean", "var", "std"):
+        acc = data.to(torch.float64) if not col.dtype.is_integer else data.to(torch.int64)
+        s = acc.sum().item() if n else (0.0 if not col.dtype.is_integer else 0)
+        out = {"sum": s, "count": n}
+        if func in ("var", "std"):
+            out["sumsq"] = float((acc.to(torch.float64) ** 2).sum().item()) if n else 0.0
+        return out
+    if func in ("count",):
+        return {"count": n}
+    if func == "size":
+        return {"count": len(col)}
+    if func == "min":
+        return {"min": data.min().item() if n else None}
+    if func == "max":
+        return {"max": data.max().item() if n else None}
+    if func == "any":
+        return {"any": bool(data.any().item()) if n else False}
+    if func == "all":
+        return {"all": bool(data.all().item()) if n else True}
+    raise NotImplementedError(f"reduce {func}")

@@ -1,0 +1,447 @@
+"""Expression evaluation over a Table batch.
+
+CPU backend: numpy (null-mask aware).  GPU backend: torch ops on HBM tensors
+with hand-written HIP kernels for datetime extraction and fused paths
+(csrc/).  Reference role: bodo/pandas/physical/expression.h (Arrow-compute
+expression eval) — redesigned to evaluate on device columns directly.
+"""
+
+from __future__ import annotations
+
+import datetime as _dt
+from typing import Dict, Optional
+
+import numpy as np
+import pandas as pd
+import torch
+
+from ..core import types as bt
+from ..core.column import Column
+from ..core.table import Table
+from ..core.types import DType, TypeKind
+from ..plan.expr import (
+    BinOp, BoolOp, Case, Cast, ColRef, Cmp, Const, DtField, Expr, IsIn,
+    IsNull, Not, StrOp, UdfMap,
+)
+
+
+def eval_expr(e: Expr, tbl: Table) -> Column:
+    col = _Evaluator(tbl).visit(e)
+    return col
+
+
+def eval_filter(cond: Expr, tbl: Table) -> Table:
+    from . import take_table
+
+    mask_col = eval_expr(cond, tbl)
+    mask = mask_col.data
+    if mask_col.mask is not None:
+        mask = mask & mask_col.mask
+    idx = torch.nonzero(mask, as_tuple=False).reshape(-1)
+    return take_table(tbl, idx)
+
+
+def project(tbl: Table, names, exprs) -> Table:
+    cols = [eval_expr(e, tbl) for e in exprs]
+    return Table(list(names), cols, len(tbl))
+
+
+class _Evaluator:
+    def __init__(self, tbl: Table):
+        self.tbl = tbl
+        self.n = len(tbl)
+        self.device = tbl.device
+
+    # ------------------------------------------------------------------
+    def visit(self, e: Expr) -> Column:
+        meth = getattr(self, f"visit_{type(e).__name__}")
+        return meth(e)
+
+    def visit_ColRef(self, e: ColRef) -> Column:
+        return self.tbl.column(e.name)
+
+    def visit_Const(self, e: Const) -> Column:
+        v, dtype = e.value, e.dtype
+        if dtype is None:
+            dtype = infer_const_dtype(v)
+        v = normalize_const(v, dtype)
+        return Column.full_const(v, dtype, self.n, self.device)
+
+    # ------------------------------------------------------------------
+    def _decode_if_dict(self, c: Column) -> Column:
+        return c
+
+    def visit_BinOp(self, e: BinOp) -> Column:
+        a, b = self.visit(e.left), self.visit(e.right)
+        return binary_arith(e.op, a, b)
+
+    def visit_Cmp(self, e: Cmp) -> Column:
+        a, b = self.visit(e.left), self.visit(e.right)
+        return compare(e.op, a, b)
+
+    def visit_BoolOp(self, e: BoolOp) -> Column:
+        a, b = self.visit(e.left), self.visit(e.right)
+        da, db = a.data, b.data
+        out = (da & db) if e.op == "and" else (da | db)
+        return Column(bt.boolean, out)
+
+    def visit_Not(self, e: Not) -> Column:
+        a = self.visit(e.operand)
+        return Column(bt.boolean, ~a.data, a.mask)
+
+    def visit_IsNull(self, e: IsNull) -> Column:
+        a = self.visit(e.operand)
+        if a.mask is not None:
+            isnull = ~a.mask
+        elif a.dtype.is_float:
+            isnull = torch.isnan(a.data)
+        else:
+            isnull = torch.zeros(len(a), dtype=torch.bool, device=self.device)
+        if e.negate:
+            isnull = ~isnull
+        return Column(bt.boolean, isnull)
+
+    def visit_IsIn(self, e: IsIn) -> Column:
+        a = self.visit(e.operand)
+        if a.dtype.kind == TypeKind.DICT:
+            dvals = a.dictionary.to_pylist()
+            hit = np.array([v in e.values for v in dvals], dtype=bool)
+            lut = torch.from_numpy(hit).to(self.device)
+            out = lut[a.data.long()]
+            if a.mask is not None:
+                out = out & a.mask
+            return Column(bt.boolean, out)
+        if a.dtype.kind == TypeKind.STRING:
+            # host round-trip for plain strings (rare: benchmark strings are dict)
+            ser = a.to_pandas()
+            res = ser.isin(list(e.values)).to_numpy()
+            return Column(bt.boolean, torch.from_numpy(res).to(self.device))
+        vals = [normalize_const(v, a.dtype) for v in e.values]
+        test = torch.tensor(vals, dtype=a.data.dtype, device=self.device)
+        out = torch.isin(a.data, test)
+        if a.mask is not None:
+            out = out & a.mask
+        return Column(bt.boolean, out)
+
+    def visit_DtField(self, e: DtField) -> Column:
+        a = self.visit(e.operand)
+        return dt_field(a, e.fld)
+
+    def visit_Cast(self, e: Cast) -> Column:
+        a = self.visit(e.operand)
+        return cast_column(a, e.to)
+
+    def visit_StrOp(self, e: StrOp) -> Column:
+        a = self.visit(e.operand)
+        return str_op(a, e.op, e.args)
+
+    def visit_UdfMap(self, e: UdfMap) -> Column:
+        a = self.visit(e.operand)
+        return udf_map(a, e.func, e.na_action)
+
+    def visit_Case(self, e: Case) -> Column:
+        other = self.visit(e.otherwise)
+        out_data = other.data.clone()
+        out_mask = None if other.mask is None else other.mask.clone()
+        # apply in reverse so the first matching cond wins
+        for cond_e, then_e in reversed(list(zip(e.conds, e.thens))):
+            c = self.visit(cond_e).data
+            t = self.visit(then_e)
+            td = t.data.to(out_data.dtype) if t.data.dtype != out_data.dtype else t.data
+            out_data = torch.where(c, td, out_data)
+            if out_mask is not None:
+                tm = t.mask if t.mask is not None else torch.ones_like(out_mask)
+                out_mask = torch.where(c, tm, out_mask)
+        return Column(other.dtype, out_data, out_mask)
+
+
+# ----------------------------------------------------------------------
+# scalar kernels
+# ----------------------------------------------------------------------
+
+def infer_const_dtype(v) -> DType:
+    if isinstance(v, bool):
+        return bt.boolean
+    if isinstance(v, int):
+        return bt.int64
+    if isinstance(v, float):
+        return bt.float64
+    if isinstance(v, (pd.Timestamp, np.datetime64, _dt.datetime)):
+        return bt.timestamp_ns
+    if isinstance(v, _dt.date):
+        return bt.date32
+    if isinstance(v, str):
+        return bt.string
+    raise TypeError(f"cannot infer dtype for constant {v!r}")
+
+
+def normalize_const(v, dtype: DType):
+    if dtype.kind == TypeKind.TIMESTAMP_NS:
+        return int(pd.Timestamp(v).value)
+    if dtype.kind == TypeKind.DATE32:
+        if isinstance(v, (int, np.integer)):
+            return int(v)
+        d = pd.Timestamp(v).date() if not isinstance(v, _dt.date) else v
+        return (d - _dt.date(1970, 1, 1)).days
+    if dtype.kind == TypeKind.BOOL:
+        return bool(v)
+    if dtype.is_integer:
+        return int(v)
+    if dtype.is_float:
+        return float(v)
+    return v
+
+
+_ARITH_RESULT_FLOAT = {"div", "pow"}
+
+
+def binary_arith(op: str, a: Column, b: Column) -> Column:
+    da, db = a.data, b.data
+    if a.dtype.is_float or b.dtype.is_float or op in _ARITH_RESULT_FLOAT:
+        target = torch.float64 if (
+            da.dtype == torch.float64 or db.dtype == torch.float64
+            or not (a.dtype.is_float or b.dtype.is_float)) else torch.float32
+        if da.dtype != target:
+            da = da.to(target)
+        if db.dtype != target:
+            db = db.to(target)
+    elif da.dtype != db.dtype:
+        t = torch.promote_types(da.dtype, db.dtype)
+        da, db = da.to(t), db.to(t)
+    if op == "add":
+        out = da + db
+    elif op == "sub":
+        out = da - db
+    elif op == "mul":
+        out = da * db
+    elif op == "div":
+        out = da / db
+    elif op == "floordiv":
+        out = torch.div(da, db, rounding_mode="floor")
+    elif op == "mod":
+        out = torch.remainder(da, db)
+    elif op == "pow":
+        out = torch.pow(da, db)
+    else:
+        raise NotImplementedError(op)
+    mask = combine_masks(a.mask, b.mask)
+    dtype = bt.from_numpy_dtype(np.dtype(str(out.dtype).replace("torch.", "")))
+    return Column(dtype, out, mask)
+
+
+def compare(op: str, a: Column, b: Column) -> Column:
+    if a.dtype.kind == TypeKind.DICT or b.dtype.kind == TypeKind.DICT:
+        return _compare_dict(op, a, b)
+    if a.dtype.kind == TypeKind.STRING or b.dtype.kind == TypeKind.STRING:
+        return _compare_string_host(op, a, b)
+    da, db = a.data, b.data
+    if da.dtype != db.dtype:
+        if da.dtype == torch.bool:
+            da = da.to(torch.int64)
+        if db.dtype == torch.bool:
+            db = db.to(torch.int64)
+        t = torch.promote_types(da.dtype, db.dtype)
+        da, db = da.to(t), db.to(t)
+    out = getattr(torch, op)(da, db)
+    # pandas semantics: comparisons involving null -> False
+    invalid = combine_masks(a.mask, b.mask)
+    if invalid is not None:
+        out = out & invalid
+    if a.dtype.is_float:
+        out = out & ~torch.isnan(a.data)
+    if b.dtype.is_float:
+        out = out & ~torch.isnan(b.data)
+    return Column(bt.boolean, out)
+
+
+def _compare_dict(op: str, a: Column, b: Column):
+    if isinstance(b, Column) and b.dtype.kind != TypeKind.DICT:
+        a, b = (a, b) if a.dtype.kind == TypeKind.DICT else (b, a)
+    # dict vs const string column
+    if b.dtype.kind == TypeKind.STRING:
+        # b should be a constant column; compare dictionary values on host
+        sval = bytes(b.data[b.offsets[0]:b.offsets[1]].cpu().numpy()).decode() if len(b) else ""
+        dvals = a.dictionary.to_pylist()
+        import operator as _op
+
+        f = {"lt": _op.lt, "le": _op.le, "gt": _op.gt, "ge": _op.ge,
+             "eq": _op.eq, "ne": _op.ne}[op]
+        hit = np.array([v is not None and f(v, sval) for v in dvals], dtype=bool)
+        lut = torch.from_numpy(hit).to(a.device)
+        out = lut[a.data.long()]
+        if a.mask is not None:
+            out = out & a.mask
+        return Column(bt.boolean, out)
+    raise NotImplementedError("dict-dict comparison")
+
+
+def _compare_string_host(op: str, a: Column, b: Column):
+    sa, sb = a.to_pandas(), b.to_pandas()
+    out = getattr(sa, op)(sb).fillna(False).to_numpy(dtype=bool)
+    return Column(bt.boolean, torch.from_numpy(out).to(a.device))
+
+
+def combine_masks(ma: Optional[torch.Tensor], mb: Optional[torch.Tensor]):
+    if ma is None:
+        return mb
+    if mb is None:
+        return ma
+    return ma & mb
+
+
+# ----------------------------------------------------------------------
+# datetime extraction
+# ----------------------------------------------------------------------
+
+_DT_OUT_TYPE = {
+    "year": bt.int16, "month": bt.int8, "day": bt.int8, "hour": bt.int8,
+    "minute": bt.int8, "second": bt.int8, "dayofweek": bt.int8,
+    "weekday": bt.int8, "dayofyear": bt.int16, "quarter": bt.int8,
+    "date": bt.date32, "normalize": bt.timestamp_ns, "floor_day": bt.timestamp_ns,
+}
+
+NS_PER_DAY = 86400 * 10**9
+
+
+def dt_field(a: Column, fld: str) -> Column:
+    if a.is_cuda:
+        from . import gpu
+
+        return gpu.dt_field(a, fld)
+    return _dt_field_cpu(a, fld)
+
+
+def _dt_field_cpu(a: Column, fld: str) -> Column:
+    vals = a.data.numpy()
+    if a.dtype.kind == TypeKind.DATE32:
+        ts = vals.astype("datetime64[D]")
+    else:
+        ts = vals.view("datetime64[ns]")
+    idx = pd.DatetimeIndex(ts)
+    if fld == "date":
+        days = (idx.normalize().asi8 // NS_PER_DAY).astype(np.int32)
+        out = np.ascontiguousarray(days)
+        dtype = bt.date32
+    elif fld in ("normalize", "floor_day"):
+        out = idx.normalize().asi8
+        dtype = bt.timestamp_ns
+    else:
+        attr = "dayofweek" if fld == "weekday" else fld
+        out = getattr(idx, attr).to_numpy()
+        dtype = _DT_OUT_TYPE[fld]
+        out = out.astype(bt.numpy_storage_dtype(dtype))
+    return Column(dtype, torch.from_numpy(out), a.mask)
+
+
+def cast_column(a: Column, to: DType) -> Column:
+    if a.dtype == to:
+        return a
+    if to.kind == TypeKind.STRING or a.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+        # host path for string casts
+        ser = a.to_pandas()
+        if to.kind == TypeKind.STRING:
+            res = ser.astype(str)
+            return Column.from_arrow(__import__("pyarrow").array(res), a.device)
+        npv = ser.to_numpy(dtype=bt.numpy_storage_dtype(to))
+        return Column(to, torch.from_numpy(npv).to(a.device))
+    if to.kind == TypeKind.TIMESTAMP_NS and a.dtype.kind == TypeKind.DATE32:
+        out = a.data.to(torch.int64) * NS_PER_DAY
+        return Column(to, out, a.mask)
+    if to.kind == TypeKind.DATE32 and a.dtype.kind == TypeKind.TIMESTAMP_NS:
+        return dt_field(a, "date")
+    out = a.data.to(bt.torch_storage_dtype(to))
+    return Column(to, out, a.mask)
+
+
+# ----------------------------------------------------------------------
+# strings
+# ----------------------------------------------------------------------
+
+def str_op(a: Column, op: str, args) -> Column:
+    import pyarrow.compute as pc
+
+    if a.dtype.kind == TypeKind.DICT:
+        # operate on the (small) dictionary, keep indices
+        d = a.dictionary
+        if op in ("lower", "upper", "strip", "title", "capitalize"):
+            f = {"lower": pc.utf8_lower, "upper": pc.utf8_upper,
+                 "strip": pc.utf8_trim_whitespace, "title": pc.utf8_title,
+                 "capitalize": pc.utf8_capitalize}[op]
+            return Column(a.dtype, a.data, a.mask, dictionary=f(d), length=len(a))
+        if op in ("contains", "startswith", "endswith", "len"):
+            if op == "len":
+                lut_np = pc.utf8_length(d).to_numpy(zero_copy_only=False).astype(np.int64)
+                ret = bt.int64
+            else:
+                pat = args[0]
+                f = {"contains": lambda x: pc.match_substring(x, pat),
+                     "startswith": lambda x: pc.starts_with(x, pat),
+                     "endswith": lambda x: pc.ends_with(x, pat)}[op]
+                lut_np = f(d).to_numpy(zero_copy_only=False).astype(bool)
+                ret = bt.boolean
+            lut = torch.from_numpy(lut_np).to(a.device)
+            out = lut[a.data.long()]
+            return Column(ret, out, a.mask)
+        raise NotImplementedError(f"str.{op} on dict column")
+    # plain strings: arrow compute on host
+    arr = a.to_arrow() if not a.is_cuda else a.to_device("cpu").to_arrow()
+    fmap = {
+        "lower": lambda x: pc.utf8_lower(x),
+        "upper": lambda x: pc.utf8_upper(x),
+        "strip": lambda x: pc.utf8_trim_whitespace(x),
+        "len": lambda x: pc.utf8_length(x),
+        "contains": lambda x: pc.match_substring(x, args[0]),
+        "startswith": lambda x: pc.starts_with(x, args[0]),
+        "endswith": lambda x: pc.ends_with(x, args[0]),
+        "title": lambda x: pc.utf8_title(x),
+        "capitalize": lambda x: pc.utf8_capitalize(x),
+        "slice": lambda x: pc.utf8_slice_codeunits(x, *args),
+    }
+    res = fmap[op](arr)
+    if res.type in (__import__("pyarrow").int32(), __import__("pyarrow").int64()):
+        res = res.cast(__import__("pyarrow").int64())
+    return Column.from_arrow(res, a.device)
+
+
+# ----------------------------------------------------------------------
+# UDF map: low-cardinality dictionary evaluation
+# ----------------------------------------------------------------------
+
+def udf_map(a: Column, func, na_action=None) -> Column:
+    if a.dtype.kind == TypeKind.DICT:
+        d = a.dictionary.to_pylist()
+        vals = [None if (v is None and na_action == "ignore") else func(v) for v in d]
+        import pyarrow as pa
+
+        if all(v is None or isinstance(v, str) for v in vals):
+            return Column(a.dtype, a.data, a.mask,
+                          dictionary=pa.array(vals, type=pa.large_string()), length=len(a))
+        lut = torch.tensor([np.nan if v is None else v for v in vals],
+                           dtype=torch.float64, device=a.device)
+        return Column(bt.float64, lut[a.data.long()], a.mask)
+    if a.dtype.is_integer or a.dtype.kind == TypeKind.BOOL:
+        # evaluate over unique values (bounded domain assumption checked)
+        uniq = torch.unique(a.data)
+        if uniq.numel() <= 1_000_000:
+            uvals = uniq.cpu().numpy()
+            res = [func(v.item() if hasattr(v, "item") else v) for v in uvals]
+            import pyarrow as pa
+
+            if all(isinstance(v, str) for v in res):
+                # produce dictionary-encoded output
+                udict, inv = np.unique(np.array(res, dtype=object), return_inverse=True)
+                inv_t = torch.from_numpy(inv.astype(np.int32)).to(a.device)
+                # index of each row's value in uniq
+                pos = torch.searchsorted(uniq, a.data)
+                codes = inv_t[pos.long()]
+                return Column(bt.dictionary, codes.to(torch.int32), a.mask,
+                              dictionary=pa.array(list(udict), type=pa.large_string()),
+                              length=len(a))
+            lut = torch.tensor([float(v) for v in res], dtype=torch.float64,
+                               device=a.device)
+            pos = torch.searchsorted(uniq, a.data)
+            return Column(bt.float64, lut[pos.long()], a.mask)
+    # general fallback: host round-trip
+    ser = a.to_pandas()
+    res = ser.map(func, na_action=na_action)
+    return Column.from_arrow(__import__("pyarrow").Array.from_pandas(res), a.device)

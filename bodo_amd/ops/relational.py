@@ -1,0 +1,203 @@
+"""Local (single-rank) relational kernels: hash groupby, hash join, distinct.
+
+CPU: pandas-backed (defines the semantics; the engine's test contract is
+"matches pandas").  GPU: hand-written gfx950 HIP kernels via csrc/ — see
+bodo_amd/ops/gpu.py.  Reference roles: bodo/libs/streaming/_groupby.cpp,
+_join.cpp, groupby/ kernels.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+import numpy as np
+import pandas as pd
+import torch
+
+from ..core import types as bt
+from ..core.column import Column
+from ..core.table import Table
+from ..core.types import TypeKind
+
+# physical agg funcs every backend implements
+PHYSICAL_AGGS = {
+    "sum", "count", "min", "max", "size", "first", "last", "prod",
+    # single-phase only (require co-located full groups):
+    "median", "nunique",
+}
+
+
+def groupby_local(tbl: Table, keys: Sequence[str],
+                  aggs: Sequence[Tuple[str, str, str]],
+                  dropna: bool = True) -> Table:
+    """Group rows of the local shard; aggs = (out_name, in_name, func)."""
+    if tbl.device.type == "cuda":
+        from . import gpu
+
+        return gpu.groupby_local(tbl, keys, aggs, dropna)
+    return _groupby_pandas(tbl, keys, aggs, dropna)
+
+
+def _groupby_pandas(tbl: Table, keys, aggs, dropna) -> Table:
+    need = list(dict.fromkeys(list(keys) + [a[1] for a in aggs if a[1]]))
+    df = tbl.select([c for c in need if tbl.has_column(c)]).to_pandas()
+    if len(df) == 0:
+        # empty shard: produce empty frame with right columns
+        out = pd.DataFrame({k: df[k] if k in df else pd.Series([], dtype="float64")
+                            for k in keys})
+        for out_name, in_name, func in aggs:
+            dt = "int64" if func in ("count", "size", "nunique") else "float64"
+            out[out_name] = pd.Series([], dtype=dt)
+        return Table.from_pandas(out, tbl.device)
+    gb = df.groupby(list(keys), dropna=dropna, sort=False, observed=True)
+    named = {}
+    for out_name, in_name, func in aggs:
+        if func == "size":
+            named[out_name] = pd.NamedAgg(column=df.columns[0] if not in_name or in_name not in df.columns else in_name, aggfunc="size")
+        else:
+            named[out_name] = pd.NamedAgg(column=in_name, aggfunc=func)
+    res = gb.agg(**named).reset_index()
+    return Table.from_pandas(res, tbl.device)
+
+
+# combiner for two-phase aggregation: how to merge partial results
+COMBINE_FUNC = {
+    "sum": "sum", "count": "sum", "size": "sum", "min": "min", "max": "max",
+    "first": "first", "last": "last", "prod": "prod",
+}
+
+
+def join_local(left: Table, right: Table, left_on: Sequence[str],
+               right_on: Sequence[str], how: str,
+               suffixes=("_x", "_y")) -> Table:
+    if left.device.type == "cuda":
+        from . import gpu
+
+        return gpu.join_local(left, right, left_on, right_on, how, suffixes)
+    return _join_pandas(left, right, left_on, right_on, how, suffixes)
+
+
+def _join_pandas(left: Table, right: Table, left_on, right_on, how, suffixes) -> Table:
+    from . import take_table
+
+    if how == "cross":
+        nl, nr = len(left), len(right)
+        li = torch.arange(nl, dtype=torch.int64).repeat_interleave(nr)
+        ri = torch.arange(nr, dtype=torch.int64).repeat(nl)
+        lt, rt = take_table(left, li), take_table(right, ri)
+        return _merge_joined(lt, rt, [], [], suffixes, how)
+    ldf = left.select(list(left_on)).to_pandas()
+    rdf = right.select(list(right_on)).to_pandas()
+    ldf.columns = [f"k{i}" for i in range(len(left_on))]
+    rdf.columns = [f"k{i}" for i in range(len(right_on))]
+    ldf["__li"] = np.arange(len(ldf), dtype=np.int64)
+    rdf["__ri"] = np.arange(len(rdf), dtype=np.int64)
+    pd_how = {"semi": "inner", "anti": "left"}.get(how, how)
+    m = ldf.merge(rdf, on=[f"k{i}" for i in range(len(left_on))], how=pd_how)
+    if how == "semi":
+        li = np.unique(m["__li"].to_numpy())
+        return take_table(left, torch.from_numpy(li).to(left.device))
+    if how == "anti":
+        miss = m[m["__ri"].isna()]["__li"].to_numpy().astype(np.int64)
+        return take_table(left, torch.from_numpy(np.unique(miss)).to(left.device))
+    li = m["__li"].to_numpy()
+    ri = m["__ri"].to_numpy()
+    return _materialize_join(left, right, li, ri, left_on, right_on, how, suffixes)
+
+
+def _materialize_join(left: Table, right: Table, li: np.ndarray, ri: np.ndarray,
+                      left_on, right_on, how, suffixes) -> Table:
+    from . import gather, take_table
+
+    dev = left.device
+    l_valid = ~pd.isna(li)
+    r_valid = ~pd.isna(ri)
+    li_t = torch.from_numpy(np.where(l_valid, li, 0).astype(np.int64)).to(dev)
+    ri_t = torch.from_numpy(np.where(r_valid, ri, 0).astype(np.int64)).to(dev)
+    l_valid_t = torch.from_numpy(l_valid.astype(bool)).to(dev)
+    r_valid_t = torch.from_numpy(r_valid.astype(bool)).to(dev)
+    lt = take_table(left, li_t)
+    rt = take_table(right, ri_t)
+    if not bool(l_valid.all()):
+        lt = _null_out(lt, l_valid_t)
+    if not bool(r_valid.all()):
+        rt = _null_out(rt, r_valid_t)
+    out = _merge_joined(lt, rt, left_on, right_on, suffixes, how)
+    # outer join: coalesce key columns
+    if how in ("outer", "right", "left"):
+        pass
+    return out
+
+
+def _null_out(tbl: Table, valid: torch.Tensor) -> Table:
+    cols = []
+    for c in tbl.columns:
+        mask = valid.clone() if c.mask is None else (c.mask & valid)
+        if c.dtype.is_float:
+            data = c.data.clone()
+            data[~valid] = float("nan")
+            cols.append(Column(c.dtype, data, None, c.offsets, c.dictionary, len(c)))
+        else:
+            cols.append(Column(c.dtype, c.data, mask, c.offsets, c.dictionary, len(c)))
+    return Table(tbl.names, cols, len(tbl))
+
+
+def _merge_joined(lt: Table, rt: Table, left_on, right_on, suffixes, how) -> Table:
+    """Column naming like pandas merge: shared key names appear once (from
+    left, coalesced for outer); clashing non-key names get suffixes."""
+    names, cols = [], []
+    shared_keys = [k for k, rk in zip(left_on, right_on) if k == rk]
+    lnames = set(lt.names)
+    rnames = set(rt.names)
+    for n, c in zip(lt.names, lt.columns):
+        if n in rnames and n not in shared_keys:
+            names.append(n + suffixes[0])
+        else:
+            names.append(n)
+        cols.append(c)
+    for n, c in zip(rt.names, rt.columns):
+        if n in shared_keys:
+            if how in ("outer", "right"):
+                # coalesce into the left key column
+                i = names.index(n)
+                cols[i] = _coalesce(cols[i], c)
+            continue
+        if n in lnames:
+            names.append(n + suffixes[1])
+        else:
+            names.append(n)
+        cols.append(c)
+    return Table(names, cols, len(lt))
+
+
+def _coalesce(a: Column, b: Column) -> Column:
+    if a.mask is None and not a.dtype.is_float:
+        return a
+    if a.dtype.is_float:
+        use_b = torch.isnan(a.data)
+        data = torch.where(use_b, b.data.to(a.data.dtype), a.data)
+        return Column(a.dtype, data, None, length=len(a))
+    use_b = ~a.mask
+    if a.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+        # host path
+        sa, sb = a.to_pandas(), b.to_pandas()
+        out = sa.where(~pd.isna(sa), sb)
+        import pyarrow as pa
+
+        return Column.from_arrow(pa.Array.from_pandas(out), a.device)
+    data = torch.where(use_b, b.data.to(a.data.dtype), a.data)
+    mask = None
+    if b.mask is not None:
+        mask = a.mask | b.mask
+    return Column(a.dtype, data, mask, length=len(a))
+
+
+def distinct_local(tbl: Table, subset: Optional[Sequence[str]] = None,
+                   keep: str = "first") -> Table:
+    if tbl.device.type == "cuda":
+        from . import gpu
+
+        return gpu.distinct_local(tbl, subset, keep)
+    df = tbl.to_pandas()
+    res = df.drop_duplicates(subset=list(subset) if subset else None, keep=keep)
+    return Table.from_pandas(res.reset_index(drop=True), tbl.device)

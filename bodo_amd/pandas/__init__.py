@@ -1,0 +1,80 @@
+"""``bodo_amd.pandas``: lazy drop-in pandas replacement (reference:
+bodo/pandas/__init__.py + base.py).  Unimplemented attributes fall back to
+real pandas with a warning (BODO_AMD_PANDAS_FALLBACK)."""
+
+from __future__ import annotations
+
+import warnings
+
+import pandas as _pd
+
+from .. import config
+from ..engine import executor as _ex
+from ..io import csv as _csv
+from ..io import parquet as _pq
+from ..plan import nodes as _pn
+from .frame import BodoDataFrame, from_pandas_df
+from .series import BodoSeries
+
+# pandas passthroughs commonly used together with the frame API
+Timestamp = _pd.Timestamp
+Timedelta = _pd.Timedelta
+NamedAgg = _pd.NamedAgg
+NA = _pd.NA
+NaT = _pd.NaT
+isna = _pd.isna
+notna = _pd.notna
+DataFrame = BodoDataFrame
+Series = BodoSeries
+
+
+def read_parquet(path, columns=None, **kwargs) -> BodoDataFrame:
+    names = _pq.schema_names(path)
+    plan = _pn.ParquetScan(str(path), tuple(columns) if columns else None,
+                           (), tuple(names))
+    return BodoDataFrame(plan, list(columns) if columns else list(names))
+
+
+def read_csv(path, **kwargs) -> BodoDataFrame:
+    options = tuple(sorted(kwargs.items(), key=lambda kv: kv[0]))
+    names = _csv.schema_names(str(path), dict(options))
+    plan = _pn.CsvScan(str(path), options, None, tuple(names))
+    return BodoDataFrame(plan, list(names))
+
+
+def from_pandas(df: _pd.DataFrame) -> BodoDataFrame:
+    return from_pandas_df(df)
+
+
+def merge(left, right, **kwargs) -> BodoDataFrame:
+    if not isinstance(left, BodoDataFrame):
+        left = from_pandas_df(left)
+    return left.merge(right, **kwargs)
+
+
+def concat(objs, axis=0, ignore_index=False, **kwargs):
+    objs = list(objs)
+    if axis in (0, "index") and all(isinstance(o, BodoDataFrame) for o in objs):
+        plans = tuple(o._lazy_plan for o in objs)
+        return BodoDataFrame(_pn.Union(plans), objs[0]._columns)
+    warnings.warn("concat: falling back to pandas")
+    mats = [o.to_pandas() if isinstance(o, (BodoDataFrame, BodoSeries)) else o
+            for o in objs]
+    return from_pandas_df(_pd.concat(mats, axis=axis, ignore_index=ignore_index,
+                                     **kwargs))
+
+
+def to_datetime(arg, **kwargs):
+    if isinstance(arg, BodoSeries):
+        from ..core import types as bt
+        from ..plan.expr import Cast
+
+        return arg._wrap(Cast(arg._expr, bt.timestamp_ns), arg.name)
+    return _pd.to_datetime(arg, **kwargs)
+
+
+def __getattr__(name):
+    if hasattr(_pd, name):
+        if config.PANDAS_FALLBACK:
+            return getattr(_pd, name)
+    raise AttributeError(f"module 'bodo_amd.pandas' has no attribute {name!r}")

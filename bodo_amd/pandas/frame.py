@@ -1,0 +1,383 @@
+"""BodoDataFrame: lazy drop-in pandas DataFrame (reference:
+bodo/pandas/frame.py:117).  Methods build logical plan nodes; materialization
+(executed on every rank's GPU shard, gathered on collect) is triggered by
+len/repr/to_pandas/fallback."""
+
+from __future__ import annotations
+
+import warnings
+from typing import Dict, List, Optional, Sequence
+
+import numpy as np
+import pandas as pd
+
+from .. import config
+from ..core.table import Table
+from ..engine import executor as ex
+from ..parallel import comm
+from ..plan import nodes as pn
+from ..plan.expr import BoolOp, ColRef, Expr, Not
+
+
+class BodoDataFrame:
+    def __init__(self, plan: pn.PlanNode, columns: Sequence[str]):
+        object.__setattr__(self, "_plan", plan)
+        object.__setattr__(self, "_columns", list(columns))
+        object.__setattr__(self, "_result", None)  # cached local shard Table
+
+    # ------------------------------------------------------------------
+    # plan / execution
+    # ------------------------------------------------------------------
+    @property
+    def _lazy_plan(self) -> pn.PlanNode:
+        return self._plan
+
+    def execute(self) -> Table:
+        """Execute the plan; returns this rank's shard and caches it."""
+        if self._result is None:
+            tbl = ex.execute(self._plan, ex.ExecutionContext())
+            object.__setattr__(self, "_result", tbl)
+            # re-root the plan on the materialized shard (ExecState.DISTRIBUTED)
+            key = ex.register_object(tbl)
+            object.__setattr__(
+                self, "_plan",
+                pn.PandasScan(key, tuple(tbl.names), distributed=True))
+            object.__setattr__(self, "_columns", list(tbl.names))
+        return self._result
+
+    def to_pandas(self) -> pd.DataFrame:
+        shard = self.execute()
+        full = comm.allgather_table(shard)
+        df = full.to_pandas()
+        return df
+
+    # aliases used by tests / fallback
+    collect = to_pandas
+
+    # ------------------------------------------------------------------
+    # basic introspection
+    # ------------------------------------------------------------------
+    @property
+    def columns(self):
+        return pd.Index(self._columns)
+
+    @property
+    def shape(self):
+        n = len(self)
+        return (n, len(self._columns))
+
+    def __len__(self) -> int:
+        shard = self.execute()
+        return int(sum(comm.allgather_obj(len(shard))))
+
+    @property
+    def empty(self) -> bool:
+        return len(self) == 0
+
+    @property
+    def dtypes(self):
+        return self.head(1).to_pandas().dtypes
+
+    def __repr__(self) -> str:
+        head = self.head(10).to_pandas()
+        return repr(head)
+
+    # ------------------------------------------------------------------
+    # column access
+    # ------------------------------------------------------------------
+    def __getitem__(self, key):
+        from .series import BodoSeries
+
+        if isinstance(key, str):
+            if key not in self._columns:
+                raise KeyError(key)
+            return BodoSeries(self._plan, ColRef(key), key, frame=self)
+        if isinstance(key, list):
+            missing = [k for k in key if k not in self._columns]
+            if missing:
+                raise KeyError(missing)
+            exprs = tuple(ColRef(k) for k in key)
+            return BodoDataFrame(pn.Projection(self._plan, tuple(key), exprs), key)
+        if isinstance(key, BodoSeries):
+            return BodoDataFrame(pn.Filter(self._plan, key._expr), self._columns)
+        if isinstance(key, pd.Series) and key.dtype == bool:
+            # host boolean mask: materialize path
+            return self._fallback("__getitem__", key)
+        raise TypeError(f"unsupported key {type(key)}")
+
+    def __setitem__(self, key: str, value):
+        from .series import BodoSeries
+
+        if isinstance(value, BodoSeries):
+            vplan = value._plan
+            if (isinstance(vplan, pn.MapPartitions)
+                    and vplan.child is self._plan):
+                # df["s"] = df.apply(f, axis=1): fuse into one partition pass
+                inner, expr_v = vplan.func, value._expr
+                col = expr_v.name if isinstance(expr_v, ColRef) else None
+
+                def _fused(pdf, *a, _inner=inner, _col=col, _key=key):
+                    res = _inner(pdf, *a)
+                    ser = res[_col] if _col is not None else res.iloc[:, 0]
+                    out = pdf.copy()
+                    out[_key] = ser.to_numpy()
+                    return out
+
+                names = list(self._columns)
+                if key not in names:
+                    names.append(key)
+                new_plan = pn.MapPartitions(self._plan, _fused, vplan.args,
+                                            tuple(names))
+                object.__setattr__(self, "_plan", new_plan)
+                object.__setattr__(self, "_columns", names)
+                object.__setattr__(self, "_result", None)
+                return
+            if vplan is not self._plan:
+                # allow setting from a series derived from the same frame
+                # lineage after assignments: rebuild on current plan if the
+                # referenced columns still exist
+                from ..plan.expr import expr_columns
+
+                refs = expr_columns(value._expr)
+                if not refs.issubset(set(self._columns)):
+                    raise ValueError(
+                        "cannot assign a Series derived from a different frame")
+            expr = value._expr
+        else:
+            from ..plan.expr import Const
+
+            expr = Const(value)
+        names = list(self._columns)
+        exprs = [ColRef(c) for c in names]
+        if key in names:
+            exprs[names.index(key)] = expr
+        else:
+            names.append(key)
+            exprs.append(expr)
+        new_plan = pn.Projection(self._plan, tuple(names), tuple(exprs))
+        object.__setattr__(self, "_plan", new_plan)
+        object.__setattr__(self, "_columns", names)
+        object.__setattr__(self, "_result", None)
+
+    def assign(self, **kwargs) -> "BodoDataFrame":
+        out = BodoDataFrame(self._plan, self._columns)
+        for k, v in kwargs.items():
+            out[k] = v(out) if callable(v) else v
+        return out
+
+    def rename(self, columns: Optional[Dict[str, str]] = None, copy=None,
+               inplace=False, **kwargs) -> "BodoDataFrame":
+        assert columns is not None
+        names = [columns.get(c, c) for c in self._columns]
+        exprs = tuple(ColRef(c) for c in self._columns)
+        plan = pn.Projection(self._plan, tuple(names), exprs)
+        if inplace:
+            object.__setattr__(self, "_plan", plan)
+            object.__setattr__(self, "_columns", names)
+            object.__setattr__(self, "_result", None)
+            return None
+        return BodoDataFrame(plan, names)
+
+    def drop(self, labels=None, columns=None, axis=0, inplace=False, **kw):
+        if columns is None and axis in (1, "columns"):
+            columns = labels
+        if columns is None:
+            return self._fallback("drop", labels=labels, axis=axis, **kw)
+        if isinstance(columns, str):
+            columns = [columns]
+        keep = [c for c in self._columns if c not in set(columns)]
+        plan = pn.Projection(self._plan, tuple(keep),
+                             tuple(ColRef(c) for c in keep))
+        if inplace:
+            object.__setattr__(self, "_plan", plan)
+            object.__setattr__(self, "_columns", keep)
+            object.__setattr__(self, "_result", None)
+            return None
+        return BodoDataFrame(plan, keep)
+
+    def __getattr__(self, name: str):
+        # column attribute access
+        cols = object.__getattribute__(self, "_columns")
+        if name in cols:
+            return self[name]
+        if name.startswith("_"):
+            raise AttributeError(name)
+        if hasattr(pd.DataFrame, name) and config.PANDAS_FALLBACK:
+            attr = getattr(pd.DataFrame, name)
+            if callable(attr):
+                def method(*args, **kwargs):
+                    return self._fallback(name, *args, **kwargs)
+
+                return method
+            warnings.warn(f"BodoDataFrame.{name}: falling back to pandas "
+                          "(materializes the result)", stacklevel=2)
+            return getattr(self.to_pandas(), name)
+        raise AttributeError(name)
+
+    def _fallback(self, name, *args, **kwargs):
+        warnings.warn(f"BodoDataFrame.{name}: falling back to pandas "
+                      "(materializes the result)", stacklevel=3)
+        pdf = self.to_pandas()
+        res = getattr(pdf, name)(*args, **kwargs)
+        return wrap_result(res)
+
+    # ------------------------------------------------------------------
+    # relational methods
+    # ------------------------------------------------------------------
+    def head(self, n: int = 5) -> "BodoDataFrame":
+        return BodoDataFrame(pn.Limit(self._plan, n), self._columns)
+
+    def tail(self, n: int = 5) -> "BodoDataFrame":
+        return BodoDataFrame(pn.Limit(self._plan, n, tail=True), self._columns)
+
+    def merge(self, right: "BodoDataFrame", how="inner", on=None, left_on=None,
+              right_on=None, suffixes=("_x", "_y"), copy=None, **kwargs):
+        if on is not None:
+            if isinstance(on, str):
+                on = [on]
+            left_on = right_on = list(on)
+        else:
+            if isinstance(left_on, str):
+                left_on = [left_on]
+            if isinstance(right_on, str):
+                right_on = [right_on]
+        if not isinstance(right, BodoDataFrame):
+            right = from_pandas_df(right)
+        plan = pn.Join(self._plan, right._plan, tuple(left_on), tuple(right_on),
+                       how, tuple(suffixes))
+        out_cols = _join_out_columns(self._columns, right._columns,
+                                     left_on, right_on, suffixes, how)
+        return BodoDataFrame(plan, out_cols)
+
+    def groupby(self, by, as_index: bool = True, dropna: bool = True,
+                sort: bool = False, observed=True):
+        from .groupby import DataFrameGroupBy
+
+        if isinstance(by, str):
+            by = [by]
+        return DataFrameGroupBy(self, list(by), as_index=as_index,
+                                dropna=dropna, sort=sort)
+
+    def sort_values(self, by, ascending=True, na_position="last",
+                    ignore_index=False, **kwargs) -> "BodoDataFrame":
+        if isinstance(by, str):
+            by = [by]
+        if isinstance(ascending, bool):
+            ascending = [ascending] * len(by)
+        plan = pn.Sort(self._plan, tuple(by), tuple(ascending), na_position)
+        return BodoDataFrame(plan, self._columns)
+
+    def drop_duplicates(self, subset=None, keep="first", **kwargs):
+        if isinstance(subset, str):
+            subset = [subset]
+        plan = pn.Distinct(self._plan, tuple(subset) if subset else None, keep)
+        return BodoDataFrame(plan, self._columns)
+
+    def sample(self, n=None, frac=None, random_state=None, **kwargs):
+        plan = pn.Sample(self._plan, n, frac, random_state)
+        return BodoDataFrame(plan, self._columns)
+
+    def map_partitions(self, func, *args) -> "BodoDataFrame":
+        plan = pn.MapPartitions(self._plan, func, tuple(args))
+        # run on empty frame to infer schema
+        try:
+            probe = func(pd.DataFrame(columns=self._columns), *args)
+            names = list(probe.columns)
+        except Exception:
+            names = list(self._columns)
+        plan = pn.MapPartitions(self._plan, func, tuple(args), tuple(names))
+        return BodoDataFrame(plan, names)
+
+    def apply(self, func, axis=0, args=(), **kwargs):
+        if axis in (1, "columns"):
+            def _part(pdf, *a):
+                res = pdf.apply(func, axis=1, args=a, **kwargs)
+                if isinstance(res, pd.Series):
+                    res = res.to_frame(name="0")
+                return res
+
+            plan = pn.MapPartitions(self._plan, _part, tuple(args), ("0",))
+            from .series import BodoSeries
+
+            return BodoSeries(plan, ColRef("0"), None)
+        return self._fallback("apply", func, axis=axis, args=args, **kwargs)
+
+    def isna(self):
+        from ..plan.expr import IsNull
+
+        names = list(self._columns)
+        exprs = tuple(IsNull(ColRef(c)) for c in names)
+        return BodoDataFrame(pn.Projection(self._plan, tuple(names), exprs), names)
+
+    def notna(self):
+        from ..plan.expr import IsNull
+
+        names = list(self._columns)
+        exprs = tuple(IsNull(ColRef(c), negate=True) for c in names)
+        return BodoDataFrame(pn.Projection(self._plan, tuple(names), exprs), names)
+
+    def dropna(self, subset=None, how="any", **kwargs):
+        from ..plan.expr import IsNull
+
+        cols = list(subset) if subset is not None else list(self._columns)
+        conds = [IsNull(ColRef(c), negate=True) for c in cols]
+        out = conds[0]
+        for c in conds[1:]:
+            out = BoolOp("and" if how == "any" else "or", out, c)
+        if how == "all":
+            # keep row if any column non-null
+            out = conds[0]
+            for c in conds[1:]:
+                out = BoolOp("or", out, c)
+        return BodoDataFrame(pn.Filter(self._plan, out), self._columns)
+
+    # ------------------------------------------------------------------
+    # IO
+    # ------------------------------------------------------------------
+    def to_parquet(self, path: str, compression="snappy", **kwargs):
+        plan = pn.ParquetWrite(self._plan, path, compression)
+        ex.execute(plan, ex.ExecutionContext())
+
+    def to_csv(self, path=None, **kwargs):
+        pdf = self.to_pandas()
+        return pdf.to_csv(path, index=False, **kwargs)
+
+    # reductions over the whole frame fall back (rare)
+    def count(self):
+        return self.to_pandas().count()
+
+    def copy(self, deep=True):
+        return BodoDataFrame(self._plan, list(self._columns))
+
+
+def _join_out_columns(lcols, rcols, left_on, right_on, suffixes, how):
+    shared_keys = [k for k, rk in zip(left_on, right_on) if k == rk]
+    out = []
+    rset, lset = set(rcols), set(lcols)
+    for c in lcols:
+        if c in rset and c not in shared_keys:
+            out.append(c + suffixes[0])
+        else:
+            out.append(c)
+    for c in rcols:
+        if c in shared_keys:
+            continue
+        if c in lset:
+            out.append(c + suffixes[1])
+        else:
+            out.append(c)
+    return out
+
+
+def wrap_result(res):
+    if isinstance(res, pd.DataFrame):
+        if isinstance(res.index, pd.RangeIndex):
+            return from_pandas_df(res)
+        return res  # keep meaningful indexes (describe(), set_index results)
+    return res
+
+
+def from_pandas_df(df: pd.DataFrame) -> BodoDataFrame:
+    key = ex.register_object(df.reset_index(drop=True))
+    plan = pn.PandasScan(key, tuple(df.columns), distributed=False)
+    return BodoDataFrame(plan, list(df.columns))

@@ -1,0 +1,223 @@
+"""Lazy groupby (reference: bodo/pandas/groupby.py)."""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import pandas as pd
+
+from ..plan import nodes as pn
+
+
+class DataFrameGroupBy:
+    def __init__(self, frame, keys: List[str], as_index=True, dropna=True,
+                 sort=False, selection: Optional[List[str]] = None):
+        self._frame = frame
+        self._keys = keys
+        self._as_index = as_index
+        self._dropna = dropna
+        self._sort = sort
+        self._selection = selection
+
+    def __getitem__(self, key):
+        if isinstance(key, str):
+            return SeriesGroupBy(self._frame, self._keys, key,
+                                 self._as_index, self._dropna, self._sort)
+        return DataFrameGroupBy(self._frame, self._keys, self._as_index,
+                                self._dropna, self._sort, list(key))
+
+    # ------------------------------------------------------------------
+    def _value_columns(self):
+        if self._selection is not None:
+            return self._selection
+        return [c for c in self._frame._columns if c not in self._keys]
+
+    def _build(self, aggs):
+        plan = pn.Aggregate(self._frame._lazy_plan, tuple(self._keys),
+                            tuple(aggs), self._as_index, self._dropna)
+        if self._sort:
+            plan = pn.Sort(plan, tuple(self._keys),
+                           tuple([True] * len(self._keys)))
+        from .frame import BodoDataFrame
+
+        cols = list(self._keys) + [a[0] for a in aggs]
+        out = BodoDataFrame(plan, cols)
+        if self._as_index:
+            return _IndexedAggResult(out, self._keys)
+        return out
+
+    def agg(self, arg=None, **kwargs):
+        aggs = []
+        if arg is None and kwargs:
+            # named aggregation: out=NamedAgg(column=..., aggfunc=...) or tuples
+            for out_name, spec in kwargs.items():
+                if isinstance(spec, pd.NamedAgg) or (isinstance(spec, tuple) and len(spec) == 2):
+                    col, func = spec
+                else:
+                    raise TypeError(f"bad agg spec {spec}")
+                aggs.append((out_name, col, _norm_func(func)))
+        elif isinstance(arg, dict):
+            for col, func in arg.items():
+                if isinstance(func, (list, tuple)):
+                    for f in func:
+                        aggs.append((f"{col}_{f}", col, _norm_func(f)))
+                else:
+                    aggs.append((col, col, _norm_func(func)))
+        elif isinstance(arg, str):
+            for col in self._value_columns():
+                aggs.append((col, col, _norm_func(arg)))
+        else:
+            raise NotImplementedError(f"agg({arg!r})")
+        return self._build(aggs)
+
+    aggregate = agg
+
+    def _simple(self, func):
+        aggs = [(c, c, func) for c in self._value_columns()]
+        return self._build(aggs)
+
+    def sum(self, numeric_only=False):
+        return self._simple("sum")
+
+    def count(self):
+        return self._simple("count")
+
+    def mean(self, numeric_only=False):
+        return self._simple("mean")
+
+    def min(self, numeric_only=False):
+        return self._simple("min")
+
+    def max(self, numeric_only=False):
+        return self._simple("max")
+
+    def median(self, numeric_only=False):
+        return self._simple("median")
+
+    def first(self):
+        return self._simple("first")
+
+    def last(self):
+        return self._simple("last")
+
+    def nunique(self):
+        return self._simple("nunique")
+
+    def var(self, ddof=1):
+        return self._simple("var")
+
+    def std(self, ddof=1):
+        return self._simple("std")
+
+    def size(self):
+        out = self._build([("size", "", "size")])
+        if isinstance(out, _IndexedAggResult):
+            ser = out._frame.to_pandas().set_index(self._keys)["size"]
+            return ser
+        return out
+
+    def apply(self, func, *args, **kwargs):
+        # general groupby-apply: shuffle rows by key then run pandas apply
+        # per shard (keys co-located so results are exact)
+        frame = self._frame
+        keys = self._keys
+
+        def _part(pdf, *a):
+            if len(pdf) == 0:
+                return pd.DataFrame()
+            res = pdf.groupby(keys, dropna=self._dropna).apply(func, *a, **kwargs)
+            return res.reset_index()
+
+        from .frame import BodoDataFrame
+
+        shuffled = pn.ShuffleByKey(frame._lazy_plan, tuple(keys))
+        try:
+            import pandas as _pd
+
+            probe = _part(frame.head(0).to_pandas())
+            names = list(probe.columns)
+        except Exception:
+            names = []
+        plan = pn.MapPartitions(shuffled, _part, (), tuple(names))
+        return BodoDataFrame(plan, names)
+
+
+class SeriesGroupBy:
+    def __init__(self, frame, keys, column, as_index, dropna, sort):
+        self._frame = frame
+        self._keys = keys
+        self._col = column
+        self._as_index = as_index
+        self._dropna = dropna
+        self._sort = sort
+
+    def _agg1(self, func):
+        gb = DataFrameGroupBy(self._frame, self._keys, self._as_index,
+                              self._dropna, self._sort, [self._col])
+        out = gb._build([(self._col, self._col, func)])
+        if isinstance(out, _IndexedAggResult):
+            pdf = out._frame.to_pandas().set_index(self._keys)[self._col]
+            return pdf
+        return out[self._col] if hasattr(out, "__getitem__") else out
+
+    def sum(self):
+        return self._agg1("sum")
+
+    def count(self):
+        return self._agg1("count")
+
+    def mean(self):
+        return self._agg1("mean")
+
+    def min(self):
+        return self._agg1("min")
+
+    def max(self):
+        return self._agg1("max")
+
+    def nunique(self):
+        return self._agg1("nunique")
+
+    def agg(self, func):
+        if isinstance(func, str):
+            return self._agg1(func)
+        raise NotImplementedError
+
+
+class _IndexedAggResult:
+    """as_index=True result: behaves like the BodoDataFrame but materializes
+    with keys as the index (collected path only)."""
+
+    def __init__(self, frame, keys):
+        self._frame = frame
+        self._keys = keys
+
+    def to_pandas(self):
+        return self._frame.to_pandas().set_index(self._keys)
+
+    def reset_index(self):
+        return self._frame
+
+    def sort_values(self, *args, **kwargs):
+        return _IndexedAggResult(self._frame.sort_values(*args, **kwargs),
+                                 self._keys)
+
+    def __getattr__(self, name):
+        return getattr(self._frame, name)
+
+    def __getitem__(self, k):
+        return self._frame[k]
+
+    def __repr__(self):
+        return repr(self.to_pandas().head(10))
+
+
+def _norm_func(f) -> str:
+    if callable(f) and hasattr(f, "__name__"):
+        name = f.__name__
+        if name in ("sum", "mean", "min", "max", "count", "size", "median",
+                    "var", "std", "prod"):
+            return name
+        raise NotImplementedError(f"custom agg func {f}")
+    m = {"average": "mean", "nunique": "nunique"}
+    return m.get(f, f)

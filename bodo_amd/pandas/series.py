@@ -1,0 +1,345 @@
+"""BodoSeries: lazy Series with .dt/.str accessors (reference:
+bodo/pandas/series.py:99).  A series is an expression over a source plan."""
+
+from __future__ import annotations
+
+import warnings
+from typing import Optional
+
+import pandas as pd
+
+from .. import config
+from ..core import types as bt
+from ..engine import executor as ex
+from ..parallel import comm
+from ..plan import nodes as pn
+from ..plan.expr import (
+    BinOp, BoolOp, Case, Cast, Cmp, ColRef, Const, DtField, Expr, IsIn,
+    IsNull, Not, StrOp, UdfMap, as_expr,
+)
+
+
+class BodoSeries:
+    def __init__(self, plan, expr: Expr, name: Optional[str], frame=None):
+        self._plan = plan
+        self._expr = expr
+        self.name = name
+        self._frame = frame
+
+    # ------------------------------------------------------------------
+    def _wrap(self, expr: Expr, name=None) -> "BodoSeries":
+        return BodoSeries(self._plan, expr, name or self.name, self._frame)
+
+    def _bin(self, op, other, reflect=False) -> "BodoSeries":
+        oe = other._expr if isinstance(other, BodoSeries) else as_expr(other)
+        e = BinOp(op, oe, self._expr) if reflect else BinOp(op, self._expr, oe)
+        return self._wrap(e, None)
+
+    def _cmp(self, op, other) -> "BodoSeries":
+        oe = other._expr if isinstance(other, BodoSeries) else as_expr(other)
+        return self._wrap(Cmp(op, self._expr, oe), None)
+
+    __add__ = lambda s, o: s._bin("add", o)
+    __radd__ = lambda s, o: s._bin("add", o, True)
+    __sub__ = lambda s, o: s._bin("sub", o)
+    __rsub__ = lambda s, o: s._bin("sub", o, True)
+    __mul__ = lambda s, o: s._bin("mul", o)
+    __rmul__ = lambda s, o: s._bin("mul", o, True)
+    __truediv__ = lambda s, o: s._bin("div", o)
+    __rtruediv__ = lambda s, o: s._bin("div", o, True)
+    __floordiv__ = lambda s, o: s._bin("floordiv", o)
+    __mod__ = lambda s, o: s._bin("mod", o)
+    __pow__ = lambda s, o: s._bin("pow", o)
+    __lt__ = lambda s, o: s._cmp("lt", o)
+    __le__ = lambda s, o: s._cmp("le", o)
+    __gt__ = lambda s, o: s._cmp("gt", o)
+    __ge__ = lambda s, o: s._cmp("ge", o)
+    __eq__ = lambda s, o: s._cmp("eq", o)
+    __ne__ = lambda s, o: s._cmp("ne", o)
+
+    def __and__(self, o):
+        oe = o._expr if isinstance(o, BodoSeries) else as_expr(o)
+        return self._wrap(BoolOp("and", self._expr, oe), None)
+
+    def __or__(self, o):
+        oe = o._expr if isinstance(o, BodoSeries) else as_expr(o)
+        return self._wrap(BoolOp("or", self._expr, oe), None)
+
+    def __invert__(self):
+        return self._wrap(Not(self._expr), None)
+
+    def __hash__(self):
+        return id(self)
+
+    # ------------------------------------------------------------------
+    def isin(self, values) -> "BodoSeries":
+        return self._wrap(IsIn(self._expr, tuple(values)), None)
+
+    def isna(self) -> "BodoSeries":
+        return self._wrap(IsNull(self._expr), None)
+
+    isnull = isna
+
+    def notna(self) -> "BodoSeries":
+        return self._wrap(IsNull(self._expr, negate=True), None)
+
+    notnull = notna
+
+    def fillna(self, value) -> "BodoSeries":
+        return self._wrap(
+            Case((IsNull(self._expr),), (as_expr(value),), self._expr), None)
+
+    def map(self, func, na_action=None) -> "BodoSeries":
+        if isinstance(func, dict):
+            d = dict(func)
+            return self.map(lambda v: d.get(v), na_action=na_action)
+        return self._wrap(UdfMap(self._expr, func, na_action), None)
+
+    apply = map
+
+    def astype(self, dtype) -> "BodoSeries":
+        target = _pd_dtype_to_bodo(dtype)
+        return self._wrap(Cast(self._expr, target), None)
+
+    def abs(self):
+        zero = Const(0)
+        return self._wrap(
+            Case((Cmp("lt", self._expr, zero),),
+                 (BinOp("sub", zero, self._expr),), self._expr), None)
+
+    def round(self, decimals=0):
+        return self.map(lambda v: None if pd.isna(v) else round(v, decimals))
+
+    def where(self, cond, other) -> "BodoSeries":
+        ce = cond._expr if isinstance(cond, BodoSeries) else as_expr(cond)
+        oe = other._expr if isinstance(other, BodoSeries) else as_expr(other)
+        return self._wrap(Case((ce,), (self._expr,), oe), None)
+
+    @property
+    def dt(self):
+        return _DtAccessor(self)
+
+    @property
+    def str(self):
+        return _StrAccessor(self)
+
+    # ------------------------------------------------------------------
+    # reductions (execute eagerly, return python scalar)
+    # ------------------------------------------------------------------
+    def _reduce(self, func: str):
+        plan = pn.Reduce(self._as_projection_plan(), (("r", "v", func),))
+        tbl = ex.execute(plan, ex.ExecutionContext())
+        ser = tbl.column("r").to_pandas()
+        return ser.iloc[0] if len(ser) else None
+
+    def sum(self):
+        return self._reduce("sum")
+
+    def mean(self):
+        return self._reduce("mean")
+
+    def min(self):
+        return self._reduce("min")
+
+    def max(self):
+        return self._reduce("max")
+
+    def count(self):
+        return self._reduce("count")
+
+    def var(self, ddof=1):
+        return self._reduce("var")
+
+    def std(self, ddof=1):
+        return self._reduce("std")
+
+    def any(self):
+        return self._reduce("any")
+
+    def all(self):
+        return self._reduce("all")
+
+    def nunique(self):
+        plan = pn.Aggregate(
+            pn.Distinct(self._as_projection_plan(), ("v",)), (),
+            (("n", "v", "count"),))
+        tbl = ex.execute(plan, ex.ExecutionContext())
+        # aggregate with no keys isn't supported by groupby; use Reduce
+        plan = pn.Reduce(pn.Distinct(self._as_projection_plan(), ("v",)),
+                         (("n", "v", "count"),))
+        tbl = ex.execute(plan, ex.ExecutionContext())
+        return int(tbl.column("n").to_pandas().iloc[0])
+
+    def unique(self):
+        plan = pn.Distinct(self._as_projection_plan(), ("v",))
+        tbl = ex.execute(plan, ex.ExecutionContext())
+        full = comm.allgather_table(tbl)
+        return full.column("v").to_pandas().to_numpy()
+
+    def value_counts(self, ascending=False, dropna=True):
+        plan = pn.Aggregate(self._as_projection_plan(), ("v",),
+                            (("count", "v", "size"),), dropna=dropna)
+        plan = pn.Sort(plan, ("count",), (ascending,))
+        tbl = ex.execute(plan, ex.ExecutionContext())
+        full = comm.allgather_table(tbl)
+        pdf = full.to_pandas()
+        # ties: pandas orders by value; match roughly
+        out = pd.Series(pdf["count"].to_numpy(), index=pdf["v"].to_numpy(),
+                        name="count")
+        out.index.name = self.name
+        return out
+
+    def _as_projection_plan(self):
+        return pn.Projection(self._plan, ("v",), (self._expr,))
+
+    # ------------------------------------------------------------------
+    def to_pandas(self) -> pd.Series:
+        plan = pn.Projection(self._plan, ("v",), (self._expr,))
+        tbl = ex.execute(plan, ex.ExecutionContext())
+        full = comm.allgather_table(tbl)
+        ser = full.column("v").to_pandas()
+        ser.name = self.name
+        return ser
+
+    def head(self, n=5):
+        plan = pn.Limit(pn.Projection(self._plan, ("v",), (self._expr,)), n)
+        tbl = ex.execute(plan, ex.ExecutionContext())
+        full = comm.allgather_table(tbl)
+        ser = full.column("v").to_pandas()
+        ser.name = self.name
+        return ser
+
+    def __len__(self):
+        return len(self.to_pandas())
+
+    def __repr__(self):
+        return repr(self.head(10))
+
+    def __getattr__(self, name):
+        if name.startswith("_"):
+            raise AttributeError(name)
+        if hasattr(pd.Series, name) and config.PANDAS_FALLBACK:
+            attr = getattr(pd.Series, name)
+            if callable(attr):
+                def method(*args, **kwargs):
+                    warnings.warn(f"BodoSeries.{name}: falling back to pandas",
+                                  stacklevel=2)
+                    return getattr(self.to_pandas(), name)(*args, **kwargs)
+
+                return method
+            warnings.warn(f"BodoSeries.{name}: falling back to pandas",
+                          stacklevel=2)
+            return getattr(self.to_pandas(), name)
+        raise AttributeError(name)
+
+
+class _DtAccessor:
+    def __init__(self, s: BodoSeries):
+        self._s = s
+
+    def _f(self, fld):
+        return self._s._wrap(DtField(self._s._expr, fld), None)
+
+    @property
+    def date(self):
+        return self._f("date")
+
+    @property
+    def year(self):
+        return self._f("year")
+
+    @property
+    def month(self):
+        return self._f("month")
+
+    @property
+    def day(self):
+        return self._f("day")
+
+    @property
+    def hour(self):
+        return self._f("hour")
+
+    @property
+    def minute(self):
+        return self._f("minute")
+
+    @property
+    def second(self):
+        return self._f("second")
+
+    @property
+    def dayofweek(self):
+        return self._f("dayofweek")
+
+    weekday = dayofweek
+
+    @property
+    def dayofyear(self):
+        return self._f("dayofyear")
+
+    @property
+    def quarter(self):
+        return self._f("quarter")
+
+    def normalize(self):
+        return self._f("normalize")
+
+    def floor(self, freq):
+        assert freq in ("D", "d"), "only day floor supported"
+        return self._f("floor_day")
+
+
+class _StrAccessor:
+    def __init__(self, s: BodoSeries):
+        self._s = s
+
+    def _f(self, op, *args):
+        return self._s._wrap(StrOp(self._s._expr, op, tuple(args)), None)
+
+    def lower(self):
+        return self._f("lower")
+
+    def upper(self):
+        return self._f("upper")
+
+    def strip(self):
+        return self._f("strip")
+
+    def title(self):
+        return self._f("title")
+
+    def capitalize(self):
+        return self._f("capitalize")
+
+    def len(self):
+        return self._f("len")
+
+    def contains(self, pat, regex=True, **kw):
+        return self._f("contains", pat)
+
+    def startswith(self, pat):
+        return self._f("startswith", pat)
+
+    def endswith(self, pat):
+        return self._f("endswith", pat)
+
+    def slice(self, start=None, stop=None, step=None):
+        return self._f("slice", start or 0, stop, step or 1)
+
+
+def _pd_dtype_to_bodo(dtype):
+    import numpy as np
+
+    m = {
+        "int8": bt.int8, "int16": bt.int16, "int32": bt.int32, "int64": bt.int64,
+        "float32": bt.float32, "float64": bt.float64, "bool": bt.boolean,
+        "str": bt.string, str: bt.string, int: bt.int64, float: bt.float64,
+        bool: bt.boolean,
+    }
+    if dtype in m:
+        return m[dtype]
+    nd = np.dtype(dtype)
+    if nd.kind == "M":
+        return bt.timestamp_ns
+    return bt.from_numpy_dtype(nd)

@@ -1,0 +1,232 @@
+"""Distributed communication layer: one process per MI355X GPU,
+``torch.distributed`` over RCCL (backend "nccl" on ROCm) for the data plane,
+gloo for CPU test runs.  Reference role: bodo/libs/_distributed.{h,cpp} +
+bodo/libs/_shuffle.cpp (MPI alltoallv) — redesigned on RCCL collectives over
+xGMI: the bulk table shuffle is one all_to_all_single per buffer, issued on
+the communication stream and overlappable with partition kernels.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+import pickle
+from typing import List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..core import types as bt
+from ..core.column import Column
+from ..core.table import Table
+from ..core.types import TypeKind
+
+
+def initialized() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def get_rank() -> int:
+    return dist.get_rank() if initialized() else 0
+
+
+def get_world_size() -> int:
+    return dist.get_world_size() if initialized() else 1
+
+
+def init_from_env(device: Optional[str] = None) -> None:
+    """Initialize the process group from torchrun-style env vars."""
+    if initialized():
+        return
+    if "RANK" not in os.environ:
+        return  # single process mode
+    backend = "nccl" if (device or "").startswith("cuda") or torch.cuda.is_available() else "gloo"
+    if backend == "nccl":
+        local = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", 0)))
+        torch.cuda.set_device(local % torch.cuda.device_count())
+    dist.init_process_group(backend, timeout=datetime.timedelta(seconds=600))
+
+
+def barrier():
+    if initialized():
+        dist.barrier()
+
+
+def allgather_obj(obj) -> list:
+    if not initialized():
+        return [obj]
+    out = [None] * get_world_size()
+    dist.all_gather_object(out, obj)
+    return out
+
+
+def bcast_obj(obj, src=0):
+    if not initialized():
+        return obj
+    lst = [obj]
+    dist.broadcast_object_list(lst, src=src)
+    return lst[0]
+
+
+def _comm_device() -> torch.device:
+    if initialized() and dist.get_backend() == "nccl":
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")
+
+
+def _exchange_counts(send: torch.Tensor) -> torch.Tensor:
+    """all_to_all of per-destination counts (send[j] rows go to rank j)."""
+    w = get_world_size()
+    dev = _comm_device()
+    send_d = send.to(dev, torch.int64)
+    recv_d = torch.empty(w, dtype=torch.int64, device=dev)
+    dist.all_to_all_single(recv_d, send_d)
+    return recv_d.cpu()
+
+
+def alltoallv_tensor(buf: torch.Tensor, send_counts: List[int],
+                     recv_counts: List[int]) -> torch.Tensor:
+    """Variable all-to-all of a 1-D tensor partitioned contiguously by dest."""
+    dev = _comm_device()
+    moved = buf.device != dev
+    src = buf.to(dev) if moved else buf
+    total = int(sum(recv_counts))
+    out = torch.empty(total, dtype=buf.dtype, device=dev)
+    dist.all_to_all_single(out, src, recv_counts, send_counts)
+    return out.to(buf.device) if moved else out
+
+
+def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
+    """Exchange rows of the local shard so row i lands on rank part_ids[i].
+
+    Partition permutation locally (stable), then one alltoallv per buffer over
+    RCCL.  xGMI note: all_to_all_single maps to grouped ncclSend/ncclRecv
+    pairwise over the 7 p2p links; per-destination contiguous packing is done
+    here by the gather kernels.
+    """
+    w = get_world_size()
+    if w == 1:
+        return tbl
+    from .. import ops
+
+    perm = torch.argsort(part_ids, stable=True)
+    counts = torch.bincount(part_ids, minlength=w)
+    packed = ops.take_table(tbl, perm)
+    send_counts = [int(c) for c in counts.tolist()]
+    recv_counts = [int(c) for c in _exchange_counts(counts).tolist()]
+    out_cols = []
+    for col in packed.columns:
+        out_cols.append(_shuffle_column(col, send_counts, recv_counts))
+    return Table(packed.names, out_cols, sum(recv_counts))
+
+
+def _shuffle_column(col: Column, send_counts, recv_counts) -> Column:
+    n_out = sum(recv_counts)
+    mask = None
+    if col.mask is not None or True:
+        # always exchange masks if any rank has one (cheap consensus via
+        # object allgather would serialize; just send a materialized mask
+        # when present locally and let others send all-ones)
+        pass
+    if col.dtype.kind == TypeKind.STRING:
+        # exchange per-row lengths, then bytes with byte-granular counts
+        lens = (col.offsets[1:] - col.offsets[:-1]).contiguous()
+        new_lens = alltoallv_tensor(lens, send_counts, recv_counts)
+        # per-destination byte counts
+        boundaries = np.cumsum([0] + list(send_counts))
+        byte_send = [int((col.offsets[boundaries[i + 1]] - col.offsets[boundaries[i]]).item())
+                     for i in range(len(send_counts))]
+        byte_recv_np = np.zeros(len(recv_counts), dtype=np.int64)
+        rb = np.cumsum([0] + list(recv_counts))
+        for i in range(len(recv_counts)):
+            byte_recv_np[i] = int(new_lens[rb[i]:rb[i + 1]].sum().item())
+        new_bytes = alltoallv_tensor(col.data, byte_send, list(byte_recv_np))
+        new_off = torch.zeros(n_out + 1, dtype=torch.int64, device=col.device)
+        torch.cumsum(new_lens, 0, out=new_off[1:])
+        new_mask = _shuffle_mask(col, send_counts, recv_counts)
+        return Column(bt.string, new_bytes, new_mask, offsets=new_off, length=n_out)
+    data = alltoallv_tensor(col.data.contiguous(), send_counts, recv_counts)
+    new_mask = _shuffle_mask(col, send_counts, recv_counts)
+    if col.dtype.kind == TypeKind.DICT:
+        # dictionaries may differ per-rank; unify by value after exchange
+        return _unify_dict_after_shuffle(col, data, new_mask, send_counts, recv_counts, n_out)
+    return Column(col.dtype, data, new_mask, length=n_out)
+
+
+def _shuffle_mask(col: Column, send_counts, recv_counts):
+    has_mask = any(allgather_obj(col.mask is not None))
+    if not has_mask:
+        return None
+    m = col.mask if col.mask is not None else torch.ones(
+        len(col), dtype=torch.bool, device=col.device)
+    return alltoallv_tensor(m.contiguous(), send_counts, recv_counts)
+
+
+def _unify_dict_after_shuffle(col: Column, codes: torch.Tensor, mask,
+                              send_counts, recv_counts, n_out) -> Column:
+    import pyarrow as pa
+
+    my_dict = col.dictionary.to_pylist()
+    all_dicts = allgather_obj(my_dict)
+    merged: List[str] = []
+    seen = {}
+    for d in all_dicts:
+        for v in d:
+            if v not in seen:
+                seen[v] = len(merged)
+                merged.append(v)
+    # remap received codes: rows from rank j used dict all_dicts[j]
+    out = codes.clone()
+    rb = np.cumsum([0] + list(recv_counts))
+    for j in range(len(recv_counts)):
+        if recv_counts[j] == 0:
+            continue
+        remap = np.array([seen[v] for v in all_dicts[j]], dtype=np.int32) \
+            if all_dicts[j] else np.zeros(0, np.int32)
+        remap_t = torch.from_numpy(remap).to(codes.device)
+        seg = codes[rb[j]:rb[j + 1]].long()
+        out[rb[j]:rb[j + 1]] = remap_t[seg] if len(remap) else seg.to(torch.int32)
+    return Column(bt.dictionary, out.to(torch.int32), mask,
+                  dictionary=pa.array(merged, type=pa.large_string()),
+                  length=n_out)
+
+
+def gather_table(tbl: Table, root: int = 0) -> Optional[Table]:
+    """Gather all shards to `root` (None elsewhere).  Reference:
+    distributed_api.py gatherv."""
+    w = get_world_size()
+    if w == 1:
+        return tbl
+    # simple + robust: arrow-serialize shards and object-gather
+    payload = tbl.to_device("cpu").to_arrow()
+    parts = gather_obj(payload, root)
+    if get_rank() != root:
+        return None
+    import pyarrow as pa
+
+    non_empty = [p for p in parts if p is not None]
+    merged = pa.concat_tables(non_empty, promote_options="permissive")
+    return Table.from_arrow(merged.combine_chunks(), tbl.device)
+
+
+def gather_obj(obj, root: int = 0) -> Optional[list]:
+    if not initialized():
+        return [obj]
+    out = [None] * get_world_size() if get_rank() == root else None
+    dist.gather_object(obj, out, dst=root)
+    return out
+
+
+def allgather_table(tbl: Table) -> Table:
+    """Replicate the concatenation of all shards on every rank (broadcast
+    join build side)."""
+    w = get_world_size()
+    if w == 1:
+        return tbl
+    payload = tbl.to_device("cpu").to_arrow()
+    parts = allgather_obj(payload)
+    import pyarrow as pa
+
+    merged = pa.concat_tables(parts, promote_options="permissive")
+    return Table.from_arrow(merged.combine_chunks(), tbl.device)

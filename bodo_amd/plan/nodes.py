@@ -1,0 +1,287 @@
+"""Logical plan nodes (reference: bodo/pandas/plan.py:305-573 LogicalOperator
+hierarchy).  A plan is an immutable tree; the optimizer rewrites it; the
+executor (bodo_amd/engine/executor.py) interprets the optimized tree against
+the rank-local shard with distributed exchanges inserted per operator."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
+
+from .expr import Expr
+
+
+class PlanNode:
+    def children(self) -> Tuple["PlanNode", ...]:
+        return ()
+
+    def with_children(self, *ch: "PlanNode") -> "PlanNode":
+        assert not ch
+        return self
+
+    # schema: list of column names this node produces (best-effort; scans
+    # fill from file metadata, other nodes derive)
+    def out_columns(self) -> Optional[List[str]]:
+        return None
+
+
+# ---------------------------------------------------------------- sources
+@dataclass(frozen=True)
+class ParquetScan(PlanNode):
+    path: str
+    columns: Optional[Tuple[str, ...]] = None  # projection pushdown target
+    filters: Tuple[Expr, ...] = ()  # filter pushdown (row-group stats + row-level)
+    schema_names: Tuple[str, ...] = ()  # discovered at plan build
+
+    def out_columns(self):
+        return list(self.columns) if self.columns else list(self.schema_names)
+
+
+@dataclass(frozen=True)
+class CsvScan(PlanNode):
+    path: str
+    options: Tuple[Tuple[str, Any], ...] = ()
+    columns: Optional[Tuple[str, ...]] = None
+    schema_names: Tuple[str, ...] = ()
+
+    def out_columns(self):
+        return list(self.columns) if self.columns else list(self.schema_names)
+
+
+@dataclass(frozen=True)
+class PandasScan(PlanNode):
+    """In-memory source: a host pandas DataFrame (replicated or to scatter) or
+    an already-distributed Table shard registered in the result registry."""
+
+    data_id: str  # key into the executor's object registry
+    names: Tuple[str, ...] = ()
+    distributed: bool = False  # True: registry holds this rank's shard
+
+    def out_columns(self):
+        return list(self.names)
+
+
+# ---------------------------------------------------------------- unary
+@dataclass(frozen=True)
+class Projection(PlanNode):
+    child: PlanNode
+    names: Tuple[str, ...] = ()
+    exprs: Tuple[Expr, ...] = ()
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Projection(ch[0], self.names, self.exprs)
+
+    def out_columns(self):
+        return list(self.names)
+
+
+@dataclass(frozen=True)
+class Filter(PlanNode):
+    child: PlanNode
+    cond: Expr = None
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Filter(ch[0], self.cond)
+
+    def out_columns(self):
+        return self.child.out_columns()
+
+
+@dataclass(frozen=True)
+class Aggregate(PlanNode):
+    child: PlanNode
+    keys: Tuple[str, ...] = ()
+    # aggs: (out_name, in_name, func); func in
+    # sum/count/mean/min/max/size/nunique/first/last/var/std/median/prod
+    aggs: Tuple[Tuple[str, str, str], ...] = ()
+    as_index: bool = False
+    dropna: bool = True
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Aggregate(ch[0], self.keys, self.aggs, self.as_index, self.dropna)
+
+    def out_columns(self):
+        return list(self.keys) + [a[0] for a in self.aggs]
+
+
+@dataclass(frozen=True)
+class Sort(PlanNode):
+    child: PlanNode
+    keys: Tuple[str, ...] = ()
+    ascending: Tuple[bool, ...] = ()
+    na_position: str = "last"
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Sort(ch[0], self.keys, self.ascending, self.na_position)
+
+    def out_columns(self):
+        return self.child.out_columns()
+
+
+@dataclass(frozen=True)
+class Limit(PlanNode):
+    child: PlanNode
+    n: int = 0
+    offset: int = 0
+    tail: bool = False
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Limit(ch[0], self.n, self.offset, self.tail)
+
+    def out_columns(self):
+        return self.child.out_columns()
+
+
+@dataclass(frozen=True)
+class Distinct(PlanNode):
+    child: PlanNode
+    subset: Optional[Tuple[str, ...]] = None
+    keep: str = "first"
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Distinct(ch[0], self.subset, self.keep)
+
+    def out_columns(self):
+        return self.child.out_columns()
+
+
+@dataclass(frozen=True)
+class Sample(PlanNode):
+    child: PlanNode
+    n: Optional[int] = None
+    frac: Optional[float] = None
+    seed: Optional[int] = None
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Sample(ch[0], self.n, self.frac, self.seed)
+
+    def out_columns(self):
+        return self.child.out_columns()
+
+
+@dataclass(frozen=True)
+class MapPartitions(PlanNode):
+    """Run a python function over each rank's shard as a pandas DataFrame
+    (reference: frame.py map_partitions / apply axis=1 fallback)."""
+
+    child: PlanNode
+    func: Callable = field(compare=False)
+    args: Tuple[Any, ...] = ()
+    names: Tuple[str, ...] = ()
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return MapPartitions(ch[0], self.func, self.args, self.names)
+
+    def out_columns(self):
+        return list(self.names) if self.names else None
+
+
+@dataclass(frozen=True)
+class ShuffleByKey(PlanNode):
+    """Explicit hash-repartition so equal keys co-locate on one rank
+    (used by groupby.apply / median paths)."""
+
+    child: PlanNode
+    keys: Tuple[str, ...] = ()
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return ShuffleByKey(ch[0], self.keys)
+
+    def out_columns(self):
+        return self.child.out_columns()
+
+
+# ---------------------------------------------------------------- binary
+@dataclass(frozen=True)
+class Join(PlanNode):
+    left: PlanNode
+    right: PlanNode
+    left_on: Tuple[str, ...] = ()
+    right_on: Tuple[str, ...] = ()
+    how: str = "inner"  # inner/left/right/outer/semi/anti/cross
+    suffixes: Tuple[str, str] = ("_x", "_y")
+
+    def children(self):
+        return (self.left, self.right)
+
+    def with_children(self, *ch):
+        return Join(ch[0], ch[1], self.left_on, self.right_on, self.how, self.suffixes)
+
+
+@dataclass(frozen=True)
+class Union(PlanNode):
+    inputs: Tuple[PlanNode, ...] = ()
+    distinct: bool = False
+
+    def children(self):
+        return self.inputs
+
+    def with_children(self, *ch):
+        return Union(tuple(ch), self.distinct)
+
+    def out_columns(self):
+        return self.inputs[0].out_columns()
+
+
+# ---------------------------------------------------------------- sinks
+@dataclass(frozen=True)
+class ParquetWrite(PlanNode):
+    child: PlanNode
+    path: str = ""
+    compression: Optional[str] = "snappy"
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return ParquetWrite(ch[0], self.path, self.compression)
+
+
+@dataclass(frozen=True)
+class Reduce(PlanNode):
+    """Whole-column reductions producing a scalar row (Series.sum() etc.)."""
+
+    child: PlanNode
+    aggs: Tuple[Tuple[str, str, str], ...] = ()  # (out, in, func)
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Reduce(ch[0], self.aggs)
+
+    def out_columns(self):
+        return [a[0] for a in self.aggs]
+
+
+def walk(node: PlanNode):
+    yield node
+    for c in node.children():
+        yield from walk(c)

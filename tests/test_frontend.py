@@ -1,0 +1,244 @@
+"""Frontend + single-rank executor tests (CPU): differential vs pandas."""
+
+import numpy as np
+import pandas as pd
+import pytest
+
+import bodo_amd.pandas as bpd
+from tests.utils import check_query
+
+
+def simple_df(n=1000, seed=0):
+    rng = np.random.default_rng(seed)
+    return pd.DataFrame({
+        "a": rng.integers(0, 10, n),
+        "b": rng.uniform(-1, 1, n),
+        "c": rng.choice(["x", "y", "z"], n),
+        "d": rng.integers(0, 2, n).astype(bool),
+    })
+
+
+def test_filter_project():
+    def q(m, df):
+        f = df[df.a > 5]
+        f["e"] = f.a * 2 + f.b
+        return f[["a", "b", "e"]]
+
+    check_query(q, {"df": simple_df()})
+
+
+def test_chained_filters():
+    def q(m, df):
+        return df[(df.a > 2) & (df.b < 0.5) | df.d]
+
+    check_query(q, {"df": simple_df()})
+
+
+def test_groupby_agg_dict():
+    def q(m, df):
+        return df.groupby("a", as_index=False).agg({"b": "sum"})
+
+    check_query(q, {"df": simple_df()}, sort_by=["a"])
+
+
+def test_groupby_multi_key():
+    def q(m, df):
+        return df.groupby(["a", "c"], as_index=False).agg(
+            s=m.NamedAgg("b", "sum"), mn=m.NamedAgg("b", "mean"),
+            mx=m.NamedAgg("b", "max"), n=m.NamedAgg("b", "count"))
+
+    check_query(q, {"df": simple_df()}, sort_by=["a", "c"])
+
+
+def test_groupby_size_min_first():
+    def q(m, df):
+        return df.groupby("c", as_index=False).agg(
+            sz=m.NamedAgg("a", "size"), mi=m.NamedAgg("b", "min"))
+
+    check_query(q, {"df": simple_df()}, sort_by=["c"])
+
+
+def test_sort_values():
+    def q(m, df):
+        return df.sort_values(["a", "b"], ascending=[True, False])
+
+    check_query(q, {"df": simple_df()})
+
+
+def test_sort_strings():
+    def q(m, df):
+        return df.sort_values(["c", "a"])[["c", "a"]]
+
+    check_query(q, {"df": simple_df()}, sort_by=None)
+
+
+def test_head():
+    def q(m, df):
+        return df.sort_values("b").head(17)
+
+    check_query(q, {"df": simple_df()})
+
+
+def test_merge_inner():
+    def q(m, left, right):
+        return m.merge(left, right, on="k", how="inner").sort_values(
+            ["k", "v1", "v2"])
+
+    rng = np.random.default_rng(3)
+    left = pd.DataFrame({"k": rng.integers(0, 20, 500), "v1": rng.uniform(0, 1, 500)})
+    right = pd.DataFrame({"k": np.arange(15), "v2": rng.uniform(0, 1, 15)})
+    check_query(q, {"left": left, "right": right})
+
+
+def test_merge_left():
+    def q(m, left, right):
+        return m.merge(left, right, on="k", how="left").sort_values(
+            ["k", "v1"])
+
+    rng = np.random.default_rng(4)
+    left = pd.DataFrame({"k": rng.integers(0, 30, 300), "v1": rng.uniform(0, 1, 300)})
+    right = pd.DataFrame({"k": np.arange(15), "v2": rng.uniform(0, 1, 15)})
+    check_query(q, {"left": left, "right": right})
+
+
+def test_merge_different_keys_suffixes():
+    def q(m, left, right):
+        return m.merge(left, right, left_on="k1", right_on="k2",
+                       how="inner").sort_values(["k1", "v_x"])
+
+    rng = np.random.default_rng(5)
+    left = pd.DataFrame({"k1": rng.integers(0, 10, 100), "v": rng.uniform(0, 1, 100)})
+    right = pd.DataFrame({"k2": np.arange(8), "v": rng.uniform(0, 1, 8)})
+    check_query(q, {"left": left, "right": right})
+
+
+def test_dt_accessor():
+    def q(m, df):
+        df["y"] = df.t.dt.year
+        df["mo"] = df.t.dt.month
+        df["h"] = df.t.dt.hour
+        df["dow"] = df.t.dt.dayofweek
+        return df[["y", "mo", "h", "dow"]]
+
+    rng = np.random.default_rng(6)
+    t = pd.to_datetime(pd.Timestamp("2020-01-01").value
+                       + rng.integers(0, 3 * 365 * 86400 * 10**9, 500))
+    check_query(q, {"df": pd.DataFrame({"t": t})})
+
+
+def test_dt_date_merge():
+    def q(m, df, w):
+        df["date"] = df.t.dt.date
+        w["date"] = w.d.dt.date
+        return df.merge(w, on="date", how="inner")[["date", "v", "p"]] \
+            .sort_values(["date", "v"])
+
+    rng = np.random.default_rng(7)
+    t = pd.to_datetime(pd.Timestamp("2021-06-01").value
+                       + rng.integers(0, 30 * 86400 * 10**9, 300))
+    d = pd.date_range("2021-06-01", periods=30)
+    check_query(q, {
+        "df": pd.DataFrame({"t": t, "v": rng.uniform(0, 1, 300)}),
+        "w": pd.DataFrame({"d": d, "p": rng.uniform(0, 1, 30)}),
+    })
+
+
+def test_isin_map():
+    def q(m, df):
+        df["w"] = df.a.isin([1, 3, 5])
+        df["lab"] = df.a.map(lambda v: "low" if v < 5 else "high")
+        return df[["a", "w", "lab"]]
+
+    check_query(q, {"df": simple_df()})
+
+
+def test_series_reductions():
+    df = simple_df()
+    b = bpd.from_pandas(df)
+    assert abs(b.b.sum() - df.b.sum()) < 1e-9
+    assert abs(b.b.mean() - df.b.mean()) < 1e-9
+    assert b.a.max() == df.a.max()
+    assert b.a.min() == df.a.min()
+    assert b.b.count() == df.b.count()
+
+
+def test_drop_duplicates():
+    def q(m, df):
+        return df.drop_duplicates(subset=["a"]).sort_values("a")[["a"]]
+
+    check_query(q, {"df": simple_df()})
+
+
+def test_rename_drop():
+    def q(m, df):
+        return df.rename(columns={"a": "alpha"}).drop(columns=["d"])
+
+    check_query(q, {"df": simple_df()})
+
+
+def test_dropna_fillna():
+    def q(m, df):
+        out = df.dropna(subset=["x"])
+        return out
+
+    rng = np.random.default_rng(8)
+    x = rng.uniform(0, 1, 200)
+    x[rng.random(200) < 0.3] = np.nan
+    check_query(q, {"df": pd.DataFrame({"x": x, "y": rng.integers(0, 5, 200)})})
+
+
+def test_value_counts():
+    df = simple_df()
+    b = bpd.from_pandas(df)
+    got = b.c.value_counts().sort_index()
+    exp = df.c.value_counts().sort_index()
+    assert (got.to_numpy() == exp.to_numpy()).all()
+
+
+def test_concat():
+    def q(m, df1, df2):
+        return m.concat([df1, df2]).sort_values(["a", "b"])
+
+    check_query(q, {"df1": simple_df(300, 1), "df2": simple_df(300, 2)})
+
+
+def test_nunique_median_groupby():
+    def q(m, df):
+        return df.groupby("a", as_index=False).agg(
+            nu=m.NamedAgg("c", "nunique"), md=m.NamedAgg("b", "median"))
+
+    check_query(q, {"df": simple_df()}, sort_by=["a"])
+
+
+def test_var_std_groupby():
+    def q(m, df):
+        return df.groupby("c", as_index=False).agg(
+            v=m.NamedAgg("b", "var"), s=m.NamedAgg("b", "std"))
+
+    check_query(q, {"df": simple_df()}, sort_by=["c"])
+
+
+def test_apply_axis1():
+    def q(m, df):
+        df["s"] = df.apply(lambda r: r["a"] * 2 + (1 if r["d"] else 0), axis=1)
+        return df[["a", "s"]]
+
+    check_query(q, {"df": simple_df(200)})
+
+
+def test_str_accessor():
+    def q(m, df):
+        df["u"] = df.c.str.upper()
+        df["has_x"] = df.c.str.contains("x")
+        return df[["c", "u", "has_x"]]
+
+    check_query(q, {"df": simple_df(200)})
+
+
+def test_fallback_describe():
+    df = simple_df(100)
+    b = bpd.from_pandas(df)
+    got = b.describe()
+    exp = df.describe()
+    pd.testing.assert_frame_equal(got.to_pandas() if hasattr(got, "to_pandas")
+                                  else got, exp)

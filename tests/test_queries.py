@@ -1,0 +1,169 @@
+"""End-to-end benchmark-query tests (reference: e2e-tests/tpch/test_q1.py,
+benchmarks/nyc_taxi/bodo/nyc_taxi_precipitation.py)."""
+
+import numpy as np
+import pandas as pd
+import pytest
+
+from tests.utils import check_query
+
+
+def make_taxi(n=20000, seed=0):
+    rng = np.random.default_rng(seed)
+    base = pd.Timestamp("2023-01-01").value
+    pickup = base + rng.integers(0, 365 * 86400 * 10**9, n)
+    trips = pd.DataFrame({
+        "hvfhs_license_num": rng.choice(["HV0002", "HV0003", "HV0004", "HV0005"], n),
+        "pickup_datetime": pd.to_datetime(pickup),
+        "PULocationID": rng.integers(1, 266, n).astype(np.int64),
+        "DOLocationID": rng.integers(1, 266, n).astype(np.int64),
+        "trip_miles": rng.exponential(3.0, n),
+    })
+    dates = pd.date_range("2023-01-01", "2023-12-31")
+    weather = pd.DataFrame({"DATE": dates,
+                            "PRCP": rng.exponential(0.05, len(dates))})
+    return trips, weather
+
+
+def nyc_taxi_q1(m, trips, weather):
+    w = weather.rename(columns={"DATE": "date", "PRCP": "precipitation"})
+    t = trips
+    w["date"] = w["date"].dt.date
+    t["date"] = t["pickup_datetime"].dt.date
+    t["month"] = t["pickup_datetime"].dt.month
+    t["hour"] = t["pickup_datetime"].dt.hour
+    t["weekday"] = t["pickup_datetime"].dt.dayofweek.isin([0, 1, 2, 3, 4])
+    mt = t.merge(w, on="date", how="inner")
+    mt["date_with_precipitation"] = mt["precipitation"] > 0.1
+
+    def get_time_bucket(tt):
+        if tt in (8, 9, 10):
+            return "morning"
+        if tt in (11, 12, 13, 14, 15):
+            return "midday"
+        if tt in (16, 17, 18):
+            return "afternoon"
+        if tt in (19, 20, 21):
+            return "evening"
+        return "other"
+
+    mt["time_bucket"] = mt.hour.map(get_time_bucket)
+    g = mt.groupby(
+        ["PULocationID", "DOLocationID", "month", "weekday",
+         "date_with_precipitation", "time_bucket"],
+        as_index=False).agg({"hvfhs_license_num": "count", "trip_miles": "mean"})
+    return g.sort_values(
+        by=["PULocationID", "DOLocationID", "month", "weekday",
+            "date_with_precipitation", "time_bucket"])
+
+
+def test_nyc_taxi_q1():
+    trips, weather = make_taxi()
+    check_query(nyc_taxi_q1, {"trips": trips, "weather": weather})
+
+
+def make_lineitem(n=30000, seed=1):
+    rng = np.random.default_rng(seed)
+    return pd.DataFrame({
+        "L_ORDERKEY": rng.integers(0, 10000, n),
+        "L_QUANTITY": rng.integers(1, 51, n).astype(np.float64),
+        "L_EXTENDEDPRICE": rng.uniform(1000, 100000, n),
+        "L_DISCOUNT": rng.uniform(0, 0.1, n).round(2),
+        "L_TAX": rng.uniform(0, 0.08, n).round(2),
+        "L_RETURNFLAG": rng.choice(["A", "N", "R"], n),
+        "L_LINESTATUS": rng.choice(["O", "F"], n),
+        "L_SHIPDATE": pd.to_datetime(
+            pd.Timestamp("1992-01-01").value
+            + rng.integers(0, 2500 * 86400 * 10**9, n)),
+    })
+
+
+def tpch_q1(m, lineitem):
+    var1 = pd.Timestamp("1998-09-02")
+    filt = lineitem[lineitem["L_SHIPDATE"] <= var1]
+    filt["DISC_PRICE"] = filt.L_EXTENDEDPRICE * (1.0 - filt.L_DISCOUNT)
+    filt["CHARGE"] = (filt.L_EXTENDEDPRICE * (1.0 - filt.L_DISCOUNT)
+                      * (1.0 + filt.L_TAX))
+    gb = filt.groupby(["L_RETURNFLAG", "L_LINESTATUS"], as_index=False)
+    agg = gb.agg(
+        SUM_QTY=m.NamedAgg("L_QUANTITY", "sum"),
+        SUM_BASE_PRICE=m.NamedAgg("L_EXTENDEDPRICE", "sum"),
+        SUM_DISC_PRICE=m.NamedAgg("DISC_PRICE", "sum"),
+        SUM_CHARGE=m.NamedAgg("CHARGE", "sum"),
+        AVG_QTY=m.NamedAgg("L_QUANTITY", "mean"),
+        AVG_PRICE=m.NamedAgg("L_EXTENDEDPRICE", "mean"),
+        AVG_DISC=m.NamedAgg("L_DISCOUNT", "mean"),
+        COUNT_ORDER=m.NamedAgg("L_ORDERKEY", "size"))
+    return agg.sort_values(["L_RETURNFLAG", "L_LINESTATUS"])
+
+
+def test_tpch_q1():
+    check_query(tpch_q1, {"lineitem": make_lineitem()})
+
+
+def test_parquet_roundtrip(tmp_path):
+    import bodo_amd.pandas as bpd
+
+    df = make_lineitem(5000)
+    p = str(tmp_path / "li.parquet")
+    df.to_parquet(p)
+    b = bpd.read_parquet(p)
+    out = tpch_q1(bpd, b).to_pandas().reset_index(drop=True)
+    exp = tpch_q1(pd, df).reset_index(drop=True)
+    for c in ("L_RETURNFLAG", "L_LINESTATUS"):
+        out[c] = out[c].astype(str)
+        exp[c] = exp[c].astype(str)
+    pd.testing.assert_frame_equal(out, exp, check_dtype=False)
+
+
+def test_parquet_pushdown_pruning(tmp_path):
+    """Column pruning + filter pushdown reach the scan node."""
+    import bodo_amd.pandas as bpd
+    from bodo_amd.engine.optimizer import optimize
+    from bodo_amd.plan import nodes as pn
+
+    df = make_lineitem(1000)
+    p = str(tmp_path / "li2.parquet")
+    df.to_parquet(p)
+    b = bpd.read_parquet(p)
+    q = b[b.L_QUANTITY > 10.0][["L_ORDERKEY", "L_QUANTITY"]]
+    plan = optimize(q._lazy_plan)
+    scans = [n for n in _walk(plan) if isinstance(n, pn.ParquetScan)]
+    assert len(scans) == 1
+    assert set(scans[0].columns) == {"L_ORDERKEY", "L_QUANTITY"}
+    assert len(scans[0].filters) == 1
+    out = q.to_pandas()
+    exp = df[df.L_QUANTITY > 10.0][["L_ORDERKEY", "L_QUANTITY"]].reset_index(drop=True)
+    pd.testing.assert_frame_equal(out, exp, check_dtype=False)
+
+
+def _walk(n):
+    yield n
+    for c in n.children():
+        yield from _walk(c)
+
+
+def test_readme_20m_style(tmp_path):
+    """Scaled version of the README quickstart config:
+    read_parquet + apply + groupby + to_parquet via bodo_amd.pandas."""
+    import bodo_amd.pandas as bpd
+
+    n = 50_000
+    rng = np.random.default_rng(2)
+    df = pd.DataFrame({
+        "A": rng.integers(0, 100, n),
+        "B": rng.uniform(0, 1, n),
+    })
+    src = str(tmp_path / "in.parquet")
+    dst = str(tmp_path / "out.parquet")
+    df.to_parquet(src)
+    b = bpd.read_parquet(src)
+    b["C"] = b.A.map(lambda x: x * 2 + 1)
+    res = b.groupby("A", as_index=False).agg(s=bpd.NamedAgg("C", "sum"),
+                                             m=bpd.NamedAgg("B", "mean"))
+    res.to_parquet(dst)
+    out = pd.read_parquet(dst).sort_values("A").reset_index(drop=True)
+    exp_c = df.A * 2 + 1
+    exp = df.assign(C=exp_c).groupby("A", as_index=False).agg(
+        s=("C", "sum"), m=("B", "mean")).sort_values("A").reset_index(drop=True)
+    pd.testing.assert_frame_equal(out, exp, check_dtype=False)
