@@ -211,7 +211,7 @@ def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:
     else:
         sample = tbl.select(list(keys)).to_device("cpu").to_pandas()
     all_samples = comm.allgather_obj(sample)
-    merged = pd.concat(all_samples, ignore_index=True)
+    merged = _decat(pd.concat(all_samples, ignore_index=True))
     merged = merged.sort_values(list(keys), ascending=asc, na_position=na_position)
     if len(merged) == 0:
         return tbl
@@ -222,13 +222,24 @@ def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:
     # host comparison on the key sample of local rows (vectorized via
     # pandas merge_sorted rank); for performance the GPU path will move to a
     # device searchsorted on normalized keys.
-    local_keys = tbl.select(list(keys)).to_device("cpu").to_pandas()
+    local_keys = _decat(tbl.select(list(keys)).to_device("cpu").to_pandas())
     part = np.zeros(n, dtype=np.int64)
     for i in range(w - 1):
         row = splitters.iloc[i]
         gt = _row_greater(local_keys, row, keys, asc, na_position)
         part = np.where(gt, i + 1, part)
     return comm.shuffle_table(tbl, torch.from_numpy(part).to(tbl.device))
+
+
+def _decat(df: pd.DataFrame) -> pd.DataFrame:
+    """Categoricals -> strings so splitter comparisons are well-defined."""
+    out = df
+    for c in df.columns:
+        if isinstance(df[c].dtype, pd.CategoricalDtype):
+            if out is df:
+                out = df.copy()
+            out[c] = out[c].astype("string")
+    return out
 
 
 def _row_greater(df: pd.DataFrame, row, keys, asc, na_position) -> np.ndarray:

@@ -1,0 +1,177 @@
+"""Multi-process distributed tests on gloo (world_size=2), CPU.
+
+Exercises the same exchange logic (hash shuffle, two-phase agg, broadcast
+join, range-partitioned sort) that runs over RCCL on MI355X.  Reference
+analog: bodo/runtests.py running pytest under mpiexec -n N.
+"""
+
+import multiprocessing as mp
+import os
+import pickle
+import sys
+import traceback
+
+import numpy as np
+import pandas as pd
+import pytest
+
+pytestmark = pytest.mark.multi_rank
+
+_PORT = [29600]
+
+
+def _worker(rank, world, port, fn, payload, out_q):
+    try:
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["BODO_AMD_DEVICE"] = "cpu"
+        import warnings
+
+        warnings.filterwarnings("ignore")
+        import torch.distributed as dist
+
+        import bodo_amd  # noqa: F401  (auto-inits process group)
+        import bodo_amd.pandas as bpd
+
+        res = fn(bpd, rank, payload)
+        got = res.to_pandas() if hasattr(res, "to_pandas") else res
+        if rank == 0:
+            out_q.put(("ok", got))
+        dist.barrier()
+        dist.destroy_process_group()
+    except Exception:
+        out_q.put(("err", traceback.format_exc()))
+
+
+def run_dist(fn, payload, world=2):
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    _PORT[0] += 1
+    port = _PORT[0] + os.getpid() % 500
+    procs = [ctx.Process(target=_worker, args=(r, world, port, fn, payload, out_q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    status, result = out_q.get(timeout=180)
+    for p in procs:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.terminate()
+    assert status == "ok", result
+    return result
+
+
+# ----------------------------------------------------------------------
+# module-level query functions (must be picklable)
+# ----------------------------------------------------------------------
+
+def _q_groupby(bpd, rank, payload):
+    df = payload["df"]
+    b = bpd.from_pandas(df)
+    return b.groupby(["a", "c"], as_index=False).agg(
+        s=bpd.NamedAgg("b", "sum"), m=bpd.NamedAgg("b", "mean"),
+        n=bpd.NamedAgg("b", "count"), mx=bpd.NamedAgg("b", "max"),
+    ).sort_values(["a", "c"])
+
+
+def _q_join(bpd, rank, payload):
+    l = bpd.from_pandas(payload["left"])
+    r = bpd.from_pandas(payload["right"])
+    return l.merge(r, on="k", how="inner").sort_values(["k", "v1", "v2"])
+
+
+def _q_sort(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).sort_values(
+        ["a", "b"], ascending=[True, False])
+
+
+def _q_taxi(bpd, rank, payload):
+    from tests.test_queries import nyc_taxi_q1
+
+    t = bpd.from_pandas(payload["trips"])
+    w = bpd.from_pandas(payload["weather"])
+    return nyc_taxi_q1(bpd, t, w)
+
+
+def _q_median(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).groupby("a", as_index=False).agg(
+        md=bpd.NamedAgg("b", "median"), nu=bpd.NamedAgg("c", "nunique"),
+    ).sort_values("a")
+
+
+def _q_distinct(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).drop_duplicates(
+        subset=["a"]).sort_values("a")[["a"]]
+
+
+def _q_limit(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).sort_values("b").head(13)
+
+
+def _df(n=3000, seed=0):
+    rng = np.random.default_rng(seed)
+    return pd.DataFrame({
+        "a": rng.integers(0, 20, n),
+        "b": rng.uniform(-1, 1, n),
+        "c": rng.choice(["x", "y", "z", "w"], n),
+    })
+
+
+def test_dist_groupby():
+    df = _df()
+    got = run_dist(_q_groupby, {"df": df}).reset_index(drop=True)
+    exp = df.groupby(["a", "c"], as_index=False).agg(
+        s=("b", "sum"), m=("b", "mean"), n=("b", "count"), mx=("b", "max"),
+    ).sort_values(["a", "c"]).reset_index(drop=True)
+    got["c"] = got["c"].astype(str)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_dist_join():
+    rng = np.random.default_rng(3)
+    left = pd.DataFrame({"k": rng.integers(0, 50, 2000),
+                         "v1": rng.uniform(0, 1, 2000)})
+    right = pd.DataFrame({"k": np.arange(40), "v2": rng.uniform(0, 1, 40)})
+    got = run_dist(_q_join, {"left": left, "right": right}).reset_index(drop=True)
+    exp = left.merge(right, on="k", how="inner").sort_values(
+        ["k", "v1", "v2"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_dist_sort():
+    df = _df(2500, 5)
+    got = run_dist(_q_sort, {"df": df}).reset_index(drop=True)
+    exp = df.sort_values(["a", "b"], ascending=[True, False]).reset_index(drop=True)
+    got["c"] = got["c"].astype(str)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_dist_taxi_q1():
+    from tests.test_queries import make_taxi, nyc_taxi_q1
+
+    trips, weather = make_taxi(8000, 7)
+    got = run_dist(_q_taxi, {"trips": trips, "weather": weather}).reset_index(drop=True)
+    exp = nyc_taxi_q1(pd, trips.copy(), weather.copy()).reset_index(drop=True)
+    got["time_bucket"] = got["time_bucket"].astype(str)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_dist_median_nunique():
+    df = _df(1500, 9)
+    got = run_dist(_q_median, {"df": df}).reset_index(drop=True)
+    exp = df.groupby("a", as_index=False).agg(
+        md=("b", "median"), nu=("c", "nunique")).sort_values("a").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_dist_distinct_limit():
+    df = _df(1200, 11)
+    got = run_dist(_q_distinct, {"df": df}).reset_index(drop=True)
+    exp = df.drop_duplicates(subset=["a"]).sort_values("a")[["a"]].reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    got2 = run_dist(_q_limit, {"df": df}).reset_index(drop=True)
+    exp2 = df.sort_values("b").head(13).reset_index(drop=True)
+    got2["c"] = got2["c"].astype(str)
+    pd.testing.assert_frame_equal(got2, exp2, check_dtype=False)
