@@ -1,0 +1,227 @@
+"""GPU kernel numerics tests: each HIP kernel vs the CPU (pandas/numpy)
+reference of the same op.  All marked gpu; run via
+``pytest tests -m gpu`` on an MI355X box."""
+
+import numpy as np
+import pandas as pd
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(autouse=True)
+def _cuda_required():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+
+
+def _to_gpu(col):
+    return col.to_device("cuda")
+
+
+def make_cols(n=10000, seed=0):
+    from bodo_amd.core.column import Column
+
+    rng = np.random.default_rng(seed)
+    cols = {
+        "i64": Column.from_numpy(rng.integers(-1000, 1000, n)),
+        "i32": Column.from_numpy(rng.integers(-100, 100, n).astype(np.int32)),
+        "f64": Column.from_numpy(rng.uniform(-1, 1, n)),
+        "b": Column.from_numpy(rng.integers(0, 2, n).astype(bool)),
+        "s": Column.from_numpy(rng.choice(["aa", "bb", "cc", "dddd"], n)),
+    }
+    import pyarrow as pa
+
+    dict_arr = pa.array(rng.choice(["u", "vv", "www"], n)).dictionary_encode()
+    cols["d"] = Column.from_arrow(dict_arr)
+    return cols
+
+
+def test_hash_matches_cpu():
+    from bodo_amd import ops
+
+    cols = make_cols()
+    for name, c in cols.items():
+        h_cpu = ops.hash_columns([c])
+        h_gpu = ops.hash_columns([_to_gpu(c)]).cpu()
+        assert torch.equal(h_cpu, h_gpu), f"hash mismatch for {name}"
+    # multi-column combine
+    combo = [cols["i64"], cols["s"], cols["f64"]]
+    h_cpu = ops.hash_columns(combo)
+    h_gpu = ops.hash_columns([_to_gpu(c) for c in combo]).cpu()
+    assert torch.equal(h_cpu, h_gpu)
+
+
+def test_hash_nulls_match():
+    from bodo_amd import ops
+    from bodo_amd.core.column import Column
+
+    rng = np.random.default_rng(1)
+    data = rng.integers(0, 50, 5000)
+    mask = rng.random(5000) > 0.2
+    c = Column.from_numpy(data, mask=mask)
+    h_cpu = ops.hash_columns([c])
+    h_gpu = ops.hash_columns([_to_gpu(c)]).cpu()
+    assert torch.equal(h_cpu, h_gpu)
+
+
+def test_dt_field_gpu():
+    from bodo_amd.core.column import Column
+    from bodo_amd.ops import gpu
+    from bodo_amd.ops.evaluate import _dt_field_cpu
+
+    rng = np.random.default_rng(2)
+    ts = (pd.Timestamp("1995-01-01").value
+          + rng.integers(0, 40 * 365 * 86400 * 10**9, 20000))
+    c = Column.from_numpy(ts.view("datetime64[ns]"))
+    for fld in ("year", "month", "day", "hour", "minute", "second",
+                "dayofweek", "dayofyear", "quarter", "date", "normalize"):
+        exp = _dt_field_cpu(c, fld)
+        got = gpu.dt_field(_to_gpu(c), fld)
+        assert torch.equal(exp.data, got.data.cpu()), fld
+
+
+def test_gather_string_gpu():
+    from bodo_amd.core.column import Column
+    from bodo_amd import ops
+
+    rng = np.random.default_rng(3)
+    vals = np.array(["", "a", "xyz", "hello world", "qq"], dtype=object)
+    c = Column.from_numpy(vals[rng.integers(0, 5, 3000)])
+    idx = torch.from_numpy(rng.integers(0, 3000, 500))
+    exp = ops.gather(c, idx)
+    got = ops.gather(_to_gpu(c), idx.cuda())
+    assert exp.to_pandas().tolist() == got.to_pandas().tolist()
+
+
+def _gb_frames(n=20000, seed=4, with_nulls=False):
+    rng = np.random.default_rng(seed)
+    df = pd.DataFrame({
+        "k1": rng.integers(0, 50, n),
+        "k2": rng.choice(["x", "y", "z"], n),
+        "v1": rng.uniform(-1, 1, n),
+        "v2": rng.integers(0, 100, n),
+    })
+    if with_nulls:
+        df.loc[rng.random(n) < 0.1, "v1"] = np.nan
+    return df
+
+
+@pytest.mark.parametrize("funcs", [
+    {"s": ("v1", "sum"), "c": ("v1", "count"), "m": ("v1", "mean")},
+    {"mi": ("v1", "min"), "mx": ("v1", "max"), "sz": ("v1", "size")},
+    {"is": ("v2", "sum"), "imi": ("v2", "min"), "imx": ("v2", "max")},
+    {"f": ("v2", "first"), "l": ("v2", "last")},
+    {"md": ("v1", "median"), "nu": ("v2", "nunique")},
+    {"vr": ("v1", "var"), "sd": ("v1", "std")},
+])
+def test_groupby_gpu_vs_pandas(funcs):
+    from bodo_amd.core.table import Table
+    from bodo_amd.ops import gpu
+
+    df = _gb_frames(with_nulls=True)
+    t = Table.from_pandas(df, device="cuda")
+    aggs = [(out, src, f) for out, (src, f) in funcs.items()]
+    got = gpu.groupby_local(t, ["k1", "k2"], aggs).to_pandas()
+    got["k2"] = got["k2"].astype(str)
+    exp = df.groupby(["k1", "k2"], sort=False, dropna=True).agg(
+        **{out: (src, f) for out, (src, f) in funcs.items()}).reset_index()
+    got = got.sort_values(["k1", "k2"]).reset_index(drop=True)
+    exp = exp.sort_values(["k1", "k2"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False,
+                                  check_exact=False, rtol=1e-9)
+
+
+def test_groupby_gpu_high_cardinality():
+    from bodo_amd.core.table import Table
+    from bodo_amd.ops import gpu
+
+    rng = np.random.default_rng(5)
+    n = 200_000
+    df = pd.DataFrame({"k": rng.integers(0, n // 2, n),
+                       "v": rng.uniform(0, 1, n)})
+    t = Table.from_pandas(df, device="cuda")
+    got = gpu.groupby_local(t, ["k"], [("s", "v", "sum"), ("n", "v", "count")]) \
+        .to_pandas().sort_values("k").reset_index(drop=True)
+    exp = df.groupby("k", sort=True).agg(s=("v", "sum"), n=("v", "count")) \
+        .reset_index()
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False,
+                                  check_exact=False, rtol=1e-9)
+
+
+@pytest.mark.parametrize("how", ["inner", "left", "semi", "anti", "outer"])
+def test_join_gpu_vs_pandas(how):
+    from bodo_amd.core.table import Table
+    from bodo_amd.ops import gpu
+
+    rng = np.random.default_rng(6)
+    left = pd.DataFrame({"k": rng.integers(0, 40, 5000),
+                         "v1": rng.uniform(0, 1, 5000)})
+    right = pd.DataFrame({"k": rng.integers(0, 60, 30),
+                          "v2": rng.uniform(0, 1, 30)}).drop_duplicates("k")
+    lt = Table.from_pandas(left, device="cuda")
+    rt = Table.from_pandas(right, device="cuda")
+    got = gpu.join_local(lt, rt, ["k"], ["k"], how).to_pandas()
+    if how in ("semi", "anti"):
+        keys = set(right["k"])
+        exp = left[left["k"].isin(keys)] if how == "semi" else left[~left["k"].isin(keys)]
+        got = got.sort_values(["k", "v1"]).reset_index(drop=True)
+        exp = exp.sort_values(["k", "v1"]).reset_index(drop=True)
+        pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+        return
+    exp = left.merge(right, on="k", how=how)
+    sort_cols = ["k", "v1", "v2"]
+    got = got.sort_values(sort_cols, na_position="last").reset_index(drop=True)
+    exp = exp.sort_values(sort_cols, na_position="last").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_distinct_gpu():
+    from bodo_amd.core.table import Table
+    from bodo_amd.ops import gpu
+
+    rng = np.random.default_rng(7)
+    df = pd.DataFrame({"a": rng.integers(0, 30, 2000),
+                       "b": rng.integers(0, 3, 2000)})
+    t = Table.from_pandas(df, device="cuda")
+    got = gpu.distinct_local(t, ["a"]).to_pandas().reset_index(drop=True)
+    exp = df.drop_duplicates(subset=["a"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_taxi_q1_gpu_end_to_end(monkeypatch):
+    monkeypatch.setenv("BODO_AMD_DEVICE", "cuda")
+    import importlib
+
+    import bodo_amd.config as cfg
+
+    importlib.reload(cfg)
+    cfg.DEVICE = "cuda"
+    import bodo_amd.pandas as bpd
+    from tests.test_queries import make_taxi, nyc_taxi_q1
+
+    trips, weather = make_taxi(50000, 17)
+    got = nyc_taxi_q1(bpd, bpd.from_pandas(trips), bpd.from_pandas(weather))
+    got = got.to_pandas().reset_index(drop=True)
+    exp = nyc_taxi_q1(pd, trips.copy(), weather.copy()).reset_index(drop=True)
+    got["time_bucket"] = got["time_bucket"].astype(str)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    cfg.DEVICE = ""
+
+
+def test_tpch_q1_gpu_end_to_end(monkeypatch):
+    import bodo_amd.config as cfg
+
+    cfg.DEVICE = "cuda"
+    import bodo_amd.pandas as bpd
+    from tests.test_queries import make_lineitem, tpch_q1
+
+    li = make_lineitem(60000, 19)
+    got = tpch_q1(bpd, bpd.from_pandas(li)).to_pandas().reset_index(drop=True)
+    exp = tpch_q1(pd, li.copy()).reset_index(drop=True)
+    for c in ("L_RETURNFLAG", "L_LINESTATUS"):
+        got[c] = got[c].astype(str)
+        exp[c] = exp[c].astype(str)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    cfg.DEVICE = ""
