@@ -75,8 +75,12 @@ class Column:
         if self.device == device:
             return self
 
+        # non_blocking only for H2D: async D2H into pageable host memory can
+        # be read by numpy before the copy completes (observed on ROCm)
+        nb = device.type == "cuda"
+
         def mv(t):
-            return None if t is None else t.to(device, non_blocking=True)
+            return None if t is None else t.to(device, non_blocking=nb)
 
         return Column(
             self.dtype, mv(self.data), mv(self.mask), mv(self.offsets),
