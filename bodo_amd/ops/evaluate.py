@@ -71,11 +71,37 @@ class _Evaluator:
     def _decode_if_dict(self, c: Column) -> Column:
         return c
 
+    def _scalar_const(self, e: Expr):
+        """Return a python scalar for numeric/temporal Const exprs (avoids
+        materializing an n-element constant column), else None."""
+        if not isinstance(e, Const):
+            return None
+        dtype = e.dtype or infer_const_dtype(e.value)
+        if dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+            return None
+        return normalize_const(e.value, dtype), dtype
+
     def visit_BinOp(self, e: BinOp) -> Column:
+        rs = self._scalar_const(e.right)
+        ls = self._scalar_const(e.left)
+        if rs is not None:
+            return binary_arith_scalar(e.op, self.visit(e.left), rs[0], rs[1],
+                                       reflect=False)
+        if ls is not None:
+            return binary_arith_scalar(e.op, self.visit(e.right), ls[0], ls[1],
+                                       reflect=True)
         a, b = self.visit(e.left), self.visit(e.right)
         return binary_arith(e.op, a, b)
 
     def visit_Cmp(self, e: Cmp) -> Column:
+        rs = self._scalar_const(e.right)
+        if rs is not None:
+            return compare_scalar(e.op, self.visit(e.left), rs[0], rs[1])
+        ls = self._scalar_const(e.left)
+        if ls is not None:
+            flip = {"lt": "gt", "le": "ge", "gt": "lt", "ge": "le",
+                    "eq": "eq", "ne": "ne"}
+            return compare_scalar(flip[e.op], self.visit(e.right), ls[0], ls[1])
         a, b = self.visit(e.left), self.visit(e.right)
         return compare(e.op, a, b)
 
@@ -227,6 +253,56 @@ def binary_arith(op: str, a: Column, b: Column) -> Column:
     mask = combine_masks(a.mask, b.mask)
     dtype = bt.from_numpy_dtype(np.dtype(str(out.dtype).replace("torch.", "")))
     return Column(dtype, out, mask)
+
+
+def binary_arith_scalar(op: str, a: Column, v, vdtype: DType,
+                        reflect: bool) -> Column:
+    da = a.data
+    float_out = a.dtype.is_float or isinstance(v, float) or op in _ARITH_RESULT_FLOAT
+    if float_out and not da.dtype.is_floating_point:
+        da = da.to(torch.float64)
+    elif a.dtype.is_float and da.dtype == torch.float32 and isinstance(v, float):
+        pass
+    x, y = (v, da) if reflect else (da, v)
+    if op == "add":
+        out = x + y
+    elif op == "sub":
+        out = x - y
+    elif op == "mul":
+        out = x * y
+    elif op == "div":
+        out = x / y
+    elif op == "floordiv":
+        out = torch.div(da, v, rounding_mode="floor") if not reflect else \
+            torch.div(torch.full_like(da, v), da, rounding_mode="floor")
+    elif op == "mod":
+        out = torch.remainder(x, y) if not reflect else torch.remainder(
+            torch.full_like(da, v), da)
+    elif op == "pow":
+        out = torch.pow(x, y) if not reflect else torch.pow(
+            torch.full_like(da, v), da)
+    else:
+        raise NotImplementedError(op)
+    if not torch.is_tensor(out):  # pragma: no cover
+        out = torch.as_tensor(out)
+    dtype = bt.from_numpy_dtype(np.dtype(str(out.dtype).replace("torch.", "")))
+    return Column(dtype, out, a.mask)
+
+
+def compare_scalar(op: str, a: Column, v, vdtype: DType) -> Column:
+    if a.dtype.kind == TypeKind.DICT or a.dtype.kind == TypeKind.STRING:
+        # string comparisons handled via column path
+        b = Column.full_const(v, vdtype, len(a), a.device)
+        return compare(op, a, b)
+    da = a.data
+    if da.dtype == torch.bool and isinstance(v, (int, float)) and not isinstance(v, bool):
+        da = da.to(torch.int64)
+    out = getattr(torch, op)(da, v)
+    if a.mask is not None:
+        out = out & a.mask
+    if a.dtype.is_float:
+        out = out & ~torch.isnan(a.data)
+    return Column(bt.boolean, out)
 
 
 def compare(op: str, a: Column, b: Column) -> Column:
@@ -420,6 +496,31 @@ def udf_map(a: Column, func, na_action=None) -> Column:
                            dtype=torch.float64, device=a.device)
         return Column(bt.float64, lut[a.data.long()], a.mask)
     if a.dtype.is_integer or a.dtype.kind == TypeKind.BOOL:
+        # small dense domain: dense LUT indexed by value (no sort; one
+        # min/max reduction + one gather — the hot path for e.g. hour buckets)
+        if len(a):
+            lo = int(a.data.min().item())
+            hi = int(a.data.max().item())
+            if hi - lo < 65536:
+                import pyarrow as pa
+
+                vals = [func(v) for v in range(lo, hi + 1)]
+                pos = (a.data.long() - lo)
+                if all(isinstance(v, str) for v in vals):
+                    udict, inv = np.unique(np.array(vals, dtype=object),
+                                           return_inverse=True)
+                    inv_t = torch.from_numpy(inv.astype(np.int32)).to(a.device)
+                    codes = inv_t[pos]
+                    return Column(bt.dictionary, codes.to(torch.int32), a.mask,
+                                  dictionary=pa.array(list(udict),
+                                                      type=pa.large_string()),
+                                  length=len(a))
+                if all(v is None or isinstance(v, (int, float, bool))
+                       for v in vals):
+                    lut = torch.tensor(
+                        [np.nan if v is None else float(v) for v in vals],
+                        dtype=torch.float64, device=a.device)
+                    return Column(bt.float64, lut[pos], a.mask)
         # evaluate over unique values (bounded domain assumption checked)
         uniq = torch.unique(a.data)
         if uniq.numel() <= 1_000_000:
