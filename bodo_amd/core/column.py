@@ -25,7 +25,8 @@ def _np_from_tensor(t: torch.Tensor) -> np.ndarray:
 
 
 class Column:
-    __slots__ = ("dtype", "data", "mask", "offsets", "dictionary", "name", "_length")
+    __slots__ = ("dtype", "data", "mask", "offsets", "dictionary", "name",
+                 "_length", "val_range")
 
     def __init__(
         self,
@@ -41,6 +42,7 @@ class Column:
         self.mask = mask  # torch.bool, True = valid, or None (all valid)
         self.offsets = offsets  # STRING only: int64 (n+1)
         self.dictionary = dictionary  # DICT only: pa.StringArray (host)
+        self.val_range = None  # optional known (lo, hi) for int columns
         if length is not None:
             self._length = length
         elif dtype.kind == TypeKind.STRING:

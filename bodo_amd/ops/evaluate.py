@@ -406,7 +406,11 @@ def _dt_field_cpu(a: Column, fld: str) -> Column:
         out = getattr(idx, attr).to_numpy()
         dtype = _DT_OUT_TYPE[fld]
         out = out.astype(bt.numpy_storage_dtype(dtype))
-    return Column(dtype, torch.from_numpy(out), a.mask)
+    res = Column(dtype, torch.from_numpy(out), a.mask)
+    from .gpu import _DT_RANGE
+
+    res.val_range = _DT_RANGE.get(fld)
+    return res
 
 
 def cast_column(a: Column, to: DType) -> Column:
@@ -499,8 +503,11 @@ def udf_map(a: Column, func, na_action=None) -> Column:
         # small dense domain: dense LUT indexed by value (no sort; one
         # min/max reduction + one gather — the hot path for e.g. hour buckets)
         if len(a):
-            lo = int(a.data.min().item())
-            hi = int(a.data.max().item())
+            if a.val_range is not None:
+                lo, hi = a.val_range
+            else:
+                lo = int(a.data.min().item())
+                hi = int(a.data.max().item())
             if hi - lo < 65536:
                 import pyarrow as pa
 

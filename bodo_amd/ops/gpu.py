@@ -96,6 +96,12 @@ _DT_FIELD_ID = {
     "normalize": 10, "floor_day": 10,
 }
 
+_DT_RANGE = {
+    "month": (1, 12), "day": (1, 31), "hour": (0, 23), "minute": (0, 59),
+    "second": (0, 59), "dayofweek": (0, 6), "weekday": (0, 6),
+    "dayofyear": (1, 366), "quarter": (1, 4),
+}
+
 _DT_OUT_KIND = {
     "year": (0, bt.int16), "month": (0, bt.int8), "day": (0, bt.int8),
     "hour": (0, bt.int8), "minute": (0, bt.int8), "second": (0, bt.int8),
@@ -115,7 +121,9 @@ def dt_field(col: Column, fld: str) -> Column:
     store = bt.torch_storage_dtype(out_dtype)
     if res.dtype != store:
         res = res.to(store)
-    return Column(out_dtype, res, col.mask)
+    out = Column(out_dtype, res, col.mask)
+    out.val_range = _DT_RANGE.get(fld)
+    return out
 
 
 def gather_string(col: Column, idx: torch.Tensor) -> Column:

@@ -308,14 +308,44 @@ __global__ void gb_insert_kernel(const ColumnDesc* __restrict__ cols, int ncols,
   }
 }
 
+// two-pass group-id assignment: pass 1 counts occupied slots per block
+// (no atomics), host cumsums 2048 values, pass 2 assigns gids via an LDS
+// counter (one global atomic total: zero).  Replaces a single-global-counter
+// atomicAdd that measured 43% of NYC-taxi Q1 step time at 300M rows.
+__global__ void gb_count_kernel(const uint32_t* __restrict__ slots,
+                                int64_t chunk, int64_t cap,
+                                int64_t* __restrict__ block_counts) {
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t stop = min(start + chunk, cap);
+  int64_t local = 0;
+  for (int64_t s = start + threadIdx.x; s < stop; s += blockDim.x) {
+    local += (slots[s] != 0u);
+  }
+  __shared__ int64_t red[256];
+  red[threadIdx.x] = local;
+  __syncthreads();
+  for (int w = blockDim.x / 2; w > 0; w >>= 1) {
+    if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) block_counts[blockIdx.x] = red[0];
+}
+
 __global__ void gb_assign_gid_kernel(const uint32_t* __restrict__ slots,
-                                     int64_t cap, int32_t* __restrict__ slot_gid,
-                                     int64_t* __restrict__ uniq_rows,
-                                     int32_t* __restrict__ counter) {
-  GRID_STRIDE_LOOP(s, cap) {
+                                     int64_t chunk, int64_t cap,
+                                     const int64_t* __restrict__ block_base,
+                                     int32_t* __restrict__ slot_gid,
+                                     int64_t* __restrict__ uniq_rows) {
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t stop = min(start + chunk, cap);
+  __shared__ int lds_cnt;
+  if (threadIdx.x == 0) lds_cnt = 0;
+  __syncthreads();
+  int64_t base = block_base[blockIdx.x];
+  for (int64_t s = start + threadIdx.x; s < stop; s += blockDim.x) {
     uint32_t v = slots[s];
     if (v != 0u) {
-      int32_t gid = atomicAdd(counter, 1);
+      int32_t gid = (int32_t)(base + atomicAdd(&lds_cnt, 1));
       slot_gid[s] = gid;
       uniq_rows[gid] = (int64_t)v - 1;
     }
@@ -350,13 +380,23 @@ std::vector<torch::Tensor> groupby_build(
   auto slot_gid = torch::empty({cap}, torch::dtype(torch::kInt32).device(dev));
   auto uniq_rows = torch::empty({n > 0 ? n : 1},
                                 torch::dtype(torch::kInt64).device(dev));
-  auto counter = torch::zeros({1}, torch::dtype(torch::kInt32).device(dev));
-  hipLaunchKernelGGL(gb_assign_gid_kernel, dim3(grid_for(cap, block)),
-                     dim3(block), 0, cur_stream(),
-                     (const uint32_t*)slots.data_ptr(), cap,
+  int nblocks = 2048;
+  int64_t chunk = (cap + nblocks - 1) / nblocks;
+  auto block_counts = torch::zeros({nblocks},
+                                   torch::dtype(torch::kInt64).device(dev));
+  hipLaunchKernelGGL(gb_count_kernel, dim3(nblocks), dim3(block), 0,
+                     cur_stream(), (const uint32_t*)slots.data_ptr(), chunk,
+                     cap, (int64_t*)block_counts.data_ptr());
+  CHECK_HIP(hipGetLastError());
+  auto block_base = torch::zeros({nblocks}, torch::dtype(torch::kInt64).device(dev));
+  auto bb_tail = block_base.slice(0, 1, nblocks);
+  at::cumsum_out(bb_tail, block_counts.slice(0, 0, nblocks - 1), 0);
+  int64_t ngroups = block_counts.sum().item<int64_t>();
+  hipLaunchKernelGGL(gb_assign_gid_kernel, dim3(nblocks), dim3(block), 0,
+                     cur_stream(), (const uint32_t*)slots.data_ptr(), chunk,
+                     cap, (const int64_t*)block_base.data_ptr(),
                      (int32_t*)slot_gid.data_ptr(),
-                     (int64_t*)uniq_rows.data_ptr(),
-                     (int32_t*)counter.data_ptr());
+                     (int64_t*)uniq_rows.data_ptr());
   CHECK_HIP(hipGetLastError());
   auto row_gid = torch::empty({n}, torch::dtype(torch::kInt32).device(dev));
   hipLaunchKernelGGL(gb_rowgid_kernel, dim3(grid_for(n, block)), dim3(block),
@@ -364,7 +404,6 @@ std::vector<torch::Tensor> groupby_build(
                      (const int32_t*)slot_gid.data_ptr(),
                      (int32_t*)row_gid.data_ptr(), n);
   CHECK_HIP(hipGetLastError());
-  int32_t ngroups = counter.cpu().item<int32_t>();
   return {row_gid, uniq_rows.slice(0, 0, ngroups)};
 }
 
