@@ -129,6 +129,10 @@ class Column:
                 offsets=torch.from_numpy(np.ascontiguousarray(offsets)).to(device),
                 length=len(arr),
             )
+        if pa.types.is_null(t):
+            # typeless all-null column: represent as float64 NaN
+            data = torch.full((len(arr),), float("nan"), dtype=torch.float64)
+            return Column(bt.float64, data.to(device), length=len(arr))
         if pa.types.is_timestamp(t):
             arr = arr.cast(pa.timestamp("ns"))
             dtype = bt.timestamp_ns

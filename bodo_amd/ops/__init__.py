@@ -27,6 +27,21 @@ from .evaluate import eval_expr, eval_filter, project  # noqa: F401
 
 def gather(col: Column, idx: torch.Tensor) -> Column:
     n = int(idx.shape[0])
+    if len(col) == 0 and n > 0:
+        # gathering from an empty column only happens for null-padded rows
+        # (outer-join unmatched side): produce an all-null column
+        if col.dtype.kind == TypeKind.STRING:
+            return Column(col.dtype,
+                          torch.zeros(0, dtype=torch.uint8, device=idx.device),
+                          torch.zeros(n, dtype=torch.bool, device=idx.device),
+                          offsets=torch.zeros(n + 1, dtype=torch.int64,
+                                              device=idx.device), length=n)
+        from ..core import types as _bt
+
+        store = _bt.torch_storage_dtype(col.dtype)
+        data = torch.zeros(n, dtype=store, device=idx.device)
+        mask = torch.zeros(n, dtype=torch.bool, device=idx.device)
+        return Column(col.dtype, data, mask, dictionary=col.dictionary, length=n)
     if col.dtype.kind == TypeKind.STRING:
         if col.is_cuda:
             from . import gpu

@@ -21,7 +21,7 @@ from ..core.table import Table
 from ..core.types import DType, TypeKind
 from ..plan.expr import (
     BinOp, BoolOp, Case, Cast, ColRef, Cmp, Const, DtField, Expr, IsIn,
-    IsNull, Not, StrOp, UdfMap,
+    IsNull, Not, RoundExpr, SemiJoinIn, StrOp, UdfMap,
 )
 
 
@@ -164,6 +164,41 @@ class _Evaluator:
     def visit_UdfMap(self, e: UdfMap) -> Column:
         a = self.visit(e.operand)
         return udf_map(a, e.func, e.na_action)
+
+    def visit_RoundExpr(self, e: RoundExpr) -> Column:
+        a = self.visit(e.operand)
+        if not a.dtype.is_float:
+            return a
+        scale = 10.0 ** e.decimals
+        out = torch.round(a.data * scale) / scale
+        return Column(a.dtype, out, a.mask)
+
+    def visit_SemiJoinIn(self, e: SemiJoinIn) -> Column:
+        # general-expression fallback: membership against the other side's
+        # allgathered distinct values (the Filter(SemiJoinIn) form is
+        # rewritten into a SEMI join by the frontend instead)
+        from ..engine import executor as _ex
+        from ..parallel import comm as _comm
+        from ..plan import nodes as _pn
+
+        plan = _pn.Distinct(
+            _pn.Projection(e.other_plan, ("v",), (e.other_expr,)), ("v",))
+        vals_tbl = _ex.execute(plan, _ex.ExecutionContext())
+        full = _comm.allgather_table(vals_tbl)
+        a = self.visit(e.operand)
+        vcol = full.column("v").to_device(a.device)
+        if a.dtype.kind in (TypeKind.STRING, TypeKind.DICT) or                 vcol.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+            vals = tuple(full.column("v").to_pandas().dropna().tolist())
+            return self.visit_IsIn(IsIn(e.operand, vals))
+        test = vcol.data
+        data = a.data
+        if data.dtype != test.dtype:
+            t = torch.promote_types(data.dtype, test.dtype)
+            data, test = data.to(t), test.to(t)
+        out = torch.isin(data, test)
+        if a.mask is not None:
+            out = out & a.mask
+        return Column(bt.boolean, out)
 
     def visit_Case(self, e: Case) -> Column:
         other = self.visit(e.otherwise)
@@ -448,13 +483,16 @@ def str_op(a: Column, op: str, args) -> Column:
                  "strip": pc.utf8_trim_whitespace, "title": pc.utf8_title,
                  "capitalize": pc.utf8_capitalize}[op]
             return Column(a.dtype, a.data, a.mask, dictionary=f(d), length=len(a))
-        if op in ("contains", "startswith", "endswith", "len"):
+        if op in ("contains", "contains_re", "match", "startswith",
+                  "endswith", "len"):
             if op == "len":
                 lut_np = pc.utf8_length(d).to_numpy(zero_copy_only=False).astype(np.int64)
                 ret = bt.int64
             else:
                 pat = args[0]
                 f = {"contains": lambda x: pc.match_substring(x, pat),
+                     "contains_re": lambda x: pc.match_substring_regex(x, pat),
+                     "match": lambda x: pc.match_substring_regex(x, "^" + pat),
                      "startswith": lambda x: pc.starts_with(x, pat),
                      "endswith": lambda x: pc.ends_with(x, pat)}[op]
                 lut_np = f(d).to_numpy(zero_copy_only=False).astype(bool)
@@ -471,6 +509,8 @@ def str_op(a: Column, op: str, args) -> Column:
         "strip": lambda x: pc.utf8_trim_whitespace(x),
         "len": lambda x: pc.utf8_length(x),
         "contains": lambda x: pc.match_substring(x, args[0]),
+        "contains_re": lambda x: pc.match_substring_regex(x, args[0]),
+        "match": lambda x: pc.match_substring_regex(x, "^" + args[0]),
         "startswith": lambda x: pc.starts_with(x, args[0]),
         "endswith": lambda x: pc.ends_with(x, args[0]),
         "title": lambda x: pc.utf8_title(x),

@@ -24,8 +24,15 @@ NA = _pd.NA
 NaT = _pd.NaT
 isna = _pd.isna
 notna = _pd.notna
-DataFrame = BodoDataFrame
 Series = BodoSeries
+
+
+def DataFrame(data=None, *args, **kwargs):
+    """Constructor-compatible factory: builds a BodoDataFrame from pandas
+    DataFrame constructor arguments (reference: bodo.pandas.DataFrame)."""
+    if isinstance(data, BodoDataFrame):
+        return data.copy()
+    return from_pandas_df(_pd.DataFrame(data, *args, **kwargs))
 
 
 def read_parquet(path, columns=None, **kwargs) -> BodoDataFrame:

@@ -20,6 +20,12 @@ from ..plan.expr import BoolOp, ColRef, Expr, Not
 
 
 class BodoDataFrame:
+    def __setattr__(self, name, value):
+        if name == "columns":
+            self._set_columns(value)
+            return
+        object.__setattr__(self, name, value)
+
     def __init__(self, plan: pn.PlanNode, columns: Sequence[str]):
         object.__setattr__(self, "_plan", plan)
         object.__setattr__(self, "_columns", list(columns))
@@ -61,6 +67,8 @@ class BodoDataFrame:
     def columns(self):
         return pd.Index(self._columns)
 
+    # (setter defined below via _set_columns)
+
     @property
     def shape(self):
         n = len(self)
@@ -99,11 +107,40 @@ class BodoDataFrame:
             exprs = tuple(ColRef(k) for k in key)
             return BodoDataFrame(pn.Projection(self._plan, tuple(key), exprs), key)
         if isinstance(key, BodoSeries):
+            from ..plan.expr import Not as _Not
+            from ..plan.expr import SemiJoinIn as _SJ
+
+            e = key._expr
+            if isinstance(e, _SJ) and isinstance(e.operand, ColRef):
+                return self._semi_filter(e, anti=False)
+            if isinstance(e, _Not) and isinstance(e.operand, _SJ)                     and isinstance(e.operand.operand, ColRef):
+                return self._semi_filter(e.operand, anti=True)
             return BodoDataFrame(pn.Filter(self._plan, key._expr), self._columns)
         if isinstance(key, pd.Series) and key.dtype == bool:
             # host boolean mask: materialize path
             return self._fallback("__getitem__", key)
         raise TypeError(f"unsupported key {type(key)}")
+
+    def _semi_filter(self, e, anti: bool):
+        other = pn.Distinct(
+            pn.Projection(e.other_plan, ("__in_v",), (e.other_expr,)),
+            ("__in_v",))
+        plan = pn.Join(self._plan, other, (e.operand.name,), ("__in_v",),
+                       "anti" if anti else "semi")
+        return BodoDataFrame(plan, self._columns)
+
+    @property
+    def loc(self):
+        return _LocIndexer(self)
+
+    def _set_columns(self, new_cols):
+        new_cols = list(new_cols)
+        assert len(new_cols) == len(self._columns)
+        exprs = tuple(ColRef(c) for c in self._columns)
+        plan = pn.Projection(self._plan, tuple(new_cols), exprs)
+        object.__setattr__(self, "_plan", plan)
+        object.__setattr__(self, "_columns", new_cols)
+        object.__setattr__(self, "_result", None)
 
     def __setitem__(self, key: str, value):
         from .series import BodoSeries
@@ -381,3 +418,31 @@ def from_pandas_df(df: pd.DataFrame) -> BodoDataFrame:
     key = ex.register_object(df.reset_index(drop=True))
     plan = pn.PandasScan(key, tuple(df.columns), distributed=False)
     return BodoDataFrame(plan, list(df.columns))
+
+
+class _LocIndexer:
+    """df.loc[:, [cols]] and df.loc[boolean_series] support."""
+
+    def __init__(self, frame: BodoDataFrame):
+        self._frame = frame
+
+    def __getitem__(self, key):
+        from .series import BodoSeries
+
+        if isinstance(key, tuple) and len(key) == 2:
+            rows, cols = key
+            out = self._frame
+            if isinstance(rows, BodoSeries):
+                out = out[rows]
+            elif not (isinstance(rows, slice) and rows == slice(None)):
+                raise NotImplementedError("loc row selection")
+            if isinstance(cols, list):
+                return out[cols]
+            if isinstance(cols, str):
+                return out[cols]
+            if isinstance(cols, slice) and cols == slice(None):
+                return out
+            raise NotImplementedError("loc column selection")
+        if isinstance(key, BodoSeries):
+            return self._frame[key]
+        raise NotImplementedError("loc")

@@ -158,7 +158,8 @@ class SeriesGroupBy:
         if isinstance(out, _IndexedAggResult):
             pdf = out._frame.to_pandas().set_index(self._keys)[self._col]
             return pdf
-        return out[self._col] if hasattr(out, "__getitem__") else out
+        # as_index=False: pandas returns a DataFrame of keys + the column
+        return out
 
     def sum(self):
         return self._agg1("sum")

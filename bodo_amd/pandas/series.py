@@ -73,6 +73,16 @@ class BodoSeries:
 
     # ------------------------------------------------------------------
     def isin(self, values) -> "BodoSeries":
+        if isinstance(values, BodoSeries):
+            # semi-join filter: df[a.isin(other_col)] becomes a SEMI join;
+            # in general expression positions it evaluates via the other
+            # side's distinct values (reference analog: runtime join filters)
+            from ..plan.expr import SemiJoinIn
+
+            return self._wrap(
+                SemiJoinIn(self._expr, values._plan, values._expr), None)
+        if isinstance(values, pd.Series):
+            values = values.tolist()
         return self._wrap(IsIn(self._expr, tuple(values)), None)
 
     def isna(self) -> "BodoSeries":
@@ -108,7 +118,9 @@ class BodoSeries:
                  (BinOp("sub", zero, self._expr),), self._expr), None)
 
     def round(self, decimals=0):
-        return self.map(lambda v: None if pd.isna(v) else round(v, decimals))
+        from ..plan.expr import RoundExpr
+
+        return self._wrap(RoundExpr(self._expr, decimals), None)
 
     def where(self, cond, other) -> "BodoSeries":
         ce = cond._expr if isinstance(cond, BodoSeries) else as_expr(cond)
@@ -316,7 +328,12 @@ class _StrAccessor:
         return self._f("len")
 
     def contains(self, pat, regex=True, **kw):
+        if regex and any(ch in pat for ch in ".*+?[](){}|\\^$"):
+            return self._f("contains_re", pat)
         return self._f("contains", pat)
+
+    def match(self, pat):
+        return self._f("match", pat)
 
     def startswith(self, pat):
         return self._f("startswith", pat)

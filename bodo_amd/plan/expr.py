@@ -219,6 +219,36 @@ class Case(Expr):
         return Case(tuple(ch[:k]), tuple(ch[k:2 * k]), ch[2 * k])
 
 
+@dataclass(frozen=True)
+class SemiJoinIn(Expr):
+    """col.isin(other_series): a semi-join filter.  The frontend rewrites
+    Filter(SemiJoinIn) / Filter(Not(SemiJoinIn)) into SEMI/ANTI joins; in
+    general expression positions it evaluates against the other side's
+    distinct values (allgathered)."""
+
+    operand: Expr
+    other_plan: object = field(compare=False, default=None)
+    other_expr: Expr = None
+
+    def children(self):
+        return (self.operand,)
+
+    def with_children(self, *ch):
+        return SemiJoinIn(ch[0], self.other_plan, self.other_expr)
+
+
+@dataclass(frozen=True)
+class RoundExpr(Expr):
+    operand: Expr
+    decimals: int = 0
+
+    def children(self):
+        return (self.operand,)
+
+    def with_children(self, *ch):
+        return RoundExpr(ch[0], self.decimals)
+
+
 def as_expr(v) -> Expr:
     if isinstance(v, Expr):
         return v
