@@ -129,6 +129,15 @@ def prune_columns(node: pn.PlanNode, required: Optional[Set[str]]) -> pn.PlanNod
         else:
             cols = tuple(sorted(required))
         return pn.CsvScan(node.path, node.options, cols, node.schema_names)
+    if isinstance(node, pn.PandasScan):
+        if required is None or not node.names:
+            return node
+        keep = [c for c in node.names if c in required]
+        if len(keep) == len(node.names) or not keep:
+            return node
+        from ..plan.expr import ColRef as _CR
+
+        return pn.Projection(node, tuple(keep), tuple(_CR(c) for c in keep))
     if isinstance(node, pn.Projection):
         if required is not None:
             keep = [(n, e) for n, e in zip(node.names, node.exprs) if n in required]
@@ -176,9 +185,13 @@ def prune_columns(node: pn.PlanNode, required: Optional[Set[str]]) -> pn.PlanNod
                     lreq.add(r)
                 if r in sr:
                     rreq.add(r)
+                # a suffixed requirement needs the base column on BOTH sides
+                # so the runtime overlap (and hence the suffix) is preserved
                 if base_l and base_l in sl and base_l in sr:
                     lreq.add(base_l)
+                    rreq.add(base_l)
                 if base_r and base_r in sr and base_r in sl:
+                    lreq.add(base_r)
                     rreq.add(base_r)
         return pn.Join(
             prune_columns(node.left, lreq), prune_columns(node.right, rreq),

@@ -47,10 +47,19 @@ def gather(col: Column, idx: torch.Tensor) -> Column:
             from . import gpu
 
             return gpu.gather_string(col, idx)
-        return _gather_string_torch(col, idx)
+        return _gather_string_arrow(col, idx)
     data = col.data[idx]
     mask = col.mask[idx] if col.mask is not None else None
     return Column(col.dtype, data, mask, dictionary=col.dictionary, length=n)
+
+
+def _gather_string_arrow(col: Column, idx: torch.Tensor) -> Column:
+    """CPU string gather through Arrow take (C++ kernel)."""
+    import pyarrow.compute as pc
+
+    arr = col.to_arrow()
+    taken = pc.take(arr, idx.numpy())
+    return Column.from_arrow(taken, col.device)
 
 
 def _gather_string_torch(col: Column, idx: torch.Tensor) -> Column:

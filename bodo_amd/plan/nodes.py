@@ -234,6 +234,25 @@ class Join(PlanNode):
     def with_children(self, *ch):
         return Join(ch[0], ch[1], self.left_on, self.right_on, self.how, self.suffixes)
 
+    def out_columns(self):
+        if self.how in ("semi", "anti"):
+            return self.left.out_columns()
+        lcols = self.left.out_columns()
+        rcols = self.right.out_columns()
+        if lcols is None or rcols is None:
+            return None
+        shared = {k for k, rk in zip(self.left_on, self.right_on) if k == rk}
+        lset, rset = set(lcols), set(rcols)
+        out = []
+        for c in lcols:
+            out.append(c + self.suffixes[0]
+                       if (c in rset and c not in shared) else c)
+        for c in rcols:
+            if c in shared:
+                continue
+            out.append(c + self.suffixes[1] if c in lset else c)
+        return out
+
 
 @dataclass(frozen=True)
 class Union(PlanNode):
