@@ -82,6 +82,7 @@ class Table:
     @staticmethod
     def from_pandas(df: pd.DataFrame, device="cpu") -> "Table":
         tbl = pa.Table.from_pandas(df, preserve_index=False)
+        tbl = dict_encode_strings(tbl)
         return Table.from_arrow(tbl, device)
 
     def to_pandas(self) -> pd.DataFrame:
@@ -115,3 +116,29 @@ class Table:
 
     def __repr__(self) -> str:  # pragma: no cover
         return f"Table({len(self)} rows x {self.num_columns} cols, device={self.device})"
+
+
+def dict_encode_strings(tbl: pa.Table, threshold: float = 0.5,
+                        sample: int = 8192) -> pa.Table:
+    """Dictionary-encode low-cardinality string columns so filters/joins/
+    groupbys run on int32 codes on device (reference: dict-encoded string
+    arrays, bodo/libs/dict_arr_ext.py)."""
+    import pyarrow.compute as pc
+
+    new_cols = []
+    changed = False
+    for i, f in enumerate(tbl.schema):
+        col = tbl.column(i)
+        if pa.types.is_string(f.type) or pa.types.is_large_string(f.type):
+            head = col.slice(0, min(sample, len(col)))
+            try:
+                nuniq = len(pc.unique(head.combine_chunks()))
+            except Exception:
+                nuniq = len(head)
+            if len(head) > 0 and nuniq <= max(1, int(len(head) * threshold)):
+                col = pc.dictionary_encode(col.combine_chunks())
+                changed = True
+        new_cols.append(col)
+    if not changed:
+        return tbl
+    return pa.table(dict(zip(tbl.column_names, new_cols)))

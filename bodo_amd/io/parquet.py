@@ -98,7 +98,9 @@ def read_shard(path: str, columns: Optional[Sequence[str]],
             schema = pa.schema([schema.field(c) for c in cols])
         out = schema.empty_table()
     # auto dict-encode low-cardinality string columns for device residency
-    out = _dict_encode_strings(out)
+    from ..core.table import dict_encode_strings
+
+    out = dict_encode_strings(out)
     return Table.from_arrow(out, ctx.device)
 
 
