@@ -71,13 +71,13 @@ class _Evaluator:
     def _decode_if_dict(self, c: Column) -> Column:
         return c
 
-    def _scalar_const(self, e: Expr):
+    def _scalar_const(self, e: Expr, allow_str: bool = False):
         """Return a python scalar for numeric/temporal Const exprs (avoids
         materializing an n-element constant column), else None."""
         if not isinstance(e, Const):
             return None
         dtype = e.dtype or infer_const_dtype(e.value)
-        if dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+        if dtype.kind in (TypeKind.STRING, TypeKind.DICT) and not allow_str:
             return None
         return normalize_const(e.value, dtype), dtype
 
@@ -94,10 +94,10 @@ class _Evaluator:
         return binary_arith(e.op, a, b)
 
     def visit_Cmp(self, e: Cmp) -> Column:
-        rs = self._scalar_const(e.right)
+        rs = self._scalar_const(e.right, allow_str=True)
         if rs is not None:
             return compare_scalar(e.op, self.visit(e.left), rs[0], rs[1])
-        ls = self._scalar_const(e.left)
+        ls = self._scalar_const(e.left, allow_str=True)
         if ls is not None:
             flip = {"lt": "gt", "le": "ge", "gt": "lt", "ge": "le",
                     "eq": "eq", "ne": "ne"}
@@ -325,8 +325,30 @@ def binary_arith_scalar(op: str, a: Column, v, vdtype: DType,
 
 
 def compare_scalar(op: str, a: Column, v, vdtype: DType) -> Column:
-    if a.dtype.kind == TypeKind.DICT or a.dtype.kind == TypeKind.STRING:
-        # string comparisons handled via column path
+    if a.dtype.kind == TypeKind.DICT and isinstance(v, str):
+        # compare the (small) dictionary on host, map through a code LUT
+        import operator as _op
+
+        f = {"lt": _op.lt, "le": _op.le, "gt": _op.gt, "ge": _op.ge,
+             "eq": _op.eq, "ne": _op.ne}[op]
+        dvals = a.dictionary.to_pylist()
+        hit = np.array([x is not None and f(x, v) for x in dvals], dtype=bool)
+        lut = torch.from_numpy(hit).to(a.device)
+        out = lut[a.data.long()]
+        if a.mask is not None:
+            out = out & a.mask
+        return Column(bt.boolean, out)
+    if a.dtype.kind == TypeKind.STRING and isinstance(v, str):
+        import pyarrow.compute as pc
+
+        arr = a.to_device("cpu").to_arrow()
+        fmap = {"eq": pc.equal, "ne": pc.not_equal, "lt": pc.less,
+                "le": pc.less_equal, "gt": pc.greater, "ge": pc.greater_equal}
+        res = fmap[op](arr, v)
+        out_np = res.to_numpy(zero_copy_only=False)
+        out_np = np.where(pd.isna(out_np), False, out_np).astype(bool)
+        return Column(bt.boolean, torch.from_numpy(out_np).to(a.device))
+    if a.dtype.kind in (TypeKind.DICT, TypeKind.STRING):
         b = Column.full_const(v, vdtype, len(a), a.device)
         return compare(op, a, b)
     da = a.data

@@ -481,6 +481,83 @@ __global__ void agg_update_kernel(const ColumnDesc col,
   }
 }
 
+// LDS pre-aggregation path for low-cardinality groupbys (TPC-H Q1 shape):
+// per-block accumulators in LDS, one global atomic per (block, group).
+// Replaces 60M global atomics on 6 slots (measured 243 ms/call) with
+// 60M LDS atomics + ~2048*6 global ones.
+#define LDS_AGG_MAX_GROUPS 2048
+
+__global__ void agg_update_lds_kernel(const ColumnDesc col,
+                                      const int32_t* __restrict__ row_gid,
+                                      int op, int ngroups,
+                                      void* __restrict__ acc,
+                                      int64_t* __restrict__ cnt, int64_t n,
+                                      double init_f, int64_t init_i) {
+  __shared__ double lacc[LDS_AGG_MAX_GROUPS];
+  __shared__ long long lcnt[LDS_AGG_MAX_GROUPS];
+  bool f64_acc = (op == AGG_SUM_F64 || op == AGG_MIN_F64 || op == AGG_MAX_F64);
+  for (int g = threadIdx.x; g < ngroups; g += blockDim.x) {
+    lacc[g] = f64_acc ? init_f : __longlong_as_double(init_i);
+    lcnt[g] = 0;
+  }
+  __syncthreads();
+  GRID_STRIDE_LOOP(i, n) {
+    int32_t g = row_gid[i];
+    if (op == AGG_SIZE) {
+      atomicAdd((unsigned long long*)&lcnt[g], 1ull);
+      continue;
+    }
+    bool valid = is_valid_at(col, i);
+    double dv = 0.0;
+    int64_t iv = 0;
+    if (valid) {
+      switch (col.dtype) {
+        case BT_INT8: iv = ((const int8_t*)col.data)[i]; dv = (double)iv; break;
+        case BT_UINT8: case BT_BOOL: iv = ((const uint8_t*)col.data)[i]; dv = (double)iv; break;
+        case BT_INT16: case BT_UINT16: iv = ((const int16_t*)col.data)[i]; dv = (double)iv; break;
+        case BT_INT32: case BT_UINT32: case BT_DATE32: case BT_DICT:
+          iv = ((const int32_t*)col.data)[i]; dv = (double)iv; break;
+        case BT_INT64: case BT_UINT64: case BT_TIMESTAMP_NS:
+          iv = ((const int64_t*)col.data)[i]; dv = (double)iv; break;
+        case BT_FLOAT32: { float f = ((const float*)col.data)[i]; valid = !(f != f); dv = (double)f; iv = (int64_t)f; break; }
+        case BT_FLOAT64: { double f = ((const double*)col.data)[i]; valid = !(f != f); dv = f; iv = (int64_t)f; break; }
+      }
+    }
+    if (!valid) continue;
+    switch (op) {
+      case AGG_SUM_F64: atomicAdd(&lacc[g], dv); break;
+      case AGG_SUM_I64: atomicAdd((unsigned long long*)&lacc[g], (unsigned long long)iv); break;
+      case AGG_COUNT: break;
+      case AGG_MIN_F64: atomic_min_f64(&lacc[g], dv); break;
+      case AGG_MAX_F64: atomic_max_f64(&lacc[g], dv); break;
+      case AGG_MIN_I64: atomic_min_i64((int64_t*)&lacc[g], iv); break;
+      case AGG_MAX_I64: atomic_max_i64((int64_t*)&lacc[g], iv); break;
+      case AGG_FIRST_ROW: atomic_min_i64((int64_t*)&lacc[g], (int64_t)i); break;
+      case AGG_LAST_ROW: atomic_max_i64((int64_t*)&lacc[g], (int64_t)i); break;
+    }
+    atomicAdd((unsigned long long*)&lcnt[g], 1ull);
+  }
+  __syncthreads();
+  // merge block partials into global accumulators
+  for (int g = threadIdx.x; g < ngroups; g += blockDim.x) {
+    long long c = lcnt[g];
+    if (cnt != nullptr && c) atomicAdd((unsigned long long*)&cnt[g], (unsigned long long)c);
+    if (op == AGG_SIZE || op == AGG_COUNT) continue;
+    if (c == 0) continue;
+    switch (op) {
+      case AGG_SUM_F64: atomicAdd((double*)acc + g, lacc[g]); break;
+      case AGG_SUM_I64: atomicAdd((unsigned long long*)acc + g,
+                                  (unsigned long long)__double_as_longlong(lacc[g])); break;
+      case AGG_MIN_F64: atomic_min_f64((double*)acc + g, lacc[g]); break;
+      case AGG_MAX_F64: atomic_max_f64((double*)acc + g, lacc[g]); break;
+      case AGG_MIN_I64: atomic_min_i64((int64_t*)acc + g, __double_as_longlong(lacc[g])); break;
+      case AGG_MAX_I64: atomic_max_i64((int64_t*)acc + g, __double_as_longlong(lacc[g])); break;
+      case AGG_FIRST_ROW: atomic_min_i64((int64_t*)acc + g, __double_as_longlong(lacc[g])); break;
+      case AGG_LAST_ROW: atomic_max_i64((int64_t*)acc + g, __double_as_longlong(lacc[g])); break;
+    }
+  }
+}
+
 std::vector<torch::Tensor> agg_update(
     torch::Tensor data, c10::optional<torch::Tensor> mask,
     c10::optional<torch::Tensor> offsets, int64_t dtype,
@@ -504,10 +581,19 @@ std::vector<torch::Tensor> agg_update(
   }
   ColumnDesc col = make_desc(data, mask, offsets, c10::nullopt, dtype, n);
   int block = 256;
-  hipLaunchKernelGGL(agg_update_kernel, dim3(grid_for(n, block)), dim3(block),
-                     0, cur_stream(), col,
-                     (const int32_t*)row_gid.data_ptr(), (int)op,
-                     acc.data_ptr(), cnt_ptr, n);
+  bool lds_ok = ngroups <= LDS_AGG_MAX_GROUPS && op != AGG_PROD_F64;
+  if (lds_ok) {
+    hipLaunchKernelGGL(agg_update_lds_kernel, dim3(grid_for(n, block)),
+                       dim3(block), 0, cur_stream(), col,
+                       (const int32_t*)row_gid.data_ptr(), (int)op,
+                       (int)ngroups, acc.data_ptr(), cnt_ptr, n,
+                       init_f, init_i);
+  } else {
+    hipLaunchKernelGGL(agg_update_kernel, dim3(grid_for(n, block)), dim3(block),
+                       0, cur_stream(), col,
+                       (const int32_t*)row_gid.data_ptr(), (int)op,
+                       acc.data_ptr(), cnt_ptr, n);
+  }
   CHECK_HIP(hipGetLastError());
   if (cnt.defined()) return {acc, cnt};
   return {acc};
