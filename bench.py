@@ -147,7 +147,7 @@ def main():
 
     def one_step():
         shard = run_query(bpd, trips, weather)
-        return len(shard)
+        return sum(comm.allgather_obj(len(shard)))
 
     for _ in range(args.warmup):
         one_step()

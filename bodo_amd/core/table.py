@@ -122,23 +122,38 @@ def dict_encode_strings(tbl: pa.Table, threshold: float = 0.5,
                         sample: int = 8192) -> pa.Table:
     """Dictionary-encode low-cardinality string columns so filters/joins/
     groupbys run on int32 codes on device (reference: dict-encoded string
-    arrays, bodo/libs/dict_arr_ext.py)."""
+    arrays, bodo/libs/dict_arr_ext.py).
+
+    SPMD-safe: each rank votes on its shard sample and the decision is the
+    AND across ranks (a per-rank decision would give different column kinds
+    per rank and deadlock the shuffle collective sequence)."""
     import pyarrow.compute as pc
 
-    new_cols = []
-    changed = False
+    want = []
     for i, f in enumerate(tbl.schema):
         col = tbl.column(i)
+        w = False
         if pa.types.is_string(f.type) or pa.types.is_large_string(f.type):
             head = col.slice(0, min(sample, len(col)))
             try:
                 nuniq = len(pc.unique(head.combine_chunks()))
             except Exception:
                 nuniq = len(head)
-            if len(head) > 0 and nuniq <= max(1, int(len(head) * threshold)):
-                col = pc.dictionary_encode(col.combine_chunks())
-                changed = True
-        new_cols.append(col)
-    if not changed:
+            # empty shards vote yes so non-empty shards decide
+            w = len(head) == 0 or nuniq <= max(1, int(len(head) * threshold))
+        want.append(w)
+    from ..parallel import comm
+
+    if comm.initialized() and comm.get_world_size() > 1:
+        votes = comm.allgather_obj(want)
+        want = [all(v[i] for v in votes) for i in range(len(want))]
+    if not any(want):
         return tbl
+    new_cols = []
+    for i, f in enumerate(tbl.schema):
+        col = tbl.column(i)
+        if want[i] and (pa.types.is_string(f.type)
+                        or pa.types.is_large_string(f.type)):
+            col = pc.dictionary_encode(col.combine_chunks())
+        new_cols.append(col)
     return pa.table(dict(zip(tbl.column_names, new_cols)))

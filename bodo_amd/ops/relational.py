@@ -41,14 +41,6 @@ def groupby_local(tbl: Table, keys: Sequence[str],
 def _groupby_pandas(tbl: Table, keys, aggs, dropna) -> Table:
     need = list(dict.fromkeys(list(keys) + [a[1] for a in aggs if a[1]]))
     df = tbl.select([c for c in need if tbl.has_column(c)]).to_pandas()
-    if len(df) == 0:
-        # empty shard: produce empty frame with right columns
-        out = pd.DataFrame({k: df[k] if k in df else pd.Series([], dtype="float64")
-                            for k in keys})
-        for out_name, in_name, func in aggs:
-            dt = "int64" if func in ("count", "size", "nunique") else "float64"
-            out[out_name] = pd.Series([], dtype=dt)
-        return Table.from_pandas(out, tbl.device)
     gb = df.groupby(list(keys), dropna=dropna, sort=False, observed=True)
     named = {}
     for out_name, in_name, func in aggs:
@@ -57,7 +49,65 @@ def _groupby_pandas(tbl: Table, keys, aggs, dropna) -> Table:
         else:
             named[out_name] = pd.NamedAgg(column=in_name, aggfunc=func)
     res = gb.agg(**named).reset_index()
-    return Table.from_pandas(res, tbl.device)
+    # SPMD type stability: an empty shard must produce the same column kinds
+    # as non-empty ranks (pandas infers `null` for empty object columns and
+    # the shuffle collective sequences would then diverge) — enforce the
+    # input column's type for keys and string-typed agg outputs.
+    import pyarrow as pa
+
+    cols, names = [], []
+    for k in keys:
+        cols.append(_col_from_pandas_typed(res[k], tbl.column(k), tbl.device))
+        names.append(k)
+    for out_name, in_name, func in aggs:
+        src = tbl.column(in_name) if in_name and tbl.has_column(in_name) else None
+        if (src is not None and src.dtype.is_string_like
+                and func in ("min", "max", "first", "last")):
+            cols.append(_col_from_pandas_typed(res[out_name], src, tbl.device))
+        else:
+            arr = pa.Array.from_pandas(res[out_name])
+            if pa.types.is_null(arr.type):
+                arr = arr.cast(pa.float64())
+            cols.append(Column.from_arrow(arr, tbl.device))
+        names.append(out_name)
+    return Table(names, cols, len(res))
+
+
+def _col_from_pandas_typed(ser: pd.Series, like: Column, device) -> Column:
+    """Convert a pandas Series enforcing the arrow type of `like`."""
+    import pyarrow as pa
+
+    k = like.dtype.kind
+    if k == TypeKind.DICT:
+        arr = pa.Array.from_pandas(ser)
+        if not pa.types.is_dictionary(arr.type):
+            arr = arr.cast(pa.large_string()).dictionary_encode()
+        return Column.from_arrow(arr, device)
+    target = _PA_TYPE_OF.get(k)
+    arr = pa.Array.from_pandas(ser, type=target)
+    return Column.from_arrow(arr, device)
+
+
+_PA_TYPE_OF = {}
+
+
+def _init_pa_types():
+    import pyarrow as pa
+
+    global _PA_TYPE_OF
+    _PA_TYPE_OF = {
+        TypeKind.INT8: pa.int8(), TypeKind.INT16: pa.int16(),
+        TypeKind.INT32: pa.int32(), TypeKind.INT64: pa.int64(),
+        TypeKind.UINT8: pa.uint8(), TypeKind.UINT16: pa.uint16(),
+        TypeKind.UINT32: pa.uint32(), TypeKind.UINT64: pa.uint64(),
+        TypeKind.FLOAT32: pa.float32(), TypeKind.FLOAT64: pa.float64(),
+        TypeKind.BOOL: pa.bool_(), TypeKind.DATE32: pa.date32(),
+        TypeKind.TIMESTAMP_NS: pa.timestamp("ns"),
+        TypeKind.STRING: pa.large_string(),
+    }
+
+
+_init_pa_types()
 
 
 # combiner for two-phase aggregation: how to merge partial results

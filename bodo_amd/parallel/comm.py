@@ -89,12 +89,25 @@ def alltoallv_tensor(buf: torch.Tensor, send_counts: List[int],
                      recv_counts: List[int]) -> torch.Tensor:
     """Variable all-to-all of a 1-D tensor partitioned contiguously by dest."""
     dev = _comm_device()
+    orig_dtype = buf.dtype
+    # bool / int16 are not wire types for gloo: exchange as byte views
+    if orig_dtype == torch.bool:
+        buf = buf.view(torch.uint8)
+    elif orig_dtype == torch.int16:
+        buf = buf.contiguous().view(torch.uint8)
+        send_counts = [c * 2 for c in send_counts]
+        recv_counts = [c * 2 for c in recv_counts]
     moved = buf.device != dev
     src = buf.to(dev) if moved else buf
     total = int(sum(recv_counts))
     out = torch.empty(total, dtype=buf.dtype, device=dev)
     dist.all_to_all_single(out, src, recv_counts, send_counts)
-    return out.to(buf.device) if moved else out
+    out = out.to(buf.device) if moved else out
+    if orig_dtype == torch.bool:
+        out = out.view(torch.bool)
+    elif orig_dtype == torch.int16:
+        out = out.view(torch.int16)
+    return out
 
 
 def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:

@@ -175,3 +175,41 @@ def test_dist_distinct_limit():
     exp2 = df.sort_values("b").head(13).reset_index(drop=True)
     got2["c"] = got2["c"].astype(str)
     pd.testing.assert_frame_equal(got2, exp2, check_dtype=False)
+
+
+def _q_tpch(bpd, rank, payload):
+    import os, sys
+
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "benchmarks"))
+    from tpch_data import gen_all
+    from tpch_queries import ALL
+
+    t = gen_all(0.01)
+    frames = {k: bpd.from_pandas(v) for k, v in t.items()}
+    qn = payload["q"]
+    res = ALL[qn](bpd, frames)
+    return res
+
+
+@pytest.mark.parametrize("qn", [1, 4, 5, 13, 16, 21, 22])
+def test_dist_tpch(qn):
+    import os, sys
+
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "benchmarks"))
+    from tpch_data import gen_all
+    from tpch_queries import ALL
+
+    got = run_dist(_q_tpch, {"q": qn}).reset_index(drop=True)
+    t = gen_all(0.01)
+    exp = ALL[qn](pd, {k: v.copy() for k, v in t.items()}).reset_index(drop=True)
+    for c in exp.columns:
+        if exp[c].dtype == object or str(exp[c].dtype) == "category":
+            exp[c] = exp[c].astype(str)
+            got[c] = got[c].astype(str)
+    cols = list(exp.columns)
+    pd.testing.assert_frame_equal(
+        got.sort_values(cols).reset_index(drop=True),
+        exp.sort_values(cols).reset_index(drop=True),
+        check_dtype=False, atol=1e-6, rtol=1e-6)
