@@ -29,7 +29,9 @@ class BodoDataFrame:
     def __init__(self, plan: pn.PlanNode, columns: Sequence[str]):
         object.__setattr__(self, "_plan", plan)
         object.__setattr__(self, "_columns", list(columns))
-        object.__setattr__(self, "_result", None)  # cached local shard Table
+        object.__setattr__(self, "_result", None)
+        object.__setattr__(self, "_remote", None)  # cached local shard Table
+        object.__setattr__(self, "_remote", None)  # spawn-mode RemoteResult
 
     # ------------------------------------------------------------------
     # plan / execution
@@ -39,23 +41,43 @@ class BodoDataFrame:
         return self._plan
 
     def execute(self) -> Table:
-        """Execute the plan; returns this rank's shard and caches it."""
-        if self._result is None:
-            tbl = ex.execute(self._plan, ex.ExecutionContext())
-            object.__setattr__(self, "_result", tbl)
-            # re-root the plan on the materialized shard (ExecState.DISTRIBUTED)
-            key = ex.register_object(tbl)
-            object.__setattr__(
-                self, "_plan",
-                pn.PandasScan(key, tuple(tbl.names), distributed=True))
-            object.__setattr__(self, "_columns", list(tbl.names))
+        """Execute the plan; returns this rank's shard and caches it
+        (SPMD/local mode only — spawn mode goes through _materialize)."""
+        self._materialize()
         return self._result
 
+    def _materialize(self):
+        """Execute the plan locally or on spawn-mode workers; re-roots the
+        plan on the materialized result (ExecState.DISTRIBUTED)."""
+        if self._result is not None or getattr(self, "_remote", None) is not None:
+            return
+        from ..engine import api
+
+        kind, res = api.materialize(self._plan)
+        if kind == "remote":
+            object.__setattr__(self, "_remote", res)
+            object.__setattr__(
+                self, "_plan",
+                pn.PandasScan(res.res_id, tuple(res.names), distributed=True))
+            object.__setattr__(self, "_columns", list(res.names))
+            return
+        tbl = res
+        object.__setattr__(self, "_result", tbl)
+        key = ex.register_object(tbl)
+        object.__setattr__(
+            self, "_plan",
+            pn.PandasScan(key, tuple(tbl.names), distributed=True))
+        object.__setattr__(self, "_columns", list(tbl.names))
+
     def to_pandas(self) -> pd.DataFrame:
-        shard = self.execute()
-        full = comm.allgather_table(shard)
-        df = full.to_pandas()
-        return df
+        self._materialize()
+        remote = getattr(self, "_remote", None)
+        if remote is not None:
+            from ..parallel import spawn
+
+            return spawn.get_spawner().gather(remote.res_id).to_pandas()
+        full = comm.allgather_table(self._result)
+        return full.to_pandas()
 
     # aliases used by tests / fallback
     collect = to_pandas
@@ -75,8 +97,11 @@ class BodoDataFrame:
         return (n, len(self._columns))
 
     def __len__(self) -> int:
-        shard = self.execute()
-        return int(sum(comm.allgather_obj(len(shard))))
+        self._materialize()
+        remote = getattr(self, "_remote", None)
+        if remote is not None:
+            return remote.length
+        return int(sum(comm.allgather_obj(len(self._result))))
 
     @property
     def empty(self) -> bool:
@@ -141,6 +166,7 @@ class BodoDataFrame:
         object.__setattr__(self, "_plan", plan)
         object.__setattr__(self, "_columns", new_cols)
         object.__setattr__(self, "_result", None)
+        object.__setattr__(self, "_remote", None)
 
     def __setitem__(self, key: str, value):
         from .series import BodoSeries
@@ -168,6 +194,7 @@ class BodoDataFrame:
                 object.__setattr__(self, "_plan", new_plan)
                 object.__setattr__(self, "_columns", names)
                 object.__setattr__(self, "_result", None)
+                object.__setattr__(self, "_remote", None)
                 return
             if vplan is not self._plan:
                 # allow setting from a series derived from the same frame
@@ -195,6 +222,7 @@ class BodoDataFrame:
         object.__setattr__(self, "_plan", new_plan)
         object.__setattr__(self, "_columns", names)
         object.__setattr__(self, "_result", None)
+        object.__setattr__(self, "_remote", None)
 
     def assign(self, **kwargs) -> "BodoDataFrame":
         out = BodoDataFrame(self._plan, self._columns)
@@ -212,6 +240,7 @@ class BodoDataFrame:
             object.__setattr__(self, "_plan", plan)
             object.__setattr__(self, "_columns", names)
             object.__setattr__(self, "_result", None)
+            object.__setattr__(self, "_remote", None)
             return None
         return BodoDataFrame(plan, names)
 
@@ -229,6 +258,7 @@ class BodoDataFrame:
             object.__setattr__(self, "_plan", plan)
             object.__setattr__(self, "_columns", keep)
             object.__setattr__(self, "_result", None)
+            object.__setattr__(self, "_remote", None)
             return None
         return BodoDataFrame(plan, keep)
 
@@ -372,8 +402,10 @@ class BodoDataFrame:
     # IO
     # ------------------------------------------------------------------
     def to_parquet(self, path: str, compression="snappy", **kwargs):
+        from ..engine import api
+
         plan = pn.ParquetWrite(self._plan, path, compression)
-        ex.execute(plan, ex.ExecutionContext())
+        api.materialize(plan)
 
     def to_csv(self, path=None, **kwargs):
         pdf = self.to_pandas()

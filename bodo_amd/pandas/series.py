@@ -139,10 +139,11 @@ class BodoSeries:
     # reductions (execute eagerly, return python scalar)
     # ------------------------------------------------------------------
     def _reduce(self, func: str):
+        from ..engine import api
+
         plan = pn.Reduce(self._as_projection_plan(), (("r", "v", func),))
-        tbl = ex.execute(plan, ex.ExecutionContext())
-        ser = tbl.column("r").to_pandas()
-        return ser.iloc[0] if len(ser) else None
+        df = api.collect(plan)
+        return df["r"].iloc[0] if len(df) else None
 
     def sum(self):
         return self._reduce("sum")
@@ -172,29 +173,26 @@ class BodoSeries:
         return self._reduce("all")
 
     def nunique(self):
-        plan = pn.Aggregate(
-            pn.Distinct(self._as_projection_plan(), ("v",)), (),
-            (("n", "v", "count"),))
-        tbl = ex.execute(plan, ex.ExecutionContext())
-        # aggregate with no keys isn't supported by groupby; use Reduce
+        from ..engine import api
+
         plan = pn.Reduce(pn.Distinct(self._as_projection_plan(), ("v",)),
                          (("n", "v", "count"),))
-        tbl = ex.execute(plan, ex.ExecutionContext())
-        return int(tbl.column("n").to_pandas().iloc[0])
+        df = api.collect(plan)
+        return int(df["n"].iloc[0])
 
     def unique(self):
+        from ..engine import api
+
         plan = pn.Distinct(self._as_projection_plan(), ("v",))
-        tbl = ex.execute(plan, ex.ExecutionContext())
-        full = comm.allgather_table(tbl)
-        return full.column("v").to_pandas().to_numpy()
+        return api.collect(plan)["v"].to_numpy()
 
     def value_counts(self, ascending=False, dropna=True):
+        from ..engine import api
+
         plan = pn.Aggregate(self._as_projection_plan(), ("v",),
                             (("count", "v", "size"),), dropna=dropna)
         plan = pn.Sort(plan, ("count",), (ascending,))
-        tbl = ex.execute(plan, ex.ExecutionContext())
-        full = comm.allgather_table(tbl)
-        pdf = full.to_pandas()
+        pdf = api.collect(plan)
         # ties: pandas orders by value; match roughly
         out = pd.Series(pdf["count"].to_numpy(), index=pdf["v"].to_numpy(),
                         name="count")
@@ -206,18 +204,18 @@ class BodoSeries:
 
     # ------------------------------------------------------------------
     def to_pandas(self) -> pd.Series:
+        from ..engine import api
+
         plan = pn.Projection(self._plan, ("v",), (self._expr,))
-        tbl = ex.execute(plan, ex.ExecutionContext())
-        full = comm.allgather_table(tbl)
-        ser = full.column("v").to_pandas()
+        ser = api.collect(plan)["v"]
         ser.name = self.name
         return ser
 
     def head(self, n=5):
+        from ..engine import api
+
         plan = pn.Limit(pn.Projection(self._plan, ("v",), (self._expr,)), n)
-        tbl = ex.execute(plan, ex.ExecutionContext())
-        full = comm.allgather_table(tbl)
-        ser = full.column("v").to_pandas()
+        ser = api.collect(plan)["v"]
         ser.name = self.name
         return ser
 
