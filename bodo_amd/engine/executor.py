@@ -309,6 +309,13 @@ def _exec_sample(node: pn.Sample, ctx) -> Table:
     return ops.take_table(child, torch.sort(pos).values)
 
 
+def _exec_window(node: pn.Window, ctx) -> Table:
+    from .window import exec_window
+
+    child = _exec(node.child, ctx)
+    return exec_window(node, ctx, child)
+
+
 def _exec_shuffle_by_key(node: pn.ShuffleByKey, ctx) -> Table:
     child = _exec(node.child, ctx)
     if ctx.world == 1:
@@ -442,6 +449,7 @@ _HANDLERS = {
     pn.Sample: _exec_sample,
     pn.MapPartitions: _exec_map_partitions,
     pn.ShuffleByKey: _exec_shuffle_by_key,
+    pn.Window: _exec_window,
     pn.Join: _exec_join,
     pn.Union: _exec_union,
     pn.ParquetWrite: _exec_parquet_write,

@@ -160,6 +160,13 @@ def prune_columns(node: pn.PlanNode, required: Optional[Set[str]]) -> pn.PlanNod
     if isinstance(node, pn.Reduce):
         child_req = {a[1] for a in node.aggs if a[1]}
         return node.with_children(prune_columns(node.child, child_req))
+    if isinstance(node, pn.Window):
+        child_req = None
+        if required is not None:
+            child_req = (set(required) | set(node.keys) | set(node.order_by)
+                         | {s[1] for s in node.specs if s[1]})
+            child_req -= {s[0] for s in node.specs}
+        return node.with_children(prune_columns(node.child, child_req))
     if isinstance(node, (pn.Sort, pn.Limit, pn.Distinct, pn.Sample)):
         req2 = None
         if required is not None:

@@ -196,6 +196,32 @@ class BodoDataFrame:
                 object.__setattr__(self, "_result", None)
                 object.__setattr__(self, "_remote", None)
                 return
+            if (isinstance(vplan, pn.Window) and vplan.child is self._plan
+                    and isinstance(value._expr, ColRef)
+                    and value._expr.name == vplan.specs[0][0]):
+                # df["r"] = df.groupby(k)[c].shift()/rank()/...: rename the
+                # window output column to the assignment target
+                spec = vplan.specs[0]
+                new_specs = ((key,) + spec[1:],) + vplan.specs[1:]
+                names = list(self._columns)
+                if key not in names:
+                    names.append(key)
+                    new_plan = pn.Window(self._plan, vplan.keys,
+                                         vplan.order_by, vplan.ascending,
+                                         new_specs)
+                else:
+                    # overwrite existing column: window then project
+                    w_plan = pn.Window(self._plan, vplan.keys, vplan.order_by,
+                                       vplan.ascending, vplan.specs)
+                    exprs = [ColRef(c) if c != key else ColRef(spec[0])
+                             for c in names]
+                    new_plan = pn.Projection(w_plan, tuple(names),
+                                             tuple(exprs))
+                object.__setattr__(self, "_plan", new_plan)
+                object.__setattr__(self, "_columns", names)
+                object.__setattr__(self, "_result", None)
+                object.__setattr__(self, "_remote", None)
+                return
             if vplan is not self._plan:
                 # allow setting from a series derived from the same frame
                 # lineage after assignments: rebuild on current plan if the

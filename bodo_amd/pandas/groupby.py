@@ -116,6 +116,28 @@ class DataFrameGroupBy:
             return ser
         return out
 
+    def _window(self, specs, order_by=(), ascending=()):
+        from .frame import BodoDataFrame
+
+        plan = pn.Window(self._frame._lazy_plan, tuple(self._keys),
+                         tuple(order_by), tuple(ascending), tuple(specs))
+        cols = list(self._frame._columns) + [s[0] for s in specs]
+        return BodoDataFrame(plan, cols)
+
+    def transform(self, func):
+        from .series import BodoSeries
+        from ..plan.expr import ColRef
+
+        cols = self._value_columns()
+        specs = [(f"__t_{c}", c, f"transform_{func}", None) for c in cols]
+        out = self._window(specs)
+        proj = out[[f"__t_{c}" for c in cols]]
+        return proj.rename(columns={f"__t_{c}": c for c in cols})
+
+    def cumcount(self):
+        out = self._window([("__cc", "", "cumcount", None)])
+        return out["__cc"]
+
     def apply(self, func, *args, **kwargs):
         # general groupby-apply: shuffle rows by key then run pandas apply
         # per shard (keys co-located so results are exact)
@@ -183,6 +205,32 @@ class SeriesGroupBy:
         if isinstance(func, str):
             return self._agg1(func)
         raise NotImplementedError
+
+    def _window1(self, func, in_name=None, arg=None, order_by=(), ascending=()):
+        from .frame import BodoDataFrame
+
+        plan = pn.Window(self._frame._lazy_plan, tuple(self._keys),
+                         tuple(order_by), tuple(ascending),
+                         ((f"__w", in_name if in_name is not None else self._col,
+                           func, arg),))
+        cols = list(self._frame._columns) + ["__w"]
+        return BodoDataFrame(plan, cols)["__w"]
+
+    def transform(self, func):
+        return self._window1(f"transform_{func}")
+
+    def shift(self, periods=1):
+        return self._window1("shift", arg=periods)
+
+    def cumsum(self):
+        return self._window1("cumsum")
+
+    def cumcount(self):
+        return self._window1("cumcount")
+
+    def rank(self, method="min", ascending=True):
+        return self._window1("rank", arg=method,
+                             ascending=(ascending,))
 
 
 class _IndexedAggResult:

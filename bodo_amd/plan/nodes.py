@@ -218,6 +218,34 @@ class ShuffleByKey(PlanNode):
         return self.child.out_columns()
 
 
+@dataclass(frozen=True)
+class Window(PlanNode):
+    """Window/transform functions over partitions (reference:
+    bodo/libs/window/_window_calculator.cpp).  specs = (out_name, in_name,
+    func, arg) with func in transform_sum/transform_mean/transform_min/
+    transform_max/transform_count/row_number/rank/dense_rank/shift/cumsum/
+    cumcount."""
+
+    child: PlanNode
+    keys: Tuple[str, ...] = ()
+    order_by: Tuple[str, ...] = ()
+    ascending: Tuple[bool, ...] = ()
+    specs: Tuple[Tuple[str, str, str, Any], ...] = ()
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Window(ch[0], self.keys, self.order_by, self.ascending,
+                      self.specs)
+
+    def out_columns(self):
+        base = self.child.out_columns()
+        if base is None:
+            return None
+        return list(base) + [s[0] for s in self.specs]
+
+
 # ---------------------------------------------------------------- binary
 @dataclass(frozen=True)
 class Join(PlanNode):

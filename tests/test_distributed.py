@@ -213,3 +213,22 @@ def test_dist_tpch(qn):
         got.sort_values(cols).reset_index(drop=True),
         exp.sort_values(cols).reset_index(drop=True),
         check_dtype=False, atol=1e-6, rtol=1e-6)
+
+
+def _q_window(bpd, rank, payload):
+    b = bpd.from_pandas(payload["df"])
+    b["gs"] = b.groupby("a")["b"].transform("sum")
+    b["cs"] = b.groupby("a")["b"].cumsum()
+    b["r"] = b.groupby("a")["b"].rank(method="min")
+    return b[["a", "b", "gs", "cs", "r"]]
+
+
+def test_dist_window():
+    df = _df(2000, 21)
+    got = run_dist(_q_window, {"df": df}).reset_index(drop=True)
+    exp = df.copy()
+    exp["gs"] = exp.groupby("a")["b"].transform("sum")
+    exp["cs"] = exp.groupby("a")["b"].cumsum()
+    exp["r"] = exp.groupby("a")["b"].rank(method="min")
+    exp = exp[["a", "b", "gs", "cs", "r"]].reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)

@@ -242,3 +242,29 @@ def test_fallback_describe():
     exp = df.describe()
     pd.testing.assert_frame_equal(got.to_pandas() if hasattr(got, "to_pandas")
                                   else got, exp)
+
+
+def test_groupby_transform():
+    def q(m, df):
+        df["gs"] = df.groupby("a")["b"].transform("sum")
+        df["gm"] = df.groupby("a")["b"].transform("mean")
+        return df[["a", "gs", "gm"]]
+
+    check_query(q, {"df": simple_df(800)})
+
+
+def test_groupby_shift_cumsum():
+    def q(m, df):
+        df["sh"] = df.groupby("a")["b"].shift(1)
+        df["cs"] = df.groupby("a")["b"].cumsum()
+        return df[["a", "sh", "cs"]]
+
+    check_query(q, {"df": simple_df(500)})
+
+
+def test_groupby_rank():
+    def q(m, df):
+        df["r"] = df.groupby("a")["b"].rank(method="min")
+        return df[["a", "b", "r"]]
+
+    check_query(q, {"df": simple_df(400)})
