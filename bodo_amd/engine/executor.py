@@ -133,7 +133,7 @@ def _exec_aggregate(node: pn.Aggregate, ctx) -> Table:
     aggs = list(node.aggs)
     if ctx.world == 1:
         return rel.groupby_local(child, keys, aggs, node.dropna)
-    if any(a[2] in SINGLE_PHASE_AGGS for a in aggs):
+    if any(a[2] in SINGLE_PHASE_AGGS or callable(a[2]) for a in aggs):
         # shuffle raw rows by key hash, then single local groupby
         shuffled = _shuffle_by_keys(child, keys, ctx)
         return rel.groupby_local(shuffled, keys, aggs, node.dropna)

@@ -25,8 +25,16 @@ def read_shard(path: str, options: dict, columns: Optional[Sequence[str]], ctx) 
     opts = _reader_opts(options)
     if columns:
         use = opts.get("usecols")
-        opts["usecols"] = list(columns) if use is None else [c for c in use if c in columns]
+        if use is None:
+            opts["usecols"] = list(columns)
+        elif all(isinstance(c, str) for c in use):
+            opts["usecols"] = [c for c in use if c in columns]
+        # integer usecols: keep the user's selection; prune after reading
     df = pd.read_csv(path, **opts)
+    if columns:
+        keep = [c for c in df.columns if c in set(columns)]
+        if keep and len(keep) < len(df.columns):
+            df = df[keep]
     n = len(df)
     w, r = ctx.world, ctx.rank
     base, rem = divmod(n, w)

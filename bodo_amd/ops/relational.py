@@ -32,6 +32,10 @@ def groupby_local(tbl: Table, keys: Sequence[str],
                   dropna: bool = True) -> Table:
     """Group rows of the local shard; aggs = (out_name, in_name, func)."""
     if tbl.device.type == "cuda":
+        if any(callable(a[2]) for a in aggs):
+            # custom python agg: host pandas per co-located shard (the
+            # @jit-to-HIP lowering is the native path for these)
+            return _groupby_pandas(tbl, keys, aggs, dropna)
         from . import gpu
 
         return gpu.groupby_local(tbl, keys, aggs, dropna)

@@ -261,12 +261,12 @@ class _IndexedAggResult:
         return repr(self.to_pandas().head(10))
 
 
-def _norm_func(f) -> str:
+def _norm_func(f):
     if callable(f) and hasattr(f, "__name__"):
         name = f.__name__
         if name in ("sum", "mean", "min", "max", "count", "size", "median",
                     "var", "std", "prod"):
             return name
-        raise NotImplementedError(f"custom agg func {f}")
+        return f  # custom callable: single-phase pandas agg on shuffled groups
     m = {"average": "mean", "nunique": "nunique"}
     return m.get(f, f)

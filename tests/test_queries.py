@@ -167,3 +167,53 @@ def test_readme_20m_style(tmp_path):
     exp = df.assign(C=exp_c).groupby("A", as_index=False).agg(
         s=("C", "sum"), m=("B", "mean")).sort_values("A").reset_index(drop=True)
     pd.testing.assert_frame_equal(out, exp, check_dtype=False)
+
+
+def test_tpcxbb_q26_style(tmp_path):
+    """TPCx-BB Q26 shape (reference: e2e-tests/tpcx-bb/TPCxBB_q26.py):
+    csv read with sep/usecols/names/dtype + merge + custom-callable aggs."""
+    import bodo_amd.pandas as bpd
+
+    rng = np.random.default_rng(4)
+    n = 20000
+    ss = pd.DataFrame({
+        "junk1": np.zeros(n, dtype=np.int64),
+        "junk2": np.zeros(n, dtype=np.int64),
+        "ss_item_sk": rng.integers(1, 500, n),
+        "ss_customer_sk": rng.integers(1, 300, n),
+    })
+    item = pd.DataFrame({
+        "i_item_sk": np.arange(1, 501),
+        "i_class_id": rng.integers(1, 16, 500).astype(np.int32),
+        "i_category": rng.choice(["Books", "Music", "Home"], 500),
+    })
+    ss_path = str(tmp_path / "ss.dat")
+    ss.to_csv(ss_path, sep="|", header=False, index=False)
+
+    def q26(m, ss_frame, item_frame, category, item_count):
+        item2 = item_frame[item_frame["i_category"] == category]
+        sale_items = ss_frame.merge(item2, left_on="ss_item_sk",
+                                    right_on="i_item_sk")
+
+        def id1(x):
+            return (x == 1).sum()
+
+        def id2(x):
+            return (x == 2).sum()
+
+        agg = sale_items.groupby("ss_customer_sk", as_index=False).agg(
+            cnt=m.NamedAgg("ss_item_sk", "count"),
+            c1=m.NamedAgg("i_class_id", id1),
+            c2=m.NamedAgg("i_class_id", id2))
+        agg = agg[agg.cnt > item_count]
+        return agg.sort_values("ss_customer_sk")
+
+    b_ss = bpd.read_csv(ss_path, sep="|",
+                        names=["junk1", "junk2", "ss_item_sk", "ss_customer_sk"],
+                        usecols=[2, 3],
+                        dtype={"ss_item_sk": np.int64, "ss_customer_sk": np.int64})
+    got = q26(bpd, b_ss, bpd.from_pandas(item), "Books", 5).to_pandas()
+    exp = q26(pd, ss[["ss_item_sk", "ss_customer_sk"]].copy(), item.copy(),
+              "Books", 5).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got.reset_index(drop=True), exp,
+                                  check_dtype=False)
