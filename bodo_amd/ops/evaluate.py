@@ -201,7 +201,36 @@ class _Evaluator:
         return Column(bt.boolean, out)
 
     def visit_Case(self, e: Case) -> Column:
+        # string-valued CASE: build a small dictionary + int32 codes
+        branch_vals = list(e.thens) + [e.otherwise]
+        if all(isinstance(v, Const) and isinstance(v.value, str)
+               for v in branch_vals):
+            import pyarrow as pa
+
+            vals = [v.value for v in branch_vals]
+            uniq = list(dict.fromkeys(vals))
+            code_of = {v: i for i, v in enumerate(uniq)}
+            codes = torch.full((self.n,), code_of[vals[-1]],
+                               dtype=torch.int32, device=self.device)
+            for cond_e, then_e in reversed(list(zip(e.conds, e.thens))):
+                c = self.visit(cond_e).data
+                codes = torch.where(
+                    c, torch.tensor(code_of[then_e.value], dtype=torch.int32,
+                                    device=self.device), codes)
+            return Column(bt.dictionary, codes, None,
+                          dictionary=pa.array(uniq, type=pa.large_string()),
+                          length=self.n)
         other = self.visit(e.otherwise)
+        if other.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+            # general string CASE: host path
+            ser = other.to_pandas()
+            for cond_e, then_e in reversed(list(zip(e.conds, e.thens))):
+                c = self.visit(cond_e).data.cpu().numpy()
+                tv = self.visit(then_e).to_pandas()
+                ser = ser.where(~c, tv)
+            import pyarrow as pa
+
+            return Column.from_arrow(pa.Array.from_pandas(ser), self.device)
         out_data = other.data.clone()
         out_mask = None if other.mask is None else other.mask.clone()
         # apply in reverse so the first matching cond wins

@@ -1,0 +1,46 @@
+"""bodo_amd.sql: SQL frontend (reference: BodoSQL BodoSQLContext,
+BodoSQL/bodosql/context.py:504) — parses SQL natively (no JVM/Calcite) and
+executes it on the same logical-plan/columnar-HIP engine as the pandas API.
+"""
+
+from __future__ import annotations
+
+from typing import Dict
+
+
+class BodoSQLContext:
+    def __init__(self, tables: Dict[str, object]):
+        from ..pandas.frame import BodoDataFrame, from_pandas_df
+
+        import pandas as pd
+
+        self.tables = {}
+        for name, t in tables.items():
+            if isinstance(t, pd.DataFrame):
+                t = from_pandas_df(t)
+            self.tables[name] = t
+
+    def add_or_replace_view(self, name: str, table) -> "BodoSQLContext":
+        new = dict(self.tables)
+        new[name] = table
+        return BodoSQLContext(new)
+
+    def remove_view(self, name: str) -> "BodoSQLContext":
+        new = dict(self.tables)
+        new.pop(name, None)
+        return BodoSQLContext(new)
+
+    def sql(self, query: str):
+        from ..pandas.frame import BodoDataFrame
+        from .parser import Parser
+        from .planner import Planner
+
+        q = Parser(query).parse()
+        plan, names = Planner(self.tables).plan(q)
+        return BodoDataFrame(plan, names)
+
+    # convenience parity alias
+    def convert_to_pandas(self, query: str) -> str:  # pragma: no cover
+        raise NotImplementedError(
+            "bodo_amd executes SQL directly on the C++ backend path; "
+            "pandas-codegen output is not produced")

@@ -1,0 +1,497 @@
+"""SQL tokenizer + recursive-descent parser producing a small AST.
+
+Replaces the reference's Calcite JVM frontend (BodoSQL/calcite_sql, SURVEY
+§2.8: "the Calcite JVM can be replaced by reusing the same logical-plan
+builder") with a native parser for the analytic subset the engine executes:
+SELECT / FROM / JOIN / WHERE / GROUP BY / HAVING / ORDER BY / LIMIT,
+standard expressions, CASE, IN, BETWEEN, LIKE, EXTRACT, CAST.
+"""
+
+from __future__ import annotations
+
+import re
+from dataclasses import dataclass, field
+from typing import Any, List, Optional, Tuple
+
+KEYWORDS = {
+    "select", "distinct", "from", "where", "group", "by", "having", "order",
+    "limit", "as", "and", "or", "not", "in", "between", "like", "is", "null",
+    "case", "when", "then", "else", "end", "cast", "extract", "join", "inner",
+    "left", "right", "full", "outer", "cross", "on", "asc", "desc", "date",
+    "interval", "union", "all", "exists", "true", "false", "substring", "for",
+}
+
+_TOKEN_RE = re.compile(r"""
+    (?P<ws>\s+)
+  | (?P<num>\d+\.\d*|\.\d+|\d+)
+  | (?P<str>'(?:[^']|'')*')
+  | (?P<id>[A-Za-z_][A-Za-z_0-9]*)
+  | (?P<op><>|<=|>=|!=|\|\||[-+*/%(),.<>=])
+""", re.VERBOSE)
+
+
+@dataclass
+class Tok:
+    kind: str  # num, str, id, kw, op
+    value: str
+
+
+def tokenize(sql: str) -> List[Tok]:
+    out = []
+    pos = 0
+    sql = re.sub(r"--[^\n]*", " ", sql)
+    while pos < len(sql):
+        m = _TOKEN_RE.match(sql, pos)
+        if not m:
+            raise SyntaxError(f"bad SQL at: {sql[pos:pos + 30]!r}")
+        pos = m.end()
+        if m.lastgroup == "ws":
+            continue
+        v = m.group()
+        if m.lastgroup == "id":
+            low = v.lower()
+            out.append(Tok("kw", low) if low in KEYWORDS else Tok("id", v))
+        elif m.lastgroup == "str":
+            out.append(Tok("str", v[1:-1].replace("''", "'")))
+        else:
+            out.append(Tok(m.lastgroup, v))
+    return out
+
+
+# ---------------------------------------------------------------- AST
+@dataclass
+class Col:
+    table: Optional[str]
+    name: str
+
+
+@dataclass
+class Lit:
+    value: Any
+    kind: str = "auto"  # num/str/date/bool/null
+
+
+@dataclass
+class Bin:
+    op: str
+    left: Any
+    right: Any
+
+
+@dataclass
+class Un:
+    op: str
+    operand: Any
+
+
+@dataclass
+class Func:
+    name: str
+    args: List[Any]
+    distinct: bool = False
+    star: bool = False
+
+
+@dataclass
+class CaseE:
+    whens: List[Tuple[Any, Any]]
+    els: Optional[Any]
+
+
+@dataclass
+class InE:
+    operand: Any
+    values: List[Any]
+    negated: bool = False
+
+
+@dataclass
+class BetweenE:
+    operand: Any
+    lo: Any
+    hi: Any
+    negated: bool = False
+
+
+@dataclass
+class LikeE:
+    operand: Any
+    pattern: str
+    negated: bool = False
+
+
+@dataclass
+class IsNullE:
+    operand: Any
+    negated: bool = False
+
+
+@dataclass
+class CastE:
+    operand: Any
+    to: str
+
+
+@dataclass
+class ExtractE:
+    fld: str
+    operand: Any
+
+
+@dataclass
+class TableRef:
+    name: str
+    alias: Optional[str]
+
+
+@dataclass
+class JoinClause:
+    kind: str  # inner/left/right/full/cross
+    table: TableRef
+    on: Optional[Any]
+
+
+@dataclass
+class SelectItem:
+    expr: Any
+    alias: Optional[str]
+    star: bool = False
+
+
+@dataclass
+class Query:
+    items: List[SelectItem]
+    distinct: bool
+    table: Optional[TableRef]
+    joins: List[JoinClause]
+    where: Optional[Any]
+    group_by: List[Any]
+    having: Optional[Any]
+    order_by: List[Tuple[Any, bool]]
+    limit: Optional[int]
+
+
+class Parser:
+    def __init__(self, sql: str):
+        self.toks = tokenize(sql)
+        self.i = 0
+
+    # ------------------------------------------------------------- utils
+    def peek(self) -> Optional[Tok]:
+        return self.toks[self.i] if self.i < len(self.toks) else None
+
+    def next(self) -> Tok:
+        t = self.peek()
+        if t is None:
+            raise SyntaxError("unexpected end of SQL")
+        self.i += 1
+        return t
+
+    def accept_kw(self, *kws) -> Optional[str]:
+        t = self.peek()
+        if t and t.kind == "kw" and t.value in kws:
+            self.i += 1
+            return t.value
+        return None
+
+    def expect_kw(self, kw):
+        if not self.accept_kw(kw):
+            raise SyntaxError(f"expected {kw.upper()} at {self.peek()}")
+
+    def accept_op(self, op) -> bool:
+        t = self.peek()
+        if t and t.kind == "op" and t.value == op:
+            self.i += 1
+            return True
+        return False
+
+    def expect_op(self, op):
+        if not self.accept_op(op):
+            raise SyntaxError(f"expected {op!r} at {self.peek()}")
+
+    # ------------------------------------------------------------- query
+    def parse(self) -> Query:
+        q = self.parse_select()
+        if self.peek() is not None:
+            raise SyntaxError(f"trailing tokens at {self.peek()}")
+        return q
+
+    def parse_select(self) -> Query:
+        self.expect_kw("select")
+        distinct = bool(self.accept_kw("distinct"))
+        items = [self.parse_select_item()]
+        while self.accept_op(","):
+            items.append(self.parse_select_item())
+        table, joins = None, []
+        if self.accept_kw("from"):
+            table = self.parse_table_ref()
+            while True:
+                t = self.peek()
+                if t is None:
+                    break
+                if t.kind == "op" and t.value == ",":
+                    self.i += 1
+                    joins.append(JoinClause("cross", self.parse_table_ref(), None))
+                    continue
+                kind = None
+                if self.accept_kw("inner"):
+                    kind = "inner"
+                elif self.accept_kw("left"):
+                    self.accept_kw("outer")
+                    kind = "left"
+                elif self.accept_kw("right"):
+                    self.accept_kw("outer")
+                    kind = "right"
+                elif self.accept_kw("full"):
+                    self.accept_kw("outer")
+                    kind = "outer"
+                elif self.accept_kw("cross"):
+                    kind = "cross"
+                if kind is None:
+                    if self.peek() and self.peek().kind == "kw" \
+                            and self.peek().value == "join":
+                        kind = "inner"
+                    else:
+                        break
+                self.expect_kw("join")
+                tr = self.parse_table_ref()
+                on = None
+                if kind != "cross" and self.accept_kw("on"):
+                    on = self.parse_expr()
+                joins.append(JoinClause(kind, tr, on))
+        where = self.parse_expr() if self.accept_kw("where") else None
+        group_by: List[Any] = []
+        if self.accept_kw("group"):
+            self.expect_kw("by")
+            group_by.append(self.parse_expr())
+            while self.accept_op(","):
+                group_by.append(self.parse_expr())
+        having = self.parse_expr() if self.accept_kw("having") else None
+        order_by: List[Tuple[Any, bool]] = []
+        if self.accept_kw("order"):
+            self.expect_kw("by")
+            while True:
+                e = self.parse_expr()
+                asc = True
+                if self.accept_kw("desc"):
+                    asc = False
+                else:
+                    self.accept_kw("asc")
+                order_by.append((e, asc))
+                if not self.accept_op(","):
+                    break
+        limit = None
+        if self.accept_kw("limit"):
+            limit = int(self.next().value)
+        return Query(items, distinct, table, joins, where, group_by, having,
+                     order_by, limit)
+
+    def parse_table_ref(self) -> TableRef:
+        t = self.next()
+        if t.kind != "id":
+            raise SyntaxError(f"expected table name, got {t}")
+        alias = None
+        nt = self.peek()
+        if nt and nt.kind == "id":
+            alias = self.next().value
+        elif self.accept_kw("as"):
+            alias = self.next().value
+        return TableRef(t.value, alias)
+
+    def parse_select_item(self) -> SelectItem:
+        if self.accept_op("*"):
+            return SelectItem(None, None, star=True)
+        e = self.parse_expr()
+        alias = None
+        if self.accept_kw("as"):
+            alias = self.next().value
+        else:
+            nt = self.peek()
+            if nt and nt.kind == "id":
+                alias = self.next().value
+        return SelectItem(e, alias)
+
+    # --------------------------------------------------------- expressions
+    def parse_expr(self):
+        return self.parse_or()
+
+    def parse_or(self):
+        left = self.parse_and()
+        while self.accept_kw("or"):
+            left = Bin("or", left, self.parse_and())
+        return left
+
+    def parse_and(self):
+        left = self.parse_not()
+        while self.accept_kw("and"):
+            left = Bin("and", left, self.parse_not())
+        return left
+
+    def parse_not(self):
+        if self.accept_kw("not"):
+            return Un("not", self.parse_not())
+        return self.parse_predicate()
+
+    def parse_predicate(self):
+        left = self.parse_add()
+        t = self.peek()
+        negated = False
+        if t and t.kind == "kw" and t.value == "not":
+            nxt = self.toks[self.i + 1] if self.i + 1 < len(self.toks) else None
+            if nxt and nxt.kind == "kw" and nxt.value in ("in", "like", "between"):
+                self.i += 1
+                negated = True
+                t = self.peek()
+        if t and t.kind == "op" and t.value in ("=", "<>", "!=", "<", "<=", ">", ">="):
+            self.i += 1
+            opmap = {"=": "eq", "<>": "ne", "!=": "ne", "<": "lt",
+                     "<=": "le", ">": "gt", ">=": "ge"}
+            return Bin(opmap[t.value], left, self.parse_add())
+        if t and t.kind == "kw" and t.value == "in":
+            self.i += 1
+            self.expect_op("(")
+            vals = [self.parse_expr()]
+            while self.accept_op(","):
+                vals.append(self.parse_expr())
+            self.expect_op(")")
+            return InE(left, vals, negated)
+        if t and t.kind == "kw" and t.value == "between":
+            self.i += 1
+            lo = self.parse_add()
+            self.expect_kw("and")
+            hi = self.parse_add()
+            return BetweenE(left, lo, hi, negated)
+        if t and t.kind == "kw" and t.value == "like":
+            self.i += 1
+            pat = self.next()
+            return LikeE(left, pat.value, negated)
+        if t and t.kind == "kw" and t.value == "is":
+            self.i += 1
+            neg = bool(self.accept_kw("not"))
+            self.expect_kw("null")
+            return IsNullE(left, neg)
+        return left
+
+    def parse_add(self):
+        left = self.parse_mul()
+        while True:
+            t = self.peek()
+            if t and t.kind == "op" and t.value in ("+", "-"):
+                self.i += 1
+                left = Bin("add" if t.value == "+" else "sub", left,
+                           self.parse_mul())
+            else:
+                return left
+
+    def parse_mul(self):
+        left = self.parse_unary()
+        while True:
+            t = self.peek()
+            if t and t.kind == "op" and t.value in ("*", "/", "%"):
+                self.i += 1
+                op = {"*": "mul", "/": "div", "%": "mod"}[t.value]
+                left = Bin(op, left, self.parse_unary())
+            else:
+                return left
+
+    def parse_unary(self):
+        if self.accept_op("-"):
+            return Bin("sub", Lit(0, "num"), self.parse_unary())
+        if self.accept_op("+"):
+            return self.parse_unary()
+        return self.parse_primary()
+
+    def parse_primary(self):
+        t = self.next()
+        if t.kind == "num":
+            v = float(t.value) if "." in t.value else int(t.value)
+            return Lit(v, "num")
+        if t.kind == "str":
+            return Lit(t.value, "str")
+        if t.kind == "kw":
+            if t.value == "date":
+                s = self.next()
+                return Lit(s.value, "date")
+            if t.value == "interval":
+                s = self.next()  # '3' or '3 month'
+                unit_t = self.peek()
+                unit = None
+                if unit_t and unit_t.kind == "id":
+                    unit = self.next().value.lower()
+                parts = s.value.split()
+                qty = int(parts[0])
+                if unit is None and len(parts) > 1:
+                    unit = parts[1].lower()
+                return Lit((qty, (unit or "day").rstrip("s")), "interval")
+            if t.value == "null":
+                return Lit(None, "null")
+            if t.value in ("true", "false"):
+                return Lit(t.value == "true", "bool")
+            if t.value == "case":
+                whens = []
+                els = None
+                while self.accept_kw("when"):
+                    c = self.parse_expr()
+                    self.expect_kw("then")
+                    v = self.parse_expr()
+                    whens.append((c, v))
+                if self.accept_kw("else"):
+                    els = self.parse_expr()
+                self.expect_kw("end")
+                return CaseE(whens, els)
+            if t.value == "cast":
+                self.expect_op("(")
+                e = self.parse_expr()
+                self.expect_kw("as")
+                ty = self.next().value.lower()
+                # swallow precision args: decimal(12,2)
+                if self.accept_op("("):
+                    while not self.accept_op(")"):
+                        self.next()
+                self.expect_op(")")
+                return CastE(e, ty)
+            if t.value == "extract":
+                self.expect_op("(")
+                fld = self.next().value.lower()
+                self.expect_kw("from")
+                e = self.parse_expr()
+                self.expect_op(")")
+                return ExtractE(fld, e)
+            if t.value == "substring":
+                self.expect_op("(")
+                e = self.parse_expr()
+                self.expect_kw("from")
+                start = self.parse_expr()
+                length = None
+                if self.accept_kw("for"):
+                    length = self.parse_expr()
+                self.expect_op(")")
+                return Func("substring", [e, start, length])
+            if t.value == "exists":
+                raise SyntaxError("EXISTS subqueries not supported yet")
+            raise SyntaxError(f"unexpected keyword {t.value!r}")
+        if t.kind == "op" and t.value == "(":
+            e = self.parse_expr()
+            self.expect_op(")")
+            return e
+        if t.kind == "id":
+            nt = self.peek()
+            if nt and nt.kind == "op" and nt.value == "(":
+                self.i += 1
+                distinct = bool(self.accept_kw("distinct"))
+                if self.accept_op("*"):
+                    self.expect_op(")")
+                    return Func(t.value.lower(), [], star=True)
+                args = []
+                if not self.accept_op(")"):
+                    args.append(self.parse_expr())
+                    while self.accept_op(","):
+                        args.append(self.parse_expr())
+                    self.expect_op(")")
+                return Func(t.value.lower(), args, distinct=distinct)
+            if nt and nt.kind == "op" and nt.value == ".":
+                self.i += 1
+                col = self.next()
+                return Col(t.value, col.value)
+            return Col(None, t.value)
+        raise SyntaxError(f"unexpected token {t}")

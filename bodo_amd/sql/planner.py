@@ -1,0 +1,493 @@
+"""SQL AST -> logical plan (reference: BodoSQL plan_conversion.py converting
+Calcite RelNodes into bodo.pandas LazyPlan; here the parser AST maps straight
+onto our plan nodes)."""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import pandas as pd
+
+from ..core import types as bt
+from ..plan import expr as ex
+from ..plan import nodes as pn
+from . import parser as ast
+
+AGG_FUNCS = {"sum": "sum", "avg": "mean", "count": "count", "min": "min",
+             "max": "max", "stddev": "std", "variance": "var",
+             "median": "median"}
+
+CAST_TYPES = {
+    "int": bt.int64, "integer": bt.int64, "bigint": bt.int64,
+    "smallint": bt.int16, "double": bt.float64, "float": bt.float64,
+    "real": bt.float32, "decimal": bt.float64, "numeric": bt.float64,
+    "varchar": bt.string, "char": bt.string, "text": bt.string,
+    "date": bt.date32, "timestamp": bt.timestamp_ns, "boolean": bt.boolean,
+}
+
+
+class Scope:
+    """Maps (alias, column) -> internal plan column name."""
+
+    def __init__(self):
+        self.entries: List[Tuple[str, str, str]] = []  # alias, col, internal
+
+    def add(self, alias: str, col: str, internal: str):
+        self.entries.append((alias, col, internal))
+
+    def resolve(self, table: Optional[str], name: str) -> str:
+        cands = []
+        for alias, col, internal in self.entries:
+            if col.lower() == name.lower() and (table is None
+                                                or alias.lower() == table.lower()):
+                cands.append(internal)
+        if not cands:
+            raise KeyError(f"unknown column {table + '.' if table else ''}{name}")
+        if len(set(cands)) > 1:
+            raise KeyError(f"ambiguous column {name}")
+        return cands[0]
+
+    def tables_of(self, e) -> set:
+        """Set of internal columns referenced by an AST expression."""
+        cols = set()
+
+        def walk(x):
+            if isinstance(x, ast.Col):
+                cols.add(self.resolve(x.table, x.name))
+            for f in getattr(x, "__dataclass_fields__", {}):
+                v = getattr(x, f)
+                if isinstance(v, (list, tuple)):
+                    for i in v:
+                        if hasattr(i, "__dataclass_fields__"):
+                            walk(i)
+                        elif isinstance(i, tuple):
+                            for j in i:
+                                if hasattr(j, "__dataclass_fields__"):
+                                    walk(j)
+                elif hasattr(v, "__dataclass_fields__"):
+                    walk(v)
+
+        walk(e)
+        return cols
+
+
+class Planner:
+    def __init__(self, tables: Dict[str, "object"]):
+        self.tables = {k.lower(): v for k, v in tables.items()}
+        self._counter = 0
+
+    def _uniq(self, base: str) -> str:
+        self._counter += 1
+        return f"{base}__{self._counter}"
+
+    # ------------------------------------------------------------------
+    def plan(self, q: ast.Query):
+        scope = Scope()
+        plan = None
+        joined_cols: set = set()
+        where_conjuncts = _split_conjuncts(q.where) if q.where else []
+        used_conjuncts = [False] * len(where_conjuncts)
+
+        refs = ([q.table] if q.table else []) + [j.table for j in q.joins]
+        join_kinds = ["base"] + [j.kind for j in q.joins]
+        join_ons = [None] + [j.on for j in q.joins]
+
+        for idx, tr in enumerate(refs):
+            sub_plan, sub_cols = self._table_plan(tr, scope)
+            if plan is None:
+                plan = sub_plan
+                joined_cols |= sub_cols
+                continue
+            kind = join_kinds[idx]
+            on = join_ons[idx]
+            eq_pairs: List[Tuple[str, str]] = []
+            post: List[ast.Bin] = []
+            if on is not None:
+                for c in _split_conjuncts(on):
+                    pair = self._equi_pair(c, scope, joined_cols, sub_cols)
+                    if pair:
+                        eq_pairs.append(pair)
+                    else:
+                        post.append(c)
+            if kind == "cross" or not eq_pairs:
+                # pull applicable equi conditions from WHERE (FROM a, b style)
+                for ci, c in enumerate(where_conjuncts):
+                    if used_conjuncts[ci]:
+                        continue
+                    pair = self._equi_pair(c, scope, joined_cols, sub_cols)
+                    if pair:
+                        eq_pairs.append(pair)
+                        used_conjuncts[ci] = True
+                if eq_pairs:
+                    kind = "inner" if kind == "cross" else kind
+            if eq_pairs:
+                lks = tuple(p[0] for p in eq_pairs)
+                rks = tuple(p[1] for p in eq_pairs)
+                plan = pn.Join(plan, sub_plan, lks, rks,
+                               kind if kind != "base" else "inner")
+            else:
+                plan = pn.Join(plan, sub_plan, (), (), "cross")
+            joined_cols |= sub_cols
+            for c in post:
+                plan = pn.Filter(plan, self.expr(c, scope))
+        if plan is None:
+            raise ValueError("SELECT without FROM not supported")
+        # remaining WHERE
+        rest = [c for ci, c in enumerate(where_conjuncts) if not used_conjuncts[ci]]
+        if rest:
+            cond = rest[0]
+            for c in rest[1:]:
+                cond = ast.Bin("and", cond, c)
+            plan = pn.Filter(plan, self.expr(cond, scope))
+
+        # ------------------------------------------------- aggregation
+        has_agg = any(_has_agg(it.expr) for it in q.items if not it.star) \
+            or (q.having is not None and _has_agg(q.having)) or q.group_by
+        out_names: List[str] = []
+        out_exprs: List[ex.Expr] = []
+        if has_agg:
+            plan, scope2, key_map, agg_map = self._aggregate(
+                plan, scope, q)
+            # build output projection over the agg result
+            for it in q.items:
+                if it.star:
+                    raise ValueError("SELECT * with GROUP BY not supported")
+                name = it.alias or _default_name(it.expr)
+                out_names.append(name)
+                out_exprs.append(self._post_agg_expr(it.expr, key_map, agg_map,
+                                                     scope))
+            if q.having is not None:
+                plan = pn.Filter(plan, self._post_agg_expr(
+                    q.having, key_map, agg_map, scope))
+            proj = pn.Projection(plan, tuple(out_names), tuple(out_exprs))
+            plan = proj
+            order_scope = dict(zip(out_names, out_names))
+            order_resolver = lambda e: self._post_agg_expr(
+                e, key_map, agg_map, scope) if not isinstance(e, ast.Col) \
+                or (e.table is None and e.name in out_names and False) else None
+        else:
+            for it in q.items:
+                if it.star:
+                    for alias, col, internal in scope.entries:
+                        out_names.append(col)
+                        out_exprs.append(ex.ColRef(internal))
+                    continue
+                name = it.alias or _default_name(it.expr)
+                out_names.append(name)
+                out_exprs.append(self.expr(it.expr, scope))
+            plan = pn.Projection(plan, tuple(out_names), tuple(out_exprs))
+        if q.distinct:
+            plan = pn.Distinct(plan, None)
+        if q.order_by:
+            keys, asc = [], []
+            extra = 0
+            for e, a in q.order_by:
+                if isinstance(e, ast.Lit) and e.kind == "num":
+                    keys.append(out_names[int(e.value) - 1])
+                elif isinstance(e, ast.Col) and e.table is None \
+                        and e.name in out_names:
+                    keys.append(e.name)
+                elif isinstance(e, ast.Col):
+                    keys.append(e.name)  # hope it's an output name
+                else:
+                    raise ValueError("ORDER BY expressions must be output "
+                                     "columns or positions")
+                asc.append(a)
+            plan = pn.Sort(plan, tuple(keys), tuple(asc))
+        if q.limit is not None:
+            plan = pn.Limit(plan, q.limit)
+        return plan, out_names
+
+    # ------------------------------------------------------------------
+    def _table_plan(self, tr: ast.TableRef, scope: Scope):
+        name = tr.name.lower()
+        if name not in self.tables:
+            raise KeyError(f"unknown table {tr.name}")
+        frame = self.tables[name]
+        alias = (tr.alias or tr.name).lower()
+        cols = list(frame._columns)
+        internal = []
+        need_rename = False
+        for c in cols:
+            iname = f"{alias}_{c}" if any(
+                e[1].lower() == c.lower() for e in scope.entries) else c
+            if iname != c:
+                need_rename = True
+            internal.append(iname)
+            scope.add(alias, c, iname)
+        sub = frame._lazy_plan
+        if need_rename:
+            sub = pn.Projection(sub, tuple(internal),
+                                tuple(ex.ColRef(c) for c in cols))
+        return sub, set(internal)
+
+    def _equi_pair(self, c, scope: Scope, left_cols: set, right_cols: set):
+        if isinstance(c, ast.Bin) and c.op == "eq" \
+                and isinstance(c.left, ast.Col) and isinstance(c.right, ast.Col):
+            try:
+                l = scope.resolve(c.left.table, c.left.name)
+                r = scope.resolve(c.right.table, c.right.name)
+            except KeyError:
+                return None
+            if l in left_cols and r in right_cols:
+                return (l, r)
+            if r in left_cols and l in right_cols:
+                return (r, l)
+        return None
+
+    # ------------------------------------------------------------------
+    def _aggregate(self, plan, scope: Scope, q: ast.Query):
+        # 1. group keys: plain columns stay; computed keys get a pre-projection
+        pre_names = [e[2] for e in scope.entries]
+        pre_exprs = {n: ex.ColRef(n) for n in pre_names}
+        keys = []
+        key_map = {}  # AST repr -> key column name
+        for g in q.group_by:
+            if isinstance(g, ast.Col):
+                internal = scope.resolve(g.table, g.name)
+                keys.append(internal)
+                key_map[_ast_key(g)] = internal
+            else:
+                kname = self._uniq("__gk")
+                pre_exprs[kname] = self.expr(g, scope)
+                pre_names.append(kname)
+                keys.append(kname)
+                key_map[_ast_key(g)] = kname
+        # 2. collect aggregate calls from select/having
+        agg_specs = []  # (out_name, in_name or "", func)
+        agg_map = {}
+
+        def collect(e):
+            if isinstance(e, ast.Func) and e.name in AGG_FUNCS:
+                k = _ast_key(e)
+                if k in agg_map:
+                    return
+                out = self._uniq("__agg")
+                if e.star or not e.args:
+                    agg_specs.append((out, "", "size"))
+                else:
+                    arg = e.args[0]
+                    if isinstance(arg, ast.Col):
+                        in_name = scope.resolve(arg.table, arg.name)
+                    else:
+                        in_name = self._uniq("__ain")
+                        pre_exprs[in_name] = self.expr(arg, scope)
+                        pre_names.append(in_name)
+                    func = AGG_FUNCS[e.name]
+                    if e.distinct and e.name == "count":
+                        func = "nunique"
+                    agg_specs.append((out, in_name, func))
+                agg_map[k] = out
+                return
+            for f in getattr(e, "__dataclass_fields__", {}):
+                v = getattr(e, f)
+                if isinstance(v, (list, tuple)):
+                    for i in v:
+                        if hasattr(i, "__dataclass_fields__"):
+                            collect(i)
+                        elif isinstance(i, tuple):
+                            for j in i:
+                                if hasattr(j, "__dataclass_fields__"):
+                                    collect(j)
+                elif hasattr(v, "__dataclass_fields__"):
+                    collect(v)
+
+        for it in q.items:
+            if not it.star:
+                collect(it.expr)
+        if q.having is not None:
+            collect(q.having)
+        pre = pn.Projection(plan, tuple(pre_names),
+                            tuple(pre_exprs[n] for n in pre_names))
+        if keys:
+            agg = pn.Aggregate(pre, tuple(keys), tuple(agg_specs))
+        else:
+            agg = pn.Reduce(pre, tuple(agg_specs))
+        return agg, scope, key_map, agg_map
+
+    def _post_agg_expr(self, e, key_map, agg_map, scope: Scope) -> ex.Expr:
+        k = _ast_key(e)
+        if k in agg_map:
+            return ex.ColRef(agg_map[k])
+        if k in key_map:
+            return ex.ColRef(key_map[k])
+        if isinstance(e, ast.Col):
+            internal = scope.resolve(e.table, e.name)
+            if internal in key_map.values():
+                return ex.ColRef(internal)
+            raise ValueError(f"column {e.name} is neither grouped nor aggregated")
+        if isinstance(e, ast.Lit):
+            return self.expr(e, scope)
+        if isinstance(e, ast.Bin):
+            return ex.BinOp(e.op, self._post_agg_expr(e.left, key_map, agg_map, scope),
+                            self._post_agg_expr(e.right, key_map, agg_map, scope)) \
+                if e.op in ("add", "sub", "mul", "div", "mod") else \
+                (ex.BoolOp(e.op, self._post_agg_expr(e.left, key_map, agg_map, scope),
+                           self._post_agg_expr(e.right, key_map, agg_map, scope))
+                 if e.op in ("and", "or") else
+                 ex.Cmp(e.op, self._post_agg_expr(e.left, key_map, agg_map, scope),
+                        self._post_agg_expr(e.right, key_map, agg_map, scope)))
+        if isinstance(e, ast.Func) and e.name == "round":
+            return ex.RoundExpr(self._post_agg_expr(e.args[0], key_map, agg_map,
+                                                    scope),
+                                int(e.args[1].value) if len(e.args) > 1 else 0)
+        if isinstance(e, ast.CastE):
+            return ex.Cast(self._post_agg_expr(e.operand, key_map, agg_map,
+                                               scope), CAST_TYPES[e.to])
+        raise NotImplementedError(f"post-agg expr {e}")
+
+    # ------------------------------------------------------------------
+    def expr(self, e, scope: Scope) -> ex.Expr:
+        if isinstance(e, ast.Col):
+            return ex.ColRef(scope.resolve(e.table, e.name))
+        if isinstance(e, ast.Lit):
+            return _lit_expr(e)
+        if isinstance(e, ast.Bin):
+            l, r = self.expr(e.left, scope), self.expr(e.right, scope)
+            if e.op in ("and", "or"):
+                return ex.BoolOp(e.op, l, r)
+            if e.op in ("eq", "ne", "lt", "le", "gt", "ge"):
+                return ex.Cmp(e.op, l, r)
+            # date +/- interval
+            if e.op in ("add", "sub") and isinstance(e.right, ast.Lit) \
+                    and e.right.kind == "interval":
+                qty, unit = e.right.value
+                delta = _interval_to_timestamp_delta(e.left, qty, unit, e.op)
+                if delta is not None:
+                    return delta
+                raise NotImplementedError("interval arithmetic on columns")
+            return ex.BinOp(e.op, l, r)
+        if isinstance(e, ast.Un):
+            assert e.op == "not"
+            return ex.Not(self.expr(e.operand, scope))
+        if isinstance(e, ast.InE):
+            vals = tuple(v.value for v in e.values)
+            out = ex.IsIn(self.expr(e.operand, scope), vals)
+            return ex.Not(out) if e.negated else out
+        if isinstance(e, ast.BetweenE):
+            lo = ex.Cmp("ge", self.expr(e.operand, scope), self.expr(e.lo, scope))
+            hi = ex.Cmp("le", self.expr(e.operand, scope), self.expr(e.hi, scope))
+            out = ex.BoolOp("and", lo, hi)
+            return ex.Not(out) if e.negated else out
+        if isinstance(e, ast.LikeE):
+            out = _like_expr(self.expr(e.operand, scope), e.pattern)
+            return ex.Not(out) if e.negated else out
+        if isinstance(e, ast.IsNullE):
+            return ex.IsNull(self.expr(e.operand, scope), negate=e.negated)
+        if isinstance(e, ast.CastE):
+            if e.to not in CAST_TYPES:
+                raise NotImplementedError(f"CAST to {e.to}")
+            return ex.Cast(self.expr(e.operand, scope), CAST_TYPES[e.to])
+        if isinstance(e, ast.ExtractE):
+            return ex.DtField(self.expr(e.operand, scope), e.fld)
+        if isinstance(e, ast.CaseE):
+            conds = tuple(self.expr(c, scope) for c, _ in e.whens)
+            thens = tuple(self.expr(v, scope) for _, v in e.whens)
+            els = self.expr(e.els, scope) if e.els is not None else ex.Const(None, bt.float64)
+            return ex.Case(conds, thens, els)
+        if isinstance(e, ast.Func):
+            name = e.name
+            if name in ("year", "month", "day", "hour", "minute", "second",
+                        "quarter"):
+                return ex.DtField(self.expr(e.args[0], scope), name)
+            if name == "upper":
+                return ex.StrOp(self.expr(e.args[0], scope), "upper")
+            if name == "lower":
+                return ex.StrOp(self.expr(e.args[0], scope), "lower")
+            if name == "length":
+                return ex.StrOp(self.expr(e.args[0], scope), "len")
+            if name == "trim":
+                return ex.StrOp(self.expr(e.args[0], scope), "strip")
+            if name == "substring":
+                arg, start, ln = e.args
+                s = int(start.value) - 1
+                stop = None if ln is None else s + int(ln.value)
+                return ex.StrOp(self.expr(arg, scope), "slice", (s, stop, 1))
+            if name == "round":
+                return ex.RoundExpr(self.expr(e.args[0], scope),
+                                    int(e.args[1].value) if len(e.args) > 1 else 0)
+            if name == "abs":
+                inner = self.expr(e.args[0], scope)
+                zero = ex.Const(0)
+                return ex.Case((ex.Cmp("lt", inner, zero),),
+                               (ex.BinOp("sub", zero, inner),), inner)
+            if name == "coalesce":
+                args = [self.expr(a, scope) for a in e.args]
+                out = args[-1]
+                for a in reversed(args[:-1]):
+                    out = ex.Case((ex.IsNull(a, negate=True),), (a,), out)
+                return out
+            raise NotImplementedError(f"SQL function {name}")
+        raise NotImplementedError(f"expr {e}")
+
+
+def _interval_to_timestamp_delta(left_ast, qty, unit, op):
+    """Constant date +/- interval folds to a Timestamp constant."""
+    if isinstance(left_ast, ast.Lit) and left_ast.kind == "date":
+        base = pd.Timestamp(left_ast.value)
+        off = pd.DateOffset(**{unit + "s": qty})
+        res = base + off if op == "add" else base - off
+        return ex.Const(pd.Timestamp(res), bt.timestamp_ns)
+    return None
+
+
+def _lit_expr(e: ast.Lit) -> ex.Expr:
+    if e.kind == "date":
+        return ex.Const(pd.Timestamp(e.value), bt.timestamp_ns)
+    if e.kind == "null":
+        return ex.Const(None, bt.float64)
+    return ex.Const(e.value)
+
+
+def _like_expr(operand: ex.Expr, pattern: str) -> ex.Expr:
+    body = pattern.strip("%")
+    if "%" not in body and "_" not in body:
+        if pattern.startswith("%") and pattern.endswith("%"):
+            return ex.StrOp(operand, "contains", (body,))
+        if pattern.endswith("%"):
+            return ex.StrOp(operand, "startswith", (body,))
+        if pattern.startswith("%"):
+            return ex.StrOp(operand, "endswith", (body,))
+        return ex.Cmp("eq", operand, ex.Const(pattern, bt.string))
+    import re as _re
+
+    rx = _re.escape(pattern).replace("%", ".*").replace("_", ".")
+    return ex.StrOp(operand, "contains_re", ("^" + rx + "$",))
+
+
+def _split_conjuncts(e) -> list:
+    if isinstance(e, ast.Bin) and e.op == "and":
+        return _split_conjuncts(e.left) + _split_conjuncts(e.right)
+    return [e]
+
+
+def _has_agg(e) -> bool:
+    if isinstance(e, ast.Func) and e.name in AGG_FUNCS:
+        return True
+    for f in getattr(e, "__dataclass_fields__", {}):
+        v = getattr(e, f)
+        if isinstance(v, (list, tuple)):
+            for i in v:
+                if hasattr(i, "__dataclass_fields__") and _has_agg(i):
+                    return True
+                if isinstance(i, tuple):
+                    for j in i:
+                        if hasattr(j, "__dataclass_fields__") and _has_agg(j):
+                            return True
+        elif hasattr(v, "__dataclass_fields__") and _has_agg(v):
+            return True
+    return False
+
+
+def _ast_key(e) -> str:
+    return repr(e)
+
+
+def _default_name(e) -> str:
+    if isinstance(e, ast.Col):
+        return e.name
+    if isinstance(e, ast.Func):
+        return e.name
+    if isinstance(e, ast.ExtractE):
+        return e.fld
+    return "expr"

@@ -1,0 +1,212 @@
+"""SQL frontend tests: SQL results vs equivalent pandas (reference:
+BodoSQL test suites, e.g. test_agg_groupby.py / test_tpch-style)."""
+
+import os
+import sys
+
+import numpy as np
+import pandas as pd
+import pytest
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))), "benchmarks"))
+
+from bodo_amd.sql import BodoSQLContext  # noqa: E402
+
+
+def _df(n=2000, seed=0):
+    rng = np.random.default_rng(seed)
+    return pd.DataFrame({
+        "a": rng.integers(0, 10, n),
+        "b": rng.uniform(-1, 1, n),
+        "c": rng.choice(["x", "y", "z"], n),
+        "t": pd.to_datetime(pd.Timestamp("1995-01-01").value
+                            + rng.integers(0, 4 * 365 * 86400 * 10**9, n)),
+    })
+
+
+def run_sql(sql, tables):
+    bc = BodoSQLContext(tables)
+    return bc.sql(sql).to_pandas()
+
+
+def _cmp(got, exp, sort=True):
+    got = got.reset_index(drop=True).copy()
+    exp = exp.reset_index(drop=True).copy()
+    for c in exp.columns:
+        if exp[c].dtype == object or str(exp[c].dtype) == "category":
+            exp[c] = exp[c].astype(str)
+            got[c] = got[c].astype(str)
+    if sort:
+        cols = list(exp.columns)
+        got = got.sort_values(cols).reset_index(drop=True)
+        exp = exp.sort_values(cols).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False,
+                                  atol=1e-8, rtol=1e-8)
+
+
+def test_select_where():
+    df = _df()
+    got = run_sql("SELECT a, b FROM t1 WHERE a > 5 AND b < 0.5", {"t1": df})
+    exp = df[(df.a > 5) & (df.b < 0.5)][["a", "b"]]
+    _cmp(got, exp)
+
+
+def test_select_exprs():
+    df = _df()
+    got = run_sql(
+        "SELECT a + 1 AS a1, b * 2.0 AS b2, UPPER(c) AS cu FROM t1", {"t1": df})
+    exp = pd.DataFrame({"a1": df.a + 1, "b2": df.b * 2.0,
+                        "cu": df.c.str.upper()})
+    _cmp(got, exp)
+
+
+def test_group_by_agg():
+    df = _df()
+    got = run_sql(
+        "SELECT a, SUM(b) AS s, AVG(b) AS m, COUNT(*) AS n "
+        "FROM t1 GROUP BY a ORDER BY a", {"t1": df})
+    exp = df.groupby("a", as_index=False).agg(
+        s=("b", "sum"), m=("b", "mean"), n=("b", "size")).sort_values("a")
+    _cmp(got, exp, sort=False)
+
+
+def test_join_where_style():
+    rng = np.random.default_rng(5)
+    left = pd.DataFrame({"k": rng.integers(0, 30, 800),
+                         "v": rng.uniform(0, 1, 800)})
+    right = pd.DataFrame({"kk": np.arange(20), "w": rng.uniform(0, 1, 20)})
+    got = run_sql(
+        "SELECT k, v, w FROM l, r WHERE l.k = r.kk AND v > 0.5",
+        {"l": left, "r": right})
+    exp = left.merge(right, left_on="k", right_on="kk")
+    exp = exp[exp.v > 0.5][["k", "v", "w"]]
+    _cmp(got, exp)
+
+
+def test_explicit_join_on():
+    rng = np.random.default_rng(6)
+    left = pd.DataFrame({"k": rng.integers(0, 30, 500),
+                         "v": rng.uniform(0, 1, 500)})
+    right = pd.DataFrame({"kk": np.arange(25), "w": rng.uniform(0, 1, 25)})
+    got = run_sql(
+        "SELECT k, v, w FROM l JOIN r ON l.k = r.kk", {"l": left, "r": right})
+    exp = left.merge(right, left_on="k", right_on="kk")[["k", "v", "w"]]
+    _cmp(got, exp)
+
+
+def test_between_in_like_case():
+    df = _df()
+    got = run_sql(
+        "SELECT a, CASE WHEN b > 0 THEN 'pos' ELSE 'neg' END AS sign "
+        "FROM t1 WHERE a BETWEEN 2 AND 7 AND c IN ('x', 'y') "
+        "AND c LIKE 'x%'", {"t1": df})
+    f = df[(df.a >= 2) & (df.a <= 7) & df.c.isin(["x", "y"])
+           & df.c.str.startswith("x")]
+    exp = pd.DataFrame({"a": f.a,
+                        "sign": np.where(f.b > 0, "pos", "neg")})
+    _cmp(got, exp)
+
+
+def test_dates_extract():
+    df = _df()
+    got = run_sql(
+        "SELECT EXTRACT(year FROM t) AS y, COUNT(*) AS n FROM t1 "
+        "WHERE t >= DATE '1996-01-01' AND t < DATE '1997-01-01' "
+        "GROUP BY EXTRACT(year FROM t)", {"t1": df})
+    f = df[(df.t >= pd.Timestamp("1996-01-01"))
+           & (df.t < pd.Timestamp("1997-01-01"))]
+    exp = f.groupby(f.t.dt.year.rename("y"), as_index=False).size() \
+        .rename(columns={"size": "n"})
+    _cmp(got, exp)
+
+
+def test_tpch_q6_sql():
+    from tpch_data import gen_all
+
+    t = gen_all(0.02)
+    got = run_sql(
+        "SELECT SUM(l_extendedprice * l_discount) AS revenue "
+        "FROM lineitem WHERE l_shipdate >= DATE '1996-01-01' "
+        "AND l_shipdate < DATE '1996-01-01' + INTERVAL '1 year' "
+        "AND l_discount BETWEEN 0.08 AND 0.1 AND l_quantity < 24",
+        {"lineitem": t["lineitem"]})
+    li = t["lineitem"]
+    f = li[(li.L_SHIPDATE >= pd.Timestamp("1996-01-01"))
+           & (li.L_SHIPDATE < pd.Timestamp("1997-01-01"))
+           & (li.L_DISCOUNT >= 0.08) & (li.L_DISCOUNT <= 0.1)
+           & (li.L_QUANTITY < 24)]
+    exp_val = (f.L_EXTENDEDPRICE * f.L_DISCOUNT).sum()
+    assert abs(got["revenue"].iloc[0] - exp_val) < 1e-6
+
+
+def test_tpch_q1_sql():
+    from tpch_data import gen_all
+
+    t = gen_all(0.02)
+    got = run_sql("""
+        SELECT l_returnflag, l_linestatus,
+               SUM(l_quantity) AS sum_qty,
+               SUM(l_extendedprice) AS sum_base_price,
+               SUM(l_extendedprice * (1 - l_discount)) AS sum_disc_price,
+               SUM(l_extendedprice * (1 - l_discount) * (1 + l_tax)) AS sum_charge,
+               AVG(l_quantity) AS avg_qty,
+               AVG(l_extendedprice) AS avg_price,
+               AVG(l_discount) AS avg_disc,
+               COUNT(*) AS count_order
+        FROM lineitem
+        WHERE l_shipdate <= DATE '1998-09-02'
+        GROUP BY l_returnflag, l_linestatus
+        ORDER BY l_returnflag, l_linestatus
+    """, {"lineitem": t["lineitem"]})
+    li = t["lineitem"]
+    f = li[li.L_SHIPDATE <= pd.Timestamp("1998-09-02")].copy()
+    f["dp"] = f.L_EXTENDEDPRICE * (1 - f.L_DISCOUNT)
+    f["ch"] = f.dp * (1 + f.L_TAX)
+    exp = f.groupby(["L_RETURNFLAG", "L_LINESTATUS"], as_index=False).agg(
+        sum_qty=("L_QUANTITY", "sum"), sum_base_price=("L_EXTENDEDPRICE", "sum"),
+        sum_disc_price=("dp", "sum"), sum_charge=("ch", "sum"),
+        avg_qty=("L_QUANTITY", "mean"), avg_price=("L_EXTENDEDPRICE", "mean"),
+        avg_disc=("L_DISCOUNT", "mean"), count_order=("L_QUANTITY", "size"),
+    ).sort_values(["L_RETURNFLAG", "L_LINESTATUS"])
+    exp.columns = ["l_returnflag", "l_linestatus"] + list(exp.columns[2:])
+    _cmp(got, exp, sort=False)
+
+
+def test_tpch_q3_sql():
+    from tpch_data import gen_all
+
+    t = gen_all(0.02)
+    got = run_sql("""
+        SELECT o_orderkey,
+               SUM(l_extendedprice * (1 - l_discount)) AS revenue,
+               o_orderdate, o_shippriority
+        FROM customer, orders, lineitem
+        WHERE c_mktsegment = 'BUILDING' AND c_custkey = o_custkey
+          AND l_orderkey = o_orderkey AND o_orderdate < DATE '1995-03-15'
+          AND l_shipdate > DATE '1995-03-15'
+        GROUP BY o_orderkey, o_orderdate, o_shippriority
+        ORDER BY revenue DESC, o_orderdate LIMIT 10
+    """, {"customer": t["customer"], "orders": t["orders"],
+          "lineitem": t["lineitem"]})
+    from tpch_queries import q3
+
+    exp = q3(pd, {k: v.copy() for k, v in t.items()})
+    exp = exp.rename(columns={"L_ORDERKEY": "o_orderkey", "REVENUE": "revenue",
+                              "O_ORDERDATE": "o_orderdate",
+                              "O_SHIPPRIORITY": "o_shippriority"})
+    _cmp(got, exp[["o_orderkey", "revenue", "o_orderdate",
+                   "o_shippriority"]], sort=False)
+
+
+def test_having_distinct_limit():
+    df = _df()
+    got = run_sql(
+        "SELECT a, COUNT(*) AS n FROM t1 GROUP BY a HAVING COUNT(*) > 150 "
+        "ORDER BY a", {"t1": df})
+    g = df.groupby("a", as_index=False).size().rename(columns={"size": "n"})
+    exp = g[g.n > 150].sort_values("a")
+    _cmp(got, exp, sort=False)
+    got2 = run_sql("SELECT DISTINCT c FROM t1 ORDER BY c", {"t1": df})
+    exp2 = pd.DataFrame({"c": np.sort(df.c.unique())})
+    _cmp(got2, exp2, sort=False)
