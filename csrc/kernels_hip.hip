@@ -762,7 +762,10 @@ std::vector<torch::Tensor> join_probe(
 
 // ---------------------------------------------------------------------
 
+torch::Tensor gemm_f32(torch::Tensor A, torch::Tensor B);  // gemm.hip
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm_f32", &gemm_f32, "f32 MFMA GEMM (v_mfma_f32_16x16x4_f32)");
   m.def("hash_columns", &hash_columns, "multi-column row hash");
   m.def("dt_field", &dt_field, "datetime field extraction");
   m.def("gather_string", &gather_string, "string column gather");
