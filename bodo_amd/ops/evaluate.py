@@ -590,6 +590,13 @@ def udf_map(a: Column, func, na_action=None) -> Column:
         lut = torch.tensor([np.nan if v is None else v for v in vals],
                            dtype=torch.float64, device=a.device)
         return Column(bt.float64, lut[a.data.long()], a.mask)
+    if a.is_cuda and a.dtype.is_float:
+        # numeric UDF on device: lower to a hipRTC-compiled HIP kernel
+        from ..jit.hip_udf import try_hip_udf
+
+        res = try_hip_udf(func, a.data)
+        if res is not None:
+            return Column(bt.float64, res, a.mask)
     if a.dtype.is_integer or a.dtype.kind == TypeKind.BOOL:
         # small dense domain: dense LUT indexed by value (no sort; one
         # min/max reduction + one gather — the hot path for e.g. hour buckets)
