@@ -64,14 +64,25 @@ def execute(plan: pn.PlanNode, ctx: Optional[ExecutionContext] = None) -> Table:
 
     ctx = ctx or ExecutionContext()
     plan = optimize(plan)
-    return _exec(plan, ctx)
+    out = _exec(plan, ctx)
+    from ..utils import query_profile as qp
+
+    qp.finish_query()
+    return out
 
 
 def _exec(node: pn.PlanNode, ctx: ExecutionContext) -> Table:
     h = _HANDLERS.get(type(node))
     if h is None:
         raise NotImplementedError(f"no executor for {type(node).__name__}")
-    return h(node, ctx)
+    from ..utils import query_profile as qp
+    from ..utils import tracing
+
+    name = type(node).__name__
+    with tracing.Event(f"exec.{name}"), qp.OpTimer(name) as t:
+        out = h(node, ctx)
+        t.rows_out = len(out) if out is not None else -1
+    return out
 
 
 # ---------------------------------------------------------------- sources

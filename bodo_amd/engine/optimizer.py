@@ -50,6 +50,11 @@ def push_filters(node: pn.PlanNode) -> pn.PlanNode:
             pushable = [c for c in conjuncts if _arrow_convertible(c)]
             rest = [c for c in conjuncts if not _arrow_convertible(c)]
             if pushable:
+                from ..user_logging import log_message
+
+                log_message("Filter Pushdown",
+                            f"pushed {len(pushable)} filter(s) into parquet "
+                            f"scan of {child.path}")
                 new_scan = pn.ParquetScan(
                     child.path, child.columns,
                     child.filters + tuple(pushable), child.schema_names)

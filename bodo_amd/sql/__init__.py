@@ -30,13 +30,23 @@ class BodoSQLContext:
         new.pop(name, None)
         return BodoSQLContext(new)
 
+    _plan_cache = {}  # (reference: BodoSqlPlanCache, bodo/sql_plan_cache.py)
+
     def sql(self, query: str):
         from ..pandas.frame import BodoDataFrame
         from .parser import Parser
         from .planner import Planner
 
-        q = Parser(query).parse()
-        plan, names = Planner(self.tables).plan(q)
+        key = (query, tuple(sorted((n, id(t._lazy_plan))
+                                   for n, t in self.tables.items())))
+        hit = BodoSQLContext._plan_cache.get(key)
+        if hit is None:
+            q = Parser(query).parse()
+            hit = Planner(self.tables).plan(q)
+            if len(BodoSQLContext._plan_cache) > 256:
+                BodoSQLContext._plan_cache.clear()
+            BodoSQLContext._plan_cache[key] = hit
+        plan, names = hit
         return BodoDataFrame(plan, names)
 
     # convenience parity alias

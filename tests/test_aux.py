@@ -1,0 +1,68 @@
+"""Auxiliary subsystem tests: tracing, query profile, user logging, plan
+cache (reference: bodo/utils/tracing.pyx, _query_profile_collector,
+user_logging, sql_plan_cache)."""
+
+import json
+import os
+
+import numpy as np
+import pandas as pd
+
+import bodo_amd.pandas as bpd
+
+
+def test_tracing(tmp_path):
+    from bodo_amd.utils import tracing
+
+    tracing.start_tracing()
+    df = pd.DataFrame({"a": np.arange(100), "b": np.arange(100) * 0.5})
+    b = bpd.from_pandas(df)
+    b.groupby("a", as_index=False).agg(s=bpd.NamedAgg("b", "sum")).to_pandas()
+    with tracing.Event("custom", foo=1):
+        pass
+    agg = tracing.aggregate_events()
+    assert any(e["name"].startswith("exec.") for e in agg)
+    p = str(tmp_path / "trace.json")
+    tracing.dump(p)
+    data = json.load(open(p))
+    assert len(data["traceEvents"]) > 0
+    tracing.stop_tracing()
+
+
+def test_query_profile(tmp_path):
+    from bodo_amd.utils import query_profile as qp
+
+    qp.clear()
+    qp.enable(str(tmp_path))
+    df = pd.DataFrame({"a": np.arange(50), "b": np.arange(50) * 2.0})
+    b = bpd.from_pandas(df)
+    b[b.a > 10].to_pandas()
+    recs = qp.get_records()
+    assert any(r["operator"] == "Filter" for r in recs)
+    qp.flush(str(tmp_path))
+    assert os.path.exists(str(tmp_path / "query_profile_rank0.json"))
+    qp.clear()
+
+
+def test_user_logging(capsys):
+    import logging
+
+    from bodo_amd import user_logging
+
+    user_logging.set_verbose_level(2)
+    user_logging.log_message("Test", "hello %s", "world")
+    user_logging.set_verbose_level(0)
+    out = capsys.readouterr().out
+    assert "hello world" in out
+
+
+def test_sql_plan_cache():
+    from bodo_amd.sql import BodoSQLContext
+
+    df = pd.DataFrame({"a": np.arange(30), "b": np.arange(30) * 1.5})
+    bc = BodoSQLContext({"t": df})
+    r1 = bc.sql("SELECT a FROM t WHERE a > 5")
+    n0 = len(BodoSQLContext._plan_cache)
+    r2 = bc.sql("SELECT a FROM t WHERE a > 5")
+    assert len(BodoSQLContext._plan_cache) == n0
+    assert r1._lazy_plan is r2._lazy_plan
