@@ -64,6 +64,17 @@ def _expr_to_arrow(e: Expr):
 
 def read_shard(path: str, columns: Optional[Sequence[str]],
                filters: Sequence[Expr], ctx) -> Table:
+    if ctx.device.type == "cuda" and not filters:
+        # on-GPU decode fast path (uncompressed PLAIN / RLE_DICTIONARY)
+        from . import parquet_gpu
+
+        try:
+            t = parquet_gpu.read_shard_gpu(path, list(columns) if columns
+                                           else None, ctx)
+        except Exception:
+            t = None
+        if t is not None:
+            return t
     d = _dataset(path)
     frags = list(d.get_fragments())
     # explode into row-group pieces and block-assign to ranks

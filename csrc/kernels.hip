@@ -760,11 +760,75 @@ std::vector<torch::Tensor> join_probe(
 }
 
 // ---------------------------------------------------------------------
+// Parquet RLE/bit-packed hybrid expansion (on-GPU parquet decode path;
+// reference role: cudf::io::read_parquet in gpu_read_parquet.h, here a
+// hand-written gfx950 kernel).  Run table parsed on host from page bytes:
+// kind 0 = RLE run (value repeated count times), kind 1 = bit-packed group
+// (count values packed at `bitwidth` starting at bit_offset).
+
+struct RleRun {
+  int64_t out_start;
+  int32_t count;
+  int32_t kind;
+  int64_t value_or_bitoff;
+};
+
+__global__ void rle_expand_kernel(const RleRun* __restrict__ runs, int n_runs,
+                                  const uint8_t* __restrict__ packed,
+                                  int bitwidth, int32_t* __restrict__ out) {
+  // one wave per run slot, grid-stride over runs; lanes fill values
+  int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  for (int64_t r = wave_id; r < n_runs; r += n_waves) {
+    RleRun run = runs[r];
+    if (run.kind == 0) {
+      int32_t v = (int32_t)run.value_or_bitoff;
+      for (int64_t i = lane; i < run.count; i += WAVE) {
+        out[run.out_start + i] = v;
+      }
+    } else {
+      int64_t bit0 = run.value_or_bitoff;
+      for (int64_t i = lane; i < run.count; i += WAVE) {
+        int64_t bp = bit0 + i * bitwidth;
+        int64_t byte = bp >> 3;
+        int shift = (int)(bp & 7);
+        // up to 32-bit reads cover bitwidth <= 24; parquet dict ids fit
+        uint32_t w = (uint32_t)packed[byte]
+                     | ((uint32_t)packed[byte + 1] << 8)
+                     | ((uint32_t)packed[byte + 2] << 16)
+                     | ((uint32_t)packed[byte + 3] << 24);
+        out[run.out_start + i] = (int32_t)((w >> shift)
+                                           & ((1u << bitwidth) - 1));
+      }
+    }
+  }
+}
+
+torch::Tensor rle_expand(torch::Tensor runs_blob, int64_t n_runs,
+                         torch::Tensor packed, int64_t bitwidth,
+                         int64_t n_out) {
+  auto dev = packed.device();
+  auto out = torch::empty({n_out}, torch::dtype(torch::kInt32).device(dev));
+  if (n_runs) {
+    int block = 256;
+    int waves_per_block = block / WAVE;
+    int grid = (int)std::min<int64_t>(
+        (n_runs + waves_per_block - 1) / waves_per_block, 2048);
+    hipLaunchKernelGGL(rle_expand_kernel, dim3(grid), dim3(block), 0,
+                       cur_stream(), (const RleRun*)runs_blob.data_ptr(),
+                       (int)n_runs, (const uint8_t*)packed.data_ptr(),
+                       (int)bitwidth, (int32_t*)out.data_ptr());
+    CHECK_HIP(hipGetLastError());
+  }
+  return out;
+}
 
 torch::Tensor gemm_f32(torch::Tensor A, torch::Tensor B);  // gemm.hip
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_f32", &gemm_f32, "f32 MFMA GEMM (v_mfma_f32_16x16x4_f32)");
+  m.def("rle_expand", &rle_expand, "parquet RLE/bit-packed hybrid expand");
   m.def("hash_columns", &hash_columns, "multi-column row hash");
   m.def("dt_field", &dt_field, "datetime field extraction");
   m.def("gather_string", &gather_string, "string column gather");
