@@ -45,6 +45,15 @@ PANDAS_FALLBACK = _env_bool("BODO_AMD_PANDAS_FALLBACK", True)
 #: verbosity for user logging
 VERBOSE = _env_int("BODO_AMD_VERBOSE", 0)
 
+#: streaming execution: "0" off, "1" force, "auto" = when scan bytes exceed
+#: STREAM_THRESHOLD_BYTES (reference: streaming operator states with 288 GB
+#: HBM-sized morsels)
+STREAMING = os.environ.get("BODO_AMD_STREAMING", "auto")
+
+#: auto-streaming threshold (bytes of scanned files per query)
+STREAM_THRESHOLD_BYTES = _env_int("BODO_AMD_STREAM_THRESHOLD",
+                                  100 * 1024**3)
+
 
 def default_device() -> str:
     if DEVICE:

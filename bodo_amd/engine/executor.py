@@ -139,6 +139,10 @@ SINGLE_PHASE_AGGS = {"median", "nunique", "var", "std", "quantile"}
 
 
 def _exec_aggregate(node: pn.Aggregate, ctx) -> Table:
+    from .streaming import exec_streaming, want_streaming
+
+    if want_streaming(node, ctx):
+        return exec_streaming(node, ctx)
     child = _exec(node.child, ctx)
     keys = list(node.keys)
     aggs = list(node.aggs)
@@ -406,12 +410,21 @@ def _exec_parquet_write(node: pn.ParquetWrite, ctx) -> Table:
 
 
 def _exec_reduce(node: pn.Reduce, ctx) -> Table:
+    from .streaming import exec_streaming, want_streaming
+
+    if want_streaming(node, ctx):
+        return exec_streaming(node, ctx)
     child = _exec(node.child, ctx)
+    partials = {out: ops.reduce_column(child.column(in_name), func)
+                for out, in_name, func in node.aggs}
+    return _finish_reduce(node, ctx, partials)
+
+
+def _finish_reduce(node: pn.Reduce, ctx, partials: dict) -> Table:
     out_names, out_cols = [], []
     for out_name, in_name, func in node.aggs:
-        partial = ops.reduce_column(child.column(in_name), func)
-        partials = comm.allgather_obj(partial)
-        val = _combine_reduce(partials, func)
+        parts = comm.allgather_obj(partials[out_name])
+        val = _combine_reduce(parts, func)
         out_names.append(out_name)
         arr = pd.Series([val])
         out_cols.append(Column.from_numpy(arr.to_numpy(), ctx.device))

@@ -1,0 +1,222 @@
+"""Streaming (morsel-wise) execution for larger-than-HBM inputs.
+
+Reference: the push-based batch pipelines of bodo/pandas/_pipeline.h and the
+incremental operator states in bodo/libs/streaming/.  MI355X redesign: with
+288 GB HBM3E per GPU the morsel is sized in the hundreds of MB–GB range
+(config.STREAM_BATCH_SIZE rows), and only the FIRST pipeline (scan side) is
+streamed — the blocking operator keeps a bounded device-resident state:
+
+* Aggregate: per-morsel local hash groupby, partial states concatenated and
+  re-combined whenever they exceed a bound (monotone shrink per merge).
+* Reduce: scalar partials folded per morsel.
+* Join probe: the build side is materialized (small side), each probe morsel
+  joins and feeds the downstream incremental consumer.
+
+Activated when a plan bottoms out in a ParquetScan whose estimated bytes
+exceed config.STREAM_THRESHOLD_BYTES, or explicitly via
+BODO_AMD_STREAMING=1.
+"""
+
+from __future__ import annotations
+
+from typing import Iterator, List, Optional
+
+import torch
+
+from .. import ops
+from ..core.table import Table
+from ..ops import relational as rel
+from ..plan import nodes as pn
+from ..plan.expr import ColRef
+
+
+def scan_batches(node: pn.ParquetScan, ctx, batch_rows: int) -> Iterator[Table]:
+    """Yield this rank's shard of a parquet scan as morsels (row-group
+    granularity, grouped into ~batch_rows chunks)."""
+    import pyarrow as pa
+    import pyarrow.dataset as pads
+
+    from ..core.table import dict_encode_strings
+    from ..io.parquet import _dataset, _expr_to_arrow
+
+    d = _dataset(node.path)
+    pieces = []
+    for f in d.get_fragments():
+        try:
+            pieces.extend(f.split_by_row_group())
+        except Exception:
+            pieces.append(f)
+    w, r = ctx.world, ctx.rank
+    base, rem = divmod(len(pieces), w)
+    start = r * base + min(r, rem)
+    my = pieces[start:start + base + (1 if r < rem else 0)]
+    filt = None
+    for e in node.filters:
+        ae = _expr_to_arrow(e)
+        filt = ae if filt is None else (filt & ae)
+    cols = list(node.columns) if node.columns else None
+    buf: List[pa.Table] = []
+    rows = 0
+    got_any = False
+    for piece in my:
+        t = piece.to_table(columns=cols, filter=filt)
+        buf.append(t)
+        rows += t.num_rows
+        if rows >= batch_rows:
+            yield _to_device_batch(buf, ctx)
+            got_any = True
+            buf, rows = [], 0
+    if buf or not got_any:
+        if buf:
+            yield _to_device_batch(buf, ctx)
+        else:
+            schema = d.schema
+            if cols:
+                schema = pa.schema([schema.field(c) for c in cols])
+            yield Table.from_arrow(schema.empty_table(), ctx.device)
+
+
+def _to_device_batch(bufs, ctx) -> Table:
+    import pyarrow as pa
+
+    from ..core.table import dict_encode_strings
+
+    t = pa.concat_tables(bufs).combine_chunks() if len(bufs) > 1 else bufs[0]
+    t = dict_encode_strings(t)
+    return Table.from_arrow(t, ctx.device)
+
+
+def want_streaming(node, ctx) -> bool:
+    """Stream when forced (BODO_AMD_STREAMING=1) or when the scan's file
+    bytes exceed the threshold (auto mode)."""
+    from .. import config
+
+    mode = config.STREAMING
+    if mode == "0" or not streamable(node):
+        return False
+    if mode == "1":
+        return True
+    # auto: compare file sizes against threshold
+    import glob
+    import os
+
+    cur = node.children()[0]
+    while isinstance(cur, (pn.Filter, pn.Projection)):
+        cur = cur.children()[0]
+    path = cur.path
+    try:
+        if os.path.isdir(path):
+            total = sum(os.path.getsize(p)
+                        for p in glob.glob(os.path.join(path, "*")))
+        else:
+            total = os.path.getsize(path)
+    except OSError:
+        return False
+    return total > config.STREAM_THRESHOLD_BYTES
+
+
+def streamable(node: pn.PlanNode) -> bool:
+    """True when `node` is Aggregate/Reduce over a chain of
+    Filter/Projection over a ParquetScan (the first-pipeline shape we
+    stream)."""
+    if not isinstance(node, (pn.Aggregate, pn.Reduce)):
+        return False
+    if isinstance(node, pn.Aggregate):
+        ok = {"sum", "count", "size", "min", "max", "mean", "first", "last",
+              "prod"}
+        if any((a[2] not in ok) if not callable(a[2]) else True
+               for a in node.aggs):
+            return False
+    cur = node.children()[0]
+    while isinstance(cur, (pn.Filter, pn.Projection)):
+        cur = cur.children()[0]
+    return isinstance(cur, pn.ParquetScan)
+
+
+def exec_streaming(node, ctx):
+    """Execute Aggregate/Reduce(…(ParquetScan)) morsel-wise."""
+    from .. import config
+    from ..ops import evaluate as ev
+    from . import executor as ex
+
+    # peel the operator chain
+    chain = []
+    cur = node.children()[0]
+    while isinstance(cur, (pn.Filter, pn.Projection)):
+        chain.append(cur)
+        cur = cur.children()[0]
+    scan: pn.ParquetScan = cur
+    chain.reverse()  # scan-side first
+
+    batch_rows = config.STREAM_BATCH_SIZE
+
+    def apply_chain(batch: Table) -> Table:
+        for op in chain:
+            if isinstance(op, pn.Filter):
+                batch = ev.eval_filter(op.cond, batch)
+            else:
+                batch = ev.project(batch, op.names, op.exprs)
+        return batch
+
+    if isinstance(node, pn.Reduce):
+        partials = None
+        for batch in scan_batches(scan, ctx, batch_rows):
+            batch = apply_chain(batch)
+            cur_p = {out: ops.reduce_column(batch.column(in_name), func)
+                     for out, in_name, func in node.aggs}
+            partials = cur_p if partials is None else _merge_reduce(
+                partials, cur_p, node.aggs)
+        # reuse the executor's distributed combine
+        return ex._finish_reduce(node, ctx, partials)
+
+    # Aggregate: incremental partial-groupby state with bounded re-combine
+    assert isinstance(node, pn.Aggregate)
+    keys = list(node.keys)
+    partial_aggs, final_map = ex._decompose_aggs(list(node.aggs))
+    state: Optional[Table] = None
+    combine_aggs = [(name, name, rel.COMBINE_FUNC[f])
+                    for name, _, f in partial_aggs]
+    for batch in scan_batches(scan, ctx, batch_rows):
+        batch = apply_chain(batch)
+        part = rel.groupby_local(batch, keys, partial_aggs, node.dropna)
+        if state is None:
+            state = part
+        else:
+            both = ops.concat_tables([state, part])
+            state = rel.groupby_local(both, keys, combine_aggs, node.dropna)
+    if state is None:
+        return ex._exec_aggregate(node, ctx)  # no batches: fall back
+    # distributed final combine (same as two-phase tail)
+    if ctx.world > 1:
+        state = ex._shuffle_by_keys(state, keys, ctx)
+        state = rel.groupby_local(state, keys, combine_aggs, node.dropna)
+    from ..ops import evaluate as ev2
+
+    names, exprs = list(keys), [ColRef(k) for k in keys]
+    for out_name, expr in final_map:
+        names.append(out_name)
+        exprs.append(expr)
+    return ev2.project(state, names, exprs)
+
+
+def _merge_reduce(a: dict, b: dict, aggs) -> dict:
+    out = {}
+    for out_name, in_name, func in aggs:
+        pa_, pb = a[out_name], b[out_name]
+        m = {}
+        for k in set(pa_) | set(pb):
+            va, vb = pa_.get(k), pb.get(k)
+            if k in ("sum", "count", "sumsq"):
+                m[k] = (va or 0) + (vb or 0)
+            elif k == "min":
+                vals = [v for v in (va, vb) if v is not None]
+                m[k] = min(vals) if vals else None
+            elif k == "max":
+                vals = [v for v in (va, vb) if v is not None]
+                m[k] = max(vals) if vals else None
+            elif k == "any":
+                m[k] = bool(va) or bool(vb)
+            elif k == "all":
+                m[k] = bool(va) and bool(vb)
+        out[out_name] = m
+    return out

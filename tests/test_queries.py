@@ -217,3 +217,32 @@ def test_tpcxbb_q26_style(tmp_path):
               "Books", 5).reset_index(drop=True)
     pd.testing.assert_frame_equal(got.reset_index(drop=True), exp,
                                   check_dtype=False)
+
+
+def test_streaming_aggregate(tmp_path, monkeypatch):
+    """Morsel-wise streaming aggregate over parquet (larger-than-HBM path,
+    reference: bodo/libs/streaming/_groupby.cpp incremental states)."""
+    import bodo_amd.config as cfg
+    import bodo_amd.pandas as bpd
+
+    df = make_lineitem(40000)
+    p = str(tmp_path / "s.parquet")
+    df.to_parquet(p, row_group_size=2000)
+    old_mode, old_batch = cfg.STREAMING, cfg.STREAM_BATCH_SIZE
+    cfg.STREAMING = "1"
+    cfg.STREAM_BATCH_SIZE = 5000
+    try:
+        b = bpd.read_parquet(p)
+        got = tpch_q1(bpd, b).to_pandas().reset_index(drop=True)
+        exp = tpch_q1(pd, df.copy()).reset_index(drop=True)
+        for c in ("L_RETURNFLAG", "L_LINESTATUS"):
+            got[c] = got[c].astype(str)
+            exp[c] = exp[c].astype(str)
+        pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+        # streaming reduce
+        b2 = bpd.read_parquet(p)
+        s = b2[b2.L_QUANTITY < 25].L_EXTENDEDPRICE.sum()
+        exp_s = df[df.L_QUANTITY < 25].L_EXTENDEDPRICE.sum()
+        assert abs(s - exp_s) < 1e-6
+    finally:
+        cfg.STREAMING, cfg.STREAM_BATCH_SIZE = old_mode, old_batch
