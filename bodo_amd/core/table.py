@@ -118,8 +118,8 @@ class Table:
         return f"Table({len(self)} rows x {self.num_columns} cols, device={self.device})"
 
 
-def dict_encode_strings(tbl: pa.Table, threshold: float = 0.5,
-                        sample: int = 8192) -> pa.Table:
+def dict_encode_strings(tbl: pa.Table, threshold: float = 0.7,
+                        sample: int = 8192, small_table: int = 65536) -> pa.Table:
     """Dictionary-encode low-cardinality string columns so filters/joins/
     groupbys run on int32 codes on device (reference: dict-encoded string
     arrays, bodo/libs/dict_arr_ext.py).
@@ -139,8 +139,11 @@ def dict_encode_strings(tbl: pa.Table, threshold: float = 0.5,
                 nuniq = len(pc.unique(head.combine_chunks()))
             except Exception:
                 nuniq = len(head)
-            # empty shards vote yes so non-empty shards decide
-            w = len(head) == 0 or nuniq <= max(1, int(len(head) * threshold))
+            # empty shards vote yes so non-empty shards decide; small tables
+            # always encode (tiny dictionaries, and their values are the
+            # common equality-filter targets, e.g. nation/region names)
+            w = (len(head) == 0 or len(col) <= small_table
+                 or nuniq <= max(1, int(len(head) * threshold)))
         want.append(w)
     from ..parallel import comm
 
