@@ -529,11 +529,18 @@ def str_op(a: Column, op: str, args) -> Column:
     if a.dtype.kind == TypeKind.DICT:
         # operate on the (small) dictionary, keep indices
         d = a.dictionary
-        if op in ("lower", "upper", "strip", "title", "capitalize"):
+        if op in ("lower", "upper", "strip", "title", "capitalize", "slice"):
             f = {"lower": pc.utf8_lower, "upper": pc.utf8_upper,
                  "strip": pc.utf8_trim_whitespace, "title": pc.utf8_title,
-                 "capitalize": pc.utf8_capitalize}[op]
-            return Column(a.dtype, a.data, a.mask, dictionary=f(d), length=len(a))
+                 "capitalize": pc.utf8_capitalize,
+                 "slice": lambda x: pc.utf8_slice_codeunits(
+                     x, args[0], None if len(args) < 2 or args[1] is None
+                     else args[1], args[2] if len(args) > 2 and args[2] else 1),
+                 }[op]
+            nd = f(d)
+            if not __import__("pyarrow").types.is_large_string(nd.type):
+                nd = nd.cast(__import__("pyarrow").large_string())
+            return Column(a.dtype, a.data, a.mask, dictionary=nd, length=len(a))
         if op in ("contains", "contains_re", "match", "startswith",
                   "endswith", "len"):
             if op == "len":
