@@ -537,9 +537,22 @@ def str_op(a: Column, op: str, args) -> Column:
                      x, args[0], None if len(args) < 2 or args[1] is None
                      else args[1], args[2] if len(args) > 2 and args[2] else 1),
                  }[op]
+            import pyarrow as pa
+
             nd = f(d)
-            if not __import__("pyarrow").types.is_large_string(nd.type):
-                nd = nd.cast(__import__("pyarrow").large_string())
+            if not pa.types.is_large_string(nd.type):
+                nd = nd.cast(pa.large_string())
+            vals = nd.to_pylist()
+            if len(set(vals)) != len(vals):
+                # transformed values collapsed: re-unify the dictionary
+                uniq = list(dict.fromkeys(vals))
+                code_of = {v: i for i, v in enumerate(uniq)}
+                remap = torch.tensor([code_of[v] for v in vals],
+                                     dtype=torch.int32, device=a.device)
+                codes = remap[a.data.long()]
+                return Column(a.dtype, codes, a.mask,
+                              dictionary=pa.array(uniq, type=pa.large_string()),
+                              length=len(a))
             return Column(a.dtype, a.data, a.mask, dictionary=nd, length=len(a))
         if op in ("contains", "contains_re", "match", "startswith",
                   "endswith", "len"):
