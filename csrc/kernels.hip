@@ -287,18 +287,22 @@ std::vector<torch::Tensor> gather_string(torch::Tensor bytes,
 __global__ void gb_insert_kernel(const ColumnDesc* __restrict__ cols, int ncols,
                                  const uint64_t* __restrict__ hashes,
                                  uint32_t* __restrict__ slots, uint64_t cap_mask,
-                                 uint32_t* __restrict__ row_slot, int64_t n) {
+                                 uint32_t* __restrict__ row_slot, int64_t n,
+                                 int64_t row0, const uint64_t* __restrict__ all_hashes) {
+  // rows [row0, row0+n) of the table; hashes/row_slot pre-offset by caller,
+  // all_hashes/cols indexed globally (slots store global row+1)
   GRID_STRIDE_LOOP(i, n) {
     uint64_t h = hashes[i];
+    int64_t gi = row0 + i;
     uint64_t s = h & cap_mask;
     while (true) {
-      uint32_t old = atomicCAS(&slots[s], 0u, (uint32_t)(i + 1));
+      uint32_t old = atomicCAS(&slots[s], 0u, (uint32_t)(gi + 1));
       if (old == 0u) {  // we claimed the slot: new group
         row_slot[i] = (uint32_t)s;
         break;
       }
       int64_t cand = (int64_t)old - 1;
-      if (hashes[cand] == h && rows_eq(cols, ncols, cand, i)) {
+      if (all_hashes[cand] == h && rows_eq(cols, ncols, cand, gi)) {
         row_slot[i] = (uint32_t)s;
         break;
       }
@@ -365,17 +369,48 @@ std::vector<torch::Tensor> groupby_build(
     std::vector<int64_t> dtypes, int64_t n, torch::Tensor hashes) {
   auto dev = datas[0].device();
   auto ds = build_descset(datas, masks, offsets, auxs, dtypes, n);
+  // Start with an Infinity-Cache-resident table (2^26 slots = 256 MB):
+  // the random atomicCAS probes then hit L3 instead of HBM (guide §2).
+  // Insert in chunks, checking the load factor between chunks; grow 8x and
+  // re-insert when it crosses 1/2 (few groupbys exceed 32M groups/rank).
   int64_t cap = 16;
   while (cap < 2 * n) cap <<= 1;
-  auto slots = torch::zeros({cap}, torch::dtype(torch::kInt32).device(dev));
-  auto row_slot = torch::empty({n}, torch::dtype(torch::kInt32).device(dev));
+  if (cap > (1 << 26)) cap = 1 << 26;
   int block = 256;
-  hipLaunchKernelGGL(gb_insert_kernel, dim3(grid_for(n, block)), dim3(block),
-                     0, cur_stream(), ds.ptr(), (int)datas.size(),
-                     (const uint64_t*)hashes.data_ptr(),
-                     (uint32_t*)slots.data_ptr(), (uint64_t)(cap - 1),
-                     (uint32_t*)row_slot.data_ptr(), n);
-  CHECK_HIP(hipGetLastError());
+  auto row_slot = torch::empty({n}, torch::dtype(torch::kInt32).device(dev));
+  torch::Tensor slots;
+  const int64_t chunk_rows = 1 << 27;
+  while (true) {
+    slots = torch::zeros({cap}, torch::dtype(torch::kInt32).device(dev));
+    bool grown = false;
+    for (int64_t start = 0; start < n; start += chunk_rows) {
+      int64_t cnt = std::min(chunk_rows, n - start);
+      hipLaunchKernelGGL(gb_insert_kernel, dim3(grid_for(cnt, block)),
+                         dim3(block), 0, cur_stream(), ds.ptr(),
+                         (int)datas.size(),
+                         (const uint64_t*)hashes.data_ptr() + start,
+                         (uint32_t*)slots.data_ptr(), (uint64_t)(cap - 1),
+                         (uint32_t*)row_slot.data_ptr() + start, cnt, start,
+                         (const uint64_t*)hashes.data_ptr());
+      CHECK_HIP(hipGetLastError());
+      if (start + cnt < n && cap < 2 * n) {
+        int64_t occupied = (slots != 0).sum().item<int64_t>();
+        if (occupied * 2 > cap) {
+          int64_t want = cap * 8;
+          while (want < 2 * std::min(n, occupied * 8)) want <<= 1;
+          if (want > 2 * n) {
+            int64_t c2 = 16;
+            while (c2 < 2 * n) c2 <<= 1;
+            want = c2;
+          }
+          cap = want;
+          grown = true;
+          break;  // rebuild from scratch at the bigger capacity
+        }
+      }
+    }
+    if (!grown) break;
+  }
   auto slot_gid = torch::empty({cap}, torch::dtype(torch::kInt32).device(dev));
   auto uniq_rows = torch::empty({n > 0 ? n : 1},
                                 torch::dtype(torch::kInt64).device(dev));
