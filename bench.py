@@ -51,11 +51,15 @@ def make_trips_shard(n, rank, device):
     lic = torch.randint(0, 4, (n,), generator=g, dtype=torch.int32, device=dev)
     lic_dict = pa.array(["HV0002", "HV0003", "HV0004", "HV0005"],
                         type=pa.large_string())
+    pu_col = Column(bt.int64, pu)
+    pu_col.val_range = (1, N_LOCATIONS)
+    do_col = Column(bt.int64, do)
+    do_col.val_range = (1, N_LOCATIONS)
     cols = {
         "hvfhs_license_num": Column(bt.dictionary, lic, dictionary=lic_dict),
         "pickup_datetime": Column(bt.timestamp_ns, pickup),
-        "PULocationID": Column(bt.int64, pu),
-        "DOLocationID": Column(bt.int64, do),
+        "PULocationID": pu_col,
+        "DOLocationID": do_col,
         "trip_miles": Column(bt.float64, miles),
     }
     return Table(list(cols), list(cols.values()), n)

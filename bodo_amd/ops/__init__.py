@@ -50,7 +50,9 @@ def gather(col: Column, idx: torch.Tensor) -> Column:
         return _gather_string_arrow(col, idx)
     data = col.data[idx]
     mask = col.mask[idx] if col.mask is not None else None
-    return Column(col.dtype, data, mask, dictionary=col.dictionary, length=n)
+    out = Column(col.dtype, data, mask, dictionary=col.dictionary, length=n)
+    out.val_range = col.val_range
+    return out
 
 
 def _gather_string_arrow(col: Column, idx: torch.Tensor) -> Column:
@@ -143,8 +145,12 @@ def concat_columns(cols: Sequence[Column]) -> Column:
         for d in datas[1:]:
             t = torch.promote_types(t, d.dtype)
         datas = [d.to(t) for d in datas]
-    return Column(first.dtype, torch.cat(datas), masks,
-                  dictionary=first.dictionary, length=n)
+    out = Column(first.dtype, torch.cat(datas), masks,
+                 dictionary=first.dictionary, length=n)
+    if all(c.val_range is not None for c in cols):
+        out.val_range = (min(c.val_range[0] for c in cols),
+                         max(c.val_range[1] for c in cols))
+    return out
 
 
 def concat_tables(tables: Sequence[Table]) -> Table:

@@ -164,11 +164,42 @@ def _keys_valid_mask(cols: Sequence[Column]) -> Optional[torch.Tensor]:
     return m
 
 
+def _try_pack_keys(key_cols: Sequence[Column]) -> Optional[Column]:
+    """Pack small-range integer/dict/bool keys into ONE int64 column so the
+    hash-table equality check reads 8 bytes instead of one random read per
+    key column (reference analog: key normalization in _hash_join.cpp)."""
+    shift = 0
+    packed = None
+    for c in key_cols:
+        if c.mask is not None:
+            return None
+        k = c.dtype.kind
+        if k == TypeKind.DICT:
+            lo, hi = 0, max(0, len(c.dictionary) - 1)
+        elif k == TypeKind.BOOL:
+            lo, hi = 0, 1
+        elif (c.dtype.is_integer or k == TypeKind.DATE32) and c.val_range:
+            lo, hi = c.val_range
+        else:
+            return None
+        width = max(1, int(hi - lo + 1).bit_length())
+        if shift + width > 63:
+            return None
+        part = (c.data.long() - lo) << shift
+        packed = part if packed is None else (packed | part)
+        shift += width
+    if packed is None:
+        return None
+    return Column(bt.int64, packed)
+
+
 def groupby_build(key_cols: Sequence[Column]) -> Tuple[torch.Tensor, torch.Tensor]:
     K = kernels()
     n = len(key_cols[0])
-    h = hash_columns(key_cols)
-    datas, masks, offsets, auxs, dtypes = _col_args(key_cols)
+    pk = _try_pack_keys(key_cols)
+    build_cols = [pk] if pk is not None else list(key_cols)
+    h = hash_columns(build_cols)
+    datas, masks, offsets, auxs, dtypes = _col_args(build_cols)
     row_gid, uniq_rows = K.groupby_build(datas, masks, offsets, auxs, dtypes, n, h)
     return row_gid, uniq_rows
 
@@ -193,11 +224,127 @@ def groupby_local(tbl: Table, keys: Sequence[str],
     ngroups = int(uniq_rows.numel())
     out_names = list(keys)
     out_cols = [_gather_any(tbl.column(k), uniq_rows) for k in keys]
+    # fused path: batch simple aggs (count/size/sum/mean/min/max) into one
+    # kernel pass sharing the row_gid read; complex aggs go one-by-one
+    FUSABLE = {"count", "size", "sum", "mean", "min", "max"}
+    fused_batch = []
+    results = {}
+
+    def flush_fused():
+        if not fused_batch:
+            return
+        _run_fused(fused_batch, row_gid, ngroups, results)
+        fused_batch.clear()
+
     for out_name, in_name, func in aggs:
         col = tbl.column(in_name) if in_name and tbl.has_column(in_name) else None
-        out_cols.append(_agg_one(col, row_gid, ngroups, func, uniq_rows, tbl))
+        if func in FUSABLE and (col is None or col.dtype.kind != TypeKind.STRING):
+            fused_batch.append((out_name, col, func))
+            if len(fused_batch) == 4:
+                flush_fused()
+        else:
+            results[out_name] = _agg_one(col, row_gid, ngroups, func,
+                                         uniq_rows, tbl)
+    flush_fused()
+    for out_name, in_name, func in aggs:
+        out_cols.append(results[out_name])
         out_names.append(out_name)
     return Table(out_names, out_cols, ngroups)
+
+
+def _run_fused(batch, row_gid, ngroups, results):
+    """Launch one agg_update_fused for up to 4 (out_name, col, func)."""
+    K = kernels()
+    datas, masks, dtypes, ops_, init_fs, init_is, wants, posts =         [], [], [], [], [], [], [], []
+    for out_name, col, func in batch:
+        is_float = col is not None and col.dtype.is_float
+        nullable = col is not None and ((col.mask is not None) or is_float)
+        if func == "size" or col is None:
+            ops_.append(_AGG_OP["size"])
+            init_fs.append(0.0)
+            init_is.append(0)
+            wants.append(1)
+            posts.append(("cnt_i64", None))
+        elif func == "count":
+            ops_.append(_AGG_OP["count"])
+            init_fs.append(0.0)
+            init_is.append(0)
+            wants.append(1)
+            posts.append(("cnt_i64", None))
+        elif func in ("sum", "mean"):
+            if is_float or func == "mean":
+                ops_.append(_AGG_OP["sum_f64"])
+                init_fs.append(0.0)
+                init_is.append(0)
+                wants.append(1)
+                posts.append(("mean" if func == "mean" else "acc_f64", None))
+            else:
+                ops_.append(_AGG_OP["sum_i64"])
+                init_fs.append(0.0)
+                init_is.append(0)
+                wants.append(0)
+                posts.append(("acc_i64", None))
+        else:  # min / max
+            if is_float:
+                ops_.append(_AGG_OP[f"{func}_f64"])
+                init_fs.append(_F64_MAX if func == "min" else -_F64_MAX)
+                init_is.append(0)
+                wants.append(1)
+                posts.append(("minmax_f64", None))
+            else:
+                ops_.append(_AGG_OP[f"{func}_i64"])
+                init_fs.append(0.0)
+                init_is.append(_I64_MAX if func == "min" else _I64_MIN)
+                wants.append(1)
+                posts.append(("minmax_i64", col))
+        if col is None:
+            data = torch.zeros(int(row_gid.numel()), dtype=torch.int8,
+                               device=row_gid.device)
+            masks.append(None)
+            dtypes.append(int(TypeKind.INT8))
+        else:
+            data = col.data
+            if data.dtype == torch.bool:
+                data = data.view(torch.uint8)
+            masks.append(None if col.mask is None else col.mask.view(torch.uint8))
+            dtypes.append(int(col.dtype.kind))
+        datas.append(data)
+    flat = K.agg_update_fused(datas, masks, dtypes, ops_, init_fs, init_is,
+                              wants, row_gid, ngroups)
+    for k, (out_name, col, func) in enumerate(batch):
+        acc, cnt = flat[2 * k], flat[2 * k + 1]
+        kind, extra = posts[k]
+        nullable = col is not None and ((col.mask is not None)
+                                        or col.dtype.is_float)
+        if kind == "cnt_i64":
+            results[out_name] = Column(bt.int64, cnt)
+        elif kind == "acc_f64":
+            results[out_name] = Column(bt.float64, acc)
+        elif kind == "acc_i64":
+            results[out_name] = Column(bt.int64, acc)
+        elif kind == "mean" and func == "mean":
+            results[out_name] = Column(bt.float64, acc / cnt.to(torch.float64))
+        elif kind == "mean":
+            results[out_name] = Column(bt.float64, acc)
+        elif kind == "minmax_f64":
+            empty = cnt == 0
+            out = torch.where(empty, torch.full_like(acc, float("nan")), acc)
+            results[out_name] = Column(bt.float64, out)
+        else:  # minmax_i64
+            if nullable:
+                results[out_name] = Column(bt.float64, torch.where(
+                    cnt == 0,
+                    torch.full((ngroups,), float("nan"), dtype=torch.float64,
+                               device=acc.device),
+                    acc.to(torch.float64)))
+            elif col.dtype.kind == TypeKind.TIMESTAMP_NS:
+                results[out_name] = Column(bt.timestamp_ns, acc)
+            elif col.dtype.kind == TypeKind.DATE32:
+                results[out_name] = Column(bt.date32, acc.to(torch.int32))
+            elif col.dtype.kind == TypeKind.BOOL:
+                results[out_name] = Column(bt.boolean, acc.to(torch.bool))
+            else:
+                results[out_name] = Column(col.dtype, acc.to(col.data.dtype))
 
 
 def _gather_any(col: Column, idx: torch.Tensor) -> Column:

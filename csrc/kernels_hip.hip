@@ -594,6 +594,106 @@ __global__ void agg_update_lds_kernel(const ColumnDesc col,
   }
 }
 
+// fused multi-aggregate: one pass over row_gid updating up to 4 aggregates
+// (taxi Q1: count + sum + count in ONE read of the gid array)
+#define MAX_FUSED_AGGS 4
+
+struct FusedAgg {
+  ColumnDesc col;
+  int op;
+  void* acc;
+  int64_t* cnt;
+};
+
+__global__ void agg_update_fused_kernel(FusedAgg a0, FusedAgg a1, FusedAgg a2,
+                                        FusedAgg a3, int n_aggs,
+                                        const int32_t* __restrict__ row_gid,
+                                        int64_t n) {
+  FusedAgg aggs[MAX_FUSED_AGGS] = {a0, a1, a2, a3};
+  GRID_STRIDE_LOOP(i, n) {
+    int32_t g = row_gid[i];
+    for (int k = 0; k < n_aggs; ++k) {
+      const ColumnDesc& col = aggs[k].col;
+      int op = aggs[k].op;
+      void* acc = aggs[k].acc;
+      int64_t* cnt = aggs[k].cnt;
+      if (op == AGG_SIZE) {
+        atomicAdd((unsigned long long*)&cnt[g], 1ull);
+        continue;
+      }
+      bool valid = is_valid_at(col, i);
+      double dv = 0.0;
+      int64_t iv = 0;
+      if (valid) {
+        switch (col.dtype) {
+          case BT_INT8: iv = ((const int8_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_UINT8: case BT_BOOL: iv = ((const uint8_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_INT16: case BT_UINT16: iv = ((const int16_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_INT32: case BT_UINT32: case BT_DATE32: case BT_DICT:
+            iv = ((const int32_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_INT64: case BT_UINT64: case BT_TIMESTAMP_NS:
+            iv = ((const int64_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_FLOAT32: { float f = ((const float*)col.data)[i]; valid = !(f != f); dv = (double)f; iv = (int64_t)f; break; }
+          case BT_FLOAT64: { double f = ((const double*)col.data)[i]; valid = !(f != f); dv = f; iv = (int64_t)f; break; }
+        }
+      }
+      if (!valid) continue;
+      switch (op) {
+        case AGG_SUM_F64: atomicAdd((double*)acc + g, dv); break;
+        case AGG_SUM_I64: atomicAdd((unsigned long long*)acc + g, (unsigned long long)iv); break;
+        case AGG_COUNT: break;
+        case AGG_MIN_F64: atomic_min_f64((double*)acc + g, dv); break;
+        case AGG_MAX_F64: atomic_max_f64((double*)acc + g, dv); break;
+        case AGG_MIN_I64: atomic_min_i64((int64_t*)acc + g, iv); break;
+        case AGG_MAX_I64: atomic_max_i64((int64_t*)acc + g, iv); break;
+      }
+      if (cnt != nullptr) atomicAdd((unsigned long long*)&cnt[g], 1ull);
+    }
+  }
+}
+
+// returns per agg: [acc, cnt?] pairs flattened
+std::vector<torch::Tensor> agg_update_fused(
+    std::vector<torch::Tensor> datas,
+    std::vector<c10::optional<torch::Tensor>> masks,
+    std::vector<int64_t> dtypes, std::vector<int64_t> ops,
+    std::vector<double> init_fs, std::vector<int64_t> init_is,
+    std::vector<int64_t> want_counts,
+    torch::Tensor row_gid, int64_t ngroups) {
+  auto dev = row_gid.device();
+  int64_t n = row_gid.numel();
+  int n_aggs = (int)datas.size();
+  TORCH_CHECK(n_aggs >= 1 && n_aggs <= MAX_FUSED_AGGS);
+  std::vector<torch::Tensor> out;
+  FusedAgg fas[MAX_FUSED_AGGS] = {};
+  for (int k = 0; k < n_aggs; ++k) {
+    int64_t op = ops[k];
+    bool f64_acc = (op == AGG_SUM_F64 || op == AGG_MIN_F64 || op == AGG_MAX_F64);
+    torch::Tensor acc = f64_acc
+        ? torch::full({ngroups}, init_fs[k], torch::dtype(torch::kFloat64).device(dev))
+        : torch::full({ngroups}, init_is[k], torch::dtype(torch::kInt64).device(dev));
+    torch::Tensor cnt;
+    if (want_counts[k] || op == AGG_SIZE || op == AGG_COUNT) {
+      cnt = torch::zeros({ngroups}, torch::dtype(torch::kInt64).device(dev));
+    }
+    fas[k].col = make_desc(datas[k], masks[k], c10::nullopt, c10::nullopt,
+                           dtypes[k], n);
+    fas[k].op = (int)op;
+    fas[k].acc = acc.data_ptr();
+    fas[k].cnt = cnt.defined() ? (int64_t*)cnt.data_ptr() : nullptr;
+    out.push_back(acc);
+    out.push_back(cnt.defined() ? cnt : torch::Tensor());
+  }
+  int block = 256;
+  hipLaunchKernelGGL(agg_update_fused_kernel, dim3(grid_for(n, block)),
+                     dim3(block), 0, cur_stream(), fas[0], fas[1], fas[2],
+                     fas[3], n_aggs, (const int32_t*)row_gid.data_ptr(), n);
+  CHECK_HIP(hipGetLastError());
+  std::vector<torch::Tensor> cleaned;
+  for (auto& t : out) cleaned.push_back(t.defined() ? t : torch::empty({0}));
+  return cleaned;
+}
+
 std::vector<torch::Tensor> agg_update(
     torch::Tensor data, c10::optional<torch::Tensor> mask,
     c10::optional<torch::Tensor> offsets, int64_t dtype,
@@ -865,6 +965,7 @@ torch::Tensor gemm_f32(torch::Tensor A, torch::Tensor B);  // gemm.hip
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_f32", &gemm_f32, "f32 MFMA GEMM (v_mfma_f32_16x16x4_f32)");
   m.def("rle_expand", &rle_expand, "parquet RLE/bit-packed hybrid expand");
+  m.def("agg_update_fused", &agg_update_fused, "fused multi-aggregate update");
   m.def("hash_columns", &hash_columns, "multi-column row hash");
   m.def("dt_field", &dt_field, "datetime field extraction");
   m.def("gather_string", &gather_string, "string column gather");
