@@ -197,9 +197,10 @@ def groupby_build(key_cols: Sequence[Column]) -> Tuple[torch.Tensor, torch.Tenso
     K = kernels()
     n = len(key_cols[0])
     pk = _try_pack_keys(key_cols)
-    build_cols = [pk] if pk is not None else list(key_cols)
-    h = hash_columns(build_cols)
-    datas, masks, offsets, auxs, dtypes = _col_args(build_cols)
+    if pk is not None:
+        return K.groupby_build_packed(pk.data)
+    h = hash_columns(key_cols)
+    datas, masks, offsets, auxs, dtypes = _col_args(key_cols)
     row_gid, uniq_rows = K.groupby_build(datas, masks, offsets, auxs, dtypes, n, h)
     return row_gid, uniq_rows
 
@@ -548,15 +549,20 @@ def join_local(left: Table, right: Table, left_on: Sequence[str],
                                torch.full((int(unmatched.numel()),), -1,
                                           dtype=torch.int64, device=out_probe.device)])
         out_build = torch.cat([out_build, unmatched])
-    # materialize
-    p_valid = out_probe >= 0
-    b_valid = out_build >= 0
-    pt = take_table(probe, out_probe.clamp(min=0))
-    btb = take_table(build, out_build.clamp(min=0))
-    if not bool(p_valid.all().item()):
-        pt = _null_out(pt, p_valid)
-    if not bool(b_valid.all().item()):
-        btb = _null_out(btb, b_valid)
+    # materialize (inner joins produce only valid pairs by construction;
+    # skip the 1B-row .all() reductions/syncs there)
+    if how == "inner":
+        pt = take_table(probe, out_probe)
+        btb = take_table(build, out_build)
+    else:
+        p_valid = out_probe >= 0
+        b_valid = out_build >= 0
+        pt = take_table(probe, out_probe.clamp(min=0))
+        btb = take_table(build, out_build.clamp(min=0))
+        if not bool(p_valid.all().item()):
+            pt = _null_out(pt, p_valid)
+        if not bool(b_valid.all().item()):
+            btb = _null_out(btb, b_valid)
     if swap:
         # right join: probe side was the right table
         return _merge_joined(btb, pt, list(right_on), list(left_on), suffixes, how)

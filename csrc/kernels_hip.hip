@@ -335,6 +335,161 @@ __global__ void gb_count_kernel(const uint32_t* __restrict__ slots,
   if (threadIdx.x == 0) block_counts[blockIdx.x] = red[0];
 }
 
+__global__ void gb_rowgid_kernel(const uint32_t* __restrict__ row_slot,
+                                 const int32_t* __restrict__ slot_gid,
+                                 int32_t* __restrict__ row_gid, int64_t n);
+
+// Packed-key fast path: keys pre-packed into one u64 (< 2^63), slots hold
+// the KEY itself (CAS, sentinel ~0) so probing never dereferences candidate
+// rows or a hash array; the slot hash is computed in-kernel.
+#define PACKED_EMPTY 0xFFFFFFFFFFFFFFFFULL
+
+__global__ void gb_insert_packed_kernel(const int64_t* __restrict__ keys,
+                                        unsigned long long* __restrict__ slot_keys,
+                                        uint32_t* __restrict__ slot_rows,
+                                        uint64_t cap_mask,
+                                        uint32_t* __restrict__ row_slot,
+                                        int64_t n) {
+  GRID_STRIDE_LOOP(i, n) {
+    unsigned long long key = (unsigned long long)keys[i];
+    uint64_t s = mix64(key) & cap_mask;
+    while (true) {
+      unsigned long long old = slot_keys[s];
+      if (old == key) {
+        row_slot[i] = (uint32_t)s;
+        break;
+      }
+      if (old == PACKED_EMPTY) {
+        old = atomicCAS(&slot_keys[s], PACKED_EMPTY, key);
+        if (old == PACKED_EMPTY) {
+          slot_rows[s] = (uint32_t)(i + 1);
+          row_slot[i] = (uint32_t)s;
+          break;
+        }
+        if (old == key) {
+          row_slot[i] = (uint32_t)s;
+          break;
+        }
+      }
+      s = (s + 1) & cap_mask;
+    }
+  }
+}
+
+__global__ void gb_assign_gid_packed_kernel(
+    const unsigned long long* __restrict__ slot_keys,
+    const uint32_t* __restrict__ slot_rows, int64_t chunk, int64_t cap,
+    const int64_t* __restrict__ block_base, int32_t* __restrict__ slot_gid,
+    int64_t* __restrict__ uniq_rows) {
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t stop = min(start + chunk, cap);
+  __shared__ int lds_cnt;
+  if (threadIdx.x == 0) lds_cnt = 0;
+  __syncthreads();
+  int64_t base = block_base[blockIdx.x];
+  for (int64_t s = start + threadIdx.x; s < stop; s += blockDim.x) {
+    if (slot_keys[s] != PACKED_EMPTY) {
+      int32_t gid = (int32_t)(base + atomicAdd(&lds_cnt, 1));
+      slot_gid[s] = gid;
+      uniq_rows[gid] = (int64_t)slot_rows[s] - 1;
+    }
+  }
+}
+
+__global__ void gb_count_packed_kernel(
+    const unsigned long long* __restrict__ slot_keys, int64_t chunk,
+    int64_t cap, int64_t* __restrict__ block_counts) {
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t stop = min(start + chunk, cap);
+  int64_t local = 0;
+  for (int64_t s = start + threadIdx.x; s < stop; s += blockDim.x) {
+    local += (slot_keys[s] != PACKED_EMPTY);
+  }
+  __shared__ int64_t red[256];
+  red[threadIdx.x] = local;
+  __syncthreads();
+  for (int w = blockDim.x / 2; w > 0; w >>= 1) {
+    if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) block_counts[blockIdx.x] = red[0];
+}
+
+std::vector<torch::Tensor> groupby_build_packed(torch::Tensor keys) {
+  auto dev = keys.device();
+  int64_t n = keys.numel();
+  int block = 256;
+  // 2^25 slots x 8 B = 256 MB: Infinity-Cache resident; grow on pressure
+  int64_t cap = 16;
+  while (cap < 2 * n) cap <<= 1;
+  if (cap > (1 << 25)) cap = 1 << 25;
+  auto row_slot = torch::empty({n}, torch::dtype(torch::kInt32).device(dev));
+  torch::Tensor slot_keys, slot_rows;
+  const int64_t chunk_rows = 1 << 27;
+  while (true) {
+    slot_keys = torch::full({cap}, -1, torch::dtype(torch::kInt64).device(dev));
+    slot_rows = torch::empty({cap}, torch::dtype(torch::kInt32).device(dev));
+    bool grown = false;
+    for (int64_t start = 0; start < n; start += chunk_rows) {
+      int64_t cnt = std::min(chunk_rows, n - start);
+      hipLaunchKernelGGL(gb_insert_packed_kernel, dim3(grid_for(cnt, block)),
+                         dim3(block), 0, cur_stream(),
+                         (const int64_t*)keys.data_ptr() + start,
+                         (unsigned long long*)slot_keys.data_ptr(),
+                         (uint32_t*)slot_rows.data_ptr(), (uint64_t)(cap - 1),
+                         (uint32_t*)row_slot.data_ptr() + start, cnt);
+      CHECK_HIP(hipGetLastError());
+      if (start + cnt < n && cap < 2 * n) {
+        int64_t occupied = (slot_keys != -1).sum().item<int64_t>();
+        if (occupied * 2 > cap) {
+          int64_t want = cap * 8;
+          while (want < 2 * std::min(n, occupied * 8)) want <<= 1;
+          if (want > 2 * n) {
+            int64_t c2 = 16;
+            while (c2 < 2 * n) c2 <<= 1;
+            want = c2;
+          }
+          cap = want;
+          grown = true;
+          break;
+        }
+      }
+    }
+    if (!grown) break;
+  }
+  int nblocks = 2048;
+  int64_t chunk = (cap + nblocks - 1) / nblocks;
+  auto block_counts = torch::zeros({nblocks},
+                                   torch::dtype(torch::kInt64).device(dev));
+  hipLaunchKernelGGL(gb_count_packed_kernel, dim3(nblocks), dim3(block), 0,
+                     cur_stream(),
+                     (const unsigned long long*)slot_keys.data_ptr(), chunk,
+                     cap, (int64_t*)block_counts.data_ptr());
+  CHECK_HIP(hipGetLastError());
+  auto block_base = torch::zeros({nblocks}, torch::dtype(torch::kInt64).device(dev));
+  auto bb_tail = block_base.slice(0, 1, nblocks);
+  at::cumsum_out(bb_tail, block_counts.slice(0, 0, nblocks - 1), 0);
+  int64_t ngroups = block_counts.sum().item<int64_t>();
+  auto slot_gid = torch::empty({cap}, torch::dtype(torch::kInt32).device(dev));
+  auto uniq_rows = torch::empty({std::max<int64_t>(ngroups, 1)},
+                                torch::dtype(torch::kInt64).device(dev));
+  hipLaunchKernelGGL(gb_assign_gid_packed_kernel, dim3(nblocks), dim3(block),
+                     0, cur_stream(),
+                     (const unsigned long long*)slot_keys.data_ptr(),
+                     (const uint32_t*)slot_rows.data_ptr(), chunk, cap,
+                     (const int64_t*)block_base.data_ptr(),
+                     (int32_t*)slot_gid.data_ptr(),
+                     (int64_t*)uniq_rows.data_ptr());
+  CHECK_HIP(hipGetLastError());
+  auto row_gid = torch::empty({n}, torch::dtype(torch::kInt32).device(dev));
+  hipLaunchKernelGGL(gb_rowgid_kernel, dim3(grid_for(n, block)), dim3(block),
+                     0, cur_stream(), (const uint32_t*)row_slot.data_ptr(),
+                     (const int32_t*)slot_gid.data_ptr(),
+                     (int32_t*)row_gid.data_ptr(), n);
+  CHECK_HIP(hipGetLastError());
+  return {row_gid, uniq_rows.slice(0, 0, std::max<int64_t>(ngroups, 0))};
+}
+
 __global__ void gb_assign_gid_kernel(const uint32_t* __restrict__ slots,
                                      int64_t chunk, int64_t cap,
                                      const int64_t* __restrict__ block_base,
@@ -966,6 +1121,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_f32", &gemm_f32, "f32 MFMA GEMM (v_mfma_f32_16x16x4_f32)");
   m.def("rle_expand", &rle_expand, "parquet RLE/bit-packed hybrid expand");
   m.def("agg_update_fused", &agg_update_fused, "fused multi-aggregate update");
+  m.def("groupby_build_packed", &groupby_build_packed,
+        "packed-u64-key hash groupby build");
   m.def("hash_columns", &hash_columns, "multi-column row hash");
   m.def("dt_field", &dt_field, "datetime field extraction");
   m.def("gather_string", &gather_string, "string column gather");
