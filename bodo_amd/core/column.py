@@ -74,7 +74,8 @@ class Column:
     # ------------------------------------------------------------------
     def to_device(self, device) -> "Column":
         device = torch.device(device)
-        if self.device == device:
+        cur = self.device
+        if cur == device or (cur.type == device.type and device.index is None):
             return self
 
         # non_blocking only for H2D: async D2H into pageable host memory can
@@ -84,10 +85,12 @@ class Column:
         def mv(t):
             return None if t is None else t.to(device, non_blocking=nb)
 
-        return Column(
+        out = Column(
             self.dtype, mv(self.data), mv(self.mask), mv(self.offsets),
             self.dictionary, self._length,
         )
+        out.val_range = self.val_range
+        return out
 
     # ------------------------------------------------------------------
     # arrow interop
