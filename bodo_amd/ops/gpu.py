@@ -260,12 +260,16 @@ def _run_fused(batch, row_gid, ngroups, results):
     for out_name, col, func in batch:
         is_float = col is not None and col.dtype.is_float
         nullable = col is not None and ((col.mask is not None) or is_float)
-        if func == "size" or col is None:
+        col_for_data = col
+        if func == "size" or col is None or (
+                func == "count" and col.mask is None and not is_float):
+            # count over a non-nullable column == size: no column read
             ops_.append(_AGG_OP["size"])
             init_fs.append(0.0)
             init_is.append(0)
             wants.append(1)
             posts.append(("cnt_i64", None))
+            col_for_data = None
         elif func == "count":
             ops_.append(_AGG_OP["count"])
             init_fs.append(0.0)
@@ -298,17 +302,17 @@ def _run_fused(batch, row_gid, ngroups, results):
                 init_is.append(_I64_MAX if func == "min" else _I64_MIN)
                 wants.append(1)
                 posts.append(("minmax_i64", col))
-        if col is None:
-            data = torch.zeros(int(row_gid.numel()), dtype=torch.int8,
-                               device=row_gid.device)
+        if col_for_data is None:
+            data = torch.zeros(1, dtype=torch.int8, device=row_gid.device)
             masks.append(None)
             dtypes.append(int(TypeKind.INT8))
         else:
-            data = col.data
+            data = col_for_data.data
             if data.dtype == torch.bool:
                 data = data.view(torch.uint8)
-            masks.append(None if col.mask is None else col.mask.view(torch.uint8))
-            dtypes.append(int(col.dtype.kind))
+            masks.append(None if col_for_data.mask is None
+                         else col_for_data.mask.view(torch.uint8))
+            dtypes.append(int(col_for_data.dtype.kind))
         datas.append(data)
     flat = K.agg_update_fused(datas, masks, dtypes, ops_, init_fs, init_is,
                               wants, row_gid, ngroups)
@@ -552,7 +556,15 @@ def join_local(left: Table, right: Table, left_on: Sequence[str],
     # materialize (inner joins produce only valid pairs by construction;
     # skip the 1B-row .all() reductions/syncs there)
     if how == "inner":
-        pt = take_table(probe, out_probe)
+        elide = False
+        if int(out_probe.numel()) == n_probe and n_probe > 0:
+            # identity iff strictly increasing (out_probe is sorted runs of
+            # repeated probe ids; strict monotonic + length n => arange(n))
+            elide = bool((out_probe[1:] > out_probe[:-1]).all().item())
+        if elide:
+            pt = probe  # every probe row matched exactly once: skip gather
+        else:
+            pt = take_table(probe, out_probe)
         btb = take_table(build, out_build)
     else:
         p_valid = out_probe >= 0
