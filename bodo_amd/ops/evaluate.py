@@ -143,8 +143,15 @@ class _Evaluator:
             res = ser.isin(list(e.values)).to_numpy()
             return Column(bt.boolean, torch.from_numpy(res).to(self.device))
         vals = [normalize_const(v, a.dtype) for v in e.values]
-        test = torch.tensor(vals, dtype=a.data.dtype, device=self.device)
-        out = torch.isin(a.data, test)
+        if len(vals) <= 16:
+            # explicit OR of equalities: avoids torch.isin's 1B x k
+            # broadcast + dim-reduce (measured 4x7ms per query at 1B rows)
+            out = a.data == vals[0]
+            for v in vals[1:]:
+                out |= a.data == v
+        else:
+            test = torch.tensor(vals, dtype=a.data.dtype, device=self.device)
+            out = torch.isin(a.data, test)
         if a.mask is not None:
             out = out & a.mask
         return Column(bt.boolean, out)

@@ -281,8 +281,11 @@ def _run_fused(batch, row_gid, ngroups, results):
                 ops_.append(_AGG_OP["sum_f64"])
                 init_fs.append(0.0)
                 init_is.append(0)
-                wants.append(1)
-                posts.append(("mean" if func == "mean" else "acc_f64", None))
+                shared = (func == "mean" and share_size
+                          and clean_float.get(id(col), False))
+                wants.append(0 if shared else 1)
+                posts.append(("mean_shared" if shared else
+                              ("mean" if func == "mean" else "acc_f64"), None))
             else:
                 ops_.append(_AGG_OP["sum_i64"])
                 init_fs.append(0.0)
@@ -316,6 +319,11 @@ def _run_fused(batch, row_gid, ngroups, results):
         datas.append(data)
     flat = K.agg_update_fused(datas, masks, dtypes, ops_, init_fs, init_is,
                               wants, row_gid, ngroups)
+    shared_cnt = None
+    for k in range(len(batch)):
+        if posts[k][0] == "cnt_i64":
+            shared_cnt = flat[2 * k + 1]
+            break
     for k, (out_name, col, func) in enumerate(batch):
         acc, cnt = flat[2 * k], flat[2 * k + 1]
         kind, extra = posts[k]
@@ -327,6 +335,9 @@ def _run_fused(batch, row_gid, ngroups, results):
             results[out_name] = Column(bt.float64, acc)
         elif kind == "acc_i64":
             results[out_name] = Column(bt.int64, acc)
+        elif kind == "mean_shared":
+            results[out_name] = Column(
+                bt.float64, acc / shared_cnt.to(torch.float64))
         elif kind == "mean" and func == "mean":
             results[out_name] = Column(bt.float64, acc / cnt.to(torch.float64))
         elif kind == "mean":
