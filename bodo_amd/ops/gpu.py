@@ -256,7 +256,19 @@ def groupby_local(tbl: Table, keys: Sequence[str],
 def _run_fused(batch, row_gid, ngroups, results):
     """Launch one agg_update_fused for up to 4 (out_name, col, func)."""
     K = kernels()
-    datas, masks, dtypes, ops_, init_fs, init_is, wants, posts =         [], [], [], [], [], [], [], []
+    # means over provably-NaN-free unmasked float columns can reuse a shared
+    # group-size count instead of carrying their own valid-count atomics
+    clean_float = {}
+    for _, col, func in batch:
+        if func == "mean" and col is not None and col.mask is None \
+                and col.dtype.is_float and id(col) not in clean_float:
+            clean_float[id(col)] = not bool(torch.isnan(col.data).any().item())
+    share_size = any(clean_float.values()) and any(
+        f == "size" or (f == "count" and c is not None and c.mask is None
+                        and not c.dtype.is_float)
+        for _, c, f in batch)
+    datas, masks, dtypes, ops_, init_fs, init_is, wants, posts = \
+        [], [], [], [], [], [], [], []
     for out_name, col, func in batch:
         is_float = col is not None and col.dtype.is_float
         nullable = col is not None and ((col.mask is not None) or is_float)
