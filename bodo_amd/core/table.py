@@ -70,8 +70,13 @@ class Table:
     # ------------------------------------------------------------------
     @staticmethod
     def from_arrow(tbl: pa.Table, device="cpu") -> "Table":
-        cols = [Column.from_arrow(tbl.column(i).combine_chunks(), device)
-                for i in range(tbl.num_columns)]
+        cols = []
+        for i in range(tbl.num_columns):
+            c = tbl.column(i)
+            # avoid 32-bit offset overflow combining >2GB string chunks
+            if pa.types.is_string(c.type):
+                c = c.cast(pa.large_string())
+            cols.append(Column.from_arrow(c.combine_chunks(), device))
         return Table(tbl.column_names, cols, tbl.num_rows)
 
     def to_arrow(self) -> pa.Table:
@@ -134,6 +139,8 @@ def dict_encode_strings(tbl: pa.Table, threshold: float = 0.7,
         col = tbl.column(i)
         w = False
         if pa.types.is_string(f.type) or pa.types.is_large_string(f.type):
+            if pa.types.is_string(f.type):
+                col = col.cast(pa.large_string())
             head = col.slice(0, min(sample, len(col)))
             try:
                 nuniq = len(pc.unique(head.combine_chunks()))
