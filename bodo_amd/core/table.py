@@ -164,6 +164,8 @@ def dict_encode_strings(tbl: pa.Table, threshold: float = 0.7,
         col = tbl.column(i)
         if want[i] and (pa.types.is_string(f.type)
                         or pa.types.is_large_string(f.type)):
+            if pa.types.is_string(f.type):
+                col = col.cast(pa.large_string())
             col = pc.dictionary_encode(col.combine_chunks())
         new_cols.append(col)
     return pa.table(dict(zip(tbl.column_names, new_cols)))
