@@ -42,7 +42,10 @@ def _unif(idx, salt, lo, hi):
 
 
 def _pick(idx, salt, values):
-    return np.asarray(values, dtype=object)[_uni(idx, salt, 0, len(values) - 1)]
+    """Categorical (codes + small category list): no per-row python objects,
+    scales to SF100 row counts."""
+    codes = _uni(idx, salt, 0, len(values) - 1).astype(np.int32)
+    return pd.Categorical.from_codes(codes, categories=list(values))
 
 
 NATIONS = [
@@ -94,7 +97,10 @@ def _comment_pool(n_pool, salt, trigger=None, trig_rate=0.012):
         hit = rng.choice(n_pool, k, replace=False)
         mids = rng.choice(WORDS, k)
         pool[hit] = [f"{trigger[0]} {m} {trigger[1]}" for m in mids]
-    return pool
+    # de-duplicate (random sequences can collide; Categorical categories
+    # must be unique) while keeping order
+    uniq = list(dict.fromkeys(pool.tolist()))
+    return np.array(uniq, dtype=object)
 
 
 def _bounds(n, rank, world):
@@ -137,6 +143,8 @@ def gen_table(name: str, sf: float, rank: int = 0, world: int = 1) -> pd.DataFra
         })
     if name == "supplier":
         pool = _comment_pool(5000, 42, ("Customer", "Complaints"))
+        com = pd.Categorical.from_codes(
+            _uni(i, 3, 0, len(pool) - 1).astype(np.int32), categories=list(pool))
         return pd.DataFrame({
             "S_SUPPKEY": key,
             "S_NAME": np.char.add("Supplier#",
@@ -145,7 +153,7 @@ def gen_table(name: str, sf: float, rank: int = 0, world: int = 1) -> pd.DataFra
             "S_NATIONKEY": _uni(i, 1, 0, 24),
             "S_PHONE": _phone(i, 1),
             "S_ACCTBAL": _unif(i, 2, -999.99, 9999.99).round(2),
-            "S_COMMENT": pool[_uni(i, 3, 0, len(pool) - 1)],
+            "S_COMMENT": com,
         })
     if name == "customer":
         nk = _uni(i, 10, 0, 24)
@@ -161,24 +169,36 @@ def gen_table(name: str, sf: float, rank: int = 0, world: int = 1) -> pd.DataFra
             "C_COMMENT": np.char.add("com ", key.astype("U12")).astype(object),
         })
     if name == "part":
-        t1, t2, t3 = _pick(i, 20, TYPE1), _pick(i, 21, TYPE2), _pick(i, 22, TYPE3)
-        ptype = np.char.add(np.char.add(np.char.add(
-            t1.astype("U8"), " "), np.char.add(t2.astype("U9"), " ")),
-            t3.astype("U6")).astype(object)
-        c1, c2 = _pick(i, 23, CONT1), _pick(i, 24, CONT2)
-        cont = np.char.add(np.char.add(c1.astype("U5"), " "),
-                           c2.astype("U4")).astype(object)
-        brand = np.char.add("Brand#", np.char.add(
-            _uni(i, 25, 1, 5).astype("U1"), _uni(i, 26, 1, 5).astype("U1"))
-        ).astype(object)
-        nm1, nm2 = _pick(i, 27, COLORS), _pick(i, 28, COLORS)
-        pname = np.char.add(np.char.add(nm1.astype("U12"), " "),
-                            nm2.astype("U12")).astype(object)
+        # composite categoricals: build the small cross-product category
+        # lists once and index them with combined codes
+        ptype_cats = [f"{a} {b} {c}" for a in TYPE1 for b in TYPE2
+                      for c in TYPE3]
+        pt_code = (_uni(i, 20, 0, len(TYPE1) - 1) * len(TYPE2)
+                   + _uni(i, 21, 0, len(TYPE2) - 1)) * len(TYPE3)             + _uni(i, 22, 0, len(TYPE3) - 1)
+        ptype = pd.Categorical.from_codes(pt_code.astype(np.int32),
+                                          categories=ptype_cats)
+        cont_cats = [f"{a} {b}" for a in CONT1 for b in CONT2]
+        ct_code = _uni(i, 23, 0, len(CONT1) - 1) * len(CONT2)             + _uni(i, 24, 0, len(CONT2) - 1)
+        cont = pd.Categorical.from_codes(ct_code.astype(np.int32),
+                                         categories=cont_cats)
+        brand_cats = [f"Brand#{a}{b}" for a in range(1, 6)
+                      for b in range(1, 6)]
+        br_code = (_uni(i, 25, 1, 5) - 1) * 5 + (_uni(i, 26, 1, 5) - 1)
+        brand = pd.Categorical.from_codes(br_code.astype(np.int32),
+                                          categories=brand_cats)
+        pname_cats = [f"{a} {b}" for a in COLORS for b in COLORS]
+        pn_code = _uni(i, 27, 0, len(COLORS) - 1) * len(COLORS)             + _uni(i, 28, 0, len(COLORS) - 1)
+        pname = pd.Categorical.from_codes(pn_code.astype(np.int32),
+                                          categories=pname_cats)
+        nm1 = pd.Categorical.from_codes(
+            _uni(i, 27, 0, len(COLORS) - 1).astype(np.int32),
+            categories=COLORS)
         return pd.DataFrame({
             "P_PARTKEY": key,
             "P_NAME": pname,
-            "P_MFGR": np.char.add("Manufacturer#",
-                                  _uni(i, 29, 1, 5).astype("U1")).astype(object),
+            "P_MFGR": pd.Categorical.from_codes(
+                (_uni(i, 29, 1, 5) - 1).astype(np.int32),
+                categories=[f"Manufacturer#{k}" for k in range(1, 6)]),
             "P_BRAND": brand,
             "P_TYPE": ptype,
             "P_SIZE": _uni(i, 30, 1, 50),
@@ -202,6 +222,9 @@ def gen_table(name: str, sf: float, rank: int = 0, world: int = 1) -> pd.DataFra
     if name == "orders":
         ncust = ROWS["customer"](sf)
         pool = _comment_pool(8000, 77, ("special", "requests"))
+        com = pd.Categorical.from_codes(
+            _uni(i, 56, 0, len(pool) - 1).astype(np.int32),
+            categories=list(pool))
         return pd.DataFrame({
             "O_ORDERKEY": key,
             # custkeys divisible by 3 never order (q22's not-exists branch;
@@ -211,10 +234,11 @@ def gen_table(name: str, sf: float, rank: int = 0, world: int = 1) -> pd.DataFra
             "O_TOTALPRICE": _unif(i, 52, 1000.0, 500000.0).round(2),
             "O_ORDERDATE": _date_col(i, 53),
             "O_ORDERPRIORITY": _pick(i, 54, PRIORITIES),
-            "O_CLERK": np.char.add("Clerk#", _uni(i, 55, 1, 1000)
-                                   .astype("U6")).astype(object),
+            "O_CLERK": pd.Categorical.from_codes(
+                (_uni(i, 55, 1, 1000) - 1).astype(np.int32),
+                categories=[f"Clerk#{k}" for k in range(1, 1001)]),
             "O_SHIPPRIORITY": np.zeros(len(i), dtype=np.int64),
-            "O_COMMENT": pool[_uni(i, 56, 0, len(pool) - 1)],
+            "O_COMMENT": com,
         })
     if name == "lineitem":
         nparts = ROWS["part"](sf)
@@ -264,8 +288,10 @@ def _phone(i, salt, nationkey=None):
     p1 = _uni(i, salt + 100, 100, 999)
     p2 = _uni(i, salt + 101, 100, 999)
     p3 = _uni(i, salt + 102, 1000, 9999)
-    return np.array([f"{a}-{b}-{c}-{d}" for a, b, c, d in
-                     zip(cc, p1, p2, p3)], dtype=object)
+    out = np.char.add(np.char.add(np.char.add(np.char.add(
+        np.char.add(np.char.add(cc.astype("U2"), "-"), p1.astype("U3")), "-"),
+        p2.astype("U3")), "-"), p3.astype("U4"))
+    return out.astype(object)
 
 
 def _ps_suppkey(partkey, j, nsupp):
@@ -274,6 +300,15 @@ def _ps_suppkey(partkey, j, nsupp):
 
 TABLES = ["region", "nation", "supplier", "customer", "part", "partsupp",
           "orders", "lineitem"]
+
+
+def gen_arrow(name: str, sf: float, rank: int = 0, world: int = 1):
+    """Arrow table without pandas block consolidation (2x faster, no 2D
+    stacking copies) — the path the GPU benchmarks load through."""
+    import pyarrow as pa
+
+    df = gen_table(name, sf, rank, world)
+    return pa.Table.from_pandas(df, preserve_index=False)
 
 
 def gen_all(sf: float, rank: int = 0, world: int = 1):

@@ -203,7 +203,9 @@ def test_dist_tpch(qn):
 
     got = run_dist(_q_tpch, {"q": qn}).reset_index(drop=True)
     t = gen_all(0.01)
-    exp = ALL[qn](pd, {k: v.copy() for k, v in t.items()}).reset_index(drop=True)
+    from tests.test_tpch import _decat_df
+
+    exp = ALL[qn](pd, {k: _decat_df(v) for k, v in t.items()}).reset_index(drop=True)
     for c in exp.columns:
         if exp[c].dtype == object or str(exp[c].dtype) == "category":
             exp[c] = exp[c].astype(str)

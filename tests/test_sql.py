@@ -123,8 +123,9 @@ def test_dates_extract():
 
 def test_tpch_q6_sql():
     from tpch_data import gen_all
+    from tests.test_tpch import _decat_df
 
-    t = gen_all(0.02)
+    t = {k: _decat_df(v) for k, v in gen_all(0.02).items()}
     got = run_sql(
         "SELECT SUM(l_extendedprice * l_discount) AS revenue "
         "FROM lineitem WHERE l_shipdate >= DATE '1996-01-01' "
@@ -142,8 +143,9 @@ def test_tpch_q6_sql():
 
 def test_tpch_q1_sql():
     from tpch_data import gen_all
+    from tests.test_tpch import _decat_df
 
-    t = gen_all(0.02)
+    t = {k: _decat_df(v) for k, v in gen_all(0.02).items()}
     got = run_sql("""
         SELECT l_returnflag, l_linestatus,
                SUM(l_quantity) AS sum_qty,
@@ -190,8 +192,9 @@ def test_tpch_q3_sql():
     """, {"customer": t["customer"], "orders": t["orders"],
           "lineitem": t["lineitem"]})
     from tpch_queries import q3
+    from tests.test_tpch import _decat_df
 
-    exp = q3(pd, {k: v.copy() for k, v in t.items()})
+    exp = q3(pd, {k: _decat_df(v) for k, v in t.items()})
     exp = exp.rename(columns={"L_ORDERKEY": "o_orderkey", "REVENUE": "revenue",
                               "O_ORDERDATE": "o_orderdate",
                               "O_SHIPPRIORITY": "o_shippriority"})
