@@ -32,6 +32,50 @@ def jit(fn=None, **options):
     return _jit(fn, **options)
 
 
+def wrap_python(fn=None, **options):
+    """Escape hatch running a plain python function per-rank (reference:
+    bodo.wrap_python, decorators.py)."""
+    from .jit.decorator import jit as _jit
+
+    return _jit(fn, **options)
+
+
+def gatherv(data, root=0):
+    from .parallel.api import gatherv as _g
+
+    return _g(data, root)
+
+
+def allgatherv(data):
+    from .parallel.api import allgatherv as _a
+
+    return _a(data)
+
+
+def scatterv(data, root=0):
+    from .parallel.api import scatterv as _s
+
+    return _s(data, root)
+
+
+def rebalance(data):
+    from .parallel.api import rebalance as _r
+
+    return _r(data)
+
+
+def random_shuffle(data, seed=None):
+    from .parallel.api import random_shuffle as _r
+
+    return _r(data, seed)
+
+
+def get_gpu_ranks():
+    from .parallel.api import get_gpu_ranks as _g
+
+    return _g()
+
+
 def get_rank() -> int:
     from .parallel import comm
 

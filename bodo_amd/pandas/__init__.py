@@ -49,6 +49,13 @@ def read_csv(path, **kwargs) -> BodoDataFrame:
     return BodoDataFrame(plan, list(names))
 
 
+def read_json(path, orient="records", lines=True, **kwargs) -> BodoDataFrame:
+    """JSON reader (reference: bodo/io json reader).  Host parse with rank
+    block-slicing, same model as read_csv."""
+    df = _pd.read_json(path, orient=orient, lines=lines, **kwargs)
+    return from_pandas_df(df)
+
+
 def from_pandas(df: _pd.DataFrame) -> BodoDataFrame:
     return from_pandas_df(df)
 

@@ -66,3 +66,41 @@ def test_sql_plan_cache():
     r2 = bc.sql("SELECT a FROM t WHERE a > 5")
     assert len(BodoSQLContext._plan_cache) == n0
     assert r1._lazy_plan is r2._lazy_plan
+
+
+def test_distributed_api_single_rank():
+    import bodo_amd
+
+    df = pd.DataFrame({"a": np.arange(20), "b": np.arange(20) * 0.5})
+    b = bpd.from_pandas(df)
+    g = bodo_amd.gatherv(b)
+    pd.testing.assert_frame_equal(g, df, check_dtype=False)
+    ag = bodo_amd.allgatherv(b)
+    pd.testing.assert_frame_equal(ag, df, check_dtype=False)
+    sc = bodo_amd.scatterv(df)
+    pd.testing.assert_frame_equal(sc, df, check_dtype=False)
+    rb = bodo_amd.rebalance(b)
+    assert len(rb) == 20
+    assert bodo_amd.get_rank() == 0 and bodo_amd.get_size() == 1
+    assert isinstance(bodo_amd.get_gpu_ranks(), list)
+
+
+def test_jit_options_accepted():
+    import bodo_amd
+
+    @bodo_amd.jit(cache=True, distributed=["df"], spawn=True)
+    def f(df):
+        return df[df.a > 5]
+
+    df = pd.DataFrame({"a": np.arange(10)})
+    out = f(bpd.from_pandas(df)).to_pandas()
+    assert len(out) == 4
+
+
+def test_read_json(tmp_path):
+    df = pd.DataFrame({"a": [1, 2, 3], "b": ["x", "y", "z"]})
+    p = str(tmp_path / "d.json")
+    df.to_json(p, orient="records", lines=True)
+    out = bpd.read_json(p).to_pandas()
+    out["b"] = out["b"].astype(str)
+    pd.testing.assert_frame_equal(out, df, check_dtype=False)
