@@ -27,7 +27,7 @@ if "RANK" in os.environ and "MASTER_ADDR" in os.environ:
 
 
 def jit(fn=None, **options):
-    from .jit.decorator import jit as _jit
+    from .compiler.decorator import jit as _jit
 
     return _jit(fn, **options)
 
@@ -35,7 +35,7 @@ def jit(fn=None, **options):
 def wrap_python(fn=None, **options):
     """Escape hatch running a plain python function per-rank (reference:
     bodo.wrap_python, decorators.py)."""
-    from .jit.decorator import jit as _jit
+    from .compiler.decorator import jit as _jit
 
     return _jit(fn, **options)
 

@@ -619,7 +619,7 @@ def udf_map(a: Column, func, na_action=None) -> Column:
         return Column(bt.float64, lut[a.data.long()], a.mask)
     if a.is_cuda and a.dtype.is_float:
         # numeric UDF on device: lower to a hipRTC-compiled HIP kernel
-        from ..jit.hip_udf import try_hip_udf
+        from ..compiler.hip_udf import try_hip_udf
 
         res = try_hip_udf(func, a.data)
         if res is not None:

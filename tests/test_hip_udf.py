@@ -8,7 +8,7 @@ import pandas as pd
 import pytest
 import torch
 
-from bodo_amd.jit.hip_udf import translate_udf
+from bodo_amd.compiler.hip_udf import translate_udf
 
 
 def test_translate_lambda():
@@ -45,7 +45,7 @@ def test_translate_unsupported_returns_none():
 
 @pytest.mark.gpu
 def test_hip_udf_matches_pandas():
-    from bodo_amd.jit.hip_udf import try_hip_udf
+    from bodo_amd.compiler.hip_udf import try_hip_udf
 
     rng = np.random.default_rng(0)
     vals = rng.uniform(-2, 2, 100000)
