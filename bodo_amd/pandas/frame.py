@@ -366,6 +366,20 @@ class BodoDataFrame:
         plan = pn.Distinct(self._plan, tuple(subset) if subset else None, keep)
         return BodoDataFrame(plan, self._columns)
 
+    def nlargest(self, n, columns, keep="first"):
+        if isinstance(columns, str):
+            columns = [columns]
+        plan = pn.Limit(pn.Sort(self._plan, tuple(columns),
+                                tuple([False] * len(columns))), n)
+        return BodoDataFrame(plan, self._columns)
+
+    def nsmallest(self, n, columns, keep="first"):
+        if isinstance(columns, str):
+            columns = [columns]
+        plan = pn.Limit(pn.Sort(self._plan, tuple(columns),
+                                tuple([True] * len(columns))), n)
+        return BodoDataFrame(plan, self._columns)
+
     def sample(self, n=None, frac=None, random_state=None, **kwargs):
         plan = pn.Sample(self._plan, n, frac, random_state)
         return BodoDataFrame(plan, self._columns)

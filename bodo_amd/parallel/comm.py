@@ -46,6 +46,29 @@ def init_from_env(device: Optional[str] = None) -> None:
         local = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", 0)))
         torch.cuda.set_device(local % torch.cuda.device_count())
     dist.init_process_group(backend, timeout=datetime.timedelta(seconds=600))
+    _install_failfast_hook()
+
+
+def _install_failfast_hook():
+    """Fail-fast on rank divergence (reference: bodo/__init__.py:6-75
+    replaces sys.excepthook with an MPI_Abort-with-timeout guard): an
+    uncaught exception on one rank exits the process hard so the launcher
+    tears the job down instead of the healthy ranks hanging in the next
+    collective until the 600s timeout."""
+    import sys
+    import traceback
+
+    prev = sys.excepthook
+
+    def hook(tp, val, tb):
+        try:
+            prev(tp, val, tb)
+        except Exception:
+            traceback.print_exception(tp, val, tb)
+        sys.stderr.flush()
+        os._exit(1)
+
+    sys.excepthook = hook
 
 
 def barrier():

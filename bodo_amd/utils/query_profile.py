@@ -45,13 +45,21 @@ class OpTimer:
 
     def __enter__(self):
         self.t0 = time.perf_counter()
+        self.mem0 = _device_allocated()
         return self
 
     def __exit__(self, *exc):
         if _FORCED[0] or _DIR:
             record_operator(self.name, time.perf_counter() - self.t0,
-                            self.rows_in, self.rows_out)
+                            self.rows_in, self.rows_out,
+                            hbm_delta_bytes=_device_allocated() - self.mem0)
         return False
+
+
+def _device_allocated() -> int:
+    import torch
+
+    return torch.cuda.memory_allocated() if torch.cuda.is_available() else 0
 
 
 def finish_query():

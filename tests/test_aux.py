@@ -104,3 +104,39 @@ def test_read_json(tmp_path):
     out = bpd.read_json(p).to_pandas()
     out["b"] = out["b"].astype(str)
     pd.testing.assert_frame_equal(out, df, check_dtype=False)
+
+
+def test_profile_aggregation_cli(tmp_path):
+    import json as _json
+    import subprocess
+    import sys
+
+    from bodo_amd.utils import query_profile as qp
+
+    qp.clear()
+    qp.enable(str(tmp_path))
+    b = bpd.from_pandas(pd.DataFrame({"a": np.arange(40)}))
+    b[b.a > 3].to_pandas()
+    qp.flush(str(tmp_path))
+    out = subprocess.run([sys.executable, "-m",
+                          "bodo_amd.utils.aggregate_query_profiles",
+                          str(tmp_path)], capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    data = _json.loads(out.stdout)
+    assert data["n_ranks"] == 1 and data["operators"]
+    qp.clear()
+
+
+def test_nlargest():
+    df = pd.DataFrame({"a": np.arange(100), "b": np.arange(100)[::-1]})
+    b = bpd.from_pandas(df)
+    got = b.nlargest(5, "b").to_pandas().reset_index(drop=True)
+    exp = df.nlargest(5, "b").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_allocation_tracking():
+    from bodo_amd.utils.allocation_tracking import get_allocation_stats
+
+    s = get_allocation_stats()
+    assert set(s) == {"allocated_bytes", "reserved_bytes", "peak_bytes"}
