@@ -242,7 +242,10 @@ class _HiprtcRuntime:
         return buf.raw
 
     def get_kernel(self, source: str):
-        key = hash(source)
+        return self.get_kernel_named(source, b"udf_kernel")
+
+    def get_kernel_named(self, source: str, name: bytes):
+        key = (hash(source), name)
         if key in self._modules:
             return self._modules[key]
         code = self.compile(source)
@@ -250,11 +253,23 @@ class _HiprtcRuntime:
         r = self.hip.hipModuleLoadData(ctypes.byref(module), code)
         assert r == 0, f"hipModuleLoadData failed: {r}"
         fn = ctypes.c_void_p()
-        r = self.hip.hipModuleGetFunction(ctypes.byref(fn), module,
-                                          b"udf_kernel")
+        r = self.hip.hipModuleGetFunction(ctypes.byref(fn), module, name)
         assert r == 0, f"hipModuleGetFunction failed: {r}"
         self._modules[key] = fn
         return fn
+
+    def launch_generic(self, fn, n: int, ptr_args, stream: int,
+                       block: int = 256):
+        """Launch kernel(long long n, <pointers...>)."""
+        holders = [ctypes.c_longlong(n)] + [ctypes.c_void_p(p)
+                                            for p in ptr_args]
+        arr = (ctypes.c_void_p * len(holders))(*[
+            ctypes.cast(ctypes.byref(h), ctypes.c_void_p) for h in holders])
+        grid = min((n + block - 1) // block, 2048) or 1
+        r = self.hip.hipModuleLaunchKernel(
+            fn, int(grid), 1, 1, block, 1, 1, 0,
+            ctypes.c_void_p(stream), arr, None)
+        assert r == 0, f"hipModuleLaunchKernel failed: {r}"
 
     def launch(self, fn, in_ptr: int, out_ptr: int, n: int, stream: int):
         args = (ctypes.c_void_p(in_ptr), ctypes.c_void_p(out_ptr),

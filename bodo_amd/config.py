@@ -50,6 +50,9 @@ VERBOSE = _env_int("BODO_AMD_VERBOSE", 0)
 #: HBM-sized morsels)
 STREAMING = os.environ.get("BODO_AMD_STREAMING", "auto")
 
+#: fuse projection/filter expressions into one hipRTC kernel on GPU
+FUSE_EXPR = _env_bool("BODO_AMD_FUSE_EXPR", True)
+
 #: auto-streaming threshold (bytes of scanned files per query)
 STREAM_THRESHOLD_BYTES = _env_int("BODO_AMD_STREAM_THRESHOLD",
                                   100 * 1024**3)

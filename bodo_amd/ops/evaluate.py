@@ -33,7 +33,7 @@ def eval_expr(e: Expr, tbl: Table) -> Column:
 def eval_filter(cond: Expr, tbl: Table) -> Table:
     from . import take_table
 
-    mask_col = eval_expr(cond, tbl)
+    mask_col = _eval_many(tbl, [cond])[0]
     mask = mask_col.data
     if mask_col.mask is not None:
         mask = mask & mask_col.mask
@@ -42,8 +42,25 @@ def eval_filter(cond: Expr, tbl: Table) -> Table:
 
 
 def project(tbl: Table, names, exprs) -> Table:
-    cols = [eval_expr(e, tbl) for e in exprs]
+    cols = _eval_many(tbl, list(exprs))
     return Table(list(names), cols, len(tbl))
+
+
+def _eval_many(tbl: Table, exprs):
+    """Evaluate a list of expressions, fusing the numeric subset into one
+    hipRTC kernel pass on GPU (config.FUSE_EXPR)."""
+    from .. import config
+
+    if config.FUSE_EXPR and tbl.device.type == "cuda":
+        try:
+            from ..compiler.expr_fuse import try_fuse_exprs
+
+            fused = try_fuse_exprs(tbl, exprs)
+        except Exception:
+            fused = None
+        if fused is not None:
+            return fused
+    return [eval_expr(e, tbl) for e in exprs]
 
 
 class _Evaluator:
