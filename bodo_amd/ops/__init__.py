@@ -64,26 +64,6 @@ def _gather_string_arrow(col: Column, idx: torch.Tensor) -> Column:
     return Column.from_arrow(taken, col.device)
 
 
-def _gather_string_torch(col: Column, idx: torch.Tensor) -> Column:
-    offsets = col.offsets
-    idx = idx.long()
-    starts = offsets[idx]
-    lens = offsets[idx + 1] - starts
-    total = int(lens.sum().item())
-    new_off = torch.zeros(len(idx) + 1, dtype=torch.int64, device=idx.device)
-    torch.cumsum(lens, 0, out=new_off[1:])
-    if total > 0:
-        rows = torch.repeat_interleave(
-            torch.arange(len(idx), device=idx.device), lens)
-        within = torch.arange(total, device=idx.device) - new_off[rows]
-        src_pos = starts[rows] + within
-        data = col.data[src_pos]
-    else:
-        data = torch.zeros(0, dtype=torch.uint8, device=idx.device)
-    mask = col.mask[idx] if col.mask is not None else None
-    return Column(bt.string, data, mask, offsets=new_off, length=len(idx))
-
-
 def take_table(tbl: Table, idx: torch.Tensor) -> Table:
     return Table(tbl.names, [gather(c, idx) for c in tbl.columns], int(idx.shape[0]))
 

@@ -29,8 +29,7 @@ class BodoDataFrame:
     def __init__(self, plan: pn.PlanNode, columns: Sequence[str]):
         object.__setattr__(self, "_plan", plan)
         object.__setattr__(self, "_columns", list(columns))
-        object.__setattr__(self, "_result", None)
-        object.__setattr__(self, "_remote", None)  # cached local shard Table
+        object.__setattr__(self, "_result", None)  # cached local shard Table
         object.__setattr__(self, "_remote", None)  # spawn-mode RemoteResult
 
     # ------------------------------------------------------------------
