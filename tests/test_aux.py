@@ -140,3 +140,19 @@ def test_allocation_tracking():
 
     s = get_allocation_stats()
     assert set(s) == {"allocated_bytes", "reserved_bytes", "peak_bytes"}
+
+
+def test_bodo_alias_package():
+    import bodo
+    import bodo.pandas as bpd2
+
+    df = pd.DataFrame({"a": np.arange(10)})
+    b = bpd2.from_pandas(df)
+    out = b[b.a > 4].to_pandas()
+    assert len(out) == 5
+
+    @bodo.jit
+    def f(d):
+        return d[d.a > 7]
+
+    assert len(f(b).to_pandas()) == 2
