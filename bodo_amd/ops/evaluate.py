@@ -21,7 +21,7 @@ from ..core.table import Table
 from ..core.types import DType, TypeKind
 from ..plan.expr import (
     BinOp, BoolOp, Case, Cast, ColRef, Cmp, Const, DtField, Expr, IsIn,
-    IsNull, Not, RoundExpr, SemiJoinIn, StrOp, UdfMap,
+    IsNull, Not, RoundExpr, ScalarSubquery, SemiJoinIn, StrOp, UdfMap,
 )
 
 
@@ -110,7 +110,25 @@ class _Evaluator:
         a, b = self.visit(e.left), self.visit(e.right)
         return binary_arith(e.op, a, b)
 
+    def _subquery_scalar(self, e: Expr):
+        if not isinstance(e, ScalarSubquery):
+            return None
+        from ..engine import api
+
+        df = api.collect(e.plan)
+        v = df[e.col].iloc[0] if len(df) else None
+        import pandas as pd
+
+        if v is None or pd.isna(v):
+            return float("nan"), bt.float64
+        if hasattr(v, "item"):
+            v = v.item()
+        return normalize_const(v, infer_const_dtype(v)), infer_const_dtype(v)
+
     def visit_Cmp(self, e: Cmp) -> Column:
+        ss = self._subquery_scalar(e.right)
+        if ss is not None:
+            return compare_scalar(e.op, self.visit(e.left), ss[0], ss[1])
         rs = self._scalar_const(e.right, allow_str=True)
         if rs is not None:
             return compare_scalar(e.op, self.visit(e.left), rs[0], rs[1])

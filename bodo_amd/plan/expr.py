@@ -238,6 +238,18 @@ class SemiJoinIn(Expr):
 
 
 @dataclass(frozen=True)
+class ScalarSubquery(Expr):
+    """Uncorrelated scalar subquery: executed lazily at evaluation time
+    (SPMD-consistent), compared via the scalar fast path."""
+
+    plan: object = field(compare=False, default=None)
+    col: str = ""
+
+    def children(self):
+        return ()
+
+
+@dataclass(frozen=True)
 class RoundExpr(Expr):
     operand: Expr
     decimals: int = 0

@@ -139,9 +139,30 @@ class ExtractE:
 
 
 @dataclass
+class SubqueryE:
+    """Scalar subquery (SELECT agg ...)."""
+
+    query: "Query"
+
+
+@dataclass
+class InSubquery:
+    operand: Any
+    query: "Query"
+    negated: bool = False
+
+
+@dataclass
+class ExistsE:
+    query: "Query"
+    negated: bool = False
+
+
+@dataclass
 class TableRef:
     name: str
     alias: Optional[str]
+    subquery: Optional["Query"] = None
 
 
 @dataclass
@@ -287,6 +308,17 @@ class Parser:
                      order_by, limit)
 
     def parse_table_ref(self) -> TableRef:
+        if self.accept_op("("):
+            q = self.parse_select()
+            self.expect_op(")")
+            alias = None
+            if self.accept_kw("as"):
+                alias = self.next().value
+            else:
+                nt = self.peek()
+                if nt and nt.kind == "id":
+                    alias = self.next().value
+            return TableRef(alias or "__subq", alias, subquery=q)
         t = self.next()
         if t.kind != "id":
             raise SyntaxError(f"expected table name, got {t}")
@@ -350,6 +382,11 @@ class Parser:
         if t and t.kind == "kw" and t.value == "in":
             self.i += 1
             self.expect_op("(")
+            nt = self.peek()
+            if nt and nt.kind == "kw" and nt.value == "select":
+                q = self.parse_select()
+                self.expect_op(")")
+                return InSubquery(left, q, negated)
             vals = [self.parse_expr()]
             while self.accept_op(","):
                 vals.append(self.parse_expr())
@@ -468,9 +505,17 @@ class Parser:
                 self.expect_op(")")
                 return Func("substring", [e, start, length])
             if t.value == "exists":
-                raise SyntaxError("EXISTS subqueries not supported yet")
+                self.expect_op("(")
+                q = self.parse_select()
+                self.expect_op(")")
+                return ExistsE(q)
             raise SyntaxError(f"unexpected keyword {t.value!r}")
         if t.kind == "op" and t.value == "(":
+            nt = self.peek()
+            if nt and nt.kind == "kw" and nt.value == "select":
+                q = self.parse_select()
+                self.expect_op(")")
+                return SubqueryE(q)
             e = self.parse_expr()
             self.expect_op(")")
             return e

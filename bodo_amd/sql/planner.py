@@ -27,25 +27,41 @@ CAST_TYPES = {
 
 
 class Scope:
-    """Maps (alias, column) -> internal plan column name."""
+    """Maps (alias, column) -> internal plan column name; `parent` links a
+    subquery scope to the outer query for correlation detection."""
 
-    def __init__(self):
+    def __init__(self, parent: Optional["Scope"] = None):
         self.entries: List[Tuple[str, str, str]] = []  # alias, col, internal
+        self.parent = parent
 
     def add(self, alias: str, col: str, internal: str):
         self.entries.append((alias, col, internal))
 
-    def resolve(self, table: Optional[str], name: str) -> str:
+    def resolve_local(self, table: Optional[str], name: str) -> Optional[str]:
         cands = []
         for alias, col, internal in self.entries:
             if col.lower() == name.lower() and (table is None
                                                 or alias.lower() == table.lower()):
                 cands.append(internal)
         if not cands:
-            raise KeyError(f"unknown column {table + '.' if table else ''}{name}")
+            return None
         if len(set(cands)) > 1:
             raise KeyError(f"ambiguous column {name}")
         return cands[0]
+
+    def resolve(self, table: Optional[str], name: str) -> str:
+        r = self.resolve_local(table, name)
+        if r is not None:
+            return r
+        if self.parent is not None:
+            return self.parent.resolve(table, name)
+        raise KeyError(f"unknown column {table + '.' if table else ''}{name}")
+
+    def is_outer(self, table: Optional[str], name: str) -> bool:
+        """True when the column resolves only in an enclosing scope."""
+        if self.resolve_local(table, name) is not None:
+            return False
+        return self.parent is not None
 
     def tables_of(self, e) -> set:
         """Set of internal columns referenced by an AST expression."""
@@ -107,8 +123,18 @@ class Planner:
                     pair = self._equi_pair(c, scope, joined_cols, sub_cols)
                     if pair:
                         eq_pairs.append(pair)
-                    else:
+                        continue
+                    refs_c = {scope.resolve(x.table, x.name)
+                              for x in _col_refs(c)}
+                    if kind in ("left", "inner") and refs_c <= sub_cols:
+                        # condition touches only the new (right) table:
+                        # pre-filter it (required for LEFT join semantics)
+                        sub_plan = pn.Filter(sub_plan, self.expr(c, scope))
+                    elif kind == "inner":
                         post.append(c)
+                    else:
+                        raise NotImplementedError(
+                            "outer-join ON condition across both sides")
             if kind == "cross" or not eq_pairs:
                 # pull applicable equi conditions from WHERE (FROM a, b style)
                 for ci, c in enumerate(where_conjuncts):
@@ -132,11 +158,16 @@ class Planner:
                 plan = pn.Filter(plan, self.expr(c, scope))
         if plan is None:
             raise ValueError("SELECT without FROM not supported")
-        # remaining WHERE
+        # remaining WHERE: subquery predicates become plan transforms
         rest = [c for ci, c in enumerate(where_conjuncts) if not used_conjuncts[ci]]
-        if rest:
-            cond = rest[0]
-            for c in rest[1:]:
+        plain = []
+        for c in rest:
+            handled, plan = self._apply_subquery_pred(c, plan, scope)
+            if not handled:
+                plain.append(c)
+        if plain:
+            cond = plain[0]
+            for c in plain[1:]:
                 cond = ast.Bin("and", cond, c)
             plan = pn.Filter(plan, self.expr(cond, scope))
 
@@ -200,12 +231,16 @@ class Planner:
 
     # ------------------------------------------------------------------
     def _table_plan(self, tr: ast.TableRef, scope: Scope):
-        name = tr.name.lower()
-        if name not in self.tables:
-            raise KeyError(f"unknown table {tr.name}")
-        frame = self.tables[name]
         alias = (tr.alias or tr.name).lower()
-        cols = list(frame._columns)
+        if tr.subquery is not None:
+            sub, cols = self.plan(tr.subquery)
+        else:
+            name = tr.name.lower()
+            if name not in self.tables:
+                raise KeyError(f"unknown table {tr.name}")
+            frame = self.tables[name]
+            sub = frame._lazy_plan
+            cols = list(frame._columns)
         internal = []
         need_rename = False
         for c in cols:
@@ -215,7 +250,6 @@ class Planner:
                 need_rename = True
             internal.append(iname)
             scope.add(alias, c, iname)
-        sub = frame._lazy_plan
         if need_rename:
             sub = pn.Projection(sub, tuple(internal),
                                 tuple(ex.ColRef(c) for c in cols))
@@ -234,6 +268,234 @@ class Planner:
             if r in left_cols and l in right_cols:
                 return (r, l)
         return None
+
+    # ------------------------------------------------------------------
+    # subqueries (reference: BodoSQL/Calcite decorrelation rule sets)
+    # ------------------------------------------------------------------
+    def _apply_subquery_pred(self, c, plan, scope: Scope):
+        """IN/EXISTS/correlated-scalar predicates -> semi/anti/agg joins.
+        Returns (handled, new_plan)."""
+        neg = False
+        inner = c
+        while isinstance(inner, ast.Un) and inner.op == "not":
+            neg = not neg
+            inner = inner.operand
+        if isinstance(inner, ast.InSubquery):
+            return True, self._in_subquery(inner.operand, inner.query,
+                                           neg ^ inner.negated, plan, scope)
+        if isinstance(inner, ast.ExistsE):
+            return True, self._exists_subquery(inner.query,
+                                               neg ^ inner.negated, plan,
+                                               scope)
+        if isinstance(inner, ast.Bin) and inner.op in (
+                "lt", "le", "gt", "ge", "eq", "ne"):
+            l, r = inner.left, inner.right
+            flip = {"lt": "gt", "le": "ge", "gt": "lt", "ge": "le",
+                    "eq": "eq", "ne": "ne"}
+            if isinstance(l, ast.SubqueryE) and not isinstance(r, ast.SubqueryE):
+                l, r = r, l
+                inner = ast.Bin(flip[inner.op], l, r)
+            if isinstance(inner.right, ast.SubqueryE):
+                if neg:
+                    return False, plan  # NOT (x < (subq)): keep expr path
+                return True, self._scalar_cmp_subquery(
+                    inner.op, inner.left, inner.right.query, plan, scope)
+        return False, plan
+
+    def _plan_inner(self, q: ast.Query, outer: Scope):
+        """Plan a subquery's FROM+WHERE; returns (plan, scope, corr_pairs)
+        with correlation equalities removed from the filter."""
+        scope = Scope(parent=outer)
+        plan = None
+        joined = set()
+        conjs = _split_conjuncts(q.where) if q.where is not None else []
+        used = [False] * len(conjs)
+        refs = ([q.table] if q.table else []) + [j.table for j in q.joins]
+        kinds = ["base"] + [j.kind for j in q.joins]
+        ons = [None] + [j.on for j in q.joins]
+        for idx, tr in enumerate(refs):
+            sub_plan, sub_cols = self._table_plan(tr, scope)
+            if plan is None:
+                plan = sub_plan
+                joined |= sub_cols
+                continue
+            eq = []
+            if ons[idx] is not None:
+                for cj in _split_conjuncts(ons[idx]):
+                    p = self._equi_pair(cj, scope, joined, sub_cols)
+                    if p is None:
+                        raise NotImplementedError("non-equi subquery join")
+                    eq.append(p)
+            for ci, cj in enumerate(conjs):
+                if used[ci]:
+                    continue
+                p = self._equi_pair(cj, scope, joined, sub_cols)
+                if p:
+                    eq.append(p)
+                    used[ci] = True
+            if eq:
+                plan = pn.Join(plan, sub_plan, tuple(x[0] for x in eq),
+                               tuple(x[1] for x in eq), "inner")
+            else:
+                plan = pn.Join(plan, sub_plan, (), (), "cross")
+            joined |= sub_cols
+        corr_pairs = []
+        inner_keep = []
+        for ci, cj in enumerate(conjs):
+            if used[ci]:
+                continue
+            pair = self._corr_pair(cj, scope)
+            if pair is not None:
+                corr_pairs.append(pair)
+                continue
+            handled, plan = self._apply_subquery_pred(cj, plan, scope)
+            if not handled:
+                inner_keep.append(cj)
+        if inner_keep:
+            cond = inner_keep[0]
+            for cj in inner_keep[1:]:
+                cond = ast.Bin("and", cond, cj)
+            plan = pn.Filter(plan, self.expr(cond, scope))
+        return plan, scope, corr_pairs
+
+    def _corr_pair(self, cj, sub_scope: Scope):
+        """inner_col = outer_col -> (outer_internal, inner_internal)."""
+        if not (isinstance(cj, ast.Bin) and cj.op == "eq"
+                and isinstance(cj.left, ast.Col)
+                and isinstance(cj.right, ast.Col)):
+            return None
+        l, r = cj.left, cj.right
+        l_outer = sub_scope.is_outer(l.table, l.name)
+        r_outer = sub_scope.is_outer(r.table, r.name)
+        if l_outer == r_outer:
+            return None
+        o, i = (l, r) if l_outer else (r, l)
+        return (sub_scope.parent.resolve(o.table, o.name),
+                sub_scope.resolve_local(i.table, i.name))
+
+    def _try_plan_full(self, q: ast.Query):
+        """Full planner for an (assumed) uncorrelated subquery; None when it
+        references enclosing columns (KeyError) or uses unsupported shapes."""
+        try:
+            return self.plan(q)
+        except (KeyError, NotImplementedError, ValueError):
+            return None
+
+    def _in_subquery(self, operand, q: ast.Query, negated, plan, scope):
+        if q.group_by or q.having is not None or any(
+                _has_agg(it.expr) for it in q.items if not it.star):
+            full = self._try_plan_full(q)
+            if full is None:
+                raise NotImplementedError("correlated grouped IN subquery")
+            sub_plan, out_names = full
+            op_expr = self.expr(operand, scope)
+            if not isinstance(op_expr, ex.ColRef):
+                raise NotImplementedError("IN subquery on computed operand")
+            pfx = self._uniq("__sq")
+            renamed = (f"{pfx}_{out_names[0]}",)
+            proj = pn.Projection(sub_plan, renamed,
+                                 (ex.ColRef(out_names[0]),))
+            dis = pn.Distinct(proj, renamed)
+            return pn.Join(plan, dis, (op_expr.name,), renamed,
+                           "anti" if negated else "semi")
+        sub_plan, sub_scope, corr = self._plan_inner(q, scope)
+        item = q.items[0]
+        val_internal = self._subq_value_col(item, sub_scope, sub_plan)
+        sub_plan = val_internal[1]
+        val_col = val_internal[0]
+        op_expr = self.expr(operand, scope)
+        if not isinstance(op_expr, ex.ColRef):
+            raise NotImplementedError("IN subquery on computed operand")
+        pfx = self._uniq("__sq")
+        keep = [val_col] + [i for _, i in corr]
+        renamed = [f"{pfx}_{c}" for c in keep]
+        proj = pn.Projection(sub_plan, tuple(renamed),
+                             tuple(ex.ColRef(c) for c in keep))
+        dis = pn.Distinct(proj, tuple(renamed))
+        lks = (op_expr.name,) + tuple(o for o, _ in corr)
+        rks = tuple(renamed)
+        return pn.Join(plan, dis, lks, rks, "anti" if negated else "semi")
+
+    def _exists_subquery(self, q: ast.Query, negated, plan, scope):
+        sub_plan, sub_scope, corr = self._plan_inner(q, scope)
+        if not corr:
+            raise NotImplementedError("uncorrelated EXISTS")
+        pfx = self._uniq("__sq")
+        keep = [i for _, i in corr]
+        renamed = [f"{pfx}_{c}" for c in keep]
+        proj = pn.Projection(sub_plan, tuple(renamed),
+                             tuple(ex.ColRef(c) for c in keep))
+        dis = pn.Distinct(proj, tuple(renamed))
+        lks = tuple(o for o, _ in corr)
+        return pn.Join(plan, dis, lks, tuple(renamed),
+                       "anti" if negated else "semi")
+
+    def _scalar_cmp_subquery(self, op, outer_expr, q: ast.Query, plan, scope):
+        full = self._try_plan_full(q)
+        if full is not None:
+            sub_plan, out_names = full
+            return pn.Filter(plan, ex.Cmp(
+                op, self.expr(outer_expr, scope),
+                ex.ScalarSubquery(sub_plan, out_names[0])))
+        sub_plan, sub_scope, corr = self._plan_inner(q, scope)
+        item = q.items[0]
+        agg_ast, rebuild = _extract_single_agg(item.expr)
+        if agg_ast is None:
+            raise NotImplementedError("scalar subquery must be an aggregate")
+        keys = tuple(i for _, i in corr)
+        val, aplan = self._subq_agg_plan(agg_ast, sub_scope, sub_plan, keys)
+        if rebuild is not None:
+            # arithmetic around the aggregate (e.g. 0.2 * avg(x))
+            wrapped = self._uniq("__sw")
+            aplan = pn.Projection(
+                aplan, tuple(list(keys) + [wrapped]),
+                tuple([ex.ColRef(k) for k in keys]
+                      + [rebuild(ex.ColRef(val), self, sub_scope)]))
+            val = wrapped
+        if not corr:
+            return pn.Filter(plan, ex.Cmp(
+                op, self.expr(outer_expr, scope),
+                ex.ScalarSubquery(aplan, val)))
+        pfx = self._uniq("__sq")
+        renamed = [f"{pfx}_{c}" for c in keys] + [f"{pfx}_{val}"]
+        proj = pn.Projection(aplan, tuple(renamed),
+                             tuple(ex.ColRef(c) for c in list(keys) + [val]))
+        joined = pn.Join(plan, proj, tuple(o for o, _ in corr),
+                         tuple(renamed[:-1]), "inner")
+        return pn.Filter(joined, ex.Cmp(
+            op, self.expr(outer_expr, scope), ex.ColRef(renamed[-1])))
+
+    def _subq_agg_plan(self, agg_ast, sub_scope, sub_plan, keys):
+        func = AGG_FUNCS[agg_ast.name]
+        out = self._uniq("__sv")
+        if agg_ast.star or not agg_ast.args:
+            return out, pn.Aggregate(sub_plan, tuple(keys),
+                                     ((out, "", "size"),)) if keys else                 (out, pn.Reduce(sub_plan, ((out, "", "size"),)))
+        arg = agg_ast.args[0]
+        if isinstance(arg, ast.Col):
+            in_name = sub_scope.resolve(arg.table, arg.name)
+            pre = sub_plan
+        else:
+            in_name = self._uniq("__ain")
+            cols = [e[2] for e in sub_scope.entries]
+            pre = pn.Projection(
+                sub_plan, tuple(cols + [in_name]),
+                tuple([ex.ColRef(c) for c in cols]
+                      + [self.expr(arg, sub_scope)]))
+        if keys:
+            return out, pn.Aggregate(pre, tuple(keys), ((out, in_name, func),))
+        return out, pn.Reduce(pre, ((out, in_name, func),))
+
+    def _subq_value_col(self, item, sub_scope, sub_plan):
+        if isinstance(item.expr, ast.Col):
+            return sub_scope.resolve(item.expr.table, item.expr.name), sub_plan
+        name = self._uniq("__sqv")
+        cols = [e[2] for e in sub_scope.entries]
+        proj = pn.Projection(
+            sub_plan, tuple(cols + [name]),
+            tuple([ex.ColRef(c) for c in cols]
+                  + [self.expr(item.expr, sub_scope)]))
+        return name, proj
 
     # ------------------------------------------------------------------
     def _aggregate(self, plan, scope: Scope, q: ast.Query):
@@ -334,6 +596,12 @@ class Planner:
         if isinstance(e, ast.CastE):
             return ex.Cast(self._post_agg_expr(e.operand, key_map, agg_map,
                                                scope), CAST_TYPES[e.to])
+        if isinstance(e, ast.SubqueryE):
+            full = self._try_plan_full(e.query)
+            if full is None:
+                raise NotImplementedError("correlated subquery in HAVING")
+            sub_plan, out_names = full
+            return ex.ScalarSubquery(sub_plan, out_names[0])
         raise NotImplementedError(f"post-agg expr {e}")
 
     # ------------------------------------------------------------------
@@ -453,6 +721,52 @@ def _like_expr(operand: ex.Expr, pattern: str) -> ex.Expr:
 
     rx = _re.escape(pattern).replace("%", ".*").replace("_", ".")
     return ex.StrOp(operand, "contains_re", ("^" + rx + "$",))
+
+
+def _extract_single_agg(e):
+    """(agg_func_ast, rebuild) where rebuild(value_expr, planner, scope)
+    re-applies any arithmetic wrapper; (None, None) if there is no single
+    aggregate call."""
+    if isinstance(e, ast.Func) and e.name in AGG_FUNCS:
+        return e, None
+    if isinstance(e, ast.Bin) and e.op in ("add", "sub", "mul", "div"):
+        for side, other, right in ((e.left, e.right, True),
+                                   (e.right, e.left, False)):
+            if isinstance(side, ast.Func) and side.name in AGG_FUNCS                     and isinstance(other, ast.Lit):
+                op = e.op
+
+                def rebuild(val, planner, scope, _op=op, _lit=other,
+                            _agg_left=not right):
+                    lit = planner.expr(_lit, scope)
+                    if _agg_left:
+                        return ex.BinOp(_op, lit, val)
+                    return ex.BinOp(_op, val, lit)
+
+                return side, rebuild
+    return None, None
+
+
+def _col_refs(e) -> list:
+    out = []
+
+    def walk(x):
+        if isinstance(x, ast.Col):
+            out.append(x)
+        for f in getattr(x, "__dataclass_fields__", {}):
+            v = getattr(x, f)
+            if isinstance(v, (list, tuple)):
+                for i in v:
+                    if hasattr(i, "__dataclass_fields__"):
+                        walk(i)
+            elif hasattr(v, "__dataclass_fields__"):
+                walk(v)
+
+    walk(e)
+    return out
+
+
+def sub_scope_outer(scope, col, q):
+    return scope.is_outer(col.table, col.name)
 
 
 def _split_conjuncts(e) -> list:
