@@ -246,3 +246,43 @@ def test_streaming_aggregate(tmp_path, monkeypatch):
         assert abs(s - exp_s) < 1e-6
     finally:
         cfg.STREAMING, cfg.STREAM_BATCH_SIZE = old_mode, old_batch
+
+
+def test_tpcxbb_q05_style():
+    """TPCx-BB Q05 shape: per-user category-click pivot via SQL aggregation
+    feeding a distributed logistic regression (BASELINE config 5)."""
+    import bodo_amd.pandas as bpd
+    from bodo_amd.ml import LogisticRegression
+    from bodo_amd.sql import BodoSQLContext
+
+    rng = np.random.default_rng(11)
+    n = 30000
+    clicks = pd.DataFrame({
+        "wcs_user_sk": rng.integers(1, 2000, n),
+        "i_category_id": rng.integers(1, 8, n),
+    })
+    users = pd.DataFrame({
+        "c_customer_sk": np.arange(1, 2001),
+        "c_education": rng.integers(0, 5, 2000),
+    })
+    bc = BodoSQLContext({"clicks": clicks, "users": users})
+    feats = bc.sql("""
+        SELECT wcs_user_sk,
+               SUM(CASE WHEN i_category_id = 1 THEN 1 ELSE 0 END) AS cat1,
+               SUM(CASE WHEN i_category_id = 2 THEN 1 ELSE 0 END) AS cat2,
+               SUM(CASE WHEN i_category_id = 3 THEN 1 ELSE 0 END) AS cat3,
+               COUNT(*) AS total
+        FROM clicks GROUP BY wcs_user_sk
+    """)
+    joined = feats.merge(bpd.from_pandas(users), left_on="wcs_user_sk",
+                         right_on="c_customer_sk")
+    pdf = joined.to_pandas()
+    X = pdf[["cat1", "cat2", "cat3", "total"]].to_numpy(dtype=np.float64)
+    y = (pdf["c_education"] >= 3).to_numpy(dtype=np.float64)
+    m = LogisticRegression(lr=0.5, max_iter=100).fit(X, y)
+    proba = m.predict_proba(X)
+    assert proba.shape[0] == len(pdf) and np.isfinite(proba).all()
+    # sanity vs pandas-computed features
+    exp = clicks.groupby("wcs_user_sk").size()
+    got = pdf.set_index("wcs_user_sk")["total"].sort_index()
+    assert (got == exp.loc[got.index]).all()
