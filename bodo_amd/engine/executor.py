@@ -214,9 +214,26 @@ def _exec_sort(node: pn.Sort, ctx) -> Table:
 def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:
     """Sample-based range partitioning so rank r holds globally-contiguous key
     range r (PSRS step 1-3; reference: streaming/_sort.h reservoir sampling +
-    bounds)."""
+    bounds).  Packable key sets partition entirely on device (sampled packed
+    keys -> allgathered splitters -> device searchsorted)."""
     w = ctx.world
     n = len(tbl)
+    packed = ops.pack_ordered_keys([tbl.column(k) for k in keys], asc)
+    if packed is not None:
+        k = min(n, 64 * w)
+        if n > 0:
+            pos = torch.randint(0, n, (k,), device=tbl.device) if n > k else                 torch.arange(n, device=tbl.device)
+            sample = packed[pos].cpu().numpy()
+        else:
+            sample = np.zeros(0, dtype=np.int64)
+        merged = np.sort(np.concatenate(comm.allgather_obj(sample)))
+        if len(merged) == 0:
+            return tbl
+        q = [min(int(len(merged) * (i + 1) / w), len(merged) - 1)
+             for i in range(w - 1)]
+        splitters = torch.from_numpy(merged[q].copy()).to(tbl.device)
+        part = torch.searchsorted(splitters, packed, right=True)
+        return comm.shuffle_table(tbl, part)
     # sample up to 64*w rows of the key columns
     k = min(n, 64 * w)
     if k > 0:

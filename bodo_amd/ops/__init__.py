@@ -259,9 +259,14 @@ def partition_indices(hashes: torch.Tensor, nparts: int) -> Tuple[torch.Tensor, 
 
 def sort_indices(cols: Sequence[Column], ascending: Sequence[bool],
                  na_position: str = "last") -> torch.Tensor:
-    """Stable lexicographic argsort (last key first, stable iterations)."""
+    """Stable lexicographic argsort (last key first, stable iterations).
+    Small-range key sets pack into ONE order-preserving u64 -> a single
+    argsort instead of one stable pass per key."""
     device = cols[0].device if cols else torch.device("cpu")
     n = len(cols[0]) if cols else 0
+    packed = pack_ordered_keys(cols, ascending)
+    if packed is not None:
+        return torch.argsort(packed, stable=True)
     idx = torch.arange(n, dtype=torch.int64, device=device)
     for c, asc in reversed(list(zip(cols, ascending))):
         keys = _sort_key_tensor(c, asc, na_position)
@@ -269,6 +274,51 @@ def sort_indices(cols: Sequence[Column], ascending: Sequence[bool],
         order = torch.argsort(k, stable=True)
         idx = idx[order]
     return idx
+
+
+def pack_ordered_keys(cols: Sequence[Column],
+                      ascending: Sequence[bool]) -> Optional[torch.Tensor]:
+    """Pack sort keys into one int64 whose numeric order equals the
+    lexicographic (keys, ascending) order.  First key occupies the most
+    significant bits; DICT codes are rank-remapped by dictionary value
+    order; descending keys are bit-flipped within their range.  Returns
+    None when any key is unsuitable (mask, float, wide range, strings)."""
+    if not cols:
+        return None
+    widths = []
+    parts = []
+    for c, asc in zip(cols, ascending):
+        if c.mask is not None:
+            return None
+        k = c.dtype.kind
+        if k == TypeKind.DICT:
+            import pyarrow.compute as pc
+
+            order = pc.array_sort_indices(c.dictionary).to_numpy()
+            rank = np.empty(len(order), dtype=np.int64)
+            rank[order] = np.arange(len(order))
+            vals = torch.from_numpy(rank).to(c.device)[c.data.long()]
+            lo, hi = 0, max(0, len(order) - 1)
+        elif k == TypeKind.BOOL:
+            vals = c.data.long()
+            lo, hi = 0, 1
+        elif (c.dtype.is_integer or k == TypeKind.DATE32) and c.val_range:
+            lo, hi = c.val_range
+            vals = c.data.long()
+        else:
+            return None
+        width = max(1, int(hi - lo + 1).bit_length())
+        v = vals - lo
+        if not asc:
+            v = (hi - lo) - v
+        parts.append(v)
+        widths.append(width)
+    if sum(widths) > 63:
+        return None
+    packed = None
+    for v, w in zip(parts, widths):
+        packed = v if packed is None else ((packed << w) | v)
+    return packed
 
 
 def _sort_key_tensor(c: Column, asc: bool, na_position: str) -> torch.Tensor:

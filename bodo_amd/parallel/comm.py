@@ -151,20 +151,20 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
     packed = ops.take_table(tbl, perm)
     send_counts = [int(c) for c in counts.tolist()]
     recv_counts = [int(c) for c in _exchange_counts(counts).tolist()]
+    # ONE consensus round for mask presence across all columns (a per-column
+    # object allgather serializes w latencies at 8 ranks)
+    local_has_mask = [c.mask is not None for c in packed.columns]
+    votes = allgather_obj(local_has_mask)
+    has_mask = [any(v[i] for v in votes) for i in range(len(local_has_mask))]
     out_cols = []
-    for col in packed.columns:
-        out_cols.append(_shuffle_column(col, send_counts, recv_counts))
+    for col, hm in zip(packed.columns, has_mask):
+        out_cols.append(_shuffle_column(col, send_counts, recv_counts, hm))
     return Table(packed.names, out_cols, sum(recv_counts))
 
 
-def _shuffle_column(col: Column, send_counts, recv_counts) -> Column:
+def _shuffle_column(col: Column, send_counts, recv_counts,
+                    has_mask: bool) -> Column:
     n_out = sum(recv_counts)
-    mask = None
-    if col.mask is not None or True:
-        # always exchange masks if any rank has one (cheap consensus via
-        # object allgather would serialize; just send a materialized mask
-        # when present locally and let others send all-ones)
-        pass
     if col.dtype.kind == TypeKind.STRING:
         # exchange per-row lengths, then bytes with byte-granular counts
         lens = (col.offsets[1:] - col.offsets[:-1]).contiguous()
@@ -180,18 +180,17 @@ def _shuffle_column(col: Column, send_counts, recv_counts) -> Column:
         new_bytes = alltoallv_tensor(col.data, byte_send, list(byte_recv_np))
         new_off = torch.zeros(n_out + 1, dtype=torch.int64, device=col.device)
         torch.cumsum(new_lens, 0, out=new_off[1:])
-        new_mask = _shuffle_mask(col, send_counts, recv_counts)
+        new_mask = _shuffle_mask(col, send_counts, recv_counts, has_mask)
         return Column(bt.string, new_bytes, new_mask, offsets=new_off, length=n_out)
     data = alltoallv_tensor(col.data.contiguous(), send_counts, recv_counts)
-    new_mask = _shuffle_mask(col, send_counts, recv_counts)
+    new_mask = _shuffle_mask(col, send_counts, recv_counts, has_mask)
     if col.dtype.kind == TypeKind.DICT:
         # dictionaries may differ per-rank; unify by value after exchange
         return _unify_dict_after_shuffle(col, data, new_mask, send_counts, recv_counts, n_out)
     return Column(col.dtype, data, new_mask, length=n_out)
 
 
-def _shuffle_mask(col: Column, send_counts, recv_counts):
-    has_mask = any(allgather_obj(col.mask is not None))
+def _shuffle_mask(col: Column, send_counts, recv_counts, has_mask: bool):
     if not has_mask:
         return None
     m = col.mask if col.mask is not None else torch.ones(
