@@ -316,7 +316,16 @@ extern "C" __global__ void fused(long long n, {', '.join(params)}) {{
         from .hip_udf import _runtime
 
         rt = _runtime()
-        fn = rt.get_kernel_named(src, b"fused")
+        import os as _os
+        if _os.environ.get("BODO_AMD_FUSE_LOG"):
+            import sys as _sys
+            import time as _time
+            _t0 = _time.perf_counter()
+            fn = rt.get_kernel_named(src, b"fused")
+            print(f"[fuse-compile] {1e3 * (_time.perf_counter() - _t0):.0f}ms "
+                  f"sig={sig[:160]}", file=_sys.stderr, flush=True)
+        else:
+            fn = rt.get_kernel_named(src, b"fused")
         entry = (fn, [t for _, (_, t) in emitted])
         if len(_KERNEL_CACHE) > 512:
             _KERNEL_CACHE.clear()

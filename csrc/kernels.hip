@@ -757,6 +757,8 @@ struct FusedAgg {
   int op;
   void* acc;
   int64_t* cnt;
+  double init_f;
+  int64_t init_i;
 };
 
 __global__ void agg_update_fused_kernel(FusedAgg a0, FusedAgg a1, FusedAgg a2,
@@ -806,6 +808,88 @@ __global__ void agg_update_fused_kernel(FusedAgg a0, FusedAgg a1, FusedAgg a2,
   }
 }
 
+// LDS pre-aggregation variant of the fused kernel for low-cardinality keys
+// (TPC-H Q1: 6 groups, 8 aggregates).  Without it 60M rows x 4 aggs of
+// global atomics on 6 slots measured 274 ms/launch; with block-local LDS
+// accumulators the global traffic is ~grid*ngroups atomics.
+__global__ void agg_update_fused_lds_kernel(
+    FusedAgg a0, FusedAgg a1, FusedAgg a2, FusedAgg a3, int n_aggs,
+    int ngroups, const int32_t* __restrict__ row_gid, int64_t n) {
+  FusedAgg aggs[MAX_FUSED_AGGS] = {a0, a1, a2, a3};
+  extern __shared__ double smem[];
+  double* lacc = smem;                              // n_aggs * ngroups
+  long long* lcnt = (long long*)(smem + (size_t)n_aggs * ngroups);
+  for (int j = threadIdx.x; j < n_aggs * ngroups; j += blockDim.x) {
+    int k = j / ngroups;
+    int op = aggs[k].op;
+    bool f64_acc = (op == AGG_SUM_F64 || op == AGG_MIN_F64 || op == AGG_MAX_F64);
+    lacc[j] = f64_acc ? aggs[k].init_f : __longlong_as_double(aggs[k].init_i);
+    lcnt[j] = 0;
+  }
+  __syncthreads();
+  GRID_STRIDE_LOOP(i, n) {
+    int32_t g = row_gid[i];
+    for (int k = 0; k < n_aggs; ++k) {
+      double* la = lacc + (size_t)k * ngroups;
+      long long* lc = lcnt + (size_t)k * ngroups;
+      int op = aggs[k].op;
+      if (op == AGG_SIZE) {
+        atomicAdd((unsigned long long*)&lc[g], 1ull);
+        continue;
+      }
+      const ColumnDesc& col = aggs[k].col;
+      bool valid = is_valid_at(col, i);
+      double dv = 0.0;
+      int64_t iv = 0;
+      if (valid) {
+        switch (col.dtype) {
+          case BT_INT8: iv = ((const int8_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_UINT8: case BT_BOOL: iv = ((const uint8_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_INT16: case BT_UINT16: iv = ((const int16_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_INT32: case BT_UINT32: case BT_DATE32: case BT_DICT:
+            iv = ((const int32_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_INT64: case BT_UINT64: case BT_TIMESTAMP_NS:
+            iv = ((const int64_t*)col.data)[i]; dv = (double)iv; break;
+          case BT_FLOAT32: { float f = ((const float*)col.data)[i]; valid = !(f != f); dv = (double)f; iv = (int64_t)f; break; }
+          case BT_FLOAT64: { double f = ((const double*)col.data)[i]; valid = !(f != f); dv = f; iv = (int64_t)f; break; }
+        }
+      }
+      if (!valid) continue;
+      switch (op) {
+        case AGG_SUM_F64: atomicAdd(&la[g], dv); break;
+        case AGG_SUM_I64: atomicAdd((unsigned long long*)&la[g], (unsigned long long)iv); break;
+        case AGG_COUNT: break;
+        case AGG_MIN_F64: atomic_min_f64(&la[g], dv); break;
+        case AGG_MAX_F64: atomic_max_f64(&la[g], dv); break;
+        case AGG_MIN_I64: atomic_min_i64((int64_t*)&la[g], iv); break;
+        case AGG_MAX_I64: atomic_max_i64((int64_t*)&la[g], iv); break;
+      }
+      atomicAdd((unsigned long long*)&lc[g], 1ull);
+    }
+  }
+  __syncthreads();
+  for (int j = threadIdx.x; j < n_aggs * ngroups; j += blockDim.x) {
+    int k = j / ngroups;
+    int g = j - k * ngroups;
+    int op = aggs[k].op;
+    void* acc = aggs[k].acc;
+    long long c = lcnt[j];
+    if (aggs[k].cnt != nullptr && c)
+      atomicAdd((unsigned long long*)&aggs[k].cnt[g], (unsigned long long)c);
+    if (op == AGG_SIZE || op == AGG_COUNT) continue;
+    if (c == 0) continue;
+    switch (op) {
+      case AGG_SUM_F64: atomicAdd((double*)acc + g, lacc[j]); break;
+      case AGG_SUM_I64: atomicAdd((unsigned long long*)acc + g,
+                                  (unsigned long long)__double_as_longlong(lacc[j])); break;
+      case AGG_MIN_F64: atomic_min_f64((double*)acc + g, lacc[j]); break;
+      case AGG_MAX_F64: atomic_max_f64((double*)acc + g, lacc[j]); break;
+      case AGG_MIN_I64: atomic_min_i64((int64_t*)acc + g, __double_as_longlong(lacc[j])); break;
+      case AGG_MAX_I64: atomic_max_i64((int64_t*)acc + g, __double_as_longlong(lacc[j])); break;
+    }
+  }
+}
+
 // returns per agg: [acc, cnt?] pairs flattened
 std::vector<torch::Tensor> agg_update_fused(
     std::vector<torch::Tensor> datas,
@@ -835,13 +919,23 @@ std::vector<torch::Tensor> agg_update_fused(
     fas[k].op = (int)op;
     fas[k].acc = acc.data_ptr();
     fas[k].cnt = cnt.defined() ? (int64_t*)cnt.data_ptr() : nullptr;
+    fas[k].init_f = init_fs[k];
+    fas[k].init_i = init_is[k];
     out.push_back(acc);
     out.push_back(cnt.defined() ? cnt : torch::Tensor());
   }
   int block = 256;
-  hipLaunchKernelGGL(agg_update_fused_kernel, dim3(grid_for(n, block)),
-                     dim3(block), 0, cur_stream(), fas[0], fas[1], fas[2],
-                     fas[3], n_aggs, (const int32_t*)row_gid.data_ptr(), n);
+  size_t lds_bytes = (size_t)n_aggs * (size_t)ngroups * 16;
+  if (lds_bytes <= 48 * 1024) {
+    hipLaunchKernelGGL(agg_update_fused_lds_kernel, dim3(grid_for(n, block)),
+                       dim3(block), lds_bytes, cur_stream(), fas[0], fas[1],
+                       fas[2], fas[3], n_aggs, (int)ngroups,
+                       (const int32_t*)row_gid.data_ptr(), n);
+  } else {
+    hipLaunchKernelGGL(agg_update_fused_kernel, dim3(grid_for(n, block)),
+                       dim3(block), 0, cur_stream(), fas[0], fas[1], fas[2],
+                       fas[3], n_aggs, (const int32_t*)row_gid.data_ptr(), n);
+  }
   CHECK_HIP(hipGetLastError());
   std::vector<torch::Tensor> cleaned;
   for (auto& t : out) cleaned.push_back(t.defined() ? t : torch::empty({0}));
