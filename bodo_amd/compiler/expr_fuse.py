@@ -66,6 +66,12 @@ class _Fuser:
         self.inputs: Dict[str, Tuple[int, Column]] = {}  # name -> (idx, col)
         self.lines: List[str] = []
         self.tmp = 0
+        # scalar constants passed as kernel ARGUMENTS (not source literals)
+        # so one compiled kernel serves every constant value — without this,
+        # data-dependent filter scalars (TPC-H q11/q15/q22 computed
+        # thresholds, which differ by 1 ULP run-to-run from atomic-add
+        # ordering) forced a ~93 ms hipRTC recompile inside the timed query.
+        self.scalars: List[Tuple[str, object]] = []  # (ctype, ctypes value)
 
     def input_of(self, name: str) -> str:
         col = self.tbl.column(name)
@@ -103,10 +109,13 @@ class _Fuser:
             v = normalize_const(e.value, dtype)
             if isinstance(v, bool):
                 return ("true" if v else "false"), "bool"
+            name = f"c{len(self.scalars)}"
             if isinstance(v, int):
-                return f"{v}LL", "i64"
+                self.scalars.append(("long long", ctypes.c_longlong(v)))
+                return name, "i64"
             if isinstance(v, float):
-                return repr(v), "double"
+                self.scalars.append(("double", ctypes.c_double(float(v))))
+                return name, "double"
             raise _NotFusable
         if isinstance(e, BinOp):
             l, lt = self.emit(e.left)
@@ -259,7 +268,18 @@ def _expr_signature(e: Expr, tbl: Table) -> str:
     if isinstance(e, ColRef):
         return f"col({e.name}:{tbl.column(e.name).dtype.kind})"
     if isinstance(e, Const):
-        return f"const({e.value!r})"
+        # keyed by the emitted TYPE only: the value is a kernel argument,
+        # so one kernel serves all constant values of this structure
+        from ..ops.evaluate import infer_const_dtype, normalize_const
+
+        try:
+            v = normalize_const(e.value, e.dtype or infer_const_dtype(e.value))
+        except Exception:
+            return f"const({e.value!r})"
+        if isinstance(v, bool):
+            return f"const(bool:{v})"
+        return "const(i64)" if isinstance(v, int) else \
+            ("const(f64)" if isinstance(v, float) else f"const({v!r})")
     parts = [type(e).__name__]
     for f in getattr(e, "__dataclass_fields__", {}):
         v = getattr(e, f)
@@ -297,6 +317,8 @@ def try_fuse_exprs(tbl: Table, exprs: List[Expr]) -> Optional[List[Column]]:
         for idx, col in in_items:
             ct = _CTYPE_OF_KIND[col.dtype.kind]
             params.append(f"const {ct}* __restrict__ in{idx}")
+        for j, (ct, _) in enumerate(fuser.scalars):
+            params.append(f"{ct} c{j}")
         for j, (_, (cexpr, t)) in enumerate(emitted):
             params.append(f"{_OUT_CTYPE[t]}* __restrict__ out{j}")
         body_lines = "\n    ".join(fuser.lines)
@@ -339,7 +361,8 @@ extern "C" __global__ void fused(long long n, {', '.join(params)}) {{
                 for o in outs]
     from .hip_udf import _runtime
 
-    _runtime().launch_generic(fn, n, ptrs + out_ptrs,
+    _runtime().launch_generic(fn, n,
+                              ptrs + [v for _, v in fuser.scalars] + out_ptrs,
                               torch.cuda.current_stream().cuda_stream)
     result: List[Optional[Column]] = [None] * len(exprs)
     for (i, e), o, t in zip(compute, outs, out_types):

@@ -260,9 +260,12 @@ class _HiprtcRuntime:
 
     def launch_generic(self, fn, n: int, ptr_args, stream: int,
                        block: int = 256):
-        """Launch kernel(long long n, <pointers...>)."""
-        holders = [ctypes.c_longlong(n)] + [ctypes.c_void_p(p)
-                                            for p in ptr_args]
+        """Launch kernel(long long n, <args...>).  Entries of ptr_args that
+        are already ctypes scalars (c_double/c_longlong/...) are passed by
+        value; plain ints are device pointers."""
+        holders = [ctypes.c_longlong(n)] + [
+            p if isinstance(p, ctypes._SimpleCData) else ctypes.c_void_p(p)
+            for p in ptr_args]
         arr = (ctypes.c_void_p * len(holders))(*[
             ctypes.cast(ctypes.byref(h), ctypes.c_void_p) for h in holders])
         grid = min((n + block - 1) // block, 2048) or 1
