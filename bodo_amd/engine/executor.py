@@ -355,6 +355,20 @@ def _exec_shuffle_by_key(node: pn.ShuffleByKey, ctx) -> Table:
     return _shuffle_by_keys(child, list(node.keys), ctx)
 
 
+def _exec_rowid(node: pn.RowId, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    n = len(child)
+    base = 0
+    if ctx.world > 1:
+        lengths = comm.allgather_obj(n)
+        base = sum(lengths[:ctx.rank])
+    rid = torch.arange(base, base + n, dtype=torch.int64, device=child.device)
+    from ..core import types as bt
+    from ..core.column import Column
+
+    return child.with_column(node.name, Column(bt.int64, rid))
+
+
 def _exec_map_partitions(node: pn.MapPartitions, ctx) -> Table:
     child = _exec(node.child, ctx)
     pdf = child.to_pandas()
@@ -490,6 +504,7 @@ _HANDLERS = {
     pn.Sample: _exec_sample,
     pn.MapPartitions: _exec_map_partitions,
     pn.ShuffleByKey: _exec_shuffle_by_key,
+    pn.RowId: _exec_rowid,
     pn.Window: _exec_window,
     pn.Join: _exec_join,
     pn.Union: _exec_union,

@@ -210,6 +210,9 @@ def prune_columns(node: pn.PlanNode, required: Optional[Set[str]]) -> pn.PlanNod
             node.left_on, node.right_on, node.how, node.suffixes)
     if isinstance(node, pn.Union):
         return node.with_children(*[prune_columns(c, required) for c in node.inputs])
+    if isinstance(node, pn.RowId):
+        req2 = None if required is None else set(required) - {node.name}
+        return node.with_children(prune_columns(node.child, req2))
     # unknown shape: don't prune below
     ch = node.children()
     if not ch:

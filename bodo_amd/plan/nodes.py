@@ -201,6 +201,27 @@ class MapPartitions(PlanNode):
 
 
 @dataclass(frozen=True)
+class RowId(PlanNode):
+    """Append a globally-unique int64 row-id column (rank-offset arange).
+    Used by the SQL planner's row-id decorrelation of EXISTS subqueries with
+    non-equality correlated predicates (reference analog: Calcite's
+    RelDecorrelator as used by BodoSQL)."""
+
+    child: PlanNode
+    name: str = "__rid"
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return RowId(ch[0], self.name)
+
+    def out_columns(self):
+        cols = self.child.out_columns()
+        return None if cols is None else list(cols) + [self.name]
+
+
+@dataclass(frozen=True)
 class ShuffleByKey(PlanNode):
     """Explicit hash-repartition so equal keys co-locate on one rank
     (used by groupby.apply / median paths)."""

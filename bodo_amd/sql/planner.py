@@ -243,9 +243,23 @@ class Planner:
             cols = list(frame._columns)
         internal = []
         need_rename = False
+
+        def _taken(name):
+            # collision check climbs to enclosing scopes so a subquery's
+            # columns get distinct internal names from the outer query's
+            # (needed by row-id decorrelation, where both sides of the
+            # decorrelating join can come from the SAME base table)
+            s = scope
+            while s is not None:
+                for e in s.entries:
+                    if e[1].lower() == name.lower() or \
+                            e[2].lower() == name.lower():
+                        return True
+                s = s.parent
+            return False
+
         for c in cols:
-            iname = f"{alias}_{c}" if any(
-                e[1].lower() == c.lower() for e in scope.entries) else c
+            iname = f"{alias}_{c}" if _taken(c) else c
             if iname != c:
                 need_rename = True
             internal.append(iname)
@@ -340,6 +354,7 @@ class Planner:
                 plan = pn.Join(plan, sub_plan, (), (), "cross")
             joined |= sub_cols
         corr_pairs = []
+        corr_extra = []  # correlated conjuncts that are NOT equality pairs
         inner_keep = []
         for ci, cj in enumerate(conjs):
             if used[ci]:
@@ -347,6 +362,9 @@ class Planner:
             pair = self._corr_pair(cj, scope)
             if pair is not None:
                 corr_pairs.append(pair)
+                continue
+            if self._refs_outer(cj, scope):
+                corr_extra.append(cj)
                 continue
             handled, plan = self._apply_subquery_pred(cj, plan, scope)
             if not handled:
@@ -356,7 +374,24 @@ class Planner:
             for cj in inner_keep[1:]:
                 cond = ast.Bin("and", cond, cj)
             plan = pn.Filter(plan, self.expr(cond, scope))
-        return plan, scope, corr_pairs
+        return plan, scope, corr_pairs, corr_extra
+
+    def _refs_outer(self, node, scope: Scope) -> bool:
+        """True when the AST references any enclosing-scope column."""
+        if isinstance(node, ast.Col):
+            return scope.is_outer(node.table, node.name)
+        if isinstance(node, ast.Query):
+            # nested subqueries resolve through their own child Scope; their
+            # columns are not this scope's correlation concern
+            return False
+        for f in getattr(node, "__dataclass_fields__", {}):
+            v = getattr(node, f)
+            items = v if isinstance(v, (tuple, list)) else (v,)
+            for x in items:
+                if hasattr(x, "__dataclass_fields__") and \
+                        self._refs_outer(x, scope):
+                    return True
+        return False
 
     def _corr_pair(self, cj, sub_scope: Scope):
         """inner_col = outer_col -> (outer_internal, inner_internal)."""
@@ -398,7 +433,9 @@ class Planner:
             dis = pn.Distinct(proj, renamed)
             return pn.Join(plan, dis, (op_expr.name,), renamed,
                            "anti" if negated else "semi")
-        sub_plan, sub_scope, corr = self._plan_inner(q, scope)
+        sub_plan, sub_scope, corr, corr_extra = self._plan_inner(q, scope)
+        if corr_extra:
+            raise NotImplementedError("non-equality correlated IN subquery")
         item = q.items[0]
         val_internal = self._subq_value_col(item, sub_scope, sub_plan)
         sub_plan = val_internal[1]
@@ -417,18 +454,50 @@ class Planner:
         return pn.Join(plan, dis, lks, rks, "anti" if negated else "semi")
 
     def _exists_subquery(self, q: ast.Query, negated, plan, scope):
-        sub_plan, sub_scope, corr = self._plan_inner(q, scope)
-        if not corr:
+        sub_plan, sub_scope, corr, corr_extra = self._plan_inner(q, scope)
+        if not corr and not corr_extra:
             raise NotImplementedError("uncorrelated EXISTS")
-        pfx = self._uniq("__sq")
-        keep = [i for _, i in corr]
-        renamed = [f"{pfx}_{c}" for c in keep]
-        proj = pn.Projection(sub_plan, tuple(renamed),
-                             tuple(ex.ColRef(c) for c in keep))
-        dis = pn.Distinct(proj, tuple(renamed))
+        if not corr_extra:
+            pfx = self._uniq("__sq")
+            keep = [i for _, i in corr]
+            renamed = [f"{pfx}_{c}" for c in keep]
+            proj = pn.Projection(sub_plan, tuple(renamed),
+                                 tuple(ex.ColRef(c) for c in keep))
+            dis = pn.Distinct(proj, tuple(renamed))
+            lks = tuple(o for o, _ in corr)
+            return pn.Join(plan, dis, lks, tuple(renamed),
+                           "anti" if negated else "semi")
+        # row-id decorrelation (TPC-H q21 shape: EXISTS with an extra
+        # non-equality correlated predicate, e.g. l2.l_suppkey <> l1.l_suppkey;
+        # reference analog: Calcite RelDecorrelator used by BodoSQL):
+        # tag the outer rows with a global row id, inner-join outer x inner on
+        # the equality correlations, filter by the non-equality predicates
+        # (both sides' columns are in scope post-join, with distinct internal
+        # names), then semi/anti join the tagged outer plan on the surviving
+        # row ids.
+        if not corr:
+            raise NotImplementedError(
+                "EXISTS with only non-equality correlation (cross-product "
+                "decorrelation not supported)")
+        rid = self._uniq("__rid")
+        tagged = pn.RowId(plan, rid)
         lks = tuple(o for o, _ in corr)
-        return pn.Join(plan, dis, lks, tuple(renamed),
-                       "anti" if negated else "semi")
+        rks = tuple(i for _, i in corr)
+        jn = pn.Join(tagged, sub_plan, lks, rks, "inner")
+        cond = self.expr(corr_extra[0], sub_scope)
+        for cj in corr_extra[1:]:
+            cond = ex.BoolOp("and", cond, self.expr(cj, sub_scope))
+        flt = pn.Filter(jn, cond)
+        proj = pn.Projection(flt, (rid,), (ex.ColRef(rid),))
+        dis = pn.Distinct(proj, (rid,))
+        out = pn.Join(tagged, dis, (rid,), (rid,), "anti" if negated else "semi")
+        # drop the row-id column so downstream output shape is unchanged
+        cols = plan.out_columns()
+        if cols is None:
+            return out
+        keep_cols = tuple(cols)
+        return pn.Projection(out, keep_cols,
+                             tuple(ex.ColRef(c) for c in keep_cols))
 
     def _scalar_cmp_subquery(self, op, outer_expr, q: ast.Query, plan, scope):
         full = self._try_plan_full(q)
@@ -437,7 +506,10 @@ class Planner:
             return pn.Filter(plan, ex.Cmp(
                 op, self.expr(outer_expr, scope),
                 ex.ScalarSubquery(sub_plan, out_names[0])))
-        sub_plan, sub_scope, corr = self._plan_inner(q, scope)
+        sub_plan, sub_scope, corr, corr_extra = self._plan_inner(q, scope)
+        if corr_extra:
+            raise NotImplementedError(
+                "non-equality correlated scalar subquery")
         item = q.items[0]
         agg_ast, rebuild = _extract_single_agg(item.expr)
         if agg_ast is None:
