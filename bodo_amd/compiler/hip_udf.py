@@ -220,6 +220,39 @@ class _HiprtcRuntime:
         self._modules = {}
 
     def compile(self, source: str) -> bytes:
+        """hipRTC compile with a durable on-disk code cache keyed by
+        sha256(source)+arch (reference: @bodo.jit(cache=True) durable
+        compiles, bodo/tests/caching_tests/)."""
+        cache_path = None
+        from .. import config as cfg
+
+        if cfg.KERNEL_CACHE_DIR:
+            import hashlib
+            import os
+
+            key = hashlib.sha256(b"gfx950|" + source.encode()).hexdigest()
+            cache_path = os.path.join(cfg.KERNEL_CACHE_DIR, key + ".hsaco")
+            try:
+                with open(cache_path, "rb") as f:
+                    return f.read()
+            except OSError:
+                pass
+        code = self._compile_nocache(source)
+        if cache_path is not None:
+            try:
+                import os
+                import tempfile
+
+                os.makedirs(cfg.KERNEL_CACHE_DIR, exist_ok=True)
+                fd, tmp = tempfile.mkstemp(dir=cfg.KERNEL_CACHE_DIR)
+                with os.fdopen(fd, "wb") as f:
+                    f.write(code)
+                os.replace(tmp, cache_path)  # atomic under concurrent ranks
+            except OSError:
+                pass
+        return code
+
+    def _compile_nocache(self, source: str) -> bytes:
         rtc = self.rtc
         prog = ctypes.c_void_p()
         r = rtc.hiprtcCreateProgram(ctypes.byref(prog),

@@ -57,6 +57,13 @@ FUSE_EXPR = _env_bool("BODO_AMD_FUSE_EXPR", True)
 STREAM_THRESHOLD_BYTES = _env_int("BODO_AMD_STREAM_THRESHOLD",
                                   100 * 1024**3)
 
+#: on-disk cache for hipRTC-compiled UDF/fused-expression code objects
+#: (the reference's @bodo.jit(cache=True) durable-compile analog,
+#: bodo/tests/caching_tests); "" disables
+KERNEL_CACHE_DIR = os.environ.get(
+    "BODO_AMD_KERNEL_CACHE",
+    os.path.join(os.path.expanduser("~"), ".cache", "bodo_amd_kernels"))
+
 
 def default_device() -> str:
     if DEVICE:
