@@ -440,6 +440,14 @@ def _exec_parquet_write(node: pn.ParquetWrite, ctx) -> Table:
     return Table([], [], 0)
 
 
+def _exec_iceberg_write(node: pn.IcebergWrite, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    from ..io import iceberg
+
+    iceberg.write_iceberg(child, node.path, node.mode, ctx)
+    return Table([], [], 0)
+
+
 def _exec_reduce(node: pn.Reduce, ctx) -> Table:
     from .streaming import exec_streaming, want_streaming
 
@@ -509,5 +517,6 @@ _HANDLERS = {
     pn.Join: _exec_join,
     pn.Union: _exec_union,
     pn.ParquetWrite: _exec_parquet_write,
+    pn.IcebergWrite: _exec_iceberg_write,
     pn.Reduce: _exec_reduce,
 }

@@ -26,6 +26,11 @@ def schema_names(path: str) -> Tuple[str, ...]:
 
 
 def _dataset(path: str) -> pads.Dataset:
+    from . import iceberg
+
+    files = iceberg.resolve_scan_path(path)
+    if files is not None:
+        return pads.dataset(files, format="parquet")
     return pads.dataset(path, format="parquet")
 
 
@@ -64,7 +69,10 @@ def _expr_to_arrow(e: Expr):
 
 def read_shard(path: str, columns: Optional[Sequence[str]],
                filters: Sequence[Expr], ctx) -> Table:
-    if ctx.device.type == "cuda" and not filters:
+    from . import iceberg
+
+    if ctx.device.type == "cuda" and not filters and \
+            iceberg.resolve_scan_path(path) is None:
         # on-GPU decode fast path (uncompressed PLAIN / RLE_DICTIONARY)
         from . import parquet_gpu
 

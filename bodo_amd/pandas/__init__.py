@@ -42,6 +42,21 @@ def read_parquet(path, columns=None, **kwargs) -> BodoDataFrame:
     return BodoDataFrame(plan, list(columns) if columns else list(names))
 
 
+def read_iceberg(table_uri, snapshot_id=None, columns=None,
+                 **kwargs) -> BodoDataFrame:
+    """Read a filesystem Iceberg table, optionally time-travelling to an
+    older snapshot (reference: bodo/pandas/base.py read_iceberg:313)."""
+    from ..io import iceberg as _ib
+
+    path = str(table_uri)
+    if snapshot_id is not None:
+        path = f"{path}@{_ib.SNAP_PREFIX}{int(snapshot_id)}"
+    names = _pq.schema_names(path)
+    plan = _pn.ParquetScan(path, tuple(columns) if columns else None,
+                           (), tuple(names))
+    return BodoDataFrame(plan, list(columns) if columns else list(names))
+
+
 def read_csv(path, **kwargs) -> BodoDataFrame:
     options = tuple(sorted(kwargs.items(), key=lambda kv: kv[0]))
     names = _csv.schema_names(str(path), dict(options))

@@ -446,6 +446,15 @@ class BodoDataFrame:
         plan = pn.ParquetWrite(self._plan, path, compression)
         api.materialize(plan)
 
+    def to_iceberg(self, path: str, mode: str = "create", **kwargs):
+        """Transactional snapshot write to a filesystem Iceberg table
+        (reference: frame.py to_iceberg; metadata is committed only after
+        every rank's data files land)."""
+        from ..engine import api
+
+        plan = pn.IcebergWrite(self._plan, path, mode)
+        api.materialize(plan)
+
     def to_csv(self, path=None, **kwargs):
         pdf = self.to_pandas()
         return pdf.to_csv(path, index=False, **kwargs)

@@ -333,6 +333,22 @@ class ParquetWrite(PlanNode):
 
 
 @dataclass(frozen=True)
+class IcebergWrite(PlanNode):
+    """Transactional snapshot write to a filesystem Iceberg table
+    (reference: frame.py to_iceberg, bodo/io/iceberg/write.py)."""
+
+    child: PlanNode
+    path: str = ""
+    mode: str = "create"  # create | replace | append
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return IcebergWrite(ch[0], self.path, self.mode)
+
+
+@dataclass(frozen=True)
 class Reduce(PlanNode):
     """Whole-column reductions producing a scalar row (Series.sum() etc.)."""
 
