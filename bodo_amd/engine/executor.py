@@ -496,6 +496,11 @@ def _combine_reduce(partials: List[dict], func: str):
         return any(p["any"] for p in partials)
     if func == "all":
         return all(p["all"] for p in partials)
+    if func == "prod":
+        out = 1.0
+        for p in partials:
+            out *= p["prod"]
+        return out
     raise NotImplementedError(func)
 
 

@@ -414,4 +414,6 @@ def reduce_column(col: Column, func: str):
         return {"any": bool(data.any().item()) if n else False}
     if func == "all":
         return {"all": bool(data.all().item()) if n else True}
+    if func == "prod":
+        return {"prod": data.to(torch.float64).prod().item() if n else 1.0}
     raise NotImplementedError(f"reduce {func}")

@@ -201,7 +201,7 @@ class _Evaluator:
 
     def visit_StrOp(self, e: StrOp) -> Column:
         a = self.visit(e.operand)
-        return str_op(a, e.op, e.args)
+        return str_op(a, e.op, e.args, dict(e.kwargs))
 
     def visit_UdfMap(self, e: UdfMap) -> Column:
         a = self.visit(e.operand)
@@ -502,6 +502,10 @@ _DT_OUT_TYPE = {
     "minute": bt.int8, "second": bt.int8, "dayofweek": bt.int8,
     "weekday": bt.int8, "dayofyear": bt.int16, "quarter": bt.int8,
     "date": bt.date32, "normalize": bt.timestamp_ns, "floor_day": bt.timestamp_ns,
+    "is_month_start": bt.boolean, "is_month_end": bt.boolean,
+    "is_quarter_start": bt.boolean, "is_quarter_end": bt.boolean,
+    "is_year_start": bt.boolean, "is_year_end": bt.boolean,
+    "days_in_month": bt.int8, "daysinmonth": bt.int8,
 }
 
 NS_PER_DAY = 86400 * 10**9
@@ -511,7 +515,12 @@ def dt_field(a: Column, fld: str) -> Column:
     if a.is_cuda:
         from . import gpu
 
-        return gpu.dt_field(a, fld)
+        try:
+            return gpu.dt_field(a, fld)
+        except (NotImplementedError, KeyError, AssertionError):
+            # fields without a device kernel yet: host round-trip
+            res = _dt_field_cpu(a.to_device("cpu"), fld)
+            return res.to_device(a.device)
     return _dt_field_cpu(a, fld)
 
 
@@ -531,7 +540,8 @@ def _dt_field_cpu(a: Column, fld: str) -> Column:
         dtype = bt.timestamp_ns
     else:
         attr = "dayofweek" if fld == "weekday" else fld
-        out = getattr(idx, attr).to_numpy()
+        raw = getattr(idx, attr)
+        out = raw.to_numpy() if hasattr(raw, "to_numpy") else np.asarray(raw)
         dtype = _DT_OUT_TYPE[fld]
         out = out.astype(bt.numpy_storage_dtype(dtype))
     res = Column(dtype, torch.from_numpy(out), a.mask)
@@ -565,7 +575,42 @@ def cast_column(a: Column, to: DType) -> Column:
 # strings
 # ----------------------------------------------------------------------
 
-def str_op(a: Column, op: str, args) -> Column:
+def _str_dict_generic(a: Column, op: str, args, kwargs) -> Column:
+    """Apply any pandas .str method to a DICT column's (small) dictionary:
+    string results re-encode as a dictionary, scalar results become a LUT
+    gather over the codes."""
+    import pyarrow as pa
+
+    vals = pd.Series(a.dictionary.to_pylist(), dtype="object")
+    res = getattr(vals.str, op)(*args, **(kwargs or {}))
+    if res.dtype == object or isinstance(res.dtype, pd.StringDtype):
+        out_vals = res.tolist()
+        if any(isinstance(v, (list, tuple)) for v in out_vals):
+            raise NotImplementedError(f"str.{op} returns lists")
+        uniq = list(dict.fromkeys(out_vals))
+        if len(uniq) != len(out_vals):
+            code_of = {v: i for i, v in enumerate(uniq)}
+            remap = torch.tensor([code_of[v] for v in out_vals],
+                                 dtype=torch.int32, device=a.device)
+            codes = remap[a.data.long()]
+            return Column(a.dtype, codes, a.mask,
+                          dictionary=pa.array(uniq, type=pa.large_string()),
+                          length=len(a))
+        return Column(a.dtype, a.data, a.mask,
+                      dictionary=pa.array(out_vals, type=pa.large_string()),
+                      length=len(a))
+    if res.dtype == bool:
+        lut = torch.from_numpy(res.to_numpy()).to(a.device)
+        return Column(bt.boolean, lut[a.data.long()], a.mask)
+    arr = res.to_numpy()
+    if np.issubdtype(arr.dtype, np.integer):
+        lut = torch.from_numpy(arr.astype(np.int64)).to(a.device)
+        return Column(bt.int64, lut[a.data.long()], a.mask)
+    lut = torch.from_numpy(arr.astype(np.float64)).to(a.device)
+    return Column(bt.float64, lut[a.data.long()], a.mask)
+
+
+def str_op(a: Column, op: str, args, kwargs=None) -> Column:
     import pyarrow.compute as pc
 
     if a.dtype.kind == TypeKind.DICT:
@@ -613,7 +658,7 @@ def str_op(a: Column, op: str, args) -> Column:
             lut = torch.from_numpy(lut_np).to(a.device)
             out = lut[a.data.long()]
             return Column(ret, out, a.mask)
-        raise NotImplementedError(f"str.{op} on dict column")
+        return _str_dict_generic(a, op, args, kwargs)
     # plain strings: arrow compute on host
     arr = a.to_arrow() if not a.is_cuda else a.to_device("cpu").to_arrow()
     fmap = {
@@ -630,6 +675,15 @@ def str_op(a: Column, op: str, args) -> Column:
         "capitalize": lambda x: pc.utf8_capitalize(x),
         "slice": lambda x: pc.utf8_slice_codeunits(x, *args),
     }
+    if op not in fmap:
+        # generic pandas .str on host (replace/zfill/pad/isdigit/...)
+        import pyarrow as pa
+
+        ser = arr.to_pandas().astype("object")
+        res_s = getattr(ser.str, op)(*args, **(kwargs or {}))
+        if any(isinstance(v, (list, tuple)) for v in res_s.head(64).tolist()):
+            raise NotImplementedError(f"str.{op} returns lists")
+        return Column.from_arrow(pa.Array.from_pandas(res_s), a.device)
     res = fmap[op](arr)
     if res.type in (__import__("pyarrow").int32(), __import__("pyarrow").int64()):
         res = res.cast(__import__("pyarrow").int64())

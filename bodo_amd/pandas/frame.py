@@ -437,6 +437,76 @@ class BodoDataFrame:
                 out = BoolOp("or", out, c)
         return BodoDataFrame(pn.Filter(self._plan, out), self._columns)
 
+    def astype(self, dtype) -> "BodoDataFrame":
+        from .series import _pd_dtype_to_bodo
+        from ..plan.expr import Cast
+
+        per_col = dtype if isinstance(dtype, dict) else \
+            {c: dtype for c in self._columns}
+        exprs = tuple(
+            Cast(ColRef(c), _pd_dtype_to_bodo(per_col[c]))
+            if c in per_col else ColRef(c) for c in self._columns)
+        return BodoDataFrame(
+            pn.Projection(self._plan, tuple(self._columns), exprs),
+            list(self._columns))
+
+    def fillna(self, value) -> "BodoDataFrame":
+        from ..plan.expr import Case, IsNull, as_expr
+
+        per_col = value if isinstance(value, dict) else \
+            {c: value for c in self._columns}
+        exprs = tuple(
+            Case((IsNull(ColRef(c)),), (as_expr(per_col[c]),), ColRef(c))
+            if c in per_col else ColRef(c) for c in self._columns)
+        return BodoDataFrame(
+            pn.Projection(self._plan, tuple(self._columns), exprs),
+            list(self._columns))
+
+    def nunique(self):
+        return pd.Series({c: self[c].nunique() for c in self._columns})
+
+    def describe(self):
+        from ..engine import api
+
+        pdf_head = api.collect(pn.Limit(self._plan, 1))
+        num = [c for c in self._columns
+               if pd.api.types.is_numeric_dtype(pdf_head[c].dtype)]
+        return pd.DataFrame({c: self[c].describe() for c in num})
+
+    def query(self, expr: str, **kwargs) -> "BodoDataFrame":
+        """pandas query-string filter: the expression is rewritten to the
+        operator form (and->&, or->|, not->~) and evaluated lazily against
+        this frame's columns (reference: frame.py query)."""
+        import ast as _ast
+
+        tree = _ast.parse(expr, mode="eval")
+
+        class _Rw(_ast.NodeTransformer):
+            def visit_BoolOp(self, node):
+                self.generic_visit(node)
+                op = _ast.BitAnd() if isinstance(node.op, _ast.And) \
+                    else _ast.BitOr()
+                out = node.values[0]
+                for v in node.values[1:]:
+                    out = _ast.BinOp(left=out, op=op, right=v)
+                return out
+
+            def visit_UnaryOp(self, node):
+                self.generic_visit(node)
+                if isinstance(node.op, _ast.Not):
+                    return _ast.UnaryOp(op=_ast.Invert(), operand=node.operand)
+                return node
+
+        new = _ast.fix_missing_locations(_Rw().visit(tree))
+        ns = {c: self[c] for c in self._columns if c.isidentifier()}
+        ns.update(kwargs.get("local_dict") or {})
+        mask = eval(compile(new, "<query>", "eval"), {"__builtins__": {}}, ns)
+        return self[mask]
+
+    @property
+    def iloc(self):
+        return _ILoc(self)
+
     # ------------------------------------------------------------------
     # IO
     # ------------------------------------------------------------------
@@ -498,6 +568,34 @@ def from_pandas_df(df: pd.DataFrame) -> BodoDataFrame:
     key = ex.register_object(df.reset_index(drop=True))
     plan = pn.PandasScan(key, tuple(df.columns), distributed=False)
     return BodoDataFrame(plan, list(df.columns))
+
+
+class _ILoc:
+    """df.iloc[:n] / df.iloc[:n, :] head-style slicing stays lazy (Limit
+    plan); anything else materializes through the pandas fallback."""
+
+    def __init__(self, frame: BodoDataFrame):
+        self._frame = frame
+
+    def __getitem__(self, key):
+        rows = key[0] if isinstance(key, tuple) else key
+        if isinstance(rows, slice) and rows.start in (None, 0) and \
+                rows.step in (None, 1) and rows.stop is not None and \
+                rows.stop >= 0:
+            out = BodoDataFrame(pn.Limit(self._frame._plan, rows.stop),
+                                list(self._frame._columns))
+            if isinstance(key, tuple) and len(key) == 2:
+                cols = key[1]
+                if isinstance(cols, slice) and cols == slice(None):
+                    return out
+                if isinstance(cols, list):
+                    names = [self._frame._columns[i] for i in cols] \
+                        if all(isinstance(i, int) for i in cols) else cols
+                    return out[names]
+                if isinstance(cols, int):
+                    return out[self._frame._columns[cols]]
+            return out
+        return self._frame.to_pandas().iloc[key]
 
 
 class _LocIndexer:

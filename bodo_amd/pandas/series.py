@@ -199,6 +199,87 @@ class BodoSeries:
         out.index.name = self.name
         return out
 
+    def prod(self):
+        return self._reduce("prod")
+
+    product = prod
+
+    def quantile(self, q=0.5, interpolation="linear"):
+        """Exact quantile: gathers the projected column (distributed shards
+        allgather) and interpolates with numpy.  For terabyte-scale inputs
+        prefer .sample(...)-based estimates; groupby('...').median() stays
+        fully distributed."""
+        ser = self.dropna().to_pandas()
+        if hasattr(q, "__len__"):
+            return pd.Series([ser.quantile(x, interpolation=interpolation)
+                              for x in q], index=list(q), name=self.name)
+        return ser.quantile(q, interpolation=interpolation)
+
+    def median(self):
+        return self.quantile(0.5)
+
+    def describe(self):
+        cnt = self.count()
+        qs = self.quantile([0.25, 0.5, 0.75])
+        vals = [cnt, self.mean(), self.std(), self.min(),
+                qs.iloc[0], qs.iloc[1], qs.iloc[2], self.max()]
+        return pd.Series(vals, index=["count", "mean", "std", "min", "25%",
+                                      "50%", "75%", "max"], name=self.name)
+
+    def mode(self):
+        vc = self.value_counts(ascending=False)
+        if not len(vc):
+            return pd.Series([], dtype="object", name=self.name)
+        top = vc[vc == vc.max()].index.to_numpy()
+        return pd.Series(sorted(top), name=self.name)
+
+    def between(self, left, right, inclusive="both"):
+        lo = Cmp("ge" if inclusive in ("both", "left") else "gt",
+                 self._expr, as_expr(left))
+        hi = Cmp("le" if inclusive in ("both", "right") else "lt",
+                 self._expr, as_expr(right))
+        return self._wrap(BoolOp("and", lo, hi), None)
+
+    def clip(self, lower=None, upper=None):
+        e = self._expr
+        conds, thens = [], []
+        if lower is not None:
+            conds.append(Cmp("lt", e, as_expr(lower)))
+            thens.append(as_expr(lower))
+        if upper is not None:
+            conds.append(Cmp("gt", e, as_expr(upper)))
+            thens.append(as_expr(upper))
+        if not conds:
+            return self
+        return self._wrap(Case(tuple(conds), tuple(thens), e), None)
+
+    def dropna(self) -> "BodoSeries":
+        filt = pn.Filter(self._as_projection_plan(),
+                         IsNull(ColRef("v"), negate=True))
+        return BodoSeries(filt, ColRef("v"), self.name)
+
+    def sort_values(self, ascending=True) -> "BodoSeries":
+        plan = pn.Sort(self._as_projection_plan(), ("v",), (bool(ascending),))
+        return BodoSeries(plan, ColRef("v"), self.name)
+
+    def nlargest(self, n=5) -> pd.Series:
+        from ..engine import api
+
+        plan = pn.Limit(pn.Sort(self._as_projection_plan(), ("v",), (False,)),
+                        n)
+        ser = api.collect(plan)["v"]
+        ser.name = self.name
+        return ser
+
+    def nsmallest(self, n=5) -> pd.Series:
+        from ..engine import api
+
+        plan = pn.Limit(pn.Sort(self._as_projection_plan(), ("v",), (True,)),
+                        n)
+        ser = api.collect(plan)["v"]
+        ser.name = self.name
+        return ser
+
     def _as_projection_plan(self):
         return pn.Projection(self._plan, ("v",), (self._expr,))
 
@@ -299,6 +380,52 @@ class _DtAccessor:
         assert freq in ("D", "d"), "only day floor supported"
         return self._f("floor_day")
 
+    @property
+    def is_month_start(self):
+        return self._f("is_month_start")
+
+    @property
+    def is_month_end(self):
+        return self._f("is_month_end")
+
+    @property
+    def is_quarter_start(self):
+        return self._f("is_quarter_start")
+
+    @property
+    def is_quarter_end(self):
+        return self._f("is_quarter_end")
+
+    @property
+    def is_year_start(self):
+        return self._f("is_year_start")
+
+    @property
+    def is_year_end(self):
+        return self._f("is_year_end")
+
+    @property
+    def days_in_month(self):
+        return self._f("days_in_month")
+
+    daysinmonth = days_in_month
+
+    def month_name(self, locale=None):
+        # 12-value LUT over the month field (dense-domain UdfMap builds a
+        # dictionary column, so this stays device-resident)
+        import calendar
+
+        return self._s._wrap(
+            UdfMap(DtField(self._s._expr, "month"),
+                   lambda m: calendar.month_name[int(m)], None), None)
+
+    def day_name(self, locale=None):
+        import calendar
+
+        return self._s._wrap(
+            UdfMap(DtField(self._s._expr, "dayofweek"),
+                   lambda d: calendar.day_name[int(d)], None), None)
+
 
 class _StrAccessor:
     def __init__(self, s: BodoSeries):
@@ -341,6 +468,32 @@ class _StrAccessor:
 
     def slice(self, start=None, stop=None, step=None):
         return self._f("slice", start or 0, stop, step or 1)
+
+    def replace(self, pat, repl, regex=True, **kw):
+        return self._s._wrap(
+            StrOp(self._s._expr, "replace", (pat, repl),
+                  (("regex", bool(regex)),)), None)
+
+    # methods whose pandas result is a list/frame per row (need explode
+    # semantics) fall back via the Series-level pandas fallback
+    _UNSUPPORTED = {"split", "rsplit", "extract", "extractall", "findall",
+                    "get_dummies", "partition", "rpartition", "cat"}
+
+    def __getattr__(self, op):
+        """Any other pandas .str method lowers to a StrOp evaluated on the
+        dictionary (DICT columns: one host call over the unique values) or
+        the host string array (reference: BodoSeries str accessor,
+        bodo/pandas/series.py)."""
+        if op.startswith("_") or op in self._UNSUPPORTED or \
+                not hasattr(pd.Series([], dtype="object").str, op):
+            raise AttributeError(op)
+
+        def method(*args, **kwargs):
+            return self._s._wrap(
+                StrOp(self._s._expr, op, tuple(args),
+                      tuple(sorted(kwargs.items()))), None)
+
+        return method
 
 
 def _pd_dtype_to_bodo(dtype):

@@ -190,11 +190,13 @@ class UdfMap(Expr):
 
 @dataclass(frozen=True)
 class StrOp(Expr):
-    """String method: op in lower/upper/len/strip/contains/startswith/endswith."""
+    """String method (pandas .str surface); kwargs as a tuple of pairs so
+    the node stays hashable."""
 
     operand: Expr
     op: str
     args: Tuple[Any, ...] = ()
+    kwargs: Tuple[Tuple[str, Any], ...] = ()
 
     def children(self):
         return (self.operand,)

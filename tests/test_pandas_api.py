@@ -1,0 +1,115 @@
+"""Differential tests for the extended pandas API surface (reference:
+bodo/pandas frame.py/series.py method inventory)."""
+
+import numpy as np
+import pandas as pd
+import pytest
+
+import bodo_amd.pandas as bpd
+
+
+def _decat(obj):
+    if isinstance(obj, pd.DataFrame):
+        out = obj.copy()
+        for c in out.columns:
+            if isinstance(out[c].dtype, pd.CategoricalDtype):
+                out[c] = out[c].astype(object)
+        return out
+    if isinstance(obj.dtype, pd.CategoricalDtype):
+        return obj.astype(object)
+    return obj
+
+
+@pytest.fixture()
+def df():
+    rng = np.random.default_rng(0)
+    return pd.DataFrame({
+        "a": rng.integers(0, 10, 200),
+        "b": np.where(rng.random(200) < 0.1, np.nan, rng.random(200) * 100),
+        "c": rng.choice(["xx", "yy", "zz", "ww"], 200),
+        "t": pd.to_datetime("2020-01-01")
+        + pd.to_timedelta(rng.integers(0, 3 * 365, 200), unit="D"),
+    })
+
+
+def test_series_reductions(df):
+    b = bpd.from_pandas(df)
+    assert abs(b.b.median() - df.b.median()) < 1e-9
+    assert abs(b.b.quantile(0.3) - df.b.quantile(0.3)) < 1e-9
+    assert abs(b.a.prod() - float(df.a.astype("float64").prod())) < 1e-6
+    pd.testing.assert_series_equal(b.b.describe(), df.b.describe(),
+                                   check_dtype=False)
+
+
+def test_series_transforms(df):
+    b = bpd.from_pandas(df)
+    pd.testing.assert_series_equal(b.b.between(20, 60).to_pandas(),
+                                   df.b.between(20, 60), check_names=False,
+                                   check_dtype=False)
+    pd.testing.assert_series_equal(b.b.clip(10, 90).to_pandas(),
+                                   df.b.clip(10, 90), check_names=False,
+                                   check_dtype=False)
+    assert list(b.b.nlargest(5)) == list(df.b.nlargest(5))
+    assert list(b.b.nsmallest(5)) == list(df.b.nsmallest(5))
+    assert sorted(b.b.dropna().to_pandas()) == sorted(df.b.dropna())
+    assert list(b.c.mode()) == list(df.c.mode())
+
+
+def test_frame_astype_fillna(df):
+    b = bpd.from_pandas(df)
+    out = b.astype({"a": "float64"}).to_pandas()
+    assert out["a"].dtype == np.float64
+    f = b.fillna({"b": -1.0}).to_pandas()
+    exp = df.fillna({"b": -1.0})
+    pd.testing.assert_series_equal(f["b"], exp["b"], check_dtype=False)
+
+
+def test_frame_query_iloc_describe(df):
+    b = bpd.from_pandas(df)
+    got = _decat(b.query("a > 3 and b < 80").to_pandas())
+    exp = df.query("a > 3 and b < 80").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got.reset_index(drop=True), exp,
+                                  check_dtype=False)
+    pd.testing.assert_frame_equal(_decat(b.iloc[:7].to_pandas()), df.iloc[:7],
+                                  check_dtype=False)
+    pd.testing.assert_frame_equal(b.describe(), df[["a", "b"]].describe(),
+                                  check_dtype=False)
+    assert dict(b.nunique()) == dict(df.nunique())
+
+
+def test_str_generic(df):
+    b = bpd.from_pandas(df)
+    cases = [
+        ("replace", ("x", "Q"), {}),
+        ("zfill", (5,), {}),
+        ("isalpha", (), {}),
+        ("find", ("y",), {}),
+        ("rstrip", ("wz",), {}),
+        ("repeat", (2,), {}),
+    ]
+    for op, args, kw in cases:
+        got = _decat(getattr(b.c.str, op)(*args, **kw).to_pandas())
+        exp = getattr(df.c.str, op)(*args, **kw)
+        pd.testing.assert_series_equal(got, exp, check_names=False,
+                                       check_dtype=False)
+
+
+def test_dt_extended(df):
+    b = bpd.from_pandas(df)
+    for fld in ["is_month_start", "is_month_end", "is_quarter_start",
+                "is_year_start", "is_year_end", "days_in_month"]:
+        got = getattr(b.t.dt, fld).to_pandas().to_numpy()
+        exp = getattr(df.t.dt, fld).to_numpy()
+        assert (got == exp).all(), fld
+    got = _decat(b.t.dt.month_name().to_pandas()).astype(str)
+    assert (got.to_numpy() == df.t.dt.month_name().to_numpy()).all()
+    got = _decat(b.t.dt.day_name().to_pandas()).astype(str)
+    assert (got.to_numpy() == df.t.dt.day_name().to_numpy()).all()
+
+
+def test_series_sort_values(df):
+    b = bpd.from_pandas(df)
+    got = b.a.sort_values().to_pandas().to_numpy()
+    assert (got == np.sort(df.a.to_numpy())).all()
+    got = b.a.sort_values(ascending=False).to_pandas().to_numpy()
+    assert (got == np.sort(df.a.to_numpy())[::-1]).all()
