@@ -470,7 +470,8 @@ class BodoDataFrame:
 
         pdf_head = api.collect(pn.Limit(self._plan, 1))
         num = [c for c in self._columns
-               if pd.api.types.is_numeric_dtype(pdf_head[c].dtype)]
+               if pd.api.types.is_numeric_dtype(pdf_head[c].dtype)
+               and not pd.api.types.is_bool_dtype(pdf_head[c].dtype)]
         return pd.DataFrame({c: self[c].describe() for c in num})
 
     def query(self, expr: str, **kwargs) -> "BodoDataFrame":
