@@ -369,6 +369,37 @@ def _exec_rowid(node: pn.RowId, ctx) -> Table:
     return child.with_column(node.name, Column(bt.int64, rid))
 
 
+def _exec_rolling(node: pn.Rolling, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    w = int(node.window)
+    pdf = child.to_pandas()
+    halo = 0
+    if ctx.world > 1 and w > 1:
+        # halo exchange: each rank publishes its last (w-1) rows; a rank
+        # prepends rows from preceding ranks until it holds w-1 of them
+        # (walking further back when a predecessor shard is shorter)
+        tail = pdf.tail(w - 1)
+        tails = comm.allgather_obj(tail)
+        pieces, have, i = [], 0, ctx.rank - 1
+        while have < w - 1 and i >= 0:
+            t = tails[i]
+            take = t.tail(w - 1 - have)
+            pieces.insert(0, take)
+            have += len(take)
+            if len(t) >= w - 1:
+                break
+            i -= 1
+        if pieces:
+            halo = sum(len(p) for p in pieces)
+            pdf = pd.concat(pieces + [pdf], ignore_index=True)
+    out = {}
+    for out_name, in_name, func in node.specs:
+        r = pdf[in_name].rolling(w, min_periods=node.min_periods)
+        out[out_name] = getattr(r, func)()
+    res = pd.DataFrame(out).iloc[halo:].reset_index(drop=True)
+    return Table.from_pandas(res, ctx.device)
+
+
 def _exec_map_partitions(node: pn.MapPartitions, ctx) -> Table:
     child = _exec(node.child, ctx)
     pdf = child.to_pandas()
@@ -518,6 +549,7 @@ _HANDLERS = {
     pn.MapPartitions: _exec_map_partitions,
     pn.ShuffleByKey: _exec_shuffle_by_key,
     pn.RowId: _exec_rowid,
+    pn.Rolling: _exec_rolling,
     pn.Window: _exec_window,
     pn.Join: _exec_join,
     pn.Union: _exec_union,

@@ -508,6 +508,9 @@ class BodoDataFrame:
     def iloc(self):
         return _ILoc(self)
 
+    def rolling(self, window, min_periods=None, **kwargs):
+        return _RollingFrame(self, int(window), min_periods)
+
     # ------------------------------------------------------------------
     # IO
     # ------------------------------------------------------------------
@@ -569,6 +572,34 @@ def from_pandas_df(df: pd.DataFrame) -> BodoDataFrame:
     key = ex.register_object(df.reset_index(drop=True))
     plan = pn.PandasScan(key, tuple(df.columns), distributed=False)
     return BodoDataFrame(plan, list(df.columns))
+
+
+class _RollingFrame:
+    """df.rolling(w): per-column rolling aggregation (Rolling plan node with
+    distributed halo exchange; reference: hiframes rolling)."""
+
+    _FUNCS = ("sum", "mean", "min", "max", "count", "std", "var", "median")
+
+    def __init__(self, frame: BodoDataFrame, window: int, min_periods):
+        self._frame = frame
+        self._window = window
+        self._min_periods = min_periods
+
+    def _agg(self, func):
+        from ..engine import api
+
+        head = api.collect(pn.Limit(self._frame._plan, 1))
+        cols = [c for c in self._frame._columns
+                if pd.api.types.is_numeric_dtype(head[c].dtype)]
+        specs = tuple((c, c, func) for c in cols)
+        plan = pn.Rolling(self._frame._plan, self._window, self._min_periods,
+                          specs)
+        return BodoDataFrame(plan, list(cols))
+
+    def __getattr__(self, name):
+        if name in self._FUNCS:
+            return lambda: self._agg(name)
+        raise AttributeError(name)
 
 
 class _ILoc:

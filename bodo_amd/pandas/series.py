@@ -280,6 +280,9 @@ class BodoSeries:
         ser.name = self.name
         return ser
 
+    def rolling(self, window, min_periods=None, **kwargs):
+        return _RollingSeries(self, int(window), min_periods)
+
     def _as_projection_plan(self):
         return pn.Projection(self._plan, ("v",), (self._expr,))
 
@@ -511,3 +514,25 @@ def _pd_dtype_to_bodo(dtype):
     if nd.kind == "M":
         return bt.timestamp_ns
     return bt.from_numpy_dtype(nd)
+
+
+class _RollingSeries:
+    """s.rolling(w): rolling aggregation over the series (Rolling plan node
+    with distributed halo exchange)."""
+
+    _FUNCS = ("sum", "mean", "min", "max", "count", "std", "var", "median")
+
+    def __init__(self, s: BodoSeries, window: int, min_periods):
+        self._s = s
+        self._window = window
+        self._min_periods = min_periods
+
+    def _agg(self, func):
+        plan = pn.Rolling(self._s._as_projection_plan(), self._window,
+                          self._min_periods, (("v", "v", func),))
+        return BodoSeries(plan, ColRef("v"), self._s.name)
+
+    def __getattr__(self, name):
+        if name in self._FUNCS:
+            return lambda: self._agg(name)
+        raise AttributeError(name)

@@ -222,6 +222,28 @@ class RowId(PlanNode):
 
 
 @dataclass(frozen=True)
+class Rolling(PlanNode):
+    """Rolling-window aggregation over the global row order.  Distributed
+    execution exchanges a (window-1)-row halo from preceding ranks so shard
+    boundaries produce exactly the single-process result (reference:
+    hiframes rolling support in @bodo.jit)."""
+
+    child: PlanNode
+    window: int = 2
+    min_periods: Optional[int] = None
+    specs: Tuple[Tuple[str, str, str], ...] = ()  # (out, in, func)
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Rolling(ch[0], self.window, self.min_periods, self.specs)
+
+    def out_columns(self):
+        return [s[0] for s in self.specs]
+
+
+@dataclass(frozen=True)
 class ShuffleByKey(PlanNode):
     """Explicit hash-repartition so equal keys co-locate on one rank
     (used by groupby.apply / median paths)."""

@@ -234,3 +234,55 @@ def test_dist_window():
     exp["r"] = exp.groupby("a")["b"].rank(method="min")
     exp = exp[["a", "b", "gs", "cs", "r"]].reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_rolling(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).rolling(7).mean()
+
+
+def test_dist_rolling():
+    """Halo exchange must make shard boundaries invisible."""
+    df = _df(1000, 11)
+    got = run_dist(_q_rolling, {"df": df}).reset_index(drop=True)
+    exp = df[["a", "b"]].rolling(7).mean().reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_exists_noneq(bpd, rank, payload):
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": bpd.from_pandas(payload["df"])})
+    return bc.sql(
+        "select a, count(*) as n from t t1 "
+        "where exists (select * from t t2 "
+        "              where t2.a = t1.a and t2.b <> t1.b) "
+        "group by a order by a")
+
+
+def test_dist_exists_rowid_decorrelation():
+    """q21-shape EXISTS with non-equality correlation on 2 ranks (the RowId
+    plan node must produce globally unique ids across shards)."""
+    df = _df(800, 13)
+    got = run_dist(_q_exists_noneq, {"df": df}).reset_index(drop=True)
+    has_other = df.groupby("a")["b"].transform("nunique") > 1
+    keep = df[has_other]
+    exp = keep.groupby("a").size().reset_index(name="n").sort_values(
+        "a").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_iceberg(bpd, rank, payload):
+    b = bpd.from_pandas(payload["df"])
+    b.to_iceberg(payload["path"])
+    return bpd.read_iceberg(payload["path"])
+
+
+def test_dist_iceberg_roundtrip(tmp_path):
+    """2-rank transactional write then parallel read-back."""
+    df = _df(600, 17)
+    p = str(tmp_path / "ice_tbl")
+    got = run_dist(_q_iceberg, {"df": df, "path": p})
+    got = got.sort_values(["a", "b"]).reset_index(drop=True)
+    got["c"] = got["c"].astype(str)
+    exp = df.sort_values(["a", "b"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
