@@ -19,6 +19,7 @@ KEYWORDS = {
     "case", "when", "then", "else", "end", "cast", "extract", "join", "inner",
     "left", "right", "full", "outer", "cross", "on", "asc", "desc", "date",
     "interval", "union", "all", "exists", "true", "false", "substring", "for",
+    "intersect", "except", "over", "partition",
 }
 
 _TOKEN_RE = re.compile(r"""
@@ -159,6 +160,27 @@ class ExistsE:
 
 
 @dataclass
+class WindowE:
+    """func(args) OVER (PARTITION BY ... ORDER BY ...)."""
+
+    func: str
+    args: List[Any]
+    partition_by: List[Any]
+    order_by: List[Tuple[Any, bool]]
+    star: bool = False
+
+
+@dataclass
+class SetOpQ:
+    """UNION / INTERSECT / EXCEPT of two query expressions."""
+
+    op: str
+    all: bool
+    left: Any
+    right: Any
+
+
+@dataclass
 class TableRef:
     name: str
     alias: Optional[str]
@@ -231,11 +253,26 @@ class Parser:
             raise SyntaxError(f"expected {op!r} at {self.peek()}")
 
     # ------------------------------------------------------------- query
-    def parse(self) -> Query:
-        q = self.parse_select()
+    def parse(self):
+        q = self.parse_query_expr()
         if self.peek() is not None:
             raise SyntaxError(f"trailing tokens at {self.peek()}")
         return q
+
+    def parse_query_expr(self):
+        q = self.parse_select()
+        while True:
+            op = None
+            for kw in ("union", "intersect", "except"):
+                if self.accept_kw(kw):
+                    op = kw
+                    break
+            if op is None:
+                return q
+            all_ = bool(self.accept_kw("all"))
+            self.accept_kw("distinct")
+            r = self.parse_select()
+            q = SetOpQ(op, all_, q, r)
 
     def parse_select(self) -> Query:
         self.expect_kw("select")
@@ -309,7 +346,7 @@ class Parser:
 
     def parse_table_ref(self) -> TableRef:
         if self.accept_op("("):
-            q = self.parse_select()
+            q = self.parse_query_expr()
             self.expect_op(")")
             alias = None
             if self.accept_kw("as"):
@@ -524,15 +561,41 @@ class Parser:
             if nt and nt.kind == "op" and nt.value == "(":
                 self.i += 1
                 distinct = bool(self.accept_kw("distinct"))
+                star = False
+                args = []
                 if self.accept_op("*"):
                     self.expect_op(")")
-                    return Func(t.value.lower(), [], star=True)
-                args = []
-                if not self.accept_op(")"):
+                    star = True
+                elif not self.accept_op(")"):
                     args.append(self.parse_expr())
                     while self.accept_op(","):
                         args.append(self.parse_expr())
                     self.expect_op(")")
+                if self.accept_kw("over"):
+                    self.expect_op("(")
+                    part, order = [], []
+                    if self.accept_kw("partition"):
+                        self.expect_kw("by")
+                        part.append(self.parse_expr())
+                        while self.accept_op(","):
+                            part.append(self.parse_expr())
+                    if self.accept_kw("order"):
+                        self.expect_kw("by")
+                        while True:
+                            e = self.parse_expr()
+                            asc = True
+                            if self.accept_kw("desc"):
+                                asc = False
+                            else:
+                                self.accept_kw("asc")
+                            order.append((e, asc))
+                            if not self.accept_op(","):
+                                break
+                    self.expect_op(")")
+                    return WindowE(t.value.lower(), args, part, order,
+                                   star=star)
+                if star:
+                    return Func(t.value.lower(), [], star=True)
                 return Func(t.value.lower(), args, distinct=distinct)
             if nt and nt.kind == "op" and nt.value == ".":
                 self.i += 1

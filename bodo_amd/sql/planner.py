@@ -97,7 +97,9 @@ class Planner:
         return f"{base}__{self._counter}"
 
     # ------------------------------------------------------------------
-    def plan(self, q: ast.Query):
+    def plan(self, q):
+        if isinstance(q, ast.SetOpQ):
+            return self._plan_setop(q)
         scope = Scope()
         plan = None
         joined_cols: set = set()
@@ -197,6 +199,8 @@ class Planner:
                 e, key_map, agg_map, scope) if not isinstance(e, ast.Col) \
                 or (e.table is None and e.name in out_names and False) else None
         else:
+            if any(_has_window(it.expr) for it in q.items if not it.star):
+                plan = self._plan_windows(plan, scope, q)
             for it in q.items:
                 if it.star:
                     for alias, col, internal in scope.entries:
@@ -230,6 +234,141 @@ class Planner:
         return plan, out_names
 
     # ------------------------------------------------------------------
+    def _plan_windows(self, plan, scope: Scope, q: ast.Query):
+        """OVER-clause planning (reference: BodoSQL window RexOver lowering
+        onto bodo/libs/window/_window_calculator.cpp): each WindowE in the
+        select list is replaced by a reference to a Window plan-node output
+        column; windows sharing (partition, order) run in one node."""
+        pre_names = [e[2] for e in scope.entries]
+        pre_exprs = {n: ex.ColRef(n) for n in pre_names}
+        need_pre = False
+
+        def as_col(e, base):
+            nonlocal need_pre
+            if isinstance(e, ast.Col):
+                return scope.resolve(e.table, e.name)
+            name = self._uniq(base)
+            pre_exprs[name] = self.expr(e, scope)
+            pre_names.append(name)
+            need_pre = True
+            return name
+
+        groups: Dict[tuple, list] = {}  # (keys, order, asc) -> specs
+
+        def register(w: ast.WindowE) -> str:
+            nonlocal need_pre
+            keys = tuple(as_col(pe, "__wk") for pe in w.partition_by)
+            if not keys:
+                kname = self._uniq("__wk")
+                pre_exprs[kname] = ex.Const(1)
+                pre_names.append(kname)
+                need_pre = True
+                keys = (kname,)
+            order = tuple(as_col(oe, "__wo") for oe, _ in w.order_by)
+            asc = tuple(a for _, a in w.order_by)
+            out = self._uniq("__win")
+            fn = w.func
+            arg_col = ""
+            if w.args and not w.star:
+                arg_col = as_col(w.args[0], "__wa")
+            if fn == "row_number":
+                spec = (out, "", "row_number", None)
+            elif fn in ("rank", "dense_rank"):
+                spec = (out, "", fn, None)
+            elif fn in ("lag", "lead"):
+                n = 1
+                if len(w.args) > 1 and isinstance(w.args[1], ast.Lit):
+                    n = int(w.args[1].value)
+                spec = (out, arg_col, "shift", n if fn == "lag" else -n)
+            elif fn in AGG_FUNCS or fn == "count":
+                if order:
+                    if fn == "sum":
+                        spec = (out, arg_col, "cumsum", None)
+                    else:
+                        raise NotImplementedError(
+                            f"{fn.upper()} OVER (... ORDER BY) not supported")
+                elif fn == "count" and (w.star or not w.args):
+                    spec = (out, "", "transform_size", None)
+                else:
+                    spec = (out, arg_col,
+                            f"transform_{AGG_FUNCS.get(fn, fn)}", None)
+            else:
+                raise NotImplementedError(f"window function {fn}")
+            groups.setdefault((keys, order, asc), []).append(spec)
+            return out
+
+        def rewrite(e):
+            if isinstance(e, ast.WindowE):
+                name = register(e)
+                scope.add("", name, name)
+                return ast.Col(None, name)
+            if isinstance(e, (ast.Query, ast.SetOpQ)):
+                return e
+            for f in getattr(e, "__dataclass_fields__", {}):
+                v = getattr(e, f)
+                if isinstance(v, list):
+                    setattr(e, f, [rewrite(x) if hasattr(
+                        x, "__dataclass_fields__") else x for x in v])
+                elif hasattr(v, "__dataclass_fields__"):
+                    setattr(e, f, rewrite(v))
+            return e
+
+        for it in q.items:
+            if not it.star:
+                it.expr = rewrite(it.expr)
+        if need_pre:
+            plan = pn.Projection(
+                plan, tuple(pre_names),
+                tuple(pre_exprs[n] for n in pre_names))
+        for (keys, order, asc), specs in groups.items():
+            plan = pn.Window(plan, keys, order, asc, tuple(specs))
+        return plan
+
+    def _plan_setop(self, q: "ast.SetOpQ"):
+        """UNION [ALL] / INTERSECT / EXCEPT.  A trailing ORDER BY/LIMIT
+        parsed into the right arm applies to the whole set expression
+        (standard SQL binding)."""
+        order_by, limit = [], None
+        rq = q.right
+        if isinstance(rq, ast.Query) and (rq.order_by or rq.limit is not None):
+            order_by, limit = rq.order_by, rq.limit
+            rq.order_by, rq.limit = [], None
+        lplan, lnames = self.plan(q.left)
+        rplan, rnames = self.plan(rq)
+        if len(lnames) != len(rnames):
+            raise ValueError("set operation arms have different column counts")
+        # positional alignment: right arm takes the left arm's column names
+        rplan = pn.Projection(rplan, tuple(lnames),
+                              tuple(ex.ColRef(c) for c in rnames))
+        if q.op == "union":
+            out = pn.Union((lplan, rplan), distinct=not q.all)
+        else:
+            if q.all:
+                raise NotImplementedError(f"{q.op.upper()} ALL")
+            pfx = self._uniq("__set")
+            renamed = [f"{pfx}_{c}" for c in lnames]
+            rproj = pn.Projection(rplan, tuple(renamed),
+                                  tuple(ex.ColRef(c) for c in lnames))
+            rdis = pn.Distinct(rproj, tuple(renamed))
+            out = pn.Join(pn.Distinct(lplan, None), rdis, tuple(lnames),
+                          tuple(renamed),
+                          "semi" if q.op == "intersect" else "anti")
+        if order_by:
+            keys, asc = [], []
+            for e, a in order_by:
+                if isinstance(e, ast.Lit) and e.kind == "num":
+                    keys.append(lnames[int(e.value) - 1])
+                elif isinstance(e, ast.Col) and e.table is None:
+                    keys.append(e.name)
+                else:
+                    raise ValueError("set-op ORDER BY must be output "
+                                     "columns or positions")
+                asc.append(a)
+            out = pn.Sort(out, tuple(keys), tuple(asc))
+        if limit is not None:
+            out = pn.Limit(out, limit)
+        return out, lnames
+
     def _table_plan(self, tr: ast.TableRef, scope: Scope):
         alias = (tr.alias or tr.name).lower()
         if tr.subquery is not None:
@@ -877,3 +1016,19 @@ def _default_name(e) -> str:
     if isinstance(e, ast.ExtractE):
         return e.fld
     return "expr"
+
+
+def _has_window(e) -> bool:
+    if isinstance(e, ast.WindowE):
+        return True
+    if isinstance(e, (ast.Query,)):
+        return False
+    for f in getattr(e, "__dataclass_fields__", {}):
+        v = getattr(e, f)
+        if isinstance(v, (list, tuple)):
+            for i in v:
+                if hasattr(i, "__dataclass_fields__") and _has_window(i):
+                    return True
+        elif hasattr(v, "__dataclass_fields__") and _has_window(v):
+            return True
+    return False

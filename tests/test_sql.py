@@ -213,3 +213,71 @@ def test_having_distinct_limit():
     got2 = run_sql("SELECT DISTINCT c FROM t1 ORDER BY c", {"t1": df})
     exp2 = pd.DataFrame({"c": np.sort(df.c.unique())})
     _cmp(got2, exp2, sort=False)
+
+
+def test_set_operations():
+    rng = np.random.default_rng(4)
+    df = pd.DataFrame({"k": rng.integers(0, 6, 60), "v": rng.random(60)})
+    bc = BodoSQLContext({"t": df})
+    u = bc.sql("select k from t where v > 0.5 union "
+               "select k from t where v <= 0.5 order by k").to_pandas()
+    assert u.k.tolist() == sorted(df.k.unique().tolist())
+    ua = bc.sql("select k from t union all select k from t").to_pandas()
+    assert len(ua) == 2 * len(df)
+    i = bc.sql("select k from t where v > 0.3 intersect "
+               "select k from t where v < 0.7").to_pandas()
+    assert set(i.k) == set(df[df.v > 0.3].k) & set(df[df.v < 0.7].k)
+    e = bc.sql("select k from t except "
+               "select k from t where v > 0.2").to_pandas()
+    assert set(e.k) == set(df.k) - set(df[df.v > 0.2].k)
+
+
+def test_window_functions():
+    rng = np.random.default_rng(5)
+    df = pd.DataFrame({"k": rng.integers(0, 5, 80),
+                       "v": rng.random(80).round(3),
+                       "o": rng.permutation(80)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql(
+        "select k, v, row_number() over (partition by k order by o) as rn, "
+        "sum(v) over (partition by k) as tot, "
+        "rank() over (partition by k order by v) as rk, "
+        "count(*) over (partition by k) as cnt "
+        "from t order by k, rn").to_pandas()
+    exp = df.copy()
+    exp["rn"] = exp.sort_values("o").groupby("k").cumcount() + 1
+    exp["tot"] = exp.groupby("k")["v"].transform("sum")
+    exp["rk"] = exp.groupby("k")["v"].rank(method="min")
+    exp["cnt"] = exp.groupby("k")["v"].transform("size")
+    exp = exp.sort_values(["k", "rn"]).reset_index(drop=True)[
+        ["k", "v", "rn", "tot", "rk", "cnt"]]
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_window_lag_running_sum():
+    rng = np.random.default_rng(6)
+    df = pd.DataFrame({"k": rng.integers(0, 3, 50), "v": rng.random(50),
+                       "o": rng.permutation(50)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql(
+        "select k, o, lag(v) over (partition by k order by o) as pv, "
+        "lead(v, 2) over (partition by k order by o) as nv, "
+        "sum(v) over (partition by k order by o) as rs "
+        "from t order by k, o").to_pandas()
+    sdf = df.sort_values(["k", "o"]).reset_index(drop=True)
+    exp = sdf[["k", "o"]].copy()
+    exp["pv"] = sdf.groupby("k")["v"].shift(1)
+    exp["nv"] = sdf.groupby("k")["v"].shift(-2)
+    exp["rs"] = sdf.groupby("k")["v"].cumsum()
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_window_no_partition():
+    df = pd.DataFrame({"v": [3.0, 1.0, 2.0, 5.0, 4.0]})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select v, rank() over (order by v) as rk from t "
+                 "order by v").to_pandas()
+    exp = df.copy()
+    exp["rk"] = exp.v.rank(method="min")
+    exp = exp.sort_values("v").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
