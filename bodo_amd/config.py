@@ -69,3 +69,8 @@ def default_device() -> str:
     if DEVICE:
         return DEVICE
     return "cuda" if torch.cuda.is_available() else "cpu"
+
+#: out-of-core budget override in bytes (0 = auto: half of free HBM on GPU,
+#: unlimited on CPU).  See engine/ooc.py — partition-splitting hash operators
+#: with host-DRAM staging when a local groupby/join exceeds the budget.
+OOC_BYTES = _env_int("BODO_AMD_OOC_BYTES", 0)

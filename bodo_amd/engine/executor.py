@@ -146,15 +146,17 @@ def _exec_aggregate(node: pn.Aggregate, ctx) -> Table:
     child = _exec(node.child, ctx)
     keys = list(node.keys)
     aggs = list(node.aggs)
+    from . import ooc
+
     if ctx.world == 1:
-        return rel.groupby_local(child, keys, aggs, node.dropna)
+        return ooc.groupby_local(child, keys, aggs, node.dropna)
     if any(a[2] in SINGLE_PHASE_AGGS or callable(a[2]) for a in aggs):
         # shuffle raw rows by key hash, then single local groupby
         shuffled = _shuffle_by_keys(child, keys, ctx)
-        return rel.groupby_local(shuffled, keys, aggs, node.dropna)
+        return ooc.groupby_local(shuffled, keys, aggs, node.dropna)
     # two-phase: local pre-agg -> shuffle groups -> combine
     partial_aggs, final_map = _decompose_aggs(aggs)
-    local = rel.groupby_local(child, keys, partial_aggs, node.dropna)
+    local = ooc.groupby_local(child, keys, partial_aggs, node.dropna)
     shuffled = _shuffle_by_keys(local, keys, ctx)
     combine_aggs = [(name, name, rel.COMBINE_FUNC[f]) for name, _, f in partial_aggs]
     combined = rel.groupby_local(shuffled, keys, combine_aggs, node.dropna)
@@ -417,10 +419,12 @@ BROADCAST_JOIN_THRESHOLD = 256 * 1024 * 1024  # bytes (reference: gpu_join.h:92)
 def _exec_join(node: pn.Join, ctx) -> Table:
     left = _exec(node.left, ctx)
     right = _exec(node.right, ctx)
+    from . import ooc
+
     if ctx.world == 1 or node.how == "cross":
         if ctx.world > 1 and node.how == "cross":
             right = comm.allgather_table(right)
-        return rel.join_local(left, right, node.left_on, node.right_on,
+        return ooc.join_local(left, right, node.left_on, node.right_on,
                               node.how, node.suffixes)
     # broadcast the smaller side when cheap and semantics allow
     lsize = sum(comm.allgather_obj(left.nbytes()))
@@ -443,7 +447,7 @@ def _exec_join(node: pn.Join, ctx) -> Table:
     rp = torch.where(rp < 0, rp + ctx.world, rp)
     lshuf = comm.shuffle_table(left, lp)
     rshuf = comm.shuffle_table(right, rp)
-    return rel.join_local(lshuf, rshuf, node.left_on, node.right_on,
+    return ooc.join_local(lshuf, rshuf, node.left_on, node.right_on,
                           node.how, node.suffixes)
 
 
