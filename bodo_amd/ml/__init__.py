@@ -225,3 +225,143 @@ def torch_train(model, dataset, loss_fn, optimizer_cls=torch.optim.SGD,
             loss.backward()
             opt.step()
     return model
+
+
+class StandardScaler:
+    """Distributed standardization: per-feature mean/var from all-reduced
+    shard partials (reference: sklearn_ext.py preprocessing overloads)."""
+
+    def __init__(self, with_mean: bool = True, with_std: bool = True):
+        self.with_mean = with_mean
+        self.with_std = with_std
+        self.mean_: Optional[np.ndarray] = None
+        self.scale_: Optional[np.ndarray] = None
+
+    def fit(self, X):
+        from .. import config
+
+        dev = torch.device(config.default_device())
+        Xl = _as_matrix(X, dev).to(torch.float64)
+        parts = torch.zeros(2 * Xl.shape[1] + 1, dtype=torch.float64,
+                            device=Xl.device)
+        parts[0] = Xl.shape[0]
+        parts[1:Xl.shape[1] + 1] = Xl.sum(dim=0)
+        parts[Xl.shape[1] + 1:] = (Xl * Xl).sum(dim=0)
+        _allreduce_(parts)
+        n = parts[0].item()
+        s = parts[1:Xl.shape[1] + 1]
+        ss = parts[Xl.shape[1] + 1:]
+        mean = s / n
+        var = torch.clamp(ss / n - mean * mean, min=0.0)
+        self.mean_ = mean.cpu().numpy()
+        self.scale_ = np.sqrt(var.cpu().numpy())
+        self.scale_[self.scale_ == 0.0] = 1.0
+        return self
+
+    def transform(self, X):
+        from .. import config
+
+        dev = torch.device(config.default_device())
+        Xl = _as_matrix(X, dev).to(torch.float64)
+        if self.with_mean:
+            Xl = Xl - torch.as_tensor(self.mean_, device=Xl.device)
+        if self.with_std:
+            Xl = Xl / torch.as_tensor(self.scale_, device=Xl.device)
+        return Xl.cpu().numpy()
+
+    def fit_transform(self, X):
+        return self.fit(X).transform(X)
+
+
+class KMeans:
+    """Distributed Lloyd's iteration: the assignment step is one GEMM
+    (X @ C^T, rocBLAS/MFMA on device) per iteration; centroid sums and
+    counts are all-reduced over RCCL (reference: sklearn_ext.py KMeans
+    fit-with-allreduce pattern)."""
+
+    def __init__(self, n_clusters: int = 8, max_iter: int = 300,
+                 tol: float = 1e-6, random_state: Optional[int] = 0):
+        self.n_clusters = n_clusters
+        self.max_iter = max_iter
+        self.tol = tol
+        self.random_state = random_state
+        self.cluster_centers_: Optional[np.ndarray] = None
+        self.inertia_: Optional[float] = None
+        self.n_iter_: int = 0
+
+    def _init_centers(self, Xl: torch.Tensor) -> torch.Tensor:
+        """Deterministic farthest-point init (greedy k-means++): first
+        center is a seeded sample, each next center is the globally
+        farthest point from the chosen set (one allgather of (dist, row)
+        candidates per center)."""
+        k = self.n_clusters
+        world = comm.get_world_size() if comm.initialized() else 1
+        dev = Xl.device
+
+        def global_pick(val: float, vec: np.ndarray) -> np.ndarray:
+            if world == 1:
+                return vec
+            cands = comm.allgather_obj((float(val), vec))
+            return max(cands, key=lambda t: t[0])[1]
+
+        g = torch.Generator().manual_seed(self.random_state or 0)
+        first_i = int(torch.randint(max(Xl.shape[0], 1), (1,),
+                                    generator=g).item()) % max(Xl.shape[0], 1)
+        first = global_pick(Xl.shape[0], Xl[first_i].cpu().numpy())
+        centers = [torch.as_tensor(first, device=dev, dtype=Xl.dtype)]
+        d2min = ((Xl - centers[0]) ** 2).sum(dim=1)
+        for _ in range(1, k):
+            j = int(torch.argmax(d2min).item())
+            vec = global_pick(d2min[j].item(), Xl[j].cpu().numpy())
+            c = torch.as_tensor(vec, device=dev, dtype=Xl.dtype)
+            centers.append(c)
+            d2min = torch.minimum(d2min, ((Xl - c) ** 2).sum(dim=1))
+        return torch.stack(centers)
+
+    def fit(self, X):
+        from .. import config
+
+        dev = torch.device(config.default_device())
+        Xl = _as_matrix(X, dev).to(torch.float32)
+        C = self._init_centers(Xl)
+        x2 = (Xl * Xl).sum(dim=1, keepdim=True)
+        k, d = C.shape
+        for it in range(self.max_iter):
+            c2 = (C * C).sum(dim=1)
+            d2 = x2 - 2.0 * (Xl @ C.t()) + c2
+            assign = torch.argmin(d2, dim=1)
+            sums = torch.zeros((k, d), dtype=torch.float64, device=dev)
+            sums.index_add_(0, assign, Xl.to(torch.float64))
+            counts = torch.bincount(assign, minlength=k).to(torch.float64)
+            local_inertia = torch.gather(
+                d2, 1, assign.view(-1, 1)).clamp_min(0).sum()
+            packed = torch.cat([sums.reshape(-1), counts,
+                                local_inertia.to(torch.float64).reshape(1)])
+            _allreduce_(packed)
+            sums = packed[:k * d].reshape(k, d)
+            counts = packed[k * d:k * d + k]
+            self.inertia_ = float(packed[-1].item())
+            newC = torch.where(counts.view(-1, 1) > 0,
+                               (sums / counts.clamp(min=1.0).view(-1, 1)),
+                               C.to(torch.float64)).to(torch.float32)
+            shift = float(((newC - C) ** 2).sum().item())
+            C = newC
+            self.n_iter_ = it + 1
+            if shift <= self.tol:
+                break
+        self.cluster_centers_ = C.cpu().numpy()
+        return self
+
+    def predict(self, X) -> np.ndarray:
+        from .. import config
+
+        dev = torch.device(config.default_device())
+        Xl = _as_matrix(X, dev).to(torch.float32)
+        C = torch.as_tensor(self.cluster_centers_, device=dev)
+        d2 = (Xl * Xl).sum(dim=1, keepdim=True) - 2.0 * (Xl @ C.t()) \
+            + (C * C).sum(dim=1)
+        return torch.argmin(d2, dim=1).cpu().numpy()
+
+    def fit_predict(self, X) -> np.ndarray:
+        self.fit(X)
+        return self.predict(X)

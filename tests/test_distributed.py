@@ -286,3 +286,35 @@ def test_dist_iceberg_roundtrip(tmp_path):
     got["c"] = got["c"].astype(str)
     exp = df.sort_values(["a", "b"]).reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_kmeans(bpd, rank, payload):
+    import numpy as np
+
+    from bodo_amd.ml import KMeans, StandardScaler
+
+    X = payload["X"]
+    n = len(X)
+    # block-shard the rows like the engine does
+    w, r = 2, rank
+    base, rem = divmod(n, w)
+    start = r * base + min(r, rem)
+    stop = start + base + (1 if r < rem else 0)
+    shard = X[start:stop]
+    sc = StandardScaler().fit(shard)
+    km = KMeans(n_clusters=4, random_state=1).fit(shard)
+    return {"mean": sc.mean_, "scale": sc.scale_,
+            "centers": np.sort(km.cluster_centers_.round(0), axis=0),
+            "inertia": km.inertia_}
+
+
+def test_dist_ml():
+    rng = np.random.default_rng(21)
+    centers = np.array([[0.0, 0.0], [10.0, 0.0], [0.0, 10.0], [10.0, 10.0]])
+    X = np.concatenate([rng.normal(c, 0.5, size=(200, 2)) for c in centers])
+    X = X[rng.permutation(len(X))].astype(np.float32)
+    got = run_dist(_q_kmeans, {"X": X})
+    np.testing.assert_allclose(got["mean"], X.mean(axis=0), rtol=1e-4)
+    np.testing.assert_allclose(got["scale"], X.std(axis=0), rtol=1e-3)
+    exp_centers = np.sort(centers, axis=0)
+    np.testing.assert_allclose(got["centers"], exp_centers, atol=1.0)

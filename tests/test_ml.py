@@ -77,3 +77,39 @@ def test_linreg_gpu():
         assert np.allclose(m.coef_, w, atol=1e-2)
     finally:
         cfg.DEVICE = ""
+
+
+def test_standard_scaler():
+    from sklearn.preprocessing import StandardScaler as SkScaler
+
+    from bodo_amd.ml import StandardScaler
+
+    rng = np.random.default_rng(7)
+    X = rng.random((500, 4)) * np.array([1.0, 10.0, 0.1, 5.0]) + 3.0
+    ours = StandardScaler().fit(X)
+    ref = SkScaler().fit(X)
+    # inputs pass through the float32 device-matrix path
+    np.testing.assert_allclose(ours.mean_, ref.mean_, rtol=1e-6)
+    np.testing.assert_allclose(ours.scale_, ref.scale_, rtol=1e-5)
+    np.testing.assert_allclose(ours.transform(X), ref.transform(X),
+                               rtol=1e-4, atol=1e-5)
+
+
+def test_kmeans_quality():
+    from sklearn.cluster import KMeans as SkKMeans
+
+    from bodo_amd.ml import KMeans
+
+    rng = np.random.default_rng(8)
+    centers = np.array([[0.0, 0.0], [10.0, 0.0], [0.0, 10.0], [10.0, 10.0]])
+    X = np.concatenate([rng.normal(c, 0.5, size=(300, 2)) for c in centers])
+    km = KMeans(n_clusters=4, random_state=3).fit(X.astype(np.float32))
+    sk = SkKMeans(n_clusters=4, n_init=10, random_state=3).fit(X)
+    # well-separated blobs: both must find the 4 true centers
+    ours = np.sort(km.cluster_centers_.round(0), axis=0)
+    theirs = np.sort(sk.cluster_centers_.round(0), axis=0)
+    np.testing.assert_allclose(np.sort(ours.flatten()),
+                               np.sort(theirs.flatten()), atol=1.0)
+    assert km.inertia_ <= sk.inertia_ * 1.2
+    pred = km.predict(X.astype(np.float32))
+    assert len(np.unique(pred)) == 4
