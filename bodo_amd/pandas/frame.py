@@ -511,6 +511,51 @@ class BodoDataFrame:
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingFrame(self, int(window), min_periods)
 
+    def melt(self, id_vars=None, value_vars=None, var_name="variable",
+             value_name="value") -> "BodoDataFrame":
+        """Wide-to-long unpivot as a lazy plan: one projection per value
+        column (variable = constant) unioned together — stays distributed
+        (reference: frame.py melt)."""
+        from ..plan.expr import Const
+
+        ids = list(id_vars) if id_vars is not None else []
+        if isinstance(id_vars, str):
+            ids = [id_vars]
+        vals = list(value_vars) if value_vars is not None else \
+            [c for c in self._columns if c not in ids]
+        if isinstance(value_vars, str):
+            vals = [value_vars]
+        names = tuple(ids + [var_name, value_name])
+        parts = []
+        for v in vals:
+            exprs = tuple([ColRef(c) for c in ids]
+                          + [Const(v), ColRef(v)])
+            parts.append(pn.Projection(self._plan, names, exprs))
+        plan = parts[0] if len(parts) == 1 else pn.Union(tuple(parts), False)
+        return BodoDataFrame(plan, list(names))
+
+    def pivot_table(self, values=None, index=None, columns=None,
+                    aggfunc="mean"):
+        """Distributed groupby([index, columns]) then a host pivot of the
+        (small) aggregated result (reference: frame.py pivot_table)."""
+        assert index is not None and columns is not None
+        idx = [index] if isinstance(index, str) else list(index)
+        cols = [columns] if isinstance(columns, str) else list(columns)
+        if values is None:
+            values = [c for c in self._columns
+                      if c not in idx + cols]
+        vals = [values] if isinstance(values, str) else list(values)
+        agg_kwargs = {v: (v, aggfunc) for v in vals}
+        small = self.groupby(idx + cols, as_index=False).agg(
+            **{k: pd.NamedAgg(c, f) for k, (c, f) in agg_kwargs.items()})
+        pdf = small.to_pandas()
+        for c in pdf.columns:
+            if isinstance(pdf[c].dtype, pd.CategoricalDtype):
+                pdf[c] = pdf[c].astype(object)
+        out_vals = values if isinstance(values, str) else vals
+        return pdf.pivot_table(values=out_vals, index=idx, columns=cols,
+                               aggfunc="first")
+
     # ------------------------------------------------------------------
     # IO
     # ------------------------------------------------------------------

@@ -113,3 +113,24 @@ def test_series_sort_values(df):
     assert (got == np.sort(df.a.to_numpy())).all()
     got = b.a.sort_values(ascending=False).to_pandas().to_numpy()
     assert (got == np.sort(df.a.to_numpy())[::-1]).all()
+
+
+def test_melt(df):
+    b = bpd.from_pandas(df)
+    got = b.melt(id_vars=["a"], value_vars=["b"]).to_pandas()
+    got["variable"] = got["variable"].astype(str)
+    got = got.sort_values(["a", "value"]).reset_index(drop=True)
+    exp = df.melt(id_vars=["a"], value_vars=["b"]).sort_values(
+        ["a", "value"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_pivot_table(df):
+    b = bpd.from_pandas(df)
+    got = b.pivot_table(values="b", index="a", columns="c", aggfunc="mean")
+    exp = df.pivot_table(values="b", index="a", columns="c", aggfunc="mean")
+    got.index = got.index.astype(exp.index.dtype)
+    got.columns = [str(c) for c in got.columns]
+    exp.columns = [str(c) for c in exp.columns]
+    pd.testing.assert_frame_equal(got.sort_index(), exp.sort_index(),
+                                  check_dtype=False, check_names=False)
