@@ -74,7 +74,68 @@ def push_filters(node: pn.PlanNode) -> pn.PlanNode:
                 return pn.Projection(
                     push_filters(pn.Filter(child.child, newcond)),
                     child.names, child.exprs)
+        if isinstance(child, pn.Join):
+            pushed = _push_into_join(child, conjuncts)
+            if pushed is not None:
+                return pushed
     return node
+
+
+def _push_into_join(join: pn.Join, conjuncts: List[Expr]):
+    """Push row-local conjuncts below a join to the side whose columns cover
+    them (the single biggest intermediate-size lever: TPC-H q5/q8 join
+    chains materialize 37-60M rows before late filters without it;
+    reference analog: DuckDB FilterPushdown used via the vendored
+    optimizer).  Preserved-side rules: a left join keeps left-side pushes
+    only, a right join right-side only; FULL OUTER pushes nothing."""
+    from ..plan.expr import ScalarSubquery, SemiJoinIn
+
+    lcols = join.left.out_columns()
+    rcols = join.right.out_columns()
+    if lcols is None or rcols is None:
+        return None
+    sl, sr = set(lcols), set(rcols)
+    overlap = sl & sr
+    left_ok = join.how in ("inner", "left", "semi", "anti", "cross")
+    right_ok = join.how in ("inner", "right", "cross")
+
+    def has_subplan(e) -> bool:
+        if isinstance(e, (ScalarSubquery, SemiJoinIn)):
+            return True
+        return any(has_subplan(c) for c in e.children())
+
+    push_l: List[Expr] = []
+    push_r: List[Expr] = []
+    keep: List[Expr] = []
+    for cj in conjuncts:
+        try:
+            refs = expr_columns(cj)
+        except Exception:
+            refs = set()
+        if refs and not (refs & overlap) and not has_subplan(cj):
+            if refs <= sl and left_ok:
+                push_l.append(cj)
+                continue
+            if refs <= sr and right_ok:
+                push_r.append(cj)
+                continue
+        keep.append(cj)
+    if not push_l and not push_r:
+        return None
+    from ..user_logging import log_message
+
+    log_message("Filter Pushdown",
+                f"pushed {len(push_l)}+{len(push_r)} filter(s) below "
+                f"{join.how} join")
+    nl = push_filters(pn.Filter(join.left, _join_and(push_l))) \
+        if push_l else join.left
+    nr = push_filters(pn.Filter(join.right, _join_and(push_r))) \
+        if push_r else join.right
+    new_join = pn.Join(nl, nr, join.left_on, join.right_on, join.how,
+                       join.suffixes)
+    if keep:
+        return pn.Filter(new_join, _join_and(keep))
+    return new_join
 
 
 def _split_and(e: Expr) -> List[Expr]:

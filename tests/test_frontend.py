@@ -268,3 +268,38 @@ def test_groupby_rank():
         return df[["a", "b", "r"]]
 
     check_query(q, {"df": simple_df(400)})
+
+
+def test_filter_pushdown_below_join():
+    """The optimizer must move side-local filters below a join."""
+    from bodo_amd.engine.optimizer import optimize
+    from bodo_amd.plan import expr as ex
+    from bodo_amd.plan import nodes as pn
+
+    left = pn.PandasScan("L", ("a", "b"), distributed=False)
+    right = pn.PandasScan("R", ("k", "c"), distributed=False)
+    j = pn.Join(left, right, ("a",), ("k",), "inner")
+    f = pn.Filter(j, ex.BoolOp(
+        "and", ex.Cmp("gt", ex.ColRef("b"), ex.Const(1)),
+        ex.Cmp("lt", ex.ColRef("c"), ex.Const(5))))
+    opt = optimize(f)
+    assert isinstance(opt, pn.Join), type(opt)
+    def has_filter(n):
+        if isinstance(n, pn.Filter):
+            return True
+        return any(has_filter(c) for c in n.children())
+    assert has_filter(opt.left) and has_filter(opt.right)
+
+
+def test_filter_stays_above_left_join_right_filter():
+    from bodo_amd.engine.optimizer import optimize
+    from bodo_amd.plan import expr as ex
+    from bodo_amd.plan import nodes as pn
+
+    left = pn.PandasScan("L", ("a", "b"), distributed=False)
+    right = pn.PandasScan("R", ("k", "c"), distributed=False)
+    j = pn.Join(left, right, ("a",), ("k",), "left")
+    f = pn.Filter(j, ex.Cmp("lt", ex.ColRef("c"), ex.Const(5)))
+    opt = optimize(f)
+    # right-side filter must NOT move below a LEFT join
+    assert isinstance(opt, pn.Filter), type(opt)
