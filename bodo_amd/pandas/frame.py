@@ -511,6 +511,16 @@ class BodoDataFrame:
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingFrame(self, int(window), min_periods)
 
+    def explain(self, optimized: bool = True) -> str:
+        """Pretty-print this frame's logical plan (reference: plan dumps at
+        tracing_level>=2, bodo/pandas/plan.py:1090-1096)."""
+        plan = self._plan
+        if optimized:
+            from ..engine.optimizer import optimize
+
+            plan = optimize(plan)
+        return pn.explain(plan)
+
     def melt(self, id_vars=None, value_vars=None, var_name="variable",
              value_name="value") -> "BodoDataFrame":
         """Wide-to-long unpivot as a lazy plan: one projection per value

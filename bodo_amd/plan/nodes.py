@@ -391,3 +391,33 @@ def walk(node: PlanNode):
     yield node
     for c in node.children():
         yield from walk(c)
+
+
+def explain(node: PlanNode, indent: int = 0) -> str:
+    """Pretty-print a plan tree (reference: plan dumps at
+    tracing_level>=2, bodo/pandas/plan.py:1090)."""
+    pad = "  " * indent
+    name = type(node).__name__
+    detail = ""
+    if isinstance(node, Projection):
+        detail = f" cols={list(node.names)[:8]}{'...' if len(node.names) > 8 else ''}"
+    elif isinstance(node, Filter):
+        detail = f" cond={node.cond!r}"[:120]
+    elif isinstance(node, Aggregate):
+        detail = f" keys={list(node.keys)} aggs={[a[0] for a in node.aggs]}"
+    elif isinstance(node, Join):
+        detail = f" how={node.how} on={list(node.left_on)}={list(node.right_on)}"
+    elif isinstance(node, Sort):
+        detail = f" keys={list(node.keys)}"
+    elif isinstance(node, ParquetScan):
+        detail = f" path={node.path} cols={node.columns} nfilters={len(node.filters)}"
+    elif isinstance(node, PandasScan):
+        detail = f" id={node.data_id[:8]} cols={list(node.names)[:6]}"
+    elif isinstance(node, Limit):
+        detail = f" n={node.n}"
+    elif isinstance(node, Window):
+        detail = f" keys={list(node.keys)} specs={[s[0] for s in node.specs]}"
+    lines = [f"{pad}{name}{detail}"]
+    for c in node.children():
+        lines.append(explain(c, indent + 1))
+    return "\n".join(lines)

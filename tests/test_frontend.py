@@ -303,3 +303,11 @@ def test_filter_stays_above_left_join_right_filter():
     opt = optimize(f)
     # right-side filter must NOT move below a LEFT join
     assert isinstance(opt, pn.Filter), type(opt)
+
+
+def test_explain():
+    df = simple_df(50)
+    b = bpd.from_pandas(df)
+    txt = b[b.a > 3][["a", "b"]].explain()
+    assert "Filter" in txt and "PandasScan" in txt
+    assert isinstance(txt, str) and len(txt.splitlines()) >= 2
