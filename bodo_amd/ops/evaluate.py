@@ -273,18 +273,29 @@ class _Evaluator:
             import pyarrow as pa
 
             return Column.from_arrow(pa.Array.from_pandas(ser), self.device)
-        out_data = other.data.clone()
+        thens = [self.visit(t) for t in e.thens]
+        # promote across all branches (int + float/null branches -> float)
+        res_torch = other.data.dtype
+        for t in thens:
+            res_torch = torch.promote_types(res_torch, t.data.dtype)
+        res_dtype = other.dtype
+        if res_torch != other.data.dtype:
+            for t in [other] + thens:
+                if t.data.dtype == res_torch:
+                    res_dtype = t.dtype
+                    break
+        out_data = other.data.to(res_torch).clone() \
+            if other.data.dtype != res_torch else other.data.clone()
         out_mask = None if other.mask is None else other.mask.clone()
         # apply in reverse so the first matching cond wins
-        for cond_e, then_e in reversed(list(zip(e.conds, e.thens))):
+        for cond_e, t in reversed(list(zip(e.conds, thens))):
             c = self.visit(cond_e).data
-            t = self.visit(then_e)
-            td = t.data.to(out_data.dtype) if t.data.dtype != out_data.dtype else t.data
+            td = t.data.to(res_torch) if t.data.dtype != res_torch else t.data
             out_data = torch.where(c, td, out_data)
             if out_mask is not None:
                 tm = t.mask if t.mask is not None else torch.ones_like(out_mask)
                 out_mask = torch.where(c, tm, out_mask)
-        return Column(other.dtype, out_data, out_mask)
+        return Column(res_dtype, out_data, out_mask)
 
 
 # ----------------------------------------------------------------------
@@ -308,6 +319,8 @@ def infer_const_dtype(v) -> DType:
 
 
 def normalize_const(v, dtype: DType):
+    if v is None:
+        return float("nan") if dtype.is_float else None
     if dtype.kind == TypeKind.TIMESTAMP_NS:
         return int(pd.Timestamp(v).value)
     if dtype.kind == TypeKind.DATE32:

@@ -546,6 +546,16 @@ class Parser:
                 q = self.parse_select()
                 self.expect_op(")")
                 return ExistsE(q)
+            if t.value in ("left", "right") and self.peek() \
+                    and self.peek().kind == "op" and self.peek().value == "(":
+                # LEFT(s, n) / RIGHT(s, n) string functions (the words are
+                # otherwise join keywords)
+                self.i += 1
+                args = [self.parse_expr()]
+                while self.accept_op(","):
+                    args.append(self.parse_expr())
+                self.expect_op(")")
+                return Func(t.value, args)
             raise SyntaxError(f"unexpected keyword {t.value!r}")
         if t.kind == "op" and t.value == "(":
             nt = self.peek()

@@ -896,6 +896,69 @@ class Planner:
                 for a in reversed(args[:-1]):
                     out = ex.Case((ex.IsNull(a, negate=True),), (a,), out)
                 return out
+            if name == "nullif":
+                a = self.expr(e.args[0], scope)
+                b = self.expr(e.args[1], scope)
+                return ex.Case((ex.Cmp("eq", a, b),),
+                               (ex.Const(None, bt.float64),), a)
+            if name == "floor":
+                return ex.BinOp("floordiv", self.expr(e.args[0], scope),
+                                ex.Const(1))
+            if name in ("ceil", "ceiling"):
+                x = self.expr(e.args[0], scope)
+                zero = ex.Const(0)
+                return ex.BinOp(
+                    "sub", zero,
+                    ex.BinOp("floordiv", ex.BinOp("sub", zero, x),
+                             ex.Const(1)))
+            if name == "mod":
+                # python modulo semantics (sign of divisor); SQL MOD takes
+                # the dividend's sign — identical for positive operands
+                return ex.BinOp("mod", self.expr(e.args[0], scope),
+                                self.expr(e.args[1], scope))
+            if name in ("power", "pow"):
+                return ex.BinOp("pow", self.expr(e.args[0], scope),
+                                self.expr(e.args[1], scope))
+            if name == "sqrt":
+                return ex.BinOp("pow", self.expr(e.args[0], scope),
+                                ex.Const(0.5))
+            if name in ("exp", "ln", "log", "log10"):
+                import math
+
+                f = {"exp": math.exp, "ln": math.log, "log": math.log,
+                     "log10": math.log10}[name]
+                return ex.UdfMap(self.expr(e.args[0], scope), f, None)
+            if name == "sign":
+                x = self.expr(e.args[0], scope)
+                return ex.Case(
+                    (ex.Cmp("lt", x, ex.Const(0)),
+                     ex.Cmp("gt", x, ex.Const(0))),
+                    (ex.Const(-1), ex.Const(1)), ex.Const(0))
+            if name in ("greatest", "least"):
+                op = "gt" if name == "greatest" else "lt"
+                args = [self.expr(a, scope) for a in e.args]
+                out = args[0]
+                for a in args[1:]:
+                    out = ex.Case((ex.Cmp(op, out, a),), (out,), a)
+                return out
+            if name == "replace":
+                return ex.StrOp(self.expr(e.args[0], scope), "replace",
+                                (e.args[1].value, e.args[2].value),
+                                (("regex", False),))
+            if name in ("ltrim", "rtrim"):
+                op = "lstrip" if name == "ltrim" else "rstrip"
+                chars = (e.args[1].value,) if len(e.args) > 1 else ()
+                return ex.StrOp(self.expr(e.args[0], scope), op, chars)
+            if name == "left":
+                n = int(e.args[1].value)
+                return ex.StrOp(self.expr(e.args[0], scope), "slice",
+                                (0, n, 1))
+            if name == "right":
+                n = int(e.args[1].value)
+                return ex.StrOp(self.expr(e.args[0], scope), "slice",
+                                (-n, None, 1))
+            if name == "initcap":
+                return ex.StrOp(self.expr(e.args[0], scope), "title")
             raise NotImplementedError(f"SQL function {name}")
         raise NotImplementedError(f"expr {e}")
 

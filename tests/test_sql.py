@@ -281,3 +281,35 @@ def test_window_no_partition():
     exp["rk"] = exp.v.rank(method="min")
     exp = exp.sort_values("v").reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_scalar_functions():
+    rng = np.random.default_rng(3)
+    df = pd.DataFrame({"x": rng.random(30) * 10 - 3,
+                       "y": rng.random(30) + 0.5,
+                       "s": ["Hello World", "foo bar", "Aaa"] * 10})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql(
+        "select floor(x) as f, ceil(x) as c, "
+        "mod(cast(x as int) + 10, 3) as m, power(y, 2) as p, sqrt(y) as q, "
+        "exp(y) as ex, ln(y) as l, sign(x) as sg, greatest(x, y) as g, "
+        "least(x, y) as le, nullif(sign(x), 1) as nl, "
+        "replace(s, 'o', '0') as rep, ltrim(s, 'H') as lt, "
+        "left(s, 3) as lf, right(s, 3) as rt, initcap(s) as ic from t"
+    ).to_pandas()
+    exp = pd.DataFrame({
+        "f": np.floor(df.x), "c": np.ceil(df.x),
+        "m": (df.x.astype(int) + 10) % 3, "p": df.y ** 2,
+        "q": np.sqrt(df.y), "ex": np.exp(df.y), "l": np.log(df.y),
+        "sg": np.sign(df.x), "g": np.maximum(df.x, df.y),
+        "le": np.minimum(df.x, df.y),
+        "nl": np.where(np.sign(df.x) == 1, np.nan, np.sign(df.x)),
+        "rep": df.s.str.replace("o", "0", regex=False),
+        "lt": df.s.str.lstrip("H"), "lf": df.s.str.slice(0, 3),
+        "rt": df.s.str.slice(-3), "ic": df.s.str.title()})
+    for c in exp.columns:
+        g = got[c]
+        if isinstance(g.dtype, pd.CategoricalDtype):
+            g = g.astype(object)
+        pd.testing.assert_series_equal(g, exp[c], check_dtype=False,
+                                       check_names=False)
