@@ -57,6 +57,27 @@ def read_iceberg(table_uri, snapshot_id=None, columns=None,
     return BodoDataFrame(plan, list(columns) if columns else list(names))
 
 
+def read_sql(sql, con, **kwargs) -> BodoDataFrame:
+    """Database read through any DB-API connection or a sqlite path/URI
+    (reference: bodo/ir/sql_ext.py SqlReader; the Snowflake fast path is a
+    connector away — no network in this build).  The result distributes by
+    block-slicing rows across ranks."""
+    import sqlite3
+
+    close = None
+    if isinstance(con, str):
+        path = con[len("sqlite://"):] if con.startswith("sqlite://") else con
+        con = sqlite3.connect(path)
+        close = con
+    try:
+        df = _pd.read_sql_query(sql, con, **kwargs) \
+            if hasattr(_pd, "read_sql_query") else _pd.read_sql(sql, con)
+    finally:
+        if close is not None:
+            close.close()
+    return from_pandas(df)
+
+
 def read_csv(path, **kwargs) -> BodoDataFrame:
     options = tuple(sorted(kwargs.items(), key=lambda kv: kv[0]))
     names = _csv.schema_names(str(path), dict(options))

@@ -588,6 +588,33 @@ class BodoDataFrame:
         pdf = self.to_pandas()
         return pdf.to_csv(path, index=False, **kwargs)
 
+    def to_json(self, path=None, orient="records", lines=True, **kwargs):
+        pdf = self.to_pandas()
+        return pdf.to_json(path, orient=orient, lines=lines, **kwargs)
+
+    def to_sql(self, name, con, if_exists="fail", **kwargs):
+        """Write through a DB-API connection or sqlite path (reference:
+        to_sql via the Snowflake writer; gathered then written by rank 0)."""
+        import sqlite3
+
+        from ..parallel import comm
+
+        pdf = self.to_pandas()
+        if comm.get_rank() != 0:
+            return
+        close = None
+        if isinstance(con, str):
+            p = con[len("sqlite://"):] if con.startswith("sqlite://") else con
+            con = sqlite3.connect(p)
+            close = con
+        try:
+            pdf.to_sql(name, con, if_exists=if_exists, index=False, **kwargs)
+            if close is not None:
+                con.commit()
+        finally:
+            if close is not None:
+                close.close()
+
     # reductions over the whole frame fall back (rare)
     def count(self):
         return self.to_pandas().count()

@@ -134,3 +134,23 @@ def test_pivot_table(df):
     exp.columns = [str(c) for c in exp.columns]
     pd.testing.assert_frame_equal(got.sort_index(), exp.sort_index(),
                                   check_dtype=False, check_names=False)
+
+
+def test_read_sql_roundtrip(tmp_path, df):
+    dbp = str(tmp_path / "t.db")
+    b = bpd.from_pandas(df[["a", "b", "c"]])
+    b.to_sql("tab", dbp, if_exists="replace")
+    got = bpd.read_sql("select a, b, c from tab where a > 3", dbp).to_pandas()
+    got = _decat(got)
+    exp = df[df.a > 3][["a", "b", "c"]].reset_index(drop=True)
+    pd.testing.assert_frame_equal(
+        got.sort_values(["a", "b"]).reset_index(drop=True),
+        exp.sort_values(["a", "b"]).reset_index(drop=True), check_dtype=False)
+
+
+def test_to_json(tmp_path, df):
+    p = str(tmp_path / "out.json")
+    bpd.from_pandas(df[["a", "b"]]).to_json(p)
+    got = pd.read_json(p, orient="records", lines=True)
+    pd.testing.assert_frame_equal(got, df[["a", "b"]].reset_index(drop=True),
+                                  check_dtype=False)
