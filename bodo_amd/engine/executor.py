@@ -371,6 +371,70 @@ def _exec_rowid(node: pn.RowId, ctx) -> Table:
     return child.with_column(node.name, Column(bt.int64, rid))
 
 
+_CUM_IDENT = {"cumsum": 0.0, "cumprod": 1.0,
+              "cummin": float("inf"), "cummax": float("-inf")}
+
+
+def _exec_cumulative(node: pn.Cumulative, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    n = len(child)
+    names, cols = [], []
+    for out_name, in_name, func in node.specs:
+        col = child.column(in_name)
+        data = col.data
+        invalid = None
+        if col.dtype.is_float:
+            invalid = torch.isnan(data)
+        if col.mask is not None:
+            miss = ~col.mask
+            invalid = miss if invalid is None else (invalid | miss)
+        work = data.to(torch.float64) if invalid is not None else data
+        ident = _CUM_IDENT[func]
+        if invalid is not None:
+            work = torch.where(invalid, torch.tensor(
+                ident, dtype=work.dtype, device=work.device), work)
+        if func == "cumsum":
+            local = torch.cumsum(work, 0)
+        elif func == "cumprod":
+            local = torch.cumprod(work, 0)
+        elif func == "cummin":
+            local = torch.cummin(work, 0).values
+        else:
+            local = torch.cummax(work, 0).values
+        if ctx.world > 1:
+            total = local[-1].item() if n else ident
+            totals = comm.allgather_obj(total)
+            prefix = ident
+            for t in totals[:ctx.rank]:
+                if func == "cumsum":
+                    prefix += t
+                elif func == "cumprod":
+                    prefix *= t
+                elif func == "cummin":
+                    prefix = min(prefix, t)
+                else:
+                    prefix = max(prefix, t)
+            if prefix != ident:
+                pt = torch.tensor(prefix, dtype=local.dtype,
+                                  device=local.device)
+                if func == "cumsum":
+                    local = local + pt
+                elif func == "cumprod":
+                    local = local * pt
+                elif func == "cummin":
+                    local = torch.minimum(local, pt)
+                else:
+                    local = torch.maximum(local, pt)
+        out_dtype = col.dtype
+        if invalid is not None:
+            local = torch.where(invalid, torch.tensor(
+                float("nan"), dtype=local.dtype, device=local.device), local)
+            out_dtype = bt.float64
+        names.append(out_name)
+        cols.append(Column(out_dtype, local))
+    return Table(names, cols, n)
+
+
 def _exec_rolling(node: pn.Rolling, ctx) -> Table:
     child = _exec(node.child, ctx)
     w = int(node.window)
@@ -554,6 +618,7 @@ _HANDLERS = {
     pn.ShuffleByKey: _exec_shuffle_by_key,
     pn.RowId: _exec_rowid,
     pn.Rolling: _exec_rolling,
+    pn.Cumulative: _exec_cumulative,
     pn.Window: _exec_window,
     pn.Join: _exec_join,
     pn.Union: _exec_union,

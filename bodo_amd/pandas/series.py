@@ -283,6 +283,22 @@ class BodoSeries:
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingSeries(self, int(window), min_periods)
 
+    def _cum(self, func) -> "BodoSeries":
+        plan = pn.Cumulative(self._as_projection_plan(), (("v", "v", func),))
+        return BodoSeries(plan, ColRef("v"), self.name)
+
+    def cumsum(self):
+        return self._cum("cumsum")
+
+    def cumprod(self):
+        return self._cum("cumprod")
+
+    def cummin(self):
+        return self._cum("cummin")
+
+    def cummax(self):
+        return self._cum("cummax")
+
     def _as_projection_plan(self):
         return pn.Projection(self._plan, ("v",), (self._expr,))
 

@@ -244,6 +244,26 @@ class Rolling(PlanNode):
 
 
 @dataclass(frozen=True)
+class Cumulative(PlanNode):
+    """Global-order cumulative ops (cumsum/cumprod/cummin/cummax) over the
+    distributed row order: local scan + an exscan of shard totals
+    (reference: dist_exscan in bodo/transforms/distributed_pass.py,
+    _distributed.h Exscan)."""
+
+    child: PlanNode
+    specs: Tuple[Tuple[str, str, str], ...] = ()  # (out, in, func)
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Cumulative(ch[0], self.specs)
+
+    def out_columns(self):
+        return [s[0] for s in self.specs]
+
+
+@dataclass(frozen=True)
 class ShuffleByKey(PlanNode):
     """Explicit hash-repartition so equal keys co-locate on one rank
     (used by groupby.apply / median paths)."""

@@ -318,3 +318,18 @@ def test_dist_ml():
     np.testing.assert_allclose(got["scale"], X.std(axis=0), rtol=1e-3)
     exp_centers = np.sort(centers, axis=0)
     np.testing.assert_allclose(got["centers"], exp_centers, atol=1.0)
+
+
+def _q_cumsum(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).b.cumsum()
+
+
+def test_dist_cumsum():
+    """Global cumulative sum must be exact across shard boundaries
+    (exscan of shard totals)."""
+    df = _df(700, 23)
+    got = run_dist(_q_cumsum, {"df": df})
+    got = got if isinstance(got, pd.Series) else got.iloc[:, 0]
+    exp = df.b.cumsum().reset_index(drop=True)
+    pd.testing.assert_series_equal(got.reset_index(drop=True), exp,
+                                   check_names=False, check_dtype=False)

@@ -154,3 +154,15 @@ def test_to_json(tmp_path, df):
     got = pd.read_json(p, orient="records", lines=True)
     pd.testing.assert_frame_equal(got, df[["a", "b"]].reset_index(drop=True),
                                   check_dtype=False)
+
+
+def test_cumulative(df):
+    b = bpd.from_pandas(df)
+    for f in ["cumsum", "cumprod", "cummin", "cummax"]:
+        got = getattr(b.b, f)().to_pandas()
+        exp = getattr(df.b, f)().reset_index(drop=True)
+        pd.testing.assert_series_equal(got, exp, check_names=False,
+                                       check_dtype=False)
+    got = b.a.cumsum().to_pandas()
+    pd.testing.assert_series_equal(got, df.a.cumsum().reset_index(drop=True),
+                                   check_names=False, check_dtype=False)
