@@ -371,6 +371,57 @@ def _exec_rowid(node: pn.RowId, ctx) -> Table:
     return child.with_column(node.name, Column(bt.int64, rid))
 
 
+def _exec_shift(node: pn.Shift, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    n = len(child)
+    k = int(node.periods)
+    names, cols = [], []
+    for out_name, in_name in node.specs:
+        col = child.column(in_name)
+        ser = col.to_pandas()
+        if ctx.world > 1 and k != 0:
+            # exchange |k| boundary rows: for k>0 take the tail of preceding
+            # ranks, for k<0 the head of following ranks
+            if k > 0:
+                edge = ser.tail(min(k, n))
+            else:
+                edge = ser.head(min(-k, n))
+            edges = comm.allgather_obj(edge)
+            if k > 0:
+                pieces, have, i = [], 0, ctx.rank - 1
+                while have < k and i >= 0:
+                    t = edges[i].tail(k - have)
+                    pieces.insert(0, t)
+                    have += len(t)
+                    if len(edges[i]) >= k:
+                        break
+                    i -= 1
+                halo = pd.concat(pieces) if pieces else ser.iloc[0:0]
+                full = pd.concat([halo, ser], ignore_index=True)
+                res = full.shift(k).iloc[len(halo):].reset_index(drop=True)
+            else:
+                m = -k
+                pieces, have, i = [], 0, ctx.rank + 1
+                while have < m and i < ctx.world:
+                    t = edges[i].head(m - have)
+                    pieces.append(t)
+                    have += len(t)
+                    if len(edges[i]) >= m:
+                        break
+                    i += 1
+                halo = pd.concat(pieces) if pieces else ser.iloc[0:0]
+                full = pd.concat([ser, halo], ignore_index=True)
+                res = full.shift(k).iloc[:n].reset_index(drop=True)
+        else:
+            res = ser.shift(k)
+        names.append(out_name)
+        cols.append(Column.from_numpy(res.to_numpy(), ctx.device))
+    out = child
+    for nm, c in zip(names, cols):
+        out = out.with_column(nm, c)
+    return out
+
+
 _CUM_IDENT = {"cumsum": 0.0, "cumprod": 1.0,
               "cummin": float("inf"), "cummax": float("-inf")}
 
@@ -619,6 +670,7 @@ _HANDLERS = {
     pn.RowId: _exec_rowid,
     pn.Rolling: _exec_rolling,
     pn.Cumulative: _exec_cumulative,
+    pn.Shift: _exec_shift,
     pn.Window: _exec_window,
     pn.Join: _exec_join,
     pn.Union: _exec_union,

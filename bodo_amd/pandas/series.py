@@ -299,6 +299,17 @@ class BodoSeries:
     def cummax(self):
         return self._cum("cummax")
 
+    def shift(self, periods=1) -> "BodoSeries":
+        plan = pn.Shift(self._as_projection_plan(), int(periods),
+                        (("__sh", "v"),))
+        return BodoSeries(plan, ColRef("__sh"), self.name)
+
+    def diff(self, periods=1) -> "BodoSeries":
+        plan = pn.Shift(self._as_projection_plan(), int(periods),
+                        (("__sh", "v"),))
+        return BodoSeries(plan, BinOp("sub", ColRef("v"), ColRef("__sh")),
+                          self.name)
+
     def _as_projection_plan(self):
         return pn.Projection(self._plan, ("v",), (self._expr,))
 

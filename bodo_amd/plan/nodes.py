@@ -244,6 +244,30 @@ class Rolling(PlanNode):
 
 
 @dataclass(frozen=True)
+class Shift(PlanNode):
+    """Global-order shift: boundary rows exchange between neighboring
+    ranks (reference: dist shift in bodo/libs/array_kernels.py +
+    Sendrecv in _distributed.h)."""
+
+    child: PlanNode
+    periods: int = 1
+    specs: Tuple[Tuple[str, str], ...] = ()  # (out, in)
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Shift(ch[0], self.periods, self.specs)
+
+    def out_columns(self):
+        # passes the child's columns through plus the shifted outputs so
+        # expressions can combine original and shifted values (diff)
+        cols = self.child.out_columns()
+        outs = [s[0] for s in self.specs]
+        return None if cols is None else list(cols) + outs
+
+
+@dataclass(frozen=True)
 class Cumulative(PlanNode):
     """Global-order cumulative ops (cumsum/cumprod/cummin/cummax) over the
     distributed row order: local scan + an exscan of shard totals

@@ -333,3 +333,15 @@ def test_dist_cumsum():
     exp = df.b.cumsum().reset_index(drop=True)
     pd.testing.assert_series_equal(got.reset_index(drop=True), exp,
                                    check_names=False, check_dtype=False)
+
+
+def _q_shift(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).b.shift(3)
+
+
+def test_dist_shift():
+    df = _df(500, 29)
+    got = run_dist(_q_shift, {"df": df})
+    exp = df.b.shift(3).reset_index(drop=True)
+    pd.testing.assert_series_equal(got.reset_index(drop=True), exp,
+                                   check_names=False, check_dtype=False)

@@ -166,3 +166,14 @@ def test_cumulative(df):
     got = b.a.cumsum().to_pandas()
     pd.testing.assert_series_equal(got, df.a.cumsum().reset_index(drop=True),
                                    check_names=False, check_dtype=False)
+
+
+def test_shift_diff(df):
+    b = bpd.from_pandas(df)
+    for k in [1, 3, -2]:
+        pd.testing.assert_series_equal(
+            b.b.shift(k).to_pandas(), df.b.shift(k).reset_index(drop=True),
+            check_names=False, check_dtype=False)
+    pd.testing.assert_series_equal(
+        b.b.diff().to_pandas(), df.b.diff().reset_index(drop=True),
+        check_names=False, check_dtype=False)
