@@ -615,6 +615,34 @@ class BodoDataFrame:
             if close is not None:
                 close.close()
 
+    def _frame_reduce(self, func):
+        from ..engine import api
+
+        head = api.collect(pn.Limit(self._plan, 1))
+        num = [c for c in self._columns
+               if pd.api.types.is_numeric_dtype(head[c].dtype)]
+        plan = pn.Reduce(self._plan, tuple((c, c, func) for c in num))
+        row = api.collect(plan)
+        return pd.Series({c: row[c].iloc[0] for c in num})
+
+    def sum(self, numeric_only=True):
+        return self._frame_reduce("sum")
+
+    def mean(self, numeric_only=True):
+        return self._frame_reduce("mean")
+
+    def min(self, numeric_only=True):
+        return self._frame_reduce("min")
+
+    def max(self, numeric_only=True):
+        return self._frame_reduce("max")
+
+    def std(self, numeric_only=True, ddof=1):
+        return self._frame_reduce("std")
+
+    def var(self, numeric_only=True, ddof=1):
+        return self._frame_reduce("var")
+
     # reductions over the whole frame fall back (rare)
     def count(self):
         return self.to_pandas().count()

@@ -66,6 +66,13 @@ class DataFrameGroupBy:
         elif isinstance(arg, str):
             for col in self._value_columns():
                 aggs.append((col, col, _norm_func(arg)))
+        elif isinstance(arg, (list, tuple)):
+            for col in self._value_columns():
+                for f in arg:
+                    aggs.append((f"{col}_{f}", col, _norm_func(f)))
+        elif callable(arg):
+            for col in self._value_columns():
+                aggs.append((col, col, arg))
         else:
             raise NotImplementedError(f"agg({arg!r})")
         return self._build(aggs)
@@ -203,6 +210,15 @@ class SeriesGroupBy:
 
     def agg(self, func):
         if isinstance(func, str):
+            return self._agg1(func)
+        if isinstance(func, (list, tuple)):
+            from .frame import BodoDataFrame
+
+            aggs = tuple((f, self._col, _norm_func(f)) for f in func)
+            plan = pn.Aggregate(self._frame._lazy_plan, tuple(self._keys),
+                                aggs)
+            return BodoDataFrame(plan, list(self._keys) + list(func))
+        if callable(func):
             return self._agg1(func)
         raise NotImplementedError
 

@@ -177,3 +177,31 @@ def test_shift_diff(df):
     pd.testing.assert_series_equal(
         b.b.diff().to_pandas(), df.b.diff().reset_index(drop=True),
         check_names=False, check_dtype=False)
+
+
+def test_groupby_list_agg(df):
+    b = bpd.from_pandas(df)
+    got = b.groupby("a", as_index=False)["b"].agg(["sum", "max"]).to_pandas()
+    got = got.sort_values("a").reset_index(drop=True)
+    exp = df.groupby("a", as_index=False)["b"].agg(["sum", "max"]).sort_values(
+        "a").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_frame_reductions(df):
+    b = bpd.from_pandas(df)
+    pd.testing.assert_series_equal(b.sum(), df[["a", "b"]].sum(),
+                                   check_dtype=False)
+    pd.testing.assert_series_equal(b.mean(), df[["a", "b"]].mean(),
+                                   check_dtype=False)
+    pd.testing.assert_series_equal(b.min(), df[["a", "b"]].min(),
+                                   check_dtype=False)
+
+
+def test_series_rank(df):
+    b = bpd.from_pandas(df)
+    for m in ["average", "min", "dense"]:
+        got = b.b.rank(method=m).to_pandas()
+        exp = df.b.rank(method=m).reset_index(drop=True)
+        pd.testing.assert_series_equal(got, exp, check_names=False,
+                                       check_dtype=False)
