@@ -56,6 +56,11 @@ class ExecutionContext:
         self.device = torch.device(device)
         self.rank = comm.get_rank()
         self.world = comm.get_world_size()
+        # per-execution memo of repeated identical subtrees (reference
+        # analog: LogicalMaterializedCTE / bfs_duplicate dedup,
+        # bodo/pandas/plan.py:103): q21-style self-join queries re-derive
+        # the same scan+filter chain several times otherwise
+        self.memo: Dict[object, Table] = {}
 
 
 def execute(plan: pn.PlanNode, ctx: Optional[ExecutionContext] = None) -> Table:
@@ -71,10 +76,27 @@ def execute(plan: pn.PlanNode, ctx: Optional[ExecutionContext] = None) -> Table:
     return out
 
 
+_NO_MEMO = (pn.Sample, pn.MapPartitions, pn.ParquetWrite, pn.IcebergWrite)
+
+
+def _memo_key(node: pn.PlanNode):
+    if isinstance(node, _NO_MEMO):
+        return None
+    try:
+        hash(node)
+    except TypeError:
+        return None
+    return node
+
+
 def _exec(node: pn.PlanNode, ctx: ExecutionContext) -> Table:
     h = _HANDLERS.get(type(node))
     if h is None:
         raise NotImplementedError(f"no executor for {type(node).__name__}")
+    key = _memo_key(node)
+    memo = getattr(ctx, "memo", None)
+    if memo is not None and key is not None and key in memo:
+        return memo[key]
     from ..utils import query_profile as qp
     from ..utils import tracing
 
@@ -82,6 +104,8 @@ def _exec(node: pn.PlanNode, ctx: ExecutionContext) -> Table:
     with tracing.Event(f"exec.{name}"), qp.OpTimer(name) as t:
         out = h(node, ctx)
         t.rows_out = len(out) if out is not None else -1
+    if memo is not None and key is not None:
+        memo[key] = out
     return out
 
 
