@@ -272,3 +272,45 @@ def test_gpu_parquet_dict_int_decode(tmp_path):
     t = parquet_gpu.read_shard_gpu(p, None, ctx)
     assert t is not None
     pd.testing.assert_frame_equal(t.to_pandas(), df, check_dtype=False)
+
+
+@pytest.mark.gpu
+def test_gpu_cumulative_and_shift():
+    import bodo_amd.config as cfg
+
+    cfg.DEVICE = "cuda"
+    rng = np.random.default_rng(11)
+    x = rng.random(200_000)
+    x[rng.random(200_000) < 0.05] = np.nan
+    df = pd.DataFrame({"x": x})
+    b = bpd.from_pandas(df)
+    for f in ["cumsum", "cummin", "cummax"]:
+        got = getattr(b.x, f)().to_pandas()
+        exp = getattr(df.x, f)().reset_index(drop=True)
+        pd.testing.assert_series_equal(got, exp, check_names=False,
+                                       check_dtype=False)
+    pd.testing.assert_series_equal(
+        b.x.shift(5).to_pandas(), df.x.shift(5).reset_index(drop=True),
+        check_names=False, check_dtype=False)
+
+
+@pytest.mark.gpu
+def test_gpu_window_sql():
+    import bodo_amd.config as cfg
+    from bodo_amd.sql import BodoSQLContext
+
+    cfg.DEVICE = "cuda"
+    rng = np.random.default_rng(12)
+    df = pd.DataFrame({"k": rng.integers(0, 50, 100_000),
+                       "v": rng.random(100_000),
+                       "o": rng.permutation(100_000)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select k, sum(v) over (partition by k) as tot, "
+                 "row_number() over (partition by k order by o) as rn "
+                 "from t order by k, rn limit 1000").to_pandas()
+    exp = df.copy()
+    exp["tot"] = exp.groupby("k")["v"].transform("sum")
+    exp["rn"] = exp.sort_values("o").groupby("k").cumcount() + 1
+    exp = exp.sort_values(["k", "rn"]).reset_index(drop=True)[
+        ["k", "tot", "rn"]].head(1000)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
