@@ -75,10 +75,52 @@ def push_filters(node: pn.PlanNode) -> pn.PlanNode:
                     push_filters(pn.Filter(child.child, newcond)),
                     child.names, child.exprs)
         if isinstance(child, pn.Join):
+            anti = _left_isnull_to_anti(child, conjuncts)
+            if anti is not None:
+                return push_filters(anti)
             pushed = _push_into_join(child, conjuncts)
             if pushed is not None:
                 return pushed
     return node
+
+
+def _left_isnull_to_anti(join: pn.Join, conjuncts: List[Expr]):
+    """LEFT JOIN + `right_key IS NULL` filter -> ANTI join (reference
+    analog: Calcite's outer-join-to-anti rule; TPC-H q22 shape).  Only a
+    JOIN KEY being null proves non-match, so the rule keys on right_on
+    columns; surviving rows have every right column null, so they are
+    re-added as null constants to keep the output schema."""
+    from ..plan.expr import IsNull
+
+    if join.how != "left" or not join.right_on:
+        return None
+    lcols = join.left.out_columns()
+    rcols = join.right.out_columns()
+    if lcols is None or rcols is None or (set(lcols) & set(rcols)):
+        return None  # suffixed overlap: keep it simple, skip
+    hit = None
+    for i, cj in enumerate(conjuncts):
+        if isinstance(cj, IsNull) and not cj.negate and \
+                isinstance(cj.operand, ColRef) and \
+                cj.operand.name in join.right_on and \
+                cj.operand.name in rcols:
+            hit = i
+            break
+    if hit is None:
+        return None
+    from ..user_logging import log_message
+
+    log_message("Join Rewrite", "LEFT JOIN + IS NULL(right key) -> ANTI join")
+    anti = pn.Join(join.left, join.right, join.left_on, join.right_on,
+                   "anti", join.suffixes)
+    out_cols = join.out_columns()
+    exprs = tuple(ColRef(c) if c in set(lcols) else Const(None)
+                  for c in out_cols)
+    proj = pn.Projection(anti, tuple(out_cols), exprs)
+    rest = conjuncts[:hit] + conjuncts[hit + 1:]
+    if rest:
+        return pn.Filter(proj, _join_and(rest))
+    return proj
 
 
 def _push_into_join(join: pn.Join, conjuncts: List[Expr]):

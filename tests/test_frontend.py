@@ -311,3 +311,35 @@ def test_explain():
     txt = b[b.a > 3][["a", "b"]].explain()
     assert "Filter" in txt and "PandasScan" in txt
     assert isinstance(txt, str) and len(txt.splitlines()) >= 2
+
+
+def test_left_isnull_anti_rewrite():
+    """LEFT JOIN + IS NULL(right key) must plan as an ANTI join and match
+    pandas (TPC-H q22 shape)."""
+    from bodo_amd.engine.optimizer import optimize
+    from bodo_amd.plan import nodes as pn
+
+    rng = np.random.default_rng(31)
+    left = pd.DataFrame({"k": rng.integers(0, 50, 400),
+                         "v": rng.uniform(0, 1, 400)})
+    right = pd.DataFrame({"rk": rng.permutation(100)[:30],
+                          "w": rng.uniform(0, 1, 30)})
+    b = bpd.from_pandas(left).merge(bpd.from_pandas(right), left_on="k",
+                                    right_on="rk", how="left")
+    b = b[b.rk.isnull()]
+    opt = optimize(b._plan)
+
+    def find_join(n):
+        if isinstance(n, pn.Join):
+            return n
+        for c in n.children():
+            j = find_join(c)
+            if j is not None:
+                return j
+        return None
+
+    assert find_join(opt).how == "anti"
+    got = b.to_pandas().sort_values(["k", "v"]).reset_index(drop=True)
+    exp = left.merge(right, left_on="k", right_on="rk", how="left")
+    exp = exp[exp.rk.isnull()].sort_values(["k", "v"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
