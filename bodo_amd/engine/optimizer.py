@@ -114,7 +114,9 @@ def _left_isnull_to_anti(join: pn.Join, conjuncts: List[Expr]):
     anti = pn.Join(join.left, join.right, join.left_on, join.right_on,
                    "anti", join.suffixes)
     out_cols = join.out_columns()
-    exprs = tuple(ColRef(c) if c in set(lcols) else Const(None)
+    from ..core import types as _bt
+
+    exprs = tuple(ColRef(c) if c in set(lcols) else Const(None, _bt.float64)
                   for c in out_cols)
     proj = pn.Projection(anti, tuple(out_cols), exprs)
     rest = conjuncts[:hit] + conjuncts[hit + 1:]
