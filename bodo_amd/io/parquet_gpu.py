@@ -154,104 +154,152 @@ class _ChunkReader:
         f.seek(start)
         self.buf = f.read(chunk_meta.total_compressed_size)
 
-    def decode(self, device) -> Optional[Tuple[torch.Tensor, Optional[pa.Array]]]:
-        """Returns (device values or codes, dictionary or None); None if the
-        chunk needs the host fallback."""
-        if self.meta.compression != "UNCOMPRESSED":
+    def decode(self, device):
+        """Returns (device values or codes, dictionary or None, validity
+        mask or None); None if the chunk needs the host fallback.
+        SNAPPY pages decompress on host (pyarrow codec) before device
+        expansion; nulls build a device mask from the definition-level RLE
+        and scatter the dense values into place."""
+        comp = self.meta.compression
+        codec = None
+        if comp == "SNAPPY":
+            codec = pa.Codec("snappy")
+        elif comp != "UNCOMPRESSED":
             return None
         if self.phys in ("BOOLEAN", "INT96", "FIXED_LEN_BYTE_ARRAY"):
             return None  # bit-packed plain / legacy types: host fallback
-        if self.meta.statistics is not None and \
-                (self.meta.statistics.null_count or 0) > 0:
-            return None
         pos = 0
         n = len(self.buf)
         dict_vals: Optional[np.ndarray] = None
-        plain_parts: List[np.ndarray] = []
-        idx_parts: List[torch.Tensor] = []
-        total = 0
-        dev_buf = None
+        parts: List[Tuple[object, Optional[torch.Tensor], str]] = []
+        has_nulls = False
         while pos < n:
             t = ThriftCompact(self.buf, pos)
             hdr = t.read_struct()
             body = t.pos
             ptype = hdr.get(1)
             comp_size = hdr.get(3)
+            raw = self.buf[body:body + comp_size]
+            if codec is not None and ptype in (PAGE_DICT, PAGE_DATA):
+                pb = codec.decompress(
+                    raw, decompressed_size=hdr.get(2)).to_pybytes()                     if hasattr(codec.decompress(b"\x00", decompressed_size=1),
+                               "to_pybytes") else codec.decompress(
+                        raw, decompressed_size=hdr.get(2))
+            else:
+                pb = raw
+            if isinstance(pb, pa.Buffer):
+                pb = pb.to_pybytes()
             if ptype == PAGE_DICT:
                 dph = hdr.get(7, {})
                 nv = dph.get(1, 0)
-                dict_vals = self._decode_plain(body, nv)
+                dict_vals = self._decode_plain(pb, 0, nv)
             elif ptype == PAGE_DATA:
                 dph = hdr.get(5, {})
                 nv = dph.get(1, 0)
                 enc = dph.get(2, ENC_PLAIN)
-                p = body
-                # definition levels (max_def==1 when field is nullable):
-                # RLE block prefixed by 4-byte length; all-valid required
+                p = 0
+                mask_t: Optional[torch.Tensor] = None
+                n_valid = nv
                 if self._max_def() > 0:
-                    (lvl_len,) = struct.unpack_from("<i", self.buf, p)
-                    lv_runs, _ = _parse_rle_runs(self.buf, p + 4,
-                                                 p + 4 + lvl_len, 1, nv)
-                    if any(k == 1 or v != 1 for _, _, k, v in lv_runs):
-                        return None  # page has nulls -> host fallback
+                    (lvl_len,) = struct.unpack_from("<i", pb, p)
+                    lv_runs, got = _parse_rle_runs(pb, p + 4, p + 4 + lvl_len,
+                                                   1, nv)
                     p += 4 + lvl_len
+                    if got != nv:
+                        return None
+                    if any(k == 1 or v != 1 for _, _, k, v in lv_runs):
+                        # page has nulls (or bit-packed def levels): expand
+                        # the level stream on device into a validity mask
+                        import bodo_amd_kernels as K
+
+                        pb_dev = _dev_bytes(pb, device)
+                        blob = torch.from_numpy(
+                            _runs_blob(lv_runs).view(np.uint8)).to(device)
+                        lv = K.rle_expand(blob, len(lv_runs), pb_dev, 1, nv)
+                        mask_t = lv.to(torch.bool)
+                        n_valid = int(mask_t.sum().item())
+                        has_nulls = True
                 if enc == ENC_PLAIN:
-                    plain_parts.append(self._decode_plain(p, nv))
+                    if self.phys == "BYTE_ARRAY" and mask_t is not None:
+                        return None  # dense plain strings + nulls: host
+                    dense = self._decode_plain(pb, p, n_valid)
+                    parts.append((dense, mask_t, "plain"))
                 elif enc in (ENC_RLE_DICT, ENC_PLAIN_DICT):
-                    bitwidth = self.buf[p]
+                    bitwidth = pb[p]
                     if bitwidth > 24:
                         return None
-                    if dev_buf is None:
-                        padded = np.frombuffer(self.buf, dtype=np.uint8)
-                        padded = np.concatenate(
-                            [padded, np.zeros(8, dtype=np.uint8)])
-                        dev_buf = torch.from_numpy(padded).to(device)
                     if bitwidth == 0:
-                        idx_parts.append(torch.zeros(
-                            nv, dtype=torch.int32, device=device))
+                        codes = torch.zeros(n_valid, dtype=torch.int32,
+                                            device=device)
                     else:
-                        runs, got = _parse_rle_runs(
-                            self.buf, p + 1, body + comp_size, bitwidth, nv)
-                        if got != nv:
+                        runs, got = _parse_rle_runs(pb, p + 1, len(pb),
+                                                    bitwidth, n_valid)
+                        if got != n_valid:
                             return None
                         import bodo_amd_kernels as K
 
+                        pb_dev = _dev_bytes(pb, device)
                         blob = torch.from_numpy(
                             _runs_blob(runs).view(np.uint8)).to(device)
-                        idx = K.rle_expand(blob, len(runs), dev_buf,
-                                           int(bitwidth), nv)
-                        idx_parts.append(idx)
+                        codes = K.rle_expand(blob, len(runs), pb_dev,
+                                             int(bitwidth), n_valid)
+                    parts.append((codes, mask_t, "codes"))
                 else:
                     return None
-                total += nv
             pos = body + comp_size
-        if plain_parts and idx_parts:
+        kinds = {k for _, _, k in parts}
+        if len(kinds) != 1:
             return None
-        if idx_parts:
-            if dict_vals is None:
-                return None
-            codes = torch.cat(idx_parts) if len(idx_parts) > 1 else idx_parts[0]
-            return codes, dict_vals
-        if plain_parts:
-            vals = np.concatenate(plain_parts) if len(plain_parts) > 1 \
-                else plain_parts[0]
-            return torch.from_numpy(np.ascontiguousarray(vals)).to(device), None
-        return None
+        kind = kinds.pop()
+        if kind == "codes" and dict_vals is None:
+            return None
+        # assemble per page: scatter dense values into null positions
+        out_parts: List[torch.Tensor] = []
+        mask_parts: List[torch.Tensor] = []
+        for dense, mask_t, k in parts:
+            if k == "plain":
+                vals_t = torch.from_numpy(
+                    np.ascontiguousarray(dense)).to(device)
+            else:
+                vals_t = dense
+            if mask_t is None:
+                out_parts.append(vals_t)
+                mask_parts.append(torch.ones(len(vals_t), dtype=torch.bool,
+                                             device=device)
+                                  if has_nulls else None)
+            else:
+                full = torch.zeros(len(mask_t), dtype=vals_t.dtype,
+                                   device=device)
+                full[mask_t] = vals_t
+                out_parts.append(full)
+                mask_parts.append(mask_t)
+        vals = torch.cat(out_parts) if len(out_parts) > 1 else out_parts[0]
+        mask = None
+        if has_nulls:
+            mask = torch.cat(mask_parts) if len(mask_parts) > 1 \
+                else mask_parts[0]
+        return vals, dict_vals, mask
 
     def _max_def(self) -> int:
         return self.max_def
 
-    def _decode_plain(self, pos: int, nv: int):
+    def _decode_plain(self, buf: bytes, pos: int, nv: int):
         if self.phys == "BYTE_ARRAY":
             out = []
             p = pos
             for _ in range(nv):
-                (ln,) = struct.unpack_from("<i", self.buf, p)
-                out.append(self.buf[p + 4:p + 4 + ln].decode())
+                (ln,) = struct.unpack_from("<i", buf, p)
+                out.append(buf[p + 4:p + 4 + ln].decode())
                 p += 4 + ln
             return np.array(out, dtype=object)
         npdt = _PHYS_NP[self.phys]
-        return np.frombuffer(self.buf, dtype=npdt, count=nv, offset=pos)
+        return np.frombuffer(buf, dtype=npdt, count=nv, offset=pos)
+
+
+def _dev_bytes(pb: bytes, device) -> torch.Tensor:
+    arr = np.frombuffer(pb, dtype=np.uint8)
+    padded = np.concatenate([arr, np.zeros(8, dtype=np.uint8)])
+    return torch.from_numpy(padded).to(device)
 
 
 def read_shard_gpu(path: str, columns, ctx) -> Optional[Table]:
@@ -312,8 +360,11 @@ def _read_row_group_gpu(fp: str, rg: int, columns, ctx) -> Optional[Table]:
                 res = None
             if res is None:
                 return None
-            vals, dict_vals = res
-            cols.append(_to_column(vals, dict_vals, field, ctx.device))
+            vals, dict_vals, mask = res
+            col = _to_column(vals, dict_vals, field, ctx.device)
+            if mask is not None:
+                col.mask = mask
+            cols.append(col)
             out_names.append(cname)
     ordered = [n for n in names if n in out_names]
     tbl = Table(out_names, cols)
