@@ -248,11 +248,20 @@ class _ChunkReader:
                     return None
             pos = body + comp_size
         kinds = {k for _, _, k in parts}
-        if len(kinds) != 1:
+        if not kinds:
             return None
-        kind = kinds.pop()
-        if kind == "codes" and dict_vals is None:
+        if "codes" in kinds and dict_vals is None:
             return None
+        # all-dict chunks stay dictionary-encoded; chunks that switched to
+        # PLAIN mid-way (dictionary overflow on high-cardinality data)
+        # decode their dict pages to values so both page kinds concatenate
+        use_dict_col = kinds == {"codes"}
+        dict_dev = None
+        if not use_dict_col and "codes" in kinds:
+            if dict_vals is None or dict_vals.dtype == object:
+                return None  # mixed string pages: host fallback
+            dict_dev = torch.from_numpy(
+                np.ascontiguousarray(dict_vals)).to(device)
         # assemble per page: scatter dense values into null positions
         out_parts: List[torch.Tensor] = []
         mask_parts: List[torch.Tensor] = []
@@ -260,8 +269,10 @@ class _ChunkReader:
             if k == "plain":
                 vals_t = torch.from_numpy(
                     np.ascontiguousarray(dense)).to(device)
-            else:
+            elif use_dict_col:
                 vals_t = dense
+            else:
+                vals_t = dict_dev[dense.long()]
             if mask_t is None:
                 out_parts.append(vals_t)
                 mask_parts.append(torch.ones(len(vals_t), dtype=torch.bool,
@@ -278,7 +289,7 @@ class _ChunkReader:
         if has_nulls:
             mask = torch.cat(mask_parts) if len(mask_parts) > 1 \
                 else mask_parts[0]
-        return vals, dict_vals, mask
+        return vals, (dict_vals if use_dict_col else None), mask
 
     def _max_def(self) -> int:
         return self.max_def
