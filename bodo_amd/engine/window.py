@@ -32,7 +32,7 @@ TRANSFORM_AGGS = {"transform_sum": "sum", "transform_mean": "mean",
                   "transform_count": "count", "transform_size": "size"}
 
 ORDERED_FUNCS = {"row_number", "rank", "dense_rank", "shift", "cumsum",
-                 "cumcount"}
+                 "cumcount", "first_value", "last_value", "ntile"}
 
 
 def exec_window(node: pn.Window, ctx, child: Table) -> Table:
@@ -128,6 +128,14 @@ def _ordered_local(tbl: Table, keys, order_by, ascending, specs) -> Table:
             res = res.reindex(pdf.index)
         elif func == "cumsum":
             res = gbs[in_name].cumsum().reindex(pdf.index)
+        elif func in ("first_value", "last_value"):
+            which = "first" if func == "first_value" else "last"
+            res = gbs[in_name].transform(which).reindex(pdf.index)
+        elif func == "ntile":
+            k = int(arg or 1)
+            rn = gbs.cumcount().reindex(pdf.index)
+            size = gb[keys[0]].transform("size")
+            res = (rn * k) // size + 1
         else:
             raise NotImplementedError(func)
         out_cols[out_name] = res

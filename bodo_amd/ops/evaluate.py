@@ -99,6 +99,8 @@ class _Evaluator:
         return normalize_const(e.value, dtype), dtype
 
     def visit_BinOp(self, e: BinOp) -> Column:
+        if e.op == "concat":
+            return self._concat(e)
         rs = self._scalar_const(e.right)
         ls = self._scalar_const(e.left)
         if rs is not None:
@@ -109,6 +111,34 @@ class _Evaluator:
                                        reflect=True)
         a, b = self.visit(e.left), self.visit(e.right)
         return binary_arith(e.op, a, b)
+
+    def _concat(self, e: BinOp) -> Column:
+        """String concatenation (SQL || / CONCAT).  DICT column + string
+        constant stays device-resident (dictionary transform); other
+        combinations concatenate on host."""
+        import pyarrow as pa
+
+        for col_e, const_e, right in ((e.left, e.right, True),
+                                      (e.right, e.left, False)):
+            if isinstance(const_e, Const) and isinstance(const_e.value, str):
+                a = self.visit(col_e)
+                cv = const_e.value
+                if a.dtype.kind == TypeKind.DICT:
+                    vals = [None if v is None else
+                            (v + cv if right else cv + v)
+                            for v in a.dictionary.to_pylist()]
+                    return Column(a.dtype, a.data, a.mask,
+                                  dictionary=pa.array(
+                                      vals, type=pa.large_string()),
+                                  length=len(a))
+                sa = a.to_pandas().astype(object)
+                res = (sa + cv) if right else (cv + sa)
+                return Column.from_arrow(pa.Array.from_pandas(res),
+                                         self.device)
+        a, b = self.visit(e.left), self.visit(e.right)
+        sa = a.to_pandas().astype(object)
+        sb = b.to_pandas().astype(object)
+        return Column.from_arrow(pa.Array.from_pandas(sa + sb), self.device)
 
     def _subquery_scalar(self, e: Expr):
         if not isinstance(e, ScalarSubquery):

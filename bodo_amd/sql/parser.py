@@ -454,6 +454,9 @@ class Parser:
                 self.i += 1
                 left = Bin("add" if t.value == "+" else "sub", left,
                            self.parse_mul())
+            elif t and t.kind == "op" and t.value == "||":
+                self.i += 1
+                left = Bin("concat", left, self.parse_mul())
             else:
                 return left
 

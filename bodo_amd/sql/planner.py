@@ -273,6 +273,11 @@ class Planner:
                 arg_col = as_col(w.args[0], "__wa")
             if fn == "row_number":
                 spec = (out, "", "row_number", None)
+            elif fn in ("first_value", "last_value"):
+                spec = (out, arg_col, fn, None)
+            elif fn == "ntile":
+                k = int(w.args[0].value) if w.args else 1
+                spec = (out, "", "ntile", k)
             elif fn in ("rank", "dense_rank"):
                 spec = (out, "", fn, None)
             elif fn in ("lag", "lead"):
@@ -959,6 +964,12 @@ class Planner:
                                 (-n, None, 1))
             if name == "initcap":
                 return ex.StrOp(self.expr(e.args[0], scope), "title")
+            if name == "concat":
+                args = [self.expr(a, scope) for a in e.args]
+                out = args[0]
+                for a in args[1:]:
+                    out = ex.BinOp("concat", out, a)
+                return out
             raise NotImplementedError(f"SQL function {name}")
         raise NotImplementedError(f"expr {e}")
 

@@ -313,3 +313,37 @@ def test_scalar_functions():
             g = g.astype(object)
         pd.testing.assert_series_equal(g, exp[c], check_dtype=False,
                                        check_names=False)
+
+
+def test_string_concat():
+    rng = np.random.default_rng(8)
+    df = pd.DataFrame({"a": rng.choice(["x", "y", "z"], 30),
+                       "b": [f"n{i}" for i in range(30)]})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select a || '-' || b as ab, concat(b, ':', a) as ba "
+                 "from t").to_pandas()
+    for c in got.columns:
+        if got[c].dtype.name == "category":
+            got[c] = got[c].astype(str)
+    exp = pd.DataFrame({"ab": df.a + "-" + df.b, "ba": df.b + ":" + df.a})
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_window_first_last_ntile():
+    rng = np.random.default_rng(9)
+    df = pd.DataFrame({"k": rng.integers(0, 4, 60), "v": rng.random(60),
+                       "o": rng.permutation(60)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql(
+        "select k, o, first_value(v) over (partition by k order by o) as fv, "
+        "last_value(v) over (partition by k order by o) as lv, "
+        "ntile(3) over (partition by k order by o) as nt "
+        "from t order by k, o").to_pandas()
+    sdf = df.sort_values(["k", "o"]).reset_index(drop=True)
+    exp = sdf[["k", "o"]].copy()
+    exp["fv"] = sdf.groupby("k")["v"].transform("first")
+    exp["lv"] = sdf.groupby("k")["v"].transform("last")
+    rn = sdf.groupby("k").cumcount()
+    size = sdf.groupby("k")["v"].transform("size")
+    exp["nt"] = (rn * 3) // size + 1
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
