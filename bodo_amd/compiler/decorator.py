@@ -19,6 +19,31 @@ _KNOWN_OPTIONS = {
 }
 
 
+def _df_lib_call(fn, args, kwargs):
+    """DataFrame-library execution of a jitted function (reference:
+    check_func df-lib mode, bodo/tests/utils.py:236-243): globals bound to
+    the real pandas module rebind to bodo_amd.pandas for the call, and
+    pandas DataFrame arguments become lazy distributed frames, so the
+    function body plans through the engine instead of eager pandas."""
+    import pandas as real_pd
+
+    import bodo_amd.pandas as bpd
+
+    g = getattr(fn, "__globals__", None)
+    replaced = []
+    if isinstance(g, dict):
+        replaced = [k for k, v in g.items() if v is real_pd]
+    conv = [bpd.from_pandas(a) if isinstance(a, real_pd.DataFrame) else a
+            for a in args]
+    try:
+        for k in replaced:
+            g[k] = bpd
+        return fn(*conv, **kwargs)
+    finally:
+        for k in replaced:
+            g[k] = real_pd
+
+
 def jit(fn=None, **options):
     """Accepts the reference's jit option surface (decorators.py:183-230);
     the df-library execution model makes most of them no-ops here, while
@@ -57,7 +82,7 @@ def jit(fn=None, **options):
                         r0["res_id"], r0["names"], r0["length"]))
                     return out
                 return r0.get("value")
-        return fn(*args, **kwargs)
+        return _df_lib_call(fn, args, kwargs)
 
     wrapper._is_bodo_jit = True
     wrapper.py_func = fn

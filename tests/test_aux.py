@@ -156,3 +156,31 @@ def test_bodo_alias_package():
         return d[d.a > 7]
 
     assert len(f(b).to_pandas()) == 2
+
+
+def test_jit_df_lib_substitution():
+    """@jit rebinds pandas globals to the lazy engine for the call and
+    converts DataFrame args to distributed frames (reference: df-lib mode,
+    bodo/tests/utils.py check_func)."""
+    import numpy as np
+    import pandas as pd
+
+    import bodo_amd
+
+    @bodo_amd.jit
+    def pipeline(df):
+        f = df[df.a > 3]
+        return f.groupby("c", as_index=False).agg(s=pd.NamedAgg("b", "sum"))
+
+    rng = np.random.default_rng(0)
+    df = pd.DataFrame({"a": rng.integers(0, 10, 500), "b": rng.random(500),
+                       "c": rng.choice(["x", "y"], 500)})
+    out = pipeline(df)
+    assert type(out).__name__ == "BodoDataFrame"
+    got = out.to_pandas()
+    got["c"] = got["c"].astype(str)
+    got = got.sort_values("c").reset_index(drop=True)
+    exp = df[df.a > 3].groupby("c", as_index=False).agg(
+        s=("b", "sum")).sort_values("c").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    assert pd.__name__ == "pandas"  # globals restored after the call
