@@ -625,11 +625,32 @@ def _str_dict_generic(a: Column, op: str, args, kwargs) -> Column:
     import pyarrow as pa
 
     vals = pd.Series(a.dictionary.to_pylist(), dtype="object")
-    res = getattr(vals.str, op)(*args, **(kwargs or {}))
+    if op == "split_get":
+        pat, idx = args
+        res = vals.str.split(pat).str.get(idx)
+    else:
+        res = getattr(vals.str, op)(*args, **(kwargs or {}))
     if res.dtype == object or isinstance(res.dtype, pd.StringDtype):
         out_vals = res.tolist()
         if any(isinstance(v, (list, tuple)) for v in out_vals):
             raise NotImplementedError(f"str.{op} returns lists")
+        # missing results (e.g. split().get(i) past the end) come back as
+        # float NaN: they become validity-mask nulls, never dictionary
+        # entries (arrow dicts reject null categories)
+        out_vals = [v if isinstance(v, str) else None for v in out_vals]
+        if any(v is None for v in out_vals):
+            uniq = list(dict.fromkeys(v for v in out_vals if v is not None))
+            code_of = {v: i for i, v in enumerate(uniq)}
+            remap = torch.tensor([code_of.get(v, 0) for v in out_vals],
+                                 dtype=torch.int32, device=a.device)
+            null_lut = torch.tensor([v is None for v in out_vals],
+                                    dtype=torch.bool, device=a.device)
+            codes = remap[a.data.long()]
+            nulls = null_lut[a.data.long()]
+            mask = ~nulls if a.mask is None else (a.mask & ~nulls)
+            return Column(a.dtype, codes, mask,
+                          dictionary=pa.array(uniq, type=pa.large_string()),
+                          length=len(a))
         uniq = list(dict.fromkeys(out_vals))
         if len(uniq) != len(out_vals):
             code_of = {v: i for i, v in enumerate(uniq)}
@@ -723,7 +744,11 @@ def str_op(a: Column, op: str, args, kwargs=None) -> Column:
         import pyarrow as pa
 
         ser = arr.to_pandas().astype("object")
-        res_s = getattr(ser.str, op)(*args, **(kwargs or {}))
+        if op == "split_get":
+            pat, idx = args
+            res_s = ser.str.split(pat).str.get(idx)
+        else:
+            res_s = getattr(ser.str, op)(*args, **(kwargs or {}))
         if any(isinstance(v, (list, tuple)) for v in res_s.head(64).tolist()):
             raise NotImplementedError(f"str.{op} returns lists")
         return Column.from_arrow(pa.Array.from_pandas(res_s), a.device)

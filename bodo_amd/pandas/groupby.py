@@ -123,6 +123,20 @@ class DataFrameGroupBy:
             return ser
         return out
 
+    def head(self, n=5):
+        """First n rows per group in encounter order (row_number window +
+        filter; reference: groupby.head)."""
+        from .frame import BodoDataFrame
+        from ..plan.expr import Cmp, ColRef, Const
+
+        w = pn.Window(self._frame._lazy_plan, tuple(self._keys), (), (),
+                      (("__rn", "", "row_number", None),))
+        filt = pn.Filter(w, Cmp("le", ColRef("__rn"), Const(int(n))))
+        cols = list(self._frame._columns)
+        proj = pn.Projection(filt, tuple(cols),
+                             tuple(ColRef(c) for c in cols))
+        return BodoDataFrame(proj, cols)
+
     def _window(self, specs, order_by=(), ascending=()):
         from .frame import BodoDataFrame
 

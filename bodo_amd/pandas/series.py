@@ -513,9 +513,15 @@ class _StrAccessor:
             StrOp(self._s._expr, "replace", (pat, repl),
                   (("regex", bool(regex)),)), None)
 
+    def split(self, pat=None, n=-1, regex=None):
+        """Returns a handle supporting .str.get(i) / .str[i], lowered to a
+        single fused split+get pass (list-valued intermediates never
+        materialize; reference: BodoSQL split_part kernel)."""
+        return _SplitResult(self._s, pat)
+
     # methods whose pandas result is a list/frame per row (need explode
     # semantics) fall back via the Series-level pandas fallback
-    _UNSUPPORTED = {"split", "rsplit", "extract", "extractall", "findall",
+    _UNSUPPORTED = {"rsplit", "extract", "extractall", "findall",
                     "get_dummies", "partition", "rpartition", "cat"}
 
     def __getattr__(self, op):
@@ -550,6 +556,25 @@ def _pd_dtype_to_bodo(dtype):
     if nd.kind == "M":
         return bt.timestamp_ns
     return bt.from_numpy_dtype(nd)
+
+
+class _SplitResult:
+    """Lazy result of .str.split(pat): only element access is supported,
+    fused into one split_get string op."""
+
+    def __init__(self, s: BodoSeries, pat):
+        self._s = s
+        self._pat = pat
+
+    @property
+    def str(self):
+        return self
+
+    def get(self, i):
+        return self._s._wrap(
+            StrOp(self._s._expr, "split_get", (self._pat, int(i))), None)
+
+    __getitem__ = get
 
 
 class _RollingSeries:

@@ -205,3 +205,24 @@ def test_series_rank(df):
         exp = df.b.rank(method=m).reset_index(drop=True)
         pd.testing.assert_series_equal(got, exp, check_names=False,
                                        check_dtype=False)
+
+
+def test_str_split_get(df):
+    b = bpd.from_pandas(df)
+    src = pd.DataFrame({"s": ["a-b-c", "x-y", "solo", None] * 50})
+    bs = bpd.from_pandas(src)
+    got = bs.s.str.split("-").str.get(1).to_pandas()
+    exp = src.s.str.split("-").str.get(1)
+    got = _decat(got).where(lambda x: x.notna(), np.nan)
+    exp = exp.where(exp.notna(), np.nan)
+    pd.testing.assert_series_equal(got, exp, check_names=False,
+                                   check_dtype=False)
+
+
+def test_groupby_head(df):
+    b = bpd.from_pandas(df)
+    got = _decat(b.groupby("a").head(3).to_pandas())
+    got = got.sort_values(["a", "b"]).reset_index(drop=True)
+    exp = df.groupby("a").head(3).sort_values(["a", "b"]).reset_index(
+        drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
