@@ -323,7 +323,29 @@ def _row_greater(df: pd.DataFrame, row, keys, asc, na_position) -> np.ndarray:
     return gt
 
 
+_TOPK_MAX = 1_000_000
+
+
 def _exec_limit(node: pn.Limit, ctx) -> Table:
+    # top-k fast path: Limit(Sort) with small n computes each rank's local
+    # top-k and resolves globally on rank-replicated candidates — no
+    # range-partition exchange of the full table (reference analog: TopN,
+    # bodo/pandas/plan.py LogicalTopN:474)
+    if isinstance(node.child, pn.Sort) and not node.tail and \
+            node.offset == 0 and 0 < node.n <= _TOPK_MAX and ctx.world > 1:
+        srt = node.child
+        child = _exec(srt.child, ctx)
+        asc = list(srt.ascending) or [True] * len(srt.keys)
+        cols = [child.column(k) for k in srt.keys]
+        idx = ops.sort_indices(cols, asc, srt.na_position)
+        local_top = ops.take_table(child, idx[:node.n])
+        full = comm.allgather_table(local_top)
+        cols = [full.column(k) for k in srt.keys]
+        idx = ops.sort_indices(cols, asc, srt.na_position)
+        merged = ops.take_table(full, idx[:node.n])
+        # keep the result distributed: this rank's block of the k rows
+        start, stop = _block_bounds(len(merged), ctx.world, ctx.rank)
+        return ops.slice_table(merged, start, stop)
     child = _exec(node.child, ctx)
     n = node.n
     if ctx.world == 1:

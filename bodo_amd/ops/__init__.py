@@ -81,8 +81,42 @@ def slice_table(tbl: Table, start: int, stop: int) -> Table:
 # concat
 # ----------------------------------------------------------------------
 
+def _is_all_null(c: Column) -> bool:
+    if len(c) == 0:
+        return True
+    if c.mask is not None:
+        return not bool(c.mask.any().item())
+    if c.dtype.is_float:
+        return bool(torch.isnan(c.data).all().item())
+    return False
+
+
+def _null_like(templ: Column, n: int) -> Column:
+    """An all-null column with the template's dtype/shape (grouping-set
+    arms where a key is rolled up)."""
+    dev = templ.device
+    mask = torch.zeros(n, dtype=torch.bool, device=dev)
+    if templ.dtype.kind == TypeKind.STRING:
+        return Column(bt.string, torch.zeros(0, dtype=torch.uint8, device=dev),
+                      mask, offsets=torch.zeros(n + 1, dtype=torch.int64,
+                                                device=dev), length=n)
+    if templ.dtype.kind == TypeKind.DICT:
+        return Column(bt.dictionary,
+                      torch.zeros(n, dtype=torch.int32, device=dev), mask,
+                      dictionary=templ.dictionary, length=n)
+    return Column(templ.dtype, torch.zeros(n, dtype=templ.data.dtype,
+                                           device=dev), mask)
+
+
 def concat_columns(cols: Sequence[Column]) -> Column:
     cols = [c for c in cols]
+    kinds = {c.dtype.kind for c in cols}
+    if len(kinds) > 1:
+        # mixed kinds: only sensible when the odd ones are all-null
+        # (UNION of grouping-set arms) — replace them with typed nulls
+        templ = next((c for c in cols if not _is_all_null(c)), cols[0])
+        cols = [c if c.dtype.kind == templ.dtype.kind
+                else _null_like(templ, len(c)) for c in cols]
     first = cols[0]
     if len(cols) == 1:
         return first

@@ -212,6 +212,7 @@ class Query:
     having: Optional[Any]
     order_by: List[Tuple[Any, bool]]
     limit: Optional[int]
+    grouping_sets: Optional[List[List[Any]]] = None
 
 
 class Parser:
@@ -319,11 +320,56 @@ class Parser:
                 joins.append(JoinClause(kind, tr, on))
         where = self.parse_expr() if self.accept_kw("where") else None
         group_by: List[Any] = []
+        grouping_sets = None
         if self.accept_kw("group"):
             self.expect_kw("by")
-            group_by.append(self.parse_expr())
-            while self.accept_op(","):
+            t = self.peek()
+            word = t.value.lower() if t and t.kind == "id" else ""
+            nxt = self.toks[self.i + 1] if self.i + 1 < len(self.toks) else None
+            if word in ("rollup", "cube") and nxt and nxt.kind == "op" \
+                    and nxt.value == "(":
+                self.i += 1
+                self.expect_op("(")
                 group_by.append(self.parse_expr())
+                while self.accept_op(","):
+                    group_by.append(self.parse_expr())
+                self.expect_op(")")
+                if word == "rollup":
+                    grouping_sets = [group_by[:i]
+                                     for i in range(len(group_by), -1, -1)]
+                else:  # cube: the full powerset
+                    import itertools as _it
+
+                    grouping_sets = [list(c) for r in
+                                     range(len(group_by), -1, -1)
+                                     for c in _it.combinations(group_by, r)]
+            elif word == "grouping" and nxt and nxt.kind == "id" \
+                    and nxt.value.lower() == "sets":
+                self.i += 2
+                self.expect_op("(")
+                grouping_sets = []
+                while True:
+                    self.expect_op("(")
+                    gset = []
+                    if not self.accept_op(")"):
+                        gset.append(self.parse_expr())
+                        while self.accept_op(","):
+                            gset.append(self.parse_expr())
+                        self.expect_op(")")
+                    grouping_sets.append(gset)
+                    if not self.accept_op(","):
+                        break
+                self.expect_op(")")
+                seen = []
+                for gset in grouping_sets:
+                    for g in gset:
+                        if repr(g) not in {repr(x) for x in seen}:
+                            seen.append(g)
+                group_by = seen
+            else:
+                group_by.append(self.parse_expr())
+                while self.accept_op(","):
+                    group_by.append(self.parse_expr())
         having = self.parse_expr() if self.accept_kw("having") else None
         order_by: List[Tuple[Any, bool]] = []
         if self.accept_kw("order"):
@@ -342,7 +388,7 @@ class Parser:
         if self.accept_kw("limit"):
             limit = int(self.next().value)
         return Query(items, distinct, table, joins, where, group_by, having,
-                     order_by, limit)
+                     order_by, limit, grouping_sets)
 
     def parse_table_ref(self) -> TableRef:
         if self.accept_op("("):

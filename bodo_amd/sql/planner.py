@@ -178,7 +178,9 @@ class Planner:
             or (q.having is not None and _has_agg(q.having)) or q.group_by
         out_names: List[str] = []
         out_exprs: List[ex.Expr] = []
-        if has_agg:
+        if has_agg and getattr(q, "grouping_sets", None) is not None:
+            plan, out_names = self._plan_grouping_sets(plan, scope, q)
+        elif has_agg:
             plan, scope2, key_map, agg_map = self._aggregate(
                 plan, scope, q)
             # build output projection over the agg result
@@ -328,6 +330,56 @@ class Planner:
         for (keys, order, asc), specs in groups.items():
             plan = pn.Window(plan, keys, order, asc, tuple(specs))
         return plan
+
+    def _plan_grouping_sets(self, plan, scope: Scope, q: ast.Query):
+        """ROLLUP / CUBE / GROUPING SETS: one aggregate per grouping set,
+        rolled-up keys project as typed nulls, arms UNION ALL (reference:
+        BodoSQL grouping-sets lowering via Calcite Aggregate.groupSets)."""
+        import copy
+
+        full_keys = {_ast_key(g) for g in q.group_by}
+        gall = None
+        arms = []
+        out_names: List[str] = []
+        for gset in q.grouping_sets:
+            q2 = copy.copy(q)
+            q2.grouping_sets = None
+            q2.group_by = list(gset)
+            arm_plan = plan
+            if not gset:
+                # grand-total arm: constant grouping key, dropped on output
+                if gall is None:
+                    gall = self._uniq("__gall")
+                    scope.add("", gall, gall)
+                cols = [e[2] for e in scope.entries if e[2] != gall]
+                arm_plan = pn.Projection(
+                    plan, tuple(cols + [gall]),
+                    tuple([ex.ColRef(c) for c in cols] + [ex.Const(1)]))
+                q2.group_by = [ast.Col(None, gall)]
+            aplan, scope2, key_map, agg_map = self._aggregate(
+                arm_plan, scope, q2)
+            if q.having is not None:
+                aplan = pn.Filter(aplan, self._post_agg_expr(
+                    q.having, key_map, agg_map, scope))
+            names, exprs = [], []
+            for it in q.items:
+                if it.star:
+                    raise ValueError("SELECT * with GROUPING SETS")
+                name = it.alias or _default_name(it.expr)
+                k = _ast_key(it.expr)
+                if k in full_keys and k not in key_map:
+                    e2 = ex.Const(None, bt.float64)
+                else:
+                    try:
+                        e2 = self._post_agg_expr(it.expr, key_map, agg_map,
+                                                 scope)
+                    except KeyError:
+                        e2 = ex.Const(None, bt.float64)
+                names.append(name)
+                exprs.append(e2)
+            arms.append(pn.Projection(aplan, tuple(names), tuple(exprs)))
+            out_names = names
+        return pn.Union(tuple(arms), False), out_names
 
     def _plan_setop(self, q: "ast.SetOpQ"):
         """UNION [ALL] / INTERSECT / EXCEPT.  A trailing ORDER BY/LIMIT

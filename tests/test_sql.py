@@ -347,3 +347,33 @@ def test_window_first_last_ntile():
     size = sdf.groupby("k")["v"].transform("size")
     exp["nt"] = (rn * 3) // size + 1
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_rollup_cube_grouping_sets():
+    rng = np.random.default_rng(11)
+    df = pd.DataFrame({"a": rng.choice(["x", "y"], 60),
+                       "b": rng.integers(0, 3, 60),
+                       "v": rng.random(60)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select a, b, sum(v) as s from t group by rollup(a, b) "
+                 "order by a, b").to_pandas()
+    lvl2 = df.groupby(["a", "b"], as_index=False)["v"].sum().rename(
+        columns={"v": "s"})
+    lvl1 = df.groupby(["a"], as_index=False)["v"].sum().rename(
+        columns={"v": "s"})
+    lvl1["b"] = np.nan
+    lvl0 = pd.DataFrame({"a": [None], "b": [np.nan], "s": [df.v.sum()]})
+    exp = pd.concat([lvl2, lvl1, lvl0], ignore_index=True)[["a", "b", "s"]]
+    got["a"] = got["a"].astype(object).where(lambda x: x.notna(), None)
+    got = got.sort_values(["a", "b"], na_position="last").reset_index(
+        drop=True)
+    exp = exp.sort_values(["a", "b"], na_position="last").reset_index(
+        drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    # cube has 4 arm shapes; grand total appears once
+    cu = bc.sql("select a, b, count(*) as n from t "
+                "group by cube(a, b)").to_pandas()
+    assert cu.n.sum() == 4 * len(df)
+    gs = bc.sql("select a, sum(v) as s from t "
+                "group by grouping sets ((a), ()) order by a").to_pandas()
+    assert len(gs) == 3
