@@ -549,6 +549,8 @@ _DT_OUT_TYPE = {
     "is_quarter_start": bt.boolean, "is_quarter_end": bt.boolean,
     "is_year_start": bt.boolean, "is_year_end": bt.boolean,
     "days_in_month": bt.int8, "daysinmonth": bt.int8,
+    "trunc_month": bt.timestamp_ns, "trunc_year": bt.timestamp_ns,
+    "trunc_quarter": bt.timestamp_ns, "trunc_week": bt.timestamp_ns,
 }
 
 NS_PER_DAY = 86400 * 10**9
@@ -580,6 +582,11 @@ def _dt_field_cpu(a: Column, fld: str) -> Column:
         dtype = bt.date32
     elif fld in ("normalize", "floor_day"):
         out = idx.normalize().asi8
+        dtype = bt.timestamp_ns
+    elif fld.startswith("trunc_"):
+        unit = {"trunc_month": "M", "trunc_year": "Y", "trunc_quarter": "Q",
+                "trunc_week": "W"}[fld]
+        out = idx.to_period(unit).to_timestamp().asi8
         dtype = bt.timestamp_ns
     else:
         attr = "dayofweek" if fld == "weekday" else fld

@@ -465,6 +465,27 @@ class _DtAccessor:
             UdfMap(DtField(self._s._expr, "dayofweek"),
                    lambda d: calendar.day_name[int(d)], None), None)
 
+    def to_period(self, freq):
+        fld = {"M": "trunc_month", "Y": "trunc_year", "Q": "trunc_quarter",
+               "W": "trunc_week", "D": "floor_day"}.get(freq)
+        if fld is None:
+            raise NotImplementedError(f"to_period({freq!r})")
+        return self._f(fld)
+
+    def __getattr__(self, name):
+        """Anything else materializes through the pandas .dt accessor with
+        a fallback warning (mirrors the Series-level fallback)."""
+        if name.startswith("_"):
+            raise AttributeError(name)
+        import warnings
+
+        host = self._s.to_pandas().dt
+        if not hasattr(host, name):
+            raise AttributeError(name)
+        warnings.warn(f"Series.dt.{name}: falling back to pandas",
+                      stacklevel=2)
+        return getattr(host, name)
+
 
 class _StrAccessor:
     def __init__(self, s: BodoSeries):

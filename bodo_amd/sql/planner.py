@@ -915,7 +915,9 @@ class Planner:
                 raise NotImplementedError(f"CAST to {e.to}")
             return ex.Cast(self.expr(e.operand, scope), CAST_TYPES[e.to])
         if isinstance(e, ast.ExtractE):
-            return ex.DtField(self.expr(e.operand, scope), e.fld)
+            fld = {"dow": "dayofweek", "doy": "dayofyear",
+                   "isodow": "dayofweek"}.get(e.fld, e.fld)
+            return ex.DtField(self.expr(e.operand, scope), fld)
         if isinstance(e, ast.CaseE):
             conds = tuple(self.expr(c, scope) for c, _ in e.whens)
             thens = tuple(self.expr(v, scope) for _, v in e.whens)
@@ -1016,6 +1018,26 @@ class Planner:
                                 (-n, None, 1))
             if name == "initcap":
                 return ex.StrOp(self.expr(e.args[0], scope), "title")
+            if name == "date_trunc":
+                unit = e.args[0].value.lower().rstrip("s")
+                fld = {"day": "floor_day", "month": "trunc_month",
+                       "year": "trunc_year", "quarter": "trunc_quarter",
+                       "week": "trunc_week"}[unit]
+                return ex.DtField(self.expr(e.args[1], scope), fld)
+            if name in ("current_date", "current_timestamp", "now",
+                        "getdate"):
+                # SPMD-stable: every rank must see the SAME constant, so the
+                # value broadcasts from rank 0 when a process group is up
+                import pandas as _pdm
+
+                from ..parallel import comm as _comm
+
+                v = _pdm.Timestamp.now()
+                if _comm.initialized() and _comm.get_world_size() > 1:
+                    v = _comm.allgather_obj(v)[0]
+                if name == "current_date":
+                    v = v.normalize()
+                return ex.Const(v, bt.timestamp_ns)
             if name == "concat":
                 args = [self.expr(a, scope) for a in e.args]
                 out = args[0]

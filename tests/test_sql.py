@@ -377,3 +377,22 @@ def test_rollup_cube_grouping_sets():
     gs = bc.sql("select a, sum(v) as s from t "
                 "group by grouping sets ((a), ()) order by a").to_pandas()
     assert len(gs) == 3
+
+
+def test_date_trunc_extract_dow():
+    rng = np.random.default_rng(12)
+    ts = pd.to_datetime("2021-01-01") + pd.to_timedelta(
+        rng.integers(0, 900, 80), unit="D")
+    df = pd.DataFrame({"t": ts, "v": rng.random(80)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select date_trunc('month', t) as m, sum(v) as s from t "
+                 "group by date_trunc('month', t) order by m").to_pandas()
+    exp = df.groupby(df.t.dt.to_period("M").dt.to_timestamp())["v"].sum() \
+        .reset_index()
+    exp.columns = ["m", "s"]
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    g2 = bc.sql("select extract(dow from t) as d, count(*) as n from t "
+                "group by extract(dow from t) order by d").to_pandas()
+    e2 = df.groupby(df.t.dt.dayofweek).size().reset_index()
+    e2.columns = ["d", "n"]
+    pd.testing.assert_frame_equal(g2, e2, check_dtype=False)
