@@ -32,7 +32,9 @@ TRANSFORM_AGGS = {"transform_sum": "sum", "transform_mean": "mean",
                   "transform_count": "count", "transform_size": "size"}
 
 ORDERED_FUNCS = {"row_number", "rank", "dense_rank", "shift", "cumsum",
-                 "cumcount", "first_value", "last_value", "ntile"}
+                 "cumcount", "first_value", "last_value", "ntile",
+                 "rolling_sum", "rolling_mean", "rolling_min", "rolling_max",
+                 "rolling_count"}
 
 
 def exec_window(node: pn.Window, ctx, child: Table) -> Table:
@@ -131,6 +133,12 @@ def _ordered_local(tbl: Table, keys, order_by, ascending, specs) -> Table:
         elif func in ("first_value", "last_value"):
             which = "first" if func == "first_value" else "last"
             res = gbs[in_name].transform(which).reindex(pdf.index)
+        elif func.startswith("rolling_"):
+            base = func[len("rolling_"):]
+            r = gbs[in_name].rolling(int(arg), min_periods=1)
+            res = getattr(r, base)()
+            res = res.droplevel(list(range(len(keys))))
+            res = res.reindex(pdf.index)
         elif func == "ntile":
             k = int(arg or 1)
             rn = gbs.cumcount().reindex(pdf.index)

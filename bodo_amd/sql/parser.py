@@ -161,13 +161,16 @@ class ExistsE:
 
 @dataclass
 class WindowE:
-    """func(args) OVER (PARTITION BY ... ORDER BY ...)."""
+    """func(args) OVER (PARTITION BY ... ORDER BY ... [ROWS BETWEEN ...]).
+    frame_preceding: None = no frame clause; -1 = UNBOUNDED PRECEDING;
+    n >= 0 = n PRECEDING (frame end is always CURRENT ROW)."""
 
     func: str
     args: List[Any]
     partition_by: List[Any]
     order_by: List[Tuple[Any, bool]]
     star: bool = False
+    frame_preceding: Optional[int] = None
 
 
 @dataclass
@@ -650,9 +653,25 @@ class Parser:
                             order.append((e, asc))
                             if not self.accept_op(","):
                                 break
+                    frame = None
+                    nt = self.peek()
+                    if nt and nt.kind == "id" and nt.value.lower() == "rows":
+                        self.i += 1
+                        self.expect_kw("between")
+                        ft = self.next()
+                        if ft.kind == "id" and ft.value.lower() == "unbounded":
+                            self.next()  # PRECEDING
+                            frame = -1
+                        else:
+                            frame = int(ft.value)
+                            self.next()  # PRECEDING
+                        self.expect_kw("and")
+                        cur = self.next()  # CURRENT
+                        assert cur.value.lower() == "current", cur
+                        self.next()  # ROW
                     self.expect_op(")")
                     return WindowE(t.value.lower(), args, part, order,
-                                   star=star)
+                                   star=star, frame_preceding=frame)
                 if star:
                     return Func(t.value.lower(), [], star=True)
                 return Func(t.value.lower(), args, distinct=distinct)

@@ -288,7 +288,15 @@ class Planner:
                     n = int(w.args[1].value)
                 spec = (out, arg_col, "shift", n if fn == "lag" else -n)
             elif fn in AGG_FUNCS or fn == "count":
-                if order:
+                frame = getattr(w, "frame_preceding", None)
+                if order and frame is not None and frame >= 0:
+                    # ROWS BETWEEN n PRECEDING AND CURRENT ROW
+                    base = AGG_FUNCS.get(fn, fn)
+                    if base not in ("sum", "mean", "min", "max", "count"):
+                        raise NotImplementedError(
+                            f"{fn.upper()} with a window frame")
+                    spec = (out, arg_col, f"rolling_{base}", frame + 1)
+                elif order:
                     if fn == "sum":
                         spec = (out, arg_col, "cumsum", None)
                     else:

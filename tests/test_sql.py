@@ -396,3 +396,23 @@ def test_date_trunc_extract_dow():
     e2 = df.groupby(df.t.dt.dayofweek).size().reset_index()
     e2.columns = ["d", "n"]
     pd.testing.assert_frame_equal(g2, e2, check_dtype=False)
+
+
+def test_window_frames():
+    rng = np.random.default_rng(13)
+    df = pd.DataFrame({"k": rng.integers(0, 4, 60), "v": rng.random(60),
+                       "o": rng.permutation(60)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql(
+        "select k, o, sum(v) over (partition by k order by o "
+        "rows between 2 preceding and current row) as rs, "
+        "min(v) over (partition by k order by o "
+        "rows between 1 preceding and current row) as rm "
+        "from t order by k, o").to_pandas()
+    sdf = df.sort_values(["k", "o"]).reset_index(drop=True)
+    exp = sdf[["k", "o"]].copy()
+    exp["rs"] = sdf.groupby("k")["v"].rolling(
+        3, min_periods=1).sum().droplevel(0).sort_index()
+    exp["rm"] = sdf.groupby("k")["v"].rolling(
+        2, min_periods=1).min().droplevel(0).sort_index()
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
