@@ -345,3 +345,62 @@ def test_dist_shift():
     exp = df.b.shift(3).reset_index(drop=True)
     pd.testing.assert_series_equal(got.reset_index(drop=True), exp,
                                    check_names=False, check_dtype=False)
+
+
+def _q_setops_window(bpd, rank, payload):
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": bpd.from_pandas(payload["df"])})
+    return bc.sql(
+        "select a, rank() over (partition by a order by b) as rk from t "
+        "where b > 0.5 union all "
+        "select a, rank() over (partition by a order by b) as rk from t "
+        "where b <= 0.5 order by a, rk")
+
+
+def test_dist_setops_window():
+    df = _df(600, 41)
+    got = run_dist(_q_setops_window, {"df": df}).reset_index(drop=True)
+    hi = df[df.b > 0.5].copy()
+    hi["rk"] = hi.groupby("a")["b"].rank(method="min")
+    lo = df[df.b <= 0.5].copy()
+    lo["rk"] = lo.groupby("a")["b"].rank(method="min")
+    exp = pd.concat([hi, lo])[["a", "rk"]].sort_values(
+        ["a", "rk"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_rollup(bpd, rank, payload):
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": bpd.from_pandas(payload["df"])})
+    return bc.sql("select c, sum(b) as s from t "
+                  "group by rollup(c) order by c")
+
+
+def test_dist_rollup():
+    df = _df(500, 43)
+    got = run_dist(_q_rollup, {"df": df}).reset_index(drop=True)
+    per = df.groupby("c")["b"].sum().reset_index()
+    per.columns = ["c", "s"]
+    tot = pd.DataFrame({"c": [None], "s": [df.b.sum()]})
+    exp = pd.concat([per, tot], ignore_index=True)
+    got["c"] = got["c"].astype(object).where(lambda x: x.notna(), None)
+    key = got["c"].astype(str)
+    got = got.iloc[key.argsort().to_numpy()].reset_index(drop=True)
+    key = exp["c"].astype(str)
+    exp = exp.iloc[key.argsort().to_numpy()].reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_topk(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).sort_values(
+        "b", ascending=False).head(25)
+
+
+def test_dist_topk():
+    df = _df(3000, 47)
+    got = run_dist(_q_topk, {"df": df}).reset_index(drop=True)
+    exp = df.sort_values("b", ascending=False).head(25).reset_index(drop=True)
+    got["c"] = got["c"].astype(str)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
