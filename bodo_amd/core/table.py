@@ -62,7 +62,24 @@ class Table:
         return self.columns[0].device
 
     def to_device(self, device) -> "Table":
-        return Table(self.names, [c.to_device(device) for c in self.columns], self._length)
+        dev = torch.device(device)
+        if dev == self.device or (dev.type == self.device.type
+                                  and dev.type == "cuda"):
+            return self
+        # device copies cache per target: repeated query runs over the same
+        # registered source must not re-upload the table each execution
+        cache = getattr(self, "_dev_cache", None)
+        if cache is None:
+            cache = {}
+            object.__setattr__(self, "_dev_cache", cache)
+        key = str(dev)
+        hit = cache.get(key)
+        if hit is not None:
+            return hit
+        out = Table(self.names,
+                    [c.to_device(device) for c in self.columns], self._length)
+        cache[key] = out
+        return out
 
     def nbytes(self) -> int:
         return sum(c.nbytes() for c in self.columns)

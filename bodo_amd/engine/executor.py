@@ -131,7 +131,10 @@ def _exec_pandas_scan(node: pn.PandasScan, ctx) -> Table:
     if isinstance(obj, Table):
         tbl = obj
     else:
+        # convert once: repeated executions over the same registered source
+        # (one per query run otherwise) were ~50ms of arrow round-trips
         tbl = Table.from_pandas(obj)
+        _OBJECT_REGISTRY[node.data_id] = tbl
     # replicated host data: slice this rank's block (1D distribution)
     n = len(tbl)
     w, r = ctx.world, ctx.rank
