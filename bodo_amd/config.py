@@ -74,3 +74,8 @@ def default_device() -> str:
 #: unlimited on CPU).  See engine/ooc.py — partition-splitting hash operators
 #: with host-DRAM staging when a local groupby/join exceeds the budget.
 OOC_BYTES = _env_int("BODO_AMD_OOC_BYTES", 0)
+
+#: synchronize the device after every operator and surface async HIP errors
+#: at the operator that caused them (reference analog: DEBUG_PIPELINE /
+#: compute-sanitizer harness)
+DEBUG_SYNC = _env_bool("BODO_AMD_DEBUG_SYNC", False)

@@ -103,6 +103,11 @@ def _exec(node: pn.PlanNode, ctx: ExecutionContext) -> Table:
     name = type(node).__name__
     with tracing.Event(f"exec.{name}"), qp.OpTimer(name) as t:
         out = h(node, ctx)
+        from .. import config as _cfg
+
+        if _cfg.DEBUG_SYNC and ctx.device.type == "cuda":
+            # surface async HIP faults at the operator that queued them
+            torch.cuda.synchronize()
         t.rows_out = len(out) if out is not None else -1
     if memo is not None and key is not None:
         memo[key] = out
