@@ -615,6 +615,23 @@ class BodoDataFrame:
             if close is not None:
                 close.close()
 
+    def corr(self, numeric_only=True) -> pd.DataFrame:
+        """Pairwise Pearson correlation of numeric columns via distributed
+        co-moment reductions."""
+        from ..engine import api
+
+        head = api.collect(pn.Limit(self._plan, 1))
+        num = [c for c in self._columns
+               if pd.api.types.is_numeric_dtype(head[c].dtype)
+               and not pd.api.types.is_bool_dtype(head[c].dtype)]
+        out = pd.DataFrame(np.eye(len(num)), index=num, columns=num)
+        for i, a in enumerate(num):
+            for b2 in num[i + 1:]:
+                v = self[a].corr(self[b2])
+                out.loc[a, b2] = v
+                out.loc[b2, a] = v
+        return out
+
     def _frame_reduce(self, func):
         from ..engine import api
 

@@ -280,6 +280,44 @@ class BodoSeries:
         ser.name = self.name
         return ser
 
+    def _pair_moments(self, other: "BodoSeries"):
+        """Distributed co-moments in ONE pass: n, Σx, Σy, Σxy, Σx², Σy²
+        (reference analog: sklearn_ext/array_kernels corr via allreduce)."""
+        from ..engine import api
+
+        x, y = self._expr, other._expr
+        # pairwise-complete observations (pandas semantics): drop rows where
+        # either side is null before the moment sums
+        flt = pn.Filter(self._plan, BoolOp(
+            "and", IsNull(x, negate=True), IsNull(y, negate=True)))
+        proj = pn.Projection(flt, ("x", "y", "xy", "xx", "yy"),
+                             (x, y, BinOp("mul", x, y), BinOp("mul", x, x),
+                              BinOp("mul", y, y)))
+        red = pn.Reduce(proj, (("n", "x", "count"), ("sx", "x", "sum"),
+                               ("sy", "y", "sum"), ("sxy", "xy", "sum"),
+                               ("sxx", "xx", "sum"), ("syy", "yy", "sum")))
+        row = api.collect(red)
+        return (float(row["n"].iloc[0]), float(row["sx"].iloc[0]),
+                float(row["sy"].iloc[0]), float(row["sxy"].iloc[0]),
+                float(row["sxx"].iloc[0]), float(row["syy"].iloc[0]))
+
+    def cov(self, other: "BodoSeries", ddof: int = 1) -> float:
+        n, sx, sy, sxy, _, _ = self._pair_moments(other)
+        if n <= ddof:
+            return float("nan")
+        return (sxy - sx * sy / n) / (n - ddof)
+
+    def corr(self, other: "BodoSeries") -> float:
+        n, sx, sy, sxy, sxx, syy = self._pair_moments(other)
+        if n < 2:
+            return float("nan")
+        cov = sxy - sx * sy / n
+        vx = sxx - sx * sx / n
+        vy = syy - sy * sy / n
+        if vx <= 0 or vy <= 0:
+            return float("nan")
+        return cov / (vx ** 0.5 * vy ** 0.5)
+
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingSeries(self, int(window), min_periods)
 

@@ -226,3 +226,12 @@ def test_groupby_head(df):
     exp = df.groupby("a").head(3).sort_values(["a", "b"]).reset_index(
         drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_corr_cov(df):
+    b = bpd.from_pandas(df)
+    exp = df[["a", "b"]].astype(float)
+    assert abs(b.a.corr(b.b) - exp.a.corr(exp.b)) < 1e-9
+    assert abs(b.a.cov(b.b) - exp.a.cov(exp.b)) < 1e-9
+    pd.testing.assert_frame_equal(b.corr(), df[["a", "b"]].corr(),
+                                  check_dtype=False, atol=1e-9)

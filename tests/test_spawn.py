@@ -94,3 +94,27 @@ def test_spawn_jit_and_scalars():
         print("SPAWN_JIT_OK", len(res))
     """)
     assert "SPAWN_JIT_OK" in out
+
+
+def test_spawn_sql():
+    """BodoSQLContext in spawn mode: the plan executes on the worker group
+    and gathers through the control plane."""
+    out = run_spawn_script("""
+        import warnings; warnings.filterwarnings("ignore")
+        import numpy as np, pandas as pd
+        from bodo_amd.sql import BodoSQLContext
+
+        rng = np.random.default_rng(15)
+        df = pd.DataFrame({"k": rng.integers(0, 6, 400),
+                           "v": rng.random(400)})
+        bc = BodoSQLContext({"t": df})
+        got = bc.sql("select k, sum(v) as s, count(*) as n from t "
+                     "group by k order by k").to_pandas()
+        exp = df.groupby("k", as_index=False).agg(
+            s=("v", "sum"), n=("v", "size")).sort_values(
+            "k").reset_index(drop=True)
+        pd.testing.assert_frame_equal(got.reset_index(drop=True), exp,
+                                      check_dtype=False)
+        print("SPAWN_SQL_OK")
+    """)
+    assert "SPAWN_SQL_OK" in out
