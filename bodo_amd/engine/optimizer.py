@@ -164,6 +164,33 @@ def _push_into_join(join: pn.Join, conjuncts: List[Expr]):
                 push_r.append(cj)
                 continue
         keep.append(cj)
+    # predicate inference for OR-of-ANDs (TPC-H q7's nation pairs:
+    # (n1=A and n2=B) or (n1=B and n2=A)): each side's implied
+    # disjunction pushes as a WEAKER derived filter while the original
+    # stays above the join
+    for cj in keep:
+        if not (isinstance(cj, BoolOp) and cj.op == "or"):
+            continue
+        disjuncts = _split_or(cj)
+        for side_set, ok, bucket in ((sl, left_ok, push_l),
+                                     (sr, right_ok, push_r)):
+            if not ok:
+                continue
+            per_disjunct = []
+            for d in disjuncts:
+                side_conj = [c for c in _split_and(d)
+                             if expr_columns(c) and
+                             expr_columns(c) <= side_set
+                             and not has_subplan(c)]
+                if not side_conj:
+                    per_disjunct = None
+                    break
+                per_disjunct.append(_join_and(side_conj))
+            if per_disjunct:
+                derived = per_disjunct[0]
+                for d in per_disjunct[1:]:
+                    derived = BoolOp("or", derived, d)
+                bucket.append(derived)
     if not push_l and not push_r:
         return None
     from ..user_logging import log_message
@@ -180,6 +207,12 @@ def _push_into_join(join: pn.Join, conjuncts: List[Expr]):
     if keep:
         return pn.Filter(new_join, _join_and(keep))
     return new_join
+
+
+def _split_or(e: Expr) -> List[Expr]:
+    if isinstance(e, BoolOp) and e.op == "or":
+        return _split_or(e.left) + _split_or(e.right)
+    return [e]
 
 
 def _split_and(e: Expr) -> List[Expr]:
