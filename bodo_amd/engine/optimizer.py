@@ -20,8 +20,23 @@ from ..plan.expr import (
 def optimize(plan: pn.PlanNode) -> pn.PlanNode:
     plan = fuse_projections(plan)
     plan = push_filters(plan)
+    plan = push_limits(plan)
     plan = prune_columns(plan, None)
     return plan
+
+
+def push_limits(node: pn.PlanNode) -> pn.PlanNode:
+    """head(n) after a computed projection evaluates the expressions over
+    only the surviving rows: Limit(Projection) -> Projection(Limit)."""
+    node = node.with_children(*[push_limits(c) for c in node.children()]) \
+        if node.children() else node
+    if isinstance(node, pn.Limit) and not node.tail and \
+            isinstance(node.child, pn.Projection):
+        proj = node.child
+        return pn.Projection(
+            pn.Limit(proj.child, node.n, node.offset, node.tail),
+            proj.names, proj.exprs)
+    return node
 
 
 # ---------------------------------------------------------------- fusion
