@@ -127,6 +127,21 @@ class BodoSeries:
         oe = other._expr if isinstance(other, BodoSeries) else as_expr(other)
         return self._wrap(Case((ce,), (self._expr,), oe), None)
 
+    def __getitem__(self, key):
+        if isinstance(key, BodoSeries):
+            # boolean-mask filter: s[s > 0]
+            filt = pn.Filter(
+                pn.Projection(self._plan, ("v", "__m"),
+                              (self._expr, key._expr)),
+                ColRef("__m"))
+            proj = pn.Projection(filt, ("v",), (ColRef("v"),))
+            return BodoSeries(proj, ColRef("v"), self.name)
+        if isinstance(key, slice) and key.start in (None, 0) and \
+                key.step in (None, 1) and key.stop is not None:
+            plan = pn.Limit(self._as_projection_plan(), key.stop)
+            return BodoSeries(plan, ColRef("v"), self.name)
+        return self.to_pandas()[key]
+
     @property
     def dt(self):
         return _DtAccessor(self)

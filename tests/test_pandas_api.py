@@ -235,3 +235,13 @@ def test_corr_cov(df):
     assert abs(b.a.cov(b.b) - exp.a.cov(exp.b)) < 1e-9
     pd.testing.assert_frame_equal(b.corr(), df[["a", "b"]].corr(),
                                   check_dtype=False, atol=1e-9)
+
+
+def test_series_getitem_mask(df):
+    b = bpd.from_pandas(df)
+    got = b.b[b.a > 5].to_pandas()
+    exp = df.b[df.a > 5].reset_index(drop=True)
+    pd.testing.assert_series_equal(got, exp, check_names=False,
+                                   check_dtype=False)
+    got2 = b.a[:7].to_pandas()
+    assert list(got2) == list(df.a[:7])
