@@ -168,8 +168,16 @@ class DataFrameGroupBy:
         def _part(pdf, *a):
             if len(pdf) == 0:
                 return pd.DataFrame()
-            res = pdf.groupby(keys, dropna=self._dropna).apply(func, *a, **kwargs)
-            return res.reset_index()
+            res = pdf.groupby(keys, dropna=self._dropna,
+                              sort=False).apply(func, *a, **kwargs)
+            if isinstance(res, pd.Series):
+                return res.reset_index()
+            # frame-returning funcs: rows already carry the key columns
+            # (pandas group_keys flattening)
+            try:
+                return res.reset_index()
+            except ValueError:
+                return res.reset_index(drop=True)
 
         from .frame import BodoDataFrame
 

@@ -245,3 +245,20 @@ def test_series_getitem_mask(df):
                                    check_dtype=False)
     got2 = b.a[:7].to_pandas()
     assert list(got2) == list(df.a[:7])
+
+
+def test_groupby_apply_frame_return(df):
+    import warnings
+
+    b = bpd.from_pandas(df)
+
+    def top2(g):
+        return g.nlargest(2, "b")
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        got = _decat(b.groupby("a").apply(top2).to_pandas())
+        got = got.sort_values(["a", "b"]).reset_index(drop=True)
+        exp = df.groupby("a", sort=False).apply(top2).reset_index(drop=True)
+    exp = exp.sort_values(["a", "b"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got[exp.columns], exp, check_dtype=False)
