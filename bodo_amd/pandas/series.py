@@ -333,6 +333,33 @@ class BodoSeries:
             return float("nan")
         return cov / (vx ** 0.5 * vy ** 0.5)
 
+    def pct_change(self, periods=1) -> "BodoSeries":
+        plan = pn.Shift(self._as_projection_plan(), int(periods),
+                        (("__sh", "v"),))
+        expr = BinOp("div", BinOp("sub", ColRef("v"), ColRef("__sh")),
+                     ColRef("__sh"))
+        return BodoSeries(plan, expr, self.name)
+
+    def duplicated(self, keep="first") -> "BodoSeries":
+        """True for repeats of an earlier (keep='first') occurrence —
+        per-value running count via the window machinery."""
+        if keep != "first":
+            raise NotImplementedError("duplicated(keep!='first')")
+        plan = pn.Window(self._as_projection_plan(), ("v",), (), (),
+                         (("__cc", "", "cumcount", None),))
+        return BodoSeries(plan, Cmp("gt", ColRef("__cc"), Const(0)),
+                          self.name)
+
+    def to_frame(self, name=None):
+        from .frame import BodoDataFrame
+
+        out_name = name or self.name or "0"
+        plan = pn.Projection(self._plan, (out_name,), (self._expr,))
+        return BodoDataFrame(plan, [out_name])
+
+    def rename(self, name, **kwargs) -> "BodoSeries":
+        return BodoSeries(self._plan, self._expr, name, self._frame)
+
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingSeries(self, int(window), min_periods)
 

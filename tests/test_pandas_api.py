@@ -262,3 +262,15 @@ def test_groupby_apply_frame_return(df):
         exp = df.groupby("a", sort=False).apply(top2).reset_index(drop=True)
     exp = exp.sort_values(["a", "b"]).reset_index(drop=True)
     pd.testing.assert_frame_equal(got[exp.columns], exp, check_dtype=False)
+
+
+def test_series_pct_change_duplicated(df):
+    b = bpd.from_pandas(df)
+    pd.testing.assert_series_equal(
+        b.b.pct_change().to_pandas(),
+        df.b.pct_change(fill_method=None).reset_index(drop=True),
+        check_names=False, check_dtype=False)
+    pd.testing.assert_series_equal(
+        b.a.duplicated().to_pandas(), df.a.duplicated().reset_index(drop=True),
+        check_names=False, check_dtype=False)
+    assert list(b.b.to_frame("x").to_pandas().columns) == ["x"]
