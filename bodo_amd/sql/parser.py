@@ -388,10 +388,17 @@ class Parser:
                 if not self.accept_op(","):
                     break
         limit = None
+        limit_offset = 0
         if self.accept_kw("limit"):
             limit = int(self.next().value)
-        return Query(items, distinct, table, joins, where, group_by, having,
-                     order_by, limit, grouping_sets)
+            nt = self.peek()
+            if nt and nt.kind == "id" and nt.value.lower() == "offset":
+                self.i += 1
+                limit_offset = int(self.next().value)
+        q = Query(items, distinct, table, joins, where, group_by, having,
+                  order_by, limit, grouping_sets)
+        q.limit_offset = limit_offset
+        return q
 
     def parse_table_ref(self) -> TableRef:
         if self.accept_op("("):
@@ -554,10 +561,18 @@ class Parser:
             if t.value in ("true", "false"):
                 return Lit(t.value == "true", "bool")
             if t.value == "case":
+                # simple form: CASE <expr> WHEN v THEN r ... rewrites each
+                # WHEN into an equality against the operand
+                operand = None
+                nt = self.peek()
+                if not (nt and nt.kind == "kw" and nt.value == "when"):
+                    operand = self.parse_expr()
                 whens = []
                 els = None
                 while self.accept_kw("when"):
                     c = self.parse_expr()
+                    if operand is not None:
+                        c = Bin("eq", operand, c)
                     self.expect_kw("then")
                     v = self.parse_expr()
                     whens.append((c, v))

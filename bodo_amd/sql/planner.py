@@ -232,7 +232,8 @@ class Planner:
                 asc.append(a)
             plan = pn.Sort(plan, tuple(keys), tuple(asc))
         if q.limit is not None:
-            plan = pn.Limit(plan, q.limit)
+            plan = pn.Limit(plan, q.limit,
+                            getattr(q, "limit_offset", 0) or 0)
         return plan, out_names
 
     # ------------------------------------------------------------------

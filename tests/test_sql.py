@@ -416,3 +416,13 @@ def test_window_frames():
     exp["rm"] = sdf.groupby("k")["v"].rolling(
         2, min_periods=1).min().droplevel(0).sort_index()
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_simple_case_and_offset():
+    df = pd.DataFrame({"a": range(20), "g": [1, 2, 3, 4] * 5})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select a from t order by a limit 5 offset 3").to_pandas()
+    assert got.a.tolist() == [3, 4, 5, 6, 7]
+    g2 = bc.sql("select g, case g when 1 then 'one' when 2 then 'two' "
+                "else 'many' end as lab from t order by a limit 4").to_pandas()
+    assert [str(x) for x in g2.lab] == ["one", "two", "many", "many"]
