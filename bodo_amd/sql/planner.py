@@ -217,7 +217,7 @@ class Planner:
             plan = pn.Distinct(plan, None)
         if q.order_by:
             keys, asc = [], []
-            extra = 0
+            hidden: List[str] = []
             for e, a in q.order_by:
                 if isinstance(e, ast.Lit) and e.kind == "num":
                     keys.append(out_names[int(e.value) - 1])
@@ -225,15 +225,38 @@ class Planner:
                         and e.name in out_names:
                     keys.append(e.name)
                 elif isinstance(e, ast.Col):
-                    keys.append(e.name)  # hope it's an output name
+                    # ORDER BY a source column not in the SELECT list:
+                    # carry it as a hidden projection column through the
+                    # sort and drop it afterwards (standard SQL)
+                    internal = None
+                    if not has_agg and not q.distinct and \
+                            isinstance(plan, pn.Projection):
+                        try:
+                            internal = scope.resolve(e.table, e.name)
+                        except KeyError:
+                            internal = None
+                    if internal is not None and internal not in hidden:
+                        hidden.append(internal)
+                        keys.append(internal)
+                    elif internal is not None:
+                        keys.append(internal)
+                    else:
+                        keys.append(e.name)  # hope it's an output name
                 else:
                     raise ValueError("ORDER BY expressions must be output "
                                      "columns or positions")
                 asc.append(a)
+            if hidden:
+                plan = pn.Projection(
+                    plan.child, plan.names + tuple(hidden),
+                    plan.exprs + tuple(ex.ColRef(h) for h in hidden))
             plan = pn.Sort(plan, tuple(keys), tuple(asc))
         if q.limit is not None:
             plan = pn.Limit(plan, q.limit,
                             getattr(q, "limit_offset", 0) or 0)
+        if q.order_by and hidden:
+            plan = pn.Projection(plan, tuple(out_names),
+                                 tuple(ex.ColRef(n) for n in out_names))
         return plan, out_names
 
     # ------------------------------------------------------------------
