@@ -476,6 +476,45 @@ def _exec_shift(node: pn.Shift, ctx) -> Table:
     return out
 
 
+def _exec_fill(node: "pn.Fill", ctx) -> Table:
+    """ffill/bfill over the global row order: local fill plus the nearest
+    valid value carried across rank boundaries (reference: array_kernels
+    fillna forward/backward with border exchange)."""
+    child = _exec(node.child, ctx)
+    names, cols = [], []
+    for out_name, in_name in node.specs:
+        ser = child.column(in_name).to_pandas()
+        filled = ser.ffill() if node.forward else ser.bfill()
+        if ctx.world > 1:
+            if node.forward:
+                last = filled.iloc[-1] if len(filled) else None
+                edges = comm.allgather_obj(
+                    None if last is None or pd.isna(last) else last)
+                carry = None
+                for v in edges[:ctx.rank]:
+                    if v is not None:
+                        carry = v
+                if carry is not None:
+                    filled = filled.fillna(carry)
+            else:
+                first = filled.iloc[0] if len(filled) else None
+                edges = comm.allgather_obj(
+                    None if first is None or pd.isna(first) else first)
+                carry = None
+                for v in edges[ctx.rank + 1:]:
+                    if v is not None:
+                        carry = v
+                        break
+                if carry is not None:
+                    filled = filled.fillna(carry)
+        names.append(out_name)
+        cols.append(Column.from_numpy(filled.to_numpy(), ctx.device))
+    out = child
+    for nm, c in zip(names, cols):
+        out = out.with_column(nm, c)
+    return out
+
+
 _CUM_IDENT = {"cumsum": 0.0, "cumprod": 1.0,
               "cummin": float("inf"), "cummax": float("-inf")}
 
@@ -725,6 +764,7 @@ _HANDLERS = {
     pn.Rolling: _exec_rolling,
     pn.Cumulative: _exec_cumulative,
     pn.Shift: _exec_shift,
+    pn.Fill: _exec_fill,
     pn.Window: _exec_window,
     pn.Join: _exec_join,
     pn.Union: _exec_union,

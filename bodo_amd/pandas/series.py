@@ -333,6 +333,14 @@ class BodoSeries:
             return float("nan")
         return cov / (vx ** 0.5 * vy ** 0.5)
 
+    def ffill(self) -> "BodoSeries":
+        plan = pn.Fill(self._as_projection_plan(), True, (("v", "v"),))
+        return BodoSeries(plan, ColRef("v"), self.name)
+
+    def bfill(self) -> "BodoSeries":
+        plan = pn.Fill(self._as_projection_plan(), False, (("v", "v"),))
+        return BodoSeries(plan, ColRef("v"), self.name)
+
     def pct_change(self, periods=1) -> "BodoSeries":
         plan = pn.Shift(self._as_projection_plan(), int(periods),
                         (("__sh", "v"),))

@@ -268,6 +268,28 @@ class Shift(PlanNode):
 
 
 @dataclass(frozen=True)
+class Fill(PlanNode):
+    """Forward/backward fill over the global row order with cross-rank
+    boundary carry (reference: array_kernels fillna method='ffill')."""
+
+    child: PlanNode
+    forward: bool = True
+    specs: Tuple[Tuple[str, str], ...] = ()  # (out, in)
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Fill(ch[0], self.forward, self.specs)
+
+    def out_columns(self):
+        cols = self.child.out_columns()
+        outs = [s[0] for s in self.specs]
+        return None if cols is None else list(cols) + [o for o in outs
+                                                       if o not in cols]
+
+
+@dataclass(frozen=True)
 class Cumulative(PlanNode):
     """Global-order cumulative ops (cumsum/cumprod/cummin/cummax) over the
     distributed row order: local scan + an exscan of shard totals

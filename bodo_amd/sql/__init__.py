@@ -37,6 +37,29 @@ class BodoSQLContext:
         from .parser import Parser
         from .planner import Planner
 
+        stripped = query.strip()
+        low = stripped.lower()
+        if low.startswith("explain"):
+            from ..engine.optimizer import optimize
+            from ..plan.nodes import explain as _explain
+
+            inner = self.sql(stripped[len("explain"):])
+            return _explain(optimize(inner._plan))
+        if low.startswith("create table") or low.startswith(
+                "create or replace table"):
+            # CTAS: CREATE [OR REPLACE] TABLE name AS SELECT ...
+            # (reference: BodoSQL DDL executed directly, context.py:504)
+            import re as _re
+
+            m = _re.match(r"create\s+(?:or\s+replace\s+)?table\s+(\w+)\s+as"
+                          r"\s+(.*)$", stripped,
+                          _re.IGNORECASE | _re.DOTALL)
+            if not m:
+                raise ValueError(f"unsupported DDL: {stripped[:60]}")
+            name, select = m.group(1), m.group(2)
+            self.tables[name.lower()] = self.sql(select)
+            BodoSQLContext._plan_cache.clear()
+            return None
         key = (query, tuple(sorted((n, id(t._lazy_plan))
                                    for n, t in self.tables.items())))
         hit = BodoSQLContext._plan_cache.get(key)

@@ -404,3 +404,20 @@ def test_dist_topk():
     exp = df.sort_values("b", ascending=False).head(25).reset_index(drop=True)
     got["c"] = got["c"].astype(str)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_ffill(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).x.ffill()
+
+
+def test_dist_ffill():
+    """Boundary carry: a leading-null shard must fill from the previous
+    rank's last valid value."""
+    rng = np.random.default_rng(51)
+    x = rng.random(600)
+    x[rng.random(600) < 0.4] = np.nan
+    df = pd.DataFrame({"x": x})
+    got = run_dist(_q_ffill, {"df": df})
+    exp = df.x.ffill().reset_index(drop=True)
+    pd.testing.assert_series_equal(got.reset_index(drop=True), exp,
+                                   check_names=False)

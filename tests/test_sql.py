@@ -426,3 +426,30 @@ def test_simple_case_and_offset():
     g2 = bc.sql("select g, case g when 1 then 'one' when 2 then 'two' "
                 "else 'many' end as lab from t order by a limit 4").to_pandas()
     assert [str(x) for x in g2.lab] == ["one", "two", "many", "many"]
+
+
+def test_explain_and_ctas():
+    df = pd.DataFrame({"a": range(10), "b": np.arange(10) * 0.5})
+    bc = BodoSQLContext({"t": df})
+    txt = bc.sql("explain select a from t where b > 2 order by a")
+    assert isinstance(txt, str) and "Filter" in txt
+    bc.sql("create table big as select a, b from t where b > 2")
+    out = bc.sql("select count(*) as n from big").to_pandas()
+    assert out.n.iloc[0] == len(df[df.b > 2])
+
+
+def test_series_ffill_bfill():
+    import bodo_amd.pandas as bpd
+
+    rng = np.random.default_rng(19)
+    x = rng.random(80)
+    x[rng.random(80) < 0.3] = np.nan
+    x[:3] = np.nan
+    df = pd.DataFrame({"x": x})
+    b = bpd.from_pandas(df)
+    pd.testing.assert_series_equal(b.x.ffill().to_pandas(),
+                                   df.x.ffill().reset_index(drop=True),
+                                   check_names=False)
+    pd.testing.assert_series_equal(b.x.bfill().to_pandas(),
+                                   df.x.bfill().reset_index(drop=True),
+                                   check_names=False)
