@@ -697,8 +697,12 @@ def _exec_reduce(node: pn.Reduce, ctx) -> Table:
     if want_streaming(node, ctx):
         return exec_streaming(node, ctx)
     child = _exec(node.child, ctx)
-    partials = {out: ops.reduce_column(child.column(in_name), func)
-                for out, in_name, func in node.aggs}
+    partials = {}
+    for out, in_name, func in node.aggs:
+        if not in_name:  # COUNT(*) / size over the whole table
+            partials[out] = {"count": len(child)}
+        else:
+            partials[out] = ops.reduce_column(child.column(in_name), func)
     return _finish_reduce(node, ctx, partials)
 
 
