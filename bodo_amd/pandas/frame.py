@@ -632,6 +632,28 @@ class BodoDataFrame:
                 out.loc[b2, a] = v
         return out
 
+    def memory_usage(self, index=False, deep=False) -> pd.Series:
+        tbl = self.execute()
+        return pd.Series({n: c.nbytes() for n, c in
+                          zip(tbl.names, tbl.columns)})
+
+    def agg(self, arg):
+        """Frame-level eager aggregation: {'col': 'func'} or
+        {'col': ['f1', 'f2']}."""
+        if isinstance(arg, dict):
+            out = {}
+            for col, f in arg.items():
+                if isinstance(f, (list, tuple)):
+                    out[col] = {x: getattr(self[col], x)() for x in f}
+                else:
+                    out[col] = getattr(self[col], f)()
+            if all(not isinstance(v, dict) for v in out.values()):
+                return pd.Series(out)
+            return pd.DataFrame(out)
+        if isinstance(arg, str):
+            return self._frame_reduce(arg)
+        raise NotImplementedError(f"df.agg({arg!r})")
+
     def _frame_reduce(self, func):
         from ..engine import api
 

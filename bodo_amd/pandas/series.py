@@ -68,6 +68,12 @@ class BodoSeries:
     def __invert__(self):
         return self._wrap(Not(self._expr), None)
 
+    def __neg__(self):
+        return self._wrap(BinOp("sub", Const(0), self._expr), None)
+
+    def __abs__(self):
+        return self.abs()
+
     def __hash__(self):
         return id(self)
 
@@ -213,6 +219,16 @@ class BodoSeries:
                         name="count")
         out.index.name = self.name
         return out
+
+    def agg(self, func):
+        if isinstance(func, str):
+            return getattr(self, func)()
+        if isinstance(func, (list, tuple)):
+            return pd.Series({f: getattr(self, f)() for f in func},
+                             name=self.name)
+        raise NotImplementedError(f"Series.agg({func!r})")
+
+    aggregate = agg
 
     def prod(self):
         return self._reduce("prod")
