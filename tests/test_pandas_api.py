@@ -274,3 +274,14 @@ def test_series_pct_change_duplicated(df):
         b.a.duplicated().to_pandas(), df.a.duplicated().reset_index(drop=True),
         check_names=False, check_dtype=False)
     assert list(b.b.to_frame("x").to_pandas().columns) == ["x"]
+
+
+def test_agg_extras(df):
+    b = bpd.from_pandas(df)
+    assert list((-b.a).to_pandas()) == list(-df.a)
+    assert abs(b.b.agg("sum") - df.b.sum()) < 1e-9
+    ser = b.b.agg(["sum", "mean"])
+    assert abs(ser["sum"] - df.b.sum()) < 1e-9
+    ag = b.agg({"a": "max", "b": "sum"})
+    assert ag["a"] == df.a.max() and abs(ag["b"] - df.b.sum()) < 1e-9
+    assert b.memory_usage()["a"] > 0
