@@ -629,6 +629,11 @@ def _exec_join(node: pn.Join, ctx) -> Table:
     right = _exec(node.right, ctx)
     from . import ooc
 
+    if ctx.world > 1 and node.how != "cross" and node.left_on:
+        from . import join_filter
+
+        left, right = join_filter.apply_runtime_filters(
+            left, right, node.left_on, node.right_on, node.how, ctx)
     if ctx.world == 1 or node.how == "cross":
         if ctx.world > 1 and node.how == "cross":
             right = comm.allgather_table(right)

@@ -76,6 +76,23 @@ def barrier():
         dist.barrier()
 
 
+def allreduce_max_(t) -> None:
+    """In-place elementwise MAX all-reduce (bloom bitset merges,
+    reference: or_u64_kernel + bitset exchange in gpu_bloom_filter.cu)."""
+    if not initialized() or get_world_size() == 1:
+        return
+    import torch
+
+    wire = t
+    moved = False
+    if dist.get_backend() == "gloo" and t.is_cuda:
+        wire = t.cpu()
+        moved = True
+    dist.all_reduce(wire, op=dist.ReduceOp.MAX)
+    if moved:
+        t.copy_(wire.to(t.device))
+
+
 def allgather_obj(obj) -> list:
     if not initialized():
         return [obj]

@@ -421,3 +421,29 @@ def test_dist_ffill():
     exp = df.x.ffill().reset_index(drop=True)
     pd.testing.assert_series_equal(got.reset_index(drop=True), exp,
                                    check_names=False)
+
+
+def _q_rtf_join(bpd, rank, payload):
+    import bodo_amd.engine.join_filter as jf
+
+    jf.MIN_PROBE_ROWS = 16  # force the runtime filter at test scale
+    l = bpd.from_pandas(payload["left"])
+    r = bpd.from_pandas(payload["right"])
+    return l.merge(r, on="k", how=payload["how"]).sort_values(
+        ["k", "v1"])
+
+
+def test_dist_runtime_join_filter():
+    """Bloom + min/max probe pruning must not change any join result."""
+    rng = np.random.default_rng(53)
+    left = pd.DataFrame({"k": rng.integers(0, 100_000, 5000),
+                         "v1": rng.uniform(0, 1, 5000)})
+    right = pd.DataFrame({"k": rng.permutation(100_000)[:40],
+                          "v2": rng.uniform(0, 1, 40)})
+    for how in ("inner", "left"):
+        got = run_dist(_q_rtf_join,
+                       {"left": left, "right": right, "how": how})
+        got = got.reset_index(drop=True)
+        exp = left.merge(right, on="k", how=how).sort_values(
+            ["k", "v1"]).reset_index(drop=True)
+        pd.testing.assert_frame_equal(got, exp, check_dtype=False)
