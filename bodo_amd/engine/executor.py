@@ -240,9 +240,9 @@ def _exec_sort(node: pn.Sort, ctx) -> Table:
     asc = list(node.ascending) or [True] * len(node.keys)
     if ctx.world > 1:
         child = _range_partition(child, list(node.keys), asc, node.na_position, ctx)
-    cols = [child.column(k) for k in node.keys]
-    idx = ops.sort_indices(cols, asc, node.na_position)
-    return ops.take_table(child, idx)
+    from . import ooc
+
+    return ooc.sort_local(child, list(node.keys), asc, node.na_position)
 
 
 def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:

@@ -103,3 +103,61 @@ def join_local(left: Table, right: Table, left_on, right_on, how,
                              left_on, right_on, how, suffixes)
         outs.append(res.to_device("cpu"))
     return concat_tables([o.to_device(device) for o in outs])
+
+
+def sort_local(tbl: Table, keys, ascending, na_position) -> Table:
+    """Local sort with partition-splitting when over budget: rows range-
+    partition by sampled key bounds into host-staged runs, each run sorts
+    on device, and runs concatenate in bound order — no merge pass needed
+    (reference analog: external k-way merge sort, streaming/_sort.h:237;
+    the range split replaces the merge)."""
+    from ..ops import concat_tables, sort_indices, take_table
+
+    budget = budget_bytes(tbl.device)
+    if budget is None or tbl.nbytes() <= budget or not keys:
+        idx = sort_indices([tbl.column(k) for k in keys], ascending,
+                           na_position)
+        return take_table(tbl, idx)
+    nparts = _npartitions(tbl.nbytes(), budget)
+    device = tbl.device
+    # sample bounds on the first key only (ties stay within a run and are
+    # resolved by the per-run full-key sort; equal-key spans across run
+    # boundaries remain key-sorted because runs are contiguous ranges)
+    first = tbl.column(keys[0])
+    n = len(tbl)
+    take = min(65536, n)
+    step = max(1, n // take)
+    sample = first.data[::step]
+    order = torch.argsort(sample, stable=True)
+    svals = sample[order]
+    qs = [svals[int(i * (len(svals) - 1) / nparts)] for i in
+          range(1, nparts)]
+    asc0 = ascending[0] if ascending else True
+    part = torch.zeros(n, dtype=torch.int64, device=device)
+    for b in qs:
+        part += (first.data > b).to(torch.int64) if asc0 else \
+            (first.data < b).to(torch.int64)
+    # null/NaN first keys route to the na_position run so the per-run sort
+    # places them globally first/last
+    invalid = None
+    if first.dtype.is_float:
+        invalid = torch.isnan(first.data)
+    if first.mask is not None:
+        miss = ~first.mask
+        invalid = miss if invalid is None else (invalid | miss)
+    if invalid is not None and bool(invalid.any().item()):
+        na_run = nparts - 1 if na_position == "last" else 0
+        part = torch.where(invalid, torch.tensor(na_run, device=device),
+                           part)
+    runs = []
+    for p in range(nparts):
+        idx = torch.nonzero(part == p, as_tuple=False).reshape(-1)
+        runs.append(take_table(tbl, idx).to_device("cpu"))
+    del tbl
+    outs = []
+    for run in runs:
+        dev_run = run.to_device(device)
+        idx = sort_indices([dev_run.column(k) for k in keys], ascending,
+                           na_position)
+        outs.append(take_table(dev_run, idx).to_device("cpu"))
+    return concat_tables([o.to_device(device) for o in outs])

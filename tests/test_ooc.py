@@ -78,3 +78,25 @@ def test_budget_off_uses_single_pass(monkeypatch):
     import torch
 
     assert ooc.budget_bytes(torch.device("cpu")) is None
+
+
+def test_ooc_sort(tiny_budget):
+    df = _df(30000, 7)
+    got = bpd.from_pandas(df).sort_values(["k", "v"]).to_pandas()
+    got = got.reset_index(drop=True)
+    got["c"] = got["c"].astype(str)
+    exp = df.sort_values(["k", "v"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_ooc_sort_nans(tiny_budget):
+    import numpy as np
+
+    rng = np.random.default_rng(8)
+    v = rng.random(20000)
+    v[rng.random(20000) < 0.05] = np.nan
+    df = pd.DataFrame({"v": v, "i": np.arange(20000)})
+    got = bpd.from_pandas(df).sort_values("v").to_pandas().reset_index(
+        drop=True)
+    exp = df.sort_values("v").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got[["v"]], exp[["v"]], check_dtype=False)
