@@ -212,7 +212,8 @@ class BodoSeries:
 
         plan = pn.Aggregate(self._as_projection_plan(), ("v",),
                             (("count", "v", "size"),), dropna=dropna)
-        plan = pn.Sort(plan, ("count",), (ascending,))
+        # deterministic tie order: value ascending within equal counts
+        plan = pn.Sort(plan, ("count", "v"), (ascending, True))
         pdf = api.collect(plan)
         # ties: pandas orders by value; match roughly
         out = pd.Series(pdf["count"].to_numpy(), index=pdf["v"].to_numpy(),

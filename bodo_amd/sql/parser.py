@@ -202,6 +202,7 @@ class SelectItem:
     expr: Any
     alias: Optional[str]
     star: bool = False
+    star_table: Optional[str] = None  # qualified star: t.*
 
 
 @dataclass
@@ -426,6 +427,14 @@ class Parser:
     def parse_select_item(self) -> SelectItem:
         if self.accept_op("*"):
             return SelectItem(None, None, star=True)
+        t = self.peek()
+        nt = self.toks[self.i + 1] if self.i + 1 < len(self.toks) else None
+        n3 = self.toks[self.i + 2] if self.i + 2 < len(self.toks) else None
+        if t and t.kind == "id" and nt and nt.kind == "op" \
+                and nt.value == "." and n3 and n3.kind == "op" \
+                and n3.value == "*":
+            self.i += 3
+            return SelectItem(None, None, star=True, star_table=t.value)
         e = self.parse_expr()
         alias = None
         if self.accept_kw("as"):

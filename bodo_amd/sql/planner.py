@@ -205,7 +205,11 @@ class Planner:
                 plan = self._plan_windows(plan, scope, q)
             for it in q.items:
                 if it.star:
+                    want = getattr(it, "star_table", None)
                     for alias, col, internal in scope.entries:
+                        if want is not None and \
+                                alias.lower() != want.lower():
+                            continue
                         out_names.append(col)
                         out_exprs.append(ex.ColRef(internal))
                     continue

@@ -453,3 +453,13 @@ def test_series_ffill_bfill():
     pd.testing.assert_series_equal(b.x.bfill().to_pandas(),
                                    df.x.bfill().reset_index(drop=True),
                                    check_names=False)
+
+
+def test_qualified_star():
+    l = pd.DataFrame({"k": [1, 2], "a": [10, 20]})
+    r = pd.DataFrame({"k": [1, 2], "b": ["x", "y"]})
+    bc = BodoSQLContext({"l": l, "r": r})
+    out = bc.sql("select l.*, r.b from l join r on l.k = r.k "
+                 "order by l.k").to_pandas()
+    assert list(out.columns) == ["k", "a", "b"]
+    assert out.a.tolist() == [10, 20]
