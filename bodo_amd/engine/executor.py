@@ -629,11 +629,6 @@ def _exec_join(node: pn.Join, ctx) -> Table:
     right = _exec(node.right, ctx)
     from . import ooc
 
-    if ctx.world > 1 and node.how != "cross" and node.left_on:
-        from . import join_filter
-
-        left, right = join_filter.apply_runtime_filters(
-            left, right, node.left_on, node.right_on, node.how, ctx)
     if ctx.world == 1 or node.how == "cross":
         if ctx.world > 1 and node.how == "cross":
             right = comm.allgather_table(right)
@@ -651,7 +646,13 @@ def _exec_join(node: pn.Join, ctx) -> Table:
         out = rel.join_local(left_full, right, node.left_on, node.right_on,
                              node.how, node.suffixes)
         return out
-    # hash-shuffle both sides by join keys
+    # hash-shuffle both sides by join keys; runtime filters prune the probe
+    # side first so non-matching rows never enter the exchange (broadcast
+    # joins above skip this: they have no shuffle to save)
+    from . import join_filter
+
+    left, right = join_filter.apply_runtime_filters(
+        left, right, node.left_on, node.right_on, node.how, ctx)
     lh = ops.hash_columns([left.column(k) for k in node.left_on])
     rh = ops.hash_columns([right.column(k) for k in node.right_on])
     lp = torch.remainder(lh, ctx.world)

@@ -424,9 +424,11 @@ def test_dist_ffill():
 
 
 def _q_rtf_join(bpd, rank, payload):
+    import bodo_amd.engine.executor as exm
     import bodo_amd.engine.join_filter as jf
 
     jf.MIN_PROBE_ROWS = 16  # force the runtime filter at test scale
+    exm.BROADCAST_JOIN_THRESHOLD = 0  # force the shuffle path
     l = bpd.from_pandas(payload["left"])
     r = bpd.from_pandas(payload["right"])
     return l.merge(r, on="k", how=payload["how"]).sort_values(
