@@ -224,7 +224,9 @@ class Column:
             arr = arr.astype("timedelta64[ns]").view("int64")
             dtype = bt.int64
         elif arr.dtype == object or arr.dtype.kind == "U":
-            return Column.from_arrow(pa.array(arr), device)
+            # from_pandas treats float NaN placeholders in object arrays as
+            # nulls (pa.array would infer double from a leading NaN)
+            return Column.from_arrow(pa.Array.from_pandas(arr), device)
         else:
             dtype = bt.from_numpy_dtype(arr.dtype)
             if arr.dtype.kind == "u" and arr.dtype.itemsize > 1:

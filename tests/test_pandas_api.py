@@ -285,3 +285,13 @@ def test_agg_extras(df):
     ag = b.agg({"a": "max", "b": "sum"})
     assert ag["a"] == df.a.max() and abs(ag["b"] - df.b.sum()) < 1e-9
     assert b.memory_usage()["a"] > 0
+
+
+def test_string_shift_ffill():
+    src = pd.DataFrame({"s": ["a", None, "b", None, None, "c"] * 10})
+    b = bpd.from_pandas(src)
+    got = _decat(b.s.shift(1).to_pandas()).astype(object)
+    exp = src.s.shift(1).astype(object)
+    assert (got.fillna("~") == exp.fillna("~")).all()
+    got = _decat(b.s.ffill().to_pandas()).astype(str)
+    assert (got.to_numpy() == src.s.ffill().astype(str).to_numpy()).all()
