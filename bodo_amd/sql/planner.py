@@ -183,6 +183,9 @@ class Planner:
         elif has_agg:
             plan, scope2, key_map, agg_map = self._aggregate(
                 plan, scope, q)
+            if any(_has_window(it.expr) for it in q.items if not it.star):
+                plan = self._plan_windows_post_agg(plan, scope, q, key_map,
+                                                   agg_map)
             # build output projection over the agg result
             for it in q.items:
                 if it.star:
@@ -416,6 +419,108 @@ class Planner:
             arms.append(pn.Projection(aplan, tuple(names), tuple(exprs)))
             out_names = names
         return pn.Union(tuple(arms), False), out_names
+
+    def _plan_windows_post_agg(self, plan, scope: Scope, q: ast.Query,
+                               key_map, agg_map):
+        """Window functions over aggregate output (RANK() OVER (ORDER BY
+        SUM(v))): _aggregate already computed every aggregate referenced
+        inside OVER(...), so partition/order/arg expressions resolve via
+        the post-agg maps and the Window node runs on the agg plan."""
+        pre_names = list(key_map.values()) + list(agg_map.values())
+        pre_exprs = {n: ex.ColRef(n) for n in pre_names}
+        need_pre = False
+
+        def conv(e):
+            return self._post_agg_expr(e, key_map, agg_map, scope)
+
+        def as_col(e, base):
+            nonlocal need_pre
+            k = _ast_key(e)
+            if k in key_map:
+                return key_map[k]
+            if k in agg_map:
+                return agg_map[k]
+            name = self._uniq(base)
+            pre_exprs[name] = conv(e)
+            pre_names.append(name)
+            need_pre = True
+            return name
+
+        groups: Dict[tuple, list] = {}
+
+        def register(w: ast.WindowE) -> str:
+            nonlocal need_pre
+            keys = tuple(as_col(pe, "__wk") for pe in w.partition_by)
+            if not keys:
+                kname = self._uniq("__wk")
+                pre_exprs[kname] = ex.Const(1)
+                pre_names.append(kname)
+                need_pre = True
+                keys = (kname,)
+            order = tuple(as_col(oe, "__wo") for oe, _ in w.order_by)
+            asc = tuple(a for _, a in w.order_by)
+            out = self._uniq("__win")
+            fn = w.func
+            arg_col = ""
+            if w.args and not w.star:
+                arg_col = as_col(w.args[0], "__wa")
+            if fn == "row_number":
+                spec = (out, "", "row_number", None)
+            elif fn in ("rank", "dense_rank"):
+                spec = (out, "", fn, None)
+            elif fn in ("first_value", "last_value"):
+                spec = (out, arg_col, fn, None)
+            elif fn == "ntile":
+                k = int(w.args[0].value) if w.args else 1
+                spec = (out, "", "ntile", k)
+            elif fn in ("lag", "lead"):
+                n = 1
+                if len(w.args) > 1 and isinstance(w.args[1], ast.Lit):
+                    n = int(w.args[1].value)
+                spec = (out, arg_col, "shift", n if fn == "lag" else -n)
+            elif fn in AGG_FUNCS or fn == "count":
+                if order:
+                    if fn == "sum":
+                        spec = (out, arg_col, "cumsum", None)
+                    else:
+                        raise NotImplementedError(
+                            f"{fn.upper()} OVER ORDER BY over aggregates")
+                elif fn == "count" and (w.star or not w.args):
+                    spec = (out, "", "transform_size", None)
+                else:
+                    spec = (out, arg_col,
+                            f"transform_{AGG_FUNCS.get(fn, fn)}", None)
+            else:
+                raise NotImplementedError(f"window function {fn}")
+            groups.setdefault((keys, order, asc), []).append(spec)
+            return out
+
+        def rewrite(e):
+            if isinstance(e, ast.WindowE):
+                name = register(e)
+                col = ast.Col(None, name)
+                key_map[_ast_key(col)] = name
+                return col
+            if isinstance(e, (ast.Query, ast.SetOpQ)):
+                return e
+            for f in getattr(e, "__dataclass_fields__", {}):
+                v = getattr(e, f)
+                if isinstance(v, list):
+                    setattr(e, f, [rewrite(x) if hasattr(
+                        x, "__dataclass_fields__") else x for x in v])
+                elif hasattr(v, "__dataclass_fields__"):
+                    setattr(e, f, rewrite(v))
+            return e
+
+        for it in q.items:
+            if not it.star:
+                it.expr = rewrite(it.expr)
+        if need_pre:
+            plan = pn.Projection(plan, tuple(pre_names),
+                                 tuple(pre_exprs[n] for n in pre_names))
+        for (keys, order, asc), specs in groups.items():
+            plan = pn.Window(plan, keys, order, asc, tuple(specs))
+        return plan
 
     def _plan_setop(self, q: "ast.SetOpQ"):
         """UNION [ALL] / INTERSECT / EXCEPT.  A trailing ORDER BY/LIMIT

@@ -463,3 +463,22 @@ def test_qualified_star():
                  "order by l.k").to_pandas()
     assert list(out.columns) == ["k", "a", "b"]
     assert out.a.tolist() == [10, 20]
+
+
+def test_window_over_aggregates():
+    rng = np.random.default_rng(22)
+    df = pd.DataFrame({"k": rng.integers(0, 8, 200),
+                       "g": rng.choice(["x", "y"], 200),
+                       "v": rng.random(200)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql(
+        "select g, k, sum(v) as s, "
+        "rank() over (partition by g order by sum(v) desc) as rk "
+        "from t group by g, k order by g, rk").to_pandas()
+    agg = df.groupby(["g", "k"], as_index=False)["v"].sum().rename(
+        columns={"v": "s"})
+    agg["rk"] = agg.groupby("g")["s"].rank(method="min", ascending=False)
+    exp = agg.sort_values(["g", "rk"]).reset_index(drop=True)[
+        ["g", "k", "s", "rk"]]
+    got["g"] = got["g"].astype(str)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
