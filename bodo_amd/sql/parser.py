@@ -31,6 +31,11 @@ _TOKEN_RE = re.compile(r"""
 """, re.VERBOSE)
 
 
+#: identifiers that start a clause and therefore can never be an implicit
+#: table alias (dialect words kept out of KEYWORDS for expression use)
+_NON_ALIAS_IDS = {"qualify", "offset", "rows", "window", "fetch"}
+
+
 @dataclass
 class Tok:
     kind: str  # num, str, id, kw, op
@@ -217,6 +222,7 @@ class Query:
     order_by: List[Tuple[Any, bool]]
     limit: Optional[int]
     grouping_sets: Optional[List[List[Any]]] = None
+    qualify: Optional[Any] = None  # Snowflake-dialect window filter
 
 
 class Parser:
@@ -375,6 +381,11 @@ class Parser:
                 while self.accept_op(","):
                     group_by.append(self.parse_expr())
         having = self.parse_expr() if self.accept_kw("having") else None
+        qualify = None
+        qt = self.peek()
+        if qt and qt.kind == "id" and qt.value.lower() == "qualify":
+            self.i += 1
+            qualify = self.parse_expr()
         order_by: List[Tuple[Any, bool]] = []
         if self.accept_kw("order"):
             self.expect_kw("by")
@@ -397,7 +408,7 @@ class Parser:
                 self.i += 1
                 limit_offset = int(self.next().value)
         q = Query(items, distinct, table, joins, where, group_by, having,
-                  order_by, limit, grouping_sets)
+                  order_by, limit, grouping_sets, qualify)
         q.limit_offset = limit_offset
         return q
 
@@ -410,7 +421,8 @@ class Parser:
                 alias = self.next().value
             else:
                 nt = self.peek()
-                if nt and nt.kind == "id":
+                if nt and nt.kind == "id" \
+                        and nt.value.lower() not in _NON_ALIAS_IDS:
                     alias = self.next().value
             return TableRef(alias or "__subq", alias, subquery=q)
         t = self.next()
@@ -418,7 +430,7 @@ class Parser:
             raise SyntaxError(f"expected table name, got {t}")
         alias = None
         nt = self.peek()
-        if nt and nt.kind == "id":
+        if nt and nt.kind == "id" and nt.value.lower() not in _NON_ALIAS_IDS:
             alias = self.next().value
         elif self.accept_kw("as"):
             alias = self.next().value

@@ -175,7 +175,9 @@ class Planner:
 
         # ------------------------------------------------- aggregation
         has_agg = any(_has_agg(it.expr) for it in q.items if not it.star) \
-            or (q.having is not None and _has_agg(q.having)) or q.group_by
+            or (q.having is not None and _has_agg(q.having)) or q.group_by \
+            or (getattr(q, "qualify", None) is not None
+                and _has_agg(q.qualify))
         out_names: List[str] = []
         out_exprs: List[ex.Expr] = []
         if has_agg and getattr(q, "grouping_sets", None) is not None:
@@ -183,7 +185,8 @@ class Planner:
         elif has_agg:
             plan, scope2, key_map, agg_map = self._aggregate(
                 plan, scope, q)
-            if any(_has_window(it.expr) for it in q.items if not it.star):
+            if any(_has_window(it.expr) for it in q.items if not it.star) \
+                    or getattr(q, "qualify", None) is not None:
                 plan = self._plan_windows_post_agg(plan, scope, q, key_map,
                                                    agg_map)
             # build output projection over the agg result
@@ -204,7 +207,8 @@ class Planner:
                 e, key_map, agg_map, scope) if not isinstance(e, ast.Col) \
                 or (e.table is None and e.name in out_names and False) else None
         else:
-            if any(_has_window(it.expr) for it in q.items if not it.star):
+            if any(_has_window(it.expr) for it in q.items if not it.star) \
+                    or (getattr(q, "qualify", None) is not None):
                 plan = self._plan_windows(plan, scope, q)
             for it in q.items:
                 if it.star:
@@ -362,12 +366,18 @@ class Planner:
         for it in q.items:
             if not it.star:
                 it.expr = rewrite(it.expr)
+        qual = getattr(q, "qualify", None)
+        if qual is not None:
+            q.qualify = rewrite(qual)
         if need_pre:
             plan = pn.Projection(
                 plan, tuple(pre_names),
                 tuple(pre_exprs[n] for n in pre_names))
         for (keys, order, asc), specs in groups.items():
             plan = pn.Window(plan, keys, order, asc, tuple(specs))
+        if qual is not None:
+            plan = pn.Filter(plan, self.expr(q.qualify, scope))
+            q.qualify = None
         return plan
 
     def _plan_grouping_sets(self, plan, scope: Scope, q: ast.Query):
@@ -515,11 +525,17 @@ class Planner:
         for it in q.items:
             if not it.star:
                 it.expr = rewrite(it.expr)
+        qual = getattr(q, "qualify", None)
+        if qual is not None:
+            q.qualify = rewrite(qual)
         if need_pre:
             plan = pn.Projection(plan, tuple(pre_names),
                                  tuple(pre_exprs[n] for n in pre_names))
         for (keys, order, asc), specs in groups.items():
             plan = pn.Window(plan, keys, order, asc, tuple(specs))
+        if qual is not None:
+            plan = pn.Filter(plan, conv(q.qualify))
+            q.qualify = None
         return plan
 
     def _plan_setop(self, q: "ast.SetOpQ"):

@@ -482,3 +482,32 @@ def test_window_over_aggregates():
         ["g", "k", "s", "rk"]]
     got["g"] = got["g"].astype(str)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_qualify():
+    rng = np.random.default_rng(23)
+    df = pd.DataFrame({"g": rng.choice(["x", "y", "z"], 300),
+                       "k": rng.integers(0, 40, 300), "v": rng.random(300)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql(
+        "select g, k, v from t qualify row_number() over "
+        "(partition by g order by v desc) <= 2 "
+        "order by g, v desc").to_pandas()
+    sdf = df.sort_values(["g", "v"], ascending=[True, False])
+    exp = sdf.groupby("g").head(2).sort_values(
+        ["g", "v"], ascending=[True, False]).reset_index(drop=True)
+    got["g"] = got["g"].astype(str)
+    pd.testing.assert_frame_equal(got, exp[["g", "k", "v"]],
+                                  check_dtype=False)
+    got2 = bc.sql(
+        "select g, k, sum(v) as s from t group by g, k "
+        "qualify rank() over (partition by g order by sum(v) desc) <= 3 "
+        "order by g, s desc").to_pandas()
+    agg = df.groupby(["g", "k"], as_index=False)["v"].sum().rename(
+        columns={"v": "s"})
+    agg["rk"] = agg.groupby("g")["s"].rank(method="min", ascending=False)
+    exp2 = agg[agg.rk <= 3].sort_values(
+        ["g", "s"], ascending=[True, False]).reset_index(drop=True)[
+        ["g", "k", "s"]]
+    got2["g"] = got2["g"].astype(str)
+    pd.testing.assert_frame_equal(got2, exp2, check_dtype=False)
