@@ -1048,7 +1048,15 @@ class Planner:
                 delta = _interval_to_timestamp_delta(e.left, qty, unit, e.op)
                 if delta is not None:
                     return delta
-                raise NotImplementedError("interval arithmetic on columns")
+                # fixed-length units add as nanosecond offsets on columns
+                ns_per = {"day": 86400 * 10**9, "hour": 3600 * 10**9,
+                          "minute": 60 * 10**9, "second": 10**9,
+                          "week": 7 * 86400 * 10**9}
+                if unit in ns_per:
+                    off = ex.Const(int(qty) * ns_per[unit], bt.int64)
+                    return ex.BinOp(e.op, l, off)
+                raise NotImplementedError(
+                    f"interval '{unit}' arithmetic on columns")
             return ex.BinOp(e.op, l, r)
         if isinstance(e, ast.Un):
             assert e.op == "not"

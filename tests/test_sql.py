@@ -511,3 +511,12 @@ def test_qualify():
         ["g", "k", "s"]]
     got2["g"] = got2["g"].astype(str)
     pd.testing.assert_frame_equal(got2, exp2, check_dtype=False)
+
+
+def test_column_interval_arithmetic():
+    ts = pd.date_range("2021-01-01", periods=20)
+    df = pd.DataFrame({"t": ts, "d": pd.date_range("2021-01-05", periods=20)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select t from t where t + interval '3' day < d").to_pandas()
+    exp = df[df.t + pd.Timedelta(days=3) < df.d][["t"]].reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
