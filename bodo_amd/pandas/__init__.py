@@ -114,6 +114,45 @@ def concat(objs, axis=0, ignore_index=False, **kwargs):
                                      **kwargs))
 
 
+def get_dummies(data, prefix=None, columns=None, dtype=None,
+                **kwargs) -> BodoDataFrame:
+    """One-hot encode (reference: pd_dataframe_ext get_dummies overload):
+    category values come from a distributed unique() pass, then each output
+    column is one lazy equality projection — the frame never leaves the
+    device."""
+    from ..plan.expr import Cmp, ColRef, Const
+    from .series import BodoSeries
+
+    if isinstance(data, BodoSeries):
+        vals = [v for v in data.unique().tolist() if v is not None
+                and v == v]
+        vals = sorted(map(str, vals))
+        names = [f"{prefix}_{v}" if prefix else str(v) for v in vals]
+        exprs = tuple(Cmp("eq", data._expr, Const(v)) for v in vals)
+        plan = _pn.Projection(data._plan, tuple(names), exprs)
+        return BodoDataFrame(plan, list(names))
+    if isinstance(data, BodoDataFrame):
+        cats = list(columns) if columns is not None else [
+            c for c in data._columns
+            if not _pd.api.types.is_numeric_dtype(data.head(1)[c].dtype)]
+        names, exprs = [], []
+        for c in data._columns:
+            if c not in cats:
+                names.append(c)
+                exprs.append(ColRef(c))
+        for c in cats:
+            ser = data[c]
+            vals = sorted(str(v) for v in ser.unique().tolist()
+                          if v is not None and v == v)
+            for v in vals:
+                names.append(f"{c}_{v}")
+                exprs.append(Cmp("eq", ser._expr, Const(v)))
+        plan = _pn.Projection(data._plan, tuple(names), tuple(exprs))
+        return BodoDataFrame(plan, list(names))
+    return _pd.get_dummies(data, prefix=prefix, columns=columns,
+                           dtype=dtype, **kwargs)
+
+
 def to_datetime(arg, **kwargs):
     if isinstance(arg, BodoSeries):
         from ..core import types as bt

@@ -295,3 +295,13 @@ def test_string_shift_ffill():
     assert (got.fillna("~") == exp.fillna("~")).all()
     got = _decat(b.s.ffill().to_pandas()).astype(str)
     assert (got.to_numpy() == src.s.ffill().astype(str).to_numpy()).all()
+
+
+def test_get_dummies(df):
+    b = bpd.from_pandas(df)
+    got = bpd.get_dummies(b.c).to_pandas().astype(bool)
+    exp = pd.get_dummies(df.c).astype(bool).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    g2 = bpd.get_dummies(b[["a", "c"]], columns=["c"]).to_pandas()
+    e2 = pd.get_dummies(df[["a", "c"]], columns=["c"])
+    assert list(g2.columns) == list(e2.columns)
