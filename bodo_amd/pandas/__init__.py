@@ -153,6 +153,47 @@ def get_dummies(data, prefix=None, columns=None, dtype=None,
                            dtype=dtype, **kwargs)
 
 
+def cut(x, bins, labels=None, right=True, **kwargs):
+    """Binning.  labels=False stays lazy (bin indices from a comparison
+    chain); labelled output falls back to pandas (interval categoricals)."""
+    from ..plan.expr import Case, Cmp, Const
+    from .series import BodoSeries
+
+    if isinstance(x, BodoSeries) and labels is False \
+            and not isinstance(bins, int):
+        edges = list(bins)
+        op = "le" if right else "lt"
+        conds, thens = [], []
+        # below the first edge (or on it for right=False) -> NaN
+        conds.append(Cmp("le" if right else "lt", x._expr, Const(edges[0])))
+        thens.append(Const(None, None))
+        for i in range(1, len(edges)):
+            conds.append(Cmp(op, x._expr, Const(edges[i])))
+            thens.append(Const(i - 1))
+        return x._wrap(Case(tuple(conds), tuple(thens), Const(None, None)),
+                       None)
+    if isinstance(x, BodoSeries):
+        return _pd.cut(x.to_pandas(), bins, labels=labels, right=right,
+                       **kwargs)
+    return _pd.cut(x, bins, labels=labels, right=right, **kwargs)
+
+
+def qcut(x, q, labels=None, **kwargs):
+    """Quantile binning: edges from the exact distributed quantile pass,
+    then cut()."""
+    from .series import BodoSeries
+
+    if isinstance(x, BodoSeries):
+        n = q if isinstance(q, int) else len(q) - 1
+        probs = [i / n for i in range(n + 1)] if isinstance(q, int) else list(q)
+        edges = [x.quantile(p) for p in probs]
+        edges[0] = edges[0] - 1e-9  # include the minimum (pandas semantics)
+        if labels is False:
+            return cut(x, edges, labels=False)
+        return _pd.qcut(x.to_pandas(), q, labels=labels, **kwargs)
+    return _pd.qcut(x, q, labels=labels, **kwargs)
+
+
 def to_datetime(arg, **kwargs):
     if isinstance(arg, BodoSeries):
         from ..core import types as bt

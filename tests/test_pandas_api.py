@@ -305,3 +305,16 @@ def test_get_dummies(df):
     g2 = bpd.get_dummies(b[["a", "c"]], columns=["c"]).to_pandas()
     e2 = pd.get_dummies(df[["a", "c"]], columns=["c"])
     assert list(g2.columns) == list(e2.columns)
+
+
+def test_cut_qcut(df):
+    b = bpd.from_pandas(df)
+    bins = [0.0, 25.0, 50.0, 75.0, 100.0]
+    got = bpd.cut(b.b, bins, labels=False).to_pandas()
+    exp = pd.cut(df.b, bins, labels=False).reset_index(drop=True)
+    pd.testing.assert_series_equal(got, exp, check_names=False,
+                                   check_dtype=False)
+    gq = bpd.qcut(b.b.dropna(), 4, labels=False).to_pandas()
+    eq = pd.qcut(df.b.dropna(), 4, labels=False).reset_index(drop=True)
+    pd.testing.assert_series_equal(gq, eq.astype(float), check_names=False,
+                                   check_dtype=False)
