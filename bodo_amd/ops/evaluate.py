@@ -345,6 +345,8 @@ def infer_const_dtype(v) -> DType:
         return bt.date32
     if isinstance(v, str):
         return bt.string
+    if v is None:
+        return bt.float64  # typed null (NaN)
     raise TypeError(f"cannot infer dtype for constant {v!r}")
 
 
