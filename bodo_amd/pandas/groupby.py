@@ -123,19 +123,62 @@ class DataFrameGroupBy:
             return ser
         return out
 
-    def head(self, n=5):
-        """First n rows per group in encounter order (row_number window +
-        filter; reference: groupby.head)."""
+    def _rn_filter(self, cond_builder):
+        """Shared row_number/size window + filter + column restore."""
         from .frame import BodoDataFrame
-        from ..plan.expr import Cmp, ColRef, Const
+        from ..plan.expr import ColRef
 
         w = pn.Window(self._frame._lazy_plan, tuple(self._keys), (), (),
-                      (("__rn", "", "row_number", None),))
-        filt = pn.Filter(w, Cmp("le", ColRef("__rn"), Const(int(n))))
+                      (("__rn", "", "row_number", None),
+                       ("__sz", "", "transform_size", None)))
+        filt = pn.Filter(w, cond_builder())
         cols = list(self._frame._columns)
         proj = pn.Projection(filt, tuple(cols),
                              tuple(ColRef(c) for c in cols))
         return BodoDataFrame(proj, cols)
+
+    def head(self, n=5):
+        """First n rows per group in encounter order (row_number window +
+        filter; reference: groupby.head)."""
+        from ..plan.expr import Cmp, ColRef, Const
+
+        return self._rn_filter(
+            lambda: Cmp("le", ColRef("__rn"), Const(int(n))))
+
+    def tail(self, n=5):
+        """Last n rows per group in encounter order."""
+        from ..plan.expr import BinOp, Cmp, ColRef, Const
+
+        return self._rn_filter(lambda: Cmp(
+            "gt", ColRef("__rn"),
+            BinOp("sub", ColRef("__sz"), Const(int(n)))))
+
+    def nth(self, n):
+        """The n-th row of each group (0-based, encounter order)."""
+        from ..plan.expr import Cmp, ColRef, Const
+
+        return self._rn_filter(
+            lambda: Cmp("eq", ColRef("__rn"), Const(int(n) + 1)))
+
+    def filter(self, func, *args):
+        """Keep rows of groups where func(group_frame) is True (reference:
+        groupby.filter): groups co-locate by key hash, pandas filter runs
+        per shard."""
+        from .frame import BodoDataFrame
+
+        keys = list(self._keys)
+        dropna = self._dropna
+
+        def _part(pdf, *a):
+            if len(pdf) == 0:
+                return pdf
+            return pdf.groupby(keys, dropna=dropna,
+                               sort=False).filter(func, *a)
+
+        plan = pn.MapPartitions(
+            pn.ShuffleByKey(self._frame._lazy_plan, tuple(keys)),
+            _part, tuple(args), tuple(self._frame._columns))
+        return BodoDataFrame(plan, list(self._frame._columns))
 
     def _window(self, specs, order_by=(), ascending=()):
         from .frame import BodoDataFrame

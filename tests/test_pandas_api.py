@@ -318,3 +318,23 @@ def test_cut_qcut(df):
     eq = pd.qcut(df.b.dropna(), 4, labels=False).reset_index(drop=True)
     pd.testing.assert_series_equal(gq, eq.astype(float), check_names=False,
                                    check_dtype=False)
+
+
+def test_groupby_tail_nth_filter(df):
+    b = bpd.from_pandas(df)
+    got = _decat(b.groupby("a").tail(2).to_pandas())
+    got = got.sort_values(["a", "b"]).reset_index(drop=True)
+    exp = df.groupby("a").tail(2).sort_values(["a", "b"]).reset_index(
+        drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    got = _decat(b.groupby("a").nth(0).to_pandas())
+    got = got.sort_values(["a", "b"]).reset_index(drop=True)
+    exp = df.groupby("a", as_index=False).nth(0).sort_values(
+        ["a", "b"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    got = _decat(b.groupby("a").filter(
+        lambda g: len(g) > 15).to_pandas())
+    got = got.sort_values(["a", "b"]).reset_index(drop=True)
+    exp = df.groupby("a").filter(lambda g: len(g) > 15).sort_values(
+        ["a", "b"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
