@@ -324,7 +324,9 @@ class BodoDataFrame:
 
     def merge(self, right: "BodoDataFrame", how="inner", on=None, left_on=None,
               right_on=None, suffixes=("_x", "_y"), copy=None, **kwargs):
-        if on is not None:
+        if how == "cross":
+            left_on = right_on = []
+        elif on is not None:
             if isinstance(on, str):
                 on = [on]
             left_on = right_on = list(on)

@@ -338,3 +338,15 @@ def test_groupby_tail_nth_filter(df):
     exp = df.groupby("a").filter(lambda g: len(g) > 15).sort_values(
         ["a", "b"]).reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_cross_merge():
+    l = pd.DataFrame({"a": [1, 2]})
+    r = pd.DataFrame({"b": ["x", "y", "z"]})
+    got = _decat(bpd.from_pandas(l).merge(bpd.from_pandas(r),
+                                          how="cross").to_pandas())
+    exp = l.merge(r, how="cross")
+    pd.testing.assert_frame_equal(
+        got.sort_values(["a", "b"]).reset_index(drop=True),
+        exp.sort_values(["a", "b"]).reset_index(drop=True),
+        check_dtype=False)
