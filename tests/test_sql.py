@@ -520,3 +520,34 @@ def test_column_interval_arithmetic():
     got = bc.sql("select t from t where t + interval '3' day < d").to_pandas()
     exp = df[df.t + pd.Timedelta(days=3) < df.d][["t"]].reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_tpcxbb_q26_sql():
+    """TPCx-BB Q26 in SQL form (reference: the BodoSQL config of
+    e2e-tests/tpcx-bb): conditional-count CASE aggregates + HAVING."""
+    rng = np.random.default_rng(4)
+    n = 20000
+    ss = pd.DataFrame({"ss_item_sk": rng.integers(1, 500, n),
+                       "ss_customer_sk": rng.integers(1, 300, n)})
+    item = pd.DataFrame({
+        "i_item_sk": np.arange(1, 501),
+        "i_class_id": rng.integers(1, 16, 500).astype(np.int32),
+        "i_category": rng.choice(["Books", "Music", "Home"], 500)})
+    bc = BodoSQLContext({"store_sales": ss, "item": item})
+    got = bc.sql(
+        "select ss_customer_sk, count(ss_item_sk) as cnt, "
+        "sum(case when i_class_id = 1 then 1 else 0 end) as c1, "
+        "sum(case when i_class_id = 2 then 1 else 0 end) as c2 "
+        "from store_sales, item "
+        "where ss_item_sk = i_item_sk and i_category = 'Books' "
+        "group by ss_customer_sk having count(ss_item_sk) > 5 "
+        "order by ss_customer_sk").to_pandas()
+    sale = ss.merge(item[item.i_category == "Books"], left_on="ss_item_sk",
+                    right_on="i_item_sk")
+    agg = sale.groupby("ss_customer_sk", as_index=False).agg(
+        cnt=("ss_item_sk", "count"),
+        c1=("i_class_id", lambda x: int((x == 1).sum())),
+        c2=("i_class_id", lambda x: int((x == 2).sum())))
+    exp = agg[agg.cnt > 5].sort_values("ss_customer_sk").reset_index(
+        drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
