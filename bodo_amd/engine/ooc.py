@@ -27,6 +27,36 @@ from .. import config as cfg
 from ..ops import hash_columns, take_table
 
 _ENV = "BODO_AMD_OOC_BYTES"
+_SPILL_ENV = "BODO_AMD_SPILL_DIR"
+
+
+class _Staged:
+    """A partition staged in host DRAM or, when BODO_AMD_SPILL_DIR is set,
+    spilled to disk (the NVMe tier below host memory; reference:
+    _storage_manager.h spill tiers).  Tables pickle cleanly: tensors via
+    torch serialization, dictionaries via arrow pickling."""
+
+    def __init__(self, tbl: Table):
+        self._tbl = None
+        self._path = None
+        spill = os.environ.get(_SPILL_ENV, "")
+        if spill:
+            import tempfile
+
+            os.makedirs(spill, exist_ok=True)
+            fd, self._path = tempfile.mkstemp(dir=spill, suffix=".spill")
+            os.close(fd)
+            torch.save(tbl, self._path)
+        else:
+            self._tbl = tbl
+
+    def load(self, device) -> Table:
+        if self._path is not None:
+            tbl = torch.load(self._path, weights_only=False)
+            os.unlink(self._path)
+            self._path = None
+            return tbl.to_device(device)
+        return self._tbl.to_device(device)
 
 
 def budget_bytes(device) -> Optional[int]:
@@ -57,7 +87,7 @@ def partition_table(tbl: Table, keys, nparts: int) -> List[Table]:
     out = []
     for p in range(nparts):
         idx = torch.nonzero(part == p, as_tuple=False).reshape(-1)
-        out.append(take_table(tbl, idx).to_device("cpu"))
+        out.append(_Staged(take_table(tbl, idx).to_device("cpu")))
     return out
 
 
@@ -75,7 +105,7 @@ def groupby_local(child: Table, keys, aggs, dropna) -> Table:
     del child
     outs = []
     for p in parts:
-        res = rel.groupby_local(p.to_device(device), keys, aggs, dropna)
+        res = rel.groupby_local(p.load(device), keys, aggs, dropna)
         outs.append(res.to_device("cpu"))
     return concat_tables([o.to_device(device) for o in outs])
 
@@ -99,7 +129,7 @@ def join_local(left: Table, right: Table, left_on, right_on, how,
     del left, right
     outs = []
     for lp, rp in zip(lparts, rparts):
-        res = rel.join_local(lp.to_device(device), rp.to_device(device),
+        res = rel.join_local(lp.load(device), rp.load(device),
                              left_on, right_on, how, suffixes)
         outs.append(res.to_device("cpu"))
     return concat_tables([o.to_device(device) for o in outs])
@@ -152,11 +182,11 @@ def sort_local(tbl: Table, keys, ascending, na_position) -> Table:
     runs = []
     for p in range(nparts):
         idx = torch.nonzero(part == p, as_tuple=False).reshape(-1)
-        runs.append(take_table(tbl, idx).to_device("cpu"))
+        runs.append(_Staged(take_table(tbl, idx).to_device("cpu")))
     del tbl
     outs = []
     for run in runs:
-        dev_run = run.to_device(device)
+        dev_run = run.load(device)
         idx = sort_indices([dev_run.column(k) for k in keys], ascending,
                            na_position)
         outs.append(take_table(dev_run, idx).to_device("cpu"))

@@ -100,3 +100,21 @@ def test_ooc_sort_nans(tiny_budget):
         drop=True)
     exp = df.sort_values("v").reset_index(drop=True)
     pd.testing.assert_frame_equal(got[["v"]], exp[["v"]], check_dtype=False)
+
+
+def test_ooc_disk_spill(tiny_budget, tmp_path, monkeypatch):
+    """With BODO_AMD_SPILL_DIR set, over-budget partitions stage on disk
+    (the NVMe tier) and results stay exact."""
+    monkeypatch.setenv("BODO_AMD_SPILL_DIR", str(tmp_path / "spill"))
+    df = _df(15000, 9)
+    got = bpd.from_pandas(df).groupby("k", as_index=False).agg(
+        s=bpd.NamedAgg("v", "sum")).sort_values("k").to_pandas()
+    got = got.reset_index(drop=True)
+    exp = df.groupby("k", as_index=False).agg(
+        s=("v", "sum")).sort_values("k").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    srt = bpd.from_pandas(df).sort_values(["k", "v"]).to_pandas()
+    srt = srt.reset_index(drop=True)
+    srt["c"] = srt["c"].astype(str)
+    exp2 = df.sort_values(["k", "v"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(srt, exp2, check_dtype=False)
