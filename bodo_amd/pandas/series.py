@@ -385,6 +385,35 @@ class BodoSeries:
     def rename(self, name, **kwargs) -> "BodoSeries":
         return BodoSeries(self._plan, self._expr, name, self._frame)
 
+    def tail(self, n=5) -> pd.Series:
+        from ..engine import api
+
+        plan = pn.Limit(self._as_projection_plan(), n, tail=True)
+        ser = api.collect(plan)["v"]
+        ser.name = self.name
+        return ser
+
+    def sample(self, n=None, frac=None, random_state=None) -> "BodoSeries":
+        plan = pn.Sample(self._as_projection_plan(), n, frac, random_state)
+        return BodoSeries(plan, ColRef("v"), self.name)
+
+    def _idx_of(self, func: str) -> int:
+        from ..engine import api
+
+        extreme = self.max() if func == "max" else self.min()
+        plan = pn.RowId(self._as_projection_plan(), "__rid")
+        flt = pn.Filter(plan, Cmp("eq", ColRef("v"), Const(extreme)))
+        red = pn.Reduce(flt, (("i", "__rid", "min"),))
+        return int(api.collect(red)["i"].iloc[0])
+
+    def idxmax(self) -> int:
+        """Positional index of the first maximum (frames are positionally
+        indexed; the global row id comes from a RowId tag)."""
+        return self._idx_of("max")
+
+    def idxmin(self) -> int:
+        return self._idx_of("min")
+
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingSeries(self, int(window), min_periods)
 

@@ -350,3 +350,11 @@ def test_cross_merge():
         got.sort_values(["a", "b"]).reset_index(drop=True),
         exp.sort_values(["a", "b"]).reset_index(drop=True),
         check_dtype=False)
+
+
+def test_series_tail_idx(df):
+    b = bpd.from_pandas(df)
+    assert list(b.a.tail(4)) == list(df.a.tail(4))
+    assert b.b.idxmax() == int(df.b.idxmax())
+    assert b.b.idxmin() == int(df.b.idxmin())
+    assert len(b.b.sample(n=9, random_state=1).to_pandas()) == 9
