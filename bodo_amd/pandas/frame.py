@@ -513,6 +513,36 @@ class BodoDataFrame:
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingFrame(self, int(window), min_periods)
 
+    def filter(self, items=None, like=None, regex=None, axis=None):
+        """Column-label filtering (reference: frame.py filter)."""
+        if items is not None:
+            keep = [c for c in self._columns if c in set(items)]
+        elif like is not None:
+            keep = [c for c in self._columns if like in c]
+        elif regex is not None:
+            import re as _re
+
+            pat = _re.compile(regex)
+            keep = [c for c in self._columns if pat.search(c)]
+        else:
+            raise TypeError("must pass items, like, or regex")
+        return self[keep]
+
+    def get(self, key, default=None):
+        if isinstance(key, str):
+            return self[key] if key in self._columns else default
+        return self[key]
+
+    def reset_index(self, drop=False, **kwargs):
+        """Frames are positionally indexed; drop=True is the identity."""
+        if drop:
+            return self
+        return self._fallback("reset_index", drop=drop, **kwargs)
+
+    @property
+    def index(self):
+        return pd.RangeIndex(len(self))
+
     def explain(self, optimized: bool = True) -> str:
         """Pretty-print this frame's logical plan (reference: plan dumps at
         tracing_level>=2, bodo/pandas/plan.py:1090-1096)."""

@@ -358,3 +358,11 @@ def test_series_tail_idx(df):
     assert b.b.idxmax() == int(df.b.idxmax())
     assert b.b.idxmin() == int(df.b.idxmin())
     assert len(b.b.sample(n=9, random_state=1).to_pandas()) == 9
+
+
+def test_frame_filter_get_reset(df):
+    b = bpd.from_pandas(df)
+    assert list(b.filter(items=["a", "b"]).columns) == ["a", "b"]
+    assert list(b.filter(regex="^[ab]$").columns) == ["a", "b"]
+    assert b.get("nope", 3) == 3
+    assert b.reset_index(drop=True) is b
