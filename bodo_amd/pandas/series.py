@@ -414,6 +414,28 @@ class BodoSeries:
     def idxmin(self) -> int:
         return self._idx_of("min")
 
+    def map_partitions(self, func, *args) -> "BodoSeries":
+        """Run a python function over each rank's shard of this series
+        (reference: series.py map_partitions)."""
+
+        def _part(pdf, *a):
+            res = func(pdf["v"], *a)
+            return res.to_frame(name="v") if isinstance(res, pd.Series) \
+                else pd.DataFrame({"v": res})
+
+        plan = pn.MapPartitions(self._as_projection_plan(), _part,
+                                tuple(args), ("v",))
+        return BodoSeries(plan, ColRef("v"), self.name)
+
+    def convert_dtypes(self, **kwargs) -> "BodoSeries":
+        return self  # columns are already arrow-typed
+
+    @property
+    def ai(self):
+        raise NotImplementedError(
+            "BodoSeries.ai (LLM/embedding calls) requires network access to "
+            "a model endpoint; this build runs fully offline")
+
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingSeries(self, int(window), min_periods)
 
