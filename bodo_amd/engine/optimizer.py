@@ -20,6 +20,12 @@ from ..plan.expr import (
 def optimize(plan: pn.PlanNode) -> pn.PlanNode:
     plan = fuse_projections(plan)
     plan = push_filters(plan)
+    import os as _os
+
+    if _os.environ.get("BODO_AMD_JOIN_REORDER", "") == "1":
+        from .join_order import reorder
+
+        plan = reorder(plan)
     plan = push_limits(plan)
     plan = prune_columns(plan, None)
     return plan

