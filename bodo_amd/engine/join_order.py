@@ -23,6 +23,7 @@ from __future__ import annotations
 from typing import Dict, List, Optional, Set, Tuple
 
 from ..plan import nodes as pn
+from ..plan.expr import ColRef
 
 _BIG = float(1 << 60)
 
@@ -138,4 +139,9 @@ def reorder(node: pn.PlanNode) -> pn.PlanNode:
         cur_est = max(cur_est, est[nxt])
         in_tree.add(nxt)
         remaining.discard(nxt)
+    # restore the original output column order (user-visible for frames)
+    orig = node.out_columns()
+    if orig is not None and plan.out_columns() != list(orig):
+        plan = pn.Projection(plan, tuple(orig),
+                             tuple(ColRef(c) for c in orig))
     return plan
