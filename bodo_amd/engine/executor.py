@@ -167,7 +167,8 @@ def _exec_filter(node: pn.Filter, ctx) -> Table:
     return ev.eval_filter(node.cond, child)
 
 
-SINGLE_PHASE_AGGS = {"median", "nunique", "var", "std", "quantile"}
+SINGLE_PHASE_AGGS = {"median", "nunique", "var", "std", "quantile",
+                     "skew"}
 
 
 def _exec_aggregate(node: pn.Aggregate, ctx) -> Table:
@@ -219,7 +220,8 @@ def _decompose_aggs(aggs):
             s = add_partial(in_name, "sum")
             c = add_partial(in_name, "count")
             finals.append((out_name, BinOp("div", ColRef(s), ColRef(c))))
-        elif func in ("sum", "min", "max", "count", "size", "first", "last", "prod"):
+        elif func in ("sum", "min", "max", "count", "size", "first", "last",
+                      "prod", "any", "all"):
             p = add_partial(in_name, func)
             finals.append((out_name, ColRef(p)))
         else:

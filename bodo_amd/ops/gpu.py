@@ -470,6 +470,38 @@ def _agg_one(col: Optional[Column], row_gid: torch.Tensor, ngroups: int,
         var = (acc2 - acc * acc / c) / (c - 1)
         var = torch.where(cnt < 2, torch.full_like(var, float("nan")), var)
         return Column(bt.float64, var if func == "var" else var.sqrt())
+    if func in ("any", "all"):
+        nz = (col.data != 0).to(torch.int64)
+        mask8 = None if col.mask is None else col.mask.view(torch.uint8)
+        op = "max_i64" if func == "any" else "min_i64"
+        init = _I64_MIN if func == "any" else _I64_MAX
+        acc, cnt = K.agg_update(nz, mask8, None, int(TypeKind.INT64),
+                                row_gid, ngroups, _AGG_OP[op], 0.0, init,
+                                True)
+        default = func == "all"  # empty/all-null group: any=False, all=True
+        out = torch.where(cnt == 0,
+                          torch.tensor(default, device=acc.device),
+                          acc != 0 if func == "any" else acc != 0)
+        return Column(bt.boolean, out.to(torch.bool))
+    if func == "skew":
+        x = col.data.to(torch.float64)
+        mask8 = None if col.mask is None else col.mask.view(torch.uint8)
+        s1, cnt = K.agg_update(x, mask8, None, int(TypeKind.FLOAT64),
+                               row_gid, ngroups, _AGG_OP["sum_f64"], 0.0, 0,
+                               True)
+        s2 = K.agg_update(x * x, mask8, None, int(TypeKind.FLOAT64), row_gid,
+                          ngroups, _AGG_OP["sum_f64"], 0.0, 0, False)[0]
+        s3 = K.agg_update(x * x * x, mask8, None, int(TypeKind.FLOAT64),
+                          row_gid, ngroups, _AGG_OP["sum_f64"], 0.0, 0,
+                          False)[0]
+        n = cnt.to(torch.float64)
+        mean = s1 / n
+        m2 = s2 / n - mean * mean
+        m3 = s3 / n - 3 * mean * s2 / n + 2 * mean ** 3
+        # pandas adjusted Fisher-Pearson
+        g = torch.sqrt(n * (n - 1)) / (n - 2) * m3 / m2.clamp(min=0) ** 1.5
+        g = torch.where(n < 3, torch.full_like(g, float("nan")), g)
+        return Column(bt.float64, g)
     if func == "median":
         return _median_by_group(col, row_gid, ngroups)
     if func == "nunique":

@@ -118,6 +118,7 @@ _init_pa_types()
 COMBINE_FUNC = {
     "sum": "sum", "count": "sum", "size": "sum", "min": "min", "max": "max",
     "first": "first", "last": "last", "prod": "prod",
+    "any": "any", "all": "all",
 }
 
 

@@ -116,6 +116,15 @@ class DataFrameGroupBy:
     def std(self, ddof=1):
         return self._simple("std")
 
+    def any(self):
+        return self._simple("any")
+
+    def all(self):
+        return self._simple("all")
+
+    def skew(self, numeric_only=False):
+        return self._simple("skew")
+
     def size(self):
         out = self._build([("size", "", "size")])
         if isinstance(out, _IndexedAggResult):

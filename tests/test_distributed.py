@@ -449,3 +449,19 @@ def test_dist_runtime_join_filter():
         exp = left.merge(right, on="k", how=how).sort_values(
             ["k", "v1"]).reset_index(drop=True)
         pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_anyall(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).groupby("a", as_index=False).agg(
+        an=bpd.NamedAgg("d", "any"), al=bpd.NamedAgg("d", "all"),
+        sk=bpd.NamedAgg("b", "skew")).sort_values("a")
+
+
+def test_dist_any_all_skew():
+    df = _df(900, 61)
+    df["d"] = df.b > 0
+    got = run_dist(_q_anyall, {"df": df}).reset_index(drop=True)
+    exp = df.groupby("a", as_index=False).agg(
+        an=("d", "any"), al=("d", "all"),
+        sk=("b", "skew")).sort_values("a").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-9)

@@ -366,3 +366,17 @@ def test_frame_filter_get_reset(df):
     assert list(b.filter(regex="^[ab]$").columns) == ["a", "b"]
     assert b.get("nope", 3) == 3
     assert b.reset_index(drop=True) is b
+
+
+def test_groupby_any_all_skew(df):
+    src = df.copy()
+    src["flag"] = src.a % 2 == 0
+    b = bpd.from_pandas(src)
+    got = b.groupby("c", as_index=False).agg(
+        a=bpd.NamedAgg("flag", "any"), l=bpd.NamedAgg("flag", "all"),
+        s=bpd.NamedAgg("b", "skew")).to_pandas()
+    got = _decat(got).sort_values("c").reset_index(drop=True)
+    exp = src.groupby("c", as_index=False).agg(
+        a=("flag", "any"), l=("flag", "all"),
+        s=("b", "skew")).sort_values("c").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-9)
