@@ -375,3 +375,23 @@ def test_parquet_gpu_snappy_e2e():
             s=("v", "sum")).sort_values("k").reset_index(drop=True)
         pd.testing.assert_frame_equal(got.reset_index(drop=True), exp,
                                       check_dtype=False)
+
+
+@pytest.mark.gpu
+def test_gpu_groupby_any_all_skew():
+    import bodo_amd.config as cfg
+
+    cfg.DEVICE = "cuda"
+    rng = np.random.default_rng(34)
+    df = pd.DataFrame({"k": rng.integers(0, 20, 100_000),
+                       "f": rng.random(100_000) > 0.3,
+                       "v": rng.random(100_000) * 5})
+    b = bpd.from_pandas(df)
+    got = b.groupby("k", as_index=False).agg(
+        an=bpd.NamedAgg("f", "any"), al=bpd.NamedAgg("f", "all"),
+        sk=bpd.NamedAgg("v", "skew")).to_pandas().sort_values("k")
+    got = got.reset_index(drop=True)
+    exp = df.groupby("k", as_index=False).agg(
+        an=("f", "any"), al=("f", "all"),
+        sk=("v", "skew")).sort_values("k").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-9)
