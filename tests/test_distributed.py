@@ -464,4 +464,4 @@ def test_dist_any_all_skew():
     exp = df.groupby("a", as_index=False).agg(
         an=("d", "any"), al=("d", "all"),
         sk=("b", "skew")).sort_values("a").reset_index(drop=True)
-    pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-9)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-6)

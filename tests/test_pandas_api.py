@@ -379,4 +379,4 @@ def test_groupby_any_all_skew(df):
     exp = src.groupby("c", as_index=False).agg(
         a=("flag", "any"), l=("flag", "all"),
         s=("b", "skew")).sort_values("c").reset_index(drop=True)
-    pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-9)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-6)
