@@ -154,14 +154,15 @@ def main():
 
     pq_path = None
     if args.parquet:
-        # write this rank's shard as uncompressed parquet (untimed), then
-        # each step re-reads it with the on-GPU decoder before the query
+        # write this rank's shard as snappy parquet (untimed; the realistic
+        # on-disk layout), then each step re-reads it with the on-GPU
+        # decoder before the query
         import pyarrow.parquet as pq
 
         os.makedirs("/tmp/bodo_bench_pq", exist_ok=True)
         pq_path = f"/tmp/bodo_bench_pq/trips_rank{rank}.parquet"
         at = trips.to_device("cpu").to_arrow()
-        pq.write_table(at, pq_path, compression="NONE",
+        pq.write_table(at, pq_path, compression="SNAPPY",
                        use_dictionary=["hvfhs_license_num"],
                        row_group_size=1 << 23)
         del at
