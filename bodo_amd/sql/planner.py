@@ -1183,6 +1183,31 @@ class Planner:
                                 (-n, None, 1))
             if name == "initcap":
                 return ex.StrOp(self.expr(e.args[0], scope), "title")
+            if name in ("char_length", "character_length", "len"):
+                return ex.StrOp(self.expr(e.args[0], scope), "len")
+            if name in ("strpos", "position", "instr"):
+                # STRPOS(str, sub) / two-arg POSITION(sub, str): 1-based,
+                # 0 when absent (find returns -1)
+                if name == "position":
+                    sub, hay = e.args[0], e.args[1]
+                else:
+                    hay, sub = e.args[0], e.args[1]
+                found = ex.StrOp(self.expr(hay, scope), "find",
+                                 (sub.value,))
+                return ex.BinOp("add", found, ex.Const(1))
+            if name == "split_part":
+                # Snowflake 1-based part index
+                return ex.StrOp(self.expr(e.args[0], scope), "split_get",
+                                (e.args[1].value,
+                                 int(e.args[2].value) - 1))
+            if name in ("lpad", "rpad"):
+                n = int(e.args[1].value)
+                fill = e.args[2].value if len(e.args) > 2 else " "
+                op = "rjust" if name == "lpad" else "ljust"
+                return ex.StrOp(self.expr(e.args[0], scope), op, (n, fill))
+            if name == "repeat":
+                return ex.StrOp(self.expr(e.args[0], scope), "repeat",
+                                (int(e.args[1].value),))
             if name == "date_trunc":
                 unit = e.args[0].value.lower().rstrip("s")
                 fld = {"day": "floor_day", "month": "trunc_month",

@@ -551,3 +551,19 @@ def test_tpcxbb_q26_sql():
     exp = agg[agg.cnt > 5].sort_values("ss_customer_sk").reset_index(
         drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_more_string_functions():
+    df = pd.DataFrame({"s": ["hello world", "foo-bar-baz", "x"] * 5})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select char_length(s) as l, strpos(s, 'o') as p, "
+                 "split_part(s, '-', 2) as sp, lpad(s, 15, '*') as lp, "
+                 "repeat(s, 2) as rp from t limit 3").to_pandas()
+    for c in got.columns:
+        if got[c].dtype.name == "category":
+            got[c] = got[c].astype(object)
+    assert got.l.tolist() == [11, 11, 1]
+    assert got.p.tolist() == [5, 2, 0]
+    assert got.sp.where(got.sp.notna(), None).tolist() == [None, "bar", None]
+    assert got.lp.iloc[2] == "*" * 14 + "x"
+    assert got.rp.iloc[2] == "xx"
