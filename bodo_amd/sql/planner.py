@@ -1228,6 +1228,43 @@ class Planner:
                 if name == "current_date":
                     v = v.normalize()
                 return ex.Const(v, bt.timestamp_ns)
+            if name in ("dateadd", "timestampadd"):
+                unit = (e.args[0].value if isinstance(e.args[0], ast.Lit)
+                        else e.args[0].name).lower().rstrip("s")
+                qty = self.expr(e.args[1], scope)
+                base = self.expr(e.args[2], scope)
+                ns_per = {"day": 86400 * 10**9, "hour": 3600 * 10**9,
+                          "minute": 60 * 10**9, "second": 10**9,
+                          "week": 7 * 86400 * 10**9}
+                if unit not in ns_per:
+                    raise NotImplementedError(f"DATEADD unit {unit}")
+                return ex.BinOp("add", base,
+                                ex.BinOp("mul", qty,
+                                         ex.Const(ns_per[unit], bt.int64)))
+            if name in ("datediff", "timestampdiff"):
+                unit = (e.args[0].value if isinstance(e.args[0], ast.Lit)
+                        else e.args[0].name).lower().rstrip("s")
+                a = self.expr(e.args[1], scope)
+                b2 = self.expr(e.args[2], scope)
+                if unit == "year":
+                    return ex.BinOp("sub", ex.DtField(b2, "year"),
+                                    ex.DtField(a, "year"))
+                if unit == "month":
+                    ydiff = ex.BinOp("sub", ex.DtField(b2, "year"),
+                                     ex.DtField(a, "year"))
+                    mdiff = ex.BinOp("sub", ex.DtField(b2, "month"),
+                                     ex.DtField(a, "month"))
+                    return ex.BinOp("add", ex.BinOp("mul", ydiff,
+                                                    ex.Const(12)), mdiff)
+                ns_per = {"day": 86400 * 10**9, "hour": 3600 * 10**9,
+                          "minute": 60 * 10**9, "second": 10**9,
+                          "week": 7 * 86400 * 10**9}
+                if unit not in ns_per:
+                    raise NotImplementedError(f"DATEDIFF unit {unit}")
+                da = ex.DtField(a, "floor_day") if unit in ("day", "week")                     else a
+                db = ex.DtField(b2, "floor_day") if unit in ("day", "week")                     else b2
+                return ex.BinOp("floordiv", ex.BinOp("sub", db, da),
+                                ex.Const(ns_per[unit], bt.int64))
             if name == "concat":
                 args = [self.expr(a, scope) for a in e.args]
                 out = args[0]

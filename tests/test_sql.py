@@ -567,3 +567,25 @@ def test_more_string_functions():
     assert got.sp.where(got.sp.notna(), None).tolist() == [None, "bar", None]
     assert got.lp.iloc[2] == "*" * 14 + "x"
     assert got.rp.iloc[2] == "xx"
+
+
+def test_dateadd_datediff():
+    rng = np.random.default_rng(40)
+    a = pd.to_datetime("2021-01-15") + pd.to_timedelta(
+        rng.integers(0, 700, 30), unit="D")
+    b = a + pd.to_timedelta(rng.integers(1, 500, 30), unit="D")
+    df = pd.DataFrame({"a": a, "b": b})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql(
+        "select datediff(day, a, b) as dd, datediff(month, a, b) as dm, "
+        "datediff(year, a, b) as dy, dateadd(day, 10, a) as ad "
+        "from t").to_pandas()
+    exp_dd = (df.b.dt.normalize() - df.a.dt.normalize()).dt.days
+    exp_dm = (df.b.dt.year - df.a.dt.year) * 12 + \
+        (df.b.dt.month - df.a.dt.month)
+    assert (got.dd.to_numpy() == exp_dd.to_numpy()).all()
+    assert (got.dm.to_numpy() == exp_dm.to_numpy()).all()
+    assert (got.dy.to_numpy() == (df.b.dt.year - df.a.dt.year)
+            .to_numpy()).all()
+    assert (pd.to_datetime(got.ad).to_numpy()
+            == (df.a + pd.Timedelta(days=10)).to_numpy()).all()
