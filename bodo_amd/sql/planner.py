@@ -1114,6 +1114,22 @@ class Planner:
                 zero = ex.Const(0)
                 return ex.Case((ex.Cmp("lt", inner, zero),),
                                (ex.BinOp("sub", zero, inner),), inner)
+            if name in ("iff", "if"):
+                return ex.Case((self.expr(e.args[0], scope),),
+                               (self.expr(e.args[1], scope),),
+                               self.expr(e.args[2], scope))
+            if name == "nvl":
+                a = self.expr(e.args[0], scope)
+                return ex.Case((ex.IsNull(a, negate=True),), (a,),
+                               self.expr(e.args[1], scope))
+            if name == "nvl2":
+                a = self.expr(e.args[0], scope)
+                return ex.Case((ex.IsNull(a, negate=True),),
+                               (self.expr(e.args[1], scope),),
+                               self.expr(e.args[2], scope))
+            if name == "zeroifnull":
+                a = self.expr(e.args[0], scope)
+                return ex.Case((ex.IsNull(a),), (ex.Const(0),), a)
             if name == "coalesce":
                 args = [self.expr(a, scope) for a in e.args]
                 out = args[-1]

@@ -589,3 +589,14 @@ def test_dateadd_datediff():
             .to_numpy()).all()
     assert (pd.to_datetime(got.ad).to_numpy()
             == (df.a + pd.Timedelta(days=10)).to_numpy()).all()
+
+
+def test_iff_nvl_family():
+    df = pd.DataFrame({"v": [1.0, np.nan, 3.0], "w": [9.0, 8.0, 7.0]})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select iff(v > 2, 'big', 'small') as c, nvl(v, 0) as n, "
+                 "nvl2(v, w, -1) as n2, zeroifnull(v) as z from t").to_pandas()
+    assert [str(x) for x in got.c] == ["small", "small", "big"]
+    assert got.n.tolist() == [1.0, 0.0, 3.0]
+    assert got.n2.tolist() == [9.0, -1.0, 7.0]
+    assert got.z.tolist() == [1.0, 0.0, 3.0]
