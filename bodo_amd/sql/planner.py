@@ -945,7 +945,9 @@ class Planner:
         agg_map = {}
 
         def collect(e):
-            if isinstance(e, ast.Func) and e.name in AGG_FUNCS:
+            if isinstance(e, ast.Func) and (
+                    e.name in AGG_FUNCS
+                    or e.name in ("listagg", "string_agg")):
                 k = _ast_key(e)
                 if k in agg_map:
                     return
@@ -960,9 +962,13 @@ class Planner:
                         in_name = self._uniq("__ain")
                         pre_exprs[in_name] = self.expr(arg, scope)
                         pre_names.append(in_name)
-                    func = AGG_FUNCS[e.name]
-                    if e.distinct and e.name == "count":
-                        func = "nunique"
+                    if e.name in ("listagg", "string_agg"):
+                        sep = e.args[1].value if len(e.args) > 1 else ""
+                        func = _listagg_func(sep)
+                    else:
+                        func = AGG_FUNCS[e.name]
+                        if e.distinct and e.name == "count":
+                            func = "nunique"
                     agg_specs.append((out, in_name, func))
                 agg_map[k] = out
                 return
@@ -1377,8 +1383,19 @@ def _split_conjuncts(e) -> list:
     return [e]
 
 
+def _listagg_func(sep: str):
+    """Callable agg for LISTAGG/STRING_AGG (runs on co-located shards via
+    the single-phase path)."""
+
+    def listagg(s):
+        return sep.join(str(v) for v in s if v is not None and v == v)
+
+    return listagg
+
+
 def _has_agg(e) -> bool:
-    if isinstance(e, ast.Func) and e.name in AGG_FUNCS:
+    if isinstance(e, ast.Func) and (
+            e.name in AGG_FUNCS or e.name in ("listagg", "string_agg")):
         return True
     for f in getattr(e, "__dataclass_fields__", {}):
         v = getattr(e, f)

@@ -600,3 +600,11 @@ def test_iff_nvl_family():
     assert got.n.tolist() == [1.0, 0.0, 3.0]
     assert got.n2.tolist() == [9.0, -1.0, 7.0]
     assert got.z.tolist() == [1.0, 0.0, 3.0]
+
+
+def test_listagg():
+    df = pd.DataFrame({"k": [1, 1, 2, 2, 2], "s": ["a", "b", "c", "d", "e"]})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select k, listagg(s, ',') as l from t "
+                 "group by k order by k").to_pandas()
+    assert [str(x) for x in got.l] == ["a,b", "c,d,e"]
