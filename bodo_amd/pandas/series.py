@@ -725,7 +725,7 @@ def _pd_dtype_to_bodo(dtype):
         "int8": bt.int8, "int16": bt.int16, "int32": bt.int32, "int64": bt.int64,
         "float32": bt.float32, "float64": bt.float64, "bool": bt.boolean,
         "str": bt.string, str: bt.string, int: bt.int64, float: bt.float64,
-        bool: bt.boolean,
+        bool: bt.boolean, "category": bt.dictionary, "string": bt.string,
     }
     if dtype in m:
         return m[dtype]

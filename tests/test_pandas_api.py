@@ -380,3 +380,10 @@ def test_groupby_any_all_skew(df):
         a=("flag", "any"), l=("flag", "all"),
         s=("b", "skew")).sort_values("c").reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-6)
+
+
+def test_astype_category(df):
+    b = bpd.from_pandas(df)
+    out = b.c.astype("category").to_pandas()
+    assert out.dtype.name == "category"
+    assert sorted(out.astype(str).unique()) == sorted(df.c.unique())
