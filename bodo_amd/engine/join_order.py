@@ -20,7 +20,7 @@ default; round 2 adds distinct-count sketches to turn this on.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Set, Tuple
+from typing import List, Optional, Set, Tuple
 
 from ..plan import nodes as pn
 from ..plan.expr import ColRef
