@@ -687,7 +687,8 @@ def _exec_parquet_write(node: pn.ParquetWrite, ctx) -> Table:
     child = _exec(node.child, ctx)
     from ..io import parquet as pio
 
-    pio.write_shard(child, node.path, node.compression, ctx)
+    pio.write_shard(child, node.path, node.compression, ctx,
+                    getattr(node, "partition_cols", ()))
     return Table([], [], 0)
 
 

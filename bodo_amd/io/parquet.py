@@ -146,8 +146,19 @@ def _dict_encode_strings(tbl: pa.Table, threshold: float = 0.5,
     return pa.table(dict(zip(tbl.column_names, new_cols)))
 
 
-def write_shard(tbl: Table, path: str, compression, ctx) -> None:
+def write_shard(tbl: Table, path: str, compression, ctx,
+                partition_cols=()) -> None:
     at = tbl.to_device("cpu").to_arrow()
+    if partition_cols:
+        # hive-style partition directories; per-rank unique basenames
+        import pyarrow.dataset as pads
+
+        pads.write_dataset(
+            at, path, format="parquet", partitioning_flavor="hive",
+            partitioning=list(partition_cols),
+            basename_template=f"part-{ctx.rank:05d}-{{i}}.parquet",
+            existing_data_behavior="overwrite_or_ignore")
+        return
     if ctx.world == 1 and path.endswith(".parquet") and not os.path.isdir(path):
         pq.write_table(at, path, compression=compression)
         return

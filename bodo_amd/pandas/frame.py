@@ -601,10 +601,12 @@ class BodoDataFrame:
     # ------------------------------------------------------------------
     # IO
     # ------------------------------------------------------------------
-    def to_parquet(self, path: str, compression="snappy", **kwargs):
+    def to_parquet(self, path: str, compression="snappy",
+                   partition_cols=None, **kwargs):
         from ..engine import api
 
-        plan = pn.ParquetWrite(self._plan, path, compression)
+        plan = pn.ParquetWrite(self._plan, path, compression,
+                               tuple(partition_cols or ()))
         api.materialize(plan)
 
     def to_iceberg(self, path: str, mode: str = "create", **kwargs):

@@ -412,12 +412,14 @@ class ParquetWrite(PlanNode):
     child: PlanNode
     path: str = ""
     compression: Optional[str] = "snappy"
+    partition_cols: Tuple[str, ...] = ()
 
     def children(self):
         return (self.child,)
 
     def with_children(self, *ch):
-        return ParquetWrite(ch[0], self.path, self.compression)
+        return ParquetWrite(ch[0], self.path, self.compression,
+                            self.partition_cols)
 
 
 @dataclass(frozen=True)

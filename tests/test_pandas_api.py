@@ -387,3 +387,13 @@ def test_astype_category(df):
     out = b.c.astype("category").to_pandas()
     assert out.dtype.name == "category"
     assert sorted(out.astype(str).unique()) == sorted(df.c.unique())
+
+
+def test_to_parquet_partition_cols(tmp_path, df):
+    p = str(tmp_path / "out")
+    bpd.from_pandas(df[["a", "c", "b"]]).to_parquet(p, partition_cols=["c"])
+    back = pd.read_parquet(p)
+    assert len(back) == len(df)
+    import os
+
+    assert any(d.startswith("c=") for d in os.listdir(p))
