@@ -35,10 +35,26 @@ def DataFrame(data=None, *args, **kwargs):
     return from_pandas_df(_pd.DataFrame(data, *args, **kwargs))
 
 
-def read_parquet(path, columns=None, **kwargs) -> BodoDataFrame:
+def read_parquet(path, columns=None, filters=None,
+                 **kwargs) -> BodoDataFrame:
+    """filters accepts the pandas/pyarrow DNF-lite list form
+    [(col, op, value), ...] (ANDed) and pushes into the scan."""
+    from ..plan.expr import Cmp, ColRef, Const, IsIn
+
     names = _pq.schema_names(path)
+    push = []
+    for f in (filters or []):
+        col, op, val = f
+        opmap = {"<": "lt", "<=": "le", ">": "gt", ">=": "ge", "==": "eq",
+                 "=": "eq", "!=": "ne"}
+        if op in opmap:
+            push.append(Cmp(opmap[op], ColRef(col), Const(val)))
+        elif op == "in":
+            push.append(IsIn(ColRef(col), tuple(val)))
+        else:
+            raise NotImplementedError(f"read_parquet filter op {op!r}")
     plan = _pn.ParquetScan(str(path), tuple(columns) if columns else None,
-                           (), tuple(names))
+                           tuple(push), tuple(names))
     return BodoDataFrame(plan, list(columns) if columns else list(names))
 
 

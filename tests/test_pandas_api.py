@@ -397,3 +397,14 @@ def test_to_parquet_partition_cols(tmp_path, df):
     import os
 
     assert any(d.startswith("c=") for d in os.listdir(p))
+
+
+def test_read_parquet_filters(tmp_path, df):
+    p = str(tmp_path / "t.parquet")
+    df[["a", "b"]].to_parquet(p, row_group_size=32)
+    out = bpd.read_parquet(p, filters=[("a", ">", 4)]).to_pandas()
+    exp = df[df.a > 4][["a", "b"]].reset_index(drop=True)
+    pd.testing.assert_frame_equal(
+        out.sort_values(["a", "b"]).reset_index(drop=True),
+        exp.sort_values(["a", "b"]).reset_index(drop=True),
+        check_dtype=False)
