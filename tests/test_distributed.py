@@ -465,3 +465,19 @@ def test_dist_any_all_skew():
         an=("d", "any"), al=("d", "all"),
         sk=("b", "skew")).sort_values("a").reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-6)
+
+
+def _q_dummies(bpd, rank, payload):
+    b = bpd.from_pandas(payload["df"])
+    return bpd.get_dummies(b.c, prefix="c")
+
+
+def test_dist_get_dummies():
+    """Category values come from a distributed unique() pass, so every
+    rank must emit the same column set."""
+    df = _df(400, 71)
+    got = run_dist(_q_dummies, {"df": df}).reset_index(drop=True)
+    exp = pd.get_dummies(df.c, prefix="c").astype(bool).reset_index(drop=True)
+    exp = exp[sorted(exp.columns)]
+    pd.testing.assert_frame_equal(got[sorted(got.columns)].astype(bool), exp,
+                                  check_dtype=False)
