@@ -118,3 +118,25 @@ def test_spawn_sql():
         print("SPAWN_SQL_OK")
     """)
     assert "SPAWN_SQL_OK" in out
+
+
+def test_spawn_window_rolling():
+    out = run_spawn_script("""
+        import warnings; warnings.filterwarnings("ignore")
+        import numpy as np, pandas as pd
+        import bodo_amd.pandas as bpd
+
+        rng = np.random.default_rng(7)
+        df = pd.DataFrame({"k": rng.integers(0, 6, 3000),
+                           "v": rng.uniform(0, 1, 3000)})
+        b = bpd.from_pandas(df)
+        b["t"] = b.groupby("k")["v"].transform("sum")
+        roll = b.v.rolling(5, min_periods=1).mean().to_pandas()
+        exp_roll = df.v.rolling(5, min_periods=1).mean().reset_index(drop=True)
+        pd.testing.assert_series_equal(roll, exp_roll, check_names=False)
+        got = b.to_pandas()
+        exp_t = df.groupby("k")["v"].transform("sum")
+        assert np.allclose(got["t"].to_numpy(), exp_t.to_numpy())
+        print("SPAWN_WINDOW_OK")
+    """)
+    assert "SPAWN_WINDOW_OK" in out
