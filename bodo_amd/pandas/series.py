@@ -117,6 +117,19 @@ class BodoSeries:
         target = _pd_dtype_to_bodo(dtype)
         return self._wrap(Cast(self._expr, target), None)
 
+    def replace(self, to_replace=None, value=None) -> "BodoSeries":
+        """Value replacement (reference: series replace overloads); lowers
+        to the UdfMap dictionary/LUT machinery."""
+        if isinstance(to_replace, dict):
+            d = dict(to_replace)
+            return self.map(lambda v: d.get(v, v))
+        if isinstance(to_replace, (list, tuple)):
+            rep = set(to_replace)
+            return self.map(lambda v, _r=rep, _val=value:
+                            _val if v in _r else v)
+        return self.map(lambda v, _t=to_replace, _val=value:
+                        _val if v == _t else v)
+
     def abs(self):
         zero = Const(0)
         return self._wrap(

@@ -408,3 +408,12 @@ def test_read_parquet_filters(tmp_path, df):
         out.sort_values(["a", "b"]).reset_index(drop=True),
         exp.sort_values(["a", "b"]).reset_index(drop=True),
         check_dtype=False)
+
+
+def test_series_replace(df):
+    b = bpd.from_pandas(df)
+    got = _decat(b.c.replace({"xx": "QQ"}).to_pandas()).astype(str)
+    exp = df.c.replace({"xx": "QQ"})
+    assert (got.to_numpy() == exp.to_numpy()).all()
+    gn = b.a.replace([0, 1], -1).to_pandas()
+    assert (gn.to_numpy() == df.a.replace([0, 1], -1).to_numpy()).all()
