@@ -464,6 +464,15 @@ class BodoDataFrame:
             pn.Projection(self._plan, tuple(self._columns), exprs),
             list(self._columns))
 
+    def replace(self, to_replace=None, value=None) -> "BodoDataFrame":
+        """Per-column value replacement (scalar or mapping applied to every
+        column; reference: frame replace overloads)."""
+        exprs = tuple(self[c].replace(to_replace, value)._expr
+                      for c in self._columns)
+        return BodoDataFrame(
+            pn.Projection(self._plan, tuple(self._columns), exprs),
+            list(self._columns))
+
     def nunique(self):
         return pd.Series({c: self[c].nunique() for c in self._columns})
 

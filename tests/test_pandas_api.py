@@ -417,3 +417,9 @@ def test_series_replace(df):
     assert (got.to_numpy() == exp.to_numpy()).all()
     gn = b.a.replace([0, 1], -1).to_pandas()
     assert (gn.to_numpy() == df.a.replace([0, 1], -1).to_numpy()).all()
+
+
+def test_frame_replace():
+    df = pd.DataFrame({"a": [1, 2, 3], "b": [2.0, 5.0, 2.0]})
+    got = bpd.from_pandas(df).replace(2, 99).to_pandas()
+    pd.testing.assert_frame_equal(got, df.replace(2, 99), check_dtype=False)
