@@ -551,6 +551,7 @@ _DT_OUT_TYPE = {
     "is_quarter_start": bt.boolean, "is_quarter_end": bt.boolean,
     "is_year_start": bt.boolean, "is_year_end": bt.boolean,
     "days_in_month": bt.int8, "daysinmonth": bt.int8,
+    "is_leap_year": bt.boolean,
     "trunc_month": bt.timestamp_ns, "trunc_year": bt.timestamp_ns,
     "trunc_quarter": bt.timestamp_ns, "trunc_week": bt.timestamp_ns,
 }

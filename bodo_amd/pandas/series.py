@@ -616,6 +616,10 @@ class _DtAccessor:
     def days_in_month(self):
         return self._f("days_in_month")
 
+    @property
+    def is_leap_year(self):
+        return self._f("is_leap_year")
+
     daysinmonth = days_in_month
 
     def month_name(self, locale=None):
