@@ -473,6 +473,23 @@ class BodoDataFrame:
             pn.Projection(self._plan, tuple(self._columns), exprs),
             list(self._columns))
 
+    def round(self, decimals=0) -> "BodoDataFrame":
+        from ..plan.expr import RoundExpr
+
+        per = decimals if isinstance(decimals, dict) else             {c: decimals for c in self._columns}
+        exprs = tuple(RoundExpr(ColRef(c), int(per[c])) if c in per
+                      else ColRef(c) for c in self._columns)
+        return BodoDataFrame(
+            pn.Projection(self._plan, tuple(self._columns), exprs),
+            list(self._columns))
+
+    def value_counts(self, subset=None, ascending=False):
+        cols = list(subset) if subset is not None else list(self._columns)
+        g = self.groupby(cols, as_index=False).agg(
+            count=pd.NamedAgg(cols[0], "size")).to_pandas()
+        g = g.sort_values("count", ascending=ascending)
+        return g.set_index(cols)["count"]
+
     def nunique(self):
         return pd.Series({c: self[c].nunique() for c in self._columns})
 

@@ -423,3 +423,13 @@ def test_frame_replace():
     df = pd.DataFrame({"a": [1, 2, 3], "b": [2.0, 5.0, 2.0]})
     got = bpd.from_pandas(df).replace(2, 99).to_pandas()
     pd.testing.assert_frame_equal(got, df.replace(2, 99), check_dtype=False)
+
+
+def test_frame_round_value_counts(df):
+    b = bpd.from_pandas(df)
+    got = b[["b"]].round(1).to_pandas()
+    pd.testing.assert_frame_equal(got, df[["b"]].round(1).reset_index(
+        drop=True), check_dtype=False)
+    vc = b.value_counts(subset=["a"])
+    exp = df.value_counts(subset=["a"])
+    assert vc.sum() == exp.sum() and vc.max() == exp.max()
