@@ -327,9 +327,13 @@ class BodoSeries:
 
     def _pair_moments(self, other: "BodoSeries"):
         """Distributed co-moments in ONE pass: n, Σx, Σy, Σxy, Σx², Σy²
-        (reference analog: sklearn_ext/array_kernels corr via allreduce)."""
+        (reference analog: sklearn_ext/array_kernels corr via allreduce).
+        Requires both series to share a source plan (the common case:
+        columns of one frame); mixed-plan pairs fall back to host."""
         from ..engine import api
 
+        if other._plan is not self._plan:
+            return None
         x, y = self._expr, other._expr
         # pairwise-complete observations (pandas semantics): drop rows where
         # either side is null before the moment sums
@@ -347,13 +351,19 @@ class BodoSeries:
                 float(row["sxx"].iloc[0]), float(row["syy"].iloc[0]))
 
     def cov(self, other: "BodoSeries", ddof: int = 1) -> float:
-        n, sx, sy, sxy, _, _ = self._pair_moments(other)
+        m = self._pair_moments(other)
+        if m is None:  # different source plans: align on host
+            return self.to_pandas().cov(other.to_pandas(), ddof=ddof)
+        n, sx, sy, sxy, _, _ = m
         if n <= ddof:
             return float("nan")
         return (sxy - sx * sy / n) / (n - ddof)
 
     def corr(self, other: "BodoSeries") -> float:
-        n, sx, sy, sxy, sxx, syy = self._pair_moments(other)
+        m = self._pair_moments(other)
+        if m is None:  # different source plans: align on host
+            return self.to_pandas().corr(other.to_pandas())
+        n, sx, sy, sxy, sxx, syy = m
         if n < 2:
             return float("nan")
         cov = sxy - sx * sy / n
@@ -448,6 +458,15 @@ class BodoSeries:
         raise NotImplementedError(
             "BodoSeries.ai (LLM/embedding calls) requires network access to "
             "a model endpoint; this build runs fully offline")
+
+    def autocorr(self, lag=1) -> float:
+        """Pearson autocorrelation at the given lag (distributed shift +
+        shared-plan co-moments)."""
+        plan = pn.Shift(self._as_projection_plan(), int(lag),
+                        (("__sh", "v"),))
+        a = BodoSeries(plan, ColRef("v"), None)
+        b = BodoSeries(plan, ColRef("__sh"), None)
+        return a.corr(b)
 
     def rolling(self, window, min_periods=None, **kwargs):
         return _RollingSeries(self, int(window), min_periods)

@@ -433,3 +433,10 @@ def test_frame_round_value_counts(df):
     vc = b.value_counts(subset=["a"])
     exp = df.value_counts(subset=["a"])
     assert vc.sum() == exp.sum() and vc.max() == exp.max()
+
+
+def test_autocorr(df):
+    b = bpd.from_pandas(df)
+    exp = df.b.autocorr(1)
+    got = b.b.autocorr(1)
+    assert abs(got - exp) < 1e-9
