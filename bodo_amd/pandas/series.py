@@ -31,11 +31,33 @@ class BodoSeries:
         return BodoSeries(self._plan, expr, name or self.name, self._frame)
 
     def _bin(self, op, other, reflect=False) -> "BodoSeries":
+        if isinstance(other, BodoSeries) and other._plan is not self._plan:
+            return self._host_combine(op, other, reflect)
         oe = other._expr if isinstance(other, BodoSeries) else as_expr(other)
         e = BinOp(op, oe, self._expr) if reflect else BinOp(op, self._expr, oe)
         return self._wrap(e, None)
 
+    def _host_combine(self, op, other, reflect=False) -> "BodoSeries":
+        """Series from DIFFERENT source plans: positional host alignment
+        (expressions can only reference one plan's columns)."""
+        import operator as _op
+
+        f = getattr(_op, {"add": "add", "sub": "sub", "mul": "mul",
+                          "div": "truediv", "floordiv": "floordiv",
+                          "mod": "mod", "pow": "pow", "lt": "lt", "le": "le",
+                          "gt": "gt", "ge": "ge", "eq": "eq",
+                          "ne": "ne"}[op])
+        a = self.to_pandas().reset_index(drop=True)
+        b2 = other.to_pandas().reset_index(drop=True)
+        res = f(b2, a) if reflect else f(a, b2)
+        from .frame import from_pandas_df
+
+        fr = from_pandas_df(res.to_frame(name="v"))
+        return BodoSeries(fr._lazy_plan, ColRef("v"), self.name)
+
     def _cmp(self, op, other) -> "BodoSeries":
+        if isinstance(other, BodoSeries) and other._plan is not self._plan:
+            return self._host_combine(op, other)
         oe = other._expr if isinstance(other, BodoSeries) else as_expr(other)
         return self._wrap(Cmp(op, self._expr, oe), None)
 

@@ -440,3 +440,11 @@ def test_autocorr(df):
     exp = df.b.autocorr(1)
     got = b.b.autocorr(1)
     assert abs(got - exp) < 1e-9
+
+
+def test_cross_frame_series_arithmetic():
+    a = bpd.from_pandas(pd.DataFrame({"x": [1.0, 2.0, 3.0]}))
+    b = bpd.from_pandas(pd.DataFrame({"y": [10.0, 20.0, 30.0]}))
+    assert (a.x + b.y).to_pandas().tolist() == [11.0, 22.0, 33.0]
+    assert (b.y - a.x).to_pandas().tolist() == [9.0, 18.0, 27.0]
+    assert (a.x > b.y).to_pandas().tolist() == [False] * 3
