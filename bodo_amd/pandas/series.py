@@ -164,6 +164,16 @@ class BodoSeries:
         return self._wrap(RoundExpr(self._expr, decimals), None)
 
     def where(self, cond, other) -> "BodoSeries":
+        if any(isinstance(x, BodoSeries) and x._plan is not self._plan
+               for x in (cond, other)):
+            # cross-plan operands: positional host alignment
+            cv = cond.to_pandas().reset_index(drop=True)                 if isinstance(cond, BodoSeries) else cond
+            ov = other.to_pandas().reset_index(drop=True)                 if isinstance(other, BodoSeries) else other
+            res = self.to_pandas().reset_index(drop=True).where(cv, ov)
+            from .frame import from_pandas_df
+
+            fr = from_pandas_df(res.to_frame(name="v"))
+            return BodoSeries(fr._lazy_plan, ColRef("v"), self.name)
         ce = cond._expr if isinstance(cond, BodoSeries) else as_expr(cond)
         oe = other._expr if isinstance(other, BodoSeries) else as_expr(other)
         return self._wrap(Case((ce,), (self._expr,), oe), None)
