@@ -139,6 +139,11 @@ class BodoDataFrame:
                 return self._semi_filter(e, anti=False)
             if isinstance(e, _Not) and isinstance(e.operand, _SJ)                     and isinstance(e.operand.operand, ColRef):
                 return self._semi_filter(e.operand, anti=True)
+            if key._plan is not self._plan:
+                # mask from a different frame: positional host alignment
+                return self._fallback(
+                    "__getitem__",
+                    key.to_pandas().reset_index(drop=True))
             return BodoDataFrame(pn.Filter(self._plan, key._expr), self._columns)
         if isinstance(key, pd.Series) and key.dtype == bool:
             # host boolean mask: materialize path
