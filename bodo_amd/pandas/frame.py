@@ -234,8 +234,17 @@ class BodoDataFrame:
 
                 refs = expr_columns(value._expr)
                 if not refs.issubset(set(self._columns)):
-                    raise ValueError(
-                        "cannot assign a Series derived from a different frame")
+                    # series from an unrelated frame: positional host align
+                    pdf = self.to_pandas().reset_index(drop=True)
+                    pdf[key] = value.to_pandas().reset_index(
+                        drop=True).to_numpy()
+                    nf = from_pandas_df(pdf)
+                    object.__setattr__(self, "_plan", nf._plan)
+                    object.__setattr__(self, "_columns",
+                                       list(nf._columns))
+                    object.__setattr__(self, "_result", None)
+                    object.__setattr__(self, "_remote", None)
+                    return
             expr = value._expr
         else:
             from ..plan.expr import Const

@@ -448,3 +448,11 @@ def test_cross_frame_series_arithmetic():
     assert (a.x + b.y).to_pandas().tolist() == [11.0, 22.0, 33.0]
     assert (b.y - a.x).to_pandas().tolist() == [9.0, 18.0, 27.0]
     assert (a.x > b.y).to_pandas().tolist() == [False] * 3
+
+
+def test_cross_frame_setitem():
+    a = bpd.from_pandas(pd.DataFrame({"x": [1, 2, 3]}))
+    b = bpd.from_pandas(pd.DataFrame({"y": [10.0, 20.0, 30.0]}))
+    a["z"] = b.y
+    out = a.to_pandas()
+    assert out.z.tolist() == [10.0, 20.0, 30.0]
