@@ -124,6 +124,16 @@ class BodoSeries:
     notnull = notna
 
     def fillna(self, value) -> "BodoSeries":
+        if isinstance(value, BodoSeries):
+            if value._plan is self._plan:
+                return self._wrap(Case((IsNull(self._expr),),
+                                       (value._expr,), self._expr), None)
+            res = self.to_pandas().reset_index(drop=True).fillna(
+                value.to_pandas().reset_index(drop=True))
+            from .frame import from_pandas_df
+
+            fr = from_pandas_df(res.to_frame(name="v"))
+            return BodoSeries(fr._lazy_plan, ColRef("v"), self.name)
         return self._wrap(
             Case((IsNull(self._expr),), (as_expr(value),), self._expr), None)
 
