@@ -120,7 +120,8 @@ def merge(left, right, **kwargs) -> BodoDataFrame:
 
 def concat(objs, axis=0, ignore_index=False, **kwargs):
     objs = list(objs)
-    if axis in (0, "index") and all(isinstance(o, BodoDataFrame) for o in objs):
+    if axis in (0, "index") and all(isinstance(o, BodoDataFrame) for o in objs) \
+            and all(set(o._columns) == set(objs[0]._columns) for o in objs):
         plans = tuple(o._lazy_plan for o in objs)
         return BodoDataFrame(_pn.Union(plans), objs[0]._columns)
     warnings.warn("concat: falling back to pandas")
