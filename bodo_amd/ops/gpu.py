@@ -659,6 +659,15 @@ def distinct_local(tbl: Table, subset=None, keep: str = "first") -> Table:
             None, int(TypeKind.INT8), row_gid, int(uniq_rows.numel()),
             _AGG_OP["last_row"], 0.0, _I64_MIN, False)[0]
         rows = acc
+    elif keep is False:
+        # drop every member of any duplicated group
+        K = kernels()
+        cnt = K.agg_update(
+            torch.zeros(len(tbl), dtype=torch.int8, device=tbl.device), None,
+            None, int(TypeKind.INT8), row_gid, int(uniq_rows.numel()),
+            _AGG_OP["size"], 0.0, 0, True)[1]
+        singles = cnt[row_gid.long()] == 1
+        rows = torch.nonzero(singles, as_tuple=False).reshape(-1)
     else:
         rows = uniq_rows
     rows = torch.sort(rows).values  # preserve original row order

@@ -396,3 +396,18 @@ def test_gpu_groupby_any_all_skew():
         sk=("v", "skew")).sort_values("k").reset_index(drop=True)
     # raw-moment skew carries ~1e-8 cancellation error vs pandas' two-pass
     pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_gpu_distinct_keep_variants():
+    import bodo_amd.config as cfg
+
+    cfg.DEVICE = "cuda"
+    rng = np.random.default_rng(35)
+    df = pd.DataFrame({"a": rng.integers(0, 500, 50_000),
+                       "b": np.arange(50_000)})
+    b = bpd.from_pandas(df)
+    for keep in ["first", "last", False]:
+        got = b.drop_duplicates(subset=["a"], keep=keep).to_pandas()
+        exp = df.drop_duplicates(subset=["a"], keep=keep)
+        assert sorted(got.b.tolist()) == sorted(exp.b.tolist()), keep
