@@ -481,3 +481,16 @@ def test_dist_get_dummies():
     exp = exp[sorted(exp.columns)]
     pd.testing.assert_frame_equal(got[sorted(got.columns)].astype(bool), exp,
                                   check_dtype=False)
+
+
+def _q_distinct_keep(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).drop_duplicates(
+        subset=["a"], keep=payload["keep"])
+
+
+def test_dist_distinct_keep_variants():
+    df = _df(800, 81)
+    for keep in ["first", "last", False]:
+        got = run_dist(_q_distinct_keep, {"df": df, "keep": keep})
+        exp = df.drop_duplicates(subset=["a"], keep=keep)
+        assert sorted(got.b.tolist()) == sorted(exp.b.tolist()), keep
