@@ -494,3 +494,18 @@ def test_dist_distinct_keep_variants():
         got = run_dist(_q_distinct_keep, {"df": df, "keep": keep})
         exp = df.drop_duplicates(subset=["a"], keep=keep)
         assert sorted(got.b.tolist()) == sorted(exp.b.tolist()), keep
+
+
+def _q_offset(bpd, rank, payload):
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": bpd.from_pandas(payload["df"])})
+    return bc.sql("select b from t order by b limit 9 offset 4")
+
+
+def test_dist_limit_offset():
+    df = _df(300, 91)
+    got = run_dist(_q_offset, {"df": df}).reset_index(drop=True)
+    exp = df.sort_values("b").reset_index(drop=True).iloc[4:13][
+        ["b"]].reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
