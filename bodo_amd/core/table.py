@@ -12,8 +12,19 @@ import torch
 from .column import Column
 
 
+def _normalize_device(device) -> torch.device:
+    """Resolve a bare ``cuda`` device to an explicit index so equality checks
+    compare (type, index), not just type.  Under the 1-process-per-GPU model
+    the current device is the only visible one."""
+    dev = torch.device(device)
+    if dev.type == "cuda" and dev.index is None:
+        idx = torch.cuda.current_device() if torch.cuda.is_available() else 0
+        dev = torch.device("cuda", idx)
+    return dev
+
+
 class Table:
-    __slots__ = ("names", "columns", "_length")
+    __slots__ = ("names", "columns", "_length", "_dev_cache")
 
     def __init__(self, names: Sequence[str], columns: Sequence[Column],
                  length: Optional[int] = None):
@@ -23,6 +34,9 @@ class Table:
         if length is None:
             length = len(columns[0]) if columns else 0
         self._length = length
+        # device copies cached per target: repeated query runs over the same
+        # registered source must not re-upload the table each execution
+        self._dev_cache: Dict[str, "Table"] = {}
 
     def __len__(self) -> int:
         return self._length
@@ -62,22 +76,16 @@ class Table:
         return self.columns[0].device
 
     def to_device(self, device) -> "Table":
-        dev = torch.device(device)
-        if dev == self.device or (dev.type == self.device.type
-                                  and dev.type == "cuda"):
+        dev = _normalize_device(device)
+        if dev == _normalize_device(self.device):
             return self
-        # device copies cache per target: repeated query runs over the same
-        # registered source must not re-upload the table each execution
-        cache = getattr(self, "_dev_cache", None)
-        if cache is None:
-            cache = {}
-            object.__setattr__(self, "_dev_cache", cache)
+        cache = self._dev_cache
         key = str(dev)
         hit = cache.get(key)
         if hit is not None:
             return hit
         out = Table(self.names,
-                    [c.to_device(device) for c in self.columns], self._length)
+                    [c.to_device(dev) for c in self.columns], self._length)
         cache[key] = out
         return out
 

@@ -247,6 +247,16 @@ def _exec_sort(node: pn.Sort, ctx) -> Table:
     return ooc.sort_local(child, list(node.keys), asc, node.na_position)
 
 
+def _sample_positions(n: int, k: int, ctx, device) -> torch.Tensor:
+    """Seeded per-rank sample positions so distributed sorts are reproducible
+    across runs (splitters otherwise change run to run)."""
+    if n <= k:
+        return torch.arange(n, device=device)
+    g = torch.Generator(device="cpu")
+    g.manual_seed(0x5EED5 + 1315423911 * (ctx.rank + 1))
+    return torch.randint(0, n, (k,), generator=g).to(device)
+
+
 def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:
     """Sample-based range partitioning so rank r holds globally-contiguous key
     range r (PSRS step 1-3; reference: streaming/_sort.h reservoir sampling +
@@ -258,7 +268,7 @@ def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:
     if packed is not None:
         k = min(n, 64 * w)
         if n > 0:
-            pos = torch.randint(0, n, (k,), device=tbl.device) if n > k else                 torch.arange(n, device=tbl.device)
+            pos = _sample_positions(n, k, ctx, tbl.device)
             sample = packed[pos].cpu().numpy()
         else:
             sample = np.zeros(0, dtype=np.int64)
@@ -273,8 +283,7 @@ def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:
     # sample up to 64*w rows of the key columns
     k = min(n, 64 * w)
     if k > 0:
-        pos = torch.randint(0, n, (k,), device=tbl.device) if n > k else \
-            torch.arange(n, device=tbl.device)
+        pos = _sample_positions(n, k, ctx, tbl.device)
         sample = ops.take_table(tbl.select(list(keys)), pos).to_device("cpu").to_pandas()
     else:
         sample = tbl.select(list(keys)).to_device("cpu").to_pandas()
@@ -375,12 +384,23 @@ def _exec_limit(node: pn.Limit, ctx) -> Table:
 
 def _exec_distinct(node: pn.Distinct, ctx) -> Table:
     child = _exec(node.child, ctx)
+    if ctx.world == 1:
+        return rel.distinct_local(child, node.subset, node.keep)
+    # keep=False must see every occurrence of a key to know whether it is
+    # duplicated anywhere: a pre-shuffle local distinct destroys per-key
+    # counts (a key with 2 rows here + 1 elsewhere would wrongly survive).
+    # Shuffle the full table by key first, then drop once.
+    if node.keep is False:
+        keys = list(node.subset) if node.subset else list(child.names)
+        shuffled = _shuffle_by_keys(child, keys, ctx)
+        return rel.distinct_local(shuffled, node.subset, node.keep)
+    # keep='first'/'last': local pass is a safe reducer because the shuffle
+    # concatenates inbound shards in source-rank order, preserving global
+    # row order per key across the two passes.
     local = rel.distinct_local(child, node.subset, node.keep)
-    if ctx.world > 1:
-        keys = list(node.subset) if node.subset else list(local.names)
-        shuffled = _shuffle_by_keys(local, keys, ctx)
-        local = rel.distinct_local(shuffled, node.subset, node.keep)
-    return local
+    keys = list(node.subset) if node.subset else list(local.names)
+    shuffled = _shuffle_by_keys(local, keys, ctx)
+    return rel.distinct_local(shuffled, node.subset, node.keep)
 
 
 def _exec_sample(node: pn.Sample, ctx) -> Table:

@@ -181,10 +181,7 @@ class _ChunkReader:
             comp_size = hdr.get(3)
             raw = self.buf[body:body + comp_size]
             if codec is not None and ptype in (PAGE_DICT, PAGE_DATA):
-                pb = codec.decompress(
-                    raw, decompressed_size=hdr.get(2)).to_pybytes()                     if hasattr(codec.decompress(b"\x00", decompressed_size=1),
-                               "to_pybytes") else codec.decompress(
-                        raw, decompressed_size=hdr.get(2))
+                pb = codec.decompress(raw, decompressed_size=hdr.get(2))
             else:
                 pb = raw
             if isinstance(pb, pa.Buffer):

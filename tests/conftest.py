@@ -29,3 +29,14 @@ def _quiet_warnings():
     with warnings.catch_warnings():
         warnings.simplefilter("ignore")
         yield
+
+
+@pytest.fixture(autouse=True)
+def _restore_device_config():
+    """Tests that flip bodo_amd.config.DEVICE must not leak it into later
+    tests (round-1 hygiene finding: in-process config reloads leaked)."""
+    import bodo_amd.config as cfg
+
+    old = cfg.DEVICE
+    yield
+    cfg.DEVICE = old

@@ -496,6 +496,21 @@ def test_dist_distinct_keep_variants():
         assert sorted(got.b.tolist()) == sorted(exp.b.tolist()), keep
 
 
+def test_dist_distinct_keep_false_cross_rank_counts():
+    """A key duplicated on one rank AND present on the other must still be
+    dropped entirely under keep=False (regression: the old pre-shuffle local
+    distinct destroyed per-key counts, leaving a lone survivor)."""
+    df = pd.DataFrame({"a": [1, 2, 3, 5, 5, 4, 5, 6],
+                       "b": [10, 20, 30, 50, 51, 40, 52, 60]})
+    got = run_dist(_q_distinct_keep, {"df": df, "keep": False})
+    exp = df.drop_duplicates(subset=["a"], keep=False)
+    assert sorted(got.b.tolist()) == sorted(exp.b.tolist())
+    for keep in ["first", "last"]:
+        got = run_dist(_q_distinct_keep, {"df": df, "keep": keep})
+        exp = df.drop_duplicates(subset=["a"], keep=keep)
+        assert sorted(got.b.tolist()) == sorted(exp.b.tolist()), keep
+
+
 def _q_offset(bpd, rank, payload):
     from bodo_amd.sql import BodoSQLContext
 

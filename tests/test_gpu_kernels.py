@@ -190,13 +190,9 @@ def test_distinct_gpu():
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
 
 
-def test_taxi_q1_gpu_end_to_end(monkeypatch):
-    monkeypatch.setenv("BODO_AMD_DEVICE", "cuda")
-    import importlib
-
+def test_taxi_q1_gpu_end_to_end():
     import bodo_amd.config as cfg
 
-    importlib.reload(cfg)
     cfg.DEVICE = "cuda"
     import bodo_amd.pandas as bpd
     from tests.test_queries import make_taxi, nyc_taxi_q1
@@ -207,10 +203,9 @@ def test_taxi_q1_gpu_end_to_end(monkeypatch):
     exp = nyc_taxi_q1(pd, trips.copy(), weather.copy()).reset_index(drop=True)
     got["time_bucket"] = got["time_bucket"].astype(str)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
-    cfg.DEVICE = ""
 
 
-def test_tpch_q1_gpu_end_to_end(monkeypatch):
+def test_tpch_q1_gpu_end_to_end():
     import bodo_amd.config as cfg
 
     cfg.DEVICE = "cuda"
@@ -224,7 +219,6 @@ def test_tpch_q1_gpu_end_to_end(monkeypatch):
         got[c] = got[c].astype(str)
         exp[c] = exp[c].astype(str)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
-    cfg.DEVICE = ""
 
 
 def test_gpu_parquet_decode(tmp_path):
