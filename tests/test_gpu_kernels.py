@@ -273,6 +273,8 @@ def test_gpu_cumulative_and_shift():
     import bodo_amd.config as cfg
 
     cfg.DEVICE = "cuda"
+    import bodo_amd.pandas as bpd
+
     rng = np.random.default_rng(11)
     x = rng.random(200_000)
     x[rng.random(200_000) < 0.05] = np.nan
@@ -356,6 +358,8 @@ def test_parquet_gpu_snappy_e2e():
     import bodo_amd.config as cfg
 
     cfg.DEVICE = "cuda"
+    import bodo_amd.pandas as bpd
+
     rng = np.random.default_rng(2)
     df = pd.DataFrame({"k": rng.integers(0, 20, 300_000),
                        "v": rng.random(300_000)})
@@ -376,6 +380,8 @@ def test_gpu_groupby_any_all_skew():
     import bodo_amd.config as cfg
 
     cfg.DEVICE = "cuda"
+    import bodo_amd.pandas as bpd
+
     rng = np.random.default_rng(34)
     df = pd.DataFrame({"k": rng.integers(0, 20, 100_000),
                        "f": rng.random(100_000) > 0.3,
@@ -397,6 +403,8 @@ def test_gpu_distinct_keep_variants():
     import bodo_amd.config as cfg
 
     cfg.DEVICE = "cuda"
+    import bodo_amd.pandas as bpd
+
     rng = np.random.default_rng(35)
     df = pd.DataFrame({"a": rng.integers(0, 500, 50_000),
                        "b": np.arange(50_000)})
