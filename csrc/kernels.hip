@@ -1210,6 +1210,34 @@ torch::Tensor rle_expand(torch::Tensor runs_blob, int64_t n_runs,
 
 torch::Tensor gemm_f32(torch::Tensor A, torch::Tensor B);  // gemm.hip
 
+// parquet.hip: page-parallel on-GPU parquet decode
+void pq_decompress(torch::Tensor src, torch::Tensor pages_blob,
+                   int64_t n_pages, torch::Tensor scratch);
+std::vector<torch::Tensor> pq_def_levels(torch::Tensor scratch,
+                                         torch::Tensor pages_blob,
+                                         int64_t n_pages, int64_t bitwidth,
+                                         int64_t max_def, int64_t total_nv);
+torch::Tensor pq_expand_codes(torch::Tensor scratch, torch::Tensor pages_blob,
+                              int64_t n_pages,
+                              c10::optional<torch::Tensor> val_data_off,
+                              torch::Tensor dense_off,
+                              c10::optional<torch::Tensor> n_valid,
+                              int64_t dense_total);
+torch::Tensor pq_copy_fixed(torch::Tensor scratch, torch::Tensor pages_blob,
+                            int64_t n_pages,
+                            c10::optional<torch::Tensor> val_data_off,
+                            torch::Tensor dense_off,
+                            c10::optional<torch::Tensor> n_valid,
+                            int64_t esize, int64_t dense_total);
+std::vector<torch::Tensor> pq_byte_array_lengths(
+    torch::Tensor scratch, torch::Tensor pages_blob, int64_t n_pages,
+    c10::optional<torch::Tensor> val_data_off, torch::Tensor dense_off,
+    c10::optional<torch::Tensor> n_valid, int64_t dense_total);
+torch::Tensor pq_copy_strings(torch::Tensor scratch, torch::Tensor src_abs,
+                              torch::Tensor dst_off, torch::Tensor lengths,
+                              int64_t n, int64_t total_bytes);
+torch::Tensor pq_parse_headers(torch::Tensor chunk_buf);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_f32", &gemm_f32, "f32 MFMA GEMM (v_mfma_f32_16x16x4_f32)");
   m.def("rle_expand", &rle_expand, "parquet RLE/bit-packed hybrid expand");
@@ -1223,5 +1251,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("agg_update", &agg_update, "aggregate update");
   m.def("join_build", &join_build, "hash join build");
   m.def("join_probe", &join_probe, "hash join probe");
+  m.def("pq_decompress", &pq_decompress,
+        "page-parallel snappy decompress (one wave/page)");
+  m.def("pq_def_levels", &pq_def_levels,
+        "definition levels -> validity mask + per-page valid counts");
+  m.def("pq_expand_codes", &pq_expand_codes,
+        "RLE/bit-packed dictionary codes -> dense int32");
+  m.def("pq_copy_fixed", &pq_copy_fixed,
+        "PLAIN fixed-width page values -> dense buffer");
+  m.def("pq_byte_array_lengths", &pq_byte_array_lengths,
+        "PLAIN BYTE_ARRAY length/offset walk");
+  m.def("pq_copy_strings", &pq_copy_strings,
+        "gather string bytes to packed buffer");
+  m.def("pq_parse_headers", &pq_parse_headers,
+        "host-side thrift page-header parse of a chunk buffer");
   m.attr("_native") = true;
 }

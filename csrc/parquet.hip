@@ -1,0 +1,670 @@
+// On-GPU Parquet decode kernels for gfx950 (MI355X): page-parallel snappy
+// decompression, definition-level -> validity-mask expansion, RLE/bit-packed
+// dictionary-code expansion, fixed-width value copy and BYTE_ARRAY (string)
+// assembly.  One 64-lane wavefront per page: lane 0 parses the sequential
+// byte stream (tags / run headers / varints) and broadcasts work items with
+// __shfl; all 64 lanes execute the copies/expansions cooperatively.  Pages
+// are the parallel axis — a shard decode runs thousands of pages at once,
+// which fills the 256 CUs.
+//
+// Also hosts the C++ (host-side) page-header parser: a minimal Thrift
+// compact-protocol walk over the chunk buffer, returning a page table as
+// numpy-compatible tensors (the Python version cost ~10-30us/page).
+//
+// Reference behavioral spec (no code reuse): parquet-format spec;
+// reference role: cudf::io::read_parquet used by
+// bodo/pandas/physical/gpu_read_parquet.h and bodo/io/parquet_reader.cpp.
+
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include <cstring>
+#include <vector>
+
+#include "common.h"
+
+#define CHECK_HIP_PQ(x)                                                 \
+  do {                                                                  \
+    hipError_t e = (x);                                                 \
+    TORCH_CHECK(e == hipSuccess, "HIP error: ", hipGetErrorString(e));  \
+  } while (0)
+
+static hipStream_t pq_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+// Per data-page metadata, filled on host from page headers.
+struct PageMeta {
+  int64_t src_off;   // compressed page body offset in the chunk buffer
+  int64_t dst_off;   // decompressed offset in scratch (8-byte aligned)
+  int64_t val_off;   // first value index of this page within the chunk
+  int32_t src_len;   // compressed body length
+  int32_t dst_len;   // decompressed length
+  int32_t nv;        // value count incl. nulls
+  int32_t flags;     // bit0: page has def-level section; bit1: snappy
+};
+
+#define PQF_HAS_DEF 1
+#define PQF_SNAPPY 2
+
+// ---------------------------------------------------------------------
+// snappy decompression, one wave per page
+// ---------------------------------------------------------------------
+
+__global__ void pq_decompress_kernel(const uint8_t* __restrict__ src,
+                                     const PageMeta* __restrict__ pages,
+                                     int n_pages, uint8_t* __restrict__ dst) {
+  int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  for (int64_t p = wave; p < n_pages; p += n_waves) {
+    PageMeta pm = pages[p];
+    const uint8_t* in = src + pm.src_off;
+    uint8_t* out = dst + pm.dst_off;
+    if (!(pm.flags & PQF_SNAPPY)) {
+      // uncompressed page: cooperative copy into scratch
+      for (int64_t i = lane; i < pm.dst_len; i += WAVE) out[i] = in[i];
+      continue;
+    }
+    int64_t ip = 0, opos = 0;
+    int64_t src_len = pm.src_len, dst_len = pm.dst_len;
+    if (lane == 0) {  // skip the uncompressed-length varint preamble
+      while (in[ip] & 0x80) ip++;
+      ip++;
+    }
+    ip = __shfl(ip, 0);
+    while (ip < src_len && opos < dst_len) {
+      int64_t v0 = 0, v1 = 0;
+      if (lane == 0) {
+        uint8_t tag = in[ip];
+        int k = tag & 3;
+        if (k == 0) {  // literal
+          int64_t len = (tag >> 2) + 1;
+          int hl = 1;
+          if (len > 60) {
+            int nb = (int)(len - 60);
+            len = 0;
+            for (int i = 0; i < nb; ++i)
+              len |= (int64_t)in[ip + 1 + i] << (8 * i);
+            len += 1;
+            hl = 1 + nb;
+          }
+          v0 = (len << 8);          // kind 0 in low byte
+          v1 = ip + hl;             // literal source offset
+          ip += hl + len;
+        } else {
+          int64_t len, off;
+          int hl;
+          if (k == 1) {
+            len = ((tag >> 2) & 7) + 4;
+            off = ((int64_t)(tag >> 5) << 8) | in[ip + 1];
+            hl = 2;
+          } else if (k == 2) {
+            len = (tag >> 2) + 1;
+            off = (int64_t)in[ip + 1] | ((int64_t)in[ip + 2] << 8);
+            hl = 3;
+          } else {
+            len = (tag >> 2) + 1;
+            off = (int64_t)in[ip + 1] | ((int64_t)in[ip + 2] << 8) |
+                  ((int64_t)in[ip + 3] << 16) | ((int64_t)in[ip + 4] << 24);
+            hl = 5;
+          }
+          v0 = (len << 8) | 1;      // kind 1: copy
+          v1 = off;
+          ip += hl;
+        }
+      }
+      v0 = __shfl(v0, 0);
+      v1 = __shfl(v1, 0);
+      ip = __shfl(ip, 0);
+      int64_t len = v0 >> 8;
+      if (len <= 0 || opos + len > dst_len) break;  // corrupt: stop
+      if ((v0 & 0xFF) == 0) {
+        const uint8_t* s = in + v1;
+        for (int64_t i = lane; i < len; i += WAVE) out[opos + i] = s[i];
+      } else {
+        int64_t off = v1;
+        if (off <= 0 || off > opos) break;  // corrupt
+        const uint8_t* s = out + opos - off;
+        if (off >= len) {
+          for (int64_t i = lane; i < len; i += WAVE) out[opos + i] = s[i];
+        } else {  // overlapping copy = pattern replication
+          for (int64_t i = lane; i < len; i += WAVE) out[opos + i] = s[i % off];
+        }
+      }
+      opos += len;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// definition levels -> validity mask (+ per-page valid counts and the
+// byte offset of the values section, which is only knowable after
+// decompression because lvl_len lives inside the page)
+// ---------------------------------------------------------------------
+
+DEV_INLINE int wave_sum_i32(int v) {
+  for (int off = WAVE / 2; off > 0; off >>= 1) v += __shfl_down(v, off);
+  return __shfl(v, 0);
+}
+
+DEV_INLINE int64_t dev_varint(const uint8_t* b, int64_t* pos) {
+  int64_t out = 0;
+  int shift = 0;
+  while (true) {
+    uint8_t v = b[*pos];
+    (*pos)++;
+    out |= (int64_t)(v & 0x7F) << shift;
+    if (!(v & 0x80)) return out;
+    shift += 7;
+  }
+}
+
+__global__ void pq_def_levels_kernel(const uint8_t* __restrict__ scratch,
+                                     const PageMeta* __restrict__ pages,
+                                     int n_pages, int bitwidth, int max_def,
+                                     uint8_t* __restrict__ mask,
+                                     int32_t* __restrict__ n_valid,
+                                     int32_t* __restrict__ val_data_off) {
+  int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  for (int64_t p = wave; p < n_pages; p += n_waves) {
+    PageMeta pm = pages[p];
+    const uint8_t* base = scratch + pm.dst_off;
+    uint8_t* m = mask + pm.val_off;
+    int64_t nv = pm.nv;
+    if (!(pm.flags & PQF_HAS_DEF)) {
+      for (int64_t i = lane; i < nv; i += WAVE) m[i] = 1;
+      if (lane == 0) {
+        n_valid[p] = (int32_t)nv;
+        val_data_off[p] = 0;
+      }
+      continue;
+    }
+    int32_t lvl_len = (int32_t)base[0] | ((int32_t)base[1] << 8) |
+                      ((int32_t)base[2] << 16) | ((int32_t)base[3] << 24);
+    if (lane == 0) val_data_off[p] = 4 + lvl_len;
+    const uint8_t* lv = base + 4;
+    int64_t pos = 0, out = 0;
+    int width_bytes = (bitwidth + 7) / 8;
+    int valid = 0;
+    while (out < nv && pos < lvl_len) {
+      int64_t h = 0, val = 0, bitoff = 0, count;
+      if (lane == 0) h = dev_varint(lv, &pos);
+      h = __shfl(h, 0);
+      if (h & 1) {  // bit-packed groups
+        int64_t groups = h >> 1;
+        count = groups * 8;
+        if (count > nv - out) count = nv - out;
+        bitoff = __shfl(pos, 0) * 8;
+        int my = 0;
+        for (int64_t i = lane; i < count; i += WAVE) {
+          int64_t bp = bitoff + i * bitwidth;
+          uint32_t w = (uint32_t)lv[bp >> 3] | ((uint32_t)lv[(bp >> 3) + 1] << 8);
+          int v = (int)((w >> (bp & 7)) & ((1u << bitwidth) - 1));
+          uint8_t ok = (v == max_def);
+          m[out + i] = ok;
+          my += ok;
+        }
+        valid += wave_sum_i32(my);
+        if (lane == 0) pos += groups * bitwidth;
+        pos = __shfl(pos, 0);
+      } else {  // RLE run
+        count = h >> 1;
+        if (count > nv - out) count = nv - out;
+        if (lane == 0) {
+          val = 0;
+          for (int i = 0; i < width_bytes; ++i)
+            val |= (int64_t)lv[pos + i] << (8 * i);
+          pos += width_bytes;
+        }
+        val = __shfl(val, 0);
+        pos = __shfl(pos, 0);
+        uint8_t ok = (val == max_def);
+        for (int64_t i = lane; i < count; i += WAVE) m[out + i] = ok;
+        if (ok) valid += (int)count;
+      }
+      out += count;
+    }
+    if (lane == 0) n_valid[p] = valid;
+  }
+}
+
+// ---------------------------------------------------------------------
+// RLE/bit-packed dictionary codes -> dense int32 codes
+// ---------------------------------------------------------------------
+
+__global__ void pq_expand_codes_kernel(const uint8_t* __restrict__ scratch,
+                                       const PageMeta* __restrict__ pages,
+                                       int n_pages,
+                                       const int32_t* __restrict__ val_data_off,
+                                       const int64_t* __restrict__ dense_off,
+                                       const int32_t* __restrict__ n_valid,
+                                       int32_t* __restrict__ out) {
+  int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  for (int64_t p = wave; p < n_pages; p += n_waves) {
+    PageMeta pm = pages[p];
+    const uint8_t* base = scratch + pm.dst_off +
+                          (val_data_off ? val_data_off[p] : 0);
+    int32_t* o = out + dense_off[p];
+    int64_t nval = n_valid ? (int64_t)n_valid[p] : (int64_t)pm.nv;
+    int bitwidth = base[0];
+    const uint8_t* d = base + 1;
+    if (bitwidth == 0) {
+      for (int64_t i = lane; i < nval; i += WAVE) o[i] = 0;
+      continue;
+    }
+    int width_bytes = (bitwidth + 7) / 8;
+    int64_t pos = 0, outp = 0;
+    int64_t dlen = pm.dst_len - (base - (scratch + pm.dst_off));
+    while (outp < nval && pos < dlen) {
+      int64_t h = 0, val = 0, count;
+      if (lane == 0) h = dev_varint(d, &pos);
+      h = __shfl(h, 0);
+      if (h & 1) {
+        int64_t groups = h >> 1;
+        count = groups * 8;
+        if (count > nval - outp) count = nval - outp;
+        int64_t bitoff = __shfl(pos, 0) * 8;
+        for (int64_t i = lane; i < count; i += WAVE) {
+          int64_t bp = bitoff + i * bitwidth;
+          int64_t byte = bp >> 3;
+          uint32_t w = (uint32_t)d[byte] | ((uint32_t)d[byte + 1] << 8) |
+                       ((uint32_t)d[byte + 2] << 16) |
+                       ((uint32_t)d[byte + 3] << 24);
+          o[outp + i] = (int32_t)((w >> (bp & 7)) & ((1u << bitwidth) - 1));
+        }
+        if (lane == 0) pos += groups * bitwidth;
+        pos = __shfl(pos, 0);
+      } else {
+        count = h >> 1;
+        if (count > nval - outp) count = nval - outp;
+        if (lane == 0) {
+          val = 0;
+          for (int i = 0; i < width_bytes; ++i)
+            val |= (int64_t)d[pos + i] << (8 * i);
+          pos += width_bytes;
+        }
+        val = __shfl(val, 0);
+        pos = __shfl(pos, 0);
+        int32_t v = (int32_t)val;
+        for (int64_t i = lane; i < count; i += WAVE) o[outp + i] = v;
+      }
+      outp += count;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// PLAIN fixed-width values -> dense value buffer (byte copy; source is
+// unaligned inside the decompressed page, destination is esize-aligned)
+// ---------------------------------------------------------------------
+
+__global__ void pq_copy_fixed_kernel(const uint8_t* __restrict__ scratch,
+                                     const PageMeta* __restrict__ pages,
+                                     int n_pages,
+                                     const int32_t* __restrict__ val_data_off,
+                                     const int64_t* __restrict__ dense_off,
+                                     const int32_t* __restrict__ n_valid,
+                                     int esize, uint8_t* __restrict__ out) {
+  // one BLOCK per page slot (pages are up to ~1 MiB: need more than a wave).
+  // Source is byte-misaligned inside the decompressed page; do shift-merged
+  // aligned u64 loads (scratch is padded by 16 bytes so the +1 word read
+  // never faults) and aligned stores sized to the destination alignment.
+  for (int64_t p = blockIdx.x; p < n_pages; p += gridDim.x) {
+    PageMeta pm = pages[p];
+    const uint8_t* s = scratch + pm.dst_off +
+                       (val_data_off ? val_data_off[p] : 0);
+    uint8_t* o = out + dense_off[p] * esize;
+    int64_t nbytes = (n_valid ? (int64_t)n_valid[p] : (int64_t)pm.nv) * esize;
+    int m = (int)((uintptr_t)s & 7);
+    const uint64_t* sa = (const uint64_t*)(s - m);
+    int sh = m * 8;
+    if ((((uintptr_t)o) & 7) == 0) {
+      int64_t nw = nbytes / 8;
+      if (sh == 0) {
+        for (int64_t i = threadIdx.x; i < nw; i += blockDim.x)
+          ((uint64_t*)o)[i] = sa[i];
+      } else {
+        for (int64_t i = threadIdx.x; i < nw; i += blockDim.x)
+          ((uint64_t*)o)[i] = (sa[i] >> sh) | (sa[i + 1] << (64 - sh));
+      }
+      for (int64_t i = nw * 8 + threadIdx.x; i < nbytes; i += blockDim.x)
+        o[i] = s[i];
+    } else {  // esize 4 destination at 4-byte alignment
+      int64_t nw = nbytes / 4;
+      for (int64_t i = threadIdx.x; i < nw; i += blockDim.x) {
+        int64_t bit = (int64_t)i * 32 + sh;
+        uint64_t w = sa[bit >> 6];
+        int off = (int)(bit & 63);
+        uint32_t v = (off <= 32)
+                         ? (uint32_t)(w >> off)
+                         : (uint32_t)((w >> off) |
+                                      (sa[(bit >> 6) + 1] << (64 - off)));
+        ((uint32_t*)o)[i] = v;
+      }
+      for (int64_t i = nw * 4 + threadIdx.x; i < nbytes; i += blockDim.x)
+        o[i] = s[i];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// PLAIN BYTE_ARRAY: pass 1 walks the 4-byte length prefixes (sequential
+// per page, lane 0) emitting per-value lengths and absolute source byte
+// offsets in scratch; pass 2 copies string bytes to packed destinations.
+// ---------------------------------------------------------------------
+
+__global__ void pq_byte_array_lengths_kernel(
+    const uint8_t* __restrict__ scratch, const PageMeta* __restrict__ pages,
+    int n_pages, const int32_t* __restrict__ val_data_off,
+    const int64_t* __restrict__ dense_off, const int32_t* __restrict__ n_valid,
+    int32_t* __restrict__ lengths, int64_t* __restrict__ src_abs) {
+  int64_t wave = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  for (int64_t p = wave; p < n_pages; p += n_waves) {
+    if (lane != 0) continue;  // sequential walk; parallelism is across pages
+    PageMeta pm = pages[p];
+    int64_t voff = pm.dst_off + (val_data_off ? val_data_off[p] : 0);
+    const uint8_t* s = scratch + voff;
+    int64_t nval = n_valid ? (int64_t)n_valid[p] : (int64_t)pm.nv;
+    int32_t* L = lengths + dense_off[p];
+    int64_t* A = src_abs + dense_off[p];
+    int64_t pos = 0;
+    for (int64_t i = 0; i < nval; ++i) {
+      int32_t ln;
+      __builtin_memcpy(&ln, s + pos, 4);
+      L[i] = ln;
+      A[i] = voff + pos + 4;
+      pos += 4 + ln;
+    }
+  }
+}
+
+__global__ void pq_copy_strings_kernel(const uint8_t* __restrict__ scratch,
+                                       const int64_t* __restrict__ src_abs,
+                                       const int64_t* __restrict__ dst_off,
+                                       const int32_t* __restrict__ lengths,
+                                       int64_t n, uint8_t* __restrict__ out) {
+  GRID_STRIDE_LOOP(i, n) {
+    const uint8_t* s = scratch + src_abs[i];
+    uint8_t* o = out + dst_off[i];
+    int32_t ln = lengths[i];
+    for (int32_t k = 0; k < ln; ++k) o[k] = s[k];
+  }
+}
+
+// ---------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------
+
+static int pq_grid_waves(int64_t n_pages, int block) {
+  int waves_per_block = block / WAVE;
+  int64_t b = (n_pages + waves_per_block - 1) / waves_per_block;
+  if (b > 4096) b = 4096;
+  if (b < 1) b = 1;
+  return (int)b;
+}
+
+void pq_decompress(torch::Tensor src, torch::Tensor pages_blob,
+                   int64_t n_pages, torch::Tensor scratch) {
+  if (!n_pages) return;
+  int block = 256;
+  hipLaunchKernelGGL(pq_decompress_kernel,
+                     dim3(pq_grid_waves(n_pages, block)), dim3(block), 0,
+                     pq_stream(), (const uint8_t*)src.data_ptr(),
+                     (const PageMeta*)pages_blob.data_ptr(), (int)n_pages,
+                     (uint8_t*)scratch.data_ptr());
+  CHECK_HIP_PQ(hipGetLastError());
+}
+
+std::vector<torch::Tensor> pq_def_levels(torch::Tensor scratch,
+                                         torch::Tensor pages_blob,
+                                         int64_t n_pages, int64_t bitwidth,
+                                         int64_t max_def, int64_t total_nv) {
+  auto dev = scratch.device();
+  auto mask = torch::empty({total_nv}, torch::dtype(torch::kUInt8).device(dev));
+  auto n_valid = torch::empty({n_pages},
+                              torch::dtype(torch::kInt32).device(dev));
+  auto vdo = torch::empty({n_pages}, torch::dtype(torch::kInt32).device(dev));
+  if (n_pages) {
+    int block = 256;
+    hipLaunchKernelGGL(pq_def_levels_kernel,
+                       dim3(pq_grid_waves(n_pages, block)), dim3(block), 0,
+                       pq_stream(), (const uint8_t*)scratch.data_ptr(),
+                       (const PageMeta*)pages_blob.data_ptr(), (int)n_pages,
+                       (int)bitwidth, (int)max_def,
+                       (uint8_t*)mask.data_ptr(),
+                       (int32_t*)n_valid.data_ptr(),
+                       (int32_t*)vdo.data_ptr());
+    CHECK_HIP_PQ(hipGetLastError());
+  }
+  return {mask, n_valid, vdo};
+}
+
+torch::Tensor pq_expand_codes(torch::Tensor scratch, torch::Tensor pages_blob,
+                              int64_t n_pages,
+                              c10::optional<torch::Tensor> val_data_off,
+                              torch::Tensor dense_off,
+                              c10::optional<torch::Tensor> n_valid,
+                              int64_t dense_total) {
+  auto dev = scratch.device();
+  auto out = torch::empty({dense_total},
+                          torch::dtype(torch::kInt32).device(dev));
+  if (n_pages) {
+    int block = 256;
+    hipLaunchKernelGGL(
+        pq_expand_codes_kernel, dim3(pq_grid_waves(n_pages, block)),
+        dim3(block), 0, pq_stream(), (const uint8_t*)scratch.data_ptr(),
+        (const PageMeta*)pages_blob.data_ptr(), (int)n_pages,
+        val_data_off.has_value() ? (const int32_t*)val_data_off->data_ptr()
+                                 : nullptr,
+        (const int64_t*)dense_off.data_ptr(),
+        n_valid.has_value() ? (const int32_t*)n_valid->data_ptr() : nullptr,
+        (int32_t*)out.data_ptr());
+    CHECK_HIP_PQ(hipGetLastError());
+  }
+  return out;
+}
+
+torch::Tensor pq_copy_fixed(torch::Tensor scratch, torch::Tensor pages_blob,
+                            int64_t n_pages,
+                            c10::optional<torch::Tensor> val_data_off,
+                            torch::Tensor dense_off,
+                            c10::optional<torch::Tensor> n_valid,
+                            int64_t esize, int64_t dense_total) {
+  auto dev = scratch.device();
+  auto out = torch::empty({dense_total * esize},
+                          torch::dtype(torch::kUInt8).device(dev));
+  if (n_pages) {
+    int block = 256;
+    int grid = (int)std::min<int64_t>(n_pages, 4096);
+    hipLaunchKernelGGL(
+        pq_copy_fixed_kernel, dim3(grid), dim3(block), 0, pq_stream(),
+        (const uint8_t*)scratch.data_ptr(),
+        (const PageMeta*)pages_blob.data_ptr(), (int)n_pages,
+        val_data_off.has_value() ? (const int32_t*)val_data_off->data_ptr()
+                                 : nullptr,
+        (const int64_t*)dense_off.data_ptr(),
+        n_valid.has_value() ? (const int32_t*)n_valid->data_ptr() : nullptr,
+        (int)esize, (uint8_t*)out.data_ptr());
+    CHECK_HIP_PQ(hipGetLastError());
+  }
+  return out;
+}
+
+std::vector<torch::Tensor> pq_byte_array_lengths(
+    torch::Tensor scratch, torch::Tensor pages_blob, int64_t n_pages,
+    c10::optional<torch::Tensor> val_data_off, torch::Tensor dense_off,
+    c10::optional<torch::Tensor> n_valid, int64_t dense_total) {
+  auto dev = scratch.device();
+  auto lengths = torch::zeros({dense_total},
+                              torch::dtype(torch::kInt32).device(dev));
+  auto src_abs = torch::zeros({dense_total},
+                              torch::dtype(torch::kInt64).device(dev));
+  if (n_pages) {
+    int block = 256;
+    hipLaunchKernelGGL(
+        pq_byte_array_lengths_kernel, dim3(pq_grid_waves(n_pages, block)),
+        dim3(block), 0, pq_stream(), (const uint8_t*)scratch.data_ptr(),
+        (const PageMeta*)pages_blob.data_ptr(), (int)n_pages,
+        val_data_off.has_value() ? (const int32_t*)val_data_off->data_ptr()
+                                 : nullptr,
+        (const int64_t*)dense_off.data_ptr(),
+        n_valid.has_value() ? (const int32_t*)n_valid->data_ptr() : nullptr,
+        (int32_t*)lengths.data_ptr(), (int64_t*)src_abs.data_ptr());
+    CHECK_HIP_PQ(hipGetLastError());
+  }
+  return {lengths, src_abs};
+}
+
+torch::Tensor pq_copy_strings(torch::Tensor scratch, torch::Tensor src_abs,
+                              torch::Tensor dst_off, torch::Tensor lengths,
+                              int64_t n, int64_t total_bytes) {
+  auto dev = scratch.device();
+  auto out = torch::empty({std::max<int64_t>(total_bytes, 1)},
+                          torch::dtype(torch::kUInt8).device(dev));
+  if (n) {
+    int block = 256;
+    hipLaunchKernelGGL(pq_copy_strings_kernel, dim3(grid_for(n, block)),
+                       dim3(block), 0, pq_stream(),
+                       (const uint8_t*)scratch.data_ptr(),
+                       (const int64_t*)src_abs.data_ptr(),
+                       (const int64_t*)dst_off.data_ptr(),
+                       (const int32_t*)lengths.data_ptr(), n,
+                       (uint8_t*)out.data_ptr());
+    CHECK_HIP_PQ(hipGetLastError());
+  }
+  return out;
+}
+
+// ---------------------------------------------------------------------
+// host-side page header parse (Thrift compact protocol): the Python
+// version cost 10-30us per page; this walks a whole chunk in C++.
+// Returns int64 tensor [n_pages, 7]:
+//   0 ptype, 1 uncompressed_size, 2 compressed_size, 3 body_off,
+//   4 nv, 5 encoding, 6 def_level_encoding
+// ---------------------------------------------------------------------
+
+namespace {
+
+struct TR {
+  const uint8_t* b;
+  size_t pos, end;
+  bool ok = true;
+
+  uint64_t varint() {
+    uint64_t out = 0;
+    int shift = 0;
+    while (pos < end) {
+      uint8_t v = b[pos++];
+      out |= (uint64_t)(v & 0x7F) << shift;
+      if (!(v & 0x80)) return out;
+      shift += 7;
+    }
+    ok = false;
+    return 0;
+  }
+  int64_t zigzag() {
+    uint64_t v = varint();
+    return (int64_t)(v >> 1) ^ -(int64_t)(v & 1);
+  }
+  void skip(int ftype);
+  void skip_struct() {
+    int16_t fid = 0;
+    while (ok && pos < end) {
+      uint8_t t = b[pos++];
+      if (t == 0) return;
+      int delta = (t >> 4) & 0x0F;
+      int ft = t & 0x0F;
+      fid = delta ? fid + delta : (int16_t)zigzag();
+      skip(ft);
+    }
+  }
+};
+
+void TR::skip(int ftype) {
+  switch (ftype) {
+    case 1: case 2: return;               // bool encoded in type
+    case 3: pos += 1; return;             // byte
+    case 4: case 5: case 6: zigzag(); return;
+    case 7: pos += 8; return;             // double
+    case 8: { uint64_t n = varint(); pos += n; return; }  // binary
+    case 9: {                              // list
+      if (pos >= end) { ok = false; return; }
+      uint8_t h = b[pos++];
+      uint64_t sz = (h >> 4) & 0x0F;
+      int et = h & 0x0F;
+      if (sz == 15) sz = varint();
+      for (uint64_t i = 0; i < sz && ok; ++i) skip(et);
+      return;
+    }
+    case 12: skip_struct(); return;
+    default: ok = false; return;
+  }
+}
+
+}  // namespace
+
+torch::Tensor pq_parse_headers(torch::Tensor chunk_buf) {
+  const uint8_t* buf = (const uint8_t*)chunk_buf.data_ptr();
+  size_t n = (size_t)chunk_buf.numel();
+  std::vector<int64_t> rows;  // 7 per page
+  size_t pos = 0;
+  while (pos < n) {
+    TR t{buf, pos, n};
+    int64_t ptype = -1, usize = 0, csize = 0, nv = 0, enc = -1, defenc = -1;
+    int16_t fid = 0;
+    // walk PageHeader struct
+    bool done = false;
+    while (t.ok && !done && t.pos < n) {
+      uint8_t tb = buf[t.pos++];
+      if (tb == 0) { done = true; break; }
+      int delta = (tb >> 4) & 0x0F;
+      int ft = tb & 0x0F;
+      fid = delta ? fid + delta : (int16_t)t.zigzag();
+      switch (fid) {
+        case 1: ptype = t.zigzag(); break;
+        case 2: usize = t.zigzag(); break;
+        case 3: csize = t.zigzag(); break;
+        case 5: case 7: {  // data_page_header / dictionary_page_header
+          if (ft != 12) { t.skip(ft); break; }
+          int16_t f2 = 0;
+          bool d2 = false;
+          while (t.ok && !d2 && t.pos < n) {
+            uint8_t b2 = buf[t.pos++];
+            if (b2 == 0) { d2 = true; break; }
+            int dl2 = (b2 >> 4) & 0x0F;
+            int ft2 = b2 & 0x0F;
+            f2 = dl2 ? f2 + dl2 : (int16_t)t.zigzag();
+            if (f2 == 1 && (ft2 == 4 || ft2 == 5 || ft2 == 6)) {
+              nv = t.zigzag();
+            } else if (f2 == 2 && (ft2 == 4 || ft2 == 5 || ft2 == 6)) {
+              enc = t.zigzag();
+            } else if (fid == 5 && f2 == 3 &&
+                       (ft2 == 4 || ft2 == 5 || ft2 == 6)) {
+              defenc = t.zigzag();
+            } else {
+              t.skip(ft2);
+            }
+          }
+          break;
+        }
+        default: t.skip(ft); break;
+      }
+    }
+    if (!t.ok || !done || csize <= 0) break;
+    int64_t body = (int64_t)t.pos;
+    rows.insert(rows.end(), {ptype, usize, csize, body, nv, enc, defenc});
+    pos = (size_t)(body + csize);
+  }
+  auto out = torch::empty({(int64_t)(rows.size() / 7), 7},
+                          torch::dtype(torch::kInt64));
+  std::memcpy(out.data_ptr(), rows.data(), rows.size() * sizeof(int64_t));
+  return out;
+}

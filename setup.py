@@ -16,7 +16,7 @@ setup(
     ext_modules=[
         CUDAExtension(
             name="bodo_amd_kernels",
-            sources=["csrc/kernels.hip", "csrc/gemm.hip"],
+            sources=["csrc/kernels.hip", "csrc/gemm.hip", "csrc/parquet.hip"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],

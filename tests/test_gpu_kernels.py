@@ -413,3 +413,79 @@ def test_gpu_distinct_keep_variants():
         got = b.drop_duplicates(subset=["a"], keep=keep).to_pandas()
         exp = df.drop_duplicates(subset=["a"], keep=keep)
         assert sorted(got.b.tolist()) == sorted(exp.b.tolist()), keep
+
+
+@pytest.mark.gpu
+def test_parquet_batched_decode_layouts(tmp_path):
+    """The page-parallel batched device decode must handle every common
+    layout without falling back: snappy/uncompressed x plain/dict x
+    nulls/no-nulls x fixed/strings, multi-page chunks."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from bodo_amd.io import parquet_gpu as g
+
+    rng = np.random.default_rng(21)
+    n = 400_000
+    base = {
+        "i64": rng.integers(-10**9, 10**9, n).astype("int64"),
+        "i32": rng.integers(0, 100, n).astype("int32"),
+        "f64": rng.random(n) * 1e6,
+        "f32": rng.random(n).astype("float32"),
+        "s_plain": np.array(
+            ["s" + str(v) for v in rng.integers(0, 10**9, n)], dtype=object),
+        "s_dict": rng.choice(["alpha", "beta", "gamma", "delta"], n),
+        "ts": (np.datetime64("2023-01-01")
+               + rng.integers(0, 10**9, n).astype("timedelta64[s]")),
+    }
+    for comp in ("snappy", "none"):
+        for with_nulls in (False, True):
+            df = pd.DataFrame({k: v.copy() for k, v in base.items()})
+            if with_nulls:
+                df.loc[rng.random(n) < 0.08, "f64"] = np.nan
+                df.loc[rng.random(n) < 0.08, "s_plain"] = None
+                df.loc[rng.random(n) < 0.08, "s_dict"] = None
+            fp = str(tmp_path / f"t_{comp}_{with_nulls}.parquet")
+            pq.write_table(
+                pa.Table.from_pandas(df), fp, compression=comp,
+                use_dictionary=["s_dict", "i32"], data_page_size=128 * 1024)
+            before = g.STATS["slow"]
+            t = g._read_row_group_gpu(fp, 0, None, _ctx("cuda"))
+            assert t is not None, (comp, with_nulls)
+            assert g.STATS["slow"] == before, \
+                f"fallback used for {comp} nulls={with_nulls}"
+            out = t.to_pandas()
+            for c in out.columns:
+                if out[c].dtype.name == "category":
+                    out[c] = out[c].astype(object)
+            exp = df.copy()
+            pd.testing.assert_frame_equal(out, exp, check_dtype=False)
+
+
+@pytest.mark.gpu
+def test_parquet_batched_decode_multi_row_group(tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    import bodo_amd.config as cfg
+    from bodo_amd.io import parquet_gpu as g
+
+    cfg.DEVICE = "cuda"
+    import bodo_amd.pandas as bpd
+
+    rng = np.random.default_rng(22)
+    n = 1_000_000
+    df = pd.DataFrame({"k": rng.integers(0, 500, n),
+                       "v": rng.random(n),
+                       "s": rng.choice(["a", "bb", "ccc"], n)})
+    fp = str(tmp_path / "big.parquet")
+    pq.write_table(pa.Table.from_pandas(df), fp, compression="snappy",
+                   row_group_size=200_000)
+    b = bpd.read_parquet(fp)
+    got = b.groupby("k", as_index=False).agg(
+        s=bpd.NamedAgg("v", "sum"), c=bpd.NamedAgg("s", "count"))
+    got = got.to_pandas().sort_values("k").reset_index(drop=True)
+    exp = df.groupby("k", as_index=False).agg(
+        s=("v", "sum"), c=("s", "count")).sort_values("k").reset_index(
+        drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)

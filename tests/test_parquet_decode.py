@@ -101,3 +101,48 @@ def test_parquet_snappy_dict_nulls_mixed(tmp_path, monkeypatch):
         if out[c].dtype.name == "category":
             out[c] = out[c].astype(str)
     pd.testing.assert_frame_equal(out, df, check_dtype=False)
+
+
+def test_pq_parse_headers_matches_python(tmp_path):
+    """C++ host page-header parser vs the Python ThriftCompact walk."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    import bodo_amd_kernels as K
+    from bodo_amd.io import parquet_gpu as g
+
+    rng = np.random.default_rng(3)
+    n = 200_000
+    df = pd.DataFrame({
+        "a": rng.integers(0, 50, n).astype("int64"),
+        "b": rng.random(n),
+        "s": rng.choice(["xx", "yyy", "zzzz"], n),
+    })
+    df.loc[rng.random(n) < 0.05, "b"] = np.nan
+    fp = str(tmp_path / "t.parquet")
+    pq.write_table(pa.Table.from_pandas(df), fp, compression="snappy",
+                   data_page_size=64 * 1024)
+    pf = pq.ParquetFile(fp)
+    rgm = pf.metadata.row_group(0)
+    with open(fp, "rb") as f:
+        for ci in range(rgm.num_columns):
+            cm = rgm.column(ci)
+            r = g._ChunkReader(f, cm, cm.physical_type)
+            got = K.pq_parse_headers(
+                torch.from_numpy(
+                    np.frombuffer(r.buf, dtype=np.uint8).copy())).numpy()
+            # python reference walk
+            pos, rows = 0, []
+            while pos < len(r.buf):
+                t = g.ThriftCompact(r.buf, pos)
+                hdr = t.read_struct()
+                body = t.pos
+                ptype = hdr.get(1)
+                dph = hdr.get(5 if ptype == g.PAGE_DATA else 7, {})
+                rows.append((ptype, hdr.get(2), hdr.get(3), body,
+                             dph.get(1, 0), dph.get(2, -1),
+                             dph.get(3, -1) if ptype == g.PAGE_DATA else -1))
+                pos = body + hdr.get(3)
+            exp = np.array(rows, dtype=np.int64)
+            assert got.shape == exp.shape, (got.shape, exp.shape)
+            assert (got == exp).all(), cm.path_in_schema
