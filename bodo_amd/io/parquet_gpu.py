@@ -181,6 +181,8 @@ _UPLOADER = _PinnedUploader()
 
 # decode-path counters (tests assert the batched path actually ran)
 STATS = {"batched": 0, "slow": 0}
+# CPU tests flip this with simulated kernels monkeypatched in
+FORCE_BATCHED = False
 
 _PAGEMETA_DT = np.dtype([
     ("src_off", np.int64), ("dst_off", np.int64), ("val_off", np.int64),
@@ -229,8 +231,8 @@ class _ChunkReader:
         return col
 
     def _decode_batched(self, device, field: pa.Field) -> Optional[Column]:
-        if torch.device(device).type != "cuda":
-            return None  # batched kernels are device-only
+        if torch.device(device).type != "cuda" and not FORCE_BATCHED:
+            return None  # batched kernels are device-only (tests simulate)
         comp = self.meta.compression
         if comp not in ("SNAPPY", "UNCOMPRESSED"):
             return None
@@ -286,6 +288,7 @@ class _ChunkReader:
         metas["src_off"] = d_body
         metas["src_len"] = d_csize
         metas["dst_len"] = d_usize
+        metas["nv"] = d_nv
         dst = np.cumsum(np.concatenate([[0], (d_usize + 7) & ~7]))[:-1]
         metas["dst_off"] = dst
         val_off = np.cumsum(np.concatenate([[0], d_nv]))[:-1]
