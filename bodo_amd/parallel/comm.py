@@ -274,7 +274,7 @@ def _pack_table(tbl: Table, dev: torch.device):
         return off, u8.numel()
 
     for c in tbl.columns:
-        m = {"kind": int(c.dtype.kind), "n": len(c)}
+        m = {"dtype": c.dtype, "n": len(c)}
         if c.data is not None:
             m["data"] = add(c.data.to(dev))
             m["data_dtype"] = str(c.data.dtype).replace("torch.", "")
@@ -295,12 +295,9 @@ def _pack_table(tbl: Table, dev: torch.device):
 def _unpack_table(buf: torch.Tensor, meta: dict, device) -> Table:
     import pyarrow as pa
 
-    from ..core.types import ALL_TYPES
-
-    kinds = {int(t.kind): t for t in ALL_TYPES}
     cols = []
     for m in meta["cols"]:
-        kind = kinds[m["kind"]]
+        kind = m["dtype"]
         data = mask = offsets = None
         dictionary = None
         if "data" in m:

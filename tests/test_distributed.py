@@ -524,3 +524,62 @@ def test_dist_limit_offset():
     exp = df.sort_values("b").reset_index(drop=True).iloc[4:13][
         ["b"]].reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_pack_collectives(bpd, rank, payload):
+    """Exercise the packed-buffer allgather_table / gather_table directly
+    (strings + dict + nulls + empty-on-one-rank)."""
+    import pyarrow as pa
+    import torch
+
+    from bodo_amd.core.column import Column
+    from bodo_amd.core.table import Table
+    from bodo_amd.core import types as bt
+    from bodo_amd.parallel import comm
+
+    df = payload["df"]
+    half = len(df) // 2
+    shard = df.iloc[rank * half:(rank + 1) * half].reset_index(drop=True)
+    if payload.get("empty_rank1") and rank == 1:
+        shard = shard.iloc[:0]
+    t = Table.from_pandas(shard)
+    rep = comm.allgather_table(t)
+    g = comm.gather_table(t, root=0)
+    out = rep.to_pandas()
+    out["gathered_rows"] = len(g) if g is not None else -1
+    return out
+
+
+def test_dist_packed_table_collectives():
+    rng = np.random.default_rng(5)
+    n = 300
+    df = pd.DataFrame({
+        "a": rng.integers(0, 50, n),
+        "f": rng.random(n),
+        "s": np.array(["v" + str(i % 37) for i in range(n)], dtype=object),
+        "c": pd.Categorical(rng.choice(["x", "y", "z"], n)),
+    })
+    df.loc[rng.random(n) < 0.1, "f"] = np.nan
+    got = run_dist(_q_pack_collectives, {"df": df})
+    exp = df.iloc[:300 // 2 * 2].reset_index(drop=True)
+    assert got["gathered_rows"].iloc[0] == len(exp)
+    got2 = got.drop(columns=["gathered_rows"]).reset_index(drop=True)
+    exp2 = exp.copy()
+    for c in exp2.columns:
+        if exp2[c].dtype == object or str(exp2[c].dtype) == "category":
+            exp2[c] = exp2[c].astype(str)
+            got2[c] = got2[c].astype(str)
+    pd.testing.assert_frame_equal(got2, exp2, check_dtype=False)
+
+
+def test_dist_packed_collectives_empty_shard():
+    rng = np.random.default_rng(6)
+    n = 100
+    df = pd.DataFrame({"a": rng.integers(0, 5, n),
+                       "s": np.array([f"w{i}" for i in range(n)],
+                                     dtype=object)})
+    got = run_dist(_q_pack_collectives, {"df": df, "empty_rank1": True})
+    exp = df.iloc[:50].reset_index(drop=True)
+    assert got["gathered_rows"].iloc[0] == len(exp)
+    got2 = got.drop(columns=["gathered_rows"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got2, exp, check_dtype=False)
