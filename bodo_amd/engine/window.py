@@ -13,7 +13,7 @@ Two strategies:
 
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 import pandas as pd
@@ -101,9 +101,264 @@ def _exec_ordered(child: Table, keys, order_by, ascending, specs, ctx) -> Table:
     return out
 
 
+_ROLLING_MINMAX_CAP = 128
+
+
+def _adj_eq(c: Column, idx: torch.Tensor) -> Optional[torch.Tensor]:
+    """Bool tensor of length n-1: sorted-adjacent rows equal under pandas
+    groupby(dropna=False) semantics (null == null, NaN == NaN).  None when
+    the column kind has no device comparison (plain strings)."""
+    from ..core.types import TypeKind as TK
+
+    k = c.dtype.kind
+    if k == TK.STRING:
+        return None
+    s = c.data[idx]
+    if c.dtype.is_float:
+        eq = (s[1:] == s[:-1]) | (torch.isnan(s[1:]) & torch.isnan(s[:-1]))
+    else:
+        eq = s[1:] == s[:-1]
+    if c.mask is not None:
+        m = c.mask[idx]
+        eq = (eq & m[1:] & m[:-1]) | (~m[1:] & ~m[:-1])
+    return eq
+
+
+def _ordered_local_device(tbl: Table, keys, order_by, ascending,
+                          specs) -> Optional[Table]:
+    """Segmented device window calculator: one stable sort by
+    (partition keys, order keys), segment ids from adjacent key equality,
+    every function as segmented prefix ops — no host pandas (reference role:
+    bodo/libs/window/_window_calculator.cpp:2106, redesigned as whole-shard
+    tensor passes instead of chunked streaming)."""
+    from ..core.types import TypeKind as TK
+
+    n = len(tbl)
+    dev = tbl.device
+    if n == 0:
+        return None
+    key_cols = [tbl.column(k) for k in keys]
+    order_cols = [tbl.column(c) for c in order_by]
+    asc = list(ascending) or [True] * len(order_by)
+    if any(c.dtype.kind == TK.STRING for c in key_cols + order_cols):
+        return None
+    for _, in_name, func, arg in specs:
+        if func.startswith("rolling_") and func[8:] in ("min", "max") \
+                and int(arg or 1) > _ROLLING_MINMAX_CAP:
+            return None
+        if in_name and tbl.has_column(in_name):
+            ic = tbl.column(in_name)
+            if ic.dtype.kind in (TK.STRING, TK.DICT) and func in (
+                    "shift", "first_value", "last_value"):
+                return None  # string payloads: host path
+    idx = ops.sort_indices(key_cols + order_cols,
+                           [True] * len(keys) + asc)
+    seg_new = torch.zeros(n, dtype=torch.bool, device=dev)
+    seg_new[0] = True
+    for c in key_cols:
+        eq = _adj_eq(c, idx)
+        if eq is None:
+            return None
+        seg_new[1:] |= ~eq
+    segid = torch.cumsum(seg_new.long(), 0) - 1
+    start_pos = torch.nonzero(seg_new).squeeze(1)           # [n_segs]
+    n_segs = int(start_pos.numel())
+    pos = torch.arange(n, dtype=torch.int64, device=dev)
+    seg_start = start_pos[segid]                            # [n] start of my seg
+    row_in_seg = pos - seg_start
+    end_pos = torch.cat([start_pos[1:] - 1,
+                         torch.tensor([n - 1], device=dev)])
+    seg_size = end_pos[segid] - seg_start + 1
+
+    def sorted_vals(name, fill0=False):
+        c = tbl.column(name)
+        v = c.data[idx]
+        invalid = None
+        if c.mask is not None:
+            invalid = ~c.mask[idx]
+        if c.dtype.is_float:
+            nanv = torch.isnan(v)
+            invalid = nanv if invalid is None else (invalid | nanv)
+        if fill0 and invalid is not None:
+            v = torch.where(invalid, torch.zeros((), dtype=v.dtype,
+                                                 device=dev), v)
+        return v, invalid
+
+    def seg_cumsum(vf):
+        cs = torch.cumsum(vf.double() if vf.dtype != torch.float64 else vf, 0)
+        excl = cs - (vf.double() if vf.dtype != torch.float64 else vf)
+        return cs - excl[seg_start]
+
+    out_sorted = {}
+    for out_name, in_name, func, arg in specs:
+        if func == "row_number":
+            res, res_inv = row_in_seg + 1, None
+        elif func == "cumcount":
+            res, res_inv = row_in_seg, None
+        elif func in ("rank", "dense_rank"):
+            method = "dense" if func == "dense_rank" else (arg or "min")
+            if method not in ("dense", "min"):
+                return None
+            if order_cols:
+                r_idx, r_seg_new, r_seg_start = idx, seg_new, seg_start
+                ocol = order_cols[0]
+            else:
+                # value rank (groupby().rank()): needs its own sort by the
+                # ranked column within each partition
+                ocol = tbl.column(in_name)
+                if ocol.dtype.kind == TK.STRING:
+                    return None
+                rasc = ascending[0] if ascending else True
+                r_idx = ops.sort_indices(key_cols + [ocol],
+                                         [True] * len(keys) + [rasc])
+                r_seg_new = torch.zeros(n, dtype=torch.bool, device=dev)
+                r_seg_new[0] = True
+                for c in key_cols:
+                    eq = _adj_eq(c, r_idx)
+                    if eq is None:
+                        return None
+                    r_seg_new[1:] |= ~eq
+                r_segid = torch.cumsum(r_seg_new.long(), 0) - 1
+                r_seg_start = torch.nonzero(r_seg_new).squeeze(1)[r_segid]
+            eq = _adj_eq(ocol, r_idx)
+            if eq is None:
+                return None
+            run_new = r_seg_new.clone()
+            run_new[1:] |= ~eq
+            if func == "dense_rank":
+                rg = torch.cumsum(run_new.long(), 0)
+                res = rg - rg[r_seg_start] + 1
+            else:  # rank(min): row_number of the run start
+                run_start = torch.nonzero(run_new).squeeze(1)
+                runid = torch.cumsum(run_new.long(), 0) - 1
+                res = run_start[runid] - r_seg_start + 1
+            # pandas rank: NaN order value -> NaN rank
+            ov = ocol.data[r_idx]
+            oinv = None
+            if ocol.dtype.is_float:
+                oinv = torch.isnan(ov)
+            if ocol.mask is not None:
+                om = ~ocol.mask[r_idx]
+                oinv = om if oinv is None else (oinv | om)
+            res_inv = oinv
+            if r_idx is not idx:
+                # map through this spec's own sort back to original order
+                r_inv = torch.empty_like(r_idx)
+                r_inv[r_idx] = pos
+                res = res[r_inv][idx]
+                if res_inv is not None:
+                    res_inv = res_inv[r_inv][idx]
+        elif func == "cumsum":
+            v, inv = sorted_vals(in_name, fill0=True)
+            res = seg_cumsum(v)
+            if tbl.column(in_name).dtype.is_integer and inv is None:
+                res = res.long()
+            res_inv = inv
+        elif func == "shift":
+            k = int(arg if arg is not None else 1)
+            v, inv = sorted_vals(in_name)
+            res = torch.full_like(v, 0)
+            res_inv = torch.ones(n, dtype=torch.bool, device=dev)
+            if k >= 0:
+                if k < n:
+                    res[k:] = v[:n - k]
+                    ok = segid[k:] == segid[:n - k]
+                    if inv is not None:
+                        ok &= ~inv[:n - k]
+                    res_inv[k:] = ~ok
+            else:
+                kk = -k
+                if kk < n:
+                    res[:n - kk] = v[kk:]
+                    ok = segid[:n - kk] == segid[kk:]
+                    if inv is not None:
+                        ok &= ~inv[kk:]
+                    res_inv[:n - kk] = ~ok
+        elif func in ("first_value", "last_value"):
+            v, inv = sorted_vals(in_name)
+            at = seg_start if func == "first_value" else end_pos[segid]
+            res = v[at]
+            res_inv = inv[at] if inv is not None else None
+        elif func == "ntile":
+            k = int(arg or 1)
+            res = (row_in_seg * k) // seg_size + 1
+            res_inv = None
+        elif func.startswith("rolling_"):
+            base = func[len("rolling_"):]
+            w = int(arg)
+            v, inv = sorted_vals(in_name, fill0=True)
+            valid = torch.ones(n, dtype=torch.float64, device=dev)
+            if inv is not None:
+                valid = (~inv).double()
+            lo = torch.maximum(pos - (w - 1), seg_start)
+            if base in ("sum", "mean", "count"):
+                vf = v.double()
+                cs = torch.cumsum(vf, 0)
+                excl = cs - vf
+                rsum = cs - excl[lo]
+                cv = torch.cumsum(valid, 0)
+                excl_v = cv - valid
+                rcount = cv - excl_v[lo]
+                if base == "sum":
+                    res = rsum
+                elif base == "count":
+                    res = rcount
+                else:
+                    res = rsum / rcount
+                res_inv = rcount == 0 if base != "count" else None
+            else:  # min / max over small windows: log-free shift loop
+                cur = torch.where(valid.bool(), v.double(),
+                                  torch.full((), float("inf") if base == "min"
+                                             else float("-inf"), device=dev))
+                acc = cur.clone()
+                for t in range(1, w):
+                    sh = torch.empty_like(cur)
+                    sh[t:] = cur[:n - t]
+                    sh[:t] = float("inf") if base == "min" else float("-inf")
+                    inwin = pos - t >= lo
+                    sh = torch.where(inwin, sh,
+                                     torch.full((), float("inf")
+                                                if base == "min"
+                                                else float("-inf"),
+                                                device=dev))
+                    acc = torch.minimum(acc, sh) if base == "min" \
+                        else torch.maximum(acc, sh)
+                res = acc
+                res_inv = torch.isinf(res)
+        else:
+            return None
+        out_sorted[out_name] = (res, res_inv)
+
+    out = tbl
+    inv_perm = torch.empty_like(idx)
+    inv_perm[idx] = pos
+    for name, (res, res_inv) in out_sorted.items():
+        orig = res[inv_perm]
+        mask = None
+        if res_inv is not None:
+            mask = ~res_inv[inv_perm]
+            if orig.dtype.is_floating_point:
+                orig = torch.where(mask, orig,
+                                   torch.full((), float("nan"), device=dev,
+                                              dtype=orig.dtype))
+                mask = None
+        kind = bt.from_numpy_dtype(np.dtype(str(orig.dtype).replace(
+            "torch.", "")))
+        col = Column(kind, orig, mask)
+        out = out.with_column(name, col)
+    return out
+
+
 def _ordered_local(tbl: Table, keys, order_by, ascending, specs) -> Table:
     """Compute ordered window funcs on co-located partitions; preserves the
-    input row order of `tbl`."""
+    input row order of `tbl`.  Device path first; host pandas only for
+    layouts the tensor calculator does not cover."""
+    try:
+        dev_out = _ordered_local_device(tbl, keys, order_by, ascending, specs)
+    except Exception:
+        dev_out = None
+    if dev_out is not None:
+        return dev_out
     need = list(dict.fromkeys(
         keys + order_by + [s[1] for s in specs if s[1] and tbl.has_column(s[1])]))
     pdf = tbl.select(need).to_pandas()
