@@ -127,9 +127,11 @@ def main():
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--rows", type=int, default=TOTAL_ROWS,
                    help="total rows (dev override; judged runs use default)")
-    p.add_argument("--parquet", action="store_true",
-                   help="read the trips table from parquet each step with "
-                        "on-GPU decode (BASELINE config 2)")
+    p.add_argument("--parquet", action="store_true", default=True,
+                   help="read the trips table from snappy parquet each step "
+                        "with on-GPU decode (BASELINE config 2; the default)")
+    p.add_argument("--no-parquet", dest="parquet", action="store_false",
+                   help="skip parquet IO: query in-HBM synthetic shards")
     args = p.parse_args()
 
     import bodo_amd  # noqa: F401  (inits process group under torchrun)
@@ -159,13 +161,24 @@ def main():
         # decoder before the query
         import pyarrow.parquet as pq
 
+        from bodo_amd import ops as _ops
+
         os.makedirs("/tmp/bodo_bench_pq", exist_ok=True)
         pq_path = f"/tmp/bodo_bench_pq/trips_rank{rank}.parquet"
-        at = trips.to_device("cpu").to_arrow()
-        pq.write_table(at, pq_path, compression="SNAPPY",
-                       use_dictionary=["hvfhs_license_num"],
-                       row_group_size=1 << 23)
-        del at
+        # chunked write keeps the host staging bounded at ~1.3 GB even for
+        # the full 1B-row shard
+        CHUNK = 1 << 25
+        writer = None
+        for s in range(0, len(trips), CHUNK):
+            at = _ops.slice_table(trips, s, min(s + CHUNK, len(trips))) \
+                .to_device("cpu").to_arrow()
+            if writer is None:
+                writer = pq.ParquetWriter(
+                    pq_path, at.schema, compression="SNAPPY",
+                    use_dictionary=["hvfhs_license_num"])
+            writer.write_table(at, row_group_size=1 << 23)
+            del at
+        writer.close()
 
     def read_trips():
         if pq_path is None:
@@ -216,8 +229,8 @@ def main():
             "scaling": "strong",
             "vs_baseline": None,
             "dtype": "fp64",
-            "data": ("synthetic parquet (uncompressed, on-GPU decode in "
-                     "timed region)" if args.parquet else
+            "data": ("synthetic snappy parquet (on-GPU decode in timed "
+                     "region)" if args.parquet else
                      "synthetic (in-HBM, generation untimed; no parquet IO)"),
             "config": {
                 "model": "nyc_taxi_q1_monthly_trips_precipitation",
