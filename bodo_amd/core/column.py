@@ -147,7 +147,28 @@ class Column:
         elif pa.types.is_boolean(t):
             dtype = bt.boolean
         elif pa.types.is_decimal(t):
-            # limited: convert through float64 for now
+            if t.precision <= 18:
+                # exact money semantics: store the unscaled integer in an
+                # int64 tensor (values with p<=18 fit the low limb; the high
+                # limb is pure sign extension).  Reference role:
+                # bodo/libs/_decimal_ext.cpp int128 decimals — redesigned as
+                # scaled int64, which is native on CDNA4; p>18 falls back.
+                bufs = arr.buffers()
+                validity, data_buf = bufs[0], bufs[1]
+                pairs = np.frombuffer(data_buf, dtype=np.int64,
+                                      count=2 * (len(arr) + arr.offset))
+                lo = pairs[2 * arr.offset::2][:len(arr)]
+                mask_np = None
+                if validity is not None and arr.null_count:
+                    mask_np = _unpack_validity(validity, arr.offset, len(arr))
+                    lo = lo.copy()
+                    lo[~mask_np] = 0
+                return Column(
+                    bt.decimal128(t.precision, t.scale),
+                    torch.from_numpy(np.ascontiguousarray(lo)).to(device),
+                    None if mask_np is None
+                    else torch.from_numpy(mask_np).to(device),
+                    length=len(arr))
             arr = arr.cast(pa.float64())
             dtype = bt.float64
         else:
@@ -199,6 +220,22 @@ class Column:
             idx_arr = pa.array(idx, type=pa.int32(),
                                mask=None if mask is None else ~mask)
             return pa.DictionaryArray.from_arrays(idx_arr, self.dictionary)
+        if k == TypeKind.DECIMAL128:
+            lo = _np_from_tensor(self.data).astype(np.int64, copy=False)
+            n = len(self)
+            pairs = np.empty(2 * n, dtype=np.int64)
+            pairs[0::2] = lo
+            pairs[1::2] = np.where(lo < 0, -1, 0)  # sign-extended high limb
+            mask = None if self.mask is None else _np_from_tensor(self.mask)
+            vbuf = None
+            null_count = 0
+            if mask is not None:
+                vbuf = pa.py_buffer(
+                    np.packbits(mask, bitorder="little").tobytes())
+                null_count = int((~mask).sum())
+            return pa.Array.from_buffers(
+                pa.decimal128(self.dtype.precision, self.dtype.scale), n,
+                [vbuf, pa.py_buffer(pairs.tobytes())], null_count)
         np_data = _np_from_tensor(self.data)
         mask = None if self.mask is None else _np_from_tensor(self.mask)
         pa_type = {

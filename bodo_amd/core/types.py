@@ -118,6 +118,7 @@ _TORCH_STORAGE = {
     TypeKind.DATE32: torch.int32,
     TypeKind.TIMESTAMP_NS: torch.int64,
     TypeKind.DICT: torch.int32,
+    TypeKind.DECIMAL128: torch.int64,  # scaled int64 (exact for p <= 18)
 }
 
 
@@ -165,6 +166,7 @@ _KIND_TO_NUMPY = {
     TypeKind.DATE32: np.dtype("int32"),
     TypeKind.TIMESTAMP_NS: np.dtype("int64"),
     TypeKind.DICT: np.dtype("int32"),
+    TypeKind.DECIMAL128: np.dtype("int64"),
 }
 
 

@@ -372,7 +372,49 @@ def normalize_const(v, dtype: DType):
 _ARITH_RESULT_FLOAT = {"div", "pow"}
 
 
+def decimal_to_float(a: Column) -> Column:
+    """Exact-int decimal -> float64 value column."""
+    out = a.data.double() / float(10 ** a.dtype.scale)
+    return Column(bt.float64, out, a.mask)
+
+
+def _binary_arith_decimal(op: str, a: Column, b: Column) -> Column:
+    """Exact decimal arithmetic on the scaled-int64 representation
+    (reference role: bodo/libs/_decimal_ext.cpp add/mul).  add/sub align
+    scales; mul adds scales; anything that can leave the p<=18 envelope
+    (div, pow, scale overflow) computes in float64."""
+    DEC = TypeKind.DECIMAL128
+    ak, bk = a.dtype.kind, b.dtype.kind
+    int_ok = (ak == DEC or a.dtype.is_integer) and \
+             (bk == DEC or b.dtype.is_integer)
+    if op in ("add", "sub", "mul") and int_ok:
+        sa = a.dtype.scale if ak == DEC else 0
+        sb = b.dtype.scale if bk == DEC else 0
+        da = a.data.long() if a.data.dtype != torch.int64 else a.data
+        db = b.data.long() if b.data.dtype != torch.int64 else b.data
+        if op == "mul":
+            s = sa + sb
+            if s <= 18:
+                out = da * db
+                mask = combine_masks(a.mask, b.mask)
+                return Column(bt.decimal128(18, s), out, mask)
+        else:
+            s = max(sa, sb)
+            if sa < s:
+                da = da * (10 ** (s - sa))
+            if sb < s:
+                db = db * (10 ** (s - sb))
+            out = da + db if op == "add" else da - db
+            mask = combine_masks(a.mask, b.mask)
+            return Column(bt.decimal128(18, s), out, mask)
+    fa = decimal_to_float(a) if ak == DEC else a
+    fb = decimal_to_float(b) if bk == DEC else b
+    return binary_arith(op, fa, fb)
+
+
 def binary_arith(op: str, a: Column, b: Column) -> Column:
+    if TypeKind.DECIMAL128 in (a.dtype.kind, b.dtype.kind):
+        return _binary_arith_decimal(op, a, b)
     da, db = a.data, b.data
     if a.dtype.is_float or b.dtype.is_float or op in _ARITH_RESULT_FLOAT:
         target = torch.float64 if (
@@ -408,6 +450,27 @@ def binary_arith(op: str, a: Column, b: Column) -> Column:
 
 def binary_arith_scalar(op: str, a: Column, v, vdtype: DType,
                         reflect: bool) -> Column:
+    if a.dtype.kind == TypeKind.DECIMAL128:
+        s = a.dtype.scale
+        exact = None  # (op-ready int, result scale)
+        if not isinstance(v, bool) and isinstance(v, (int, float)):
+            if op in ("add", "sub"):
+                scaled = v * (10 ** s)
+                if float(scaled) == round(scaled):
+                    exact = (int(round(scaled)), s)
+            elif op == "mul" and float(v) == int(v):
+                exact = (int(v), s)
+        if exact is not None:
+            vi, s_out = exact
+            da = a.data
+            if op == "mul":
+                out = da * vi
+            else:
+                x, y = (vi, da) if reflect else (da, vi)
+                out = x + y if op == "add" else x - y
+            return Column(bt.decimal128(18, s_out), out, a.mask)
+        return binary_arith_scalar(op, decimal_to_float(a), v, vdtype,
+                                   reflect)
     da = a.data
     float_out = a.dtype.is_float or isinstance(v, float) or op in _ARITH_RESULT_FLOAT
     if float_out and not da.dtype.is_floating_point:
@@ -467,6 +530,15 @@ def compare_scalar(op: str, a: Column, v, vdtype: DType) -> Column:
     if a.dtype.kind in (TypeKind.DICT, TypeKind.STRING):
         b = Column.full_const(v, vdtype, len(a), a.device)
         return compare(op, a, b)
+    if a.dtype.kind == TypeKind.DECIMAL128 and isinstance(v, (int, float)) \
+            and not isinstance(v, bool):
+        scaled = v * (10 ** a.dtype.scale)
+        if float(scaled) == round(scaled):
+            out = getattr(torch, op)(a.data, int(round(scaled)))
+            if a.mask is not None:
+                out = out & a.mask
+            return Column(bt.boolean, out)
+        return compare_scalar(op, decimal_to_float(a), v, vdtype)
     da = a.data
     if da.dtype == torch.bool and isinstance(v, (int, float)) and not isinstance(v, bool):
         da = da.to(torch.int64)
@@ -483,6 +555,20 @@ def compare(op: str, a: Column, b: Column) -> Column:
         return _compare_dict(op, a, b)
     if a.dtype.kind == TypeKind.STRING or b.dtype.kind == TypeKind.STRING:
         return _compare_string_host(op, a, b)
+    DEC = TypeKind.DECIMAL128
+    if DEC in (a.dtype.kind, b.dtype.kind):
+        if a.dtype.kind == DEC and b.dtype.kind == DEC:
+            sa, sb = a.dtype.scale, b.dtype.scale
+            s = max(sa, sb)
+            da = a.data * (10 ** (s - sa)) if sa < s else a.data
+            db = b.data * (10 ** (s - sb)) if sb < s else b.data
+            out = getattr(torch, op)(da, db)
+            invalid = combine_masks(a.mask, b.mask)
+            if invalid is not None:
+                out = out & invalid
+            return Column(bt.boolean, out)
+        return compare(op, decimal_to_float(a) if a.dtype.kind == DEC else a,
+                       decimal_to_float(b) if b.dtype.kind == DEC else b)
     da, db = a.data, b.data
     if da.dtype != db.dtype:
         if da.dtype == torch.bool:

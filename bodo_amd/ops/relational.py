@@ -27,19 +27,78 @@ PHYSICAL_AGGS = {
 }
 
 
+def _normalize_decimal_aggs(tbl: Table, aggs):
+    """Aggregate decimal columns exactly on their scaled-int64 storage
+    (sum/min/max/first/last/nunique) or on a float64 view (mean/var/...),
+    then re-wrap exact results as decimal.  Reference role:
+    bodo/libs/_decimal_ext.cpp decimal agg combine."""
+    DEC = TypeKind.DECIMAL128
+    if not any(a[1] and tbl.has_column(a[1])
+               and tbl.column(a[1]).dtype.kind == DEC for a in aggs):
+        return tbl, aggs, []
+    new_cols = {}
+    out_fix = []
+    aggs2 = []
+    for out, inn, func in aggs:
+        col = tbl.column(inn) if inn and tbl.has_column(inn) else None
+        if col is None or col.dtype.kind != DEC or callable(func):
+            aggs2.append((out, inn, func))
+            continue
+        if func in ("sum", "min", "max", "first", "last", "prod"):
+            nm = f"__deci_{inn}"
+            if nm not in new_cols:
+                new_cols[nm] = Column(bt.int64, col.data, col.mask,
+                                      length=len(col))
+            aggs2.append((out, nm, func))
+            if func != "prod":
+                out_fix.append((out, col.dtype))
+        elif func in ("count", "size", "nunique", "any", "all"):
+            nm = f"__deci_{inn}"
+            if nm not in new_cols:
+                new_cols[nm] = Column(bt.int64, col.data, col.mask,
+                                      length=len(col))
+            aggs2.append((out, nm, func))
+        else:  # mean/var/std/median/skew/quantile: float64 view
+            from .evaluate import decimal_to_float
+
+            nm = f"__decf_{inn}"
+            if nm not in new_cols:
+                new_cols[nm] = decimal_to_float(col)
+            aggs2.append((out, nm, func))
+    work = tbl
+    for nm, c in new_cols.items():
+        work = work.with_column(nm, c)
+    return work, aggs2, out_fix
+
+
 def groupby_local(tbl: Table, keys: Sequence[str],
                   aggs: Sequence[Tuple[str, str, str]],
                   dropna: bool = True) -> Table:
     """Group rows of the local shard; aggs = (out_name, in_name, func)."""
+    tbl, aggs, dec_fix = _normalize_decimal_aggs(tbl, aggs)
     if tbl.device.type == "cuda":
         if any(callable(a[2]) for a in aggs):
             # custom python agg: host pandas per co-located shard (the
             # @jit-to-HIP lowering is the native path for these)
-            return _groupby_pandas(tbl, keys, aggs, dropna)
-        from . import gpu
+            out = _groupby_pandas(tbl, keys, aggs, dropna)
+        else:
+            from . import gpu
 
-        return gpu.groupby_local(tbl, keys, aggs, dropna)
-    return _groupby_pandas(tbl, keys, aggs, dropna)
+            out = gpu.groupby_local(tbl, keys, aggs, dropna)
+    else:
+        out = _groupby_pandas(tbl, keys, aggs, dropna)
+    for out_name, dec_dtype in dec_fix:
+        c = out.column(out_name)
+        if c.data.dtype == torch.int64:
+            out = out.with_column(out_name,
+                                  Column(dec_dtype, c.data, c.mask,
+                                         length=len(c)))
+        elif c.data.dtype.is_floating_point:
+            # backend widened for empty-group NaN: value = scaled/10^s
+            out = out.with_column(out_name, Column(
+                bt.float64, c.data / float(10 ** dec_dtype.scale), c.mask,
+                length=len(c)))
+    return out
 
 
 def _groupby_pandas(tbl: Table, keys, aggs, dropna) -> Table:

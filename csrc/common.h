@@ -67,7 +67,8 @@ DEV_INLINE uint64_t hash_value(const ColumnDesc& c, int64_t i) {
     case BT_INT32: case BT_DATE32:
       return mix64((uint64_t)(int64_t)((const int32_t*)c.data)[i]);
     case BT_UINT32: return mix64((uint64_t)(int64_t)((const int32_t*)c.data)[i]);
-    case BT_INT64: case BT_TIMESTAMP_NS: case BT_UINT64:
+    case BT_INT64: case BT_TIMESTAMP_NS: case BT_UINT64: case BT_DECIMAL128:
+      // DECIMAL128 is stored as scaled int64 (exact for p <= 18)
       return mix64((uint64_t)((const int64_t*)c.data)[i]);
     case BT_BOOL: return mix64((uint64_t)((const uint8_t*)c.data)[i]);
     case BT_FLOAT32: return mix64(f64_bits_norm((double)((const float*)c.data)[i]));
@@ -100,7 +101,7 @@ DEV_INLINE bool value_eq(const ColumnDesc& c, int64_t a, int64_t b) {
       return ((const int16_t*)c.data)[a] == ((const int16_t*)c.data)[b];
     case BT_INT32: case BT_UINT32: case BT_DATE32: case BT_DICT:
       return ((const int32_t*)c.data)[a] == ((const int32_t*)c.data)[b];
-    case BT_INT64: case BT_UINT64: case BT_TIMESTAMP_NS:
+    case BT_INT64: case BT_UINT64: case BT_TIMESTAMP_NS: case BT_DECIMAL128:
       return ((const int64_t*)c.data)[a] == ((const int64_t*)c.data)[b];
     case BT_FLOAT32: {
       float x = ((const float*)c.data)[a], y = ((const float*)c.data)[b];
@@ -143,7 +144,7 @@ DEV_INLINE bool rows_eq2(const ColumnDesc* ca, const ColumnDesc* cb, int ncols,
       case BT_INT32: case BT_UINT32: case BT_DATE32: case BT_DICT:
         if (((const int32_t*)x.data)[a] != ((const int32_t*)y.data)[b]) return false;
         break;
-      case BT_INT64: case BT_UINT64: case BT_TIMESTAMP_NS:
+      case BT_INT64: case BT_UINT64: case BT_TIMESTAMP_NS: case BT_DECIMAL128:
         if (((const int64_t*)x.data)[a] != ((const int64_t*)y.data)[b]) return false;
         break;
       case BT_FLOAT32: {

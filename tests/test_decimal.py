@@ -1,0 +1,140 @@
+"""decimal128 (p<=18, scaled-int64 storage) correctness: exact arrow
+round-trips, exact arithmetic/comparison/groupby-sum against a Python
+Decimal reference (reference role: bodo/libs/_decimal_ext.cpp)."""
+
+from decimal import Decimal
+
+import numpy as np
+import pandas as pd
+import pyarrow as pa
+import pytest
+
+import bodo_amd.pandas as bpd
+from bodo_amd.core.column import Column
+from bodo_amd.core.table import Table
+from bodo_amd.core.types import TypeKind
+
+
+def _dec_arr(vals, p=15, s=2):
+    return pa.array([None if v is None else Decimal(v) for v in vals],
+                    type=pa.decimal128(p, s))
+
+
+def test_decimal_arrow_roundtrip():
+    vals = ["1.25", "-3.10", None, "99999999999.99", "0.00", "-0.01"]
+    arr = _dec_arr(vals)
+    c = Column.from_arrow(arr)
+    assert c.dtype.kind == TypeKind.DECIMAL128
+    assert c.dtype.precision == 15 and c.dtype.scale == 2
+    assert c.data.tolist() == [125, -310, 0, 9999999999999, 0, -1]
+    back = c.to_arrow()
+    assert back.equals(arr)
+
+
+def test_decimal_roundtrip_p18_extremes():
+    vals = ["9999999999999999.99", "-9999999999999999.99", "0.01"]
+    arr = _dec_arr(vals, p=18, s=2)
+    c = Column.from_arrow(arr)
+    assert c.to_arrow().equals(arr)
+
+
+def _mk_frame(n=2000, seed=3):
+    rng = np.random.default_rng(seed)
+    cents = rng.integers(-10**7, 10**7, n)
+    disc = rng.integers(0, 11, n)  # 0.00 .. 0.10
+    tbl = pa.table({
+        "k": pa.array(rng.integers(0, 25, n)),
+        "price": pa.array([Decimal(int(c)) / 100 for c in cents],
+                          type=pa.decimal128(15, 2)),
+        "disc": pa.array([Decimal(int(d)) / 100 for d in disc],
+                         type=pa.decimal128(12, 2)),
+    })
+    return tbl, cents, disc, rng
+
+
+def test_decimal_arithmetic_exact():
+    tbl, cents, disc, _ = _mk_frame()
+    t = Table.from_arrow(tbl)
+    b = bpd.from_pandas(tbl.to_pandas())
+    b["total"] = b["price"] + b["price"]
+    b["rev"] = b["price"] * b["disc"]
+    out = b.execute()
+    total = out.column("total")
+    assert total.dtype.kind == TypeKind.DECIMAL128 and total.dtype.scale == 2
+    assert total.data.tolist() == (2 * cents).tolist()
+    rev = out.column("rev")
+    assert rev.dtype.scale == 4
+    assert rev.data.tolist() == (cents * disc).tolist()
+
+
+def test_decimal_filter_and_compare():
+    tbl, cents, disc, _ = _mk_frame()
+    df = tbl.to_pandas()
+    b = bpd.from_pandas(df)
+    got = b[b["price"] > 50000.00].to_pandas()
+    exp = df[df["price"].map(lambda d: d > Decimal("50000.00"))]
+    assert len(got) == len(exp)
+    got2 = b[b["price"] <= b["disc"]].to_pandas()
+    exp2 = df[[p <= d for p, d in zip(df["price"], df["disc"])]]
+    assert len(got2) == len(exp2)
+
+
+def test_decimal_groupby_sum_exact():
+    tbl, cents, disc, _ = _mk_frame()
+    df = tbl.to_pandas()
+    b = bpd.from_pandas(df)
+    got = b.groupby("k", as_index=False).agg(
+        s=bpd.NamedAgg("price", "sum"),
+        mn=bpd.NamedAgg("price", "min"),
+        mx=bpd.NamedAgg("price", "max"),
+        av=bpd.NamedAgg("price", "mean"),
+        c=bpd.NamedAgg("price", "count"),
+    ).sort_values("k").to_pandas().reset_index(drop=True)
+    ref = pd.DataFrame({"k": df["k"], "cents": cents})
+    exp = ref.groupby("k", as_index=False).agg(
+        s=("cents", "sum"), mn=("cents", "min"), mx=("cents", "max"),
+        av=("cents", "mean"), c=("cents", "count")).sort_values(
+        "k").reset_index(drop=True)
+    # exact money: sums as Decimal == exact integer cents
+    got_s = [Decimal(str(v)) if not isinstance(v, Decimal) else v
+             for v in got["s"]]
+    assert [int(v * 100) for v in got_s] == exp["s"].tolist()
+    assert [int(Decimal(str(v)) * 100) if not isinstance(v, Decimal)
+            else int(v * 100) for v in got["mn"]] == exp["mn"].tolist()
+    np.testing.assert_allclose(got["av"].astype(float),
+                               exp["av"] / 100.0, rtol=1e-9)
+    assert got["c"].tolist() == exp["c"].tolist()
+
+
+def test_decimal_sort_and_join():
+    tbl, cents, disc, rng = _mk_frame(500, 5)
+    df = tbl.to_pandas()
+    b = bpd.from_pandas(df)
+    got = b.sort_values("price").to_pandas()
+    order = np.argsort(cents, kind="stable")
+    assert [int(Decimal(str(v)) * 100) if not isinstance(v, Decimal)
+            else int(v * 100) for v in got["price"]] == \
+        cents[order].tolist()
+
+
+def test_decimal_tpch_q1_style():
+    """l_extendedprice * (1 - l_discount) revenue with exact money."""
+    tbl, cents, disc, _ = _mk_frame(3000, 7)
+    df = tbl.to_pandas()
+    b = bpd.from_pandas(df)
+    b["rev"] = b["price"] * (b["disc"] * (-1) + 1)
+    got = b.groupby("k", as_index=False).agg(
+        r=bpd.NamedAgg("rev", "sum")).sort_values("k").to_pandas()
+    exp_scaled = pd.DataFrame({
+        "k": df["k"],
+        "r": cents * (100 - disc)}).groupby("k", as_index=False).agg(
+        r=("r", "sum")).sort_values("k")
+    got_r = [int(Decimal(str(v)) * 10000) if not isinstance(v, Decimal)
+             else int(v * 10000) for v in got["r"]]
+    assert got_r == exp_scaled["r"].tolist()
+
+
+def test_decimal_p19_falls_back_to_float():
+    arr = pa.array([Decimal("1.5"), Decimal("2.5")], type=pa.decimal128(25, 2))
+    c = Column.from_arrow(arr)
+    assert c.dtype.kind == TypeKind.FLOAT64
