@@ -134,10 +134,10 @@ def want_streaming(node, ctx) -> bool:
     import glob
     import os
 
-    cur = node.children()[0]
-    while isinstance(cur, (pn.Filter, pn.Projection)):
-        cur = cur.children()[0]
-    path = cur.path
+    walked = _walk_stream_chain(node)
+    if walked is None:
+        return False
+    path = walked[1].path
     try:
         if os.path.isdir(path):
             total = sum(os.path.getsize(p)
@@ -149,10 +149,53 @@ def want_streaming(node, ctx) -> bool:
     return total > config.STREAM_THRESHOLD_BYTES
 
 
+_STREAM_JOIN_HOWS = {"inner", "left", "semi", "anti"}
+
+
+def _walk_stream_chain(node: pn.PlanNode):
+    """Follow the streamed (probe) spine below an Aggregate/Reduce:
+    Filter/Projection pass through; a Join continues down the child whose
+    subtree holds the ParquetScan while the other side becomes a resident
+    build table (reference: push pipelines with resident build sides,
+    bodo/pandas/_pipeline.h:108 + streaming/_join.h probe loop).
+    Returns (chain ops root-to-scan order reversed later, scan) or None."""
+
+    def has_scan(nd) -> bool:
+        if isinstance(nd, pn.ParquetScan):
+            return True
+        return any(has_scan(c) for c in nd.children())
+
+    chain = []
+    cur = node.children()[0]
+    depth = 0
+    while depth < 64:
+        depth += 1
+        if isinstance(cur, (pn.Filter, pn.Projection)):
+            chain.append(("op", cur))
+            cur = cur.children()[0]
+            continue
+        if isinstance(cur, pn.Join) and cur.how in _STREAM_JOIN_HOWS:
+            left, right = cur.children()
+            ls, rs = has_scan(left), has_scan(right)
+            if ls and not rs:
+                chain.append(("join_left", cur))
+                cur = left
+                continue
+            # streaming the right side only safe for inner joins
+            if rs and not ls and cur.how == "inner":
+                chain.append(("join_right", cur))
+                cur = right
+                continue
+            return None
+        break
+    if not isinstance(cur, pn.ParquetScan):
+        return None
+    return chain, cur
+
+
 def streamable(node: pn.PlanNode) -> bool:
     """True when `node` is Aggregate/Reduce over a chain of
-    Filter/Projection over a ParquetScan (the first-pipeline shape we
-    stream)."""
+    Filter/Projection/Join-with-resident-build over a ParquetScan."""
     if not isinstance(node, (pn.Aggregate, pn.Reduce)):
         return False
     if isinstance(node, pn.Aggregate):
@@ -161,35 +204,51 @@ def streamable(node: pn.PlanNode) -> bool:
         if any((a[2] not in ok) if not callable(a[2]) else True
                for a in node.aggs):
             return False
-    cur = node.children()[0]
-    while isinstance(cur, (pn.Filter, pn.Projection)):
-        cur = cur.children()[0]
-    return isinstance(cur, pn.ParquetScan)
+    return _walk_stream_chain(node) is not None
 
 
 def exec_streaming(node, ctx):
-    """Execute Aggregate/Reduce(…(ParquetScan)) morsel-wise."""
+    """Execute Aggregate/Reduce(…chain with resident-build joins…
+    (ParquetScan)) morsel-wise: build sides materialize once (replicated
+    across ranks for local probes), probe batches flow through
+    filter/project/join without ever materializing the fact table
+    (reference: streaming/_join.h probe loop + _pipeline.h push batches)."""
     from .. import config
     from ..ops import evaluate as ev
+    from ..parallel import comm
     from . import executor as ex
 
-    # peel the operator chain
+    walked = _walk_stream_chain(node)
+    assert walked is not None
+    raw_chain, scan = walked
+    # materialize every resident build side once, up front
     chain = []
-    cur = node.children()[0]
-    while isinstance(cur, (pn.Filter, pn.Projection)):
-        chain.append(cur)
-        cur = cur.children()[0]
-    scan: pn.ParquetScan = cur
+    for kind, op in raw_chain:
+        if kind == "op":
+            chain.append((kind, op, None))
+            continue
+        build_node = op.right if kind == "join_left" else op.left
+        build = ex._exec(build_node, ctx)
+        if ctx.world > 1:
+            build = comm.allgather_table(build)
+        chain.append((kind, op, build))
     chain.reverse()  # scan-side first
 
     batch_rows = config.STREAM_BATCH_SIZE
 
     def apply_chain(batch: Table) -> Table:
-        for op in chain:
-            if isinstance(op, pn.Filter):
-                batch = ev.eval_filter(op.cond, batch)
-            else:
-                batch = ev.project(batch, op.names, op.exprs)
+        for kind, op, build in chain:
+            if kind == "op":
+                if isinstance(op, pn.Filter):
+                    batch = ev.eval_filter(op.cond, batch)
+                else:
+                    batch = ev.project(batch, op.names, op.exprs)
+            elif kind == "join_left":
+                batch = rel.join_local(batch, build, op.left_on,
+                                       op.right_on, op.how, op.suffixes)
+            else:  # join_right: resident left, streamed right (inner)
+                batch = rel.join_local(build, batch, op.left_on,
+                                       op.right_on, op.how, op.suffixes)
         return batch
 
     if isinstance(node, pn.Reduce):
