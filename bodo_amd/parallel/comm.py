@@ -156,7 +156,11 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
     Partition permutation locally (stable), then one alltoallv per buffer over
     RCCL.  xGMI note: all_to_all_single maps to grouped ncclSend/ncclRecv
     pairwise over the 7 p2p links; per-destination contiguous packing is done
-    here by the gather kernels.
+    here by the gather kernels.  ALL per-destination metadata (row counts,
+    string byte counts, mask presence, dictionaries) travels in one tensor
+    exchange + one object round up front — the per-column object collectives
+    of round 1 serialized 8-rank latencies (reference role: mpi_comm_info
+    setup in bodo/libs/_shuffle.cpp done once per table).
     """
     w = get_world_size()
     if w == 1:
@@ -166,45 +170,80 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
     perm = torch.argsort(part_ids, stable=True)
     counts = torch.bincount(part_ids, minlength=w)
     packed = ops.take_table(tbl, perm)
+    str_idx = [i for i, c in enumerate(packed.columns)
+               if c.dtype.kind == TypeKind.STRING]
+    dict_idx = [i for i, c in enumerate(packed.columns)
+                if c.dtype.kind == TypeKind.DICT]
+    dev = _comm_device()
+    # one meta exchange: [rows, str-col bytes...] per destination
+    nmeta = 1 + len(str_idx)
+    meta_send = torch.zeros(w * nmeta, dtype=torch.int64)
+    meta_send[0::nmeta] = counts.cpu()
+    bnd = torch.zeros(w + 1, dtype=torch.int64)
+    torch.cumsum(counts.cpu(), 0, out=bnd[1:])
+    for j, i in enumerate(str_idx):
+        offs = packed.columns[i].offsets
+        at = offs[bnd.to(offs.device)].cpu()
+        meta_send[1 + j::nmeta] = at[1:] - at[:-1]
+    meta_dev = meta_send.to(dev)
+    meta_recv = torch.empty_like(meta_dev)
+    dist.all_to_all_single(meta_recv, meta_dev)
+    # mask presence consensus: one small MAX all-reduce, no pickle
+    hm = torch.tensor([1 if c.mask is not None else 0
+                       for c in packed.columns],
+                      dtype=torch.int64, device=dev)
+    if hm.numel():
+        dist.all_reduce(hm, op=dist.ReduceOp.MAX)
+    # dictionaries: ONE object round for all dict columns together
+    all_dicts = None
+    if dict_idx:
+        all_dicts = allgather_obj(
+            [packed.columns[i].dictionary.to_pylist() for i in dict_idx])
+    meta = meta_recv.cpu().tolist()
+    has_mask = hm.cpu().tolist() if hm.numel() else []
     send_counts = [int(c) for c in counts.tolist()]
-    recv_counts = [int(c) for c in _exchange_counts(counts).tolist()]
-    # ONE consensus round for mask presence across all columns (a per-column
-    # object allgather serializes w latencies at 8 ranks)
-    local_has_mask = [c.mask is not None for c in packed.columns]
-    votes = allgather_obj(local_has_mask)
-    has_mask = [any(v[i] for v in votes) for i in range(len(local_has_mask))]
+    recv_counts = [int(meta[r * nmeta]) for r in range(w)]
     out_cols = []
-    for col, hm in zip(packed.columns, has_mask):
-        out_cols.append(_shuffle_column(col, send_counts, recv_counts, hm))
+    sj = 0
+    dj = 0
+    for i, col in enumerate(packed.columns):
+        if col.dtype.kind == TypeKind.STRING:
+            byte_send = [int(meta_send[1 + sj + r * nmeta]) for r in range(w)]
+            byte_recv = [int(meta[r * nmeta + 1 + sj]) for r in range(w)]
+            sj += 1
+            out_cols.append(_shuffle_string_column(
+                col, send_counts, recv_counts, byte_send, byte_recv,
+                bool(has_mask[i])))
+        elif col.dtype.kind == TypeKind.DICT:
+            data = alltoallv_tensor(col.data.contiguous(), send_counts,
+                                    recv_counts)
+            new_mask = _shuffle_mask(col, send_counts, recv_counts,
+                                     bool(has_mask[i]))
+            dicts_by_rank = [d[dj] for d in all_dicts]
+            dj += 1
+            out_cols.append(_remap_dict_codes(
+                data, new_mask, dicts_by_rank, recv_counts))
+        else:
+            data = alltoallv_tensor(col.data.contiguous(), send_counts,
+                                    recv_counts)
+            new_mask = _shuffle_mask(col, send_counts, recv_counts,
+                                     bool(has_mask[i]))
+            out_cols.append(Column(col.dtype, data, new_mask,
+                                   length=sum(recv_counts)))
     return Table(packed.names, out_cols, sum(recv_counts))
 
 
-def _shuffle_column(col: Column, send_counts, recv_counts,
-                    has_mask: bool) -> Column:
+def _shuffle_string_column(col: Column, send_counts, recv_counts,
+                           byte_send, byte_recv, has_mask: bool) -> Column:
     n_out = sum(recv_counts)
-    if col.dtype.kind == TypeKind.STRING:
-        # exchange per-row lengths, then bytes with byte-granular counts
-        lens = (col.offsets[1:] - col.offsets[:-1]).contiguous()
-        new_lens = alltoallv_tensor(lens, send_counts, recv_counts)
-        # per-destination byte counts
-        boundaries = np.cumsum([0] + list(send_counts))
-        byte_send = [int((col.offsets[boundaries[i + 1]] - col.offsets[boundaries[i]]).item())
-                     for i in range(len(send_counts))]
-        byte_recv_np = np.zeros(len(recv_counts), dtype=np.int64)
-        rb = np.cumsum([0] + list(recv_counts))
-        for i in range(len(recv_counts)):
-            byte_recv_np[i] = int(new_lens[rb[i]:rb[i + 1]].sum().item())
-        new_bytes = alltoallv_tensor(col.data, byte_send, list(byte_recv_np))
-        new_off = torch.zeros(n_out + 1, dtype=torch.int64, device=col.device)
-        torch.cumsum(new_lens, 0, out=new_off[1:])
-        new_mask = _shuffle_mask(col, send_counts, recv_counts, has_mask)
-        return Column(bt.string, new_bytes, new_mask, offsets=new_off, length=n_out)
-    data = alltoallv_tensor(col.data.contiguous(), send_counts, recv_counts)
+    lens = (col.offsets[1:] - col.offsets[:-1]).contiguous()
+    new_lens = alltoallv_tensor(lens, send_counts, recv_counts)
+    new_bytes = alltoallv_tensor(col.data, byte_send, byte_recv)
+    new_off = torch.zeros(n_out + 1, dtype=torch.int64, device=col.device)
+    torch.cumsum(new_lens, 0, out=new_off[1:])
     new_mask = _shuffle_mask(col, send_counts, recv_counts, has_mask)
-    if col.dtype.kind == TypeKind.DICT:
-        # dictionaries may differ per-rank; unify by value after exchange
-        return _unify_dict_after_shuffle(col, data, new_mask, send_counts, recv_counts, n_out)
-    return Column(col.dtype, data, new_mask, length=n_out)
+    return Column(bt.string, new_bytes, new_mask, offsets=new_off,
+                  length=n_out)
 
 
 def _shuffle_mask(col: Column, send_counts, recv_counts, has_mask: bool):
@@ -215,27 +254,27 @@ def _shuffle_mask(col: Column, send_counts, recv_counts, has_mask: bool):
     return alltoallv_tensor(m.contiguous(), send_counts, recv_counts)
 
 
-def _unify_dict_after_shuffle(col: Column, codes: torch.Tensor, mask,
-                              send_counts, recv_counts, n_out) -> Column:
+def _remap_dict_codes(codes: torch.Tensor, mask, dicts_by_rank,
+                      recv_counts) -> Column:
+    """Received dict codes from rank j index rank j's dictionary; remap all
+    segments into the value-merged dictionary."""
     import pyarrow as pa
 
-    my_dict = col.dictionary.to_pylist()
-    all_dicts = allgather_obj(my_dict)
+    n_out = sum(recv_counts)
     merged: List[str] = []
     seen = {}
-    for d in all_dicts:
+    for d in dicts_by_rank:
         for v in d:
             if v not in seen:
                 seen[v] = len(merged)
                 merged.append(v)
-    # remap received codes: rows from rank j used dict all_dicts[j]
     out = codes.clone()
     rb = np.cumsum([0] + list(recv_counts))
     for j in range(len(recv_counts)):
         if recv_counts[j] == 0:
             continue
-        remap = np.array([seen[v] for v in all_dicts[j]], dtype=np.int32) \
-            if all_dicts[j] else np.zeros(0, np.int32)
+        remap = np.array([seen[v] for v in dicts_by_rank[j]], dtype=np.int32) \
+            if dicts_by_rank[j] else np.zeros(0, np.int32)
         remap_t = torch.from_numpy(remap).to(codes.device)
         seg = codes[rb[j]:rb[j + 1]].long()
         out[rb[j]:rb[j + 1]] = remap_t[seg] if len(remap) else seg.to(torch.int32)
