@@ -193,16 +193,41 @@ _PQF_HAS_DEF = 1
 _PQF_SNAPPY = 2
 
 
+def _chunk_range(chunk_meta):
+    start = chunk_meta.dictionary_page_offset
+    if start is None or start <= 0:
+        start = chunk_meta.data_page_offset
+    return int(start), int(chunk_meta.total_compressed_size)
+
+
 class _ChunkReader:
-    def __init__(self, f, chunk_meta, phys_type: str, max_def: int = 1):
+    def __init__(self, f, chunk_meta, phys_type: str, max_def: int = 1,
+                 prefetched=None):
         self.meta = chunk_meta
         self.phys = phys_type
         self.max_def = max_def
-        start = chunk_meta.dictionary_page_offset
-        if start is None or start <= 0:
-            start = chunk_meta.data_page_offset
-        f.seek(start)
-        self.buf = f.read(chunk_meta.total_compressed_size)
+        self._hdrs = None
+        self._dev = None
+        self._dict_raw = None
+        self._lazy_src = None
+        self._buf = None
+        if prefetched is not None:
+            # (hdrs np, device tensor of the raw chunk, dict page raw bytes
+            #  or None, (path, start, size) for the rare host fallback)
+            self._hdrs, self._dev, self._dict_raw, self._lazy_src = prefetched
+        else:
+            start, size = _chunk_range(chunk_meta)
+            f.seek(start)
+            self._buf = f.read(size)
+
+    @property
+    def buf(self) -> bytes:
+        if self._buf is None:
+            path, start, size = self._lazy_src
+            with open(path, "rb") as f:
+                f.seek(start)
+                self._buf = f.read(size)
+        return self._buf
 
     # ------------------------------------------------------------------
     # batched page-parallel device decode (the fast path)
@@ -240,8 +265,11 @@ class _ChunkReader:
             return None
         import bodo_amd_kernels as K
 
-        buf_np = np.frombuffer(self.buf, dtype=np.uint8)
-        hdrs = K.pq_parse_headers(torch.from_numpy(buf_np)).numpy()
+        if self._hdrs is not None:
+            hdrs = self._hdrs
+        else:
+            buf_np = np.frombuffer(self.buf, dtype=np.uint8)
+            hdrs = K.pq_parse_headers(torch.from_numpy(buf_np)).numpy()
         if hdrs.size == 0:
             return None
         ptype, usize, csize, body, nv, enc, defenc = (hdrs[:, i]
@@ -273,7 +301,8 @@ class _ChunkReader:
             if len(di) != 1:
                 return None
             i = int(di[0])
-            raw = self.buf[int(body[i]):int(body[i]) + int(csize[i])]
+            raw = (self._dict_raw if self._dict_raw is not None
+                   else self.buf[int(body[i]):int(body[i]) + int(csize[i])])
             if comp == "SNAPPY":
                 raw = pa.Codec("snappy").decompress(
                     raw, decompressed_size=int(usize[i]))
@@ -301,7 +330,11 @@ class _ChunkReader:
             flags |= _PQF_SNAPPY
         metas["flags"] = flags
 
-        src_dev = _UPLOADER.upload(buf_np, device)
+        if self._dev is not None:
+            src_dev = self._dev
+        else:
+            src_dev = _UPLOADER.upload(
+                np.frombuffer(self.buf, dtype=np.uint8), device)
         metas_dev = torch.from_numpy(metas.view(np.uint8)).to(device)
         scratch_size = int(((d_usize + 7) & ~7).sum()) + 16
         scratch = torch.empty(scratch_size, dtype=torch.uint8, device=device)
@@ -547,19 +580,139 @@ def read_shard_gpu(path: str, columns, ctx) -> Optional[Table]:
     base, rem = divmod(len(pieces), w)
     start = r * base + min(r, rem)
     my = pieces[start:start + base + (1 if r < rem else 0)]
+    if not my:
+        return None
+    if torch.device(ctx.device).type == "cuda":
+        return _read_pieces_pipelined(my, columns, ctx)
     out_tables = []
     for fp, rg in my:
         t = _read_row_group_gpu(fp, rg, columns, ctx)
         if t is None:
             return None
         out_tables.append(t)
-    if not out_tables:
-        return None
     if len(out_tables) == 1:
         return out_tables[0]
     from .. import ops
 
     return ops.concat_tables(out_tables)
+
+
+_NSLOTS = 4
+
+
+def _read_pieces_pipelined(my, columns, ctx) -> Optional[Table]:
+    """Threaded disk->pinned prefetch overlapped with device decode.
+
+    N reader threads pull chunk byte ranges straight into a pool of pinned
+    staging buffers (`f.readinto`, no intermediate copy); the consumer
+    parses page headers (C++), snips the small dictionary page, issues the
+    async H2D copy and releases the slot for reuse once the copy's event
+    fires.  Disk read, PCIe upload and decode kernels of different chunks
+    overlap (reference role: bodo/io/parquet_reader.cpp prefetch +
+    _io_cpu_thread_pool.cpp)."""
+    import queue
+    import time
+    from concurrent.futures import ThreadPoolExecutor
+
+    import bodo_amd_kernels as K
+
+    device = ctx.device
+    items = []  # (fp, rg, cm, field, start, size, first_of_rg)
+    pf_cache = {}
+    for fp, rg in my:
+        pf = pf_cache.get(fp)
+        if pf is None:
+            pf = pq.ParquetFile(fp)
+            pf_cache[fp] = pf
+        schema = pf.schema_arrow
+        rgm = pf.metadata.row_group(rg)
+        names = columns or [schema.field(i).name
+                            for i in range(len(schema.names))]
+        first = True
+        for ci in range(rgm.num_columns):
+            cm = rgm.column(ci)
+            if cm.path_in_schema not in names:
+                continue
+            start, size = _chunk_range(cm)
+            items.append((fp, rg, cm, schema.field(cm.path_in_schema),
+                          start, size, first, names))
+            first = False
+    if not items:
+        return None
+
+    slots = [{"t": None, "np": None, "ev": None} for _ in range(_NSLOTS)]
+    slot_q: "queue.Queue[int]" = queue.Queue()
+    for i in range(_NSLOTS):
+        slot_q.put(i)
+
+    def fetch(idx):
+        fp, rg, cm, field, start, size, first, names = items[idx]
+        t0 = time.perf_counter()
+        i = slot_q.get()
+        s = slots[i]
+        if s["ev"] is not None:
+            s["ev"].synchronize()  # previous H2D out of this slot is done
+            s["ev"] = None
+        if s["t"] is None or s["t"].numel() < size:
+            s["t"] = torch.empty(max(size, 1 << 24), dtype=torch.uint8,
+                                 pin_memory=True)
+            s["np"] = s["t"].numpy()
+        with open(fp, "rb") as f:
+            f.seek(start)
+            f.readinto(memoryview(s["np"][:size]))
+        STATS["t_read"] = STATS.get("t_read", 0.0) + time.perf_counter() - t0
+        return i
+
+    ex = ThreadPoolExecutor(max_workers=_NSLOTS)
+    try:
+        futs = [ex.submit(fetch, k) for k in range(len(items))]
+        rg_tables: List[Table] = []
+        cur_cols: List[Column] = []
+        cur_names: List[str] = []
+        for k, fut in enumerate(futs):
+            fp, rg, cm, field, start, size, first, names = items[k]
+            if first and cur_cols:
+                ordered = [n for n in cur_names]
+                rg_tables.append(Table(cur_names, cur_cols).select(
+                    [n for n in items[k - 1][7] if n in ordered]))
+                cur_cols, cur_names = [], []
+            i = fut.result()
+            s = slots[i]
+            view = s["np"][:size]
+            t0 = time.perf_counter()
+            hdrs = K.pq_parse_headers(torch.from_numpy(view)).numpy()
+            dict_raw = None
+            if hdrs.size and hdrs[0, 0] == PAGE_DICT:
+                b0, c0 = int(hdrs[0, 3]), int(hdrs[0, 2])
+                dict_raw = bytes(view[b0:b0 + c0])
+            dev = torch.empty(size, dtype=torch.uint8, device=device)
+            dev.copy_(s["t"][:size], non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+            s["ev"] = ev
+            slot_q.put(i)
+            reader = _ChunkReader(
+                None, cm, cm.physical_type,
+                max_def=1 if field.nullable else 0,
+                prefetched=(hdrs, dev, dict_raw, (fp, start, size)))
+            col = reader.decode_column(device, field)
+            STATS["t_decode"] = STATS.get("t_decode", 0.0) + \
+                time.perf_counter() - t0
+            if col is None:
+                return None
+            cur_cols.append(col)
+            cur_names.append(cm.path_in_schema)
+        if cur_cols:
+            last_names = items[-1][7]
+            rg_tables.append(Table(cur_names, cur_cols).select(
+                [n for n in last_names if n in cur_names]))
+    finally:
+        ex.shutdown(wait=True)
+    if len(rg_tables) == 1:
+        return rg_tables[0]
+    from .. import ops
+
+    return ops.concat_tables(rg_tables)
 
 
 def _read_row_group_gpu(fp: str, rg: int, columns, ctx) -> Optional[Table]:
