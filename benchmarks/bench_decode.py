@@ -50,6 +50,9 @@ def main():
     ctx.world, ctx.rank = 1, 0
     # warm page cache + jit
     t = g.read_shard_gpu(path, None, ctx)
+    if t is None and device == "cpu":
+        print(json.dumps({"skipped": "device decode needs a GPU"}))
+        return
     assert t is not None and len(t) == args.rows, (t and len(t), args.rows)
     if device == "cuda":
         torch.cuda.synchronize()
