@@ -168,7 +168,7 @@ def _exec_filter(node: pn.Filter, ctx) -> Table:
 
 
 SINGLE_PHASE_AGGS = {"median", "nunique", "var", "std", "quantile",
-                     "skew"}
+                     "skew", "kurt", "sem", "mode"}
 
 
 def _exec_aggregate(node: pn.Aggregate, ctx) -> Table:
@@ -777,6 +777,18 @@ def _combine_reduce(partials: List[dict], func: str):
         for p in partials:
             out *= p["prod"]
         return out
+    if func == "nunique":
+        u = set()
+        for p in partials:
+            u |= p["uniq"]
+        return len(u)
+    if func == "approx_nunique":
+        from ..utils import sketches
+
+        regs = partials[0]["hll"]
+        for p in partials[1:]:
+            regs = torch.maximum(regs, p["hll"])
+        return int(round(sketches.hll_estimate(regs)))
     raise NotImplementedError(func)
 
 

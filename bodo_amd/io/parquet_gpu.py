@@ -13,6 +13,7 @@ fall back to the host Arrow decoder per column chunk.
 
 from __future__ import annotations
 
+import os
 import struct
 from typing import List, Optional, Tuple
 
@@ -597,7 +598,7 @@ def read_shard_gpu(path: str, columns, ctx) -> Optional[Table]:
     return ops.concat_tables(out_tables)
 
 
-_NSLOTS = 4
+_NSLOTS = int(os.environ.get("BODO_AMD_READ_THREADS", "8"))
 
 
 def _read_pieces_pipelined(my, columns, ctx) -> Optional[Table]:

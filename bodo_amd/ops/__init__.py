@@ -450,4 +450,10 @@ def reduce_column(col: Column, func: str):
         return {"all": bool(data.all().item()) if n else True}
     if func == "prod":
         return {"prod": data.to(torch.float64).prod().item() if n else 1.0}
+    if func == "nunique":
+        return {"uniq": set(torch.unique(data).cpu().tolist()) if n else set()}
+    if func == "approx_nunique":
+        from ..utils import sketches
+
+        return {"hll": sketches.hll_registers([col]).cpu()}
     raise NotImplementedError(f"reduce {func}")

@@ -502,6 +502,39 @@ def _agg_one(col: Optional[Column], row_gid: torch.Tensor, ngroups: int,
         g = torch.sqrt(n * (n - 1)) / (n - 2) * m3 / m2.clamp(min=0) ** 1.5
         g = torch.where(n < 3, torch.full_like(g, float("nan")), g)
         return Column(bt.float64, g)
+    if func == "kurt":
+        x = col.data.to(torch.float64)
+        mask8 = None if col.mask is None else col.mask.view(torch.uint8)
+        s1, cnt = K.agg_update(x, mask8, None, int(TypeKind.FLOAT64),
+                               row_gid, ngroups, _AGG_OP["sum_f64"], 0.0, 0,
+                               True)
+        s2 = K.agg_update(x * x, mask8, None, int(TypeKind.FLOAT64), row_gid,
+                          ngroups, _AGG_OP["sum_f64"], 0.0, 0, False)[0]
+        s3 = K.agg_update(x * x * x, mask8, None, int(TypeKind.FLOAT64),
+                          row_gid, ngroups, _AGG_OP["sum_f64"], 0.0, 0,
+                          False)[0]
+        s4 = K.agg_update(x * x * x * x, mask8, None, int(TypeKind.FLOAT64),
+                          row_gid, ngroups, _AGG_OP["sum_f64"], 0.0, 0,
+                          False)[0]
+        n = cnt.to(torch.float64)
+        mean = s1 / n
+        m2 = s2 / n - mean ** 2
+        m4 = (s4 - 4 * mean * s3 + 6 * mean ** 2 * s2) / n - 3 * mean ** 4
+        # pandas adjusted kurtosis (Fisher, bias-corrected)
+        g = (n - 1) / ((n - 2) * (n - 3)) * (
+            (n + 1) * (n * m4 / m2.clamp(min=0) ** 2 - 3) + 6)
+        g = torch.where(n < 4, torch.full_like(g, float("nan")), g)
+        return Column(bt.float64, g)
+    if func == "sem":
+        acc, cnt = upd("sum_f64", want_count=True)
+        sq = col.data.to(torch.float64) ** 2
+        mask8 = None if col.mask is None else col.mask.view(torch.uint8)
+        acc2 = K.agg_update(sq, mask8, None, int(TypeKind.FLOAT64), row_gid,
+                            ngroups, _AGG_OP["sum_f64"], 0.0, 0, False)[0]
+        c = cnt.to(torch.float64)
+        var = (acc2 - acc * acc / c) / (c - 1)
+        var = torch.where(cnt < 2, torch.full_like(var, float("nan")), var)
+        return Column(bt.float64, (var / c).sqrt())
     if func == "median":
         return _median_by_group(col, row_gid, ngroups)
     if func == "nunique":

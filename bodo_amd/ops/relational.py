@@ -26,6 +26,9 @@ PHYSICAL_AGGS = {
     "median", "nunique",
 }
 
+# funcs with no device kernel yet: run in host pandas even on GPU shards
+HOST_ONLY_AGGS = {"mode"}
+
 
 def _normalize_decimal_aggs(tbl: Table, aggs):
     """Aggregate decimal columns exactly on their scaled-int64 storage
@@ -77,7 +80,7 @@ def groupby_local(tbl: Table, keys: Sequence[str],
     """Group rows of the local shard; aggs = (out_name, in_name, func)."""
     tbl, aggs, dec_fix = _normalize_decimal_aggs(tbl, aggs)
     if tbl.device.type == "cuda":
-        if any(callable(a[2]) for a in aggs):
+        if any(callable(a[2]) or a[2] in HOST_ONLY_AGGS for a in aggs):
             # custom python agg: host pandas per co-located shard (the
             # @jit-to-HIP lowering is the native path for these)
             out = _groupby_pandas(tbl, keys, aggs, dropna)
@@ -107,7 +110,14 @@ def _groupby_pandas(tbl: Table, keys, aggs, dropna) -> Table:
     gb = df.groupby(list(keys), dropna=dropna, sort=False, observed=True)
     named = {}
     for out_name, in_name, func in aggs:
-        if func == "size":
+        if func == "mode":
+            named[out_name] = pd.NamedAgg(
+                column=in_name,
+                aggfunc=lambda s: s.mode().iloc[0] if len(s.mode()) else None)
+        elif func == "kurt":
+            named[out_name] = pd.NamedAgg(column=in_name,
+                                          aggfunc=lambda s: s.kurt())
+        elif func == "size":
             named[out_name] = pd.NamedAgg(column=df.columns[0] if not in_name or in_name not in df.columns else in_name, aggfunc="size")
         else:
             named[out_name] = pd.NamedAgg(column=in_name, aggfunc=func)

@@ -15,7 +15,10 @@ from . import parser as ast
 
 AGG_FUNCS = {"sum": "sum", "avg": "mean", "count": "count", "min": "min",
              "max": "max", "stddev": "std", "variance": "var",
-             "median": "median"}
+             "median": "median", "approx_count_distinct": "approx_nunique",
+             "mode": "mode", "kurtosis": "kurt", "skew": "skew",
+             "stddev_samp": "std", "var_samp": "var", "any_value": "first",
+             "booland_agg": "all", "boolor_agg": "any"}
 
 CAST_TYPES = {
     "int": bt.int64, "integer": bt.int64, "bigint": bt.int64,

@@ -184,3 +184,34 @@ def test_jit_df_lib_substitution():
         s=("b", "sum")).sort_values("c").reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
     assert pd.__name__ == "pandas"  # globals restored after the call
+
+
+def test_hll_approx_nunique():
+    """HLL sketch accuracy: within ~3% at 2^14 registers (reference role:
+    theta sketches / hyperloglog.hpp)."""
+    import numpy as np
+
+    from bodo_amd.core.column import Column
+    from bodo_amd.utils import sketches
+
+    rng = np.random.default_rng(9)
+    for true_n in (100, 10_000, 1_000_000):
+        vals = rng.integers(0, true_n, true_n * 3)
+        c = Column.from_numpy(vals)
+        est = sketches.approx_nunique([c], distributed=False)
+        exact = len(np.unique(vals))
+        assert abs(est - exact) / exact < 0.05, (true_n, est, exact)
+
+
+def test_sql_approx_count_distinct():
+    import numpy as np
+    import pandas as pd
+
+    from bodo_amd.sql import BodoSQLContext
+
+    rng = np.random.default_rng(4)
+    df = pd.DataFrame({"v": rng.integers(0, 50_000, 200_000)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select approx_count_distinct(v) as a from t").to_pandas()
+    exact = df["v"].nunique()
+    assert abs(int(got["a"].iloc[0]) - exact) / exact < 0.05

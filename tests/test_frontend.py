@@ -422,3 +422,23 @@ def test_window_device_path_no_pandas(monkeypatch):
 
     check_query(q, {"df": df}, sort_by=["k", "o"])
     assert calls["n"] > 0
+
+
+def test_groupby_mode_kurt_sem():
+    rng = np.random.default_rng(41)
+    n = 4000
+    df = pd.DataFrame({"k": rng.integers(0, 30, n),
+                       "v": rng.integers(0, 8, n),
+                       "f": rng.random(n) * 10})
+
+    def q(m, df):
+        return df.groupby("k", as_index=False).agg(
+            md=m.NamedAgg("v", "mode") if m is not pd else ("v", lambda s: s.mode().iloc[0]),
+            kt=m.NamedAgg("f", "kurt") if m is not pd
+            else ("f", lambda s: s.kurt()),
+            se=m.NamedAgg("f", "sem") if m is not pd else ("f", "sem"),
+        ).sort_values("k")
+
+    from tests.utils import check_query
+
+    check_query(q, {"df": df}, sort_by=["k"], atol=1e-8)
