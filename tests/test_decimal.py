@@ -138,3 +138,26 @@ def test_decimal_p19_falls_back_to_float():
     arr = pa.array([Decimal("1.5"), Decimal("2.5")], type=pa.decimal128(25, 2))
     c = Column.from_arrow(arr)
     assert c.dtype.kind == TypeKind.FLOAT64
+
+
+@pytest.mark.gpu
+def test_decimal_gpu_groupby_and_filter():
+    import torch
+
+    import bodo_amd.config as cfg
+
+    cfg.DEVICE = "cuda"
+    tbl, cents, disc, _ = _mk_frame(100_000, 11)
+    df = tbl.to_pandas()
+    b = bpd.from_pandas(df)
+    got = b[b["price"] > 0].groupby("k", as_index=False).agg(
+        s=bpd.NamedAgg("price", "sum"),
+        c=bpd.NamedAgg("price", "count")).sort_values("k").to_pandas()
+    pos = cents > 0
+    ref = pd.DataFrame({"k": df["k"][pos], "cents": cents[pos]})
+    exp = ref.groupby("k", as_index=False).agg(
+        s=("cents", "sum"), c=("cents", "count")).sort_values("k")
+    got_s = [int(Decimal(str(v)) * 100) if not isinstance(v, Decimal)
+             else int(v * 100) for v in got["s"]]
+    assert got_s == exp["s"].tolist()
+    assert got["c"].tolist() == exp["c"].tolist()
