@@ -347,7 +347,13 @@ class _ChunkReader:
         if self.max_def > 0:
             mask_u8, n_valid, vdo = K.pq_def_levels(
                 scratch, metas_dev, n_pages, 1, 1, total_nv)
-            nvalid_total = int(n_valid.sum().item())
+            # chunk statistics carry the null count: avoids a device sync
+            # per chunk (the round-1 pipeline serializer)
+            st = self.meta.statistics
+            if st is not None and st.has_null_count:
+                nvalid_total = total_nv - int(st.null_count)
+            else:
+                nvalid_total = int(n_valid.sum().item())
             dense_off = torch.cumsum(n_valid.long(), 0) - n_valid.long()
             if nvalid_total < total_nv:
                 mask_t = mask_u8.to(torch.bool)

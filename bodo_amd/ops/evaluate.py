@@ -441,6 +441,12 @@ def binary_arith(op: str, a: Column, b: Column) -> Column:
         out = torch.remainder(da, db)
     elif op == "pow":
         out = torch.pow(da, db)
+    elif op in ("bitand", "bitor", "bitxor", "lshift", "rshift"):
+        da, db = da.long(), db.long()
+        out = {"bitand": torch.bitwise_and, "bitor": torch.bitwise_or,
+               "bitxor": torch.bitwise_xor,
+               "lshift": torch.bitwise_left_shift,
+               "rshift": torch.bitwise_right_shift}[op](da, db)
     else:
         raise NotImplementedError(op)
     mask = combine_masks(a.mask, b.mask)
@@ -478,6 +484,15 @@ def binary_arith_scalar(op: str, a: Column, v, vdtype: DType,
     elif a.dtype.is_float and da.dtype == torch.float32 and isinstance(v, float):
         pass
     x, y = (v, da) if reflect else (da, v)
+    if op in ("bitand", "bitor", "bitxor", "lshift", "rshift"):
+        da = da.long()
+        x2, y2 = (v, da) if reflect else (da, v)
+        out = {"bitand": torch.bitwise_and, "bitor": torch.bitwise_or,
+               "bitxor": torch.bitwise_xor,
+               "lshift": torch.bitwise_left_shift,
+               "rshift": torch.bitwise_right_shift}[op](
+            x2 if torch.is_tensor(x2) else torch.tensor(x2), y2)
+        return Column(bt.int64, out, a.mask)
     if op == "add":
         out = x + y
     elif op == "sub":
@@ -606,6 +621,26 @@ def _compare_dict(op: str, a: Column, b: Column):
         out = lut[a.data.long()]
         if a.mask is not None:
             out = out & a.mask
+        return Column(bt.boolean, out)
+    if a.dtype.kind == TypeKind.DICT and b.dtype.kind == TypeKind.DICT:
+        # dict vs dict: per-dictionary value-rank LUTs against the merged
+        # value order make code comparison order- and equality-correct
+        import pyarrow as pa
+
+        av = a.dictionary.to_pylist()
+        bv = b.dictionary.to_pylist()
+        merged = sorted({v for v in av + bv if v is not None})
+        rank = {v: i for i, v in enumerate(merged)}
+        la = torch.tensor([rank.get(v, -1) for v in av] or [0],
+                          dtype=torch.int64, device=a.device)
+        lb = torch.tensor([rank.get(v, -1) for v in bv] or [0],
+                          dtype=torch.int64, device=a.device)
+        ra = la[a.data.long()]
+        rb = lb[b.data.long()]
+        out = getattr(torch, op)(ra, rb)
+        invalid = combine_masks(a.mask, b.mask)
+        if invalid is not None:
+            out = out & invalid
         return Column(bt.boolean, out)
     raise NotImplementedError("dict-dict comparison")
 
@@ -776,10 +811,12 @@ def str_op(a: Column, op: str, args, kwargs=None) -> Column:
     if a.dtype.kind == TypeKind.DICT:
         # operate on the (small) dictionary, keep indices
         d = a.dictionary
-        if op in ("lower", "upper", "strip", "title", "capitalize", "slice"):
+        if op in ("lower", "upper", "strip", "title", "capitalize", "slice",
+                  "reverse"):
             f = {"lower": pc.utf8_lower, "upper": pc.utf8_upper,
                  "strip": pc.utf8_trim_whitespace, "title": pc.utf8_title,
                  "capitalize": pc.utf8_capitalize,
+                 "reverse": pc.utf8_reverse,
                  "slice": lambda x: pc.utf8_slice_codeunits(
                      x, args[0], None if len(args) < 2 or args[1] is None
                      else args[1], args[2] if len(args) > 2 and args[2] else 1),
@@ -801,8 +838,13 @@ def str_op(a: Column, op: str, args, kwargs=None) -> Column:
                               dictionary=pa.array(uniq, type=pa.large_string()),
                               length=len(a))
             return Column(a.dtype, a.data, a.mask, dictionary=nd, length=len(a))
-        if op in ("contains", "contains_re", "match", "startswith",
-                  "endswith", "len"):
+        if op in ("contains", "contains_re", "match", "match_full",
+                  "startswith", "endswith", "count_re", "len"):
+            if op == "count_re":
+                lut_np = pc.count_substring_regex(
+                    d, args[0]).to_numpy(zero_copy_only=False).astype(np.int64)
+                lut = torch.from_numpy(lut_np).to(a.device)
+                return Column(bt.int64, lut[a.data.long()], a.mask)
             if op == "len":
                 lut_np = pc.utf8_length(d).to_numpy(zero_copy_only=False).astype(np.int64)
                 ret = bt.int64
@@ -811,6 +853,8 @@ def str_op(a: Column, op: str, args, kwargs=None) -> Column:
                 f = {"contains": lambda x: pc.match_substring(x, pat),
                      "contains_re": lambda x: pc.match_substring_regex(x, pat),
                      "match": lambda x: pc.match_substring_regex(x, "^" + pat),
+                     "match_full": lambda x: pc.match_substring_regex(
+                         x, "^(?:" + pat + ")$"),
                      "startswith": lambda x: pc.starts_with(x, pat),
                      "endswith": lambda x: pc.ends_with(x, pat)}[op]
                 lut_np = f(d).to_numpy(zero_copy_only=False).astype(bool)
@@ -834,6 +878,10 @@ def str_op(a: Column, op: str, args, kwargs=None) -> Column:
         "title": lambda x: pc.utf8_title(x),
         "capitalize": lambda x: pc.utf8_capitalize(x),
         "slice": lambda x: pc.utf8_slice_codeunits(x, *args),
+        "reverse": lambda x: pc.utf8_reverse(x),
+        "count_re": lambda x: pc.count_substring_regex(x, args[0]),
+        "match_full": lambda x: pc.match_substring_regex(
+            x, "^(?:" + args[0] + ")$"),
     }
     if op not in fmap:
         # generic pandas .str on host (replace/zfill/pad/isdigit/...)
@@ -865,8 +913,26 @@ def udf_map(a: Column, func, na_action=None) -> Column:
         import pyarrow as pa
 
         if all(v is None or isinstance(v, str) for v in vals):
-            return Column(a.dtype, a.data, a.mask,
-                          dictionary=pa.array(vals, type=pa.large_string()), length=len(a))
+            new_mask = a.mask
+            codes = a.data
+            if any(v is None for v in vals):
+                # null dictionary entries break downstream categorical
+                # conversion: null those rows instead
+                isnull = torch.tensor([v is None for v in vals],
+                                      dtype=torch.bool, device=a.device)
+                row_null = isnull[codes.long()]
+                new_mask = ~row_null if new_mask is None                     else (new_mask & ~row_null)
+                vals = ["" if v is None else v for v in vals]
+            if len(set(vals)) != len(vals):
+                uniq = list(dict.fromkeys(vals))
+                code_of = {v: i for i, v in enumerate(uniq)}
+                remap = torch.tensor([code_of[v] for v in vals],
+                                     dtype=torch.int32, device=a.device)
+                codes = remap[codes.long()]
+                vals = uniq
+            return Column(a.dtype, codes, new_mask,
+                          dictionary=pa.array(vals, type=pa.large_string()),
+                          length=len(a))
         lut = torch.tensor([np.nan if v is None else v for v in vals],
                            dtype=torch.float64, device=a.device)
         return Column(bt.float64, lut[a.data.long()], a.mask)

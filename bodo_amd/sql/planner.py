@@ -1296,6 +1296,179 @@ class Planner:
                 for a in args[1:]:
                     out = ex.BinOp("concat", out, a)
                 return out
+            if name == "reverse":
+                return ex.StrOp(self.expr(e.args[0], scope), "reverse")
+            if name == "contains":
+                return ex.StrOp(self.expr(e.args[0], scope), "contains",
+                                (e.args[1].value,))
+            if name == "startswith":
+                return ex.StrOp(self.expr(e.args[0], scope), "startswith",
+                                (e.args[1].value,))
+            if name == "endswith":
+                return ex.StrOp(self.expr(e.args[0], scope), "endswith",
+                                (e.args[1].value,))
+            if name in ("regexp_like", "rlike", "regexp"):
+                # Snowflake: the pattern must match the WHOLE subject
+                return ex.StrOp(self.expr(e.args[0], scope), "match_full",
+                                (e.args[1].value,))
+            if name == "regexp_count":
+                return ex.StrOp(self.expr(e.args[0], scope), "count_re",
+                                (e.args[1].value,))
+            if name == "regexp_replace":
+                rep = e.args[2].value if len(e.args) > 2 else ""
+                return ex.StrOp(self.expr(e.args[0], scope), "replace",
+                                (e.args[1].value, rep), (("regex", True),))
+            if name == "regexp_substr":
+                import re as _re
+
+                pat = e.args[1].value
+                rx = _re.compile(pat)
+
+                def _sub(v, rx=rx):
+                    if v is None:
+                        return None
+                    m = rx.search(v)
+                    return m.group(0) if m else None
+
+                return ex.UdfMap(self.expr(e.args[0], scope), _sub, "ignore")
+            if name == "translate":
+                tbl_map = str.maketrans(e.args[1].value, e.args[2].value)
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v, _t=tbl_map: v.translate(_t),
+                                 "ignore")
+            if name == "ascii":
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v: ord(v[0]) if v else 0, "ignore")
+            if name in ("chr", "char"):
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v: chr(int(v)), "ignore")
+            if name == "space":
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v: " " * int(v), "ignore")
+            if name == "md5":
+                import hashlib
+
+                return ex.UdfMap(
+                    self.expr(e.args[0], scope),
+                    lambda v: hashlib.md5(str(v).encode()).hexdigest(),
+                    "ignore")
+            if name == "sha1":
+                import hashlib
+
+                return ex.UdfMap(
+                    self.expr(e.args[0], scope),
+                    lambda v: hashlib.sha1(str(v).encode()).hexdigest(),
+                    "ignore")
+            if name == "sha2":
+                import hashlib
+
+                bits = int(e.args[1].value) if len(e.args) > 1 else 256
+                algo = {224: hashlib.sha224, 256: hashlib.sha256,
+                        384: hashlib.sha384, 512: hashlib.sha512}[bits]
+                return ex.UdfMap(
+                    self.expr(e.args[0], scope),
+                    lambda v, _a=algo: _a(str(v).encode()).hexdigest(),
+                    "ignore")
+            if name == "base64_encode":
+                import base64
+
+                return ex.UdfMap(
+                    self.expr(e.args[0], scope),
+                    lambda v: base64.b64encode(str(v).encode()).decode(),
+                    "ignore")
+            if name == "base64_decode_string":
+                import base64
+
+                return ex.UdfMap(
+                    self.expr(e.args[0], scope),
+                    lambda v: base64.b64decode(v).decode(), "ignore")
+            if name == "hex_encode":
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v: str(v).encode().hex().upper(),
+                                 "ignore")
+            if name == "hash":
+                # deterministic per-value FNV-1a (reference role: HASH())
+                def _fnv(v):
+                    h = 0xcbf29ce484222325
+                    for b in str(v).encode():
+                        h = ((h ^ b) * 0x100000001b3) & ((1 << 64) - 1)
+                    return h - (1 << 64) if h >= (1 << 63) else h
+
+                return ex.UdfMap(self.expr(e.args[0], scope), _fnv, "ignore")
+            if name in ("truncate", "trunc") and len(e.args) <= 2 and not (
+                    e.args and isinstance(e.args[-1], ast.Lit)
+                    and isinstance(e.args[-1].value, str)):
+                import math as _math
+
+                d = int(e.args[1].value) if len(e.args) > 1 else 0
+                sc = 10.0 ** d
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v, _s=sc: _math.trunc(v * _s) / _s,
+                                 None)
+            if name in ("sin", "cos", "tan", "asin", "acos", "atan", "sinh",
+                        "cosh", "tanh", "degrees", "radians", "cbrt"):
+                import math as _math
+
+                f = {"sin": _math.sin, "cos": _math.cos, "tan": _math.tan,
+                     "asin": _math.asin, "acos": _math.acos,
+                     "atan": _math.atan, "sinh": _math.sinh,
+                     "cosh": _math.cosh, "tanh": _math.tanh,
+                     "degrees": _math.degrees, "radians": _math.radians,
+                     "cbrt": lambda v: _math.copysign(abs(v) ** (1 / 3), v),
+                     }[name]
+                return ex.UdfMap(self.expr(e.args[0], scope), f, None)
+            if name == "pi":
+                import math as _math
+
+                return ex.Const(_math.pi, bt.float64)
+            if name in ("bitand", "bitor", "bitxor"):
+                return ex.BinOp(name, self.expr(e.args[0], scope),
+                                self.expr(e.args[1], scope))
+            if name in ("bitshiftleft", "bitshiftright"):
+                op = "lshift" if name == "bitshiftleft" else "rshift"
+                return ex.BinOp(op, self.expr(e.args[0], scope),
+                                self.expr(e.args[1], scope))
+            if name == "bitnot":
+                return ex.BinOp("bitxor", self.expr(e.args[0], scope),
+                                ex.Const(-1, bt.int64))
+            if name == "nullifzero":
+                a = self.expr(e.args[0], scope)
+                return ex.Case((ex.Cmp("eq", a, ex.Const(0)),),
+                               (ex.Const(None, bt.float64),), a)
+            if name == "equal_null":
+                a = self.expr(e.args[0], scope)
+                b = self.expr(e.args[1], scope)
+                return ex.BoolOp("or", ex.Cmp("eq", a, b),
+                                 ex.BoolOp("and", ex.IsNull(a),
+                                           ex.IsNull(b)))
+            if name in ("dayofweek", "dayofweekiso"):
+                # pandas convention: Monday=0 (ISO: Monday=1)
+                f = ex.DtField(self.expr(e.args[0], scope), "dayofweek")
+                return ex.BinOp("add", f, ex.Const(1)) \
+                    if name == "dayofweekiso" else f
+            if name == "dayofyear":
+                return ex.DtField(self.expr(e.args[0], scope), "dayofyear")
+            if name in ("week", "weekofyear", "weekiso"):
+                d = ex.DtField(self.expr(e.args[0], scope), "dayofyear")
+                return ex.BinOp("add",
+                                ex.BinOp("floordiv",
+                                         ex.BinOp("sub", d, ex.Const(1)),
+                                         ex.Const(7)), ex.Const(1))
+            if name in ("dayname", "monthname"):
+                if name == "dayname":
+                    fld, names = "dayofweek", ["Mon", "Tue", "Wed", "Thu",
+                                               "Fri", "Sat", "Sun"]
+                    base = 0
+                else:
+                    fld, names = "month", ["Jan", "Feb", "Mar", "Apr", "May",
+                                           "Jun", "Jul", "Aug", "Sep", "Oct",
+                                           "Nov", "Dec"]
+                    base = 1
+                f = ex.DtField(self.expr(e.args[0], scope), fld)
+                conds = tuple(ex.Cmp("eq", f, ex.Const(i + base))
+                              for i in range(len(names) - 1))
+                thens = tuple(ex.Const(nm) for nm in names[:-1])
+                return ex.Case(conds, thens, ex.Const(names[-1]))
             raise NotImplementedError(f"SQL function {name}")
         raise NotImplementedError(f"expr {e}")
 

@@ -608,3 +608,77 @@ def test_listagg():
     got = bc.sql("select k, listagg(s, ',') as l from t "
                  "group by k order by k").to_pandas()
     assert [str(x) for x in got.l] == ["a,b", "c,d,e"]
+
+
+def test_sql_function_breadth():
+    """Snowflake-surface scalar functions (reference: BodoSQL/bodosql/kernels
+    string/regexp/crypto/bitwise modules)."""
+    import numpy as np
+
+    from bodo_amd.sql import BodoSQLContext
+
+    df = pd.DataFrame({
+        "s": ["Hello World", "abc", None, "xyzzy", "Data engine"],
+        "n": [5, -3, 7, 0, 12],
+        "f": [1.5, -2.25, 0.0, 9.75, -0.5],
+        "d": pd.to_datetime(["2023-01-15", "2023-06-30", "2023-12-25",
+                             "2024-02-29", "2023-07-04"]),
+    })
+    bc = BodoSQLContext({"t": df})
+    q = """
+    select reverse(s) as rev,
+           contains(s, 'o') as has_o,
+           startswith(s, 'He') as st,
+           regexp_like(s, '[A-Za-z ]+') as rl,
+           regexp_count(s, '[aeiou]') as vc,
+           regexp_replace(s, '[aeiou]', '_') as rr,
+           regexp_substr(s, '[A-Z][a-z]+') as rs,
+           translate(s, 'lo', '10') as tr,
+           ascii(s) as asc_,
+           md5(s) as m5,
+           sha2(s, 256) as sh,
+           base64_encode(s) as b64,
+           bitand(n, 6) as ba, bitor(n, 1) as bo, bitxor(n, 3) as bx,
+           bitshiftleft(n, 2) as bl,
+           sin(f) as sn, degrees(f) as dg,
+           trunc(f, 1) as tc,
+           nullifzero(n) as nz,
+           equal_null(s, s) as eqn,
+           dayname(d) as dn, monthname(d) as mn,
+           dayofweek(d) as dw, dayofyear(d) as dy, weekofyear(d) as wk
+    from t
+    """
+    got = bc.sql(q).to_pandas()
+    import base64 as b64mod
+    import hashlib
+    import math
+    import re
+
+    s0 = "Hello World"
+    assert got["rev"].iloc[0] == s0[::-1]
+    assert bool(got["has_o"].iloc[0]) is True
+    assert bool(got["st"].iloc[0]) is True
+    assert bool(got["rl"].iloc[0]) is True
+    assert int(got["vc"].iloc[0]) == len(re.findall("[aeiou]", s0))
+    assert got["rr"].iloc[0] == re.sub("[aeiou]", "_", s0)
+    assert got["rs"].iloc[0] == "Hello"
+    assert got["tr"].iloc[0] == s0.translate(str.maketrans("lo", "10"))
+    assert int(got["asc_"].iloc[0]) == ord("H")
+    assert got["m5"].iloc[0] == hashlib.md5(s0.encode()).hexdigest()
+    assert got["sh"].iloc[0] == hashlib.sha256(s0.encode()).hexdigest()
+    assert got["b64"].iloc[0] == b64mod.b64encode(s0.encode()).decode()
+    assert int(got["ba"].iloc[0]) == 5 & 6
+    assert int(got["bo"].iloc[0]) == 5 | 1
+    assert int(got["bx"].iloc[0]) == 5 ^ 3
+    assert int(got["bl"].iloc[0]) == 5 << 2
+    assert abs(float(got["sn"].iloc[0]) - math.sin(1.5)) < 1e-12
+    assert abs(float(got["dg"].iloc[0]) - math.degrees(1.5)) < 1e-9
+    assert abs(float(got["tc"].iloc[1]) - (-2.2)) < 1e-12
+    assert pd.isna(got["nz"].iloc[3])
+    assert int(got["nz"].iloc[0]) == 5
+    assert bool(got["eqn"].iloc[0]) is True
+    assert got["dn"].iloc[0] == "Sun"  # 2023-01-15 is a Sunday
+    assert got["mn"].iloc[1] == "Jun"
+    assert int(got["dw"].iloc[0]) == 6  # pandas Monday=0
+    assert int(got["dy"].iloc[0]) == 15
+    assert int(got["wk"].iloc[0]) == 3
