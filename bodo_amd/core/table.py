@@ -111,6 +111,7 @@ class Table:
 
     @staticmethod
     def from_pandas(df: pd.DataFrame, device="cpu") -> "Table":
+        df = _unwrap_lazy_scalars(df)
         tbl = pa.Table.from_pandas(df, preserve_index=False)
         tbl = dict_encode_strings(tbl)
         return Table.from_arrow(tbl, device)
@@ -146,6 +147,26 @@ class Table:
 
     def __repr__(self) -> str:  # pragma: no cover
         return f"Table({len(self)} rows x {self.num_columns} cols, device={self.device})"
+
+
+def _unwrap_lazy_scalars(df: pd.DataFrame) -> pd.DataFrame:
+    """Materialize BodoScalar values embedded in object columns (user code
+    like pd.DataFrame({"r": [s.sum()]}) — drop-in contract)."""
+    from ..pandas.scalar import BodoScalar
+
+    out = None
+    for c in df.columns:
+        col = df[c]
+        if col.dtype != object:
+            continue
+        probe = col if len(col) <= 1024 else col.head(64)
+        if any(isinstance(v, BodoScalar) for v in probe):
+            if out is None:
+                out = df.copy()
+            out[c] = pd.Series(
+                [v.value if isinstance(v, BodoScalar) else v for v in col],
+                index=col.index).infer_objects()
+    return out if out is not None else df
 
 
 def dict_encode_strings(tbl: pa.Table, threshold: float = 0.7,

@@ -333,6 +333,10 @@ class _Evaluator:
 # ----------------------------------------------------------------------
 
 def infer_const_dtype(v) -> DType:
+    from ..pandas.scalar import BodoScalar
+
+    if isinstance(v, BodoScalar):
+        v = v.value
     if isinstance(v, bool):
         return bt.boolean
     if isinstance(v, int):
@@ -351,6 +355,10 @@ def infer_const_dtype(v) -> DType:
 
 
 def normalize_const(v, dtype: DType):
+    from ..pandas.scalar import BodoScalar
+
+    if isinstance(v, BodoScalar):
+        v = v.value
     if v is None:
         return float("nan") if dtype.is_float else None
     if dtype.kind == TypeKind.TIMESTAMP_NS:

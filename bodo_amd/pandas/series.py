@@ -212,14 +212,14 @@ class BodoSeries:
         return _StrAccessor(self)
 
     # ------------------------------------------------------------------
-    # reductions (execute eagerly, return python scalar)
+    # reductions (lazy scalars: the plan executes on first value access —
+    # reference: BodoScalar, bodo/pandas/scalar.py:14)
     # ------------------------------------------------------------------
     def _reduce(self, func: str):
-        from ..engine import api
+        from .scalar import BodoScalar
 
         plan = pn.Reduce(self._as_projection_plan(), (("r", "v", func),))
-        df = api.collect(plan)
-        return df["r"].iloc[0] if len(df) else None
+        return BodoScalar(plan, "r")
 
     def sum(self):
         return self._reduce("sum")
@@ -306,10 +306,13 @@ class BodoSeries:
         return self.quantile(0.5)
 
     def describe(self):
+        from .scalar import _unwrap
+
         cnt = self.count()
         qs = self.quantile([0.25, 0.5, 0.75])
-        vals = [cnt, self.mean(), self.std(), self.min(),
-                qs.iloc[0], qs.iloc[1], qs.iloc[2], self.max()]
+        vals = [_unwrap(cnt), _unwrap(self.mean()), _unwrap(self.std()),
+                _unwrap(self.min()),
+                qs.iloc[0], qs.iloc[1], qs.iloc[2], _unwrap(self.max())]
         return pd.Series(vals, index=["count", "mean", "std", "min", "25%",
                                       "50%", "75%", "max"], name=self.name)
 

