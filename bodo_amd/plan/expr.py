@@ -291,3 +291,38 @@ def substitute_cols(e: Expr, mapping) -> Expr:
     if not ch:
         return e
     return e.with_children(*[substitute_cols(c, mapping) for c in ch])
+
+
+def _install_cached_hash_expr():
+    """Same per-instance hash memoization as plan nodes (expressions are
+    hashed inside every containing plan-node hash)."""
+    import sys
+
+    mod = sys.modules[__name__]
+    for name in dir(mod):
+        cls = getattr(mod, name)
+        if isinstance(cls, type) and issubclass(cls, Expr) \
+                and cls is not Expr and cls.__hash__ is not None \
+                and "__dataclass_fields__" in cls.__dict__:
+            orig = cls.__hash__
+
+            def make(orig):
+                def __hash__(self):
+                    v = self.__dict__.get("_hc")
+                    if v is None:
+                        v = orig(self)
+                        object.__setattr__(self, "_hc", v)
+                    return v
+                return __hash__
+
+            cls.__hash__ = make(orig)
+
+            def __getstate__(self):
+                d = dict(self.__dict__)
+                d.pop("_hc", None)  # process-local (salted string hashing)
+                return d
+
+            cls.__getstate__ = __getstate__
+
+
+_install_cached_hash_expr()

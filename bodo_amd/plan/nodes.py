@@ -489,3 +489,40 @@ def explain(node: PlanNode, indent: int = 0) -> str:
     for c in node.children():
         lines.append(explain(c, indent + 1))
     return "\n".join(lines)
+
+
+def _install_cached_hash():
+    """Memoize dataclass __hash__ per instance: the executor's CTE memo and
+    optimizer rules hash plan subtrees at every visit, which is O(plan^2)
+    per query with the generated recursive hash (round-2 host-overhead
+    finding: _exec tottime was dominated by hashing on q5/q22)."""
+    import sys
+
+    mod = sys.modules[__name__]
+    for name in dir(mod):
+        cls = getattr(mod, name)
+        if isinstance(cls, type) and issubclass(cls, PlanNode) \
+                and cls is not PlanNode and cls.__hash__ is not None \
+                and "__dataclass_fields__" in cls.__dict__:
+            orig = cls.__hash__
+
+            def make(orig):
+                def __hash__(self):
+                    v = self.__dict__.get("_hc")
+                    if v is None:
+                        v = orig(self)
+                        object.__setattr__(self, "_hc", v)
+                    return v
+                return __hash__
+
+            cls.__hash__ = make(orig)
+
+            def __getstate__(self):
+                d = dict(self.__dict__)
+                d.pop("_hc", None)  # process-local (salted string hashing)
+                return d
+
+            cls.__getstate__ = __getstate__
+
+
+_install_cached_hash()

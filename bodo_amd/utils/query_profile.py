@@ -45,7 +45,7 @@ class OpTimer:
 
     def __enter__(self):
         self.t0 = time.perf_counter()
-        self.mem0 = _device_allocated()
+        self.mem0 = _device_allocated() if (_FORCED[0] or _DIR) else 0
         return self
 
     def __exit__(self, *exc):
