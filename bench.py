@@ -154,6 +154,13 @@ def main():
     trips = make_trips_shard(n_local, rank, device)
     weather = make_weather()
 
+    def log(msg):
+        if rank == 0:
+            print(f"[bench +{time.perf_counter() - T0:.1f}s] {msg}",
+                  file=sys.stderr, flush=True)
+
+    T0 = time.perf_counter()
+    log(f"generated {n_local} rows on {device}")
     pq_path = None
     if args.parquet:
         # write this rank's shard as snappy parquet (untimed; the realistic
@@ -179,6 +186,7 @@ def main():
             writer.write_table(at, row_group_size=1 << 23)
             del at
         writer.close()
+        log(f"parquet written: {os.path.getsize(pq_path)} bytes")
 
     def read_trips():
         if pq_path is None:
@@ -202,6 +210,7 @@ def main():
 
     for _ in range(args.warmup):
         one_step()
+        log("warmup step done")
     comm.barrier()
     if on_gpu:
         torch.cuda.synchronize()
@@ -209,6 +218,7 @@ def main():
     out_rows = 0
     for _ in range(args.steps):
         out_rows = one_step()
+        log("timed step done")
     if on_gpu:
         torch.cuda.synchronize()
     comm.barrier()
