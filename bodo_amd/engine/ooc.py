@@ -60,15 +60,19 @@ class _Staged:
 
 
 def budget_bytes(device) -> Optional[int]:
-    """Per-operator device working-set budget; None = unlimited."""
+    """Per-operator device working-set budget; None = unlimited.  The free
+    allowance divides across concurrently-live operators (comptroller,
+    reference: _memory_budget.h OperatorComptroller)."""
+    from . import comptroller
+
     v = os.environ.get(_ENV, "")
     if v:
-        return int(v)
+        return comptroller.share(int(v))
     if getattr(cfg, "OOC_BYTES", None):
-        return int(cfg.OOC_BYTES)
+        return comptroller.share(int(cfg.OOC_BYTES))
     if device.type == "cuda":
         free, _total = torch.cuda.mem_get_info()
-        return int(free * 0.5)
+        return comptroller.share(int(free * 0.5))
     return None
 
 
@@ -95,7 +99,14 @@ def groupby_local(child: Table, keys, aggs, dropna) -> Table:
     """rel.groupby_local with partition-splitting when over budget."""
     from ..ops import concat_tables
     from ..ops import relational as rel
+    from . import comptroller
 
+    with comptroller.operator():
+        return _groupby_local_inner(child, keys, aggs, dropna,
+                                    concat_tables, rel)
+
+
+def _groupby_local_inner(child, keys, aggs, dropna, concat_tables, rel):
     budget = budget_bytes(child.device)
     if not keys or budget is None or child.nbytes() <= budget:
         return rel.groupby_local(child, keys, aggs, dropna)
@@ -117,7 +128,15 @@ def join_local(left: Table, right: Table, left_on, right_on, how,
     decompose row-exactly)."""
     from ..ops import concat_tables
     from ..ops import relational as rel
+    from . import comptroller
 
+    with comptroller.operator():
+        return _join_local_inner(left, right, left_on, right_on, how,
+                                 suffixes, concat_tables, rel)
+
+
+def _join_local_inner(left, right, left_on, right_on, how, suffixes,
+                      concat_tables, rel):
     budget = budget_bytes(left.device)
     total = left.nbytes() + right.nbytes()
     if not left_on or budget is None or total <= budget or how == "cross":

@@ -118,3 +118,19 @@ def test_ooc_disk_spill(tiny_budget, tmp_path, monkeypatch):
     srt["c"] = srt["c"].astype(str)
     exp2 = df.sort_values(["k", "v"]).reset_index(drop=True)
     pd.testing.assert_frame_equal(srt, exp2, check_dtype=False)
+
+
+def test_comptroller_divides_budget(monkeypatch):
+    """Two live operators each see half the budget (reference:
+    _memory_budget.h OperatorComptroller)."""
+    from bodo_amd.engine import comptroller, ooc
+    import torch
+
+    monkeypatch.setenv("BODO_AMD_OOC_BYTES", "1000000")
+    dev = torch.device("cpu")
+    assert ooc.budget_bytes(dev) == 1000000
+    with comptroller.operator():
+        assert ooc.budget_bytes(dev) == 1000000  # self counts as the 1 live
+        with comptroller.operator():
+            assert ooc.budget_bytes(dev) == 500000
+    assert comptroller.peak() >= 2
