@@ -26,11 +26,15 @@ def schema_names(path: str) -> Tuple[str, ...]:
 
 
 def _dataset(path: str) -> pads.Dataset:
+    from . import fs as bfs
     from . import iceberg
 
     files = iceberg.resolve_scan_path(path)
     if files is not None:
         return pads.dataset(files, format="parquet")
+    pafs, p = bfs.resolve(path)
+    if pafs is not None:
+        return pads.dataset(p, format="parquet", filesystem=pafs)
     return pads.dataset(path, format="parquet")
 
 
@@ -71,8 +75,11 @@ def read_shard(path: str, columns: Optional[Sequence[str]],
                filters: Sequence[Expr], ctx) -> Table:
     from . import iceberg
 
+    from . import fs as bfs
+
     if ctx.device.type == "cuda" and not filters and \
-            iceberg.resolve_scan_path(path) is None:
+            iceberg.resolve_scan_path(path) is None and \
+            not bfs.is_remote(path):
         # on-GPU decode fast path (uncompressed PLAIN / RLE_DICTIONARY)
         from . import parquet_gpu
 
@@ -158,6 +165,19 @@ def write_shard(tbl: Table, path: str, compression, ctx,
             partitioning=list(partition_cols),
             basename_template=f"part-{ctx.rank:05d}-{{i}}.parquet",
             existing_data_behavior="overwrite_or_ignore")
+        return
+    from . import fs as bfs
+
+    pafs, p = bfs.resolve(path)
+    if pafs is not None:
+        if ctx.world == 1 and p.endswith(".parquet"):
+            with pafs.open_output_stream(p) as f:
+                pq.write_table(at, f, compression=compression)
+            return
+        pafs.create_dir(p, recursive=True)
+        with pafs.open_output_stream(
+                p.rstrip("/") + f"/part-{ctx.rank:05d}.parquet") as f:
+            pq.write_table(at, f, compression=compression)
         return
     if ctx.world == 1 and path.endswith(".parquet") and not os.path.isdir(path):
         pq.write_table(at, path, compression=compression)
