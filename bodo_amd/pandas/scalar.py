@@ -54,6 +54,11 @@ class BodoScalar:
     def __hash__(self):
         return hash(self.value)
 
+    def __array__(self, dtype=None, copy=None):
+        import numpy as np
+
+        return np.asarray(self.value, dtype=dtype)
+
     def __round__(self, n=None):
         return round(self.value, n) if n is not None else round(self.value)
 
