@@ -74,9 +74,14 @@ class BodoDataFrame:
         if remote is not None:
             from ..parallel import spawn
 
-            return spawn.get_spawner().gather(remote.res_id).to_pandas()
-        full = comm.allgather_table(self._result)
-        return full.to_pandas()
+            out = spawn.get_spawner().gather(remote.res_id).to_pandas()
+        else:
+            full = comm.allgather_table(self._result)
+            out = full.to_pandas()
+        idx = list(getattr(self, "_index_cols", []) or [])
+        if idx:
+            out = out.set_index(idx)
+        return out
 
     # aliases used by tests / fallback
     collect = to_pandas
@@ -573,14 +578,55 @@ class BodoDataFrame:
             return self[key] if key in self._columns else default
         return self[key]
 
+    def set_index(self, keys, drop=True, append=False) -> "BodoDataFrame":
+        """Mark columns as the frame's (Multi)Index (reference:
+        frame.py set_index).  The index columns stay physical in the
+        distributed table; collection re-applies them as the pandas index.
+        Lazy ops on the frame see them as regular columns."""
+        cols = [keys] if isinstance(keys, str) else list(keys)
+        missing = [c for c in cols if c not in self._columns]
+        if missing:
+            raise KeyError(missing)
+        out = BodoDataFrame(self._plan, self._columns)
+        prev = list(getattr(self, "_index_cols", []) or []) if append else []
+        object.__setattr__(out, "_index_cols", prev + cols)
+        object.__setattr__(out, "_result", self._result)
+        return out
+
     def reset_index(self, drop=False, **kwargs):
-        """Frames are positionally indexed; drop=True is the identity."""
+        """Positional index restored; a set_index marker either rejoins the
+        columns (drop=False) or is projected away (drop=True)."""
+        idx = list(getattr(self, "_index_cols", []) or [])
+        if not idx:
+            if drop:
+                return self
+            return self._fallback("reset_index", drop=drop, **kwargs)
         if drop:
-            return self
-        return self._fallback("reset_index", drop=drop, **kwargs)
+            keep = [c for c in self._columns if c not in idx]
+            exprs = tuple(ColRef(c) for c in keep)
+            return BodoDataFrame(pn.Projection(self._plan, tuple(keep),
+                                               exprs), keep)
+        order = idx + [c for c in self._columns if c not in idx]
+        exprs = tuple(ColRef(c) for c in order)
+        return BodoDataFrame(pn.Projection(self._plan, tuple(order), exprs),
+                             order)
+
+    def sort_index(self, ascending=True, **kwargs):
+        idx = list(getattr(self, "_index_cols", []) or [])
+        if idx:
+            out = self.sort_values(idx, ascending=ascending)
+            object.__setattr__(out, "_index_cols", idx)
+            return out
+        return self
 
     @property
     def index(self):
+        idx = list(getattr(self, "_index_cols", []) or [])
+        if idx:
+            pdf = self[idx].to_pandas()
+            if len(idx) == 1:
+                return pd.Index(pdf[idx[0]], name=idx[0])
+            return pd.MultiIndex.from_frame(pdf[idx])
         return pd.RangeIndex(len(self))
 
     def explain(self, optimized: bool = True) -> str:

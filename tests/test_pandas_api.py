@@ -456,3 +456,28 @@ def test_cross_frame_setitem():
     a["z"] = b.y
     out = a.to_pandas()
     assert out.z.tolist() == [10.0, 20.0, 30.0]
+
+
+def test_set_index_reset_index_multiindex():
+    rng = np.random.default_rng(17)
+    df = pd.DataFrame({"a": rng.integers(0, 5, 50),
+                       "b": rng.integers(0, 3, 50),
+                       "v": rng.random(50)})
+    b = bpd.from_pandas(df)
+    one = b.set_index("a")
+    got = one.to_pandas()
+    assert got.index.name == "a"
+    assert list(got.columns) == ["b", "v"] or "a" in got.columns
+    # MultiIndex
+    two = b.set_index(["a", "b"])
+    got2 = two.to_pandas()
+    assert list(got2.index.names) == ["a", "b"]
+    assert list(got2.columns) == ["v"]
+    # reset back
+    r = two.reset_index().to_pandas()
+    assert list(r.columns)[:2] == ["a", "b"]
+    rd = two.reset_index(drop=True).to_pandas()
+    assert list(rd.columns) == ["v"]
+    # sort_index
+    si = two.sort_index().to_pandas()
+    assert list(si.index.get_level_values(0)) == sorted(df["a"].tolist())
