@@ -1,0 +1,337 @@
+"""Block-distributed 1-D arrays for @bodo_amd.jit numpy code.
+
+Reference role: the Numba parfor + distributed-pass path
+(bodo/transforms/distributed_pass.py:141) that rewrites array loops into
+block-range loops with MPI collectives.  MI355X redesign: the rank shard is
+a torch tensor (HBM-resident on GPU), elementwise numpy ufuncs dispatch to
+torch kernels through ``__array_ufunc__``, whole-array functions
+(sum/mean/dot/...) run locally then combine over RCCL — no IR rewriting, no
+CPU loop lowering; the numpy protocols are the compiler surface.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from ..parallel import comm
+
+# numpy ufunc name -> torch op (elementwise)
+_UFUNC_TORCH = {
+    "add": torch.add, "subtract": torch.sub, "multiply": torch.mul,
+    "true_divide": torch.div, "divide": torch.div,
+    "floor_divide": lambda a, b: torch.div(a, b, rounding_mode="floor"),
+    "power": torch.pow, "remainder": torch.remainder,
+    "negative": torch.neg, "absolute": torch.abs, "abs": torch.abs,
+    "sqrt": torch.sqrt, "exp": torch.exp, "log": torch.log,
+    "log2": torch.log2, "log10": torch.log10, "log1p": torch.log1p,
+    "expm1": torch.expm1, "sin": torch.sin, "cos": torch.cos,
+    "tan": torch.tan, "arcsin": torch.asin, "arccos": torch.acos,
+    "arctan": torch.atan, "sinh": torch.sinh, "cosh": torch.cosh,
+    "tanh": torch.tanh, "floor": torch.floor, "ceil": torch.ceil,
+    "rint": torch.round, "sign": torch.sign, "square": torch.square,
+    "greater": torch.gt, "greater_equal": torch.ge, "less": torch.lt,
+    "less_equal": torch.le, "equal": torch.eq, "not_equal": torch.ne,
+    "maximum": torch.maximum, "minimum": torch.minimum,
+    "logical_and": torch.logical_and, "logical_or": torch.logical_or,
+    "logical_not": torch.logical_not, "isnan": torch.isnan,
+    "isfinite": torch.isfinite, "isinf": torch.isinf,
+}
+
+
+def _block_bounds(n: int, w: int, r: int):
+    base, rem = divmod(n, w)
+    start = r * base + min(r, rem)
+    return start, start + base + (1 if r < rem else 0)
+
+
+def _device():
+    from .. import config
+
+    return torch.device(config.default_device())
+
+
+class DistArray:
+    """1-D block-distributed array; ``t`` is this rank's shard."""
+
+    __slots__ = ("t", "total")
+
+    def __init__(self, t: torch.Tensor, total: int):
+        self.t = t
+        self.total = total
+
+    # -------------------------------------------------------------- alloc
+    @staticmethod
+    def from_numpy(arr: np.ndarray) -> "DistArray":
+        n = len(arr)
+        s, e = _block_bounds(n, comm.get_world_size(), comm.get_rank())
+        local = np.ascontiguousarray(arr[s:e])
+        return DistArray(torch.from_numpy(local).to(_device()), n)
+
+    @staticmethod
+    def arange(n, dtype=None) -> "DistArray":
+        s, e = _block_bounds(int(n), comm.get_world_size(), comm.get_rank())
+        t = torch.arange(s, e, device=_device())
+        if dtype is not None:
+            t = t.to(_np_to_torch(dtype))
+        return DistArray(t, int(n))
+
+    @staticmethod
+    def full(n, fill, dtype=None) -> "DistArray":
+        s, e = _block_bounds(int(n), comm.get_world_size(), comm.get_rank())
+        tdt = _np_to_torch(dtype) if dtype is not None else None
+        t = torch.full((e - s,), fill, dtype=tdt, device=_device())
+        return DistArray(t, int(n))
+
+    _rand_seq = [0]
+
+    @staticmethod
+    def random(n, seed=None) -> "DistArray":
+        s, e = _block_bounds(int(n), comm.get_world_size(), comm.get_rank())
+        g = torch.Generator(device="cpu")
+        DistArray._rand_seq[0] += 1
+        base = seed if seed is not None \
+            else 0xB0D0 + 0x85EBCA6B * DistArray._rand_seq[0]
+        g.manual_seed(base + 0x9E3779B9 * comm.get_rank())
+        t = torch.rand(e - s, generator=g, dtype=torch.float64).to(_device())
+        return DistArray(t, int(n))
+
+    # -------------------------------------------------------------- basics
+    def __len__(self):
+        return self.total
+
+    @property
+    def shape(self):
+        return (self.total,)
+
+    @property
+    def size(self):
+        return self.total
+
+    @property
+    def dtype(self):
+        return np.dtype(str(self.t.dtype).replace("torch.", "")
+                        .replace("bool", "bool_")) \
+            if str(self.t.dtype) == "torch.bool" \
+            else np.dtype(str(self.t.dtype).replace("torch.", ""))
+
+    def to_numpy(self) -> np.ndarray:
+        from ..parallel import comm as c
+
+        parts = c.allgather_obj(self.t.cpu().numpy())
+        return np.concatenate(parts)
+
+    def __repr__(self):  # pragma: no cover - debug aid
+        return (f"DistArray(n={self.total}, shard={tuple(self.t.shape)}, "
+                f"device={self.t.device})")
+
+    # ------------------------------------------------------------ numpy API
+    def __array_ufunc__(self, ufunc, method, *inputs, **kwargs):
+        if method != "__call__" or kwargs.get("out") is not None:
+            return NotImplemented
+        f = _UFUNC_TORCH.get(ufunc.__name__)
+        ts = []
+        for x in inputs:
+            if isinstance(x, DistArray):
+                ts.append(x.t)
+            elif isinstance(x, (int, float, bool, np.integer, np.floating)):
+                ts.append(x)
+            else:
+                return NotImplemented
+        if f is None:
+            # shard-local numpy fallback (stays SPMD-correct: elementwise)
+            args_np = [x.cpu().numpy() if torch.is_tensor(x) else x
+                       for x in ts]
+            res = getattr(np, ufunc.__name__)(*args_np)
+            return DistArray(torch.from_numpy(np.ascontiguousarray(res))
+                             .to(self.t.device), self.total)
+        out = f(*[t if torch.is_tensor(t) or isinstance(t, (int, float, bool))
+                  else t for t in ts])
+        if not torch.is_tensor(out):
+            out = torch.as_tensor(out)
+        return DistArray(out, self.total)
+
+    def __array_function__(self, func, types, args, kwargs):
+        name = getattr(func, "__name__", "")
+        h = _ARRAY_FUNCS.get(name)
+        if h is None:
+            return NotImplemented
+        return h(*args, **kwargs)
+
+    # ------------------------------------------------------------ operators
+    def _bin(self, other, tf):
+        if isinstance(other, DistArray):
+            other = other.t
+        return DistArray(tf(self.t, other), self.total)
+
+    def __add__(self, o):
+        return self._bin(o, torch.add)
+
+    def __radd__(self, o):
+        return self._bin(o, lambda a, b: torch.add(b, a)
+                         if torch.is_tensor(b) else b + a)
+
+    def __sub__(self, o):
+        return self._bin(o, torch.sub)
+
+    def __rsub__(self, o):
+        return DistArray(o - self.t, self.total)
+
+    def __mul__(self, o):
+        return self._bin(o, torch.mul)
+
+    def __rmul__(self, o):
+        return self._bin(o, lambda a, b: a * b)
+
+    def __truediv__(self, o):
+        return self._bin(o, torch.div)
+
+    def __rtruediv__(self, o):
+        return DistArray(o / self.t, self.total)
+
+    def __pow__(self, o):
+        return self._bin(o, torch.pow)
+
+    def __mod__(self, o):
+        return self._bin(o, torch.remainder)
+
+    def __neg__(self):
+        return DistArray(-self.t, self.total)
+
+    def __abs__(self):
+        return DistArray(self.t.abs(), self.total)
+
+    def __lt__(self, o):
+        return self._bin(o, torch.lt)
+
+    def __le__(self, o):
+        return self._bin(o, torch.le)
+
+    def __gt__(self, o):
+        return self._bin(o, torch.gt)
+
+    def __ge__(self, o):
+        return self._bin(o, torch.ge)
+
+    def __eq__(self, o):
+        return self._bin(o, torch.eq)
+
+    def __ne__(self, o):
+        return self._bin(o, torch.ne)
+
+    def __and__(self, o):
+        return self._bin(o, torch.logical_and)
+
+    def __or__(self, o):
+        return self._bin(o, torch.logical_or)
+
+    def __invert__(self):
+        return DistArray(~self.t, self.total)
+
+    __hash__ = None
+
+    # ------------------------------------------------------------ reductions
+    def _reduce(self, kind):
+        t = self.t
+        if kind == "sum":
+            v = t.sum()
+        elif kind == "prod":
+            v = t.prod()
+        elif kind == "min":
+            v = t.min() if t.numel() else torch.tensor(float("inf"))
+        elif kind == "max":
+            v = t.max() if t.numel() else torch.tensor(float("-inf"))
+        else:
+            raise NotImplementedError(kind)
+        val = v.item()
+        if comm.get_world_size() > 1:
+            parts = comm.allgather_obj(val)
+            if kind == "sum":
+                val = sum(parts)
+            elif kind == "prod":
+                out = 1
+                for p in parts:
+                    out *= p
+                val = out
+            elif kind == "min":
+                val = min(parts)
+            else:
+                val = max(parts)
+        return val
+
+    def sum(self):
+        return self._reduce("sum")
+
+    def prod(self):
+        return self._reduce("prod")
+
+    def min(self):
+        return self._reduce("min")
+
+    def max(self):
+        return self._reduce("max")
+
+    def mean(self):
+        return self._reduce("sum") / max(self.total, 1)
+
+    def std(self):
+        m = self.mean()
+        var = ((self - m) ** 2)._reduce("sum") / max(self.total - 1, 1)
+        return var ** 0.5
+
+    def astype(self, dtype):
+        return DistArray(self.t.to(_np_to_torch(dtype)), self.total)
+
+    def copy(self):
+        return DistArray(self.t.clone(), self.total)
+
+    def __getitem__(self, i):
+        if isinstance(i, DistArray) and i.t.dtype == torch.bool:
+            # boolean selection -> 1D_Var result (variable block lengths,
+            # reference: Distribution.OneD_Var, distributed_analysis.py:83)
+            sel = self.t[i.t]
+            lengths = comm.allgather_obj(int(sel.numel()))
+            return DistArray(sel, int(sum(lengths)))
+        if isinstance(i, (int, np.integer)):
+            # global scalar index: owner broadcasts
+            w, r = comm.get_world_size(), comm.get_rank()
+            i = int(i) % self.total
+            s, e = _block_bounds(self.total, w, r)
+            val = self.t[i - s].item() if s <= i < e else None
+            if w > 1:
+                vals = [v for v in comm.allgather_obj(val) if v is not None]
+                val = vals[0]
+            return val
+        raise NotImplementedError(f"DistArray index {type(i)}")
+
+
+def _np_to_torch(dtype):
+    return torch.from_numpy(np.zeros(0, dtype=np.dtype(dtype))).dtype
+
+
+def _dot(a, b):
+    if isinstance(a, DistArray) and isinstance(b, DistArray):
+        local = float((a.t.double() * b.t.double()).sum().item())
+        if comm.get_world_size() > 1:
+            local = sum(comm.allgather_obj(local))
+        return local
+    return NotImplemented
+
+
+_ARRAY_FUNCS = {
+    "sum": lambda a, **k: a.sum(),
+    "prod": lambda a, **k: a.prod(),
+    "min": lambda a, **k: a.min(),
+    "max": lambda a, **k: a.max(),
+    "mean": lambda a, **k: a.mean(),
+    "std": lambda a, **k: a.std(),
+    "dot": _dot,
+    "where": lambda c, x, y: DistArray(
+        torch.where(c.t,
+                    x.t if isinstance(x, DistArray) else torch.as_tensor(
+                        x, device=c.t.device),
+                    y.t if isinstance(y, DistArray) else torch.as_tensor(
+                        y, device=c.t.device)), c.total),
+    "concatenate": lambda arrs, **k: DistArray(
+        torch.cat([a.t for a in arrs]), sum(a.total for a in arrs)),
+}

@@ -583,3 +583,31 @@ def test_dist_packed_collectives_empty_shard():
     assert got["gathered_rows"].iloc[0] == len(exp)
     got2 = got.drop(columns=["gathered_rows"]).reset_index(drop=True)
     pd.testing.assert_frame_equal(got2, exp, check_dtype=False)
+
+
+def _q_jit_numpy(bpd, rank, payload):
+    import numpy as np
+
+    import bodo_amd
+
+    @bodo_amd.jit
+    def f(a, k):
+        b = a * 2.0 + k
+        sel = b[b > 1.0]
+        return sel.sum() + np.sum(np.arange(10000))
+
+    return float(f(payload["a"], payload["k"]))
+
+
+def test_dist_jit_numpy_world2():
+    """@bodo_amd.jit numpy path at world=2: scattered args, distributed
+    creation, 1D_Var selection and global reductions must match the serial
+    result (reference: distributed_pass parfor semantics)."""
+    import numpy as np
+
+    rng = np.random.default_rng(8)
+    a = rng.random(40_000)
+    got = run_dist(_q_jit_numpy, {"a": a, "k": 0.25})
+    b = a * 2.0 + 0.25
+    exp = b[b > 1.0].sum() + np.arange(10000).sum()
+    assert abs(got - exp) < 1e-6
