@@ -203,34 +203,89 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
     has_mask = hm.cpu().tolist() if hm.numel() else []
     send_counts = [int(c) for c in counts.tolist()]
     recv_counts = [int(meta[r * nmeta]) for r in range(w)]
-    out_cols = []
+    n_out = sum(recv_counts)
+    out_cols: List[Optional[Column]] = [None] * len(packed.columns)
+    # ---- ONE packed exchange for every fixed-width column + mask ----
+    # (contiguous-split pack, reference: GpuShuffleManager
+    #  cudf::contiguous_split + per-destination Isend, gpu_utils.cpp:96-122)
+    fixed_idx = [i for i, c in enumerate(packed.columns)
+                 if c.dtype.kind != TypeKind.STRING]
+    if fixed_idx:
+        bnd_l = bnd.tolist()
+        dev0 = packed.columns[fixed_idx[0]].device if fixed_idx else None
+        views = []  # per column: (byte view [n, esize] , mask view or None)
+        row_bytes = 0
+        for i in fixed_idx:
+            c = packed.columns[i]
+            data = c.data.contiguous()
+            esize = data.element_size()
+            v = data.view(torch.uint8).view(len(c), esize) if len(c) else \
+                data.view(torch.uint8).view(0, max(esize, 1))
+            mk = None
+            if has_mask and has_mask[i]:
+                m = c.mask if c.mask is not None else torch.ones(
+                    len(c), dtype=torch.bool, device=data.device)
+                mk = m.contiguous().view(torch.uint8).view(len(c), 1)
+                row_bytes += 1
+            views.append((v, mk))
+            row_bytes += esize
+        segs = []
+        for d in range(w):
+            s0, s1 = bnd_l[d], bnd_l[d + 1]
+            for v, mk in views:
+                segs.append(v[s0:s1].reshape(-1))
+                if mk is not None:
+                    segs.append(mk[s0:s1].reshape(-1))
+        send_buf = torch.cat(segs) if segs else torch.zeros(
+            0, dtype=torch.uint8, device=dev0)
+        sbytes = [(bnd_l[d + 1] - bnd_l[d]) * row_bytes for d in range(w)]
+        rbytes = [recv_counts[d] * row_bytes for d in range(w)]
+        recv_buf = alltoallv_tensor(send_buf, sbytes, rbytes)
+        # unpack: per source rank, per column, typed views -> concat
+        col_parts = {i: [] for i in fixed_idx}
+        mask_parts = {i: [] for i in fixed_idx}
+        off = 0
+        for d in range(w):
+            rows = recv_counts[d]
+            for k, i in enumerate(fixed_idx):
+                c = packed.columns[i]
+                esize = c.data.element_size()
+                nb = rows * esize
+                col_parts[i].append(recv_buf[off:off + nb])
+                off += nb
+                if views[k][1] is not None:
+                    mask_parts[i].append(recv_buf[off:off + rows])
+                    off += rows
+        dj = 0
+        for k, i in enumerate(fixed_idx):
+            c = packed.columns[i]
+            tdt = c.data.dtype
+            data = torch.cat(col_parts[i]).view(tdt) if n_out else \
+                torch.zeros(0, dtype=tdt, device=c.device)
+            new_mask = None
+            if views[k][1] is not None:
+                new_mask = (torch.cat(mask_parts[i]).view(torch.bool)
+                            if n_out else torch.zeros(0, dtype=torch.bool,
+                                                      device=c.device))
+            if c.dtype.kind == TypeKind.DICT:
+                dicts_by_rank = [dd[dj] for dd in all_dicts]
+                dj += 1
+                out_cols[i] = _remap_dict_codes(data.contiguous(), new_mask,
+                                                dicts_by_rank, recv_counts)
+            else:
+                out_cols[i] = Column(c.dtype, data, new_mask, length=n_out)
+    # ---- strings: per-column exchanges (offsets + bytes + mask) ----
     sj = 0
-    dj = 0
     for i, col in enumerate(packed.columns):
-        if col.dtype.kind == TypeKind.STRING:
-            byte_send = [int(meta_send[1 + sj + r * nmeta]) for r in range(w)]
-            byte_recv = [int(meta[r * nmeta + 1 + sj]) for r in range(w)]
-            sj += 1
-            out_cols.append(_shuffle_string_column(
-                col, send_counts, recv_counts, byte_send, byte_recv,
-                bool(has_mask[i])))
-        elif col.dtype.kind == TypeKind.DICT:
-            data = alltoallv_tensor(col.data.contiguous(), send_counts,
-                                    recv_counts)
-            new_mask = _shuffle_mask(col, send_counts, recv_counts,
-                                     bool(has_mask[i]))
-            dicts_by_rank = [d[dj] for d in all_dicts]
-            dj += 1
-            out_cols.append(_remap_dict_codes(
-                data, new_mask, dicts_by_rank, recv_counts))
-        else:
-            data = alltoallv_tensor(col.data.contiguous(), send_counts,
-                                    recv_counts)
-            new_mask = _shuffle_mask(col, send_counts, recv_counts,
-                                     bool(has_mask[i]))
-            out_cols.append(Column(col.dtype, data, new_mask,
-                                   length=sum(recv_counts)))
-    return Table(packed.names, out_cols, sum(recv_counts))
+        if col.dtype.kind != TypeKind.STRING:
+            continue
+        byte_send = [int(meta_send[1 + sj + r * nmeta]) for r in range(w)]
+        byte_recv = [int(meta[r * nmeta + 1 + sj]) for r in range(w)]
+        sj += 1
+        out_cols[i] = _shuffle_string_column(
+            col, send_counts, recv_counts, byte_send, byte_recv,
+            bool(has_mask[i]))
+    return Table(packed.names, out_cols, n_out)
 
 
 def _shuffle_string_column(col: Column, send_counts, recv_counts,
