@@ -186,6 +186,12 @@ def main():
             writer.write_table(at, row_group_size=1 << 23)
             del at
         writer.close()
+        # the generated device table is only the write source; freeing it
+        # returns ~37 GB of HBM to the query working set at 1B rows
+        del trips
+        trips = None
+        if on_gpu:
+            torch.cuda.empty_cache()
         log(f"parquet written: {os.path.getsize(pq_path)} bytes")
 
     def read_trips():

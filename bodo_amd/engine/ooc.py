@@ -71,8 +71,12 @@ def budget_bytes(device) -> Optional[int]:
     if getattr(cfg, "OOC_BYTES", None):
         return comptroller.share(int(cfg.OOC_BYTES))
     if device.type == "cuda":
+        # available = device-free + blocks the caching allocator holds but
+        # has not handed out (mem_get_info alone under-reports after the
+        # first big query and sent steady-state runs into the spill path)
         free, _total = torch.cuda.mem_get_info()
-        return comptroller.share(int(free * 0.5))
+        cached = torch.cuda.memory_reserved() - torch.cuda.memory_allocated()
+        return comptroller.share(int((free + max(cached, 0)) * 0.5))
     return None
 
 
