@@ -167,3 +167,49 @@ def write_iceberg(tbl, path: str, mode: str, ctx) -> None:
                       str(version + 1))
     if ctx.world > 1:
         comm.barrier()
+
+
+def merge_into(path: str, source, on, when_matched: str = "update",
+               when_not_matched: str = "insert"):
+    """Copy-on-write MERGE INTO for filesystem Iceberg tables (reference:
+    bodo/io/iceberg/merge_into.py).  Upsert semantics composed from the
+    engine's distributed joins: surviving target rows = anti-join on the
+    merge keys; matched updates take the source row wholesale; unmatched
+    source rows insert.  The rewritten table commits as a new transactional
+    snapshot (mode='replace'), so readers see either the old or the new
+    snapshot, never a mix."""
+    import bodo_amd.pandas as bpd
+
+    from ..pandas.frame import BodoDataFrame
+    from ..plan import nodes as pn
+
+    keys = [on] if isinstance(on, str) else list(on)
+    if when_matched not in ("update", "delete"):
+        raise ValueError(f"when_matched={when_matched!r}")
+    if when_not_matched not in ("insert", "ignore"):
+        raise ValueError(f"when_not_matched={when_not_matched!r}")
+    tgt = bpd.read_iceberg(path)
+    cols = list(tgt.columns)
+    if not isinstance(source, BodoDataFrame):
+        source = bpd.from_pandas(source)
+    missing = [c for c in cols if c not in list(source.columns)]
+    if missing:
+        raise ValueError(f"source is missing target columns {missing}")
+    src = source[cols]
+
+    survivors = BodoDataFrame(
+        pn.Join(tgt._plan, src._plan, tuple(keys), tuple(keys), "anti"),
+        cols)
+    parts = [survivors]
+    if when_matched == "update" and when_not_matched == "insert":
+        parts.append(src)
+    elif when_matched == "update":
+        parts.append(BodoDataFrame(
+            pn.Join(src._plan, tgt._plan, tuple(keys), tuple(keys), "semi"),
+            cols))
+    elif when_not_matched == "insert":
+        parts.append(BodoDataFrame(
+            pn.Join(src._plan, tgt._plan, tuple(keys), tuple(keys), "anti"),
+            cols))
+    out = parts[0] if len(parts) == 1 else bpd.concat(parts)
+    out.to_iceberg(path, mode="replace")
