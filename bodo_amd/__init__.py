@@ -92,3 +92,42 @@ def barrier():
     from .parallel import comm
 
     comm.barrier()
+
+
+def prange(*args):
+    """Parallel range (reference: bodo.prange).  Loop bodies over
+    distributed data should be expressed as array expressions (the
+    DistArray/ufunc layer parallelizes them); prange itself iterates this
+    rank's block of the global range so explicit loops stay SPMD."""
+    import builtins
+
+    if len(args) == 1:
+        from .compiler.distarray import _block_bounds
+        from .parallel import comm
+
+        s, e = _block_bounds(int(args[0]), comm.get_world_size(),
+                             comm.get_rank())
+        return builtins.range(s, e)
+    return builtins.range(*args)
+
+
+def dist_reduce(value, op: str = "sum"):
+    """Combine a per-rank scalar across ranks (reference:
+    bodo/libs/distributed_api.py dist_reduce)."""
+    from .parallel import comm
+
+    if comm.get_world_size() == 1:
+        return value
+    parts = comm.allgather_obj(value)
+    if op == "sum":
+        return sum(parts)
+    if op == "min":
+        return min(parts)
+    if op == "max":
+        return max(parts)
+    if op == "prod":
+        out = 1
+        for p in parts:
+            out *= p
+        return out
+    raise ValueError(f"dist_reduce op {op!r}")

@@ -117,3 +117,17 @@ def test_distarray_ufunc_fallback():
     out = np.arctanh(a)  # not in the torch map: numpy shard fallback
     exp = np.arctanh(np.linspace(0.1, 0.9, 4000))
     np.testing.assert_allclose(out.to_numpy(), exp)
+
+
+def test_prange_and_dist_reduce():
+    """Explicit SPMD loop: prange iterates this rank's block; dist_reduce
+    combines (reference: get_start/get_end + dist_reduce lowering)."""
+    @bodo_amd.jit
+    def f(n):
+        acc = 0
+        for i in bodo_amd.prange(n):
+            acc += i * i
+        return bodo_amd.dist_reduce(acc, "sum")
+
+    n = 10000
+    assert f(n) == sum(i * i for i in range(n))
