@@ -617,7 +617,6 @@ def _read_pieces_pipelined(my, columns, ctx) -> Optional[Table]:
     fires.  Disk read, PCIe upload and decode kernels of different chunks
     overlap (reference role: bodo/io/parquet_reader.cpp prefetch +
     _io_cpu_thread_pool.cpp)."""
-    import queue
     import time
     from concurrent.futures import ThreadPoolExecutor
 
@@ -647,15 +646,22 @@ def _read_pieces_pipelined(my, columns, ctx) -> Optional[Table]:
     if not items:
         return None
 
+    import threading
+
     slots = [{"t": None, "np": None, "ev": None} for _ in range(_NSLOTS)]
-    slot_q: "queue.Queue[int]" = queue.Queue()
-    for i in range(_NSLOTS):
-        slot_q.put(i)
+    # DETERMINISTIC slot assignment: task k uses slot k % N, gated by a
+    # per-slot semaphore released when the consumer finishes the slot's
+    # previous occupant (task k-N).  A shared free-slot queue deadlocked:
+    # workers for tasks AHEAD of the consumer could hoard every released
+    # slot while the consumer waited on the one starved task (observed as
+    # all threads parked in slot_q.get on 600-chunk shards).
+    sems = [threading.Semaphore(1) for _ in range(_NSLOTS)]
 
     def fetch(idx):
         fp, rg, cm, field, start, size, first, names = items[idx]
         t0 = time.perf_counter()
-        i = slot_q.get()
+        i = idx % _NSLOTS
+        sems[i].acquire()
         s = slots[i]
         if s["ev"] is not None:
             s["ev"].synchronize()  # previous H2D out of this slot is done
@@ -671,12 +677,18 @@ def _read_pieces_pipelined(my, columns, ctx) -> Optional[Table]:
         return i
 
     ex = ThreadPoolExecutor(max_workers=_NSLOTS)
+    inflight: List[torch.cuda.Event] = []
     try:
         futs = [ex.submit(fetch, k) for k in range(len(items))]
         rg_tables: List[Table] = []
         cur_cols: List[Column] = []
         cur_names: List[str] = []
         for k, fut in enumerate(futs):
+            # backpressure: keep at most ~2x the slot depth of decoded
+            # chunks in flight on the GPU (unbounded enqueue ran the
+            # allocator far ahead of execution on 600+-chunk shards)
+            if len(inflight) >= 2 * _NSLOTS:
+                inflight.pop(0).synchronize()
             fp, rg, cm, field, start, size, first, names = items[k]
             if first and cur_cols:
                 ordered = [n for n in cur_names]
@@ -697,12 +709,15 @@ def _read_pieces_pipelined(my, columns, ctx) -> Optional[Table]:
             ev = torch.cuda.Event()
             ev.record()
             s["ev"] = ev
-            slot_q.put(i)
+            sems[i].release()
             reader = _ChunkReader(
                 None, cm, cm.physical_type,
                 max_def=1 if field.nullable else 0,
                 prefetched=(hdrs, dev, dict_raw, (fp, start, size)))
             col = reader.decode_column(device, field)
+            done = torch.cuda.Event()
+            done.record()
+            inflight.append(done)
             STATS["t_decode"] = STATS.get("t_decode", 0.0) + \
                 time.perf_counter() - t0
             if col is None:
