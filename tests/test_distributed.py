@@ -611,3 +611,57 @@ def test_dist_jit_numpy_world2():
     b = a * 2.0 + 0.25
     exp = b[b > 1.0].sum() + np.arange(10000).sum()
     assert abs(got - exp) < 1e-6
+
+
+def test_dist_world4_groupby_join_sort():
+    """world=4 on gloo: the packed single-collective shuffle, broadcast
+    joins and range-partitioned sort must hold beyond 2 ranks (the driver's
+    8-GPU scale run exercises this path over RCCL)."""
+    df = _df(4000, 23)
+    got = run_dist(_q_groupby, {"df": df}, world=4).reset_index(drop=True)
+    exp = df.groupby(["a", "c"], as_index=False).agg(
+        s=("b", "sum"), m=("b", "mean"), n=("b", "count"), mx=("b", "max"),
+    ).sort_values(["a", "c"]).reset_index(drop=True)
+    got["c"] = got["c"].astype(str)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+    rng = np.random.default_rng(29)
+    left = pd.DataFrame({"k": rng.integers(0, 50, 3000),
+                         "v1": rng.uniform(0, 1, 3000)})
+    right = pd.DataFrame({"k": np.arange(40), "v2": rng.uniform(0, 1, 40)})
+    got2 = run_dist(_q_join, {"left": left, "right": right},
+                    world=4).reset_index(drop=True)
+    exp2 = left.merge(right, on="k", how="inner").sort_values(
+        ["k", "v1", "v2"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(got2, exp2, check_dtype=False)
+
+    got3 = run_dist(_q_sort, {"df": df}, world=4).reset_index(drop=True)
+    exp3 = df.sort_values(["a", "b"], ascending=[True, False]).reset_index(
+        drop=True)
+    got3["c"] = got3["c"].astype(str)
+    pd.testing.assert_frame_equal(got3, exp3, check_dtype=False)
+
+
+def _q_csv_range(bpd, rank, payload):
+    return bpd.read_csv(payload["path"]).groupby("k", as_index=False).agg(
+        s=bpd.NamedAgg("v", "sum"), c=bpd.NamedAgg("s", "count")
+    ).sort_values("k")
+
+
+def test_dist_csv_byte_range_split(tmp_path):
+    """Byte-range parallel CSV: ranks read disjoint newline-aligned slices
+    (reference: _csv_json_reader.cpp byte division)."""
+    rng = np.random.default_rng(33)
+    n = 320_000
+    df = pd.DataFrame({"k": rng.integers(0, 40, n),
+                       "v": rng.random(n).round(6),
+                       "s": rng.choice(["aa", "bb", "cc"], n)})
+    p = str(tmp_path / "big.csv")
+    df.to_csv(p, index=False)
+    import os
+    assert os.path.getsize(p) >= 4 << 20, "grow n: file under range threshold"
+    got = run_dist(_q_csv_range, {"path": p}).reset_index(drop=True)
+    exp = df.groupby("k", as_index=False).agg(
+        s=("v", "sum"), c=("s", "count")).sort_values("k").reset_index(
+        drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-6)
