@@ -131,3 +131,11 @@ def dist_reduce(value, op: str = "sum"):
             out *= p
         return out
     raise ValueError(f"dist_reduce op {op!r}")
+
+
+def __getattr__(name):
+    if name == "fft":
+        from .utils import fft as _fft
+
+        return _fft
+    raise AttributeError(name)

@@ -1454,6 +1454,36 @@ class Planner:
                                 ex.BinOp("floordiv",
                                          ex.BinOp("sub", d, ex.Const(1)),
                                          ex.Const(7)), ex.Const(1))
+            if name in ("json_extract_path_text", "get_path"):
+                import json as _json
+
+                path_expr = e.args[1].value
+
+                def _jx(v, _p=path_expr):
+                    try:
+                        cur = _json.loads(v)
+                    except Exception:
+                        return None
+                    for part in str(_p).replace("[", ".").replace(
+                            "]", "").split("."):
+                        if part == "":
+                            continue
+                        if isinstance(cur, list):
+                            try:
+                                cur = cur[int(part)]
+                            except Exception:
+                                return None
+                        elif isinstance(cur, dict):
+                            cur = cur.get(part)
+                        else:
+                            return None
+                        if cur is None:
+                            return None
+                    if isinstance(cur, (dict, list)):
+                        return _json.dumps(cur)
+                    return str(cur)
+
+                return ex.UdfMap(self.expr(e.args[0], scope), _jx, "ignore")
             if name in ("dayname", "monthname"):
                 if name == "dayname":
                     fld, names = "dayofweek", ["Mon", "Tue", "Wed", "Thu",

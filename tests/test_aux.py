@@ -244,3 +244,40 @@ def test_fs_uri_routing(tmp_path):
     bpd.from_pandas(df).to_parquet(out_uri)
     back = pd.read_parquet(str(tmp_path / "out.parquet"))
     assert len(back) == len(df)
+
+
+def test_fft_and_json_extract():
+    import numpy as np
+
+    import bodo_amd
+    from bodo_amd.sql import BodoSQLContext
+
+    x = np.random.default_rng(2).random(4096)
+    got = bodo_amd.fft.fft(x)
+    np.testing.assert_allclose(got, np.fft.fft(x), rtol=1e-8, atol=1e-8)
+    back = bodo_amd.fft.ifft(got)
+    np.testing.assert_allclose(back.real, x, atol=1e-9)
+
+    import pandas as pd
+
+    df = pd.DataFrame({"j": ['{"a": {"b": 5}, "c": [1, 2]}',
+                             '{"a": {"b": "x"}}', "not json", None]})
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql("select json_extract_path_text(j, 'a.b') as v, "
+                 "json_extract_path_text(j, 'c[1]') as e from t").to_pandas()
+    assert out["v"].tolist()[:2] == ["5", "x"]
+    assert out["e"].iloc[0] == "2"
+    assert pd.isna(out["v"].iloc[2]) and pd.isna(out["v"].iloc[3])
+
+
+def test_fft_distarray():
+    import numpy as np
+
+    import bodo_amd
+    from bodo_amd.compiler.distarray import DistArray
+
+    x = np.random.default_rng(3).random(8192)
+    d = DistArray.from_numpy(x)
+    got = bodo_amd.fft.rfft(d)
+    np.testing.assert_allclose(got.to_numpy(), np.fft.rfft(x), rtol=1e-8,
+                               atol=1e-8)
