@@ -365,3 +365,7 @@ class KMeans:
     def fit_predict(self, X) -> np.ndarray:
         self.fit(X)
         return self.predict(X)
+
+
+from .gbdt import (GradientBoostingClassifier,  # noqa: E402,F401
+                   GradientBoostingRegressor)

@@ -665,3 +665,39 @@ def test_dist_csv_byte_range_split(tmp_path):
         s=("v", "sum"), c=("s", "count")).sort_values("k").reset_index(
         drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-6)
+
+
+def _q_gbdt(bpd, rank, payload):
+    import numpy as np
+
+    from bodo_amd.ml.gbdt import GradientBoostingRegressor
+
+    X, y = payload["X"], payload["y"]
+    half = len(X) // 2
+    Xs = X[rank * half:(rank + 1) * half]
+    ys = y[rank * half:(rank + 1) * half]
+    m = GradientBoostingRegressor(n_estimators=25, max_depth=4)
+    m.fit(Xs, ys)  # histograms all-reduce across the 2 ranks
+    return float(m.score(payload["Xt"], payload["yt"]))
+
+
+def test_dist_gbdt_histogram_allreduce():
+    """2-rank GBDT must match the single-process fit (histogram sums are
+    the only cross-rank state; reference: xgboost rabit AllReduce)."""
+    import numpy as np
+
+    from bodo_amd.ml.gbdt import GradientBoostingRegressor
+
+    rng = np.random.default_rng(11)
+    n = 12000
+    X = rng.random((n, 5)).astype(np.float32)
+    y = (2 * X[:, 0] - X[:, 1] ** 2
+         + 0.05 * rng.standard_normal(n)).astype(np.float32)
+    Xt = rng.random((2000, 5)).astype(np.float32)
+    yt = (2 * Xt[:, 0] - Xt[:, 1] ** 2).astype(np.float32)
+    r2_dist = run_dist(_q_gbdt, {"X": X, "y": y, "Xt": Xt, "yt": yt})
+    m = GradientBoostingRegressor(n_estimators=25, max_depth=4)
+    m.fit(X[:n // 2 * 2], y[:n // 2 * 2])
+    r2_single = m.score(Xt, yt)
+    assert abs(r2_dist - r2_single) < 0.02, (r2_dist, r2_single)
+    assert r2_dist > 0.9
