@@ -179,6 +179,16 @@ class LogisticRegression:
     def predict(self, X) -> np.ndarray:
         return (self.predict_proba(X) >= 0.5).astype(np.int64)
 
+    def score(self, X, y) -> float:
+        """Accuracy, combined across ranks (sklearn surface)."""
+        hit = float((self.predict(X) == np.asarray(y)).sum())
+        n = float(len(np.asarray(y)))
+        if comm.get_world_size() > 1:
+            parts = comm.allgather_obj((hit, n))
+            hit = sum(p[0] for p in parts)
+            n = sum(p[1] for p in parts)
+        return hit / max(n, 1.0)
+
 
 def train_test_split(X, y=None, test_size: float = 0.25, random_state=None):
     """Shard-local split (rows already distributed; reference:
