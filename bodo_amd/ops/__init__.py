@@ -65,6 +65,12 @@ def _gather_string_arrow(col: Column, idx: torch.Tensor) -> Column:
 
 
 def take_table(tbl: Table, idx: torch.Tensor) -> Table:
+    if tbl.device.type == "cuda" and len(tbl.columns) > 1:
+        from . import gpu
+
+        fused = gpu.take_table_fused(tbl, idx)
+        if fused is not None:
+            return fused
     return Table(tbl.names, [gather(c, idx) for c in tbl.columns], int(idx.shape[0]))
 
 

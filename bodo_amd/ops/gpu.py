@@ -719,3 +719,45 @@ def _empty_gb_result(tbl: Table, keys, aggs) -> Table:
                                            device=tbl.device), length=0))
         names.append(out_name)
     return T(names, cols, 0)
+
+
+def take_table_fused(tbl: Table, idx: torch.Tensor) -> Optional[Table]:
+    """All fixed-width columns (+masks) of a take materialize in ONE kernel
+    launch (join-heavy queries were launch-bound on per-column
+    index_select; reference role: cudf::gather single pass)."""
+    from . import gather as _gather
+
+    K = kernels()
+    srcs = []
+    plan = []  # (col_index, has_mask)
+    for i, c in enumerate(tbl.columns):
+        if c.dtype.kind == TypeKind.STRING:
+            plan.append((i, None))
+            continue
+        srcs.append(c.data)
+        if c.mask is not None:
+            srcs.append(c.mask)
+            plan.append((i, True))
+        else:
+            plan.append((i, False))
+    if len(srcs) < 2:
+        return None
+    outs = K.gather_multi(srcs, idx)
+    n = int(idx.numel())
+    cols = []
+    k = 0
+    for i, has_mask in plan:
+        c = tbl.columns[i]
+        if has_mask is None:
+            cols.append(_gather(c, idx))
+            continue
+        data = outs[k]
+        k += 1
+        mask = None
+        if has_mask:
+            mask = outs[k]
+            k += 1
+        out = Column(c.dtype, data, mask, dictionary=c.dictionary, length=n)
+        out.val_range = c.val_range
+        cols.append(out)
+    return Table(tbl.names, cols, n)

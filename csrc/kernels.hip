@@ -1237,6 +1237,8 @@ torch::Tensor pq_copy_strings(torch::Tensor scratch, torch::Tensor src_abs,
                               torch::Tensor dst_off, torch::Tensor lengths,
                               int64_t n, int64_t total_bytes);
 torch::Tensor pq_parse_headers(torch::Tensor chunk_buf);
+std::vector<torch::Tensor> gather_multi(std::vector<torch::Tensor> srcs,
+                                        torch::Tensor idx);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_f32", &gemm_f32, "f32 MFMA GEMM (v_mfma_f32_16x16x4_f32)");
@@ -1265,5 +1267,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "gather string bytes to packed buffer");
   m.def("pq_parse_headers", &pq_parse_headers,
         "host-side thrift page-header parse of a chunk buffer");
+  m.def("gather_multi", &gather_multi,
+        "fused multi-column fixed-width gather (one launch per table)");
   m.attr("_native") = true;
 }

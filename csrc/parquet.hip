@@ -668,3 +668,67 @@ torch::Tensor pq_parse_headers(torch::Tensor chunk_buf) {
   std::memcpy(out.data_ptr(), rows.data(), rows.size() * sizeof(int64_t));
   return out;
 }
+
+// ---------------------------------------------------------------------
+// fused multi-column gather: one launch materializes every fixed-width
+// column (and mask) of a take_table/join output instead of one torch
+// index_select per column (join-heavy queries were launch-bound;
+// reference role: the single-pass materialization in cudf::gather).
+// Layout: consecutive threads walk consecutive output rows of one column
+// so loads from idx broadcast and stores coalesce.
+// ---------------------------------------------------------------------
+
+struct GatherCol {
+  const uint8_t* src;
+  uint8_t* dst;
+  int64_t esize;
+};
+
+__global__ void gather_multi_kernel(const GatherCol* __restrict__ cols,
+                                    int ncols, const int64_t* __restrict__ idx,
+                                    int64_t n) {
+  int64_t total = (int64_t)ncols * n;
+  GRID_STRIDE_LOOP(t, total) {
+    int c = (int)(t / n);
+    int64_t i = t - (int64_t)c * n;
+    int64_t j = idx[i];
+    const GatherCol g = cols[c];
+    switch (g.esize) {
+      case 1: g.dst[i] = g.src[j]; break;
+      case 2: ((uint16_t*)g.dst)[i] = ((const uint16_t*)g.src)[j]; break;
+      case 4: ((uint32_t*)g.dst)[i] = ((const uint32_t*)g.src)[j]; break;
+      default: ((uint64_t*)g.dst)[i] = ((const uint64_t*)g.src)[j]; break;
+    }
+  }
+}
+
+std::vector<torch::Tensor> gather_multi(std::vector<torch::Tensor> srcs,
+                                        torch::Tensor idx) {
+  int64_t n = idx.numel();
+  auto dev = idx.device();
+  std::vector<torch::Tensor> outs;
+  std::vector<GatherCol> cols;
+  for (auto& s : srcs) {
+    auto o = torch::empty({n}, s.options());
+    outs.push_back(o);
+    GatherCol g;
+    g.src = (const uint8_t*)s.data_ptr();
+    g.dst = (uint8_t*)o.data_ptr();
+    g.esize = s.element_size();
+    cols.push_back(g);
+  }
+  if (n && !srcs.empty()) {
+    auto cpu = torch::from_blob((void*)cols.data(),
+                                {(int64_t)(cols.size() * sizeof(GatherCol))},
+                                torch::kUInt8);
+    auto cols_dev = cpu.to(dev, /*non_blocking=*/false);
+    int block = 256;
+    int64_t total = (int64_t)srcs.size() * n;
+    hipLaunchKernelGGL(gather_multi_kernel, dim3(grid_for(total, block)),
+                       dim3(block), 0, pq_stream(),
+                       (const GatherCol*)cols_dev.data_ptr(),
+                       (int)srcs.size(), (const int64_t*)idx.data_ptr(), n);
+    CHECK_HIP_PQ(hipGetLastError());
+  }
+  return outs;
+}

@@ -489,3 +489,29 @@ def test_parquet_batched_decode_multi_row_group(tmp_path):
         s=("v", "sum"), c=("s", "count")).sort_values("k").reset_index(
         drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+@pytest.mark.gpu
+def test_gather_multi_fused():
+    """Fused one-launch multi-column gather == per-column gathers."""
+    import bodo_amd_kernels as K
+    from bodo_amd import ops
+    from bodo_amd.core.table import Table
+
+    rng = np.random.default_rng(44)
+    n = 100_000
+    df = pd.DataFrame({
+        "a": rng.integers(-10**9, 10**9, n),
+        "b": rng.random(n),
+        "c": rng.integers(0, 100, n).astype(np.int32),
+        "f": rng.random(n).astype(np.float32),
+        "s": rng.choice(["x", "yy", "zzz"], n),
+        "bo": rng.integers(0, 2, n).astype(bool),
+    })
+    df.loc[rng.random(n) < 0.1, "b"] = np.nan
+    t = Table.from_pandas(df, device="cuda")
+    idx = torch.from_numpy(rng.integers(0, n, 30_000)).cuda()
+    fused = ops.take_table(t, idx)
+    ref = Table(t.names, [ops.gather(c, idx) for c in t.columns],
+                int(idx.numel()))
+    pd.testing.assert_frame_equal(fused.to_pandas(), ref.to_pandas())
