@@ -26,7 +26,7 @@ def _np_from_tensor(t: torch.Tensor) -> np.ndarray:
 
 class Column:
     __slots__ = ("dtype", "data", "mask", "offsets", "dictionary", "name",
-                 "_length", "val_range")
+                 "_length", "val_range", "child")
 
     def __init__(
         self,
@@ -42,10 +42,11 @@ class Column:
         self.mask = mask  # torch.bool, True = valid, or None (all valid)
         self.offsets = offsets  # STRING only: int64 (n+1)
         self.dictionary = dictionary  # DICT only: pa.StringArray (host)
+        self.child = None  # LIST only: element Column
         self.val_range = None  # optional known (lo, hi) for int columns
         if length is not None:
             self._length = length
-        elif dtype.kind == TypeKind.STRING:
+        elif dtype.kind in (TypeKind.STRING, TypeKind.LIST):
             self._length = int(offsets.shape[0]) - 1
         else:
             self._length = int(data.shape[0])
@@ -55,7 +56,7 @@ class Column:
 
     @property
     def device(self) -> torch.device:
-        t = self.offsets if self.dtype.kind == TypeKind.STRING else self.data
+        t = self.data if self.data is not None else self.offsets
         return t.device
 
     @property
@@ -67,6 +68,8 @@ class Column:
         for t in (self.data, self.mask, self.offsets):
             if t is not None:
                 n += t.numel() * t.element_size()
+        if self.child is not None:
+            n += self.child.nbytes()
         return n
 
     # ------------------------------------------------------------------
@@ -90,6 +93,8 @@ class Column:
             self.dictionary, self._length,
         )
         out.val_range = self.val_range
+        if self.child is not None:
+            out.child = self.child.to_device(device)
         return out
 
     # ------------------------------------------------------------------
@@ -132,6 +137,29 @@ class Column:
                 offsets=torch.from_numpy(np.ascontiguousarray(offsets)).to(device),
                 length=len(arr),
             )
+        if pa.types.is_list(t) or pa.types.is_large_list(t):
+            # arrow list<child>: int64 offsets + recursively converted child
+            # (reference: array_item_arr_ext.py ArrayItemArray layout)
+            arr = arr.cast(pa.large_list(arr.type.value_type)) \
+                if pa.types.is_list(t) else arr
+            offs = np.frombuffer(arr.buffers()[1], dtype=np.int64,
+                                 count=len(arr) + 1 + arr.offset)[arr.offset:]
+            base = int(offs[0])
+            offs = offs - base
+            child = Column.from_arrow(
+                arr.values.slice(base, int(offs[-1])), device)
+            mask = None
+            if arr.buffers()[0] is not None and arr.null_count:
+                mask = _unpack_validity(arr.buffers()[0], arr.offset,
+                                        len(arr))
+            out = Column(
+                bt.list_, None,
+                None if mask is None else torch.from_numpy(mask).to(device),
+                offsets=torch.from_numpy(
+                    np.ascontiguousarray(offs)).to(device),
+                length=len(arr))
+            out.child = child
+            return out
         if pa.types.is_null(t):
             # typeless all-null column: represent as float64 NaN
             data = torch.full((len(arr),), float("nan"), dtype=torch.float64)
@@ -205,6 +233,13 @@ class Column:
 
     def to_arrow(self) -> pa.Array:
         k = self.dtype.kind
+        if k == TypeKind.LIST:
+            offsets = _np_from_tensor(self.offsets)
+            child = self.child.to_arrow()
+            mask = None if self.mask is None else _np_from_tensor(self.mask)
+            return pa.LargeListArray.from_arrays(
+                pa.array(offsets, type=pa.int64()), child,
+                mask=None if mask is None else pa.array(~mask))
         if k == TypeKind.STRING:
             offsets = _np_from_tensor(self.offsets)
             data = _np_from_tensor(self.data) if self.data is not None else np.zeros(0, np.uint8)

@@ -35,6 +35,7 @@ class TypeKind(enum.IntEnum):
     UINT16 = 13
     UINT32 = 14
     UINT64 = 15
+    LIST = 16  # arrow list<child>: int64 offsets (n+1) + child column
 
 
 @dataclass(frozen=True)
@@ -96,6 +97,7 @@ date32 = DType(TypeKind.DATE32)
 timestamp_ns = DType(TypeKind.TIMESTAMP_NS)
 string = DType(TypeKind.STRING)
 dictionary = DType(TypeKind.DICT)
+list_ = DType(TypeKind.LIST)
 
 
 def decimal128(precision: int, scale: int) -> DType:

@@ -403,6 +403,11 @@ def _exec_distinct(node: pn.Distinct, ctx) -> Table:
     return rel.distinct_local(shuffled, node.subset, node.keep)
 
 
+def _exec_explode(node: pn.Explode, ctx) -> Table:
+    child = _exec(node.child, ctx)
+    return ops.explode_table(child, node.column)
+
+
 def _exec_sample(node: pn.Sample, ctx) -> Table:
     child = _exec(node.child, ctx)
     n_local = len(child)
@@ -803,6 +808,7 @@ _HANDLERS = {
     pn.Limit: _exec_limit,
     pn.Distinct: _exec_distinct,
     pn.Sample: _exec_sample,
+    pn.Explode: _exec_explode,
     pn.MapPartitions: _exec_map_partitions,
     pn.ShuffleByKey: _exec_shuffle_by_key,
     pn.RowId: _exec_rowid,

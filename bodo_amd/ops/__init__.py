@@ -27,6 +27,8 @@ from .evaluate import eval_expr, eval_filter, project  # noqa: F401
 
 def gather(col: Column, idx: torch.Tensor) -> Column:
     n = int(idx.shape[0])
+    if col.dtype.kind == TypeKind.LIST:
+        return _gather_list(col, idx)
     if len(col) == 0 and n > 0:
         # gathering from an empty column only happens for null-padded rows
         # (outer-join unmatched side): produce an all-null column
@@ -62,6 +64,73 @@ def _gather_string_arrow(col: Column, idx: torch.Tensor) -> Column:
     arr = col.to_arrow()
     taken = pc.take(arr, idx.numpy())
     return Column.from_arrow(taken, col.device)
+
+
+def _gather_list(col: Column, idx: torch.Tensor) -> Column:
+    """LIST gather: lengths gather + child range expansion (reference:
+    array_item array gather in _array_utils.cpp)."""
+    from ..core import types as _bt
+
+    dev = col.device
+    off = col.offsets
+    lens = (off[1:] - off[:-1])[idx]
+    new_off = torch.zeros(int(idx.numel()) + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(lens, 0, out=new_off[1:])
+    starts = off[:-1][idx]
+    total = int(new_off[-1].item())
+    reps = torch.repeat_interleave(starts, lens)
+    seq = torch.arange(total, device=dev) - torch.repeat_interleave(
+        new_off[:-1], lens)
+    child = gather(col.child, reps + seq) if total or len(col.child) else \
+        gather(col.child, torch.zeros(0, dtype=torch.int64, device=dev))
+    mask = col.mask[idx] if col.mask is not None else None
+    out = Column(_bt.list_, None, mask, offsets=new_off,
+                 length=int(idx.numel()))
+    out.child = child
+    return out
+
+
+def explode_table(tbl: Table, column: str) -> Table:
+    """pandas explode semantics: each list element becomes a row; empty or
+    null lists produce one row with a null value (reference role:
+    bodo/libs/_lateral.cpp FLATTEN)."""
+    from ..core import types as _bt
+
+    col = tbl.column(column)
+    if col.dtype.kind != TypeKind.LIST:
+        return tbl  # non-list explode is the identity on scalars
+    dev = tbl.device
+    off = col.offsets
+    lens = off[1:] - off[:-1]
+    valid_list = col.mask if col.mask is not None else torch.ones(
+        len(col), dtype=torch.bool, device=dev)
+    out_lens = torch.where((lens == 0) | ~valid_list,
+                           torch.ones_like(lens), lens)
+    row_idx = torch.repeat_interleave(
+        torch.arange(len(col), dtype=torch.int64, device=dev), out_lens)
+    new_off = torch.zeros(len(col) + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(out_lens, 0, out=new_off[1:])
+    total = int(new_off[-1].item())
+    pos_in_row = torch.arange(total, device=dev) - new_off[:-1][row_idx]
+    real = (lens[row_idx] > 0) & valid_list[row_idx] & \
+        (pos_in_row < lens[row_idx])
+    child_idx = (off[:-1][row_idx] + pos_in_row).clamp(
+        min=0, max=max(len(col.child) - 1, 0))
+    vals = gather(col.child, child_idx)
+    vmask = real if vals.mask is None else (vals.mask & real)
+    vals = Column(vals.dtype, vals.data, vmask, vals.offsets,
+                  vals.dictionary, total)
+    if vals.dtype.kind == TypeKind.LIST:
+        vals.child = gather(col.child, child_idx).child
+    names, cols = [], []
+    for nm, c in zip(tbl.names, tbl.columns):
+        if nm == column:
+            names.append(nm)
+            cols.append(vals)
+        else:
+            names.append(nm)
+            cols.append(gather(c, row_idx))
+    return Table(names, cols, total)
 
 
 def take_table(tbl: Table, idx: torch.Tensor) -> Table:
@@ -133,6 +202,18 @@ def concat_columns(cols: Sequence[Column]) -> Column:
             c.mask if c.mask is not None
             else torch.ones(len(c), dtype=torch.bool, device=first.device)
             for c in cols])
+    if first.dtype.kind == TypeKind.LIST:
+        from ..core import types as _bt
+
+        offs = [torch.zeros(1, dtype=torch.int64, device=first.device)]
+        base = 0
+        for c in cols:
+            offs.append(c.offsets[1:] + base)
+            base += int(c.offsets[-1].item())
+        out = Column(_bt.list_, None, masks, offsets=torch.cat(offs),
+                     length=n)
+        out.child = concat_columns([c.child for c in cols])
+        return out
     if first.dtype.kind == TypeKind.STRING:
         datas, offs, base = [], [torch.zeros(1, dtype=torch.int64, device=first.device)], 0
         for c in cols:

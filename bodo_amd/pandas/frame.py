@@ -639,6 +639,15 @@ class BodoDataFrame:
             plan = optimize(plan)
         return pn.explain(plan)
 
+    def explode(self, column: str) -> "BodoDataFrame":
+        """pandas explode over LIST columns (reference: frame explode /
+        _lateral.cpp FLATTEN): runs distributed, row-local."""
+        if isinstance(column, list):
+            if len(column) != 1:
+                return self._fallback("explode", column)
+            column = column[0]
+        return BodoDataFrame(pn.Explode(self._plan, column), self._columns)
+
     def melt(self, id_vars=None, value_vars=None, var_name="variable",
              value_name="value") -> "BodoDataFrame":
         """Wide-to-long unpivot as a lazy plan: one projection per value

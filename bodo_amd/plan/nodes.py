@@ -164,6 +164,23 @@ class Distinct(PlanNode):
 
 
 @dataclass(frozen=True)
+class Explode(PlanNode):
+    """Each element of a LIST column becomes a row (pandas explode /
+    LATERAL FLATTEN; reference: bodo/libs/_lateral.cpp)."""
+    child: PlanNode
+    column: str
+
+    def children(self):
+        return (self.child,)
+
+    def with_children(self, *ch):
+        return Explode(ch[0], self.column)
+
+    def out_columns(self):
+        return self.child.out_columns()
+
+
+@dataclass(frozen=True)
 class Sample(PlanNode):
     child: PlanNode
     n: Optional[int] = None

@@ -481,3 +481,34 @@ def test_set_index_reset_index_multiindex():
     # sort_index
     si = two.sort_index().to_pandas()
     assert list(si.index.get_level_values(0)) == sorted(df["a"].tolist())
+
+
+def test_list_columns_and_explode():
+    """LIST (arrow list<child>) columns: round trip, gather, concat and
+    pandas-semantics explode (reference: array_item_arr_ext.py +
+    _lateral.cpp FLATTEN)."""
+    import pyarrow as pa
+
+    df = pd.DataFrame({"k": [1, 2, 3, 4, 5],
+                       "l": pd.Series([[1, 2], [3], [], None, [4, 5, 6]])})
+    b = bpd.from_pandas(df)
+    got = b.explode("l").to_pandas().reset_index(drop=True)
+    exp = df.explode("l").reset_index(drop=True)
+    assert got["k"].tolist() == exp["k"].tolist()
+    gv = ["" if pd.isna(v) else float(v) for v in got["l"]]
+    ev = ["" if pd.isna(v) else float(v) for v in exp["l"]]
+    assert gv == ev
+    # explode then aggregate (the lateral-join shape)
+    out = b.explode("l").groupby("k", as_index=False).agg(
+        s=bpd.NamedAgg("l", "sum")).sort_values("k").to_pandas()
+    exp2 = df.explode("l").groupby("k", as_index=False).agg(
+        s=("l", "sum")).sort_values("k").reset_index(drop=True)
+    assert [float(v) for v in out["s"]] == [float(v) for v in exp2["s"]]
+
+
+def test_list_string_explode():
+    df = pd.DataFrame({"k": [1, 2],
+                       "l": pd.Series([["a", "bb"], ["ccc"]])})
+    b = bpd.from_pandas(df)
+    got = b.explode("l").to_pandas().reset_index(drop=True)
+    assert got["l"].astype(str).tolist() == ["a", "bb", "ccc"]
