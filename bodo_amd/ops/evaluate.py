@@ -764,6 +764,17 @@ def _str_dict_generic(a: Column, op: str, args, kwargs) -> Column:
     import pyarrow as pa
 
     vals = pd.Series(a.dictionary.to_pylist(), dtype="object")
+    if op == "split_list":
+        lists = vals.str.split(args[0] if args and args[0] is not None
+                               else None)
+        la = pa.array(lists.tolist(), type=pa.large_list(pa.large_string()))
+        from . import gather as _g
+
+        base = Column.from_arrow(la, a.device)
+        out = _g(base, a.data.long())
+        if a.mask is not None:
+            out.mask = a.mask if out.mask is None else (out.mask & a.mask)
+        return out
     if op == "split_get":
         pat, idx = args
         res = vals.str.split(pat).str.get(idx)
@@ -899,6 +910,12 @@ def str_op(a: Column, op: str, args, kwargs=None) -> Column:
         if op == "split_get":
             pat, idx = args
             res_s = ser.str.split(pat).str.get(idx)
+        elif op == "split_list":
+            lists = ser.str.split(args[0] if args and args[0] is not None
+                                  else None)
+            la = pa.array(lists.tolist(),
+                          type=pa.large_list(pa.large_string()))
+            return Column.from_arrow(la, a.device)
         else:
             res_s = getattr(ser.str, op)(*args, **(kwargs or {}))
         if any(isinstance(v, (list, tuple)) for v in res_s.head(64).tolist()):

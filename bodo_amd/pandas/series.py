@@ -817,12 +817,13 @@ def _pd_dtype_to_bodo(dtype):
 
 
 class _SplitResult:
-    """Lazy result of .str.split(pat): only element access is supported,
-    fused into one split_get string op."""
+    """Lazy result of .str.split(pat): element access fuses into one
+    split_get pass; any other use materializes the real LIST<string>
+    series (reference: str split -> array_item array)."""
 
     def __init__(self, s: BodoSeries, pat):
-        self._s = s
-        self._pat = pat
+        object.__setattr__(self, "_s", s)
+        object.__setattr__(self, "_pat", pat)
 
     @property
     def str(self):
@@ -833,6 +834,13 @@ class _SplitResult:
             StrOp(self._s._expr, "split_get", (self._pat, int(i))), None)
 
     __getitem__ = get
+
+    def _series(self) -> "BodoSeries":
+        return self._s._wrap(
+            StrOp(self._s._expr, "split_list", (self._pat,)), None)
+
+    def __getattr__(self, name):
+        return getattr(self._series(), name)
 
 
 class _RollingSeries:

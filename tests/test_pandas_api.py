@@ -512,3 +512,22 @@ def test_list_string_explode():
     b = bpd.from_pandas(df)
     got = b.explode("l").to_pandas().reset_index(drop=True)
     assert got["l"].astype(str).tolist() == ["a", "bb", "ccc"]
+
+
+def test_str_split_list_and_explode():
+    """str.split materializes as a LIST<string> series; explode after split
+    is the csv-tags lateral pattern."""
+    df = pd.DataFrame({"k": [1, 2, 3],
+                       "s": ["a,b", "c", "d,e,f"]})
+    b = bpd.from_pandas(df)
+    parts = b.s.str.split(",")
+    got = parts.to_pandas()
+    assert [list(v) for v in got] == [["a", "b"], ["c"], ["d", "e", "f"]]
+    # fused element access still works
+    first = b.s.str.split(",").get(0).to_pandas()
+    assert first.astype(str).tolist() == ["a", "c", "d"]
+    # assign + explode
+    b["parts"] = b.s.str.split(",")._series()
+    out = b.explode("parts").to_pandas()
+    assert out["parts"].astype(str).tolist() == ["a", "b", "c", "d", "e", "f"]
+    assert out["k"].tolist() == [1, 1, 2, 3, 3, 3]
