@@ -452,6 +452,101 @@ def _exec_rowid(node: pn.RowId, ctx) -> Table:
     return child.with_column(node.name, Column(bt.int64, rid))
 
 
+def _gather_halo(data: torch.Tensor, mask, k: int, ctx):
+    """Collect |k| boundary rows from neighbor ranks as tensors: for k>0
+    the tail of preceding ranks, for k<0 the head of following ranks
+    (reference: border exchange in the distributed pass for shifts)."""
+    n = int(data.numel())
+    if k > 0:
+        edge = data[-min(k, n):]
+        emask = None if mask is None else mask[-min(k, n):]
+    else:
+        edge = data[:min(-k, n)]
+        emask = None if mask is None else mask[:min(-k, n)]
+    edges = comm.allgather_obj((edge.cpu(), None if emask is None
+                                else emask.cpu()))
+    pieces = []
+    need = abs(k)
+    have = 0
+    if k > 0:
+        i = ctx.rank - 1
+        while have < need and i >= 0:
+            t, m = edges[i]
+            take = min(need - have, int(t.numel()))
+            pieces.insert(0, (t[-take:] if take else t[:0],
+                              None if m is None else m[-take:]))
+            have += take
+            if int(t.numel()) >= need:
+                break
+            i -= 1
+    else:
+        i = ctx.rank + 1
+        while have < need and i < ctx.world:
+            t, m = edges[i]
+            take = min(need - have, int(t.numel()))
+            pieces.append((t[:take], None if m is None else m[:take]))
+            have += take
+            if int(t.numel()) >= need:
+                break
+            i += 1
+    if not pieces:
+        return data[:0], None if mask is None else mask[:0]
+    hd = torch.cat([p[0] for p in pieces]).to(data.device)
+    if mask is not None or any(p[1] is not None for p in pieces):
+        hm = torch.cat([
+            p[1] if p[1] is not None
+            else torch.ones(int(p[0].numel()), dtype=torch.bool)
+            for p in pieces]).to(data.device)
+    else:
+        hm = None
+    return hd, hm
+
+
+def _shift_column_device(col: Column, k: int, ctx) -> Optional[Column]:
+    """Device shift for fixed-width columns: slice + halo, null-padded ends
+    (no host round trip)."""
+    if col.dtype.kind in (TypeKind.STRING,):
+        return None
+    n = len(col)
+    dev = col.device
+    data, mask = col.data, col.mask
+    if k == 0:
+        return col
+    halo_d, halo_m = (data[:0], None)
+    if ctx.world > 1:
+        halo_d, halo_m = _gather_halo(data, mask, k, ctx)
+    h = int(halo_d.numel())
+    out = torch.zeros(n, dtype=data.dtype, device=dev)
+    valid = torch.zeros(n, dtype=torch.bool, device=dev)
+    if k > 0:
+        pad = k - h  # rows with no predecessor anywhere: null
+        if h:
+            out[pad:pad + h] = halo_d
+            valid[pad:pad + h] = halo_m if halo_m is not None else True
+        m = n - k
+        if m > 0:
+            out[k:] = data[:m]
+            valid[k:] = mask[:m] if mask is not None else True
+    else:
+        kk = -k
+        m = n - kk
+        if m > 0:
+            out[:m] = data[kk:]
+            valid[:m] = mask[kk:] if mask is not None else True
+        if h:
+            out[m:m + h] = halo_d
+            valid[m:m + h] = halo_m if halo_m is not None else True
+    if col.dtype.kind == TypeKind.DICT:
+        res = Column(col.dtype, out.to(torch.int32), valid,
+                     dictionary=col.dictionary, length=n)
+        return res
+    if col.dtype.is_float:
+        out = torch.where(valid, out, torch.tensor(
+            float("nan"), dtype=out.dtype, device=dev))
+        return Column(col.dtype, out)
+    return Column(col.dtype, out, valid)
+
+
 def _exec_shift(node: pn.Shift, ctx) -> Table:
     child = _exec(node.child, ctx)
     n = len(child)
@@ -459,6 +554,11 @@ def _exec_shift(node: pn.Shift, ctx) -> Table:
     names, cols = [], []
     for out_name, in_name in node.specs:
         col = child.column(in_name)
+        dcol = _shift_column_device(col, k, ctx)
+        if dcol is not None:
+            names.append(out_name)
+            cols.append(dcol)
+            continue
         ser = col.to_pandas()
         if ctx.world > 1 and k != 0:
             # exchange |k| boundary rows: for k>0 take the tail of preceding
@@ -606,8 +706,93 @@ def _exec_cumulative(node: pn.Cumulative, ctx) -> Table:
     return Table(names, cols, n)
 
 
+_ROLL_DEV_FUNCS = {"sum", "mean", "count", "min", "max"}
+
+
+def _exec_rolling_device(node: pn.Rolling, ctx, child: Table) -> Optional[Table]:
+    """Global rolling via prefix sums on device with tensor halo rows
+    (reference role: window frame aggregation, _window_aggfuncs.cpp)."""
+    w = int(node.window)
+    minp = node.min_periods if node.min_periods is not None else w
+    if any(f not in _ROLL_DEV_FUNCS for _, _, f in node.specs):
+        return None
+    if w > 512:
+        return None
+    in_cols = {inn: child.column(inn) for _, inn, _ in node.specs}
+    if any(c.dtype.kind in (TypeKind.STRING, TypeKind.DICT)
+           for c in in_cols.values()):
+        return None
+    n = len(child)
+    dev = child.device
+    names, cols = [], []
+    halo_cache = {}
+    for out_name, in_name, func in node.specs:
+        c = in_cols[in_name]
+        if in_name not in halo_cache:
+            hd, hm = (c.data[:0], None)
+            if ctx.world > 1 and w > 1:
+                hd, hm = _gather_halo(c.data, c.mask, w - 1, ctx)
+            halo_cache[in_name] = (hd, hm)
+        hd, hm = halo_cache[in_name]
+        h = int(hd.numel())
+        x = torch.cat([hd, c.data]).double()
+        valid = torch.ones(h + n, dtype=torch.bool, device=dev)
+        if c.mask is not None:
+            valid = torch.cat([hm if hm is not None else torch.ones(
+                h, dtype=torch.bool, device=dev), c.mask])
+        if c.dtype.is_float:
+            valid = valid & ~torch.isnan(torch.cat([hd, c.data]))
+        xf = torch.where(valid, x, torch.zeros((), dtype=x.dtype, device=dev))
+        pos = torch.arange(h + n, device=dev)
+        lo = (pos - (w - 1)).clamp(min=0)
+        cs = torch.cumsum(xf, 0)
+        excl = cs - xf
+        cv = torch.cumsum(valid.double(), 0)
+        excl_v = cv - valid.double()
+        rcount = cv - excl_v[lo]
+        if func == "count":
+            res = rcount
+        elif func in ("sum", "mean"):
+            rsum = cs - excl[lo]
+            res = rsum if func == "sum" else rsum / rcount
+        else:
+            sent = float("inf") if func == "min" else float("-inf")
+            cur = torch.where(valid, x, torch.full(
+                (), sent, dtype=x.dtype, device=dev))
+            acc = cur.clone()
+            for t in range(1, w):
+                sh = torch.empty_like(cur)
+                sh[t:] = cur[:h + n - t]
+                sh[:t] = sent
+                inwin = pos - t >= lo
+                sh = torch.where(inwin, sh, torch.full(
+                    (), sent, dtype=x.dtype, device=dev))
+                acc = torch.minimum(acc, sh) if func == "min" else \
+                    torch.maximum(acc, sh)
+            res = acc
+        # pandas min_periods: non-null observations for value aggs, but
+        # window ROW completeness for count
+        rows_in_win = pos - lo + 1
+        if func == "count":
+            bad = rows_in_win < minp
+        else:
+            bad = (rcount < minp) | torch.isinf(res)
+        res = torch.where(bad, torch.full((), float("nan"), dtype=res.dtype,
+                                          device=dev), res)
+        names.append(out_name)
+        cols.append(Column(bt.float64, res[h:].contiguous()))
+    return Table(names, cols, n)
+
+
 def _exec_rolling(node: pn.Rolling, ctx) -> Table:
     child = _exec(node.child, ctx)
+    dev_out = None
+    try:
+        dev_out = _exec_rolling_device(node, ctx, child)
+    except Exception:
+        dev_out = None
+    if dev_out is not None:
+        return dev_out
     w = int(node.window)
     pdf = child.to_pandas()
     halo = 0

@@ -531,3 +531,23 @@ def test_str_split_list_and_explode():
     out = b.explode("parts").to_pandas()
     assert out["parts"].astype(str).tolist() == ["a", "b", "c", "d", "e", "f"]
     assert out["k"].tolist() == [1, 1, 2, 3, 3, 3]
+
+
+def test_rolling_device_path_variants():
+    """Global rolling sum/mean/min/max/count via device prefix ops (host
+    pandas only for exotic funcs)."""
+    import bodo_amd.engine.executor as E
+
+    rng = np.random.default_rng(51)
+    n = 5000
+    df = pd.DataFrame({"x": rng.random(n)})
+    df.loc[rng.random(n) < 0.1, "x"] = np.nan
+    b = bpd.from_pandas(df)
+    for f in ("sum", "mean", "min", "max", "count"):
+        for w, mp in ((7, None), (12, 3)):
+            r = b.x.rolling(w, min_periods=mp)
+            got = getattr(r, f)().to_pandas().reset_index(drop=True)
+            exp = getattr(df.x.rolling(w, min_periods=mp), f)().reset_index(
+                drop=True)
+            pd.testing.assert_series_equal(got, exp, check_names=False,
+                                           check_dtype=False, atol=1e-9)
