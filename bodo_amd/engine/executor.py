@@ -603,6 +603,68 @@ def _exec_shift(node: pn.Shift, ctx) -> Table:
     return out
 
 
+def _fill_column_device(col: Column, forward: bool, ctx) -> Optional[Column]:
+    """Device ffill/bfill: last-valid-index via cummax gather; cross-rank
+    carry as one small object round (reference: array_kernels fillna
+    forward/backward with border exchange)."""
+    if col.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+        return None
+    n = len(col)
+    dev = col.device
+    data, mask = col.data, col.mask
+    valid = torch.ones(n, dtype=torch.bool, device=dev)
+    if mask is not None:
+        valid = mask.clone()
+    if col.dtype.is_float:
+        valid &= ~torch.isnan(data)
+    pos = torch.arange(n, device=dev)
+    if forward:
+        marked = torch.where(valid, pos, torch.full_like(pos, -1))
+        last = torch.cummax(marked, 0).values
+    else:
+        # nearest valid at-or-after i == forward pass in the flipped domain
+        v_f = torch.flip(valid, [0])
+        m_f = torch.where(v_f, pos, torch.full_like(pos, -1))
+        last_f = torch.cummax(m_f, 0).values
+        last = torch.flip(torch.where(last_f >= 0, (n - 1) - last_f,
+                                      torch.full_like(last_f, -1)), [0])
+    have = last >= 0
+    out = data[last.clamp(min=0)]
+    out_valid = valid | have
+    # cross-rank carry of the edge value
+    if ctx.world > 1:
+        if forward:
+            edge = None
+            if n and bool(have[-1].item()):
+                edge = data[last[-1]].item()
+            edges = comm.allgather_obj(edge)
+            carry = None
+            for v in edges[:ctx.rank]:
+                if v is not None:
+                    carry = v
+        else:
+            edge = None
+            if n and bool(have[0].item()):
+                edge = data[last[0]].item()
+            edges = comm.allgather_obj(edge)
+            carry = None
+            for v in edges[ctx.rank + 1:]:
+                if v is not None:
+                    carry = v
+                    break
+        if carry is not None:
+            fillv = torch.tensor(carry, dtype=out.dtype, device=dev)
+            out = torch.where(out_valid, out, fillv)
+            out_valid = torch.ones_like(out_valid)
+    if col.dtype.is_float:
+        out = torch.where(out_valid, out, torch.tensor(
+            float("nan"), dtype=out.dtype, device=dev))
+        return Column(col.dtype, out)
+    if bool(out_valid.all().item()):
+        return Column(col.dtype, out)
+    return Column(col.dtype, out, out_valid)
+
+
 def _exec_fill(node: "pn.Fill", ctx) -> Table:
     """ffill/bfill over the global row order: local fill plus the nearest
     valid value carried across rank boundaries (reference: array_kernels
@@ -610,6 +672,11 @@ def _exec_fill(node: "pn.Fill", ctx) -> Table:
     child = _exec(node.child, ctx)
     names, cols = [], []
     for out_name, in_name in node.specs:
+        dcol = _fill_column_device(child.column(in_name), node.forward, ctx)
+        if dcol is not None:
+            names.append(out_name)
+            cols.append(dcol)
+            continue
         ser = child.column(in_name).to_pandas()
         filled = ser.ffill() if node.forward else ser.bfill()
         if ctx.world > 1:

@@ -551,3 +551,23 @@ def test_rolling_device_path_variants():
                 drop=True)
             pd.testing.assert_series_equal(got, exp, check_names=False,
                                            check_dtype=False, atol=1e-9)
+
+
+def test_ffill_bfill_device_differential():
+    rng = np.random.default_rng(61)
+    n = 3000
+    x = rng.random(n)
+    x[rng.random(n) < 0.3] = np.nan
+    x[:5] = np.nan  # leading nulls stay null under ffill
+    df = pd.DataFrame({"x": x, "i": rng.integers(0, 9, n).astype("float64")})
+    df.loc[rng.random(n) < 0.2, "i"] = np.nan
+    b = bpd.from_pandas(df)
+    for f in ("ffill", "bfill"):
+        got = getattr(b.x, f)().to_pandas().reset_index(drop=True)
+        exp = getattr(df.x, f)().reset_index(drop=True)
+        pd.testing.assert_series_equal(got, exp, check_names=False,
+                                       check_dtype=False)
+        got2 = getattr(b.i, f)().to_pandas().reset_index(drop=True)
+        exp2 = getattr(df.i, f)().reset_index(drop=True)
+        pd.testing.assert_series_equal(got2, exp2, check_names=False,
+                                       check_dtype=False)
