@@ -247,6 +247,29 @@ def _exec_sort(node: pn.Sort, ctx) -> Table:
     return ooc.sort_local(child, list(node.keys), asc, node.na_position)
 
 
+def _float_sort_key(col: Column, asc: bool, na_position: str):
+    """Order-preserving int64 view of a float sort key (sortable-bits
+    transform) so single-float-key range partitions stay on device
+    (round-1 finding: non-packable splitters compared per row in host
+    pandas).  NaN maps to the na_position end; descending inverts."""
+    if not col.dtype.is_float or col.mask is not None:
+        return None
+    d = col.data
+    x = d.double() if d.dtype != torch.float64 else d
+    bits = x.view(torch.int64)
+    # IEEE-754 total order: positive floats flip the sign bit; negatives
+    # flip all bits
+    sortable = torch.where(bits >= 0, bits ^ torch.tensor(
+        -0x8000000000000000, device=d.device), ~bits)
+    if not asc:
+        sortable = ~sortable  # order-reversing, overflow-free
+    nan = torch.isnan(x)
+    sent = torch.tensor(
+        (2 ** 62) if na_position == "last" else -(2 ** 62),
+        device=d.device)
+    return torch.where(nan, sent, sortable)
+
+
 def _sample_positions(n: int, k: int, ctx, device) -> torch.Tensor:
     """Seeded per-rank sample positions so distributed sorts are reproducible
     across runs (splitters otherwise change run to run)."""
@@ -265,6 +288,8 @@ def _range_partition(tbl: Table, keys, asc, na_position, ctx) -> Table:
     w = ctx.world
     n = len(tbl)
     packed = ops.pack_ordered_keys([tbl.column(k) for k in keys], asc)
+    if packed is None and len(keys) == 1:
+        packed = _float_sort_key(tbl.column(keys[0]), asc[0], na_position)
     if packed is not None:
         k = min(n, 64 * w)
         if n > 0:

@@ -701,3 +701,26 @@ def test_dist_gbdt_histogram_allreduce():
     r2_single = m.score(Xt, yt)
     assert abs(r2_dist - r2_single) < 0.02, (r2_dist, r2_single)
     assert r2_dist > 0.9
+
+
+def _q_sort_float(bpd, rank, payload):
+    return bpd.from_pandas(payload["df"]).sort_values(
+        "v", ascending=payload["asc"], na_position=payload["na"])
+
+
+def test_dist_sort_float_key_device_partition():
+    """Single-float-key distributed sorts use the sortable-bits device
+    range partition (was a host per-row comparison)."""
+    rng = np.random.default_rng(71)
+    n = 4000
+    df = pd.DataFrame({"v": rng.standard_normal(n) * 100,
+                       "k": np.arange(n)})
+    df.loc[rng.random(n) < 0.05, "v"] = np.nan
+    for asc in (True, False):
+        for na in ("last", "first"):
+            got = run_dist(_q_sort_float,
+                           {"df": df, "asc": asc, "na": na}).reset_index(
+                drop=True)
+            exp = df.sort_values("v", ascending=asc,
+                                 na_position=na).reset_index(drop=True)
+            pd.testing.assert_frame_equal(got, exp, check_dtype=False)
