@@ -102,3 +102,71 @@ def test_fuzz_sql(seed):
         exp = df[(df.c.isin(["p", "q"])) & (df.b < thr)].sort_values(
             ["a", "b"]).head(7)[["a", "b"]].reset_index(drop=True)
     pd.testing.assert_frame_equal(got, exp, check_dtype=False, atol=1e-9)
+
+
+@pytest.mark.parametrize("seed", range(20))
+def test_fuzz_round2_features(seed):
+    """Round-2 surface fuzz: decimal columns, lists+explode, window funcs,
+    shift/fill/rolling, lazy scalars mixed in one pipeline."""
+    from decimal import Decimal
+
+    import pyarrow as pa
+
+    rng = np.random.default_rng(1000 + seed)
+    n = int(rng.integers(5, 400))
+    cents = rng.integers(-10**6, 10**6, n)
+    lists = [list(map(int, rng.integers(0, 5, rng.integers(0, 4))))
+             for _ in range(n)]
+    df = pd.DataFrame({
+        "k": rng.integers(0, 6, n),
+        "o": rng.permutation(n),
+        "v": np.where(rng.random(n) < 0.2, np.nan, rng.random(n)),
+    })
+    df["m"] = pd.Series([Decimal(int(c)) / 100 for c in cents])
+    df["l"] = pd.Series(lists, dtype=object)
+    b = bpd.from_pandas(df)
+    op = seed % 5
+    if op == 0:  # decimal groupby + compare
+        got = b[b.m > 0].groupby("k", as_index=False).agg(
+            s=bpd.NamedAgg("m", "sum")).to_pandas().sort_values(
+            "k").reset_index(drop=True)
+        ref = df[[c > 0 for c in df.m]].groupby("k", as_index=False).agg(
+            s=("m", "sum")).sort_values("k").reset_index(drop=True)
+        assert [float(x) for x in got.s] == pytest.approx(
+            [float(x) for x in ref.s])
+    elif op == 1:  # explode + agg
+        got = b.explode("l").groupby("k", as_index=False).agg(
+            c=bpd.NamedAgg("l", "count")).to_pandas().sort_values(
+            "k").reset_index(drop=True)
+        ref = df.explode("l").groupby("k", as_index=False).agg(
+            c=("l", "count")).sort_values("k").reset_index(drop=True)
+        pd.testing.assert_frame_equal(got, ref, check_dtype=False)
+    elif op == 2:  # window over permuted order
+        bb = b
+        bb["r"] = bb.groupby("k")["o"].rank(method="min")
+        got = bb.to_pandas()[["k", "o", "r"]].sort_values(
+            ["k", "o"]).reset_index(drop=True)
+        ref = df.copy()
+        ref["r"] = ref.groupby("k")["o"].rank(method="min")
+        ref = ref[["k", "o", "r"]].sort_values(["k", "o"]).reset_index(
+            drop=True)
+        pd.testing.assert_frame_equal(got, ref, check_dtype=False)
+    elif op == 3:  # shift + ffill + rolling mean chained
+        k = int(rng.integers(1, 4))
+        got = b.v.shift(k).to_pandas().reset_index(drop=True)
+        ref = df.v.shift(k).reset_index(drop=True)
+        pd.testing.assert_series_equal(got, ref, check_names=False,
+                                       check_dtype=False)
+        got2 = b.v.ffill().rolling(3, min_periods=1).mean().to_pandas()
+        ref2 = df.v.ffill().rolling(3, min_periods=1).mean()
+        pd.testing.assert_series_equal(got2.reset_index(drop=True),
+                                       ref2.reset_index(drop=True),
+                                       check_names=False, check_dtype=False,
+                                       atol=1e-9)
+    else:  # lazy scalar in filter + set_index round trip
+        m = b.v.mean()
+        got = b[b.v > m].to_pandas()
+        ref = df[df.v > df.v.mean()]
+        assert len(got) == len(ref)
+        si = b.set_index(["k"]).to_pandas()
+        assert si.index.name == "k"
