@@ -783,7 +783,19 @@ def _str_dict_generic(a: Column, op: str, args, kwargs) -> Column:
     if res.dtype == object or isinstance(res.dtype, pd.StringDtype):
         out_vals = res.tolist()
         if any(isinstance(v, (list, tuple)) for v in out_vals):
-            raise NotImplementedError(f"str.{op} returns lists")
+            # list results over the dictionary -> LIST column gathered by
+            # codes (findall/rsplit on dict-encoded strings)
+            la = pa.array([None if v is None or isinstance(v, float)
+                           else [str(x) for x in v] for v in out_vals],
+                          type=pa.large_list(pa.large_string()))
+            from . import gather as _g
+
+            base = Column.from_arrow(la, a.device)
+            out = _g(base, a.data.long())
+            if a.mask is not None:
+                out.mask = a.mask if out.mask is None \
+                    else (out.mask & a.mask)
+            return out
         # missing results (e.g. split().get(i) past the end) come back as
         # float NaN: they become validity-mask nulls, never dictionary
         # entries (arrow dicts reject null categories)
@@ -919,7 +931,11 @@ def str_op(a: Column, op: str, args, kwargs=None) -> Column:
         else:
             res_s = getattr(ser.str, op)(*args, **(kwargs or {}))
         if any(isinstance(v, (list, tuple)) for v in res_s.head(64).tolist()):
-            raise NotImplementedError(f"str.{op} returns lists")
+            # list results (findall/rsplit/...) become LIST<string> columns
+            la = pa.array([None if v is None or (isinstance(v, float))
+                           else [str(x) for x in v] for v in res_s],
+                          type=pa.large_list(pa.large_string()))
+            return Column.from_arrow(la, a.device)
         return Column.from_arrow(pa.Array.from_pandas(res_s), a.device)
     res = fmap[op](arr)
     if res.type in (__import__("pyarrow").int32(), __import__("pyarrow").int64()):

@@ -777,10 +777,33 @@ class _StrAccessor:
         materialize; reference: BodoSQL split_part kernel)."""
         return _SplitResult(self._s, pat)
 
-    # methods whose pandas result is a list/frame per row (need explode
-    # semantics) fall back via the Series-level pandas fallback
-    _UNSUPPORTED = {"rsplit", "extract", "extractall", "findall",
-                    "get_dummies", "partition", "rpartition", "cat"}
+    def findall(self, pat, flags=0):
+        """LIST<string> result (reference: BodoSQL regexp kernels +
+        array_item results)."""
+        return self._s._wrap(StrOp(self._s._expr, "findall", (pat,),
+                                   (("flags", flags),)), None)
+
+    def extract(self, pat, flags=0, expand=True):
+        """Single-group extract lowers to a host regex map; multi-group /
+        expand results fall back (frame-shaped)."""
+        import re as _re
+
+        if _re.compile(pat, flags).groups != 1 or expand:
+            raise AttributeError("extract")  # series-level pandas fallback
+        rx = _re.compile(pat, flags)
+
+        def _ex(v, _rx=rx):
+            if v is None:
+                return None
+            m = _rx.search(v)
+            return m.group(1) if m else None
+
+        return self._s._wrap(UdfMap(self._s._expr, _ex, "ignore"), None)
+
+    # methods whose pandas result is frame-shaped fall back via the
+    # Series-level pandas fallback
+    _UNSUPPORTED = {"extractall", "get_dummies", "partition", "rpartition",
+                    "cat"}
 
     def __getattr__(self, op):
         """Any other pandas .str method lowers to a StrOp evaluated on the

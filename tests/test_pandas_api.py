@@ -571,3 +571,20 @@ def test_ffill_bfill_device_differential():
         exp2 = getattr(df.i, f)().reset_index(drop=True)
         pd.testing.assert_series_equal(got2, exp2, check_names=False,
                                        check_dtype=False)
+
+
+def test_str_findall_extract_rsplit():
+    df = pd.DataFrame({"s": ["a1b22c333", "x9", "nope", None]})
+    b = bpd.from_pandas(df)
+    got = b.s.str.findall(r"\d+").to_pandas()
+    exp = df.s.str.findall(r"\d+")
+    assert [None if v is None else list(v) for v in got] == \
+        [None if not isinstance(v, list) else v for v in exp]
+    got2 = b.s.str.extract(r"([a-z])\d", expand=False).to_pandas()
+    exp2 = df.s.str.extract(r"([a-z])\d", expand=False)
+    assert [None if pd.isna(v) else v for v in got2] == \
+        [None if pd.isna(v) else v for v in exp2]
+    got3 = b.s.str.rsplit("b").to_pandas()
+    exp3 = df.s.str.rsplit("b")
+    assert [None if v is None else list(v) for v in got3] == \
+        [None if not isinstance(v, list) else v for v in exp3]
