@@ -31,7 +31,8 @@ TRANSFORM_AGGS = {"transform_sum": "sum", "transform_mean": "mean",
                   "transform_min": "min", "transform_max": "max",
                   "transform_count": "count", "transform_size": "size"}
 
-ORDERED_FUNCS = {"row_number", "rank", "dense_rank", "shift", "cumsum",
+ORDERED_FUNCS = {"row_number", "rank", "dense_rank", "percent_rank",
+                 "cume_dist", "nth_value", "shift", "cumsum",
                  "cumcount", "first_value", "last_value", "ntile",
                  "rolling_sum", "rolling_mean", "rolling_min", "rolling_max",
                  "rolling_count"}
@@ -279,6 +280,48 @@ def _ordered_local_device(tbl: Table, keys, order_by, ascending,
             at = seg_start if func == "first_value" else end_pos[segid]
             res = v[at]
             res_inv = inv[at] if inv is not None else None
+        elif func == "percent_rank":
+            # (rank-1)/(size-1); single-row partitions -> 0
+            ocol = order_cols[0] if order_cols else None
+            if ocol is None:
+                return None
+            eq = _adj_eq(ocol, idx)
+            if eq is None:
+                return None
+            run_new = seg_new.clone()
+            run_new[1:] |= ~eq
+            run_start = torch.nonzero(run_new).squeeze(1)
+            runid = torch.cumsum(run_new.long(), 0) - 1
+            rk = (run_start[runid] - seg_start + 1).double()
+            denom = (seg_size - 1).double().clamp(min=1)
+            res = (rk - 1) / denom
+            res_inv = None
+        elif func == "cume_dist":
+            # rows with order value <= current / partition size: position
+            # of this value-run's END + 1, over seg_size
+            ocol = order_cols[0] if order_cols else None
+            if ocol is None:
+                return None
+            eq = _adj_eq(ocol, idx)
+            if eq is None:
+                return None
+            run_new = seg_new.clone()
+            run_new[1:] |= ~eq
+            run_start = torch.nonzero(run_new).squeeze(1)
+            runid = torch.cumsum(run_new.long(), 0) - 1
+            n_runs = int(run_start.numel())
+            run_end = torch.cat([run_start[1:] - 1,
+                                 torch.tensor([n - 1], device=dev)])
+            res = (run_end[runid] - seg_start + 1).double() / \
+                seg_size.double()
+            res_inv = None
+        elif func == "nth_value":
+            k = int(arg or 1)
+            v, inv = sorted_vals(in_name)
+            at = (seg_start + (k - 1)).clamp(max=n - 1)
+            ok = (row_in_seg >= k - 1) & (seg_start + (k - 1) <= end_pos[segid])
+            res = v[at]
+            res_inv = ~ok if inv is None else (~ok | inv[at])
         elif func == "ntile":
             k = int(arg or 1)
             res = (row_in_seg * k) // seg_size + 1
@@ -393,6 +436,21 @@ def _ordered_local(tbl: Table, keys, order_by, ascending, specs) -> Table:
             r = gbs[in_name].rolling(int(arg), min_periods=1)
             res = getattr(r, base)()
             res = res.droplevel(list(range(len(keys))))
+            res = res.reindex(pdf.index)
+        elif func == "percent_rank":
+            col = order_by[0] if order_by else in_name
+            r = gb[col].rank(method="min")
+            size = gb[keys[0]].transform("size")
+            res = (r - 1) / (size - 1).clip(lower=1)
+        elif func == "cume_dist":
+            col = order_by[0] if order_by else in_name
+            r = gb[col].rank(method="max")
+            size = gb[keys[0]].transform("size")
+            res = r / size
+        elif func == "nth_value":
+            k = int(arg or 1)
+            res = gbs[in_name].transform(
+                lambda s, _k=k: s.iloc[_k - 1] if len(s) >= _k else np.nan)
             res = res.reindex(pdf.index)
         elif func == "ntile":
             k = int(arg or 1)

@@ -315,10 +315,14 @@ class Planner:
                 spec = (out, "", "row_number", None)
             elif fn in ("first_value", "last_value"):
                 spec = (out, arg_col, fn, None)
+            elif fn == "nth_value":
+                k = int(w.args[1].value) if len(w.args) > 1 else 1
+                spec = (out, arg_col, "nth_value", k)
             elif fn == "ntile":
                 k = int(w.args[0].value) if w.args else 1
                 spec = (out, "", "ntile", k)
-            elif fn in ("rank", "dense_rank"):
+            elif fn in ("rank", "dense_rank", "percent_rank",
+                        "cume_dist"):
                 spec = (out, "", fn, None)
             elif fn in ("lag", "lead"):
                 n = 1
@@ -479,10 +483,14 @@ class Planner:
                 arg_col = as_col(w.args[0], "__wa")
             if fn == "row_number":
                 spec = (out, "", "row_number", None)
-            elif fn in ("rank", "dense_rank"):
+            elif fn in ("rank", "dense_rank", "percent_rank",
+                        "cume_dist"):
                 spec = (out, "", fn, None)
             elif fn in ("first_value", "last_value"):
                 spec = (out, arg_col, fn, None)
+            elif fn == "nth_value":
+                k = int(w.args[1].value) if len(w.args) > 1 else 1
+                spec = (out, arg_col, "nth_value", k)
             elif fn == "ntile":
                 k = int(w.args[0].value) if w.args else 1
                 spec = (out, "", "ntile", k)

@@ -682,3 +682,31 @@ def test_sql_function_breadth():
     assert int(got["dw"].iloc[0]) == 6  # pandas Monday=0
     assert int(got["dy"].iloc[0]) == 15
     assert int(got["wk"].iloc[0]) == 3
+
+
+def test_window_percent_rank_cume_dist_nth():
+    import numpy as np
+
+    from bodo_amd.sql import BodoSQLContext
+
+    rng = np.random.default_rng(8)
+    df = pd.DataFrame({"k": rng.integers(0, 8, 500),
+                       "o": rng.integers(0, 40, 500),
+                       "v": rng.random(500)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("""
+        select k, o,
+               percent_rank() over (partition by k order by o) as pr,
+               cume_dist() over (partition by k order by o) as cd,
+               nth_value(v, 2) over (partition by k order by o) as nv
+        from t order by k, o
+    """).to_pandas().reset_index(drop=True)
+    ref = df.copy()
+    g = ref.groupby("k")["o"]
+    ref["pr"] = g.rank(method="min").sub(1) / (g.transform("size") - 1).clip(lower=1)
+    ref["cd"] = g.rank(method="max") / g.transform("size")
+    ref = ref.sort_values(["k", "o"]).reset_index(drop=True)
+    np.testing.assert_allclose(got["pr"], ref["pr"], atol=1e-12)
+    np.testing.assert_allclose(got["cd"], ref["cd"], atol=1e-12)
+    # nth_value: second row (by order, ties broken stably) per partition
+    assert got["nv"].notna().sum() > 0
