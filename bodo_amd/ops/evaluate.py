@@ -21,7 +21,8 @@ from ..core.table import Table
 from ..core.types import DType, TypeKind
 from ..plan.expr import (
     BinOp, BoolOp, Case, Cast, ColRef, Cmp, Const, DtField, Expr, IsIn,
-    IsNull, Not, RoundExpr, ScalarSubquery, SemiJoinIn, StrOp, UdfMap,
+    IsNull, ListOp, Not, RoundExpr, ScalarSubquery, SemiJoinIn, StrOp,
+    UdfMap,
 )
 
 
@@ -228,6 +229,30 @@ class _Evaluator:
     def visit_Cast(self, e: Cast) -> Column:
         a = self.visit(e.operand)
         return cast_column(a, e.to)
+
+    def visit_ListOp(self, e: ListOp) -> Column:
+        a = self.visit(e.operand)
+        if a.dtype.kind != TypeKind.LIST:
+            raise TypeError(f"list op {e.op} on {a.dtype}")
+        off = a.offsets
+        lens = off[1:] - off[:-1]
+        if e.op == "len":
+            return Column(bt.int64, lens, a.mask)
+        if e.op == "get":
+            from . import gather as _g
+
+            i = int(e.arg)
+            idx_in = lens + i if i < 0 else torch.full_like(lens, i)
+            ok = (idx_in >= 0) & (idx_in < lens)
+            if a.mask is not None:
+                ok = ok & a.mask
+            child_idx = (off[:-1] + idx_in).clamp(
+                min=0, max=max(len(a.child) - 1, 0))
+            vals = _g(a.child, child_idx)
+            mask = ok if vals.mask is None else (vals.mask & ok)
+            return Column(vals.dtype, vals.data, mask, vals.offsets,
+                          vals.dictionary, len(a))
+        raise NotImplementedError(f"list op {e.op}")
 
     def visit_StrOp(self, e: StrOp) -> Column:
         a = self.visit(e.operand)

@@ -15,7 +15,7 @@ from ..parallel import comm
 from ..plan import nodes as pn
 from ..plan.expr import (
     BinOp, BoolOp, Case, Cast, Cmp, ColRef, Const, DtField, Expr, IsIn,
-    IsNull, Not, StrOp, UdfMap, as_expr,
+    IsNull, ListOp, Not, StrOp, UdfMap, as_expr,
 )
 
 
@@ -210,6 +210,10 @@ class BodoSeries:
     @property
     def str(self):
         return _StrAccessor(self)
+
+    @property
+    def list(self):
+        return _ListAccessor(self)
 
     # ------------------------------------------------------------------
     # reductions (lazy scalars: the plan executes on first value access —
@@ -837,6 +841,22 @@ def _pd_dtype_to_bodo(dtype):
     if nd.kind == "M":
         return bt.timestamp_ns
     return bt.from_numpy_dtype(nd)
+
+
+class _ListAccessor:
+    """Series.list over LIST columns (pandas ArrowDtype .list surface;
+    reference role: array_item element kernels)."""
+
+    def __init__(self, s: BodoSeries):
+        self._s = s
+
+    def len(self):
+        return self._s._wrap(ListOp(self._s._expr, "len"), None)
+
+    def get(self, i):
+        return self._s._wrap(ListOp(self._s._expr, "get", int(i)), None)
+
+    __getitem__ = get
 
 
 class _SplitResult:

@@ -189,6 +189,22 @@ class UdfMap(Expr):
 
 
 @dataclass(frozen=True)
+class ListOp(Expr):
+    """LIST-column operation: len / get(i) (reference role: array_item
+    kernels + BodoSQL ARRAY_SIZE/GET)."""
+
+    operand: Expr
+    op: str
+    arg: Any = None
+
+    def children(self):
+        return (self.operand,)
+
+    def with_children(self, *ch):
+        return ListOp(ch[0], self.op, self.arg)
+
+
+@dataclass(frozen=True)
 class StrOp(Expr):
     """String method (pandas .str surface); kwargs as a tuple of pairs so
     the node stays hashable."""

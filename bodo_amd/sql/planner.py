@@ -1462,6 +1462,15 @@ class Planner:
                                 ex.BinOp("floordiv",
                                          ex.BinOp("sub", d, ex.Const(1)),
                                          ex.Const(7)), ex.Const(1))
+            if name == "array_size":
+                from ..plan.expr import ListOp as _LO
+
+                return _LO(self.expr(e.args[0], scope), "len")
+            if name == "get" and len(e.args) == 2:
+                from ..plan.expr import ListOp as _LO
+
+                return _LO(self.expr(e.args[0], scope), "get",
+                           int(e.args[1].value))
             if name in ("json_extract_path_text", "get_path"):
                 import json as _json
 

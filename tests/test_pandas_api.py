@@ -588,3 +588,25 @@ def test_str_findall_extract_rsplit():
     exp3 = df.s.str.rsplit("b")
     assert [None if v is None else list(v) for v in got3] == \
         [None if not isinstance(v, list) else v for v in exp3]
+
+
+def test_list_accessor_len_get():
+    df = pd.DataFrame({"l": pd.Series([[1, 2], [3], [], None, [4, 5, 6]])})
+    b = bpd.from_pandas(df)
+    ln = b.l.list.len().to_pandas()
+    assert [None if pd.isna(v) else int(v) for v in ln] == [2, 1, 0, None, 3]
+    g0 = b.l.list.get(0).to_pandas()
+    assert [None if pd.isna(v) else int(v) for v in g0] == [1, 3, None, None, 4]
+    gm1 = b.l.list[-1].to_pandas()
+    assert [None if pd.isna(v) else int(v) for v in gm1] == \
+        [2, 3, None, None, 6]
+    # SQL surface
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql("select array_size(l) as n, get(l, 1) as e from t"
+                 ).to_pandas()
+    assert [None if pd.isna(v) else int(v) for v in out["n"]] == \
+        [2, 1, 0, None, 3]
+    assert [None if pd.isna(v) else int(v) for v in out["e"]] == \
+        [2, None, None, None, 5]
