@@ -119,6 +119,10 @@ def _dispatch(msg, ex, comm, cfg):
             total = sum(comm.allgather_obj(len(shard)))
             return {"kind": "frame", "res_id": res_id,
                     "names": list(shard.names), "length": total}
+        from bodo_amd.pandas.scalar import BodoScalar
+
+        if isinstance(res, BodoScalar):
+            res = res.value  # materialize worker-side lazy scalars
         return {"kind": "value", "value": res}
     raise ValueError(f"unknown command {cmd}")
 
