@@ -200,7 +200,13 @@ def dict_encode_strings(tbl: pa.Table, threshold: float = 0.7,
         want.append(w)
     from ..parallel import comm
 
-    if comm.initialized() and comm.get_world_size() > 1:
+    has_string_cols = any(
+        pa.types.is_string(f.type) or pa.types.is_large_string(f.type)
+        for f in tbl.schema)
+    # the vote is a COLLECTIVE: run it only when string columns exist, so
+    # schemaless/numeric from_pandas calls never join a collective (uneven
+    # per-rank call counts would deadlock the sequence otherwise)
+    if has_string_cols and comm.initialized() and comm.get_world_size() > 1:
         votes = comm.allgather_obj(want)
         want = [all(v[i] for v in votes) for i in range(len(want))]
     if not any(want):

@@ -724,3 +724,49 @@ def test_dist_sort_float_key_device_partition():
             exp = df.sort_values("v", ascending=asc,
                                  na_position=na).reset_index(drop=True)
             pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def _q_incremental_shuffle(bpd, rank, payload):
+    import numpy as np
+
+    from bodo_amd.core.table import Table
+    from bodo_amd.parallel.incremental import IncrementalShuffle
+
+    rng = np.random.default_rng(100 + rank)
+    received = []
+    st = IncrementalShuffle(["k"], received.append, threshold=1 << 12,
+                            cadence=3)
+    # UNEVEN batch counts across ranks: rank 0 sends 11 morsels, rank 1
+    # sends 4 — termination consensus must drain both
+    import pyarrow as pa
+
+    n_batches = 11 if rank == 0 else 4
+    total = 0
+    for i in range(n_batches):
+        at = pa.table({"k": rng.integers(0, 10, 50),
+                       "v": rng.random(50)})
+        st.append(Table.from_arrow(at))  # from_arrow: no collective vote
+        total += 50
+    st.finish()
+    import pandas as pd2
+
+    mine = pd2.concat([t.to_pandas() for t in received]) if received \
+        else pd2.DataFrame({"k": [], "v": []})
+    # every received key must hash to this rank
+    from bodo_amd import ops as _ops
+    from bodo_amd.core.column import Column
+
+    if len(mine):
+        h = _ops.hash_columns([Column.from_numpy(
+            mine["k"].to_numpy())]).numpy()
+        assert ((h % 2 + 2) % 2 == rank).all()
+    counts = sorted(mine["k"].value_counts().to_dict().items())
+    return {"rows": len(mine), "counts": counts, "sent": total}
+
+
+def test_dist_incremental_shuffle_uneven_ranks():
+    """Cadenced-round incremental shuffle with uneven morsel counts and
+    threshold flushes (reference: IncrementalShuffleState + Ibarrier
+    consensus, streaming/_shuffle.h:777)."""
+    out = run_dist(_q_incremental_shuffle, {})
+    assert out["rows"] > 0
