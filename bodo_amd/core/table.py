@@ -200,8 +200,12 @@ def dict_encode_strings(tbl: pa.Table, threshold: float = 0.7,
         want.append(w)
     from ..parallel import comm
 
+    # SPMD-symmetric condition: empty shards infer pa.null for object
+    # columns, so null-typed columns must count as potential strings or the
+    # vote participation diverges across ranks
     has_string_cols = any(
         pa.types.is_string(f.type) or pa.types.is_large_string(f.type)
+        or pa.types.is_null(f.type)
         for f in tbl.schema)
     # the vote is a COLLECTIVE: run it only when string columns exist, so
     # schemaless/numeric from_pandas calls never join a collective (uneven
