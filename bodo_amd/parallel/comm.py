@@ -172,6 +172,8 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
     packed = ops.take_table(tbl, perm)
     str_idx = [i for i, c in enumerate(packed.columns)
                if c.dtype.kind == TypeKind.STRING]
+    list_idx = [i for i, c in enumerate(packed.columns)
+                if c.dtype.kind == TypeKind.LIST]
     dict_idx = [i for i, c in enumerate(packed.columns)
                 if c.dtype.kind == TypeKind.DICT]
     dev = _comm_device()
@@ -209,7 +211,7 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
     # (contiguous-split pack, reference: GpuShuffleManager
     #  cudf::contiguous_split + per-destination Isend, gpu_utils.cpp:96-122)
     fixed_idx = [i for i, c in enumerate(packed.columns)
-                 if c.dtype.kind != TypeKind.STRING]
+                 if c.dtype.kind not in (TypeKind.STRING, TypeKind.LIST)]
     if fixed_idx:
         bnd_l = bnd.tolist()
         dev0 = packed.columns[fixed_idx[0]].device if fixed_idx else None
@@ -274,6 +276,12 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
                                                 dicts_by_rank, recv_counts)
             else:
                 out_cols[i] = Column(c.dtype, data, new_mask, length=n_out)
+    # ---- lists: recursive varlen exchanges ----
+    bnd_l2 = bnd.tolist()
+    for i in list_idx:
+        out_cols[i] = _shuffle_list_column(
+            packed.columns[i], send_counts, recv_counts, bnd_l2,
+            bool(has_mask[i]))
     # ---- strings: per-column exchanges (offsets + bytes + mask) ----
     sj = 0
     for i, col in enumerate(packed.columns):
@@ -286,6 +294,66 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
             col, send_counts, recv_counts, byte_send, byte_recv,
             bool(has_mask[i]))
     return Table(packed.names, out_cols, n_out)
+
+
+def _exchange_count_vector(vals: List[int]) -> List[int]:
+    """all-to-all of a per-destination int vector (one element per rank)."""
+    dev = _comm_device()
+    send = torch.tensor(vals, dtype=torch.int64, device=dev)
+    recv = torch.empty_like(send)
+    dist.all_to_all_single(recv, send)
+    return [int(v) for v in recv.cpu().tolist()]
+
+
+def _shuffle_list_column(col: Column, send_counts, recv_counts,
+                         bnd_rows, has_mask: bool) -> Column:
+    """LIST column exchange: per-row lengths like strings, then the child
+    column recursively (child rows are contiguous per destination after the
+    row permutation).  Reference: nested-array shuffle in
+    bodo/libs/_shuffle.cpp."""
+    from ..core import types as _bt
+
+    n_out = sum(recv_counts)
+    off = col.offsets
+    lens = (off[1:] - off[:-1]).contiguous()
+    new_lens = alltoallv_tensor(lens, send_counts, recv_counts)
+    new_off = torch.zeros(n_out + 1, dtype=torch.int64, device=col.device)
+    torch.cumsum(new_lens, 0, out=new_off[1:])
+    # child rows per destination = offsets at the row boundaries
+    at = off[torch.tensor(bnd_rows, device=off.device)]
+    child_send = (at[1:] - at[:-1]).cpu().tolist()
+    child_recv = _exchange_count_vector(child_send)
+    child = col.child
+    child_has_mask = child.mask is not None
+    if child.dtype.kind == TypeKind.STRING:
+        # child string byte counts per destination
+        coff = child.offsets
+        cb = torch.zeros(len(bnd_rows), dtype=torch.int64)
+        cume = torch.cumsum(torch.tensor([0] + child_send), 0)
+        cat = coff[cume.to(coff.device)]
+        byte_send = (cat[1:] - cat[:-1]).cpu().tolist()
+        byte_recv = _exchange_count_vector(byte_send)
+        new_child = _shuffle_string_column(child, child_send, child_recv,
+                                           byte_send, byte_recv,
+                                           child_has_mask)
+    elif child.dtype.kind == TypeKind.LIST:
+        cume = torch.cumsum(torch.tensor([0] + child_send), 0).tolist()
+        new_child = _shuffle_list_column(child, child_send, child_recv,
+                                         cume, child_has_mask)
+    else:
+        data = alltoallv_tensor(child.data.contiguous(), child_send,
+                                child_recv)
+        cmask = None
+        if child_has_mask:
+            cmask = alltoallv_tensor(child.mask.contiguous(), child_send,
+                                     child_recv)
+        new_child = Column(child.dtype, data, cmask,
+                           dictionary=child.dictionary,
+                           length=sum(child_recv))
+    new_mask = _shuffle_mask(col, send_counts, recv_counts, has_mask)
+    out = Column(_bt.list_, None, new_mask, offsets=new_off, length=n_out)
+    out.child = new_child
+    return out
 
 
 def _shuffle_string_column(col: Column, send_counts, recv_counts,
@@ -367,7 +435,7 @@ def _pack_table(tbl: Table, dev: torch.device):
             pos += pad
         return off, u8.numel()
 
-    for c in tbl.columns:
+    def col_meta(c):
         m = {"dtype": c.dtype, "n": len(c)}
         if c.data is not None:
             m["data"] = add(c.data.to(dev))
@@ -378,7 +446,12 @@ def _pack_table(tbl: Table, dev: torch.device):
             m["mask"] = add(c.mask.to(dev))
         if c.dictionary is not None:
             m["dict"] = c.dictionary.to_pylist()
-        cols_meta.append(m)
+        if c.child is not None:
+            m["child"] = col_meta(c.child)
+        return m
+
+    for c in tbl.columns:
+        cols_meta.append(col_meta(c))
     buf = (torch.cat(segs) if segs
            else torch.zeros(0, dtype=torch.uint8, device=dev))
     meta = {"names": list(tbl.names), "cols": cols_meta, "nbytes": pos,
@@ -389,8 +462,7 @@ def _pack_table(tbl: Table, dev: torch.device):
 def _unpack_table(buf: torch.Tensor, meta: dict, device) -> Table:
     import pyarrow as pa
 
-    cols = []
-    for m in meta["cols"]:
+    def mk_col(m):
         kind = m["dtype"]
         data = mask = offsets = None
         dictionary = None
@@ -408,7 +480,13 @@ def _unpack_table(buf: torch.Tensor, meta: dict, device) -> Table:
             dictionary = pa.array(m["dict"], type=pa.large_string())
         col = Column(kind, data, mask, offsets=offsets, dictionary=dictionary,
                      length=m["n"])
-        cols.append(col)
+        if "child" in m:
+            col.child = mk_col(m["child"])
+        return col
+
+    cols = []
+    for m in meta["cols"]:
+        cols.append(mk_col(m))
     t = Table(meta["names"], cols, meta["length"])
     return t.to_device(device)
 

@@ -770,3 +770,55 @@ def test_dist_incremental_shuffle_uneven_ranks():
     consensus, streaming/_shuffle.h:777)."""
     out = run_dist(_q_incremental_shuffle, {})
     assert out["rows"] > 0
+
+
+def _q_list_shuffle(bpd, rank, payload):
+    b = bpd.from_pandas(payload["df"])
+    return b.groupby("k", as_index=False).agg(
+        c=bpd.NamedAgg("l", "count")).sort_values("k")
+
+
+def _q_list_explode_shuffle(bpd, rank, payload):
+    b = bpd.from_pandas(payload["df"])
+    return b.explode("l").groupby("k", as_index=False).agg(
+        s=bpd.NamedAgg("l", "sum")).sort_values("k")
+
+
+def test_dist_list_columns_shuffle():
+    """LIST columns survive distributed shuffles (recursive varlen
+    exchange; reference: nested-array shuffle in _shuffle.cpp)."""
+    rng = np.random.default_rng(91)
+    n = 600
+    df = pd.DataFrame({"k": rng.integers(0, 12, n)})
+    df["l"] = pd.Series([list(map(int, rng.integers(0, 9, rng.integers(0, 5))))
+                         for _ in range(n)], dtype=object)
+    # window/groupby path shuffles the frame incl. the list payload
+    got = run_dist(_q_list_explode_shuffle, {"df": df}).reset_index(drop=True)
+    exp = df.explode("l").groupby("k", as_index=False).agg(
+        s=("l", "sum")).sort_values("k").reset_index(drop=True)
+    assert got["k"].tolist() == exp["k"].tolist()
+    assert [float(v) for v in got["s"]] == [float(v) for v in exp["s"]]
+
+
+def _q_list_allgather(bpd, rank, payload):
+    from bodo_amd.core.table import Table
+    from bodo_amd.parallel import comm
+
+    df = payload["df"]
+    half = len(df) // 2
+    t = Table.from_pandas(df.iloc[rank * half:(rank + 1) * half]
+                          .reset_index(drop=True))
+    rep = comm.allgather_table(t)
+    return rep.to_pandas()
+
+
+def test_dist_list_allgather():
+    rng = np.random.default_rng(93)
+    n = 100
+    df = pd.DataFrame({"k": rng.integers(0, 5, n)})
+    df["l"] = pd.Series([list(map(int, rng.integers(0, 9, rng.integers(0, 4))))
+                         for _ in range(n)], dtype=object)
+    got = run_dist(_q_list_allgather, {"df": df})
+    exp = df.iloc[:n // 2 * 2].reset_index(drop=True)
+    assert got["k"].tolist() == exp["k"].tolist()
+    assert [list(v) for v in got["l"]] == [list(v) for v in exp["l"]]
