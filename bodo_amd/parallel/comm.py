@@ -328,7 +328,6 @@ def _shuffle_list_column(col: Column, send_counts, recv_counts,
     if child.dtype.kind == TypeKind.STRING:
         # child string byte counts per destination
         coff = child.offsets
-        cb = torch.zeros(len(bnd_rows), dtype=torch.int64)
         cume = torch.cumsum(torch.tensor([0] + child_send), 0)
         cat = coff[cume.to(coff.device)]
         byte_send = (cat[1:] - cat[:-1]).cpu().tolist()
