@@ -430,7 +430,7 @@ def _exec_distinct(node: pn.Distinct, ctx) -> Table:
 
 def _exec_explode(node: pn.Explode, ctx) -> Table:
     child = _exec(node.child, ctx)
-    return ops.explode_table(child, node.column)
+    return ops.explode_table(child, node.column, pos_name=node.pos)
 
 
 def _exec_sample(node: pn.Sample, ctx) -> Table:

@@ -90,7 +90,7 @@ def _gather_list(col: Column, idx: torch.Tensor) -> Column:
     return out
 
 
-def explode_table(tbl: Table, column: str) -> Table:
+def explode_table(tbl: Table, column: str, pos_name=None) -> Table:
     """pandas explode semantics: each list element becomes a row; empty or
     null lists produce one row with a null value (reference role:
     bodo/libs/_lateral.cpp FLATTEN)."""
@@ -130,6 +130,11 @@ def explode_table(tbl: Table, column: str) -> Table:
         else:
             names.append(nm)
             cols.append(gather(c, row_idx))
+    if pos_name is not None:
+        names.append(pos_name)
+        pc = Column(_bt.int64, pos_in_row,
+                    real.clone() if (~real).any() else None)
+        cols.append(pc)
     return Table(names, cols, total)
 
 

@@ -166,15 +166,17 @@ class Distinct(PlanNode):
 @dataclass(frozen=True)
 class Explode(PlanNode):
     """Each element of a LIST column becomes a row (pandas explode /
-    LATERAL FLATTEN; reference: bodo/libs/_lateral.cpp)."""
+    LATERAL FLATTEN; reference: bodo/libs/_lateral.cpp).  pos names an
+    optional element-index output column (FLATTEN .index)."""
     child: PlanNode
     column: str
+    pos: Optional[str] = None
 
     def children(self):
         return (self.child,)
 
     def with_children(self, *ch):
-        return Explode(ch[0], self.column)
+        return Explode(ch[0], self.column, self.pos)
 
     def out_columns(self):
         return self.child.out_columns()

@@ -193,6 +193,7 @@ class TableRef:
     name: str
     alias: Optional[str]
     subquery: Optional["Query"] = None
+    flatten: Any = None  # LATERAL FLATTEN(input => expr)
 
 
 @dataclass
@@ -300,6 +301,12 @@ class Parser:
                     break
                 if t.kind == "op" and t.value == ",":
                     self.i += 1
+                    nt = self.peek()
+                    if nt and nt.kind == "id" and \
+                            nt.value.lower() == "lateral":
+                        joins.append(JoinClause(
+                            "lateral", self.parse_lateral_flatten(), None))
+                        continue
                     joins.append(JoinClause("cross", self.parse_table_ref(), None))
                     continue
                 kind = None
@@ -411,6 +418,31 @@ class Parser:
                   order_by, limit, grouping_sets, qualify)
         q.limit_offset = limit_offset
         return q
+
+    def parse_lateral_flatten(self) -> TableRef:
+        """, LATERAL FLATTEN([input =>] expr) [AS] alias  (Snowflake
+        surface; reference: bodo/libs/_lateral.cpp FLATTEN)."""
+        self.next()  # lateral
+        fn = self.next()
+        if fn.kind != "id" or fn.value.lower() != "flatten":
+            raise SyntaxError("expected FLATTEN after LATERAL")
+        self.expect_op("(")
+        nt = self.peek()
+        if nt and nt.kind == "id" and nt.value.lower() == "input":
+            self.next()
+            self.expect_op("=")
+            self.expect_op(">")
+        e = self.parse_expr()
+        self.expect_op(")")
+        alias = None
+        if self.accept_kw("as"):
+            alias = self.next().value
+        else:
+            nt = self.peek()
+            if nt and nt.kind == "id" \
+                    and nt.value.lower() not in _NON_ALIAS_IDS:
+                alias = self.next().value
+        return TableRef("__flatten", alias or "f", flatten=e)
 
     def parse_table_ref(self) -> TableRef:
         if self.accept_op("("):

@@ -114,6 +114,22 @@ class Planner:
         join_ons = [None] + [j.on for j in q.joins]
 
         for idx, tr in enumerate(refs):
+            if getattr(tr, "flatten", None) is not None:
+                # , LATERAL FLATTEN(input => expr) f  ->  Explode over the
+                # accumulated plan (reference: _lateral.cpp FLATTEN)
+                assert plan is not None, "LATERAL FLATTEN needs a base table"
+                vcol = self._uniq("__flat_v")
+                icol = self._uniq("__flat_i")
+                keep = sorted(joined_cols)
+                exprs = tuple(ex.ColRef(c) for c in keep) + (
+                    self.expr(tr.flatten, scope),)
+                plan = pn.Projection(plan, tuple(keep) + (vcol,), exprs)
+                plan = pn.Explode(plan, vcol, pos=icol)
+                alias = tr.alias or "f"
+                scope.add(alias, "value", vcol)
+                scope.add(alias, "index", icol)
+                joined_cols |= {vcol, icol}
+                continue
             sub_plan, sub_cols = self._table_plan(tr, scope)
             if plan is None:
                 plan = sub_plan
@@ -692,6 +708,22 @@ class Planner:
         kinds = ["base"] + [j.kind for j in q.joins]
         ons = [None] + [j.on for j in q.joins]
         for idx, tr in enumerate(refs):
+            if getattr(tr, "flatten", None) is not None:
+                # , LATERAL FLATTEN(input => expr) f  ->  Explode over the
+                # accumulated plan (reference: _lateral.cpp FLATTEN)
+                assert plan is not None, "LATERAL FLATTEN needs a base table"
+                vcol = self._uniq("__flat_v")
+                icol = self._uniq("__flat_i")
+                keep = sorted(joined_cols)
+                exprs = tuple(ex.ColRef(c) for c in keep) + (
+                    self.expr(tr.flatten, scope),)
+                plan = pn.Projection(plan, tuple(keep) + (vcol,), exprs)
+                plan = pn.Explode(plan, vcol, pos=icol)
+                alias = tr.alias or "f"
+                scope.add(alias, "value", vcol)
+                scope.add(alias, "index", icol)
+                joined_cols |= {vcol, icol}
+                continue
             sub_plan, sub_cols = self._table_plan(tr, scope)
             if plan is None:
                 plan = sub_plan

@@ -710,3 +710,30 @@ def test_window_percent_rank_cume_dist_nth():
     np.testing.assert_allclose(got["cd"], ref["cd"], atol=1e-12)
     # nth_value: second row (by order, ties broken stably) per partition
     assert got["nv"].notna().sum() > 0
+
+
+def test_sql_lateral_flatten():
+    """, LATERAL FLATTEN(input => col) f  (reference: _lateral.cpp)."""
+    df = pd.DataFrame({"k": [1, 2, 3],
+                       "l": pd.Series([[10, 20], [30], []])})
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("""
+        select t.k, f.value as v, f.index as i
+        from t, lateral flatten(input => t.l) f
+        order by k, i
+    """).to_pandas().reset_index(drop=True)
+    # empty lists produce a null element row (pandas explode semantics)
+    assert got["k"].tolist() == [1, 1, 2, 3]
+    assert [None if pd.isna(v) else int(v) for v in got["v"]] == \
+        [10, 20, 30, None]
+    assert [None if pd.isna(v) else int(v) for v in got["i"]] == [0, 1, 0, None]
+    # aggregate over flattened values
+    got2 = bc.sql("""
+        select k, sum(f.value) as s
+        from t, lateral flatten(t.l) f group by k order by k
+    """).to_pandas()
+    # engine contract is pandas semantics: sum of an all-null group is 0
+    assert [None if pd.isna(v) else int(v) for v in got2["s"]] == \
+        [30, 30, 0]
