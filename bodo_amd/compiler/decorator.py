@@ -48,6 +48,17 @@ def _specialize(fn):
     clone = types.FunctionType(fn.__code__, new_g, fn.__name__,
                                fn.__defaults__, fn.__closure__)
     clone.__kwdefaults__ = fn.__kwdefaults__
+    # parfor-style vectorization of elementwise prange/range loops
+    # (reference: series_pass parfors + distributed_pass._run_parfor)
+    if fn.__closure__ is None:
+        from .vectorize import vectorize_fn
+
+        try:
+            vec = vectorize_fn(fn, new_g)
+        except Exception:
+            vec = None
+        if vec is not None:
+            return vec
     return clone
 
 

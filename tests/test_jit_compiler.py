@@ -131,3 +131,59 @@ def test_prange_and_dist_reduce():
 
     n = 10000
     assert f(n) == sum(i * i for i in range(n))
+
+
+def test_prange_loop_vectorization():
+    """Elementwise prange loops rewrite to whole-array device expressions;
+    accumulators become global reductions (parfor lowering, reference:
+    distributed_pass._run_parfor)."""
+    from bodo_amd.compiler.distarray import DistArray
+
+    @bodo_amd.jit
+    def f(a, n):
+        out = np.empty(n)
+        for i in range(n):
+            out[i] = a[i] * 2.0 + i
+        acc = 0.0
+        for i in range(n):
+            acc += a[i] * a[i]
+        return out.sum() + acc
+
+    rng = np.random.default_rng(2)
+    n = 50_000
+    a = rng.random(n)
+    got = f(a, n)
+    exp = (a * 2.0 + np.arange(n)).sum() + (a * a).sum()
+    assert abs(got - exp) / abs(exp) < 1e-12
+
+
+def test_prange_nonvectorizable_falls_back():
+    @bodo_amd.jit
+    def g(n):
+        acc = 0
+        for i in bodo_amd.prange(n):
+            if i % 3 == 0:  # branch in body: stays a scalar SPMD loop
+                acc += i
+        return bodo_amd.dist_reduce(acc, "sum")
+
+    n = 3000
+    assert g(n) == sum(i for i in range(n) if i % 3 == 0)
+
+
+def test_prange_index_free_body_not_vectorized():
+    """acc += f() without touching the index must run once per iteration."""
+    calls = {"n": 0}
+
+    def bump():
+        calls["n"] += 1
+        return 1
+
+    @bodo_amd.jit
+    def h(n, bump):
+        acc = 0
+        for i in range(n):
+            acc += bump()
+        return acc
+
+    assert h(5, bump) == 5
+    assert calls["n"] == 5
