@@ -87,8 +87,8 @@ def _vectorize_loop(node: ast.For) -> Optional[List[ast.stmt]]:
             # out[i] = expr  ->  out = expr_vec
             rw = _ElemRewriter(ivar)
             val = rw.visit(copy.deepcopy(stmt.value))
-            if not rw.ok:
-                return None
+            if not rw.ok or not (rw.used_elem or rw.used_index):
+                return None  # constant/side-effect bodies stay loops
             needs_idx |= rw.used_index
             out.append(ast.Assign(
                 targets=[ast.Name(id=stmt.targets[0].value.id,
