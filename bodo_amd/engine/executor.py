@@ -100,8 +100,11 @@ def _exec(node: pn.PlanNode, ctx: ExecutionContext) -> Table:
     from ..utils import query_profile as qp
     from ..utils import tracing
 
+    from ..utils import roctx
+
     name = type(node).__name__
-    with tracing.Event(f"exec.{name}"), qp.OpTimer(name) as t:
+    with tracing.Event(f"exec.{name}"), roctx.Range(f"bodo.{name}"), \
+            qp.OpTimer(name) as t:
         out = h(node, ctx)
         from .. import config as _cfg
 

@@ -281,3 +281,20 @@ def test_fft_distarray():
     got = bodo_amd.fft.rfft(d)
     np.testing.assert_allclose(got.to_numpy(), np.fft.rfft(x), rtol=1e-8,
                                atol=1e-8)
+
+
+def test_roctx_disabled_noop_and_enabled_fallback(monkeypatch):
+    import importlib
+
+    from bodo_amd.utils import roctx
+
+    # disabled: no library load, no-ops
+    with roctx.Range("x"):
+        pass
+    # enabled: loads the ROCm roctx library (present in the image) and the
+    # push/pop pair must not raise
+    monkeypatch.setenv("BODO_AMD_ROCTX", "1")
+    importlib.reload(roctx)
+    with roctx.Range("bodo.test"):
+        roctx.range_push("inner")
+        roctx.range_pop()
