@@ -540,6 +540,11 @@ def _shift_column_device(col: Column, k: int, ctx) -> Optional[Column]:
     data, mask = col.data, col.mask
     if k == 0:
         return col
+    if n == 0:
+        # still participate in the halo allgather (SPMD sequence)
+        if ctx.world > 1:
+            _gather_halo(data, mask, k, ctx)
+        return col
     halo_d, halo_m = (data[:0], None)
     if ctx.world > 1:
         halo_d, halo_m = _gather_halo(data, mask, k, ctx)

@@ -117,3 +117,26 @@ def test_wide_frame():
     out = b[b.c1 > 2][["c0", "c30", "c59"]].to_pandas()
     exp = df[df.c1 > 2][["c0", "c30", "c59"]].reset_index(drop=True)
     pd.testing.assert_frame_equal(out, exp, check_dtype=False)
+
+
+def test_empty_and_single_row_device_ops():
+    """Empty/1-row shards through the new device paths (shift/fill/rolling/
+    explode/list/decimal) must not fault."""
+    from decimal import Decimal
+
+    e = pd.DataFrame({"x": pd.Series([], dtype="float64")})
+    be = bpd.from_pandas(e)
+    assert len(be.x.shift(2).to_pandas()) == 0
+    assert len(be.x.ffill().to_pandas()) == 0
+    assert len(be.x.rolling(3).mean().to_pandas()) == 0
+    one = pd.DataFrame({"x": [2.5],
+                        "l": pd.Series([[1, 2]]),
+                        "m": pd.Series([Decimal("1.50")])})
+    b1 = bpd.from_pandas(one)
+    assert pd.isna(b1.x.shift(1).to_pandas().iloc[0])
+    assert b1.explode("l").to_pandas()["l"].tolist() == [1, 2]
+    assert float(b1.m.sum()) == 1.5
+    empty_list = pd.DataFrame({"l": pd.Series([], dtype=object)})
+    # all-empty object column infers null type; explode is a no-op frame
+    bl = bpd.from_pandas(empty_list)
+    assert len(bl.to_pandas()) == 0
