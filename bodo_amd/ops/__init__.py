@@ -509,6 +509,10 @@ def _sort_key_tensor(c: Column, asc: bool, na_position: str) -> torch.Tensor:
 def reduce_column(col: Column, func: str):
     """Local partial reduction -> dict of partials (combined across ranks by
     the executor)."""
+    if col.dtype.kind == TypeKind.DECIMAL128:
+        from .evaluate import decimal_to_float
+
+        col = decimal_to_float(col)  # scalar reductions report values
     data, mask = col.data, col.mask
     if col.dtype.is_float:
         valid = ~torch.isnan(data)
