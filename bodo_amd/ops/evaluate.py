@@ -780,6 +780,19 @@ def _dt_field_cpu(a: Column, fld: str) -> Column:
 def cast_column(a: Column, to: DType) -> Column:
     if a.dtype == to:
         return a
+    if a.dtype.kind == TypeKind.DECIMAL128:
+        # scaled-int storage: cast through the VALUE, not the raw int
+        f = decimal_to_float(a)
+        if to.kind in (TypeKind.FLOAT64, TypeKind.FLOAT32):
+            out = f.data.to(torch.float32) if to.kind == TypeKind.FLOAT32 \
+                else f.data
+            return Column(to, out, f.mask)
+        return cast_column(f, to)
+    if to.kind == TypeKind.DECIMAL128:
+        # value -> scaled int (round-half-even like arrow casts)
+        scale = 10 ** to.scale
+        v = a.data.double() * scale
+        return Column(to, torch.round(v).to(torch.int64), a.mask)
     if to.kind == TypeKind.STRING or a.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
         # host path for string casts
         ser = a.to_pandas()

@@ -301,6 +301,10 @@ class BodoSeries:
         prefer .sample(...)-based estimates; groupby('...').median() stays
         fully distributed."""
         ser = self.dropna().to_pandas()
+        from decimal import Decimal as _D
+
+        if len(ser) and isinstance(ser.iloc[0], _D):
+            ser = ser.astype(float)
         if hasattr(q, "__len__"):
             return pd.Series([ser.quantile(x, interpolation=interpolation)
                               for x in q], index=list(q), name=self.name)
