@@ -348,6 +348,16 @@ class _Evaluator:
 
             return Column.from_arrow(pa.Array.from_pandas(ser), self.device)
         thens = [self.visit(t) for t in e.thens]
+        # decimal branches combine by VALUE unless every branch shares the
+        # decimal scale (scaled-int leak otherwise)
+        branches = [other] + thens
+        decs = [c for c in branches if c.dtype.kind == TypeKind.DECIMAL128]
+        if decs and (len(decs) != len(branches) or
+                     len({c.dtype.scale for c in decs}) != 1):
+            conv = [decimal_to_float(c)
+                    if c.dtype.kind == TypeKind.DECIMAL128 else c
+                    for c in branches]
+            other, thens = conv[0], conv[1:]
         # promote across all branches (int + float/null branches -> float)
         res_torch = other.data.dtype
         for t in thens:
@@ -377,10 +387,14 @@ class _Evaluator:
 # ----------------------------------------------------------------------
 
 def infer_const_dtype(v) -> DType:
+    import decimal as _dec
+
     from ..pandas.scalar import BodoScalar
 
     if isinstance(v, BodoScalar):
         v = v.value
+    if isinstance(v, _dec.Decimal):
+        return bt.float64  # decimal consts compare/combine by value
     if isinstance(v, bool):
         return bt.boolean
     if isinstance(v, int):
@@ -399,10 +413,14 @@ def infer_const_dtype(v) -> DType:
 
 
 def normalize_const(v, dtype: DType):
+    import decimal as _dec
+
     from ..pandas.scalar import BodoScalar
 
     if isinstance(v, BodoScalar):
         v = v.value
+    if isinstance(v, _dec.Decimal):
+        v = float(v)
     if v is None:
         return float("nan") if dtype.is_float else None
     if dtype.kind == TypeKind.TIMESTAMP_NS:
