@@ -211,8 +211,29 @@ def dict_encode_strings(tbl: pa.Table, threshold: float = 0.7,
     # schemaless/numeric from_pandas calls never join a collective (uneven
     # per-rank call counts would deadlock the sequence otherwise)
     if has_string_cols and comm.initialized() and comm.get_world_size() > 1:
-        votes = comm.allgather_obj(want)
-        want = [all(v[i] for v in votes) for i in range(len(want))]
+        votes = comm.allgather_obj((want, tbl.schema))
+        want = [all(v[0][i] for v in votes) for i in range(len(want))]
+        # schema consensus: EMPTY shards infer pa.null for object columns
+        # (strings/decimals/lists) — adopt a non-empty rank's type so every
+        # rank presents identical column kinds to the collective sequence
+        # (reference analog: the plan's empty_data schema, plan.py:44)
+        local_fields = list(tbl.schema)
+        fixed = []
+        changed = False
+        for i, f in enumerate(local_fields):
+            t_use = f.type
+            if pa.types.is_null(t_use):
+                for _, sch in votes:
+                    if not pa.types.is_null(sch.field(i).type):
+                        t_use = sch.field(i).type
+                        break
+            if t_use != f.type:
+                changed = True
+                fixed.append(tbl.column(i).cast(t_use))
+            else:
+                fixed.append(tbl.column(i))
+        if changed:
+            tbl = pa.table(dict(zip(tbl.column_names, fixed)))
     if not any(want):
         return tbl
     new_cols = []

@@ -822,3 +822,51 @@ def test_dist_list_allgather():
     exp = df.iloc[:n // 2 * 2].reset_index(drop=True)
     assert got["k"].tolist() == exp["k"].tolist()
     assert [list(v) for v in got["l"]] == [list(v) for v in exp["l"]]
+
+
+def _q_kitchen_sink_shuffle(bpd, rank, payload):
+    from bodo_amd.core.table import Table
+    from bodo_amd.parallel import comm
+    import torch
+
+    df = payload["df"]
+    if rank == 1 and payload.get("empty1"):
+        df = df.iloc[:0]
+    t = Table.from_pandas(df.reset_index(drop=True))
+    n = len(t)
+    part = torch.arange(n) % 2 if n else torch.zeros(0, dtype=torch.int64)
+    out = comm.shuffle_table(t, part)
+    back = out.to_pandas()
+    total = sum(comm.allgather_obj(len(back)))
+    return {"total": total, "cols": list(back.columns)}
+
+
+def test_dist_shuffle_all_column_kinds():
+    """Packed shuffle across every column kind (ints, floats with NaN,
+    masked ints, bool, decimal, date, timestamp, dict, plain string, list)
+    incl. a zero-row rank."""
+    from decimal import Decimal
+
+    rng = np.random.default_rng(101)
+    n = 200
+    df = pd.DataFrame({
+        "i64": rng.integers(-5, 5, n),
+        "f64": np.where(rng.random(n) < 0.2, np.nan, rng.random(n)),
+        "b": rng.integers(0, 2, n).astype(bool),
+        "s_plain": np.array([f"v{i}" for i in range(n)], dtype=object),
+        "s_dict": rng.choice(["aa", "bb"], n),
+        "ts": pd.to_datetime(1_600_000_000_000_000_000
+                             + rng.integers(0, 10**15, n)),
+    })
+    df["mask_i"] = pd.array(rng.integers(0, 9, n), dtype="Int64")
+    df.loc[rng.random(n) < 0.2, "mask_i"] = pd.NA
+    df["dec"] = pd.Series([Decimal(int(v)) / 100
+                           for v in rng.integers(-10**4, 10**4, n)])
+    df["lst"] = pd.Series([list(map(int, rng.integers(0, 5,
+                                                      rng.integers(0, 3))))
+                           for _ in range(n)], dtype=object)
+    for empty1 in (False, True):
+        out = run_dist(_q_kitchen_sink_shuffle,
+                       {"df": df, "empty1": empty1})
+        exp_total = 2 * len(df) if not empty1 else len(df)
+        assert out["total"] == exp_total, (empty1, out)
