@@ -134,3 +134,40 @@ def test_comptroller_divides_budget(monkeypatch):
         with comptroller.operator():
             assert ooc.budget_bytes(dev) == 500000
     assert comptroller.peak() >= 2
+
+
+def test_streaming_join_with_ooc_budget(tmp_path, monkeypatch):
+    """Forced morsel streaming THROUGH a join with a tiny OOC budget: the
+    SF1000 execution shape (streamed probe + partition-split local ops)."""
+    import bodo_amd.config as cfg
+
+    monkeypatch.setenv("BODO_AMD_OOC_BYTES", "200000")
+    rng = np.random.default_rng(7)
+    n = 120_000
+    li = pd.DataFrame({"okey": rng.integers(0, 3000, n),
+                       "qty": rng.random(n) * 50,
+                       "price": rng.random(n) * 1000})
+    orders = pd.DataFrame({"o_key": np.arange(3000),
+                           "flag": rng.choice(["A", "B", "C"], 3000)})
+    p = str(tmp_path / "li.parquet")
+    li.to_parquet(p, row_group_size=8000)
+    old = cfg.STREAMING, cfg.STREAM_BATCH_SIZE
+    cfg.STREAMING = "1"
+    cfg.STREAM_BATCH_SIZE = 10_000
+    try:
+        b = bpd.read_parquet(p)
+        o = bpd.from_pandas(orders)
+        got = b[b.qty < 40].merge(o, left_on="okey", right_on="o_key") \
+            .groupby("flag", as_index=False).agg(
+            s=bpd.NamedAgg("price", "sum"),
+            c=bpd.NamedAgg("qty", "count")).sort_values("flag").to_pandas()
+        exp = li[li.qty < 40].merge(orders, left_on="okey",
+                                    right_on="o_key").groupby(
+            "flag", as_index=False).agg(
+            s=("price", "sum"), c=("qty", "count")).sort_values(
+            "flag").reset_index(drop=True)
+        got["flag"] = got["flag"].astype(str)
+        pd.testing.assert_frame_equal(got.reset_index(drop=True), exp,
+                                      check_dtype=False, rtol=1e-9)
+    finally:
+        cfg.STREAMING, cfg.STREAM_BATCH_SIZE = old
