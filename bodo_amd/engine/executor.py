@@ -170,8 +170,8 @@ def _exec_filter(node: pn.Filter, ctx) -> Table:
     return ev.eval_filter(node.cond, child)
 
 
-SINGLE_PHASE_AGGS = {"median", "nunique", "var", "std", "quantile",
-                     "skew", "kurt", "sem", "mode"}
+SINGLE_PHASE_AGGS = {"median", "nunique", "approx_nunique", "var", "std",
+                     "quantile", "skew", "kurt", "sem", "mode"}
 
 
 def _exec_aggregate(node: pn.Aggregate, ctx) -> Table:
@@ -1031,8 +1031,18 @@ def _finish_reduce(node: pn.Reduce, ctx, partials: dict) -> Table:
         parts = comm.allgather_obj(partials[out_name])
         val = _combine_reduce(parts, func)
         out_names.append(out_name)
-        arr = pd.Series([val])
-        out_cols.append(Column.from_numpy(arr.to_numpy(), ctx.device))
+        ser = pd.Series([val])
+        if ser.dtype == object:  # strings / None: go through arrow
+            import pyarrow as pa
+
+            arr = pa.Array.from_pandas(ser)
+            if pa.types.is_string(arr.type):
+                arr = arr.cast(pa.large_string())
+            elif pa.types.is_null(arr.type):
+                arr = arr.cast(pa.float64())
+            out_cols.append(Column.from_arrow(arr, ctx.device))
+        else:
+            out_cols.append(Column.from_numpy(ser.to_numpy(), ctx.device))
     return Table(out_names, out_cols, 1)
 
 
@@ -1050,14 +1060,59 @@ def _combine_reduce(partials: List[dict], func: str):
     if func == "max":
         vals = [p["max"] for p in partials if p["max"] is not None]
         return max(vals) if vals else None
-    if func in ("var", "std"):
+    if func in ("var", "std", "sem"):
         c = sum(p["count"] for p in partials)
         s = sum(p["sum"] for p in partials)
         ss = sum(p["sumsq"] for p in partials)
         if c < 2:
             return float("nan")
         var = (ss - s * s / c) / (c - 1)
-        return var if func == "var" else var ** 0.5
+        if func == "var":
+            return var
+        return var ** 0.5 if func == "std" else (var / c) ** 0.5
+    if func in ("kurt", "skew"):
+        # pandas-compatible bias-adjusted central moments from raw sums
+        c = sum(p["count"] for p in partials)
+        s1 = sum(p["sum"] for p in partials)
+        s2 = sum(p["sumsq"] for p in partials)
+        s3 = sum(p["sum3"] for p in partials)
+        s4 = sum(p["sum4"] for p in partials)
+        if (func == "skew" and c < 3) or (func == "kurt" and c < 4):
+            return float("nan")
+        mu = s1 / c
+        d2 = s2 - c * mu * mu
+        if d2 <= 0:
+            return float("nan")
+        m2 = d2 / c
+        if func == "skew":
+            d3 = s3 - 3 * mu * s2 + 2 * c * mu ** 3
+            g1 = (d3 / c) / m2 ** 1.5
+            return (c * (c - 1)) ** 0.5 / (c - 2) * g1
+        d4 = s4 - 4 * mu * s3 + 6 * mu * mu * s2 - 3 * c * mu ** 4
+        g2 = (d4 / c) / (m2 * m2) - 3.0
+        return ((c + 1) * g2 + 6) * (c - 1) / ((c - 2) * (c - 3))
+    if func in ("first", "last"):
+        seq = partials if func == "first" else list(reversed(partials))
+        for p in seq:
+            if p["has"]:
+                return p["val"]
+        return None
+    if func == "median":
+        import numpy as np
+
+        vs = [p["vals"] for p in partials if p["vals"] is not None]
+        if not vs:
+            return float("nan")
+        return float(np.median(np.concatenate(vs)))
+    if func == "mode":
+        counts: dict = {}
+        for p in partials:
+            for k, v in p["counts"].items():
+                counts[k] = counts.get(k, 0) + v
+        if not counts:
+            return None
+        best = max(counts.values())
+        return min(k for k, v in counts.items() if v == best)
     if func == "any":
         return any(p["any"] for p in partials)
     if func == "all":

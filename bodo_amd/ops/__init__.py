@@ -506,13 +506,54 @@ def _sort_key_tensor(c: Column, asc: bool, na_position: str) -> torch.Tensor:
 # reductions
 # ----------------------------------------------------------------------
 
+def _reduce_wrap_val(col: Column, v):
+    """Storage scalar -> user value (ns int -> Timestamp, days -> Timestamp);
+    reductions report VALUES, never storage encodings."""
+    if v is None:
+        return None
+    import pandas as pd
+
+    if col.dtype.kind == TypeKind.TIMESTAMP_NS:
+        return pd.Timestamp(int(v))
+    if col.dtype.kind == TypeKind.DATE32:
+        return pd.Timestamp(int(v) * 86_400_000_000_000)
+    return v
+
+
+# reductions where the partial depends on actual VALUES (strings must not
+# reduce over dictionary codes)
+_VALUE_REDUCES = {"min", "max", "first", "last", "mode", "median", "nunique"}
+
+
 def reduce_column(col: Column, func: str):
     """Local partial reduction -> dict of partials (combined across ranks by
-    the executor)."""
+    the executor's _combine_reduce).  Reference role: the parallel agg
+    combine in bodo/libs/_groupby_ftypes + distributed_api dist_reduce."""
     if col.dtype.kind == TypeKind.DECIMAL128:
         from .evaluate import decimal_to_float
 
         col = decimal_to_float(col)  # scalar reductions report values
+    if col.dtype.kind in (TypeKind.STRING, TypeKind.DICT) \
+            and func in _VALUE_REDUCES:
+        import pandas as pd
+
+        vals = pd.Series(
+            [v for v in col.to_arrow().to_pylist() if v is not None],
+            dtype=object)
+        n = len(vals)
+        if func == "min":
+            return {"min": vals.min() if n else None}
+        if func == "max":
+            return {"max": vals.max() if n else None}
+        if func == "first":
+            return {"has": n > 0, "val": vals.iloc[0] if n else None}
+        if func == "last":
+            return {"has": n > 0, "val": vals.iloc[-1] if n else None}
+        if func == "mode":
+            return {"counts": vals.value_counts().to_dict()}
+        if func == "nunique":
+            return {"uniq": set(vals)}
+        raise NotImplementedError(f"reduce {func} on strings")
     data, mask = col.data, col.mask
     if col.dtype.is_float:
         valid = ~torch.isnan(data)
@@ -525,27 +566,47 @@ def reduce_column(col: Column, func: str):
     if valid is not None:
         data = data[valid]
     n = int(data.numel())
-    if func in ("sum", "mean", "var", "std"):
+    if func in ("sum", "mean", "var", "std", "sem", "kurt", "skew"):
         acc = data.to(torch.float64) if not col.dtype.is_integer else data.to(torch.int64)
         s = acc.sum().item() if n else (0.0 if not col.dtype.is_integer else 0)
         out = {"sum": s, "count": n}
-        if func in ("var", "std"):
-            out["sumsq"] = float((acc.to(torch.float64) ** 2).sum().item()) if n else 0.0
+        if func in ("var", "std", "sem", "kurt", "skew"):
+            f = acc.to(torch.float64)
+            out["sumsq"] = float((f ** 2).sum().item()) if n else 0.0
+        if func in ("kurt", "skew"):
+            out["sum3"] = float((f ** 3).sum().item()) if n else 0.0
+            out["sum4"] = float((f ** 4).sum().item()) if n else 0.0
         return out
     if func in ("count",):
         return {"count": n}
     if func == "size":
         return {"count": len(col)}
     if func == "min":
-        return {"min": data.min().item() if n else None}
+        return {"min": _reduce_wrap_val(col, data.min().item() if n else None)}
     if func == "max":
-        return {"max": data.max().item() if n else None}
+        return {"max": _reduce_wrap_val(col, data.max().item() if n else None)}
+    if func == "first":
+        return {"has": n > 0,
+                "val": _reduce_wrap_val(col, data[0].item() if n else None)}
+    if func == "last":
+        return {"has": n > 0,
+                "val": _reduce_wrap_val(col, data[-1].item() if n else None)}
     if func == "any":
         return {"any": bool(data.any().item()) if n else False}
     if func == "all":
         return {"all": bool(data.all().item()) if n else True}
     if func == "prod":
         return {"prod": data.to(torch.float64).prod().item() if n else 1.0}
+    if func == "median":
+        # exact: ship local sorted values (scalar result; the grouped median
+        # runs co-located on shuffled groups instead, relational.py)
+        v = torch.sort(data.to(torch.float64))[0].cpu().numpy() if n else None
+        return {"vals": v}
+    if func == "mode":
+        u, c = torch.unique(data, return_counts=True)
+        return {"counts": {
+            _reduce_wrap_val(col, k): int(v)
+            for k, v in zip(u.cpu().tolist(), c.cpu().tolist())}}
     if func == "nunique":
         return {"uniq": set(torch.unique(data).cpu().tolist()) if n else set()}
     if func == "approx_nunique":

@@ -78,6 +78,10 @@ def groupby_local(tbl: Table, keys: Sequence[str],
                   aggs: Sequence[Tuple[str, str, str]],
                   dropna: bool = True) -> Table:
     """Group rows of the local shard; aggs = (out_name, in_name, func)."""
+    # grouped approx_nunique runs single-phase on co-located groups, where
+    # the exact per-group count is the best possible approximation
+    aggs = [(o, i, "nunique") if f == "approx_nunique" else (o, i, f)
+            for o, i, f in aggs]
     tbl, aggs, dec_fix = _normalize_decimal_aggs(tbl, aggs)
     if tbl.device.type == "cuda":
         if any(callable(a[2]) or a[2] in HOST_ONLY_AGGS for a in aggs):
@@ -107,10 +111,18 @@ def groupby_local(tbl: Table, keys: Sequence[str],
 def _groupby_pandas(tbl: Table, keys, aggs, dropna) -> Table:
     need = list(dict.fromkeys(list(keys) + [a[1] for a in aggs if a[1]]))
     df = tbl.select([c for c in need if tbl.has_column(c)]).to_pandas()
+    # dict-encoded inputs arrive as unordered Categorical: value aggs
+    # (min/max/...) must see the string VALUES, so decode non-key columns
+    agg_ins = {a[1] for a in aggs if a[1]} - set(keys)
+    for c in agg_ins:
+        if c in df.columns and isinstance(df[c].dtype, pd.CategoricalDtype):
+            df[c] = df[c].astype(object)
     gb = df.groupby(list(keys), dropna=dropna, sort=False, observed=True)
     named = {}
     for out_name, in_name, func in aggs:
-        if func == "mode":
+        if func == "approx_nunique":  # host: exact nunique qualifies
+            named[out_name] = pd.NamedAgg(column=in_name, aggfunc="nunique")
+        elif func == "mode":
             named[out_name] = pd.NamedAgg(
                 column=in_name,
                 aggfunc=lambda s: s.mode().iloc[0] if len(s.mode()) else None)

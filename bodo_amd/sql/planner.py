@@ -1442,9 +1442,14 @@ class Planner:
 
                 d = int(e.args[1].value) if len(e.args) > 1 else 0
                 sc = 10.0 ** d
-                return ex.UdfMap(self.expr(e.args[0], scope),
-                                 lambda v, _s=sc: _math.trunc(v * _s) / _s,
-                                 None)
+
+                def _tr(v, _s=sc):
+                    # NaN/inf pass through (math.trunc raises on both)
+                    if v != v or v in (float("inf"), float("-inf")):
+                        return v
+                    return _math.trunc(v * _s) / _s
+
+                return ex.UdfMap(self.expr(e.args[0], scope), _tr, None)
             if name in ("sin", "cos", "tan", "asin", "acos", "atan", "sinh",
                         "cosh", "tanh", "degrees", "radians", "cbrt"):
                 import math as _math

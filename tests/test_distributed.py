@@ -870,3 +870,35 @@ def test_dist_shuffle_all_column_kinds():
                        {"df": df, "empty1": empty1})
         exp_total = 2 * len(df) if not empty1 else len(df)
         assert out["total"] == exp_total, (empty1, out)
+
+
+def _q_scalar_reduces(bpd, rank, payload):
+    b = bpd.from_pandas(payload["df"])
+    return {
+        "median": float(b["b"].median()),
+        "kurt": float(b["b"].kurt()),
+        "skew": float(b["b"].skew()),
+        "sem": float(b["b"].sem()),
+        "min_d": bpd.to_pandas_scalar(b["d"].min())
+        if hasattr(bpd, "to_pandas_scalar") else b["d"].min().value,
+        "max_c": str(b["c"].max()),
+    }
+
+
+def test_dist_scalar_reduces():
+    """Whole-column reductions added in round 2 (median/kurt/skew/sem, plus
+    value semantics for dict-string max and timestamp min) across 2 ranks."""
+    rng = np.random.default_rng(7)
+    n = 101
+    df = pd.DataFrame({
+        "b": np.where(rng.random(n) < 0.2, np.nan, rng.random(n) * 9 - 3),
+        "c": rng.choice(["xq", "yy", "zx", "wv"], n),
+        "d": pd.to_datetime(1.6e18 + rng.integers(0, 9e16, n)),
+    })
+    got = run_dist(_q_scalar_reduces, {"df": df})
+    assert abs(got["median"] - df["b"].median()) < 1e-9
+    assert abs(got["kurt"] - df["b"].kurt()) < 1e-9
+    assert abs(got["skew"] - df["b"].skew()) < 1e-9
+    assert abs(got["sem"] - df["b"].sem()) < 1e-9
+    assert pd.Timestamp(got["min_d"]) == df["d"].min()
+    assert got["max_c"] == df["c"].max()

@@ -737,3 +737,48 @@ def test_sql_lateral_flatten():
     # engine contract is pandas semantics: sum of an all-null group is 0
     assert [None if pd.isna(v) else int(v) for v in got2["s"]] == \
         [30, 30, 0]
+
+
+def test_global_aggregates_value_semantics():
+    """Whole-table aggregates report VALUES: dict-string min/max must not
+    leak dictionary codes, timestamps must not leak raw ns, and the
+    moment/selection reduces (median/mode/kurtosis/skew/any_value) combine
+    across shards (reference: dist_reduce, bodo/libs/distributed_api.py)."""
+    import numpy as np
+
+    df = pd.DataFrame({
+        "a": [1, 2, 3, 1, 2, 2],
+        "c": ["xq", "yy", "zx", "xq", "yy", "yy"],
+        "b": [1.0, 2.0, np.nan, 4.0, 5.5, 0.5],
+        "d": pd.to_datetime(["2024-01-05", "2023-06-01", "2025-02-02",
+                             "2024-03-03", "2023-01-01", "2024-06-06"]),
+    })
+    bc = BodoSQLContext({"t": df})
+    r = bc.sql("select max(c) as mc, min(c) as nc, median(b) as md, "
+               "mode(a) as mo, mode(c) as ms, kurtosis(b) as k, "
+               "skew(b) as sk, any_value(a) as av, min(d) as nd, "
+               "max(d) as xd from t").to_pandas().iloc[0]
+    assert r["mc"] == "zx" and r["nc"] == "xq"
+    assert abs(r["md"] - df["b"].median()) < 1e-9
+    assert r["mo"] == 2 and r["ms"] == "yy"
+    assert abs(r["k"] - df["b"].kurt()) < 1e-9
+    assert abs(r["sk"] - df["b"].skew()) < 1e-9
+    assert r["av"] == 1
+    assert pd.Timestamp(r["nd"]) == df["d"].min()
+    assert pd.Timestamp(r["xd"]) == df["d"].max()
+
+
+def test_trunc_nan_passthrough():
+    df = pd.DataFrame({"b": [1.26, float("nan"), -2.78]})
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql("select trunc(b, 1) as t from t").to_pandas()["t"]
+    assert out[0] == 1.2 and out[2] == -2.7 and np.isnan(out[1])
+
+
+def test_grouped_approx_count_distinct():
+    df = pd.DataFrame({"c": ["x", "x", "y", "y", "y"],
+                       "a": [1, 2, 3, 3, 4]})
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql("select c, approx_count_distinct(a) as n from t "
+                 "group by c order by c").to_pandas()
+    assert out["n"].tolist() == [2, 2]
