@@ -112,8 +112,10 @@ def _groupby_pandas(tbl: Table, keys, aggs, dropna) -> Table:
     need = list(dict.fromkeys(list(keys) + [a[1] for a in aggs if a[1]]))
     df = tbl.select([c for c in need if tbl.has_column(c)]).to_pandas()
     # dict-encoded inputs arrive as unordered Categorical: value aggs
-    # (min/max/...) must see the string VALUES, so decode non-key columns
-    agg_ins = {a[1] for a in aggs if a[1]} - set(keys)
+    # (min/max/...) must see the string VALUES, so decode agg inputs (a
+    # column can be key AND agg input at once; object-key groupby stays
+    # correct, and keys are re-dict-encoded on the way out)
+    agg_ins = {a[1] for a in aggs if a[1]}
     for c in agg_ins:
         if c in df.columns and isinstance(df[c].dtype, pd.CategoricalDtype):
             df[c] = df[c].astype(object)
