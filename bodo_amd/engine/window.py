@@ -33,7 +33,8 @@ TRANSFORM_AGGS = {"transform_sum": "sum", "transform_mean": "mean",
 
 ORDERED_FUNCS = {"row_number", "rank", "dense_rank", "percent_rank",
                  "cume_dist", "nth_value", "shift", "cumsum",
-                 "cumcount", "first_value", "last_value", "ntile",
+                 "cumcount", "cummin", "cummax", "cummean", "cumcount_v",
+                 "first_value", "last_value", "ntile",
                  "rolling_sum", "rolling_mean", "rolling_min", "rolling_max",
                  "rolling_count"}
 
@@ -255,6 +256,36 @@ def _ordered_local_device(tbl: Table, keys, order_by, ascending,
             if tbl.column(in_name).dtype.is_integer and inv is None:
                 res = res.long()
             res_inv = inv
+        elif func in ("cummin", "cummax", "cummean", "cumcount_v"):
+            # running frame: ROWS UNBOUNDED PRECEDING .. CURRENT ROW
+            v, inv = sorted_vals(in_name, fill0=True)
+            valid = (torch.ones(n, dtype=torch.float64, device=dev)
+                     if inv is None else (~inv).double())
+            if func == "cummean":
+                cnt = seg_cumsum(valid)
+                res = seg_cumsum(v.double()) / cnt
+                res_inv = cnt == 0
+            elif func == "cumcount_v":
+                res = seg_cumsum(valid).long()
+                res_inv = None
+            else:
+                # segmented Hillis–Steele scan: log2(n) whole-tensor passes;
+                # the segid guard at distance `off` is exact for contiguous
+                # segments and an idempotent combine
+                pad = float("inf") if func == "cummin" else float("-inf")
+                cur = torch.where(valid.bool(), v.double(),
+                                  torch.full((), pad, device=dev))
+                off = 1
+                while off < n:
+                    prev = cur
+                    cur = prev.clone()
+                    samseg = segid[off:] == segid[:n - off]
+                    comb = (torch.minimum if func == "cummin"
+                            else torch.maximum)(prev[off:], prev[:n - off])
+                    cur[off:] = torch.where(samseg, comb, prev[off:])
+                    off *= 2
+                res = cur
+                res_inv = torch.isinf(cur)
         elif func == "shift":
             k = int(arg if arg is not None else 1)
             v, inv = sorted_vals(in_name)
@@ -428,6 +459,23 @@ def _ordered_local(tbl: Table, keys, order_by, ascending, specs) -> Table:
             res = res.reindex(pdf.index)
         elif func == "cumsum":
             res = gbs[in_name].cumsum().reindex(pdf.index)
+        elif func == "cummin":
+            # expanding (not pandas cummin): SQL MIN OVER reports the frame
+            # min even at rows whose own value is NULL
+            res = gbs[in_name].transform(
+                lambda s: s.expanding(min_periods=1).min())
+            res = res.reindex(pdf.index)
+        elif func == "cummax":
+            res = gbs[in_name].transform(
+                lambda s: s.expanding(min_periods=1).max())
+            res = res.reindex(pdf.index)
+        elif func == "cummean":
+            res = gbs[in_name].transform(
+                lambda s: s.expanding(min_periods=1).mean())
+            res = res.reindex(pdf.index)
+        elif func == "cumcount_v":
+            res = gbs[in_name].transform(
+                lambda s: s.notna().cumsum()).reindex(pdf.index)
         elif func in ("first_value", "last_value"):
             which = "first" if func == "first_value" else "last"
             res = gbs[in_name].transform(which).reindex(pdf.index)

@@ -355,11 +355,20 @@ class Planner:
                             f"{fn.upper()} with a window frame")
                     spec = (out, arg_col, f"rolling_{base}", frame + 1)
                 elif order:
-                    if fn == "sum":
-                        spec = (out, arg_col, "cumsum", None)
+                    # default frame (and explicit UNBOUNDED PRECEDING ..
+                    # CURRENT ROW): running aggregate along the order
+                    if fn == "count" and (w.star or not w.args):
+                        spec = (out, "", "row_number", None)
                     else:
-                        raise NotImplementedError(
-                            f"{fn.upper()} OVER (... ORDER BY) not supported")
+                        running = {"sum": "cumsum", "min": "cummin",
+                                   "max": "cummax", "avg": "cummean",
+                                   "mean": "cummean", "count": "cumcount_v"}
+                        base = AGG_FUNCS.get(fn, fn)
+                        if base not in running:
+                            raise NotImplementedError(
+                                f"{fn.upper()} OVER (... ORDER BY) "
+                                "not supported")
+                        spec = (out, arg_col, running[base], None)
                 elif fn == "count" and (w.star or not w.args):
                     spec = (out, "", "transform_size", None)
                 else:
@@ -517,11 +526,17 @@ class Planner:
                 spec = (out, arg_col, "shift", n if fn == "lag" else -n)
             elif fn in AGG_FUNCS or fn == "count":
                 if order:
-                    if fn == "sum":
-                        spec = (out, arg_col, "cumsum", None)
+                    if fn == "count" and (w.star or not w.args):
+                        spec = (out, "", "row_number", None)
                     else:
-                        raise NotImplementedError(
-                            f"{fn.upper()} OVER ORDER BY over aggregates")
+                        running = {"sum": "cumsum", "min": "cummin",
+                                   "max": "cummax", "avg": "cummean",
+                                   "mean": "cummean", "count": "cumcount_v"}
+                        base = AGG_FUNCS.get(fn, fn)
+                        if base not in running:
+                            raise NotImplementedError(
+                                f"{fn.upper()} OVER ORDER BY over aggregates")
+                        spec = (out, arg_col, running[base], None)
                 elif fn == "count" and (w.star or not w.args):
                     spec = (out, "", "transform_size", None)
                 else:

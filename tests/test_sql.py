@@ -782,3 +782,41 @@ def test_grouped_approx_count_distinct():
     out = bc.sql("select c, approx_count_distinct(a) as n from t "
                  "group by c order by c").to_pandas()
     assert out["n"].tolist() == [2, 2]
+
+
+def test_running_window_aggregates():
+    """MIN/MAX/AVG/COUNT OVER (ORDER BY ...) running frames (SQL null
+    semantics: the frame aggregate is reported even at rows whose own value
+    is NULL).  Exercises the segmented Hillis-Steele scan calculator."""
+    import numpy as np
+
+    rng = np.random.default_rng(3)
+    n = 120
+    df = pd.DataFrame({"g": rng.choice(["p", "q"], n),
+                       "o": rng.permutation(n),
+                       "v": np.where(rng.random(n) < 0.2, np.nan,
+                                     rng.random(n) * 10)})
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql(
+        "select g, o, v, "
+        "min(v) over (partition by g order by o) as mn, "
+        "max(v) over (partition by g order by o rows between unbounded "
+        "preceding and current row) as mx, "
+        "avg(v) over (partition by g order by o) as av, "
+        "count(v) over (partition by g order by o) as cv, "
+        "count(*) over (partition by g order by o) as cs "
+        "from t order by o").to_pandas().sort_values("o").reset_index(drop=True)
+    ref = df.sort_values("o").reset_index(drop=True)
+    g = ref.groupby("g")["v"]
+    exp = {
+        "mn": g.transform(lambda s: s.expanding(1).min()),
+        "mx": g.transform(lambda s: s.expanding(1).max()),
+        "av": g.transform(lambda s: s.expanding(1).mean()),
+        "cv": g.transform(lambda s: s.notna().cumsum()),
+        "cs": ref.groupby("g").cumcount() + 1,
+    }
+    for k, e in exp.items():
+        np.testing.assert_allclose(
+            out[k].astype(float).fillna(-9e9),
+            pd.Series(e).astype(float).fillna(-9e9), atol=1e-9,
+            err_msg=k)
