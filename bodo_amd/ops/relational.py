@@ -272,8 +272,11 @@ def _null_out(tbl: Table, valid: torch.Tensor) -> Table:
     for c in tbl.columns:
         mask = valid.clone() if c.mask is None else (c.mask & valid)
         if c.dtype.is_float:
+            # fold BOTH the unmatched-row mask and the column's own null
+            # mask into NaN (dropping c.mask here turned nulls into storage
+            # fill values)
             data = c.data.clone()
-            data[~valid] = float("nan")
+            data[~mask] = float("nan")
             cols.append(Column(c.dtype, data, None, c.offsets, c.dictionary, len(c)))
         else:
             cols.append(Column(c.dtype, c.data, mask, c.offsets, c.dictionary, len(c)))
@@ -309,24 +312,51 @@ def _merge_joined(lt: Table, rt: Table, left_on, right_on, suffixes, how) -> Tab
 
 
 def _coalesce(a: Column, b: Column) -> Column:
-    if a.mask is None and not a.dtype.is_float:
-        return a
-    if a.dtype.is_float:
-        use_b = torch.isnan(a.data)
-        data = torch.where(use_b, b.data.to(a.data.dtype), a.data)
-        return Column(a.dtype, data, None, length=len(a))
-    use_b = ~a.mask
-    if a.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
-        # host path
+    """Outer-join key coalescing: a where a is non-null, else b.  Null =
+    mask OR NaN (masked storage holds arbitrary fill, never trust it); a
+    mixed int/float key pair coalesces in float64 like pandas merge."""
+    if a.dtype.kind in (TypeKind.STRING, TypeKind.DICT) \
+            or b.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+        # host path (decode Categorical: the two dictionaries differ).
+        # `a` can be a null-typed/empty-shard schema degenerate (float64
+        # all-null) while `b` is the real string column — never fall into
+        # the numeric branch then, it would emit dictionary CODES
         sa, sb = a.to_pandas(), b.to_pandas()
+        if isinstance(sa.dtype, pd.CategoricalDtype):
+            sa = sa.astype(object)
+        if isinstance(sb.dtype, pd.CategoricalDtype):
+            sb = sb.astype(object)
         out = sa.where(~pd.isna(sa), sb)
         import pyarrow as pa
 
         return Column.from_arrow(pa.Array.from_pandas(out), a.device)
-    data = torch.where(use_b, b.data.to(a.data.dtype), a.data)
-    mask = None
-    if b.mask is not None:
-        mask = a.mask | b.mask
+
+    def nulls(c: Column) -> torch.Tensor:
+        nn = torch.zeros(len(c), dtype=torch.bool, device=c.data.device)
+        if c.mask is not None:
+            nn |= ~c.mask
+        if c.dtype.is_float:
+            nn |= torch.isnan(c.data)
+        return nn
+
+    a_null = nulls(a)
+    if not bool(a_null.any()):
+        return Column(a.dtype, a.data, None, length=len(a))
+    b_null = nulls(b)
+    both = a_null & b_null
+    if a.dtype.is_float or b.dtype.is_float:
+        ad = a.data.to(torch.float64)
+        data = torch.where(a_null, b.data.to(torch.float64), ad)
+        data = torch.where(both, torch.full((), float("nan"),
+                                            dtype=torch.float64,
+                                            device=data.device), data)
+        out_dt = a.dtype if a.dtype.is_float else bt.float64
+        if out_dt.kind == TypeKind.FLOAT32:
+            data = data.to(torch.float32)
+        return Column(out_dt, data, None, length=len(a))
+    # int-like (incl. timestamps): keep exact storage, mask the both-null rows
+    data = torch.where(a_null, b.data.to(a.data.dtype), a.data)
+    mask = ~both if bool(both.any()) else None
     return Column(a.dtype, data, mask, length=len(a))
 
 

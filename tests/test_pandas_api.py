@@ -610,3 +610,41 @@ def test_list_accessor_len_get():
         [2, 1, 0, None, 3]
     assert [None if pd.isna(v) else int(v) for v in out["e"]] == \
         [2, None, None, None, 5]
+
+
+def test_outer_join_null_keys_preserved():
+    """Outer/left joins keep NULL keys as nulls — not storage fill values
+    (coalesce must honor validity masks, not only NaN payloads)."""
+    l = pd.DataFrame({"k": [1.0, np.nan, 2.0], "v": [10, 20, 30]})
+    r = pd.DataFrame({"k": [1.0, 3.0], "w": [100, 200]})
+    got = (bpd.from_pandas(l).merge(bpd.from_pandas(r), on="k", how="outer")
+           .to_pandas())
+    exp = l.merge(r, on="k", how="outer")
+    g = got.sort_values(["k", "v"], na_position="last").reset_index(drop=True)
+    e = exp.sort_values(["k", "v"], na_position="last").reset_index(drop=True)
+    pd.testing.assert_frame_equal(g, e, check_dtype=False)
+
+
+def test_outer_join_mixed_int_float_keys():
+    """int64 left key vs float64-with-NaN right key coalesces in float64
+    (casting NaN to int64 min corrupted unmatched rows)."""
+    l = pd.DataFrame({"k": [1, 2], "v": [10, 20]})
+    r = pd.DataFrame({"k": [2.0, np.nan, 4.0], "w": [7, 8, 9]})
+    got = (bpd.from_pandas(l).merge(bpd.from_pandas(r), on="k", how="outer")
+           .to_pandas())
+    exp = l.merge(r, on="k", how="outer")
+    g = got.sort_values(["k", "w"], na_position="last").reset_index(drop=True)
+    e = exp.sort_values(["k", "w"], na_position="last").reset_index(drop=True)
+    pd.testing.assert_frame_equal(g, e, check_dtype=False)
+
+
+def test_outer_join_empty_left_string_key():
+    """Empty left frame (degenerate null schema) outer-joined on a string
+    key must yield string values, not dictionary codes."""
+    l = pd.DataFrame({"k": pd.Series([], dtype=object),
+                      "v": pd.Series([], dtype=float)})
+    r = pd.DataFrame({"k": ["a", "b"], "w": [1, 2]})
+    got = (bpd.from_pandas(l).merge(bpd.from_pandas(r), on="k", how="outer")
+           .to_pandas())
+    assert sorted(map(str, got["k"])) == ["a", "b"]
+    assert sorted(got["w"]) == [1, 2]
