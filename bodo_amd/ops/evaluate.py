@@ -850,9 +850,11 @@ def _str_dict_generic(a: Column, op: str, args, kwargs) -> Column:
         if a.mask is not None:
             out.mask = a.mask if out.mask is None else (out.mask & a.mask)
         return out
-    if op == "split_get":
+    if op in ("split_get", "split_part"):
         pat, idx = args
         res = vals.str.split(pat).str.get(idx)
+        if op == "split_part":  # Snowflake: out-of-range part -> ''
+            res = res.fillna("")
     else:
         res = getattr(vals.str, op)(*args, **(kwargs or {}))
     if res.dtype == object or isinstance(res.dtype, pd.StringDtype):
@@ -994,9 +996,11 @@ def str_op(a: Column, op: str, args, kwargs=None) -> Column:
         import pyarrow as pa
 
         ser = arr.to_pandas().astype("object")
-        if op == "split_get":
+        if op in ("split_get", "split_part"):
             pat, idx = args
             res_s = ser.str.split(pat).str.get(idx)
+            if op == "split_part":  # Snowflake: '' past the end, null stays
+                res_s = res_s.mask(ser.notna() & res_s.isna(), "")
         elif op == "split_list":
             lists = ser.str.split(args[0] if args and args[0] is not None
                                   else None)

@@ -652,13 +652,19 @@ class Parser:
                 self.expect_op(")")
                 return ExtractE(fld, e)
             if t.value == "substring":
+                # SUBSTRING(x FROM s [FOR n]) and SUBSTRING(x, s[, n])
                 self.expect_op("(")
                 e = self.parse_expr()
-                self.expect_kw("from")
-                start = self.parse_expr()
                 length = None
-                if self.accept_kw("for"):
-                    length = self.parse_expr()
+                if self.accept_op(","):
+                    start = self.parse_expr()
+                    if self.accept_op(","):
+                        length = self.parse_expr()
+                else:
+                    self.expect_kw("from")
+                    start = self.parse_expr()
+                    if self.accept_kw("for"):
+                        length = self.parse_expr()
                 self.expect_op(")")
                 return Func("substring", [e, start, length])
             if t.value == "exists":
@@ -696,6 +702,14 @@ class Parser:
                 if self.accept_op("*"):
                     self.expect_op(")")
                     star = True
+                elif t.value.lower() == "position":
+                    # POSITION(sub IN str) — parse below IN-predicate level
+                    # so the keyword separates the args; the comma form
+                    # POSITION(sub, str) also lands here
+                    args.append(self.parse_add())
+                    if self.accept_kw("in") or self.accept_op(","):
+                        args.append(self.parse_expr())
+                    self.expect_op(")")
                 elif not self.accept_op(")"):
                     args.append(self.parse_expr())
                     while self.accept_op(","):

@@ -1165,10 +1165,20 @@ class Planner:
                 return ex.StrOp(self.expr(e.args[0], scope), "len")
             if name == "trim":
                 return ex.StrOp(self.expr(e.args[0], scope), "strip")
-            if name == "substring":
-                arg, start, ln = e.args
-                s = int(start.value) - 1
-                stop = None if ln is None else s + int(ln.value)
+            if name in ("substring", "substr", "mid"):
+                arg = e.args[0]
+                start = _lit_int(e.args[1])
+                ln = (_lit_int(e.args[2])
+                      if len(e.args) > 2 and e.args[2] is not None else None)
+                # Snowflake: 1-based; 0 behaves like 1; negative counts
+                # from the end
+                s = start - 1 if start > 0 else (start if start < 0 else 0)
+                if ln is None:
+                    stop = None
+                elif s < 0:
+                    stop = s + ln if s + ln < 0 else None
+                else:
+                    stop = s + ln
                 return ex.StrOp(self.expr(arg, scope), "slice", (s, stop, 1))
             if name == "round":
                 return ex.RoundExpr(self.expr(e.args[0], scope),
@@ -1276,15 +1286,17 @@ class Planner:
                                  (sub.value,))
                 return ex.BinOp("add", found, ex.Const(1))
             if name == "split_part":
-                # Snowflake 1-based part index
-                return ex.StrOp(self.expr(e.args[0], scope), "split_get",
+                # Snowflake 1-based part index; out-of-range -> ''
+                return ex.StrOp(self.expr(e.args[0], scope), "split_part",
                                 (e.args[1].value,
                                  int(e.args[2].value) - 1))
             if name in ("lpad", "rpad"):
+                # Snowflake: a string longer than n is TRUNCATED to n
                 n = int(e.args[1].value)
                 fill = e.args[2].value if len(e.args) > 2 else " "
                 op = "rjust" if name == "lpad" else "ljust"
-                return ex.StrOp(self.expr(e.args[0], scope), op, (n, fill))
+                padded = ex.StrOp(self.expr(e.args[0], scope), op, (n, fill))
+                return ex.StrOp(padded, "slice", (0, n))
             if name == "repeat":
                 return ex.StrOp(self.expr(e.args[0], scope), "repeat",
                                 (int(e.args[1].value),))
@@ -1580,6 +1592,18 @@ def _interval_to_timestamp_delta(left_ast, qty, unit, op):
         res = base + off if op == "add" else base - off
         return ex.Const(pd.Timestamp(res), bt.timestamp_ns)
     return None
+
+
+def _lit_int(e) -> int:
+    """Constant integer from a literal or a unary-minus literal (the parser
+    lowers `-3` to Bin('sub', Lit(0), Lit(3)))."""
+    if isinstance(e, ast.Lit):
+        return int(e.value)
+    if isinstance(e, ast.Bin) and e.op == "sub" \
+            and isinstance(e.left, ast.Lit) and e.left.value == 0 \
+            and isinstance(e.right, ast.Lit):
+        return -int(e.right.value)
+    raise NotImplementedError(f"constant integer expected, got {e!r}")
 
 
 def _lit_expr(e: ast.Lit) -> ex.Expr:

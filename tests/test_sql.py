@@ -564,7 +564,8 @@ def test_more_string_functions():
             got[c] = got[c].astype(object)
     assert got.l.tolist() == [11, 11, 1]
     assert got.p.tolist() == [5, 2, 0]
-    assert got.sp.where(got.sp.notna(), None).tolist() == [None, "bar", None]
+    # Snowflake SPLIT_PART: out-of-range part index -> empty string
+    assert got.sp.tolist() == ["", "bar", ""]
     assert got.lp.iloc[2] == "*" * 14 + "x"
     assert got.rp.iloc[2] == "xx"
 
@@ -820,3 +821,27 @@ def test_running_window_aggregates():
             out[k].astype(float).fillna(-9e9),
             pd.Series(e).astype(float).fillna(-9e9), atol=1e-9,
             err_msg=k)
+
+
+def test_string_function_edge_semantics():
+    """Snowflake edge semantics: SPLIT_PART out-of-range -> '', LPAD/RPAD
+    truncate long inputs, SUBSTR aliases/negative start, POSITION(x IN y)."""
+    df = pd.DataFrame({"s": ["a,b,c", "hello world", "x", None]})
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql(
+        "select split_part(s, ',', 2) as sp, lpad(s, 5, '*') as lp, "
+        "rpad(s, 5, '*') as rp, substr(s, 2, 3) as su, "
+        "substring(s, 1, 2) as su2, position('b' in s) as po "
+        "from t").to_pandas()
+    assert out["sp"].tolist()[:3] == ["b", "", ""]
+    assert out["lp"].tolist()[:3] == ["a,b,c", "hello", "****x"]
+    assert out["rp"].tolist()[:3] == ["a,b,c", "hello", "x****"]
+    assert out["su"].tolist()[:3] == [",b,", "ell", ""]
+    assert out["su2"].tolist()[:3] == ["a,", "he", "x"]
+    assert out["po"].tolist()[:3] == [3, 0, 0]
+    assert out.iloc[3].isna().all()
+    # pandas frontend keeps NaN for out-of-range .str.get
+    import bodo_amd.pandas as bpd
+
+    g = bpd.from_pandas(df)["s"].str.split(",").str.get(1).to_pandas()
+    assert g.iloc[0] == "b" and pd.isna(g.iloc[1])
