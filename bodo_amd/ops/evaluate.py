@@ -764,12 +764,44 @@ def dt_field(a: Column, fld: str) -> Column:
 
 
 def _dt_field_cpu(a: Column, fld: str) -> Column:
+    if a.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+        # TO_DATE('2024-01-01') and friends: parse strings first
+        ser = pd.to_datetime(a.to_pandas(), errors="coerce")
+        a = Column(bt.timestamp_ns,
+                   torch.from_numpy(ser.to_numpy().view("int64").copy()),
+                   torch.from_numpy(ser.notna().to_numpy()))
     vals = a.data.numpy()
     if a.dtype.kind == TypeKind.DATE32:
         ts = vals.astype("datetime64[D]")
     else:
         ts = vals.view("datetime64[ns]")
     idx = pd.DatetimeIndex(ts)
+    if fld.startswith("add_months:"):
+        # calendar month add with Snowflake month-end clamping (pandas
+        # DateOffset clamps the same way: Jan 31 + 1 mo -> Feb 29)
+        n = int(fld.split(":", 1)[1])
+        out = (idx + pd.DateOffset(months=n)).asi8
+        return _dt_res(bt.timestamp_ns, out, a, fld)
+    if fld == "last_day":
+        out = (idx + pd.offsets.MonthEnd(0)).normalize().asi8
+        return _dt_res(bt.timestamp_ns, out, a, fld)
+    if fld == "epoch_second":
+        out = idx.asi8 // 10**9
+        return _dt_res(bt.int64, out, a, fld)
+    if fld in ("dayname", "monthname"):
+        import pyarrow as pa
+
+        if fld == "dayname":
+            codes = np.clip(np.asarray(idx.dayofweek, dtype=np.int64),
+                            0, 6).astype(np.int32)
+            names = ["Mon", "Tue", "Wed", "Thu", "Fri", "Sat", "Sun"]
+        else:
+            codes = np.clip(np.asarray(idx.month, dtype=np.int64) - 1,
+                            0, 11).astype(np.int32)
+            names = ["Jan", "Feb", "Mar", "Apr", "May", "Jun", "Jul",
+                     "Aug", "Sep", "Oct", "Nov", "Dec"]
+        return Column(bt.dictionary, torch.from_numpy(codes), a.mask,
+                      dictionary=pa.array(names, type=pa.large_string()))
     if fld == "date":
         days = (idx.normalize().asi8 // NS_PER_DAY).astype(np.int32)
         out = np.ascontiguousarray(days)
@@ -789,6 +821,14 @@ def _dt_field_cpu(a: Column, fld: str) -> Column:
         dtype = _DT_OUT_TYPE[fld]
         out = out.astype(bt.numpy_storage_dtype(dtype))
     res = Column(dtype, torch.from_numpy(out), a.mask)
+    from .gpu import _DT_RANGE
+
+    res.val_range = _DT_RANGE.get(fld)
+    return res
+
+
+def _dt_res(dtype, out, a: Column, fld: str) -> Column:
+    res = Column(dtype, torch.from_numpy(np.ascontiguousarray(out)), a.mask)
     from .gpu import _DT_RANGE
 
     res.val_range = _DT_RANGE.get(fld)

@@ -1328,6 +1328,10 @@ class Planner:
                 ns_per = {"day": 86400 * 10**9, "hour": 3600 * 10**9,
                           "minute": 60 * 10**9, "second": 10**9,
                           "week": 7 * 86400 * 10**9}
+                if unit in ("month", "quarter", "year"):
+                    mult = {"month": 1, "quarter": 3, "year": 12}[unit]
+                    n = _lit_int(e.args[1]) * mult
+                    return ex.DtField(base, f"add_months:{n}")
                 if unit not in ns_per:
                     raise NotImplementedError(f"DATEADD unit {unit}")
                 return ex.BinOp("add", base,
@@ -1566,20 +1570,19 @@ class Planner:
 
                 return ex.UdfMap(self.expr(e.args[0], scope), _jx, "ignore")
             if name in ("dayname", "monthname"):
-                if name == "dayname":
-                    fld, names = "dayofweek", ["Mon", "Tue", "Wed", "Thu",
-                                               "Fri", "Sat", "Sun"]
-                    base = 0
-                else:
-                    fld, names = "month", ["Jan", "Feb", "Mar", "Apr", "May",
-                                           "Jun", "Jul", "Aug", "Sep", "Oct",
-                                           "Nov", "Dec"]
-                    base = 1
-                f = ex.DtField(self.expr(e.args[0], scope), fld)
-                conds = tuple(ex.Cmp("eq", f, ex.Const(i + base))
-                              for i in range(len(names) - 1))
-                thens = tuple(ex.Const(nm) for nm in names[:-1])
-                return ex.Case(conds, thens, ex.Const(names[-1]))
+                # dedicated dt field: a CASE lowering leaked the ELSE value
+                # into NULL rows (string CASE has no validity mask)
+                return ex.DtField(self.expr(e.args[0], scope), name)
+            if name == "last_day":
+                return ex.DtField(self.expr(e.args[0], scope), "last_day")
+            if name in ("to_date", "try_to_date", "date"):
+                return ex.DtField(self.expr(e.args[0], scope), "floor_day")
+            if name in ("epoch_second", "date_part_epoch_second"):
+                return ex.DtField(self.expr(e.args[0], scope),
+                                  "epoch_second")
+            if name == "add_months":
+                return ex.DtField(self.expr(e.args[0], scope),
+                                  f"add_months:{_lit_int(e.args[1])}")
             raise NotImplementedError(f"SQL function {name}")
         raise NotImplementedError(f"expr {e}")
 

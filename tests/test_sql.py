@@ -845,3 +845,32 @@ def test_string_function_edge_semantics():
 
     g = bpd.from_pandas(df)["s"].str.split(",").str.get(1).to_pandas()
     assert g.iloc[0] == "b" and pd.isna(g.iloc[1])
+
+
+def test_datetime_function_extensions():
+    """LAST_DAY, calendar DATEADD (month/quarter/year with month-end
+    clamping), ADD_MONTHS, TO_DATE (timestamps and strings), EPOCH_SECOND,
+    and NULL-safe MONTHNAME/DAYNAME."""
+    df = pd.DataFrame({
+        "d": pd.to_datetime(["2024-01-31 13:45:30", "2023-12-15 06:07:08",
+                             "2024-02-29 23:59:59", None]),
+        "s": ["2024-05-06", "2023-01-02", None, "2024-12-31"],
+    })
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql(
+        "select last_day(d) as ld, dateadd('month', 2, d) as am, "
+        "dateadd('year', -1, d) as ay, to_date(d) as td, to_date(s) as ts, "
+        "epoch_second(d) as ep, monthname(d) as mn, dayname(d) as dn "
+        "from t").to_pandas()
+    assert pd.Timestamp(out["ld"][0]) == pd.Timestamp("2024-01-31")
+    assert pd.Timestamp(out["ld"][2]) == pd.Timestamp("2024-02-29")
+    # Jan 31 + 2 months clamps to Mar 31 (pandas/Snowflake agree)
+    assert pd.Timestamp(out["am"][0]) == pd.Timestamp("2024-03-31 13:45:30")
+    assert pd.Timestamp(out["ay"][2]) == pd.Timestamp("2023-02-28 23:59:59")
+    assert pd.Timestamp(out["td"][1]) == pd.Timestamp("2023-12-15")
+    assert pd.Timestamp(out["ts"][3]) == pd.Timestamp("2024-12-31")
+    assert int(out["ep"][1]) == int(df["d"][1].timestamp())
+    assert out["mn"].tolist()[:3] == ["Jan", "Dec", "Feb"]
+    assert out["dn"].tolist()[:3] == ["Wed", "Fri", "Thu"]
+    assert pd.isna(out.iloc[3]["mn"]) and pd.isna(out.iloc[3]["dn"])
+    assert pd.isna(out.iloc[3]["ld"]) and pd.isna(out.iloc[3]["ep"])
