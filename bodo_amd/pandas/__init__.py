@@ -211,6 +211,46 @@ def qcut(x, q, labels=None, **kwargs):
     return _pd.qcut(x, q, labels=labels, **kwargs)
 
 
+def crosstab(index, columns, values=None, aggfunc=None,
+             rownames=None, colnames=None):
+    """Distributed crosstab: grouped count (or aggfunc) on the pair of
+    series, host pivot of the small result (reference role:
+    pd.crosstab via BodoDataFrame.pivot_table)."""
+    from .series import BodoSeries
+
+    if not isinstance(index, BodoSeries) and not isinstance(columns,
+                                                            BodoSeries):
+        return _pd.crosstab(index, columns, values=values, aggfunc=aggfunc,
+                            rownames=rownames, colnames=colnames)
+    rname = (rownames[0] if rownames else
+             getattr(index, "name", None) or "row_0")
+    cname = (colnames[0] if colnames else
+             getattr(columns, "name", None) or "col_0")
+    if cname == rname:
+        cname = f"{cname}_col"
+    df = index.to_frame(rname)
+    df[cname] = columns
+    if values is None:
+        small = df.groupby([rname, cname], as_index=False).agg(
+            __n=_pd.NamedAgg(rname, "size"))
+        pdf = small.to_pandas()
+        for c in pdf.columns:
+            if isinstance(pdf[c].dtype, _pd.CategoricalDtype):
+                pdf[c] = pdf[c].astype(object)
+        out = pdf.pivot_table(values="__n", index=rname, columns=cname,
+                              aggfunc="first", fill_value=0)
+        return out.astype("int64").sort_index()
+    df["__v"] = values
+    small = df.groupby([rname, cname], as_index=False).agg(
+        __v=_pd.NamedAgg("__v", aggfunc or "mean"))
+    pdf = small.to_pandas()
+    for c in pdf.columns:
+        if isinstance(pdf[c].dtype, _pd.CategoricalDtype):
+            pdf[c] = pdf[c].astype(object)
+    return pdf.pivot_table(values="__v", index=rname, columns=cname,
+                           aggfunc="first").sort_index()
+
+
 def to_datetime(arg, **kwargs):
     if isinstance(arg, BodoSeries):
         from ..core import types as bt

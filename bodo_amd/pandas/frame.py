@@ -675,8 +675,20 @@ class BodoDataFrame:
                     aggfunc="mean"):
         """Distributed groupby([index, columns]) then a host pivot of the
         (small) aggregated result (reference: frame.py pivot_table)."""
-        assert index is not None and columns is not None
+        assert index is not None
         idx = [index] if isinstance(index, str) else list(index)
+        if columns is None:
+            # no columns: a grouped aggregate indexed by `index`
+            vals0 = ([values] if isinstance(values, str) else
+                     list(values) if values is not None else
+                     [c for c in self._columns if c not in idx])
+            small = self.groupby(idx, as_index=False).agg(
+                **{v: pd.NamedAgg(v, aggfunc) for v in vals0})
+            pdf = small.to_pandas()
+            for c in pdf.columns:
+                if isinstance(pdf[c].dtype, pd.CategoricalDtype):
+                    pdf[c] = pdf[c].astype(object)
+            return pdf.set_index(idx).sort_index()
         cols = [columns] if isinstance(columns, str) else list(columns)
         if values is None:
             values = [c for c in self._columns

@@ -648,3 +648,20 @@ def test_outer_join_empty_left_string_key():
            .to_pandas())
     assert sorted(map(str, got["k"])) == ["a", "b"]
     assert sorted(got["w"]) == [1, 2]
+
+
+def test_crosstab_and_pivot_index_only():
+    rng = np.random.default_rng(0)
+    df = pd.DataFrame({"g": rng.choice(["a", "b", "c"], 100),
+                       "h": rng.choice(["p", "q"], 100),
+                       "y": rng.random(100)})
+    b = bpd.from_pandas(df)
+    got = bpd.crosstab(b["g"], b["h"])
+    exp = pd.crosstab(df["g"], df["h"])
+    assert (got.values == exp.values).all()
+    got2 = bpd.crosstab(b["g"], b["h"], values=b["y"], aggfunc="mean")
+    exp2 = pd.crosstab(df["g"], df["h"], values=df["y"], aggfunc="mean")
+    np.testing.assert_allclose(got2.values, exp2.values)
+    gotp = b.pivot_table(values="y", index="g", aggfunc="mean")
+    expp = df.pivot_table(values="y", index="g", aggfunc="mean")
+    np.testing.assert_allclose(gotp.values, expp.values)
