@@ -92,6 +92,12 @@ class _Evaluator:
         v = normalize_const(v, dtype)
         return Column.full_const(v, dtype, self.n, self.device)
 
+    def visit_ScalarSubquery(self, e) -> Column:
+        # projected uncorrelated scalar subquery: evaluate once (memoized),
+        # broadcast as a constant column
+        v, dtype = self._subquery_scalar(e)
+        return Column.full_const(v, dtype, self.n, self.device)
+
     # ------------------------------------------------------------------
     def _decode_if_dict(self, c: Column) -> Column:
         return c

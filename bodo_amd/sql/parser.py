@@ -186,6 +186,7 @@ class SetOpQ:
     all: bool
     left: Any
     right: Any
+    ctes: Optional[List[Tuple[str, Any]]] = None  # WITH name AS (...)
 
 
 @dataclass
@@ -224,6 +225,7 @@ class Query:
     limit: Optional[int]
     grouping_sets: Optional[List[List[Any]]] = None
     qualify: Optional[Any] = None  # Snowflake-dialect window filter
+    ctes: Optional[List[Tuple[str, Any]]] = None  # WITH name AS (...)
 
 
 class Parser:
@@ -266,9 +268,25 @@ class Parser:
 
     # ------------------------------------------------------------- query
     def parse(self):
+        ctes = None
+        t = self.peek()
+        if t and t.kind == "id" and t.value.lower() == "with":
+            # WITH name AS (query) [, name2 AS (query2)] ... SELECT ...
+            self.i += 1
+            ctes = []
+            while True:
+                name = self.next().value.lower()
+                self.expect_kw("as")
+                self.expect_op("(")
+                sub = self.parse_query_expr()
+                self.expect_op(")")
+                ctes.append((name, sub))
+                if not self.accept_op(","):
+                    break
         q = self.parse_query_expr()
         if self.peek() is not None:
             raise SyntaxError(f"trailing tokens at {self.peek()}")
+        q.ctes = ctes
         return q
 
     def parse_query_expr(self):

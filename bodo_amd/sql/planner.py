@@ -94,6 +94,10 @@ class Planner:
     def __init__(self, tables: Dict[str, "object"]):
         self.tables = {k.lower(): v for k, v in tables.items()}
         self._counter = 0
+        # WITH-clause views: name -> AST, planned lazily on first reference
+        # and cached so multiple references share one plan subtree
+        self.cte_asts: Dict[str, object] = {}
+        self._cte_cache: Dict[str, tuple] = {}
 
     def _uniq(self, base: str) -> str:
         self._counter += 1
@@ -101,6 +105,8 @@ class Planner:
 
     # ------------------------------------------------------------------
     def plan(self, q):
+        for name, sub in (getattr(q, "ctes", None) or ()):
+            self.cte_asts[name] = sub
         if isinstance(q, ast.SetOpQ):
             return self._plan_setop(q)
         scope = Scope()
@@ -191,6 +197,10 @@ class Planner:
             for c in plain[1:]:
                 cond = ast.Bin("and", cond, c)
             plan = pn.Filter(plan, self.expr(cond, scope))
+
+        # correlated scalar subqueries in the SELECT list decorrelate into
+        # LEFT JOINs before projection planning
+        plan = self._rewrite_select_subqueries(plan, scope, q)
 
         # ------------------------------------------------- aggregation
         has_agg = any(_has_agg(it.expr) for it in q.items if not it.star) \
@@ -631,11 +641,17 @@ class Planner:
             sub, cols = self.plan(tr.subquery)
         else:
             name = tr.name.lower()
-            if name not in self.tables:
+            if name in self.cte_asts:
+                if name not in self._cte_cache:
+                    self._cte_cache[name] = self.plan(self.cte_asts[name])
+                sub, cols = self._cte_cache[name]
+                cols = list(cols)
+            elif name not in self.tables:
                 raise KeyError(f"unknown table {tr.name}")
-            frame = self.tables[name]
-            sub = frame._lazy_plan
-            cols = list(frame._columns)
+            else:
+                frame = self.tables[name]
+                sub = frame._lazy_plan
+                cols = list(frame._columns)
         internal = []
         need_rename = False
 
@@ -910,6 +926,76 @@ class Planner:
         return pn.Projection(out, keep_cols,
                              tuple(ex.ColRef(c) for c in keep_cols))
 
+    def _rewrite_select_subqueries(self, plan, scope: Scope, q: ast.Query):
+        """SELECT-list scalar subqueries.  Correlated equi ones decorrelate
+        into a LEFT JOIN against the grouped inner query (NULL when no
+        match, per SQL); uncorrelated ones stay as ScalarSubquery exprs
+        handled by expr() (reference: Calcite RelDecorrelator as used by
+        BodoSQL)."""
+        state = {"plan": plan}
+
+        def rewrite(e):
+            if isinstance(e, ast.SubqueryE):
+                repl = self._decorrelate_select_subq(e.query, state, scope)
+                return repl if repl is not None else e
+            if isinstance(e, (ast.Query, ast.SetOpQ)):
+                return e
+            for f in getattr(e, "__dataclass_fields__", {}):
+                v = getattr(e, f)
+                if isinstance(v, list):
+                    setattr(e, f, [rewrite(x) if hasattr(
+                        x, "__dataclass_fields__") else x for x in v])
+                elif hasattr(v, "__dataclass_fields__"):
+                    setattr(e, f, rewrite(v))
+            return e
+
+        for it in q.items:
+            if not it.star and it.expr is not None:
+                it.expr = rewrite(it.expr)
+        return state["plan"]
+
+    def _decorrelate_select_subq(self, subq, state, scope: Scope):
+        if not isinstance(subq, ast.Query) or len(subq.items) != 1 \
+                or subq.items[0].star:
+            return None
+        if self._try_plan_full(subq) is not None:
+            return None  # uncorrelated: expr() evaluates it once
+        try:
+            sub_plan, sub_scope, corr, corr_extra = self._plan_inner(
+                subq, scope)
+        except (NotImplementedError, KeyError):
+            return None
+        if corr_extra or not corr:
+            return None
+        item = subq.items[0]
+        agg_ast, rebuild = _extract_single_agg(item.expr)
+        keys = tuple(i for _, i in corr)
+        if agg_ast is not None:
+            val, aplan = self._subq_agg_plan(agg_ast, sub_scope, sub_plan,
+                                             keys)
+            if rebuild is not None:
+                wrapped = self._uniq("__sw")
+                aplan = pn.Projection(
+                    aplan, tuple(list(keys) + [wrapped]),
+                    tuple([ex.ColRef(k) for k in keys]
+                          + [rebuild(ex.ColRef(val), self, sub_scope)]))
+                val = wrapped
+        else:
+            # bare value: SQL requires at most one inner row per outer row
+            vcol, vplan = self._subq_value_col(item, sub_scope, sub_plan)
+            val = self._uniq("__sv")
+            aplan = pn.Aggregate(vplan, keys, ((val, vcol, "first"),))
+        pfx = self._uniq("__ss")
+        renamed = [f"{pfx}_{c}" for c in keys] + [f"{pfx}_{val}"]
+        proj = pn.Projection(aplan, tuple(renamed),
+                             tuple(ex.ColRef(c) for c in list(keys) + [val]))
+        state["plan"] = pn.Join(state["plan"], proj,
+                                tuple(o for o, _ in corr),
+                                tuple(renamed[:-1]), "left")
+        name = renamed[-1]
+        scope.add("", name, name)
+        return ast.Col(None, name)
+
     def _scalar_cmp_subquery(self, op, outer_expr, q: ast.Query, plan, scope):
         full = self._try_plan_full(q)
         if full is not None:
@@ -1030,6 +1116,8 @@ class Planner:
                     agg_specs.append((out, in_name, func))
                 agg_map[k] = out
                 return
+            if isinstance(e, (ast.Query, ast.SetOpQ, ast.SubqueryE)):
+                return  # a subquery's aggregates are planned by its own plan
             for f in getattr(e, "__dataclass_fields__", {}):
                 v = getattr(e, f)
                 if isinstance(v, (list, tuple)):
@@ -1584,6 +1672,15 @@ class Planner:
                 return ex.DtField(self.expr(e.args[0], scope),
                                   f"add_months:{_lit_int(e.args[1])}")
             raise NotImplementedError(f"SQL function {name}")
+        if isinstance(e, ast.SubqueryE):
+            # uncorrelated scalar subquery in an expression position:
+            # evaluates once (memoized in the expr layer)
+            full = self._try_plan_full(e.query)
+            if full is not None:
+                sub_plan, out_names = full
+                return ex.ScalarSubquery(sub_plan, out_names[0])
+            raise NotImplementedError(
+                "correlated scalar subquery in this context")
         raise NotImplementedError(f"expr {e}")
 
 
@@ -1699,6 +1796,8 @@ def _has_agg(e) -> bool:
     if isinstance(e, ast.Func) and (
             e.name in AGG_FUNCS or e.name in ("listagg", "string_agg")):
         return True
+    if isinstance(e, (ast.Query, ast.SetOpQ)):
+        return False  # a subquery's aggregates are its own, not the outer's
     for f in getattr(e, "__dataclass_fields__", {}):
         v = getattr(e, f)
         if isinstance(v, (list, tuple)):

@@ -874,3 +874,57 @@ def test_datetime_function_extensions():
     assert out["dn"].tolist()[:3] == ["Wed", "Fri", "Thu"]
     assert pd.isna(out.iloc[3]["mn"]) and pd.isna(out.iloc[3]["dn"])
     assert pd.isna(out.iloc[3]["ld"]) and pd.isna(out.iloc[3]["ep"])
+
+
+def test_with_cte():
+    """WITH-clause views: simple, chained, joined, self-joined, under
+    set ops (reference: BodoSQL CTE planning)."""
+    rng = np.random.default_rng(1)
+    n = 300
+    df = pd.DataFrame({"g": rng.choice(["a", "b", "c", "d"], n),
+                       "x": rng.integers(0, 50, n),
+                       "y": rng.random(n) * 100})
+    o = pd.DataFrame({"g": ["a", "b", "c", "d"], "w": [1.0, 2.0, 3.0, 4.0]})
+    bc = BodoSQLContext({"t": df, "o": o})
+    got = bc.sql("with s as (select g, sum(y) as sy from t group by g) "
+                 "select g, sy from s where sy > 100").to_pandas()
+    exp = df.groupby("g", as_index=False).agg(sy=("y", "sum")).query("sy > 100")
+    assert sorted(got["sy"].round(6)) == sorted(exp["sy"].round(6).tolist())
+    got2 = bc.sql(
+        "with s as (select g, x, y from t where x > 10), "
+        "u as (select g, avg(y) as ay from s group by g) "
+        "select * from u").to_pandas()
+    exp2 = df[df.x > 10].groupby("g", as_index=False).agg(ay=("y", "mean"))
+    assert sorted(got2["ay"].round(6)) == sorted(exp2["ay"].round(6).tolist())
+    got3 = bc.sql(
+        "with s as (select g, sum(y) as sy from t group by g) "
+        "select a.g, a.sy, b.sy as sy2 from s a join s b on a.g = b.g"
+    ).to_pandas()
+    assert len(got3) == df["g"].nunique()
+    assert np.allclose(got3["sy"], got3["sy2"])
+
+
+def test_select_list_scalar_subqueries():
+    """Scalar subqueries in the SELECT list: correlated bare value /
+    aggregate (LEFT-JOIN decorrelation, NULL when unmatched) and
+    uncorrelated (evaluate-once broadcast)."""
+    rng = np.random.default_rng(1)
+    n = 80
+    df = pd.DataFrame({"g": rng.choice(["a", "b", "c", "d", "e"], n),
+                       "x": rng.integers(0, 50, n),
+                       "y": rng.random(n) * 100})
+    o = pd.DataFrame({"g": ["a", "b", "c", "d"], "w": [1.0, 2.0, 3.0, 4.0]})
+    bc = BodoSQLContext({"t": df, "o": o})
+    got = bc.sql("select g, x, (select w from o where o.g = t.g) as w "
+                 "from t").to_pandas()
+    exp = df.merge(o, on="g", how="left")
+    assert len(got) == len(exp)
+    assert got["w"].isna().sum() == exp["w"].isna().sum()
+    got2 = bc.sql("select g, (select avg(y) from t t2 where t2.g = t.g) "
+                  "as ag from t").to_pandas()
+    m = df.groupby("g")["y"].mean()
+    assert all(abs(got2["ag"][i] - m[got2["g"][i]]) < 1e-9
+               for i in range(len(got2)))
+    got3 = bc.sql("select g, sum(x) as sx, (select max(w) from o) as mw "
+                  "from t group by g").to_pandas()
+    assert (got3["mw"] == 4.0).all()
