@@ -287,25 +287,40 @@ def _ordered_local_device(tbl: Table, keys, order_by, ascending,
                 res = cur
                 res_inv = torch.isinf(cur)
         elif func == "shift":
-            k = int(arg if arg is not None else 1)
+            k, dflt = (arg if isinstance(arg, tuple)
+                       else (int(arg if arg is not None else 1), None))
+            if dflt is not None and isinstance(dflt, str):
+                return None  # string default: host path
+            k = int(k)
             v, inv = sorted_vals(in_name)
             res = torch.full_like(v, 0)
             res_inv = torch.ones(n, dtype=torch.bool, device=dev)
+            # outside = offset crosses the partition edge (LAG default
+            # applies there; a NULL source value stays NULL)
+            outside = torch.ones(n, dtype=torch.bool, device=dev)
             if k >= 0:
                 if k < n:
                     res[k:] = v[:n - k]
-                    ok = segid[k:] == segid[:n - k]
-                    if inv is not None:
-                        ok &= ~inv[:n - k]
+                    inwin = segid[k:] == segid[:n - k]
+                    outside[k:] = ~inwin
+                    ok = inwin if inv is None else (inwin & ~inv[:n - k])
                     res_inv[k:] = ~ok
             else:
                 kk = -k
                 if kk < n:
                     res[:n - kk] = v[kk:]
-                    ok = segid[:n - kk] == segid[kk:]
-                    if inv is not None:
-                        ok &= ~inv[kk:]
+                    inwin = segid[:n - kk] == segid[kk:]
+                    outside[:n - kk] = ~inwin
+                    ok = inwin if inv is None else (inwin & ~inv[kk:])
                     res_inv[:n - kk] = ~ok
+            if dflt is not None:
+                fill = torch.full((), dflt, dtype=(
+                    v.dtype if v.dtype.is_floating_point
+                    else torch.float64), device=dev)
+                if not v.dtype.is_floating_point:
+                    res = res.to(torch.float64)
+                res = torch.where(outside, fill, res)
+                res_inv = res_inv & ~outside
         elif func in ("first_value", "last_value"):
             v, inv = sorted_vals(in_name)
             at = seg_start if func == "first_value" else end_pos[segid]
@@ -455,7 +470,11 @@ def _ordered_local(tbl: Table, keys, order_by, ascending, specs) -> Table:
             res = gb[col].rank(method=method,
                                ascending=ascending[0] if ascending else True)
         elif func == "shift":
-            res = gbs[in_name].shift(arg if arg is not None else 1)
+            if isinstance(arg, tuple):
+                k, dflt = arg
+                res = gbs[in_name].shift(int(k), fill_value=dflt)
+            else:
+                res = gbs[in_name].shift(arg if arg is not None else 1)
             res = res.reindex(pdf.index)
         elif func == "cumsum":
             res = gbs[in_name].cumsum().reindex(pdf.index)

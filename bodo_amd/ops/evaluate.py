@@ -253,7 +253,7 @@ class _Evaluator:
 
     def visit_Cast(self, e: Cast) -> Column:
         a = self.visit(e.operand)
-        return cast_column(a, e.to)
+        return cast_column(a, e.to, safe=getattr(e, "safe", False))
 
     def visit_ListOp(self, e: ListOp) -> Column:
         a = self.visit(e.operand)
@@ -841,9 +841,29 @@ def _dt_res(dtype, out, a: Column, fld: str) -> Column:
     return res
 
 
-def cast_column(a: Column, to: DType) -> Column:
+def cast_column(a: Column, to: DType, safe: bool = False) -> Column:
     if a.dtype == to:
         return a
+    if safe and a.dtype.kind in (TypeKind.STRING, TypeKind.DICT):
+        # TRY_CAST: coerce failures to NULL
+        ser = a.to_pandas()
+        if to.kind in (TypeKind.FLOAT64, TypeKind.FLOAT32, TypeKind.INT64,
+                       TypeKind.INT32):
+            num = pd.to_numeric(ser, errors="coerce")
+            import pyarrow as pa
+
+            if to.kind in (TypeKind.INT64, TypeKind.INT32):
+                arr = pa.Array.from_pandas(num.round())
+                arr = arr.cast(pa.int64() if to.kind == TypeKind.INT64
+                               else pa.int32(), safe=False)
+            else:
+                arr = pa.Array.from_pandas(num.astype("float64"))
+            return Column.from_arrow(arr, a.device)
+        if to.kind == TypeKind.TIMESTAMP_NS:
+            ts = pd.to_datetime(ser, errors="coerce")
+            import pyarrow as pa
+
+            return Column.from_arrow(pa.Array.from_pandas(ts), a.device)
     if a.dtype.kind == TypeKind.DECIMAL128:
         # scaled-int storage: cast through the VALUE, not the raw int
         f = decimal_to_float(a)

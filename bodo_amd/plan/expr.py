@@ -163,12 +163,13 @@ class DtField(Expr):
 class Cast(Expr):
     operand: Expr
     to: DType
+    safe: bool = False  # TRY_CAST: NULL instead of raising on bad values
 
     def children(self):
         return (self.operand,)
 
     def with_children(self, *ch):
-        return Cast(ch[0], self.to)
+        return Cast(ch[0], self.to, self.safe)
 
 
 @dataclass(frozen=True)

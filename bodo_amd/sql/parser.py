@@ -136,6 +136,7 @@ class IsNullE:
 class CastE:
     operand: Any
     to: str
+    safe: bool = False  # TRY_CAST: NULL on conversion failure
 
 
 @dataclass
@@ -712,6 +713,19 @@ class Parser:
             return e
         if t.kind == "id":
             nt = self.peek()
+            if nt and nt.kind == "op" and nt.value == "(" \
+                    and t.value.lower() in ("try_cast", "try_to_number",
+                                            "try_to_double"):
+                self.i += 1
+                e = self.parse_expr()
+                ty = "double"
+                if self.accept_kw("as"):
+                    ty = self.next().value.lower()
+                    if self.accept_op("("):
+                        while not self.accept_op(")"):
+                            self.next()
+                self.expect_op(")")
+                return CastE(e, ty, safe=True)
             if nt and nt.kind == "op" and nt.value == "(":
                 self.i += 1
                 distinct = bool(self.accept_kw("distinct"))

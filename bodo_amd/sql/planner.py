@@ -352,9 +352,15 @@ class Planner:
                 spec = (out, "", fn, None)
             elif fn in ("lag", "lead"):
                 n = 1
-                if len(w.args) > 1 and isinstance(w.args[1], ast.Lit):
-                    n = int(w.args[1].value)
-                spec = (out, arg_col, "shift", n if fn == "lag" else -n)
+                if len(w.args) > 1:
+                    n = _lit_int(w.args[1])
+                k = n if fn == "lag" else -n
+                if len(w.args) > 2:  # LAG(x, n, default)
+                    dflt = w.args[2].value if isinstance(
+                        w.args[2], ast.Lit) else _lit_int(w.args[2])
+                    spec = (out, arg_col, "shift", (k, dflt))
+                else:
+                    spec = (out, arg_col, "shift", k)
             elif fn in AGG_FUNCS or fn == "count":
                 frame = getattr(w, "frame_preceding", None)
                 if order and frame is not None and frame >= 0:
@@ -531,9 +537,15 @@ class Planner:
                 spec = (out, "", "ntile", k)
             elif fn in ("lag", "lead"):
                 n = 1
-                if len(w.args) > 1 and isinstance(w.args[1], ast.Lit):
-                    n = int(w.args[1].value)
-                spec = (out, arg_col, "shift", n if fn == "lag" else -n)
+                if len(w.args) > 1:
+                    n = _lit_int(w.args[1])
+                k = n if fn == "lag" else -n
+                if len(w.args) > 2:
+                    dflt = w.args[2].value if isinstance(
+                        w.args[2], ast.Lit) else _lit_int(w.args[2])
+                    spec = (out, arg_col, "shift", (k, dflt))
+                else:
+                    spec = (out, arg_col, "shift", k)
             elif fn in AGG_FUNCS or fn == "count":
                 if order:
                     if fn == "count" and (w.star or not w.args):
@@ -1172,7 +1184,8 @@ class Planner:
                                 int(e.args[1].value) if len(e.args) > 1 else 0)
         if isinstance(e, ast.CastE):
             return ex.Cast(self._post_agg_expr(e.operand, key_map, agg_map,
-                                               scope), CAST_TYPES[e.to])
+                                               scope), CAST_TYPES[e.to],
+                           getattr(e, "safe", False))
         if isinstance(e, ast.SubqueryE):
             full = self._try_plan_full(e.query)
             if full is None:
@@ -1230,7 +1243,8 @@ class Planner:
         if isinstance(e, ast.CastE):
             if e.to not in CAST_TYPES:
                 raise NotImplementedError(f"CAST to {e.to}")
-            return ex.Cast(self.expr(e.operand, scope), CAST_TYPES[e.to])
+            return ex.Cast(self.expr(e.operand, scope), CAST_TYPES[e.to],
+                           getattr(e, "safe", False))
         if isinstance(e, ast.ExtractE):
             fld = {"dow": "dayofweek", "doy": "dayofyear",
                    "isodow": "dayofweek"}.get(e.fld, e.fld)
@@ -1661,6 +1675,38 @@ class Planner:
                 # dedicated dt field: a CASE lowering leaked the ELSE value
                 # into NULL rows (string CASE has no validity mask)
                 return ex.DtField(self.expr(e.args[0], scope), name)
+            if name == "decode":
+                # DECODE(e, v1, r1 [, v2, r2 ...] [, default]) -> CASE
+                operand = self.expr(e.args[0], scope)
+                rest = e.args[1:]
+                pairs, default = [], None
+                if len(rest) % 2 == 1:
+                    default = rest[-1]
+                    rest = rest[:-1]
+                for i in range(0, len(rest), 2):
+                    pairs.append((rest[i], rest[i + 1]))
+                conds = tuple(ex.Cmp("eq", operand, self.expr(v, scope))
+                              for v, _ in pairs)
+                thens = tuple(self.expr(r, scope) for _, r in pairs)
+                other = (self.expr(default, scope) if default is not None
+                         else ex.Const(None, None))
+                return ex.Case(conds, thens, other)
+            if name == "width_bucket":
+                # WIDTH_BUCKET(x, lo, hi, n): 1..n inside, 0 below, n+1 above
+                x = self.expr(e.args[0], scope)
+                lo = float(e.args[1].value)
+                hi = float(e.args[2].value)
+                nb = _lit_int(e.args[3])
+                w = (hi - lo) / nb
+                raw = ex.BinOp("floordiv",
+                               ex.BinOp("sub", x, ex.Const(lo)),
+                               ex.Const(w))
+                bucket = ex.BinOp("add", raw, ex.Const(1))
+                return ex.Case(
+                    (ex.Cmp("lt", x, ex.Const(lo)),
+                     ex.Cmp("ge", x, ex.Const(hi))),
+                    (ex.Const(0), ex.Const(nb + 1)),
+                    bucket)
             if name == "last_day":
                 return ex.DtField(self.expr(e.args[0], scope), "last_day")
             if name in ("to_date", "try_to_date", "date"):

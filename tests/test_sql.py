@@ -928,3 +928,28 @@ def test_select_list_scalar_subqueries():
     got3 = bc.sql("select g, sum(x) as sx, (select max(w) from o) as mw "
                   "from t group by g").to_pandas()
     assert (got3["mw"] == 4.0).all()
+
+
+def test_try_cast_lag_default_decode_width_bucket():
+    df = pd.DataFrame({"g": ["a", "a", "b", "b", "c"],
+                       "v": [1.5, 2.5, 3.0, 4.0, 5.5],
+                       "s": ["12", "xx", "-7", None, "3.5"],
+                       "o": [3, 1, 2, 5, 4]})
+    bc = BodoSQLContext({"t": df})
+    tc = bc.sql("select try_cast(s as double) as r from t").to_pandas()["r"]
+    assert tc.tolist()[0] == 12.0 and np.isnan(tc[1]) and tc[4] == 3.5
+    sdf = df.sort_values("o")
+    lag = bc.sql("select lag(v, 1, -1.0) over (order by o) as r "
+                 "from t").to_pandas()["r"]
+    assert lag.tolist() == sdf["v"].shift(
+        1, fill_value=-1.0).reindex(df.index).tolist()
+    lead = bc.sql("select lead(v, 2, 0.0) over (order by o) as r "
+                  "from t").to_pandas()["r"]
+    assert lead.tolist() == sdf["v"].shift(
+        -2, fill_value=0.0).reindex(df.index).tolist()
+    dec = bc.sql("select decode(g, 'a', 1, 'b', 2, 0) as r "
+                 "from t").to_pandas()["r"]
+    assert dec.tolist() == [1, 1, 2, 2, 0]
+    wb = bc.sql("select width_bucket(v, 0, 10, 5) as r from t") \
+        .to_pandas()["r"]
+    assert wb.tolist() == [1, 2, 2, 3, 3]
