@@ -902,3 +902,42 @@ def test_dist_scalar_reduces():
     assert abs(got["sem"] - df["b"].sem()) < 1e-9
     assert pd.Timestamp(got["min_d"]) == df["d"].min()
     assert got["max_c"] == df["c"].max()
+
+
+def _q_round2_sql(bpd, rank, payload):
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": payload["df"], "o": payload["o"]})
+    a = bc.sql(
+        "with s as (select g, sum(y) as sy from t group by g) "
+        "select s.g, s.sy, o.w from s join o on s.g = o.g").to_pandas()
+    b = bc.sql(
+        "select g, min(y) over (partition by g order by x) as mn "
+        "from t").to_pandas()
+    c = bc.sql(
+        "select g, x, (select w from o where o.g = t.g) as w "
+        "from t").to_pandas()
+    return {"cte": a.sort_values("g").reset_index(drop=True),
+            "win_rows": len(b), "subq_nulls": int(c["w"].isna().sum()),
+            "subq_rows": len(c)}
+
+
+def test_dist_round2_sql_features():
+    """2-rank collective symmetry of the round-2 SQL plan shapes: CTE with
+    shared subtree, running window frame, SELECT-list decorrelated
+    subquery."""
+    rng = np.random.default_rng(3)
+    n = 400
+    df = pd.DataFrame({"g": rng.choice(["a", "b", "c", "d", "e"], n),
+                       "x": rng.integers(0, 50, n),
+                       "y": rng.random(n) * 100})
+    o = pd.DataFrame({"g": ["a", "b", "c", "d"], "w": [1.0, 2.0, 3.0, 4.0]})
+    got = run_dist(_q_round2_sql, {"df": df, "o": o})
+    exp_cte = df.groupby("g", as_index=False).agg(sy=("y", "sum")) \
+        .merge(o, on="g").sort_values("g").reset_index(drop=True)
+    g = got["cte"]
+    assert np.allclose(g["sy"], exp_cte["sy"]) and np.allclose(
+        g["w"], exp_cte["w"])
+    assert got["win_rows"] == n
+    assert got["subq_rows"] == n
+    assert got["subq_nulls"] == int((df["g"] == "e").sum())
