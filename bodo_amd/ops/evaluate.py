@@ -826,7 +826,12 @@ def _dt_field_cpu(a: Column, fld: str) -> Column:
         out = raw.to_numpy() if hasattr(raw, "to_numpy") else np.asarray(raw)
         dtype = _DT_OUT_TYPE[fld]
         out = out.astype(bt.numpy_storage_dtype(dtype))
-    res = Column(dtype, torch.from_numpy(out), a.mask)
+    mask = a.mask
+    if dtype.kind == TypeKind.BOOL and mask is not None:
+        # pandas quirk: boolean dt accessors report False for NaT
+        out = out & mask.numpy()
+        mask = None
+    res = Column(dtype, torch.from_numpy(np.ascontiguousarray(out)), mask)
     from .gpu import _DT_RANGE
 
     res.val_range = _DT_RANGE.get(fld)
