@@ -128,7 +128,7 @@ class DataFrameGroupBy:
     def size(self):
         out = self._build([("size", "", "size")])
         if isinstance(out, _IndexedAggResult):
-            ser = out._frame.to_pandas().set_index(self._keys)["size"]
+            ser = _decat(out._frame.to_pandas(), self._keys).set_index(self._keys)["size"]
             return ser
         return out
 
@@ -211,6 +211,36 @@ class DataFrameGroupBy:
         out = self._window([("__cc", "", "cumcount", None)])
         return out["__cc"]
 
+    def ngroup(self):
+        """Group number in sorted-key order (pandas sort=True default):
+        row-id over the sorted distinct keys, merged back."""
+        from .frame import BodoDataFrame
+
+        d = self._frame[list(self._keys)].drop_duplicates() \
+            .sort_values(list(self._keys))
+        rid = "__ng"
+        plan = pn.RowId(d._lazy_plan, rid)
+        dk = BodoDataFrame(plan, list(self._keys) + [rid])
+        out = self._frame.merge(dk, on=list(self._keys), how="left")
+        return out[rid]
+
+    def sample(self, n=None, frac=None, random_state=None):
+        keys, dropna = self._keys, self._dropna
+
+        def _part(pdf):
+            if len(pdf) == 0:
+                return pdf
+            return pdf.groupby(keys, dropna=dropna, sort=False,
+                               group_keys=False).sample(
+                n=n, frac=frac, random_state=random_state)
+
+        from .frame import BodoDataFrame
+
+        shuffled = pn.ShuffleByKey(self._frame._lazy_plan, tuple(keys))
+        plan = pn.MapPartitions(shuffled, _part, (),
+                                tuple(self._frame._columns))
+        return BodoDataFrame(plan, list(self._frame._columns))
+
     def apply(self, func, *args, **kwargs):
         # general groupby-apply: shuffle rows by key then run pandas apply
         # per shard (keys co-located so results are exact)
@@ -259,7 +289,7 @@ class SeriesGroupBy:
                               self._dropna, self._sort, [self._col])
         out = gb._build([(self._col, self._col, func)])
         if isinstance(out, _IndexedAggResult):
-            pdf = out._frame.to_pandas().set_index(self._keys)[self._col]
+            pdf = _decat(out._frame.to_pandas(), self._keys).set_index(self._keys)[self._col]
             return pdf
         # as_index=False: pandas returns a DataFrame of keys + the column
         return out
@@ -322,6 +352,86 @@ class SeriesGroupBy:
         return self._window1("rank", arg=method,
                              ascending=(ascending,))
 
+    def _cum_skipna(self, func):
+        # pandas cummin/cummax: running extreme skipping NaN, but the
+        # output at a NaN input row is NaN (SQL reports the frame value
+        # there — mask it back)
+        from .frame import BodoDataFrame
+
+        plan = pn.Window(self._frame._lazy_plan, tuple(self._keys),
+                         (), (), (("__w", self._col, func, None),))
+        fr = BodoDataFrame(plan, list(self._frame._columns) + ["__w"])
+        return fr["__w"].where(fr[self._col].notna(), float("nan"))
+
+    def cummin(self):
+        return self._cum_skipna("cummin")
+
+    def cummax(self):
+        return self._cum_skipna("cummax")
+
+    def pct_change(self, periods=1):
+        # fill_method=None semantics (the pandas-3.0 default): NaN inputs
+        # propagate instead of being padded forward
+        from .frame import BodoDataFrame
+
+        plan = pn.Window(self._frame._lazy_plan, tuple(self._keys),
+                         (), (), (("__w", self._col, "shift", periods),))
+        fr = BodoDataFrame(plan, list(self._frame._columns) + ["__w"])
+        return fr[self._col] / fr["__w"] - 1
+
+    def _idx_of(self, which):
+        """Global row position of the group max/min (the drop-in contract:
+        frames carry positional range indexes)."""
+        from .frame import BodoDataFrame
+
+        rid = "__rid"
+        tagged = pn.RowId(self._frame._lazy_plan, rid)
+        w = pn.Window(tagged, tuple(self._keys), (), (),
+                      (("__m", self._col, f"transform_{which}", None),))
+        fr = BodoDataFrame(w, list(self._frame._columns) + [rid, "__m"])
+        sel = fr[fr[self._col] == fr["__m"]]
+        gb = SeriesGroupBy(sel, self._keys, rid, self._as_index,
+                           self._dropna, self._sort)
+        out = gb.min()  # ties: first occurrence, like pandas
+        return out.sort_index() if isinstance(out, pd.Series) else out
+
+    def idxmax(self):
+        return self._idx_of("max")
+
+    def idxmin(self):
+        return self._idx_of("min")
+
+    def expanding(self, min_periods=1):
+        return _GroupExpanding(self)
+
+    def describe(self):
+        gb = DataFrameGroupBy(self._frame, self._keys, self._as_index,
+                              self._dropna, self._sort, [self._col])
+        return gb.apply(lambda d, _c=self._col: d[_c].describe())
+
+
+class _GroupExpanding:
+    """groupby(...).col.expanding(): running aggregates in row order (the
+    pandas MultiIndex wrapper is flattened to frame row order)."""
+
+    def __init__(self, sgb: "SeriesGroupBy"):
+        self._g = sgb
+
+    def mean(self):
+        return self._g._window1("cummean")
+
+    def sum(self):
+        return self._g._window1("cumsum")
+
+    def min(self):
+        return self._g._window1("cummin")
+
+    def max(self):
+        return self._g._window1("cummax")
+
+    def count(self):
+        return self._g._window1("cumcount_v")
+
 
 class _IndexedAggResult:
     """as_index=True result: behaves like the BodoDataFrame but materializes
@@ -332,7 +442,7 @@ class _IndexedAggResult:
         self._keys = keys
 
     def to_pandas(self):
-        return self._frame.to_pandas().set_index(self._keys)
+        return _decat(self._frame.to_pandas(), self._keys).set_index(self._keys)
 
     def reset_index(self):
         return self._frame
@@ -360,3 +470,12 @@ def _norm_func(f):
         return f  # custom callable: single-phase pandas agg on shuffled groups
     m = {"average": "mean", "nunique": "nunique"}
     return m.get(f, f)
+
+def _decat(pdf, keys):
+    """Dict-encoded keys arrive as Categorical; a CategoricalIndex sorts by
+    category order, so decode key columns to values for pandas parity."""
+    for k in keys:
+        if k in pdf.columns and isinstance(pdf[k].dtype, pd.CategoricalDtype):
+            pdf = pdf.copy()
+            pdf[k] = pdf[k].astype(object)
+    return pdf

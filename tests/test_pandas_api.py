@@ -665,3 +665,35 @@ def test_crosstab_and_pivot_index_only():
     gotp = b.pivot_table(values="y", index="g", aggfunc="mean")
     expp = df.pivot_table(values="y", index="g", aggfunc="mean")
     np.testing.assert_allclose(gotp.values, expp.values)
+
+
+def test_groupby_extended_methods():
+    """ngroup/idxmax/idxmin/cummin/cummax/pct_change/sample/expanding/
+    describe on groupby (round-2 coverage)."""
+    rng = np.random.default_rng(2)
+    df = pd.DataFrame({"g": rng.choice(["a", "b", "c"], 120),
+                       "x": rng.integers(0, 50, 120),
+                       "y": np.where(rng.random(120) < 0.1, np.nan,
+                                     rng.random(120))})
+    b = bpd.from_pandas(df)
+    assert np.asarray(b.groupby("g").ngroup().to_pandas()).tolist() == \
+        df.groupby("g").ngroup().tolist()
+    for m in ("cummin", "cummax"):
+        got = getattr(b.groupby("g")["y"], m)().to_pandas()
+        want = getattr(df.groupby("g")["y"], m)()
+        np.testing.assert_allclose(got.fillna(-9e9), want.fillna(-9e9))
+    got = b.groupby("g")["y"].pct_change().to_pandas()
+    want = df.groupby("g")["y"].pct_change(fill_method=None)
+    np.testing.assert_allclose(got.fillna(-9e9), want.fillna(-9e9))
+    assert list(b.groupby("g")["y"].idxmax()) == \
+        list(df.groupby("g")["y"].idxmax())
+    assert list(b.groupby("g")["y"].idxmin()) == \
+        list(df.groupby("g")["y"].idxmin())
+    s = b.groupby("g").sample(2, random_state=0).to_pandas()
+    assert len(s) == 6
+    got = b.groupby("g")["y"].expanding().mean().to_pandas()
+    want = df.groupby("g")["y"].transform(lambda t: t.expanding(1).mean())
+    np.testing.assert_allclose(got.fillna(-9e9), want.fillna(-9e9))
+    d = b.groupby("g")["y"].describe()
+    d = d.to_pandas() if hasattr(d, "to_pandas") else d
+    assert len(d) == 3 and "mean" in d.columns
