@@ -26,7 +26,7 @@ def _np_from_tensor(t: torch.Tensor) -> np.ndarray:
 
 class Column:
     __slots__ = ("dtype", "data", "mask", "offsets", "dictionary", "name",
-                 "_length", "val_range", "child")
+                 "_length", "val_range", "child", "children")
 
     def __init__(
         self,
@@ -43,6 +43,7 @@ class Column:
         self.offsets = offsets  # STRING only: int64 (n+1)
         self.dictionary = dictionary  # DICT only: pa.StringArray (host)
         self.child = None  # LIST only: element Column
+        self.children = None  # STRUCT only: field Columns (dtype.fields names)
         self.val_range = None  # optional known (lo, hi) for int columns
         if length is not None:
             self._length = length
@@ -57,6 +58,11 @@ class Column:
     @property
     def device(self) -> torch.device:
         t = self.data if self.data is not None else self.offsets
+        if t is None:
+            if self.children:  # STRUCT: no buffers of its own
+                return self.children[0].device
+            return self.mask.device if self.mask is not None \
+                else torch.device("cpu")
         return t.device
 
     @property
@@ -70,6 +76,8 @@ class Column:
                 n += t.numel() * t.element_size()
         if self.child is not None:
             n += self.child.nbytes()
+        if self.children is not None:
+            n += sum(c.nbytes() for c in self.children)
         return n
 
     # ------------------------------------------------------------------
@@ -95,6 +103,8 @@ class Column:
         out.val_range = self.val_range
         if self.child is not None:
             out.child = self.child.to_device(device)
+        if self.children is not None:
+            out.children = [c.to_device(device) for c in self.children]
         return out
 
     # ------------------------------------------------------------------
@@ -159,6 +169,23 @@ class Column:
                     np.ascontiguousarray(offs)).to(device),
                 length=len(arr))
             out.child = child
+            return out
+        if pa.types.is_struct(t):
+            # arrow struct<fields>: recursively converted field columns +
+            # struct-level validity (reference: struct_arr_ext.py layout)
+            arr = arr.combine_chunks() if isinstance(arr, pa.ChunkedArray) else arr
+            flat = arr.flatten()  # offset/validity-adjusted field arrays
+            names = [t.field(i).name for i in range(t.num_fields)]
+            children = [Column.from_arrow(f, device) for f in flat]
+            mask = None
+            if arr.buffers()[0] is not None and arr.null_count:
+                mask = _unpack_validity(arr.buffers()[0], arr.offset,
+                                        len(arr))
+            out = Column(
+                bt.struct_(names), None,
+                None if mask is None else torch.from_numpy(mask).to(device),
+                length=len(arr))
+            out.children = children
             return out
         if pa.types.is_null(t):
             # typeless all-null column: represent as float64 NaN
@@ -240,6 +267,12 @@ class Column:
             return pa.LargeListArray.from_arrays(
                 pa.array(offsets, type=pa.int64()), child,
                 mask=None if mask is None else pa.array(~mask))
+        if k == TypeKind.STRUCT:
+            fields = [c.to_arrow() for c in self.children]
+            mask = None if self.mask is None else ~_np_from_tensor(self.mask)
+            return pa.StructArray.from_arrays(
+                fields, list(self.dtype.fields),
+                mask=None if mask is None else pa.array(mask))
         if k == TypeKind.STRING:
             offsets = _np_from_tensor(self.offsets)
             data = _np_from_tensor(self.data) if self.data is not None else np.zeros(0, np.uint8)

@@ -36,6 +36,7 @@ class TypeKind(enum.IntEnum):
     UINT32 = 14
     UINT64 = 15
     LIST = 16  # arrow list<child>: int64 offsets (n+1) + child column
+    STRUCT = 17  # arrow struct<fields>: named child columns + validity
 
 
 @dataclass(frozen=True)
@@ -44,6 +45,8 @@ class DType:
     # for DECIMAL128
     precision: int = 0
     scale: int = 0
+    # for STRUCT: field names (children carry their own dtypes)
+    fields: tuple = ()
 
     @property
     def is_numeric(self) -> bool:
@@ -102,6 +105,10 @@ list_ = DType(TypeKind.LIST)
 
 def decimal128(precision: int, scale: int) -> DType:
     return DType(TypeKind.DECIMAL128, precision, scale)
+
+
+def struct_(field_names) -> DType:
+    return DType(TypeKind.STRUCT, fields=tuple(field_names))
 
 
 # torch storage dtype for the primary data tensor of each kind

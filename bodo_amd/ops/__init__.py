@@ -29,6 +29,12 @@ def gather(col: Column, idx: torch.Tensor) -> Column:
     n = int(idx.shape[0])
     if col.dtype.kind == TypeKind.LIST:
         return _gather_list(col, idx)
+    if col.dtype.kind == TypeKind.STRUCT:
+        out = Column(col.dtype, None,
+                     col.mask[idx] if col.mask is not None else None,
+                     length=n)
+        out.children = [gather(c, idx) for c in col.children]
+        return out
     if len(col) == 0 and n > 0:
         # gathering from an empty column only happens for null-padded rows
         # (outer-join unmatched side): produce an all-null column
@@ -218,6 +224,12 @@ def concat_columns(cols: Sequence[Column]) -> Column:
         out = Column(_bt.list_, None, masks, offsets=torch.cat(offs),
                      length=n)
         out.child = concat_columns([c.child for c in cols])
+        return out
+    if first.dtype.kind == TypeKind.STRUCT:
+        out = Column(first.dtype, None, masks, length=n)
+        out.children = [
+            concat_columns([c.children[i] for c in cols])
+            for i in range(len(first.children))]
         return out
     if first.dtype.kind == TypeKind.STRING:
         datas, offs, base = [], [torch.zeros(1, dtype=torch.int64, device=first.device)], 0

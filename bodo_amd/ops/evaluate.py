@@ -257,6 +257,21 @@ class _Evaluator:
 
     def visit_ListOp(self, e: ListOp) -> Column:
         a = self.visit(e.operand)
+        if a.dtype.kind == TypeKind.STRUCT and e.op == "get":
+            # struct field extraction: GET(col, 'name') / col['name']
+            name = str(e.arg)
+            if name not in a.dtype.fields:
+                raise KeyError(f"struct has no field {name!r}; "
+                               f"fields: {a.dtype.fields}")
+            ch = a.children[a.dtype.fields.index(name)]
+            if a.mask is not None:
+                m = a.mask if ch.mask is None else (ch.mask & a.mask)
+                out = Column(ch.dtype, ch.data, m, ch.offsets,
+                             ch.dictionary, len(ch))
+                out.child = ch.child
+                out.children = ch.children
+                return out
+            return ch
         if a.dtype.kind != TypeKind.LIST:
             raise TypeError(f"list op {e.op} on {a.dtype}")
         off = a.offsets

@@ -215,6 +215,10 @@ class BodoSeries:
     def list(self):
         return _ListAccessor(self)
 
+    @property
+    def struct(self):
+        return _StructAccessor(self)
+
     # ------------------------------------------------------------------
     # reductions (lazy scalars: the plan executes on first value access —
     # reference: BodoScalar, bodo/pandas/scalar.py:14)
@@ -861,6 +865,19 @@ class _ListAccessor:
         return self._s._wrap(ListOp(self._s._expr, "get", int(i)), None)
 
     __getitem__ = get
+
+
+class _StructAccessor:
+    """Series.struct over STRUCT columns (pandas ArrowDtype .struct
+    surface; reference role: struct_arr_ext field access)."""
+
+    def __init__(self, s: BodoSeries):
+        self._s = s
+
+    def field(self, name: str):
+        return self._s._wrap(ListOp(self._s._expr, "get", str(name)), name)
+
+    __getitem__ = field
 
 
 class _SplitResult:

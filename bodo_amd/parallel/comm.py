@@ -211,7 +211,8 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
     # (contiguous-split pack, reference: GpuShuffleManager
     #  cudf::contiguous_split + per-destination Isend, gpu_utils.cpp:96-122)
     fixed_idx = [i for i, c in enumerate(packed.columns)
-                 if c.dtype.kind not in (TypeKind.STRING, TypeKind.LIST)]
+                 if c.dtype.kind not in (TypeKind.STRING, TypeKind.LIST,
+                                         TypeKind.STRUCT)]
     if fixed_idx:
         bnd_l = bnd.tolist()
         dev0 = packed.columns[fixed_idx[0]].device if fixed_idx else None
@@ -282,6 +283,11 @@ def shuffle_table(tbl: Table, part_ids: torch.Tensor) -> Table:
         out_cols[i] = _shuffle_list_column(
             packed.columns[i], send_counts, recv_counts, bnd_l2,
             bool(has_mask[i]))
+    # ---- structs: per-field exchanges (fields are row-aligned) ----
+    for i, col in enumerate(packed.columns):
+        if col.dtype.kind == TypeKind.STRUCT:
+            out_cols[i] = _shuffle_struct_column(
+                col, send_counts, recv_counts, bnd_l2, bool(has_mask[i]))
     # ---- strings: per-column exchanges (offsets + bytes + mask) ----
     sj = 0
     for i, col in enumerate(packed.columns):
@@ -352,6 +358,55 @@ def _shuffle_list_column(col: Column, send_counts, recv_counts,
     new_mask = _shuffle_mask(col, send_counts, recv_counts, has_mask)
     out = Column(_bt.list_, None, new_mask, offsets=new_off, length=n_out)
     out.child = new_child
+    return out
+
+
+def _shuffle_struct_column(col: Column, send_counts, recv_counts,
+                           bnd_rows, has_mask: bool) -> Column:
+    """STRUCT exchange: every field is row-aligned with the parent, so each
+    shuffles with the parent's counts; field mask presence uses a MAX
+    all-reduce so the collective sequence stays SPMD-symmetric even when a
+    rank's shard has no nulls (reference: struct shuffle in
+    bodo/libs/_shuffle.cpp)."""
+    n_out = sum(recv_counts)
+    dev = _comm_device()
+    flags = torch.tensor(
+        [1 if ch.mask is not None else 0 for ch in col.children],
+        dtype=torch.int64, device=dev)
+    if flags.numel():
+        dist.all_reduce(flags, op=dist.ReduceOp.MAX)
+    child_flags = [bool(v) for v in flags.cpu().tolist()]
+    new_children = []
+    for ch, ch_has_mask in zip(col.children, child_flags):
+        if ch.dtype.kind == TypeKind.STRING:
+            coff = ch.offsets
+            cat = coff[torch.tensor(bnd_rows, device=coff.device)]
+            byte_send = (cat[1:] - cat[:-1]).cpu().tolist()
+            byte_recv = _exchange_count_vector(byte_send)
+            new_children.append(_shuffle_string_column(
+                ch, send_counts, recv_counts, byte_send, byte_recv,
+                ch_has_mask))
+        elif ch.dtype.kind == TypeKind.LIST:
+            new_children.append(_shuffle_list_column(
+                ch, send_counts, recv_counts, bnd_rows, ch_has_mask))
+        elif ch.dtype.kind == TypeKind.STRUCT:
+            new_children.append(_shuffle_struct_column(
+                ch, send_counts, recv_counts, bnd_rows, ch_has_mask))
+        else:
+            data = alltoallv_tensor(ch.data.contiguous(), send_counts,
+                                    recv_counts)
+            cmask = None
+            if ch_has_mask:
+                m = ch.mask if ch.mask is not None else torch.ones(
+                    len(ch), dtype=torch.bool, device=ch.device)
+                cmask = alltoallv_tensor(m.contiguous(), send_counts,
+                                         recv_counts)
+            new_children.append(Column(ch.dtype, data, cmask,
+                                       dictionary=ch.dictionary,
+                                       length=n_out))
+    new_mask = _shuffle_mask(col, send_counts, recv_counts, has_mask)
+    out = Column(col.dtype, None, new_mask, length=n_out)
+    out.children = new_children
     return out
 
 
@@ -447,6 +502,8 @@ def _pack_table(tbl: Table, dev: torch.device):
             m["dict"] = c.dictionary.to_pylist()
         if c.child is not None:
             m["child"] = col_meta(c.child)
+        if c.children is not None:
+            m["children"] = [col_meta(ch) for ch in c.children]
         return m
 
     for c in tbl.columns:
@@ -481,6 +538,8 @@ def _unpack_table(buf: torch.Tensor, meta: dict, device) -> Table:
                      length=m["n"])
         if "child" in m:
             col.child = mk_col(m["child"])
+        if "children" in m:
+            col.children = [mk_col(x) for x in m["children"]]
         return col
 
     cols = []

@@ -1637,10 +1637,12 @@ class Planner:
 
                 return _LO(self.expr(e.args[0], scope), "len")
             if name == "get" and len(e.args) == 2:
+                # GET(list, i) for lists, GET(struct, 'field') for structs
                 from ..plan.expr import ListOp as _LO
 
+                key = e.args[1].value
                 return _LO(self.expr(e.args[0], scope), "get",
-                           int(e.args[1].value))
+                           key if isinstance(key, str) else int(key))
             if name in ("json_extract_path_text", "get_path"):
                 import json as _json
 

@@ -941,3 +941,26 @@ def test_dist_round2_sql_features():
     assert got["win_rows"] == n
     assert got["subq_rows"] == n
     assert got["subq_nulls"] == int((df["g"] == "e").sum())
+
+
+def _q_struct_shuffle(bpd, rank, payload):
+    b = bpd.from_pandas(payload["df"])
+    out = b.sort_values("k").to_pandas()  # range partition => shuffle
+    return out.reset_index(drop=True)
+
+
+def test_dist_struct_shuffle():
+    """STRUCT columns survive the packed shuffle (2 ranks): per-field
+    exchanges incl. a string field and the struct validity mask."""
+    rng = np.random.default_rng(11)
+    n = 60
+    df = pd.DataFrame({
+        "k": rng.permutation(n),
+        "st": [None if i % 7 == 0 else
+               {"x": int(i), "y": f"s{i}" if i % 3 else None}
+               for i in range(n)],
+    })
+    got = run_dist(_q_struct_shuffle, {"df": df})
+    exp = df.sort_values("k").reset_index(drop=True)
+    assert got["k"].tolist() == exp["k"].tolist()
+    assert got["st"].tolist() == exp["st"].tolist()

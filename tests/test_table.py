@@ -52,3 +52,42 @@ def test_normalize_device_cuda_index():
     d = _normalize_device("cuda")
     assert d.type == "cuda" and d.index is not None
     assert _normalize_device("cpu") == torch.device("cpu")
+
+
+def test_struct_column_roundtrip_and_ops():
+    """STRUCT columns: arrow roundtrip, gather, concat, field extraction
+    (reference: struct_arr_ext.py layout)."""
+    import pyarrow as pa
+    import torch
+
+    from bodo_amd import ops
+    from bodo_amd.core.column import Column
+
+    a = pa.array([{"x": 1, "y": "p"}, {"x": 2, "y": "q"}, None,
+                  {"x": 4, "y": None}])
+    c = Column.from_arrow(a)
+    assert len(c) == 4 and c.dtype.fields == ("x", "y")
+    assert c.to_arrow().to_pylist() == a.to_pylist()
+    g = ops.gather(c, torch.tensor([3, 1, 2]))
+    assert g.to_arrow().to_pylist() == [a[3].as_py(), a[1].as_py(), None]
+    cc = ops.concat_columns([c, g])
+    assert len(cc) == 7 and cc.to_arrow().to_pylist()[4] == a[3].as_py()
+
+
+def test_struct_frontend_and_sql():
+    import bodo_amd.pandas as bpd
+    from bodo_amd.sql import BodoSQLContext
+
+    df = pd.DataFrame({"k": [1, 2, 3, 4],
+                       "st": [{"x": 1, "y": "p"}, {"x": 2, "y": "q"},
+                              None, {"x": 4, "y": None}]})
+    b = bpd.from_pandas(df)
+    assert b["st"].struct.field("x").to_pandas().fillna(-1).tolist() == \
+        [1, 2, -1, 4]
+    bc = BodoSQLContext({"t": df})
+    o = bc.sql("select k, get(st, 'y') as y from t order by k").to_pandas()
+    assert o["y"].where(o["y"].notna(), None).tolist() == \
+        ["p", "q", None, None]
+    o2 = bc.sql("select k, st from t where k >= 2 order by k desc") \
+        .to_pandas()
+    assert o2["st"].tolist()[0] == {"x": 4, "y": None}
