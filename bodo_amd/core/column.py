@@ -170,6 +170,17 @@ class Column:
                 length=len(arr))
             out.child = child
             return out
+        if pa.types.is_map(t):
+            # arrow map<k,v> is physically list<struct<key,value>>: reuse
+            # both layouts (reference: map_arr_ext.py over array_item+struct)
+            st = pa.StructArray.from_arrays(
+                [arr.keys, arr.items], ["key", "value"])
+            mask = None
+            if arr.null_count:
+                mask = pa.array([not v for v in arr.is_valid().to_pylist()])
+            lst = pa.LargeListArray.from_arrays(
+                arr.offsets.cast(pa.int64()), st, mask=mask)
+            return Column.from_arrow(lst, device)
         if pa.types.is_struct(t):
             # arrow struct<fields>: recursively converted field columns +
             # struct-level validity (reference: struct_arr_ext.py layout)

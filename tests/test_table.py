@@ -91,3 +91,19 @@ def test_struct_frontend_and_sql():
     o2 = bc.sql("select k, st from t where k >= 2 order by k desc") \
         .to_pandas()
     assert o2["st"].tolist()[0] == {"x": 4, "y": None}
+
+
+def test_map_column_via_list_struct():
+    """MAP columns load as list<struct<key,value>> (the physical arrow
+    layout; reference: map_arr_ext.py)."""
+    import pyarrow as pa
+
+    from bodo_amd.core.column import Column
+
+    m = pa.array([[("a", 1), ("b", 2)], None, [("c", 3)]],
+                 type=pa.map_(pa.string(), pa.int64()))
+    c = Column.from_arrow(m)
+    assert len(c) == 3
+    assert c.to_arrow().to_pylist() == [
+        [{"key": "a", "value": 1}, {"key": "b", "value": 2}], None,
+        [{"key": "c", "value": 3}]]
