@@ -1002,6 +1002,15 @@ def _str_dict_generic(a: Column, op: str, args, kwargs) -> Column:
 def str_op(a: Column, op: str, args, kwargs=None) -> Column:
     import pyarrow.compute as pc
 
+    if op == "to_datetime":
+        # string -> timestamp with an explicit strptime format / errors
+        # policy (bpd.to_datetime(format=..., errors=...))
+        import pyarrow as pa
+
+        fmt, errors = args
+        ser = a.to_pandas()
+        res = pd.to_datetime(ser, format=fmt, errors=errors or "raise")
+        return Column.from_arrow(pa.Array.from_pandas(res), a.device)
     if a.dtype.kind == TypeKind.DICT:
         # operate on the (small) dictionary, keep indices
         d = a.dictionary

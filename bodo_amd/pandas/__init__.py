@@ -251,13 +251,49 @@ def crosstab(index, columns, values=None, aggfunc=None,
                            aggfunc="first").sort_index()
 
 
-def to_datetime(arg, **kwargs):
+def to_datetime(arg, format=None, errors=None, **kwargs):
     if isinstance(arg, BodoSeries):
         from ..core import types as bt
-        from ..plan.expr import Cast
+        from ..plan.expr import Cast, StrOp
 
+        if format is not None or errors is not None:
+            return arg._wrap(StrOp(arg._expr, "to_datetime",
+                                   (format, errors)), arg.name)
         return arg._wrap(Cast(arg._expr, bt.timestamp_ns), arg.name)
+    if format is not None:
+        kwargs["format"] = format
+    if errors is not None:
+        kwargs["errors"] = errors
     return _pd.to_datetime(arg, **kwargs)
+
+
+def merge_asof(left, right, on=None, by=None, direction="backward",
+               suffixes=("_x", "_y")):
+    """Distributed merge_asof: the (small) right side replicates to every
+    rank; each locally sorted left shard then matches exactly with pandas
+    merge_asof (reference: pd.merge_asof via bodo's sorted-merge path)."""
+    from .frame import BodoDataFrame
+    from ..plan import nodes as pn
+
+    if not isinstance(left, BodoDataFrame):
+        return _pd.merge_asof(left, right, on=on, by=by,
+                              direction=direction, suffixes=suffixes)
+    r_pd = right.to_pandas() if isinstance(right, BodoDataFrame) else right
+    r_pd = r_pd.sort_values(on).reset_index(drop=True)
+    sorted_left = left.sort_values(on)
+
+    def _part(pdf, _r=r_pd):
+        if len(pdf) == 0:
+            out = _pd.merge_asof(pdf, _r, on=on, by=by,
+                                 direction=direction, suffixes=suffixes)
+            return out
+        return _pd.merge_asof(pdf, _r, on=on, by=by, direction=direction,
+                              suffixes=suffixes)
+
+    probe = _part(left.head(0).to_pandas())
+    names = list(probe.columns)
+    plan = pn.MapPartitions(sorted_left._lazy_plan, _part, (), tuple(names))
+    return BodoDataFrame(plan, names)
 
 
 def __getattr__(name):

@@ -697,3 +697,19 @@ def test_groupby_extended_methods():
     d = b.groupby("g")["y"].describe()
     d = d.to_pandas() if hasattr(d, "to_pandas") else d
     assert len(d) == 3 and "mean" in d.columns
+
+
+def test_merge_asof_and_to_datetime_options():
+    l = pd.DataFrame({"t": [1, 5, 10, 3, 8], "v": [1.0, 2.0, 3.0, 4.0, 5.0]})
+    r = pd.DataFrame({"t": [0, 4, 9], "w": [10, 20, 30]})
+    got = bpd.merge_asof(bpd.from_pandas(l), bpd.from_pandas(r),
+                         on="t").to_pandas().reset_index(drop=True)
+    exp = pd.merge_asof(l.sort_values("t"), r,
+                        on="t").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+    s = bpd.from_pandas(pd.DataFrame({"s": ["01/02/2024", None]}))["s"]
+    out = bpd.to_datetime(s, format="%d/%m/%Y").to_pandas()
+    assert out.iloc[0] == pd.Timestamp("2024-02-01") and pd.isna(out.iloc[1])
+    s2 = bpd.from_pandas(pd.DataFrame({"s": ["2024-01-01", "bad"]}))["s"]
+    out2 = bpd.to_datetime(s2, errors="coerce").to_pandas()
+    assert out2.iloc[0] == pd.Timestamp("2024-01-01") and pd.isna(out2.iloc[1])
