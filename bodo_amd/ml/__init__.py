@@ -379,3 +379,150 @@ class KMeans:
 
 from .gbdt import (GradientBoostingClassifier,  # noqa: E402,F401
                    GradientBoostingRegressor)
+
+
+class MinMaxScaler:
+    """Distributed min-max scaling: per-feature min/max via all-reduce
+    (reference: sklearn_ext.py preprocessing overloads)."""
+
+    def __init__(self, feature_range=(0.0, 1.0)):
+        self.feature_range = feature_range
+        self.data_min_: Optional[np.ndarray] = None
+        self.data_max_: Optional[np.ndarray] = None
+
+    def fit(self, X):
+        from .. import config
+        from ..parallel import comm
+
+        dev = torch.device(config.default_device())
+        Xl = _as_matrix(X, dev).to(torch.float64)
+        lo = Xl.min(dim=0).values if len(Xl) else torch.full(
+            (Xl.shape[1],), float("inf"), dtype=torch.float64, device=dev)
+        hi = Xl.max(dim=0).values if len(Xl) else torch.full(
+            (Xl.shape[1],), float("-inf"), dtype=torch.float64, device=dev)
+        if comm.initialized() and comm.get_world_size() > 1:
+            import torch.distributed as dist
+
+            dist.all_reduce(lo, op=dist.ReduceOp.MIN)
+            dist.all_reduce(hi, op=dist.ReduceOp.MAX)
+        self.data_min_ = lo.cpu().numpy()
+        self.data_max_ = hi.cpu().numpy()
+        return self
+
+    def transform(self, X):
+        from .. import config
+
+        dev = torch.device(config.default_device())
+        Xl = _as_matrix(X, dev).to(torch.float64).cpu().numpy()
+        rng = self.data_max_ - self.data_min_
+        rng[rng == 0.0] = 1.0
+        a, b = self.feature_range
+        return (Xl - self.data_min_) / rng * (b - a) + a
+
+    def fit_transform(self, X):
+        return self.fit(X).transform(X)
+
+
+class LabelEncoder:
+    """Distributed label encoding: the class set is the union over shards
+    (allgather of locally unique labels)."""
+
+    def __init__(self):
+        self.classes_: Optional[np.ndarray] = None
+
+    def fit(self, y):
+        from ..parallel import comm
+
+        vals = np.asarray(y)
+        uniq = set(pd_unique_list(vals))
+        if comm.initialized() and comm.get_world_size() > 1:
+            parts = comm.allgather_obj(sorted(uniq, key=str))
+            uniq = set()
+            for p in parts:
+                uniq |= set(p)
+        self.classes_ = np.array(sorted(uniq, key=str))
+        return self
+
+    def transform(self, y):
+        lut = {v: i for i, v in enumerate(self.classes_)}
+        return np.array([lut[v] for v in np.asarray(y)], dtype=np.int64)
+
+    def fit_transform(self, y):
+        return self.fit(y).transform(y)
+
+    def inverse_transform(self, codes):
+        return self.classes_[np.asarray(codes, dtype=np.int64)]
+
+
+def pd_unique_list(vals):
+    import pandas as pd
+
+    return pd.unique(pd.Series(vals).dropna()).tolist()
+
+
+def train_test_split(X, y=None, test_size=0.25, random_state=None,
+                     shuffle=True):
+    """Shard-local split with a shared seed (each rank splits its block;
+    reference: sklearn_ext.py train_test_split overload)."""
+    n = len(X)
+    idx = np.arange(n)
+    if shuffle:
+        rng = np.random.default_rng(random_state)
+        rng.shuffle(idx)
+    n_test = int(round(n * test_size)) if test_size < 1 else int(test_size)
+    test_i, train_i = idx[:n_test], idx[n_test:]
+
+    def take(a, i):
+        if hasattr(a, "iloc"):
+            return a.iloc[i]
+        return np.asarray(a)[i]
+
+    if y is None:
+        return take(X, train_i), take(X, test_i)
+    return take(X, train_i), take(X, test_i), take(y, train_i), take(y, test_i)
+
+
+# ---------------------------------------------------------------------------
+# metrics (distributed: partials all-reduced)
+# ---------------------------------------------------------------------------
+
+def _metric_partials(vals):
+    from .. import config
+
+    dev = torch.device(config.default_device())
+    t = torch.tensor(vals, dtype=torch.float64, device=dev)
+    _allreduce_(t)
+    return t.cpu().numpy()
+
+
+def accuracy_score(y_true, y_pred) -> float:
+    yt, yp = np.asarray(y_true), np.asarray(y_pred)
+    p = _metric_partials([float((yt == yp).sum()), float(len(yt))])
+    return p[0] / p[1] if p[1] else 0.0
+
+
+def mean_squared_error(y_true, y_pred) -> float:
+    yt = np.asarray(y_true, dtype=np.float64)
+    yp = np.asarray(y_pred, dtype=np.float64)
+    p = _metric_partials([float(((yt - yp) ** 2).sum()), float(len(yt))])
+    return p[0] / p[1] if p[1] else 0.0
+
+
+def mean_absolute_error(y_true, y_pred) -> float:
+    yt = np.asarray(y_true, dtype=np.float64)
+    yp = np.asarray(y_pred, dtype=np.float64)
+    p = _metric_partials([float(np.abs(yt - yp).sum()), float(len(yt))])
+    return p[0] / p[1] if p[1] else 0.0
+
+
+def r2_score(y_true, y_pred) -> float:
+    yt = np.asarray(y_true, dtype=np.float64)
+    yp = np.asarray(y_pred, dtype=np.float64)
+    p = _metric_partials([
+        float(((yt - yp) ** 2).sum()), float(yt.sum()),
+        float((yt * yt).sum()), float(len(yt))])
+    sse, s, ss, n = p
+    if n == 0:
+        return 0.0
+    sst = ss - s * s / n
+    return 1.0 - sse / sst if sst else 0.0

@@ -113,3 +113,33 @@ def test_kmeans_quality():
     assert km.inertia_ <= sk.inertia_ * 1.2
     pred = km.predict(X.astype(np.float32))
     assert len(np.unique(pred)) == 4
+
+
+def test_preprocessing_and_metrics_vs_sklearn():
+    """MinMaxScaler/LabelEncoder/train_test_split/metrics match sklearn
+    single-rank (distributed partials all-reduce the same way)."""
+    import sklearn.metrics as skmet
+    import sklearn.preprocessing as skp
+
+    from bodo_amd import ml
+
+    rng = np.random.default_rng(0)
+    X = rng.random((100, 3)) * 10 - 5
+    y = rng.integers(0, 3, 100)
+    mm = ml.MinMaxScaler().fit(X)
+    np.testing.assert_allclose(mm.transform(X),
+                               skp.MinMaxScaler().fit_transform(X))
+    le = ml.LabelEncoder().fit(["b", "a", "c", "a"])
+    assert le.transform(["a", "c"]).tolist() == [0, 2]
+    assert le.inverse_transform([1]).tolist() == ["b"]
+    Xtr, Xte, ytr, yte = ml.train_test_split(X, y, test_size=0.25,
+                                             random_state=0)
+    assert len(Xtr) == 75 and len(Xte) == 25 and len(ytr) == 75
+    yp = (y + (rng.random(100) < 0.2)).clip(0, 2)
+    assert abs(ml.accuracy_score(y, yp)
+               - skmet.accuracy_score(y, yp)) < 1e-12
+    assert abs(ml.mean_squared_error(y, yp)
+               - skmet.mean_squared_error(y, yp)) < 1e-12
+    assert abs(ml.r2_score(y, yp) - skmet.r2_score(y, yp)) < 1e-12
+    assert abs(ml.mean_absolute_error(y, yp)
+               - skmet.mean_absolute_error(y, yp)) < 1e-12
