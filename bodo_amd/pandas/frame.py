@@ -555,8 +555,10 @@ class BodoDataFrame:
     def iloc(self):
         return _ILoc(self)
 
-    def rolling(self, window, min_periods=None, **kwargs):
-        return _RollingFrame(self, int(window), min_periods)
+    def rolling(self, window, min_periods=None, center=False, **kwargs):
+        if isinstance(window, str):
+            return _RollingHost(self, window, min_periods)
+        return _RollingFrame(self, int(window), min_periods, center)
 
     def filter(self, items=None, like=None, regex=None, axis=None):
         """Column-label filtering (reference: frame.py filter)."""
@@ -866,14 +868,17 @@ def from_pandas_df(df: pd.DataFrame) -> BodoDataFrame:
 
 class _RollingFrame:
     """df.rolling(w): per-column rolling aggregation (Rolling plan node with
-    distributed halo exchange; reference: hiframes rolling)."""
+    distributed halo exchange; reference: hiframes rolling).  center=True
+    relabels the trailing window onto its middle row (a -(w//2) shift)."""
 
     _FUNCS = ("sum", "mean", "min", "max", "count", "std", "var", "median")
 
-    def __init__(self, frame: BodoDataFrame, window: int, min_periods):
+    def __init__(self, frame: BodoDataFrame, window: int, min_periods,
+                 center: bool = False):
         self._frame = frame
         self._window = window
         self._min_periods = min_periods
+        self._center = center
 
     def _agg(self, func):
         from ..engine import api
@@ -884,7 +889,33 @@ class _RollingFrame:
         specs = tuple((c, c, func) for c in cols)
         plan = pn.Rolling(self._frame._plan, self._window, self._min_periods,
                           specs)
+        if self._center and self._window > 1:
+            plan = pn.Shift(plan, -(self._window // 2),
+                            tuple((c, c) for c in cols))
         return BodoDataFrame(plan, list(cols))
+
+    def __getattr__(self, name):
+        if name in self._FUNCS:
+            return lambda: self._agg(name)
+        raise AttributeError(name)
+
+
+class _RollingHost:
+    """Offset-string windows ('1D', '2h'): host pandas on the gathered
+    frame (time-based windows need the timestamp index; replicated result,
+    exact)."""
+
+    _FUNCS = ("sum", "mean", "min", "max", "count", "std", "var", "median")
+
+    def __init__(self, frame: BodoDataFrame, window: str, min_periods):
+        self._frame = frame
+        self._window = window
+        self._min_periods = min_periods
+
+    def _agg(self, func):
+        pdf = self._frame.to_pandas()
+        r = pdf.rolling(self._window, min_periods=self._min_periods or 1)
+        return getattr(r, func)()
 
     def __getattr__(self, name):
         if name in self._FUNCS:

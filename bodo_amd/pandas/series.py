@@ -525,8 +525,13 @@ class BodoSeries:
         b = BodoSeries(plan, ColRef("__sh"), None)
         return a.corr(b)
 
-    def rolling(self, window, min_periods=None, **kwargs):
-        return _RollingSeries(self, int(window), min_periods)
+    def rolling(self, window, min_periods=None, center=False, **kwargs):
+        if isinstance(window, str):
+            return _RollingSeriesHost(self, window, min_periods)
+        return _RollingSeries(self, int(window), min_periods, center)
+
+    def ewm(self, com=None, span=None, alpha=None, adjust=True):
+        return _EwmSeries(self, com, span, alpha, adjust)
 
     def _cum(self, func) -> "BodoSeries":
         plan = pn.Cumulative(self._as_projection_plan(), (("v", "v", func),))
@@ -867,6 +872,46 @@ class _ListAccessor:
     __getitem__ = get
 
 
+class _RollingSeriesHost:
+    """Offset-string rolling windows on a series: host pandas fallback."""
+
+    _FUNCS = ("sum", "mean", "min", "max", "count", "std", "var", "median")
+
+    def __init__(self, s, window, min_periods):
+        self._s = s
+        self._window = window
+        self._min_periods = min_periods
+
+    def __getattr__(self, name):
+        if name in self._FUNCS:
+            return lambda: getattr(self._s.to_pandas().rolling(
+                self._window, min_periods=self._min_periods or 1), name)()
+        raise AttributeError(name)
+
+
+class _EwmSeries:
+    """s.ewm(...).mean()/var()/std(): host pandas on the gathered series
+    (the recurrence needs cross-shard carry; replicated exact result)."""
+
+    def __init__(self, s, com, span, alpha, adjust):
+        self._s = s
+        self._kw = dict(com=com, span=span, alpha=alpha, adjust=adjust)
+
+    def _run(self, func):
+        kw = {k: v for k, v in self._kw.items() if v is not None or
+              k == "adjust"}
+        return getattr(self._s.to_pandas().ewm(**kw), func)()
+
+    def mean(self):
+        return self._run("mean")
+
+    def var(self):
+        return self._run("var")
+
+    def std(self):
+        return self._run("std")
+
+
 class _StructAccessor:
     """Series.struct over STRUCT columns (pandas ArrowDtype .struct
     surface; reference role: struct_arr_ext field access)."""
@@ -913,14 +958,18 @@ class _RollingSeries:
 
     _FUNCS = ("sum", "mean", "min", "max", "count", "std", "var", "median")
 
-    def __init__(self, s: BodoSeries, window: int, min_periods):
+    def __init__(self, s: BodoSeries, window: int, min_periods,
+                 center: bool = False):
         self._s = s
         self._window = window
         self._min_periods = min_periods
+        self._center = center
 
     def _agg(self, func):
         plan = pn.Rolling(self._s._as_projection_plan(), self._window,
                           self._min_periods, (("v", "v", func),))
+        if self._center and self._window > 1:
+            plan = pn.Shift(plan, -(self._window // 2), (("v", "v"),))
         return BodoSeries(plan, ColRef("v"), self._s.name)
 
     def __getattr__(self, name):

@@ -713,3 +713,25 @@ def test_merge_asof_and_to_datetime_options():
     s2 = bpd.from_pandas(pd.DataFrame({"s": ["2024-01-01", "bad"]}))["s"]
     out2 = bpd.to_datetime(s2, errors="coerce").to_pandas()
     assert out2.iloc[0] == pd.Timestamp("2024-01-01") and pd.isna(out2.iloc[1])
+
+
+def test_rolling_center_timebased_ewm():
+    rng = np.random.default_rng(3)
+    df = pd.DataFrame({"y": np.where(rng.random(100) < 0.1, np.nan,
+                                     rng.random(100)),
+                       "t": pd.date_range("2024-01-01", periods=100,
+                                          freq="6h")})
+    b = bpd.from_pandas(df)
+    got = b.rolling(7, center=True).mean().to_pandas()["y"]
+    want = df[["y"]].rolling(7, center=True).mean()["y"]
+    np.testing.assert_allclose(got.fillna(-9), want.fillna(-9))
+    gs = b["y"].rolling(7, center=True).mean().to_pandas()
+    np.testing.assert_allclose(gs.fillna(-9), want.fillna(-9))
+    g2 = b.set_index("t").rolling("1D").mean()
+    g2 = g2.to_pandas() if hasattr(g2, "to_pandas") else g2
+    w2 = df.set_index("t")[["y"]].rolling("1D").mean()
+    np.testing.assert_allclose(np.asarray(g2["y"]), w2["y"].to_numpy(),
+                               equal_nan=True)
+    ge = b["y"].ewm(alpha=0.3).mean()
+    we = df["y"].ewm(alpha=0.3).mean()
+    np.testing.assert_allclose(ge.fillna(-9), we.fillna(-9))
