@@ -145,8 +145,9 @@ class DistArray:
             res = getattr(np, ufunc.__name__)(*args_np)
             return DistArray(torch.from_numpy(np.ascontiguousarray(res))
                              .to(self.t.device), self.total)
-        out = f(*[t if torch.is_tensor(t) or isinstance(t, (int, float, bool))
-                  else t for t in ts])
+        dev = self.t.device
+        out = f(*[t if torch.is_tensor(t)
+                  else torch.as_tensor(t, device=dev) for t in ts])
         if not torch.is_tensor(out):
             out = torch.as_tensor(out)
         return DistArray(out, self.total)
@@ -274,16 +275,48 @@ class DistArray:
     def mean(self):
         return self._reduce("sum") / max(self.total, 1)
 
-    def std(self):
+    def var(self, ddof: int = 0):
         m = self.mean()
-        var = ((self - m) ** 2)._reduce("sum") / max(self.total - 1, 1)
-        return var ** 0.5
+        return ((self - m) ** 2)._reduce("sum") / max(self.total - ddof, 1)
+
+    def std(self, ddof: int = 0):  # numpy default ddof=0 at this layer
+        return self.var(ddof) ** 0.5
+
+    def cumsum(self):
+        """Distributed scan: local cumsum + exclusive prefix of shard sums
+        (reference: parfor cumsum -> dist_exscan lowering)."""
+        local = torch.cumsum(self.t.double() if self.t.dtype == torch.bool
+                             else self.t, 0)
+        if comm.get_world_size() > 1:
+            tail = local[-1].item() if local.numel() else 0
+            parts = comm.allgather_obj(tail)
+            base = sum(parts[:comm.get_rank()])
+            if base:
+                local = local + base
+        return DistArray(local, self.total)
+
+    def _gathered(self) -> torch.Tensor:
+        """Full array on every rank (order-preserving; REP result)."""
+        if comm.get_world_size() == 1:
+            return self.t
+        parts = comm.allgather_obj(self.t.cpu().numpy())
+        return torch.from_numpy(np.concatenate(parts)).to(self.t.device)
+
+    def _my_block_of(self, full: torch.Tensor) -> "DistArray":
+        w, r = comm.get_world_size(), comm.get_rank()
+        n = int(full.numel())
+        s, e = _block_bounds(n, w, r)
+        return DistArray(full[s:e], n)
 
     def astype(self, dtype):
         return DistArray(self.t.to(_np_to_torch(dtype)), self.total)
 
     def copy(self):
         return DistArray(self.t.clone(), self.total)
+
+    def __array__(self, dtype=None):
+        a = self.to_numpy()
+        return a.astype(dtype) if dtype is not None else a
 
     def __getitem__(self, i):
         if isinstance(i, DistArray) and i.t.dtype == torch.bool:
@@ -295,7 +328,13 @@ class DistArray:
         if isinstance(i, (int, np.integer)):
             # global scalar index: owner broadcasts
             w, r = comm.get_world_size(), comm.get_rank()
-            i = int(i) % self.total
+            i = int(i)
+            if i < 0:
+                i += self.total
+            if not 0 <= i < self.total:
+                # MUST raise: numpy's iteration fallback probes increasing
+                # indexes until IndexError (a modulo wrap loops forever)
+                raise IndexError(i)
             s, e = _block_bounds(self.total, w, r)
             val = self.t[i - s].item() if s <= i < e else None
             if w > 1:
@@ -318,13 +357,72 @@ def _dot(a, b):
     return NotImplemented
 
 
+def _argext(a: "DistArray", which: str):
+    """Global argmax/argmin: local winner + (value, global index) compare."""
+    w, r = comm.get_world_size(), comm.get_rank()
+    s, _ = _block_bounds(a.total, w, r)
+    if a.t.numel():
+        li = int(torch.argmax(a.t).item() if which == "max"
+                 else torch.argmin(a.t).item())
+        cand = (float(a.t[li].item()), s + li)
+    else:
+        cand = (float("-inf") if which == "max" else float("inf"), -1)
+    if w > 1:
+        cands = comm.allgather_obj(cand)
+        key = (lambda c: (c[0], -c[1])) if which == "max" else \
+            (lambda c: (-c[0], -c[1]))
+        best = max(cands, key=key)
+        return best[1]
+    return cand[1]
+
+
+def _diff(a: "DistArray", n: int = 1):
+    assert n == 1, "np.diff with n>1 unsupported"
+    w, r = comm.get_world_size(), comm.get_rank()
+    local = torch.diff(a.t) if a.t.numel() > 1 else a.t[:0]
+    if w > 1:
+        firsts = comm.allgather_obj(
+            float(a.t[0].item()) if a.t.numel() else None)
+        lasts = comm.allgather_obj(
+            float(a.t[-1].item()) if a.t.numel() else None)
+        if r > 0 and a.t.numel():
+            prev = next((lasts[i] for i in range(r - 1, -1, -1)
+                         if lasts[i] is not None), None)
+            if prev is not None:
+                b = torch.tensor([float(a.t[0].item()) - prev],
+                                 dtype=local.dtype if local.numel()
+                                 else torch.float64, device=a.t.device)
+                local = torch.cat([b, local]) if local.numel() else b
+        total = a.total - 1
+    else:
+        total = max(a.total - 1, 0)
+    return DistArray(local, total)
+
+
 _ARRAY_FUNCS = {
     "sum": lambda a, **k: a.sum(),
     "prod": lambda a, **k: a.prod(),
     "min": lambda a, **k: a.min(),
     "max": lambda a, **k: a.max(),
     "mean": lambda a, **k: a.mean(),
-    "std": lambda a, **k: a.std(),
+    "std": lambda a, **k: a.std(**k),
+    "var": lambda a, **k: a.var(**k),
+    "cumsum": lambda a, **k: a.cumsum(),
+    "sort": lambda a, **k: a._my_block_of(
+        torch.sort(a._gathered())[0]),
+    "unique": lambda a, **k: np.unique(a._gathered().cpu().numpy()),
+    "argmax": lambda a, **k: _argext(a, "max"),
+    "argmin": lambda a, **k: _argext(a, "min"),
+    "clip": lambda a, lo=None, hi=None, **k: DistArray(
+        torch.clamp(a.t, min=lo, max=hi), a.total),
+    "percentile": lambda a, q, **k: np.percentile(
+        a._gathered().cpu().numpy(), q),
+    "quantile": lambda a, q, **k: np.quantile(
+        a._gathered().cpu().numpy(), q),
+    "median": lambda a, **k: float(np.median(a._gathered().cpu().numpy())),
+    "histogram": lambda a, bins=10, range=None, **k: np.histogram(
+        a._gathered().cpu().numpy(), bins=bins, range=range),
+    "diff": _diff,
     "dot": _dot,
     "where": lambda c, x, y: DistArray(
         torch.where(c.t,
