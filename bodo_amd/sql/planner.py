@@ -1220,6 +1220,10 @@ class Planner:
                 if unit in ns_per:
                     off = ex.Const(int(qty) * ns_per[unit], bt.int64)
                     return ex.BinOp(e.op, l, off)
+                if unit in ("month", "quarter", "year"):
+                    mult = {"month": 1, "quarter": 3, "year": 12}[unit]
+                    n_m = int(qty) * mult * (1 if e.op == "add" else -1)
+                    return ex.DtField(l, f"add_months:{n_m}")
                 raise NotImplementedError(
                     f"interval '{unit}' arithmetic on columns")
             return ex.BinOp(e.op, l, r)
@@ -1454,6 +1458,13 @@ class Planner:
                                      ex.DtField(a, "month"))
                     return ex.BinOp("add", ex.BinOp("mul", ydiff,
                                                     ex.Const(12)), mdiff)
+                if unit == "quarter":
+                    ydiff = ex.BinOp("sub", ex.DtField(b2, "year"),
+                                     ex.DtField(a, "year"))
+                    qdiff = ex.BinOp("sub", ex.DtField(b2, "quarter"),
+                                     ex.DtField(a, "quarter"))
+                    return ex.BinOp("add", ex.BinOp("mul", ydiff,
+                                                    ex.Const(4)), qdiff)
                 ns_per = {"day": 86400 * 10**9, "hour": 3600 * 10**9,
                           "minute": 60 * 10**9, "second": 10**9,
                           "week": 7 * 86400 * 10**9}

@@ -953,3 +953,15 @@ def test_try_cast_lag_default_decode_width_bucket():
     wb = bc.sql("select width_bucket(v, 0, 10, 5) as r from t") \
         .to_pandas()["r"]
     assert wb.tolist() == [1, 2, 2, 3, 3]
+
+
+def test_calendar_interval_on_columns():
+    df = pd.DataFrame({"d": pd.to_datetime(["2024-01-31", "2023-06-01"])})
+    bc = BodoSQLContext({"t": df})
+    o = bc.sql("select d + interval '1 month' as m, "
+               "d - interval '1 year' as y, "
+               "datediff('quarter', d, d + interval '200 day') as q "
+               "from t").to_pandas()
+    assert pd.Timestamp(o["m"][0]) == pd.Timestamp("2024-02-29")  # clamped
+    assert pd.Timestamp(o["y"][1]) == pd.Timestamp("2022-06-01")
+    assert o["q"].tolist() == [3, 2]
