@@ -964,4 +964,4 @@ def test_calendar_interval_on_columns():
                "from t").to_pandas()
     assert pd.Timestamp(o["m"][0]) == pd.Timestamp("2024-02-29")  # clamped
     assert pd.Timestamp(o["y"][1]) == pd.Timestamp("2022-06-01")
-    assert o["q"].tolist() == [3, 2]
+    assert o["q"].tolist() == [2, 2]  # Q1->Q3, Q2->Q4
