@@ -125,6 +125,30 @@ class Column:
                 else arr.dictionary,
                 length=len(arr),
             )
+        if pa.types.is_binary(t) or pa.types.is_large_binary(t):
+            arr = arr.cast(pa.large_binary())
+            bufs = arr.buffers()
+            validity, off_buf, data_buf = bufs[0], bufs[1], bufs[2]
+            offsets = np.frombuffer(off_buf, dtype=np.int64,
+                                    count=len(arr) + 1 + arr.offset)[arr.offset:]
+            if arr.offset != 0 or offsets[0] != 0:
+                offsets = offsets - offsets[0]
+            n_bytes = int(offsets[-1])
+            base = np.frombuffer(off_buf, dtype=np.int64,
+                                 count=1 + arr.offset + len(arr))[arr.offset]
+            data = np.frombuffer(data_buf, dtype=np.uint8, count=n_bytes,
+                                 offset=int(base)) \
+                if data_buf is not None and n_bytes else np.zeros(0, np.uint8)
+            mask = None
+            if validity is not None and arr.null_count:
+                mask = _unpack_validity(validity, arr.offset, len(arr))
+            return Column(
+                bt.binary,
+                torch.from_numpy(np.ascontiguousarray(data)).to(device),
+                None if mask is None else torch.from_numpy(mask).to(device),
+                offsets=torch.from_numpy(
+                    np.ascontiguousarray(offsets)).to(device),
+                length=len(arr))
         if pa.types.is_string(t) or pa.types.is_large_string(t):
             arr = arr.cast(pa.large_string())
             bufs = arr.buffers()
@@ -288,6 +312,15 @@ class Column:
             offsets = _np_from_tensor(self.offsets)
             data = _np_from_tensor(self.data) if self.data is not None else np.zeros(0, np.uint8)
             mask = None if self.mask is None else ~_np_from_tensor(self.mask)
+            if self.dtype.precision == 1:  # binary flag
+                vbuf = pa.py_buffer(np.packbits(
+                    ~mask, bitorder="little").tobytes()) \
+                    if mask is not None else None
+                return pa.Array.from_buffers(
+                    pa.large_binary(), len(self),
+                    [vbuf, pa.py_buffer(offsets.tobytes()),
+                     pa.py_buffer(data.tobytes())],
+                    -1 if mask is not None else 0)
             return pa.LargeStringArray.from_buffers(
                 len(self), pa.py_buffer(offsets.tobytes()), pa.py_buffer(data.tobytes()),
                 pa.py_buffer(np.packbits(~mask, bitorder="little").tobytes()) if mask is not None else None,

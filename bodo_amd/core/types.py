@@ -99,6 +99,9 @@ boolean = DType(TypeKind.BOOL)
 date32 = DType(TypeKind.DATE32)
 timestamp_ns = DType(TypeKind.TIMESTAMP_NS)
 string = DType(TypeKind.STRING)
+# binary shares the STRING layout (offsets+bytes); precision=1 flags it so
+# arrow round-trips as large_binary (reference: binary_arr_ext.py)
+binary = DType(TypeKind.STRING, precision=1)
 dictionary = DType(TypeKind.DICT)
 list_ = DType(TypeKind.LIST)
 

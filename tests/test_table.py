@@ -107,3 +107,17 @@ def test_map_column_via_list_struct():
     assert c.to_arrow().to_pylist() == [
         [{"key": "a", "value": 1}, {"key": "b", "value": 2}], None,
         [{"key": "c", "value": 3}]]
+
+
+def test_binary_columns():
+    """BINARY columns share the STRING offsets+bytes layout (precision=1
+    flags arrow large_binary round-trip; reference: binary_arr_ext.py)."""
+    import bodo_amd.pandas as bpd
+
+    df = pd.DataFrame({"b": [b"ab", b"c", None, b"\x00\xffbin"],
+                       "k": [1, 2, 3, 4]})
+    b = bpd.from_pandas(df)
+    assert b.to_pandas()["b"].tolist() == df["b"].tolist()
+    assert b[b["k"] > 1].to_pandas()["b"].tolist() == df["b"].tolist()[1:]
+    out = b.sort_values("k", ascending=False).to_pandas()["b"].tolist()
+    assert out == df["b"].tolist()[::-1]
