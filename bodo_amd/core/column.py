@@ -226,7 +226,10 @@ class Column:
             # typeless all-null column: represent as float64 NaN
             data = torch.full((len(arr),), float("nan"), dtype=torch.float64)
             return Column(bt.float64, data.to(device), length=len(arr))
-        if pa.types.is_timestamp(t):
+        if pa.types.is_duration(t):
+            arr = arr.cast(pa.duration("ns"))
+            dtype = bt.duration_ns
+        elif pa.types.is_timestamp(t):
             arr = arr.cast(pa.timestamp("ns"))
             dtype = bt.timestamp_ns
         elif pa.types.is_date32(t):
@@ -353,6 +356,7 @@ class Column:
         pa_type = {
             TypeKind.DATE32: pa.date32(),
             TypeKind.TIMESTAMP_NS: pa.timestamp("ns"),
+            TypeKind.DURATION_NS: pa.duration("ns"),
             TypeKind.UINT8: pa.uint8(),
             TypeKind.UINT16: pa.uint16(),
             TypeKind.UINT32: pa.uint32(),
@@ -371,7 +375,7 @@ class Column:
             dtype = bt.timestamp_ns
         elif arr.dtype.kind == "m":
             arr = arr.astype("timedelta64[ns]").view("int64")
-            dtype = bt.int64
+            dtype = bt.duration_ns
         elif arr.dtype == object or arr.dtype.kind == "U":
             # from_pandas treats float NaN placeholders in object arrays as
             # nulls (pa.array would infer double from a leading NaN)

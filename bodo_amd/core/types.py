@@ -37,6 +37,7 @@ class TypeKind(enum.IntEnum):
     UINT64 = 15
     LIST = 16  # arrow list<child>: int64 offsets (n+1) + child column
     STRUCT = 17  # arrow struct<fields>: named child columns + validity
+    DURATION_NS = 18  # timedelta64[ns], int64 storage
 
 
 @dataclass(frozen=True)
@@ -98,6 +99,7 @@ float64 = DType(TypeKind.FLOAT64)
 boolean = DType(TypeKind.BOOL)
 date32 = DType(TypeKind.DATE32)
 timestamp_ns = DType(TypeKind.TIMESTAMP_NS)
+duration_ns = DType(TypeKind.DURATION_NS)
 string = DType(TypeKind.STRING)
 # binary shares the STRING layout (offsets+bytes); precision=1 flags it so
 # arrow round-trips as large_binary (reference: binary_arr_ext.py)
@@ -129,6 +131,7 @@ _TORCH_STORAGE = {
     TypeKind.BOOL: torch.bool,
     TypeKind.DATE32: torch.int32,
     TypeKind.TIMESTAMP_NS: torch.int64,
+    TypeKind.DURATION_NS: torch.int64,
     TypeKind.DICT: torch.int32,
     TypeKind.DECIMAL128: torch.int64,  # scaled int64 (exact for p <= 18)
 }
@@ -177,6 +180,7 @@ _KIND_TO_NUMPY = {
     TypeKind.BOOL: np.dtype("bool"),
     TypeKind.DATE32: np.dtype("int32"),
     TypeKind.TIMESTAMP_NS: np.dtype("int64"),
+    TypeKind.DURATION_NS: np.dtype("int64"),
     TypeKind.DICT: np.dtype("int32"),
     TypeKind.DECIMAL128: np.dtype("int64"),
 }

@@ -529,6 +529,8 @@ def _reduce_wrap_val(col: Column, v):
         return pd.Timestamp(int(v))
     if col.dtype.kind == TypeKind.DATE32:
         return pd.Timestamp(int(v) * 86_400_000_000_000)
+    if col.dtype.kind == TypeKind.DURATION_NS:
+        return pd.Timedelta(int(v))
     return v
 
 

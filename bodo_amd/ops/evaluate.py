@@ -542,6 +542,20 @@ def binary_arith(op: str, a: Column, b: Column) -> Column:
         raise NotImplementedError(op)
     mask = combine_masks(a.mask, b.mask)
     dtype = bt.from_numpy_dtype(np.dtype(str(out.dtype).replace("torch.", "")))
+    # temporal typing: ts - ts -> duration; ts +/- duration -> ts;
+    # duration +/- duration -> duration
+    TK = TypeKind
+    ka, kb = a.dtype.kind, b.dtype.kind
+    if op == "sub" and ka == TK.TIMESTAMP_NS and kb == TK.TIMESTAMP_NS:
+        dtype = bt.duration_ns
+    elif op in ("add", "sub") and ka == TK.TIMESTAMP_NS \
+            and kb == TK.DURATION_NS:
+        dtype = bt.timestamp_ns
+    elif op == "add" and kb == TK.TIMESTAMP_NS and ka == TK.DURATION_NS:
+        dtype = bt.timestamp_ns
+    elif op in ("add", "sub") and TK.DURATION_NS in (ka, kb) \
+            and dtype.kind == TK.INT64:
+        dtype = bt.duration_ns
     return Column(dtype, out, mask)
 
 

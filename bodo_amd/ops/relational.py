@@ -190,6 +190,7 @@ def _init_pa_types():
         TypeKind.FLOAT32: pa.float32(), TypeKind.FLOAT64: pa.float64(),
         TypeKind.BOOL: pa.bool_(), TypeKind.DATE32: pa.date32(),
         TypeKind.TIMESTAMP_NS: pa.timestamp("ns"),
+        TypeKind.DURATION_NS: pa.duration("ns"),
         TypeKind.STRING: pa.large_string(),
     }
 

@@ -121,3 +121,21 @@ def test_binary_columns():
     assert b[b["k"] > 1].to_pandas()["b"].tolist() == df["b"].tolist()[1:]
     out = b.sort_values("k", ascending=False).to_pandas()["b"].tolist()
     assert out == df["b"].tolist()[::-1]
+
+
+def test_timedelta_columns_and_arith():
+    """DURATION_NS: timedelta64 round trip; ts - ts yields timedelta
+    (value semantics, not raw ns) (reference: pd_timedelta_ext)."""
+    import bodo_amd.pandas as bpd
+
+    df = pd.DataFrame({"d1": pd.to_datetime(["2024-01-05", "2024-02-01"]),
+                       "d2": pd.to_datetime(["2024-01-01", "2024-01-15"])})
+    b = bpd.from_pandas(df)
+    delta = (b["d1"] - b["d2"]).to_pandas()
+    assert str(delta.dtype) == "timedelta64[ns]"
+    assert delta.tolist() == (df["d1"] - df["d2"]).tolist()
+    df2 = pd.DataFrame({"td": pd.to_timedelta(["1 days", "3 hours", None]),
+                        "k": [1, 2, 3]})
+    b2 = bpd.from_pandas(df2)
+    assert b2.to_pandas()["td"].tolist()[:2] == df2["td"].tolist()[:2]
+    assert pd.Timedelta(b2["td"].max().value) == pd.Timedelta("1 days")
