@@ -170,3 +170,58 @@ def test_fuzz_round2_features(seed):
         assert len(got) == len(ref)
         si = b.set_index(["k"]).to_pandas()
         assert si.index.name == "k"
+
+
+@pytest.mark.parametrize("seed", range(20))
+def test_fuzz_round2b_features(seed):
+    """Differential fuzz over the late-round-2 additions: null-key outer
+    joins, running window frames, CTEs, global value reduces."""
+    rng = np.random.default_rng(seed + 31000)
+    n = int(rng.integers(3, 200))
+    df = pd.DataFrame({
+        "g": rng.choice(["a", "b", "c"], n),
+        "k": np.where(rng.random(n) < 0.2, np.nan,
+                      rng.integers(0, 8, n).astype(float)),
+        "y": np.where(rng.random(n) < 0.15, np.nan, rng.random(n) * 10),
+        "o": rng.permutation(n),
+    })
+    op = seed % 4
+    if op == 0:
+        r = pd.DataFrame({"k": np.arange(6, dtype=float),
+                          "w": rng.random(6)})
+        got = (bpd.from_pandas(df).merge(bpd.from_pandas(r), on="k",
+                                         how="outer").to_pandas())
+        exp = df.merge(r, on="k", how="outer")
+        assert len(got) == len(exp)
+        assert got["k"].isna().sum() == exp["k"].isna().sum()
+        assert abs(np.nansum(got["w"]) - np.nansum(exp["w"])) < 1e-6
+    elif op == 1:
+        bc = BodoSQLContext({"t": df})
+        got = bc.sql("select o, min(y) over (partition by g order by o) as m"
+                     " from t").to_pandas().sort_values("o")
+        ref = df.sort_values("o").groupby("g")["y"].transform(
+            lambda s: s.expanding(1).min())
+        ref = ref.reindex(df.sort_values("o").index)
+        np.testing.assert_allclose(
+            got["m"].to_numpy(dtype=float),
+            ref.to_numpy(dtype=float), equal_nan=True)
+    elif op == 2:
+        bc = BodoSQLContext({"t": df})
+        got = bc.sql(
+            "with s as (select g, avg(y) as ay from t group by g) "
+            "select t.g, t.y, s.ay from t join s on t.g = s.g"
+        ).to_pandas()
+        assert len(got) == n
+        m = df.groupby("g")["y"].mean()
+        samp = got.head(20)
+        for _, row in samp.iterrows():
+            assert abs(row["ay"] - m[str(row["g"])]) < 1e-9
+    else:
+        b = bpd.from_pandas(df)
+        assert abs(float(b["y"].median()) - df["y"].median()) < 1e-9 \
+            or (np.isnan(float(b["y"].median()))
+                and np.isnan(df["y"].median()))
+        got_sk = float(b["y"].skew())
+        want_sk = df["y"].skew()
+        assert (np.isnan(got_sk) and np.isnan(want_sk)) \
+            or abs(got_sk - want_sk) < 1e-9
