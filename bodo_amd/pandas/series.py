@@ -549,14 +549,19 @@ class BodoSeries:
     def cummax(self):
         return self._cum("cummax")
 
-    def rank(self, method="average", ascending=True) -> "BodoSeries":
+    def rank(self, method="average", ascending=True,
+             pct=False) -> "BodoSeries":
         """Global rank via the window machinery with a constant partition
-        key (reference: array_kernels rank)."""
+        key (reference: array_kernels rank); pct divides by the non-null
+        count like pandas."""
         proj = pn.Projection(self._plan, ("v", "__k"),
                              (self._expr, Const(1)))
         plan = pn.Window(proj, ("__k",), ("v",), (bool(ascending),),
                          (("__r", "v", "rank", method),))
-        return BodoSeries(plan, ColRef("__r"), self.name)
+        r = BodoSeries(plan, ColRef("__r"), self.name)
+        if pct:
+            return r / float(self.count())
+        return r
 
     def shift(self, periods=1) -> "BodoSeries":
         plan = pn.Shift(self._as_projection_plan(), int(periods),

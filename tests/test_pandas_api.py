@@ -735,3 +735,13 @@ def test_rolling_center_timebased_ewm():
     ge = b["y"].ewm(alpha=0.3).mean()
     we = df["y"].ewm(alpha=0.3).mean()
     np.testing.assert_allclose(ge.fillna(-9), we.fillna(-9))
+
+
+def test_series_rank_pct():
+    rng = np.random.default_rng(4)
+    df = pd.DataFrame({"x": rng.integers(0, 30, 100).astype(float)})
+    df.loc[3, "x"] = np.nan
+    b = bpd.from_pandas(df)
+    got = b["x"].rank(method="min", pct=True).to_pandas()
+    want = df["x"].rank(method="min", pct=True)
+    np.testing.assert_allclose(got.fillna(-9), want.fillna(-9))
