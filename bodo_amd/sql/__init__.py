@@ -60,6 +60,36 @@ class BodoSQLContext:
             self.tables[name.lower()] = self.sql(select)
             BodoSQLContext._plan_cache.clear()
             return None
+        if low.startswith("insert into"):
+            # INSERT INTO name SELECT ... | INSERT INTO name VALUES (...)
+            import re as _re
+
+            m = _re.match(r"insert\s+into\s+(\w+)\s+(.*)$", stripped,
+                          _re.IGNORECASE | _re.DOTALL)
+            if not m:
+                raise ValueError(f"unsupported INSERT: {stripped[:60]}")
+            name, rest = m.group(1).lower(), m.group(2).strip()
+            if name not in self.tables:
+                raise KeyError(f"unknown table {name}")
+            base = self.tables[name]
+            if rest.lower().startswith("values"):
+                import ast as _pyast
+
+                import pandas as _pd
+
+                rows = _pyast.literal_eval(
+                    "[" + rest[len("values"):].strip() + "]")
+                new = _pd.DataFrame(rows, columns=list(base.columns))
+                from ..pandas import from_pandas as _fp
+
+                add = _fp(new)
+            else:
+                add = self.sql(rest)
+            from ..pandas import concat as _concat
+
+            self.tables[name] = _concat([base, add])
+            BodoSQLContext._plan_cache.clear()
+            return None
         key = (query, tuple(sorted((n, id(t._lazy_plan))
                                    for n, t in self.tables.items())))
         hit = BodoSQLContext._plan_cache.get(key)

@@ -965,3 +965,16 @@ def test_calendar_interval_on_columns():
     assert pd.Timestamp(o["m"][0]) == pd.Timestamp("2024-02-29")  # clamped
     assert pd.Timestamp(o["y"][1]) == pd.Timestamp("2022-06-01")
     assert o["q"].tolist() == [2, 2]  # Q1->Q3, Q2->Q4
+
+
+def test_insert_into():
+    """INSERT INTO ... VALUES and INSERT INTO ... SELECT (reference:
+    BodoSQL DML on registered tables)."""
+    bc = BodoSQLContext({"t": pd.DataFrame({"a": [1, 2], "s": ["x", "y"]})})
+    bc.sql("insert into t values (3, 'z'), (4, 'w')")
+    got = bc.sql("select * from t order by a").to_pandas()
+    assert got["a"].tolist() == [1, 2, 3, 4]
+    bc.sql("create table u as select a * 10 as a, s from t where a >= 3")
+    bc.sql("insert into t select a, s from u")
+    out = bc.sql("select count(*) as n, sum(a) as s from t").to_pandas()
+    assert out["n"][0] == 6 and out["s"][0] == 80
