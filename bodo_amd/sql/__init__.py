@@ -90,6 +90,56 @@ class BodoSQLContext:
             self.tables[name] = _concat([base, add])
             BodoSQLContext._plan_cache.clear()
             return None
+        if low.startswith("delete from"):
+            # DELETE FROM name [WHERE cond] — keep the complement
+            import re as _re
+
+            m = _re.match(r"delete\s+from\s+(\w+)(?:\s+where\s+(.*))?$",
+                          stripped, _re.IGNORECASE | _re.DOTALL)
+            if not m:
+                raise ValueError(f"unsupported DELETE: {stripped[:60]}")
+            name, cond = m.group(1).lower(), m.group(2)
+            if name not in self.tables:
+                raise KeyError(f"unknown table {name}")
+            if cond is None:
+                self.tables[name] = self.tables[name].head(0)
+            else:
+                self.tables[name] = self.sql(
+                    f"select * from {name} where not ({cond})")
+            BodoSQLContext._plan_cache.clear()
+            return None
+        if low.startswith("update "):
+            # UPDATE name SET c1 = e1 [, ...] [WHERE cond]
+            import re as _re
+
+            m = _re.match(r"update\s+(\w+)\s+set\s+(.*?)"
+                          r"(?:\s+where\s+(.*))?$",
+                          stripped, _re.IGNORECASE | _re.DOTALL)
+            if not m:
+                raise ValueError(f"unsupported UPDATE: {stripped[:60]}")
+            name, sets, cond = m.group(1).lower(), m.group(2), m.group(3)
+            if name not in self.tables:
+                raise KeyError(f"unknown table {name}")
+            assigns = {}
+            for part in _split_top_level_commas(sets):
+                col, _, expr = part.partition("=")
+                assigns[col.strip().lower()] = expr.strip()
+            cols = []
+            for c in self.tables[name].columns:
+                cl = str(c).lower()
+                if cl in assigns:
+                    e = assigns[cl]
+                    if cond is not None:
+                        cols.append(f"case when {cond} then ({e}) "
+                                    f"else {c} end as {c}")
+                    else:
+                        cols.append(f"({e}) as {c}")
+                else:
+                    cols.append(str(c))
+            self.tables[name] = self.sql(
+                f"select {', '.join(cols)} from {name}")
+            BodoSQLContext._plan_cache.clear()
+            return None
         key = (query, tuple(sorted((n, id(t._lazy_plan))
                                    for n, t in self.tables.items())))
         hit = BodoSQLContext._plan_cache.get(key)
@@ -107,3 +157,31 @@ class BodoSQLContext:
         raise NotImplementedError(
             "bodo_amd executes SQL directly on the C++ backend path; "
             "pandas-codegen output is not produced")
+
+
+def _split_top_level_commas(s: str):
+    """Split on commas not inside parentheses or quotes."""
+    parts, depth, cur, q = [], 0, [], None
+    for ch in s:
+        if q:
+            cur.append(ch)
+            if ch == q:
+                q = None
+            continue
+        if ch in ("'", '"'):
+            q = ch
+            cur.append(ch)
+        elif ch == "(":
+            depth += 1
+            cur.append(ch)
+        elif ch == ")":
+            depth -= 1
+            cur.append(ch)
+        elif ch == "," and depth == 0:
+            parts.append("".join(cur))
+            cur = []
+        else:
+            cur.append(ch)
+    if cur:
+        parts.append("".join(cur))
+    return parts

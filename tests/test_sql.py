@@ -978,3 +978,18 @@ def test_insert_into():
     bc.sql("insert into t select a, s from u")
     out = bc.sql("select count(*) as n, sum(a) as s from t").to_pandas()
     assert out["n"][0] == 6 and out["s"][0] == 80
+
+
+def test_update_delete():
+    bc = BodoSQLContext({"t": pd.DataFrame({"a": [1, 2, 3, 4],
+                                            "y": [1.0, 2.0, 3.0, 4.0]})})
+    bc.sql("update t set y = y * 10 where a >= 3")
+    got = bc.sql("select * from t order by a").to_pandas()
+    assert got["y"].tolist() == [1.0, 2.0, 30.0, 40.0]
+    bc.sql("delete from t where a = 2")
+    assert bc.sql("select a from t order by a").to_pandas()["a"].tolist() \
+        == [1, 3, 4]
+    bc.sql("update t set y = 0, a = a + 100")
+    got = bc.sql("select a, y from t order by a").to_pandas()
+    assert got["a"].tolist() == [101, 103, 104]
+    assert got["y"].tolist() == [0, 0, 0]
