@@ -1162,6 +1162,28 @@ def udf_map(a: Column, func, na_action=None) -> Column:
             return Column(a.dtype, codes, new_mask,
                           dictionary=pa.array(vals, type=pa.large_string()),
                           length=len(a))
+        if any(isinstance(v, (list, tuple)) for v in vals):
+            # list-returning UDF (e.g. ai.tokenize): LIST column over the
+            # dictionary, gathered by codes
+            from . import gather as _g
+
+            elems = [x for v in vals if v for x in v]
+            if any(isinstance(x, str) for x in elems):
+                inner = pa.large_string()
+            elif all(isinstance(x, (int, np.integer)) and
+                     not isinstance(x, bool) for x in elems):
+                inner = pa.int64()
+            else:
+                inner = pa.float64()
+            ltype = pa.large_list(inner)
+            la = pa.array([None if v is None else list(v) for v in vals],
+                          type=ltype)
+            base = Column.from_arrow(la, a.device)
+            out = _g(base, a.data.long())
+            if a.mask is not None:
+                out.mask = a.mask if out.mask is None \
+                    else (out.mask & a.mask)
+            return out
         lut = torch.tensor([np.nan if v is None else v for v in vals],
                            dtype=torch.float64, device=a.device)
         return Column(bt.float64, lut[a.data.long()], a.mask)

@@ -219,6 +219,10 @@ class BodoSeries:
     def struct(self):
         return _StructAccessor(self)
 
+    @property
+    def ai(self):
+        return _AiAccessor(self)
+
     # ------------------------------------------------------------------
     # reductions (lazy scalars: the plan executes on first value access —
     # reference: BodoScalar, bodo/pandas/scalar.py:14)
@@ -509,12 +513,6 @@ class BodoSeries:
 
     def convert_dtypes(self, **kwargs) -> "BodoSeries":
         return self  # columns are already arrow-typed
-
-    @property
-    def ai(self):
-        raise NotImplementedError(
-            "BodoSeries.ai (LLM/embedding calls) requires network access to "
-            "a model endpoint; this build runs fully offline")
 
     def autocorr(self, lag=1) -> float:
         """Pearson autocorrelation at the given lag (distributed shift +
@@ -915,6 +913,54 @@ class _EwmSeries:
 
     def std(self):
         return self._run("std")
+
+
+class _AiAccessor:
+    """Series.ai (reference: BodoSeriesAiMethods, bodo/pandas/series.py:1961).
+    Offline-first: every method takes a local callable/tokenizer — there is
+    no network in this environment, so remote endpoints are the caller's
+    responsibility (pass a client function)."""
+
+    def __init__(self, s: BodoSeries):
+        self._s = s
+
+    def tokenize(self, tokenizer):
+        """Token ids per row as LIST<int64>; `tokenizer` is a HuggingFace
+        tokenizer (has .encode) or any str -> list[int] callable."""
+        enc = tokenizer.encode if hasattr(tokenizer, "encode") else tokenizer
+        from ..plan.expr import UdfMap
+
+        return self._s._wrap(
+            UdfMap(self._s._expr, lambda v: list(enc(str(v))), "ignore"),
+            self._s.name)
+
+    def embed(self, model, batch_size: int = 256):
+        """Embeddings per row as LIST<float>; `model` maps a list[str] to a
+        list of vectors (batched on each shard)."""
+        def _map(pdf_col):
+            out = []
+            vals = pdf_col.tolist()
+            for i in range(0, len(vals), batch_size):
+                out.extend(model([str(v) for v in vals[i:i + batch_size]]))
+            return [list(map(float, e)) for e in out]
+
+        ser = self._s.to_pandas()
+        import pandas as pd_
+
+        res = pd_.Series(_map(ser), index=ser.index, name=self._s.name)
+        from . import from_pandas
+
+        return from_pandas(res.to_frame("v"))["v"]
+
+    def llm_generate(self, fn, **kwargs):
+        """Row-wise generation through a caller-provided client callable
+        (prompt -> completion)."""
+        from ..plan.expr import UdfMap
+
+        return self._s._wrap(
+            UdfMap(self._s._expr,
+                   lambda v, _f=fn, _k=kwargs: _f(str(v), **_k), "ignore"),
+            self._s.name)
 
 
 class _StructAccessor:

@@ -745,3 +745,19 @@ def test_series_rank_pct():
     got = b["x"].rank(method="min", pct=True).to_pandas()
     want = df["x"].rank(method="min", pct=True)
     np.testing.assert_allclose(got.fillna(-9), want.fillna(-9))
+
+
+def test_series_ai_accessor():
+    """Series.ai with local callables (reference: BodoSeriesAiMethods;
+    offline build — endpoints are caller-provided functions)."""
+    df = pd.DataFrame({"s": ["hello world", "foo", None]})
+    b = bpd.from_pandas(df)
+    out = b["s"].ai.tokenize(
+        lambda t: [ord(c) % 97 for c in t[:4]]).to_pandas()
+    assert list(out.iloc[0]) == [ord(c) % 97 for c in "hell"]
+    assert out.iloc[2] is None or pd.isna(out.iloc[2])
+    emb = b.head(2)["s"].ai.embed(
+        lambda batch: [[float(len(x)), 1.0] for x in batch]).to_pandas()
+    assert list(emb.iloc[0]) == [11.0, 1.0]
+    gen = b["s"].ai.llm_generate(lambda p: p.upper()).to_pandas()
+    assert gen.iloc[0] == "HELLO WORLD" and pd.isna(gen.iloc[2])
