@@ -111,6 +111,36 @@ def prange(*args):
     return builtins.range(*args)
 
 
+def typeof(val):
+    """Runtime type inspection (reference: bodo.typeof via numba): returns
+    the engine DType for arrays/Series values, else the python type."""
+    import numpy as np
+
+    from .core import types as bt
+
+    if isinstance(val, np.ndarray):
+        return bt.from_numpy_dtype(val.dtype)
+    try:
+        import pandas as pd
+
+        if isinstance(val, pd.Series):
+            from .core.column import Column
+
+            return Column.from_arrow(
+                __import__("pyarrow").Array.from_pandas(val.head(16))).dtype
+    except Exception:
+        pass
+    return type(val)
+
+
+def parallel_print(*args, **kwargs):
+    """Print once per rank with a rank prefix (reference:
+    bodo.parallel_print)."""
+    from .parallel import comm
+
+    print(f"[rank {comm.get_rank()}]", *args, **kwargs)
+
+
 def dist_reduce(value, op: str = "sum"):
     """Combine a per-rank scalar across ranks (reference:
     bodo/libs/distributed_api.py dist_reduce)."""

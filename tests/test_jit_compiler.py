@@ -221,3 +221,12 @@ def test_distarray_numpy_function_coverage():
 
     with _pt.raises(IndexError):
         da[500]
+
+
+def test_typeof_and_parallel_print(capsys):
+    import bodo_amd
+
+    assert str(bodo_amd.typeof(np.zeros(3, dtype=np.int32))) == "int32"
+    assert str(bodo_amd.typeof(pd.Series(["a", "b"]))) == "string"
+    bodo_amd.parallel_print("x")
+    assert "[rank 0] x" in capsys.readouterr().out
