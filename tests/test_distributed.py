@@ -964,3 +964,35 @@ def test_dist_struct_shuffle():
     exp = df.sort_values("k").reset_index(drop=True)
     assert got["k"].tolist() == exp["k"].tolist()
     assert got["st"].tolist() == exp["st"].tolist()
+
+
+def _q_ml_metrics(bpd, rank, payload):
+    import numpy as np
+
+    from bodo_amd import ml
+
+    X = payload["X"]
+    y = payload["y"]
+    # each rank takes its block (estimator inputs are per-rank shards)
+    w = 2
+    n = len(X)
+    lo, hi = rank * n // w, (rank + 1) * n // w
+    Xl, yl = X[lo:hi], y[lo:hi]
+    mm = ml.MinMaxScaler().fit(Xl)
+    acc = ml.accuracy_score(yl, np.zeros_like(yl))
+    mse = ml.mean_squared_error(yl, np.zeros_like(yl, dtype=float))
+    return {"min": mm.data_min_.tolist(), "max": mm.data_max_.tolist(),
+            "acc": acc, "mse": mse}
+
+
+def test_dist_ml_metrics_and_scaler():
+    """MinMaxScaler min/max and metrics all-reduce across 2 ranks to the
+    global values."""
+    rng = np.random.default_rng(5)
+    X = rng.random((101, 3)) * 4 - 2
+    y = rng.integers(0, 2, 101)
+    got = run_dist(_q_ml_metrics, {"X": X, "y": y})
+    assert np.allclose(got["min"], X.min(axis=0))
+    assert np.allclose(got["max"], X.max(axis=0))
+    assert abs(got["acc"] - (y == 0).mean()) < 1e-12
+    assert abs(got["mse"] - (y.astype(float) ** 2).mean()) < 1e-12
