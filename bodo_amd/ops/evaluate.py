@@ -197,10 +197,25 @@ class _Evaluator:
         return compare(e.op, a, b)
 
     def visit_BoolOp(self, e: BoolOp) -> Column:
+        # SQL three-valued logic: NULL OR TRUE = TRUE, NULL AND FALSE =
+        # FALSE, otherwise NULL propagates.  Dropping the masks here leaked
+        # the (arbitrary) storage value of null rows into filters.
         a, b = self.visit(e.left), self.visit(e.right)
         da, db = a.data, b.data
-        out = (da & db) if e.op == "and" else (da | db)
-        return Column(bt.boolean, out)
+        if a.mask is None and b.mask is None:
+            out = (da & db) if e.op == "and" else (da | db)
+            return Column(bt.boolean, out)
+        av = a.mask if a.mask is not None else torch.ones_like(da)
+        bv = b.mask if b.mask is not None else torch.ones_like(db)
+        at, bt_ = da & av, db & bv          # known TRUE
+        af, bf = (~da) & av, (~db) & bv     # known FALSE
+        if e.op == "or":
+            out = at | bt_
+            valid = at | bt_ | (av & bv)
+        else:
+            out = at & bt_
+            valid = af | bf | (av & bv)
+        return Column(bt.boolean, out, valid)
 
     def visit_Not(self, e: Not) -> Column:
         a = self.visit(e.operand)

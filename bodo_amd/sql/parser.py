@@ -124,6 +124,7 @@ class LikeE:
     operand: Any
     pattern: str
     negated: bool = False
+    ci: bool = False  # ILIKE
 
 
 @dataclass
@@ -308,6 +309,13 @@ class Parser:
     def parse_select(self) -> Query:
         self.expect_kw("select")
         distinct = bool(self.accept_kw("distinct"))
+        top_n = None
+        tt = self.peek()
+        if tt and tt.kind == "id" and tt.value.lower() == "top":
+            nt = self.toks[self.i + 1] if self.i + 1 < len(self.toks) else None
+            if nt and nt.kind == "num":
+                self.i += 2
+                top_n = int(nt.value)
         items = [self.parse_select_item()]
         while self.accept_op(","):
             items.append(self.parse_select_item())
@@ -359,8 +367,12 @@ class Parser:
         grouping_sets = None
         if self.accept_kw("group"):
             self.expect_kw("by")
-            t = self.peek()
-            word = t.value.lower() if t and t.kind == "id" else ""
+            if self.accept_kw("all"):
+                group_by = "ALL"  # resolved by the planner to non-agg items
+                t = None
+            else:
+                t = self.peek()
+            word = t.value.lower() if t is not None and t.kind == "id" else ""
             nxt = self.toks[self.i + 1] if self.i + 1 < len(self.toks) else None
             if word in ("rollup", "cube") and nxt and nxt.kind == "op" \
                     and nxt.value == "(":
@@ -402,7 +414,7 @@ class Parser:
                         if repr(g) not in {repr(x) for x in seen}:
                             seen.append(g)
                 group_by = seen
-            else:
+            elif group_by != "ALL":
                 group_by.append(self.parse_expr())
                 while self.accept_op(","):
                     group_by.append(self.parse_expr())
@@ -415,7 +427,11 @@ class Parser:
         order_by: List[Tuple[Any, bool]] = []
         if self.accept_kw("order"):
             self.expect_kw("by")
-            while True:
+            if self.accept_kw("all"):
+                order_by = "ALL"
+            else:
+                pass
+            while order_by != "ALL":
                 e = self.parse_expr()
                 asc = True
                 if self.accept_kw("desc"):
@@ -433,6 +449,8 @@ class Parser:
             if nt and nt.kind == "id" and nt.value.lower() == "offset":
                 self.i += 1
                 limit_offset = int(self.next().value)
+        if top_n is not None and limit is None:
+            limit = top_n
         q = Query(items, distinct, table, joins, where, group_by, having,
                   order_by, limit, grouping_sets, qualify)
         q.limit_offset = limit_offset
@@ -489,6 +507,18 @@ class Parser:
 
     def parse_select_item(self) -> SelectItem:
         if self.accept_op("*"):
+            nt = self.peek()
+            if nt and nt.kind == "id" and nt.value.lower() == "exclude":
+                self.i += 1
+                bracket = bool(self.accept_op("("))
+                cols = [self.next().value]
+                while self.accept_op(","):
+                    cols.append(self.next().value)
+                if bracket:
+                    self.expect_op(")")
+                it = SelectItem(None, None, star=True)
+                it.exclude = tuple(c.lower() for c in cols)
+                return it
             return SelectItem(None, None, star=True)
         t = self.peek()
         nt = self.toks[self.i + 1] if self.i + 1 < len(self.toks) else None
@@ -535,7 +565,10 @@ class Parser:
         negated = False
         if t and t.kind == "kw" and t.value == "not":
             nxt = self.toks[self.i + 1] if self.i + 1 < len(self.toks) else None
-            if nxt and nxt.kind == "kw" and nxt.value in ("in", "like", "between"):
+            if nxt and ((nxt.kind == "kw" and nxt.value in
+                         ("in", "like", "between"))
+                        or (nxt.kind == "id" and nxt.value.lower() in
+                            ("ilike", "rlike", "regexp"))):
                 self.i += 1
                 negated = True
                 t = self.peek()
@@ -565,8 +598,29 @@ class Parser:
             return BetweenE(left, lo, hi, negated)
         if t and t.kind == "kw" and t.value == "like":
             self.i += 1
+            nt = self.peek()
+            if nt and nt.value.lower() == "any":  # LIKE ANY (p1, p2, ...)
+                self.i += 1
+                self.expect_op("(")
+                pats = [self.next().value]
+                while self.accept_op(","):
+                    pats.append(self.next().value)
+                self.expect_op(")")
+                e = LikeE(left, pats[0], False)
+                for pt in pats[1:]:
+                    e = Bin("or", e, LikeE(left, pt, False))
+                return Un("not", e) if negated else e
             pat = self.next()
             return LikeE(left, pat.value, negated)
+        if t and t.kind == "id" and t.value.lower() == "ilike":
+            self.i += 1
+            pat = self.next()
+            return LikeE(left, pat.value, negated, ci=True)
+        if t and t.kind == "id" and t.value.lower() in ("rlike", "regexp"):
+            self.i += 1
+            pat = self.next()
+            e = Func("regexp_like", [left, Lit(pat.value, "str")])
+            return Un("not", e) if negated else e
         if t and t.kind == "kw" and t.value == "is":
             self.i += 1
             neg = bool(self.accept_kw("not"))

@@ -202,6 +202,10 @@ class Planner:
         # LEFT JOINs before projection planning
         plan = self._rewrite_select_subqueries(plan, scope, q)
 
+        if q.group_by == "ALL":  # Snowflake GROUP BY ALL: non-agg items
+            q.group_by = [it.expr for it in q.items
+                          if not it.star and not _has_agg(it.expr)]
+
         # ------------------------------------------------- aggregation
         has_agg = any(_has_agg(it.expr) for it in q.items if not it.star) \
             or (q.having is not None and _has_agg(q.having)) or q.group_by \
@@ -242,9 +246,12 @@ class Planner:
             for it in q.items:
                 if it.star:
                     want = getattr(it, "star_table", None)
+                    excl = set(getattr(it, "exclude", ()) or ())
                     for alias, col, internal in scope.entries:
                         if want is not None and \
                                 alias.lower() != want.lower():
+                            continue
+                        if col.lower() in excl:
                             continue
                         out_names.append(col)
                         out_exprs.append(ex.ColRef(internal))
@@ -255,6 +262,8 @@ class Planner:
             plan = pn.Projection(plan, tuple(out_names), tuple(out_exprs))
         if q.distinct:
             plan = pn.Distinct(plan, None)
+        if q.order_by == "ALL":  # Snowflake ORDER BY ALL: every output col
+            q.order_by = [(ast.Col(None, n), True) for n in out_names]
         if q.order_by:
             keys, asc = [], []
             hidden: List[str] = []
@@ -1240,7 +1249,12 @@ class Planner:
             out = ex.BoolOp("and", lo, hi)
             return ex.Not(out) if e.negated else out
         if isinstance(e, ast.LikeE):
-            out = _like_expr(self.expr(e.operand, scope), e.pattern)
+            operand = self.expr(e.operand, scope)
+            if getattr(e, "ci", False):  # ILIKE: case-insensitive
+                out = _like_expr(ex.StrOp(operand, "lower"),
+                                 e.pattern.lower())
+            else:
+                out = _like_expr(operand, e.pattern)
             return ex.Not(out) if e.negated else out
         if isinstance(e, ast.IsNullE):
             return ex.IsNull(self.expr(e.operand, scope), negate=e.negated)

@@ -993,3 +993,33 @@ def test_update_delete():
     got = bc.sql("select a, y from t order by a").to_pandas()
     assert got["a"].tolist() == [101, 103, 104]
     assert got["y"].tolist() == [0, 0, 0]
+
+
+def test_snowflake_sugar_and_three_valued_logic():
+    """ILIKE / LIKE ANY / RLIKE infix / GROUP BY ALL / ORDER BY ALL /
+    SELECT * EXCLUDE / TOP n, and NULL-correct boolean OR/AND (three-
+    valued logic: dropping masks leaked null rows into filters)."""
+    df = pd.DataFrame({"s": ["Apple", "banana", "Cherry", None],
+                       "v": [1, 2, 3, 4], "g": ["x", "x", "y", "y"]})
+    bc = BodoSQLContext({"t": df})
+    assert bc.sql("select s from t where s ilike 'a%'") \
+        .to_pandas()["s"].tolist() == ["Apple"]
+    assert bc.sql("select s from t where s like any ('A%', 'b%')") \
+        .to_pandas()["s"].tolist() == ["Apple", "banana"]
+    assert bc.sql("select s from t where s rlike '^[AC]'") \
+        .to_pandas()["s"].astype(str).sorted_values \
+        if False else True
+    got = bc.sql("select s from t where s rlike '^[AC]'").to_pandas()
+    assert sorted(map(str, got["s"])) == ["Apple", "Cherry"]
+    got = bc.sql("select g, sum(v) as sv from t group by all "
+                 "order by g").to_pandas()
+    assert got["sv"].tolist() == [3, 7]
+    got = bc.sql("select * exclude (v) from t limit 1").to_pandas()
+    assert list(got.columns) == ["s", "g"]
+    assert bc.sql("select top 2 v from t order by v desc") \
+        .to_pandas()["v"].tolist() == [4, 3]
+    # three-valued logic: NULL OR TRUE is TRUE; NULL OR FALSE filters out
+    assert bc.sql("select v from t where s like 'Z%' or v > 3") \
+        .to_pandas()["v"].tolist() == [4]
+    assert bc.sql("select s from t where s like 'A%' or s like 'b%'") \
+        .to_pandas()["s"].tolist() == ["Apple", "banana"]
