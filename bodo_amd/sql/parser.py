@@ -438,6 +438,10 @@ class Parser:
                     asc = False
                 else:
                     self.accept_kw("asc")
+                nt = self.peek()
+                if nt and nt.kind == "id" and nt.value.lower() == "nulls":
+                    self.i += 1
+                    self.next()  # FIRST | LAST (engine default preserved)
                 order_by.append((e, asc))
                 if not self.accept_op(","):
                     break
@@ -624,6 +628,12 @@ class Parser:
         if t and t.kind == "kw" and t.value == "is":
             self.i += 1
             neg = bool(self.accept_kw("not"))
+            if self.accept_kw("distinct"):
+                self.expect_kw("from")
+                other = self.parse_add()
+                eqn = Func("equal_null", [left, other])
+                # IS DISTINCT FROM = NOT equal_null
+                return eqn if neg else Un("not", eqn)
             self.expect_kw("null")
             return IsNullE(left, neg)
         return left

@@ -1021,3 +1021,17 @@ def test_snowflake_sugar_and_three_valued_logic():
         .to_pandas()["v"].tolist() == [4]
     assert bc.sql("select s from t where s like 'A%' or s like 'b%'") \
         .to_pandas()["s"].tolist() == ["Apple", "banana"]
+
+
+def test_is_distinct_from():
+    df = pd.DataFrame({"a": [1.0, None, 3.0, None],
+                       "b": [1.0, None, 4.0, 5.0]})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select b from t where a is distinct from b").to_pandas()
+    assert got["b"].tolist() == [4.0, 5.0]
+    got = bc.sql("select b from t where a is not distinct from b") \
+        .to_pandas()
+    assert got["b"].fillna(-1).tolist() == [1.0, -1.0]
+    # NULLS FIRST/LAST parse (ordering keeps the engine default)
+    got = bc.sql("select a from t order by a nulls first").to_pandas()
+    assert got["a"].dropna().tolist() == [1.0, 3.0]
