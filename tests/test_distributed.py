@@ -996,3 +996,32 @@ def test_dist_ml_metrics_and_scaler():
     assert np.allclose(got["max"], X.max(axis=0))
     assert abs(got["acc"] - (y == 0).mean()) < 1e-12
     assert abs(got["mse"] - (y.astype(float) ** 2).mean()) < 1e-12
+
+
+def _q_grouped_value_aggs(bpd, rank, payload):
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": payload["df"]})
+    out = bc.sql(
+        "select c, min(c) as mc, max(d) as xd, approx_count_distinct(a) "
+        "as ad, mode(a) as mo from t group by c order by c").to_pandas()
+    return out.reset_index(drop=True)
+
+
+def test_dist_grouped_value_aggs():
+    """2-rank: single-phase grouped value aggs added in round 2 (dict-key
+    min/max, approx_count_distinct, mode) keep collective symmetry."""
+    rng = np.random.default_rng(9)
+    n = 300
+    df = pd.DataFrame({"a": rng.integers(0, 7, n),
+                       "c": rng.choice(["x", "y", "z"], n),
+                       "d": pd.to_datetime(1.6e18 + rng.integers(0, 1e16, n))})
+    got = run_dist(_q_grouped_value_aggs, {"df": df})
+    exp = df.groupby("c").agg(
+        xd=("d", "max"), ad=("a", "nunique"),
+        mo=("a", lambda s: s.mode().iloc[0])).reset_index()
+    assert got["c"].astype(str).tolist() == exp["c"].tolist()
+    assert got["mc"].astype(str).tolist() == exp["c"].tolist()
+    assert [pd.Timestamp(v) for v in got["xd"]] == list(exp["xd"])
+    assert got["ad"].tolist() == exp["ad"].tolist()
+    assert got["mo"].tolist() == exp["mo"].tolist()
