@@ -496,12 +496,25 @@ class Planner:
         pre_names = list(key_map.values()) + list(agg_map.values())
         pre_exprs = {n: ex.ColRef(n) for n in pre_names}
         need_pre = False
+        # Snowflake scope: QUALIFY / OVER may reference SELECT-list aliases
+        # (e.g. count(*) AS n ... ORDER BY n) — substitute before resolving
+        alias_map = {it.alias.lower(): it.expr for it in q.items
+                     if it.alias and not it.star}
+
+        def dealias(e):
+            if isinstance(e, ast.Col) and e.table is None \
+                    and e.name.lower() in alias_map:
+                k = _ast_key(e)
+                if k not in key_map and k not in agg_map:
+                    return alias_map[e.name.lower()]
+            return e
 
         def conv(e):
-            return self._post_agg_expr(e, key_map, agg_map, scope)
+            return self._post_agg_expr(dealias(e), key_map, agg_map, scope)
 
         def as_col(e, base):
             nonlocal need_pre
+            e = dealias(e)
             k = _ast_key(e)
             if k in key_map:
                 return key_map[k]
@@ -1721,8 +1734,8 @@ class Planner:
             if name == "width_bucket":
                 # WIDTH_BUCKET(x, lo, hi, n): 1..n inside, 0 below, n+1 above
                 x = self.expr(e.args[0], scope)
-                lo = float(e.args[1].value)
-                hi = float(e.args[2].value)
+                lo = float(_lit_num(e.args[1]))
+                hi = float(_lit_num(e.args[2]))
                 nb = _lit_int(e.args[3])
                 w = (hi - lo) / nb
                 raw = ex.BinOp("floordiv",
@@ -1777,6 +1790,17 @@ def _lit_int(e) -> int:
             and isinstance(e.right, ast.Lit):
         return -int(e.right.value)
     raise NotImplementedError(f"constant integer expected, got {e!r}")
+
+
+def _lit_num(e) -> float:
+    """Constant number from a literal or unary-minus literal."""
+    if isinstance(e, ast.Lit):
+        return float(e.value)
+    if isinstance(e, ast.Bin) and e.op == "sub" \
+            and isinstance(e.left, ast.Lit) and e.left.value == 0 \
+            and isinstance(e.right, ast.Lit):
+        return -float(e.right.value)
+    raise NotImplementedError(f"constant number expected, got {e!r}")
 
 
 def _lit_expr(e: ast.Lit) -> ex.Expr:

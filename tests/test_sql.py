@@ -1035,3 +1035,16 @@ def test_is_distinct_from():
     # NULLS FIRST/LAST parse (ordering keeps the engine default)
     got = bc.sql("select a from t order by a nulls first").to_pandas()
     assert got["a"].dropna().tolist() == [1.0, 3.0]
+
+
+def test_qualify_alias_and_width_bucket_negatives():
+    df = pd.DataFrame({"g": list("aabbbcccc"), "x": range(-4, 5)})
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql("select g, count(*) as n from t group by g "
+                 "qualify row_number() over (order by n desc) <= 2") \
+        .to_pandas()
+    assert sorted(zip(out["g"].astype(str), out["n"])) == \
+        [("b", 3), ("c", 4)]
+    wb = bc.sql("select width_bucket(x, -4, 4, 4) as b from t") \
+        .to_pandas()["b"]
+    assert wb.tolist() == [1, 1, 2, 2, 3, 3, 4, 4, 5]
