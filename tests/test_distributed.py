@@ -1025,3 +1025,18 @@ def test_dist_grouped_value_aggs():
     assert [pd.Timestamp(v) for v in got["xd"]] == list(exp["xd"])
     assert got["ad"].tolist() == exp["ad"].tolist()
     assert got["mo"].tolist() == exp["mo"].tolist()
+
+
+def test_dist_struct_shuffle_4rank():
+    """4-rank struct shuffle incl. an empty-ish rank distribution."""
+    rng = np.random.default_rng(13)
+    n = 37  # odd count over 4 ranks -> uneven blocks
+    df = pd.DataFrame({
+        "k": rng.permutation(n),
+        "st": [None if i % 5 == 0 else {"x": int(i), "y": f"v{i}"}
+               for i in range(n)],
+    })
+    got = run_dist(_q_struct_shuffle, {"df": df}, world=4)
+    exp = df.sort_values("k").reset_index(drop=True)
+    assert got["k"].tolist() == exp["k"].tolist()
+    assert got["st"].tolist() == exp["st"].tolist()
