@@ -21,6 +21,7 @@ from ..core.table import Table
 from ..core.types import DType, TypeKind
 from ..plan.expr import (
     BinOp, BoolOp, Case, Cast, ColRef, Cmp, Const, DtField, Expr, IsIn,
+    RandomExpr,
     IsNull, ListOp, Not, RoundExpr, ScalarSubquery, SemiJoinIn, StrOp,
     UdfMap,
 )
@@ -91,6 +92,16 @@ class _Evaluator:
             dtype = infer_const_dtype(v)
         v = normalize_const(v, dtype)
         return Column.full_const(v, dtype, self.n, self.device)
+
+    def visit_RandomExpr(self, e) -> Column:
+        g = torch.Generator(device="cpu")
+        from ..parallel import comm as _c
+
+        g.manual_seed((int(e.seed) * 0x9E3779B9 + _c.get_rank())
+                      & 0x7FFFFFFF)
+        vals = torch.randint(0, 1 << 62, (self.n,), generator=g,
+                             dtype=torch.int64)
+        return Column(bt.int64, vals.to(self.device))
 
     def visit_ScalarSubquery(self, e) -> Column:
         # projected uncorrelated scalar subquery: evaluate once (memoized),

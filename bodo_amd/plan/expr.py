@@ -269,6 +269,20 @@ class ScalarSubquery(Expr):
 
 
 @dataclass(frozen=True)
+class RandomExpr(Expr):
+    """Per-row deterministic pseudo-random int64 stream (SQL RANDOM()):
+    seeded so plan re-execution reproduces the same values."""
+
+    seed: int = 0
+
+    def children(self):
+        return ()
+
+    def with_children(self, *ch):
+        return self
+
+
+@dataclass(frozen=True)
 class RoundExpr(Expr):
     operand: Expr
     decimals: int = 0

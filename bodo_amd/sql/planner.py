@@ -1286,7 +1286,73 @@ class Planner:
             els = self.expr(e.els, scope) if e.els is not None else ex.Const(None, bt.float64)
             return ex.Case(conds, thens, els)
         if isinstance(e, ast.Func):
-            name = e.name
+            name = _FN_ALIASES.get(e.name, e.name)
+            if name != e.name:
+                e = ast.Func(name, e.args, star=getattr(e, "star", False),
+                             distinct=getattr(e, "distinct", False))
+            if name == "div0":
+                # a / b, 0 when b = 0 (Snowflake DIV0)
+                a = self.expr(e.args[0], scope)
+                b2 = self.expr(e.args[1], scope)
+                return ex.Case((ex.Cmp("eq", b2, ex.Const(0)),),
+                               (ex.Const(0.0),), ex.BinOp("div", a, b2))
+            if name == "square":
+                a = self.expr(e.args[0], scope)
+                return ex.BinOp("mul", a, a)
+            if name == "factorial":
+                import math as _m
+
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v: _m.factorial(int(v)), "ignore")
+            if name in ("booland", "boolor"):
+                a = self.expr(e.args[0], scope)
+                b2 = self.expr(e.args[1], scope)
+                op = "and" if name == "booland" else "or"
+                return ex.BoolOp(op, ex.Cmp("ne", a, ex.Const(0)),
+                                 ex.Cmp("ne", b2, ex.Const(0)))
+            if name == "boolnot":
+                return ex.Cmp("eq", self.expr(e.args[0], scope),
+                              ex.Const(0))
+            if name == "strtok":
+                # STRTOK(s, delim, part): 1-based like SPLIT_PART but NULL
+                # past the end
+                delim = e.args[1].value if len(e.args) > 1 else " "
+                part = _lit_int(e.args[2]) if len(e.args) > 2 else 1
+                return ex.StrOp(self.expr(e.args[0], scope), "split_get",
+                                (delim, part - 1))
+            if name == "insert" and len(e.args) == 4:
+                # INSERT(s, pos, len, repl): splice (1-based)
+                pos = _lit_int(e.args[1])
+                ln = _lit_int(e.args[2])
+                repl = e.args[3].value
+                return ex.UdfMap(
+                    self.expr(e.args[0], scope),
+                    lambda v, _p=pos, _l=ln, _r=repl:
+                    str(v)[:_p - 1] + _r + str(v)[_p - 1 + _l:], "ignore")
+            if name == "rtrimmed_length":
+                return ex.StrOp(ex.StrOp(self.expr(e.args[0], scope),
+                                         "rstrip"), "len")
+            if name in ("to_char", "to_varchar"):
+                return ex.Cast(self.expr(e.args[0], scope), bt.string)
+            if name in ("to_number", "to_decimal", "to_numeric"):
+                return ex.Cast(self.expr(e.args[0], scope), bt.float64)
+            if name == "to_double":
+                return ex.Cast(self.expr(e.args[0], scope), bt.float64)
+            if name == "random":
+                # per-row deterministic stream; each call site gets its own
+                # seed so two random() in one query differ
+                self._counter += 1
+                return ex.RandomExpr(self._counter)
+            if name == "uniform" and len(e.args) == 3:
+                lo = _lit_num(e.args[0])
+                hi = _lit_num(e.args[1])
+                gen = self.expr(e.args[2], scope)
+                span = hi - lo
+                scaled = ex.BinOp("div", gen, ex.Const(float(1 << 63)))
+                return ex.BinOp("add", ex.Const(lo),
+                                ex.BinOp("mul", scaled, ex.Const(span)))
+            if name == "is_null_value":
+                return ex.IsNull(self.expr(e.args[0], scope))
             if name in ("year", "month", "day", "hour", "minute", "second",
                         "quarter"):
                 return ex.DtField(self.expr(e.args[0], scope), name)
@@ -1790,6 +1856,16 @@ def _lit_int(e) -> int:
             and isinstance(e.right, ast.Lit):
         return -int(e.right.value)
     raise NotImplementedError(f"constant integer expected, got {e!r}")
+
+
+# Snowflake alias names -> canonical function names
+_FN_ALIASES = {
+    "lcase": "lower", "ucase": "upper", "char_length": "length",
+    "character_length": "length", "charindex": "position",
+    "ifnull": "coalesce", "nvl": "coalesce",
+    "ceiling": "ceil", "mod_": "mod",
+    "strtok_to_array": "split", "len": "length",
+}
 
 
 def _lit_num(e) -> float:

@@ -1048,3 +1048,35 @@ def test_qualify_alias_and_width_bucket_negatives():
     wb = bc.sql("select width_bucket(x, -4, 4, 4) as b from t") \
         .to_pandas()["b"]
     assert wb.tolist() == [1, 1, 2, 2, 3, 3, 4, 4, 5]
+
+
+def test_function_aliases_batch():
+    """Snowflake alias/utility functions: LCASE/UCASE/CHARINDEX/STRTOK/
+    INSERT/TO_CHAR/TO_NUMBER/DIV0/IFNULL/BOOL*/FACTORIAL/SQUARE/RANDOM/
+    UNIFORM."""
+    df = pd.DataFrame({"s": ["a b c", "x,y", None],
+                       "v": [1.5, -2.5, 3.0], "i": [5, 10, 15]})
+    bc = BodoSQLContext({"t": df})
+    assert bc.sql("select ucase(s) as r from t").to_pandas()["r"][0] \
+        == "A B C"
+    assert bc.sql("select charindex('b', s) as r from t") \
+        .to_pandas()["r"].tolist()[:2] == [3, 0]
+    assert bc.sql("select strtok(s, ' ', 2) as r from t") \
+        .to_pandas()["r"][0] == "b"
+    assert bc.sql("select insert(s, 2, 1, 'Z') as r from t") \
+        .to_pandas()["r"][0] == "aZb c"
+    assert bc.sql("select div0(v, 0) as r from t") \
+        .to_pandas()["r"].tolist() == [0.0, 0.0, 0.0]
+    assert bc.sql("select ifnull(null, 7) as r from t") \
+        .to_pandas()["r"].tolist() == [7, 7, 7]
+    assert bc.sql("select factorial(i) as r from t") \
+        .to_pandas()["r"][0] == 120
+    assert bc.sql("select square(v) as r from t") \
+        .to_pandas()["r"].tolist() == [2.25, 6.25, 9.0]
+    r = bc.sql("select random() as r from t").to_pandas()["r"]
+    assert len(set(r)) == 3  # per-row stream
+    u = bc.sql("select uniform(0, 10, random()) as u from t") \
+        .to_pandas()["u"]
+    assert bool(((u >= 0) & (u <= 10)).all())
+    assert bc.sql("select to_number('42') as r from t") \
+        .to_pandas()["r"][0] == 42.0
