@@ -205,6 +205,26 @@ class Planner:
         if q.group_by == "ALL":  # Snowflake GROUP BY ALL: non-agg items
             q.group_by = [it.expr for it in q.items
                           if not it.star and not _has_agg(it.expr)]
+        # GROUP BY ordinal (1-based) / SELECT alias; HAVING may also use
+        # aliases (Snowflake scoping)
+        alias_map = {it.alias.lower(): it.expr for it in q.items
+                     if it.alias and not it.star}
+        if isinstance(q.group_by, list) and q.group_by:
+            def norm_gb(e):
+                if isinstance(e, ast.Lit) and e.kind == "num":
+                    return q.items[int(e.value) - 1].expr
+                if isinstance(e, ast.Col) and e.table is None \
+                        and e.name.lower() in alias_map:
+                    try:
+                        scope.resolve(None, e.name)
+                        return e  # a real column of that name wins
+                    except KeyError:
+                        return alias_map[e.name.lower()]
+                return e
+
+            q.group_by = [norm_gb(e) for e in q.group_by]
+        if q.having is not None and alias_map:
+            q.having = _sub_aliases(q.having, alias_map, scope)
 
         # ------------------------------------------------- aggregation
         has_agg = any(_has_agg(it.expr) for it in q.items if not it.star) \
@@ -1866,6 +1886,30 @@ _FN_ALIASES = {
     "ceiling": "ceil", "mod_": "mod",
     "strtok_to_array": "split", "len": "length",
 }
+
+
+def _sub_aliases(e, alias_map, scope):
+    """Replace unqualified Col refs that name a SELECT alias (and are not
+    real source columns) with the aliased expression — Snowflake HAVING /
+    GROUP BY scoping."""
+    if isinstance(e, ast.Col) and e.table is None \
+            and e.name.lower() in alias_map:
+        try:
+            scope.resolve(None, e.name)
+            return e
+        except KeyError:
+            return alias_map[e.name.lower()]
+    if isinstance(e, (ast.Query, ast.SetOpQ)):
+        return e
+    for f in getattr(e, "__dataclass_fields__", {}):
+        v = getattr(e, f)
+        if isinstance(v, list):
+            setattr(e, f, [_sub_aliases(x, alias_map, scope)
+                           if hasattr(x, "__dataclass_fields__") else x
+                           for x in v])
+        elif hasattr(v, "__dataclass_fields__"):
+            setattr(e, f, _sub_aliases(v, alias_map, scope))
+    return e
 
 
 def _lit_num(e) -> float:

@@ -1080,3 +1080,18 @@ def test_function_aliases_batch():
     assert bool(((u >= 0) & (u <= 10)).all())
     assert bc.sql("select to_number('42') as r from t") \
         .to_pandas()["r"][0] == 42.0
+
+
+def test_group_having_alias_scoping():
+    """GROUP BY ordinal / SELECT alias, HAVING alias (Snowflake scoping)."""
+    df = pd.DataFrame({"g": list("aabbbcc"), "v": range(7)})
+    bc = BodoSQLContext({"t": df})
+    got = bc.sql("select g, count(*) as n from t group by g having n > 2") \
+        .to_pandas()
+    assert got.to_dict("records") == [{"g": "b", "n": 3}]
+    got = bc.sql("select g, count(*) as n from t group by 1 "
+                 "order by 2 desc").to_pandas()
+    assert got["n"].tolist() == [3, 2, 2]
+    got = bc.sql("select g as grp, count(*) as n from t group by grp") \
+        .to_pandas()
+    assert sorted(got["n"]) == [2, 2, 3]
