@@ -999,6 +999,19 @@ def _str_dict_generic(a: Column, op: str, args, kwargs) -> Column:
                 out.mask = a.mask if out.mask is None \
                     else (out.mask & a.mask)
             return out
+        if any(isinstance(v, bytes) for v in out_vals):
+            # bytes results (.str.encode): BINARY column over the dict,
+            # gathered by codes
+            from . import gather as _g
+
+            ba = pa.array([v if isinstance(v, bytes) else None
+                           for v in out_vals], type=pa.large_binary())
+            base = Column.from_arrow(ba, a.device)
+            out = _g(base, a.data.long())
+            if a.mask is not None:
+                out.mask = a.mask if out.mask is None \
+                    else (out.mask & a.mask)
+            return out
         # missing results (e.g. split().get(i) past the end) come back as
         # float NaN: they become validity-mask nulls, never dictionary
         # entries (arrow dicts reject null categories)

@@ -761,3 +761,12 @@ def test_series_ai_accessor():
     assert list(emb.iloc[0]) == [11.0, 1.0]
     gen = b["s"].ai.llm_generate(lambda p: p.upper()).to_pandas()
     assert gen.iloc[0] == "HELLO WORLD" and pd.isna(gen.iloc[2])
+
+
+def test_str_encode_binary():
+    """.str.encode returns BINARY columns (dict and plain layouts)."""
+    df = pd.DataFrame({"s": ["Hello", None, "abc"]})
+    b = bpd.from_pandas(df)
+    got = b["s"].str.encode("utf-8").to_pandas()
+    assert got.iloc[0] == b"Hello" and pd.isna(got.iloc[1]) \
+        and got.iloc[2] == b"abc"
