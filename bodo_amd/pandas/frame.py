@@ -555,6 +555,18 @@ class BodoDataFrame:
     def iloc(self):
         return _ILoc(self)
 
+    def corrwith(self, other):
+        """Per-column Pearson correlation with a series (distributed
+        pairwise co-moments)."""
+        import pandas as _pd
+
+        out = {}
+        head = self.head(1).to_pandas()
+        for c in self._columns:
+            if _pd.api.types.is_numeric_dtype(head[c].dtype):
+                out[c] = float(self[c].corr(other))
+        return _pd.Series(out)
+
     def rolling(self, window, min_periods=None, center=False, **kwargs):
         if isinstance(window, str):
             return _RollingHost(self, window, min_periods)

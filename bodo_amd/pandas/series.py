@@ -523,6 +523,9 @@ class BodoSeries:
         b = BodoSeries(plan, ColRef("__sh"), None)
         return a.corr(b)
 
+    def dot(self, other):
+        return (self * other).sum()
+
     def rolling(self, window, min_periods=None, center=False, **kwargs):
         if isinstance(window, str):
             return _RollingSeriesHost(self, window, min_periods)

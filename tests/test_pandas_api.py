@@ -770,3 +770,13 @@ def test_str_encode_binary():
     got = b["s"].str.encode("utf-8").to_pandas()
     assert got.iloc[0] == b"Hello" and pd.isna(got.iloc[1]) \
         and got.iloc[2] == b"abc"
+
+
+def test_corrwith_and_dot():
+    rng = np.random.default_rng(1)
+    df = pd.DataFrame({"x": rng.random(50), "y": rng.random(50),
+                       "z": rng.random(50)})
+    b = bpd.from_pandas(df)
+    np.testing.assert_allclose(b[["x", "y"]].corrwith(b["z"]).values,
+                               df[["x", "y"]].corrwith(df["z"]).values)
+    assert abs(float(b["x"].dot(b["y"])) - df["x"].dot(df["y"])) < 1e-9
