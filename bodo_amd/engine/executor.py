@@ -171,7 +171,8 @@ def _exec_filter(node: pn.Filter, ctx) -> Table:
 
 
 SINGLE_PHASE_AGGS = {"median", "nunique", "approx_nunique", "var", "std",
-                     "quantile", "skew", "kurt", "sem", "mode"}
+                     "quantile", "skew", "kurt", "sem", "mode",
+                     "array_agg"}
 
 
 def _exec_aggregate(node: pn.Aggregate, ctx) -> Table:
@@ -1104,6 +1105,11 @@ def _combine_reduce(partials: List[dict], func: str):
         if not vs:
             return float("nan")
         return float(np.median(np.concatenate(vs)))
+    if func == "array_agg":
+        out = []
+        for p_ in partials:
+            out.extend(p_["vals_list"])
+        return out
     if func == "mode":
         counts: dict = {}
         for p in partials:

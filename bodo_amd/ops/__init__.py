@@ -536,7 +536,8 @@ def _reduce_wrap_val(col: Column, v):
 
 # reductions where the partial depends on actual VALUES (strings must not
 # reduce over dictionary codes)
-_VALUE_REDUCES = {"min", "max", "first", "last", "mode", "median", "nunique"}
+_VALUE_REDUCES = {"min", "max", "first", "last", "mode", "median",
+                  "nunique", "array_agg"}
 
 
 def reduce_column(col: Column, func: str):
@@ -627,4 +628,7 @@ def reduce_column(col: Column, func: str):
         from ..utils import sketches
 
         return {"hll": sketches.hll_registers([col]).cpu()}
+    if func == "array_agg":
+        return {"vals_list": [v for v in col.to_arrow().to_pylist()
+                              if v is not None]}
     raise NotImplementedError(f"reduce {func}")

@@ -21,7 +21,7 @@ from ..core.table import Table
 from ..core.types import DType, TypeKind
 from ..plan.expr import (
     BinOp, BoolOp, Case, Cast, ColRef, Cmp, Const, DtField, Expr, IsIn,
-    RandomExpr,
+    ListBuild, RandomExpr, StructBuild,
     IsNull, ListOp, Not, RoundExpr, ScalarSubquery, SemiJoinIn, StrOp,
     UdfMap,
 )
@@ -102,6 +102,37 @@ class _Evaluator:
         vals = torch.randint(0, 1 << 62, (self.n,), generator=g,
                              dtype=torch.int64)
         return Column(bt.int64, vals.to(self.device))
+
+    def visit_ListBuild(self, e) -> Column:
+        cols = [self.visit(x) for x in e.items]
+        k = len(cols)
+        n = self.n
+        dev = self.device
+        offs = torch.arange(0, (n + 1) * k, k, dtype=torch.int64,
+                            device=dev)
+        # interleave: child row r*k+j = cols[j][r]
+        datas = [c.data.to(torch.float64) if c.data.dtype.is_floating_point
+                 or any(x.data.dtype.is_floating_point for x in cols)
+                 else c.data.to(torch.int64) for c in cols]
+        child_data = torch.stack(datas, dim=1).reshape(-1)
+        child_mask = None
+        if any(c.mask is not None for c in cols):
+            ms = [c.mask if c.mask is not None
+                  else torch.ones(n, dtype=torch.bool, device=dev)
+                  for c in cols]
+            child_mask = torch.stack(ms, dim=1).reshape(-1)
+        child = Column(
+            bt.float64 if child_data.dtype.is_floating_point else bt.int64,
+            child_data, child_mask)
+        out = Column(bt.list_, None, None, offsets=offs, length=n)
+        out.child = child
+        return out
+
+    def visit_StructBuild(self, e) -> Column:
+        cols = [self.visit(x) for x in e.items]
+        out = Column(bt.struct_(e.names), None, None, length=self.n)
+        out.children = cols
+        return out
 
     def visit_ScalarSubquery(self, e) -> Column:
         # projected uncorrelated scalar subquery: evaluate once (memoized),

@@ -27,7 +27,7 @@ PHYSICAL_AGGS = {
 }
 
 # funcs with no device kernel yet: run in host pandas even on GPU shards
-HOST_ONLY_AGGS = {"mode"}
+HOST_ONLY_AGGS = {"mode", "array_agg"}
 
 
 def _normalize_decimal_aggs(tbl: Table, aggs):
@@ -131,6 +131,10 @@ def _groupby_pandas(tbl: Table, keys, aggs, dropna) -> Table:
         elif func == "kurt":
             named[out_name] = pd.NamedAgg(column=in_name,
                                           aggfunc=lambda s: s.kurt())
+        elif func == "array_agg":
+            named[out_name] = pd.NamedAgg(
+                column=in_name,
+                aggfunc=lambda s: list(s.dropna()))
         elif func == "size":
             named[out_name] = pd.NamedAgg(column=df.columns[0] if not in_name or in_name not in df.columns else in_name, aggfunc="size")
         else:

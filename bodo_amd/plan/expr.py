@@ -269,6 +269,33 @@ class ScalarSubquery(Expr):
 
 
 @dataclass(frozen=True)
+class ListBuild(Expr):
+    """ARRAY_CONSTRUCT(e1, ..): fixed-arity LIST per row."""
+
+    items: Tuple[Expr, ...] = ()
+
+    def children(self):
+        return self.items
+
+    def with_children(self, *ch):
+        return ListBuild(tuple(ch))
+
+
+@dataclass(frozen=True)
+class StructBuild(Expr):
+    """OBJECT_CONSTRUCT('k', v, ..): STRUCT column per row."""
+
+    names: Tuple[str, ...] = ()
+    items: Tuple[Expr, ...] = ()
+
+    def children(self):
+        return self.items
+
+    def with_children(self, *ch):
+        return StructBuild(self.names, tuple(ch))
+
+
+@dataclass(frozen=True)
 class RandomExpr(Expr):
     """Per-row deterministic pseudo-random int64 stream (SQL RANDOM()):
     seeded so plan re-execution reproduces the same values."""

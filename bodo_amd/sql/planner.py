@@ -16,6 +16,7 @@ from . import parser as ast
 AGG_FUNCS = {"sum": "sum", "avg": "mean", "count": "count", "min": "min",
              "max": "max", "stddev": "std", "variance": "var",
              "median": "median", "approx_count_distinct": "approx_nunique",
+             "array_agg": "array_agg",
              "mode": "mode", "kurtosis": "kurt", "skew": "skew",
              "stddev_samp": "std", "var_samp": "var", "any_value": "first",
              "booland_agg": "all", "boolor_agg": "any"}
@@ -1310,6 +1311,15 @@ class Planner:
             if name != e.name:
                 e = ast.Func(name, e.args, star=getattr(e, "star", False),
                              distinct=getattr(e, "distinct", False))
+            if name == "array_construct":
+                return ex.ListBuild(tuple(self.expr(a, scope)
+                                          for a in e.args))
+            if name == "object_construct":
+                names = tuple(str(e.args[i].value)
+                              for i in range(0, len(e.args), 2))
+                items = tuple(self.expr(e.args[i], scope)
+                              for i in range(1, len(e.args), 2))
+                return ex.StructBuild(names, items)
             if name == "div0":
                 # a / b, 0 when b = 0 (Snowflake DIV0)
                 a = self.expr(e.args[0], scope)

@@ -1095,3 +1095,23 @@ def test_group_having_alias_scoping():
     got = bc.sql("select g as grp, count(*) as n from t group by grp") \
         .to_pandas()
     assert sorted(got["n"]) == [2, 2, 3]
+
+
+def test_semi_structured_constructors_and_array_agg():
+    """ARRAY_CONSTRUCT / OBJECT_CONSTRUCT / ARRAY_AGG (Snowflake
+    semi-structured; reference: bodosql variant kernels)."""
+    df = pd.DataFrame({"g": ["a", "a", "b"], "x": [1, 2, 3],
+                       "y": [0.5, 1.5, 2.5]})
+    bc = BodoSQLContext({"t": df})
+    a = bc.sql("select array_construct(x, x*2) as a from t") \
+        .to_pandas()["a"]
+    assert [list(v) for v in a] == [[1, 2], [2, 4], [3, 6]]
+    o = bc.sql("select object_construct('p', x, 'q', y) as o from t") \
+        .to_pandas()["o"]
+    assert o.tolist() == [{"p": 1, "q": 0.5}, {"p": 2, "q": 1.5},
+                          {"p": 3, "q": 2.5}]
+    ga = bc.sql("select g, array_agg(x) as ax from t group by g "
+                "order by g").to_pandas()["ax"]
+    assert [sorted(v) for v in ga] == [[1, 2], [3]]
+    ra = bc.sql("select array_agg(x) as ax from t").to_pandas()["ax"]
+    assert sorted(ra.iloc[0]) == [1, 2, 3]
