@@ -1040,3 +1040,25 @@ def test_dist_struct_shuffle_4rank():
     exp = df.sort_values("k").reset_index(drop=True)
     assert got["k"].tolist() == exp["k"].tolist()
     assert got["st"].tolist() == exp["st"].tolist()
+
+
+def _q_array_agg(bpd, rank, payload):
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": payload["df"]})
+    out = bc.sql("select g, array_agg(x) as ax from t group by g "
+                 "order by g").to_pandas()
+    return {"lists": [sorted(v) for v in out["ax"]],
+            "g": out["g"].astype(str).tolist()}
+
+
+def test_dist_array_agg():
+    rng = np.random.default_rng(17)
+    n = 120
+    df = pd.DataFrame({"g": rng.choice(["a", "b", "c"], n),
+                       "x": rng.integers(0, 50, n)})
+    got = run_dist(_q_array_agg, {"df": df})
+    exp = df.groupby("g")["x"].apply(lambda s: sorted(s)).to_dict()
+    assert got["g"] == sorted(exp)
+    for g, lst in zip(got["g"], got["lists"]):
+        assert lst == exp[g]
