@@ -1048,6 +1048,9 @@ def _finish_reduce(node: pn.Reduce, ctx, partials: dict) -> Table:
 
 
 def _combine_reduce(partials: List[dict], func: str):
+    if callable(func):
+        ser = pd.Series([v for p in partials for v in p["vals_list"]])
+        return func(ser)
     if func in ("sum",):
         return sum(p["sum"] for p in partials)
     if func in ("count", "size"):

@@ -544,6 +544,10 @@ def reduce_column(col: Column, func: str):
     """Local partial reduction -> dict of partials (combined across ranks by
     the executor's _combine_reduce).  Reference role: the parallel agg
     combine in bodo/libs/_groupby_ftypes + distributed_api dist_reduce."""
+    if callable(func):
+        # custom python agg (LISTAGG / PERCENTILE_*): ship local values,
+        # the combine applies the callable to the concatenation
+        return {"vals_list": col.to_arrow().to_pylist()}
     if col.dtype.kind == TypeKind.DECIMAL128:
         from .evaluate import decimal_to_float
 

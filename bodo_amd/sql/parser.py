@@ -811,6 +811,24 @@ class Parser:
                     while self.accept_op(","):
                         args.append(self.parse_expr())
                     self.expect_op(")")
+                wt = self.peek()
+                if wt and wt.kind == "id" and wt.value.lower() == "within":
+                    # fn(args) WITHIN GROUP (ORDER BY e [ASC|DESC])
+                    self.i += 1
+                    self.expect_kw("group")
+                    self.expect_op("(")
+                    self.expect_kw("order")
+                    self.expect_kw("by")
+                    oe = self.parse_expr()
+                    asc = True
+                    if self.accept_kw("desc"):
+                        asc = False
+                    else:
+                        self.accept_kw("asc")
+                    self.expect_op(")")
+                    f = Func(t.value.lower(), args, distinct=distinct)
+                    f.within_order = (oe, asc)
+                    return f
                 if self.accept_kw("over"):
                     self.expect_op("(")
                     part, order = [], []

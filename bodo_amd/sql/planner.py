@@ -1146,7 +1146,8 @@ class Planner:
         def collect(e):
             if isinstance(e, ast.Func) and (
                     e.name in AGG_FUNCS
-                    or e.name in ("listagg", "string_agg")):
+                    or e.name in ("listagg", "string_agg",
+                                  "percentile_cont", "percentile_disc")):
                 k = _ast_key(e)
                 if k in agg_map:
                     return
@@ -1161,9 +1162,34 @@ class Planner:
                         in_name = self._uniq("__ain")
                         pre_exprs[in_name] = self.expr(arg, scope)
                         pre_names.append(in_name)
-                    if e.name in ("listagg", "string_agg"):
+                    if e.name in ("percentile_cont", "percentile_disc"):
+                        qv = float(e.args[0].value)
+                        interp = ("linear" if e.name == "percentile_cont"
+                                  else "nearest")
+                        wo = getattr(e, "within_order", None)
+                        if wo is not None:
+                            # value column is the WITHIN GROUP order expr
+                            oe = wo[0]
+                            if isinstance(oe, ast.Col):
+                                in_name = scope.resolve(oe.table, oe.name)
+                            else:
+                                in_name = self._uniq("__ain")
+                                pre_exprs[in_name] = self.expr(oe, scope)
+                                pre_names.append(in_name)
+                        func = (lambda s_, _q=qv, _i=interp:
+                                s_.quantile(_q, interpolation=_i))
+                    elif e.name in ("listagg", "string_agg"):
                         sep = e.args[1].value if len(e.args) > 1 else ""
-                        func = _listagg_func(sep)
+                        wo = getattr(e, "within_order", None)
+                        if wo is not None:
+                            if _ast_key(wo[0]) != _ast_key(e.args[0]):
+                                raise NotImplementedError(
+                                    "LISTAGG WITHIN GROUP ordering by a "
+                                    "different column")
+                            func = _listagg_func(sep, sort=True,
+                                                 ascending=wo[1])
+                        else:
+                            func = _listagg_func(sep)
                     else:
                         func = AGG_FUNCS[e.name]
                         if e.distinct and e.name == "count":
@@ -2009,19 +2035,24 @@ def _split_conjuncts(e) -> list:
     return [e]
 
 
-def _listagg_func(sep: str):
+def _listagg_func(sep: str, sort: bool = False, ascending: bool = True):
     """Callable agg for LISTAGG/STRING_AGG (runs on co-located shards via
-    the single-phase path)."""
+    the single-phase path); WITHIN GROUP (ORDER BY <same col>) sorts."""
 
     def listagg(s):
-        return sep.join(str(v) for v in s if v is not None and v == v)
+        vals = [v for v in s if v is not None and v == v]
+        if sort:
+            vals = sorted(vals, reverse=not ascending)
+        return sep.join(str(v) for v in vals)
 
     return listagg
 
 
 def _has_agg(e) -> bool:
     if isinstance(e, ast.Func) and (
-            e.name in AGG_FUNCS or e.name in ("listagg", "string_agg")):
+            e.name in AGG_FUNCS or e.name in (
+                "listagg", "string_agg", "percentile_cont",
+                "percentile_disc")):
         return True
     if isinstance(e, (ast.Query, ast.SetOpQ)):
         return False  # a subquery's aggregates are its own, not the outer's

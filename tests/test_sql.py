@@ -1115,3 +1115,23 @@ def test_semi_structured_constructors_and_array_agg():
     assert [sorted(v) for v in ga] == [[1, 2], [3]]
     ra = bc.sql("select array_agg(x) as ax from t").to_pandas()["ax"]
     assert sorted(ra.iloc[0]) == [1, 2, 3]
+
+
+def test_within_group_percentile_listagg():
+    """WITHIN GROUP: ordered LISTAGG, PERCENTILE_CONT/DISC (grouped and
+    global, callable reduce across ranks)."""
+    df = pd.DataFrame({"g": ["a", "a", "b"], "s": ["y", "x", "z"],
+                       "v": [1.0, 3.0, 2.0]})
+    bc = BodoSQLContext({"t": df})
+    la = bc.sql("select g, listagg(s, ',') within group (order by s) as l "
+                "from t group by g order by g").to_pandas()["l"]
+    assert la.tolist() == ["x,y", "z"]
+    p = bc.sql("select percentile_cont(0.5) within group (order by v) "
+               "as p from t").to_pandas()["p"]
+    assert p.tolist() == [2.0]
+    pg = bc.sql("select g, percentile_cont(0.5) within group "
+                "(order by v) as p from t group by g order by g") \
+        .to_pandas()["p"]
+    assert pg.tolist() == [2.0, 2.0]
+    gl = bc.sql("select listagg(s, '|') as l from t").to_pandas()["l"]
+    assert gl.tolist() == ["y|x|z"]
