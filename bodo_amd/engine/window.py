@@ -29,12 +29,14 @@ from ..plan import nodes as pn
 
 TRANSFORM_AGGS = {"transform_sum": "sum", "transform_mean": "mean",
                   "transform_min": "min", "transform_max": "max",
-                  "transform_count": "count", "transform_size": "size"}
+                  "transform_count": "count", "transform_size": "size",
+                  "transform_nunique": "nunique"}
 
 ORDERED_FUNCS = {"row_number", "rank", "dense_rank", "percent_rank",
                  "cume_dist", "nth_value", "shift", "cumsum",
                  "cumcount", "cummin", "cummax", "cummean", "cumcount_v",
-                 "first_value", "last_value", "ntile",
+                 "first_value", "last_value", "first_value_ig", "last_value_ig",
+                 "ntile",
                  "rolling_sum", "rolling_mean", "rolling_min", "rolling_max",
                  "rolling_count"}
 
@@ -145,6 +147,8 @@ def _ordered_local_device(tbl: Table, keys, order_by, ascending,
     if any(c.dtype.kind == TK.STRING for c in key_cols + order_cols):
         return None
     for _, in_name, func, arg in specs:
+        if func in ("first_value_ig", "last_value_ig"):
+            return None  # ignore-nulls variants: host path
         if func.startswith("rolling_") and func[8:] in ("min", "max") \
                 and int(arg or 1) > _ROLLING_MINMAX_CAP:
             return None
@@ -498,6 +502,14 @@ def _ordered_local(tbl: Table, keys, order_by, ascending, specs) -> Table:
         elif func in ("first_value", "last_value"):
             which = "first" if func == "first_value" else "last"
             res = gbs[in_name].transform(which).reindex(pdf.index)
+        elif func in ("first_value_ig", "last_value_ig"):
+            # IGNORE NULLS: partition first/last non-null value
+            take_first = func.startswith("first")
+            res = gbs[in_name].transform(
+                lambda s, _f=take_first: (
+                    (s.dropna().iloc[0] if _f else s.dropna().iloc[-1])
+                    if s.notna().any() else np.nan))
+            res = res.reindex(pdf.index)
         elif func.startswith("rolling_"):
             base = func[len("rolling_"):]
             r = gbs[in_name].rolling(int(arg), min_periods=1)

@@ -178,6 +178,7 @@ class WindowE:
     order_by: List[Tuple[Any, bool]]
     star: bool = False
     frame_preceding: Optional[int] = None
+    distinct: bool = False
 
 
 @dataclass
@@ -811,6 +812,13 @@ class Parser:
                     while self.accept_op(","):
                         args.append(self.parse_expr())
                     self.expect_op(")")
+                ig = self.peek()
+                ignore_nulls = False
+                if ig and ig.kind == "id" and ig.value.lower() in (
+                        "ignore", "respect"):
+                    self.i += 1
+                    self.next()  # NULLS
+                    ignore_nulls = ig.value.lower() == "ignore"
                 wt = self.peek()
                 if wt and wt.kind == "id" and wt.value.lower() == "within":
                     # fn(args) WITHIN GROUP (ORDER BY e [ASC|DESC])
@@ -862,12 +870,21 @@ class Parser:
                             frame = int(ft.value)
                             self.next()  # PRECEDING
                         self.expect_kw("and")
-                        cur = self.next()  # CURRENT
-                        assert cur.value.lower() == "current", cur
-                        self.next()  # ROW
+                        cur = self.next()  # CURRENT | UNBOUNDED
+                        if cur.value.lower() == "unbounded":
+                            self.next()  # FOLLOWING: whole-partition frame
+                            frame = None if frame == -1 else frame
+                            # start=UNBOUNDED end=UNBOUNDED -> no frame
+                            # (partition-wide); bounded start keeps trailing
+                        else:
+                            assert cur.value.lower() == "current", cur
+                            self.next()  # ROW
                     self.expect_op(")")
-                    return WindowE(t.value.lower(), args, part, order,
-                                   star=star, frame_preceding=frame)
+                    w = WindowE(t.value.lower(), args, part, order,
+                                star=star, frame_preceding=frame,
+                                distinct=distinct)
+                    w.ignore_nulls = ignore_nulls
+                    return w
                 if star:
                     return Func(t.value.lower(), [], star=True)
                 return Func(t.value.lower(), args, distinct=distinct)

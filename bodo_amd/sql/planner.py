@@ -370,7 +370,10 @@ class Planner:
             if fn == "row_number":
                 spec = (out, "", "row_number", None)
             elif fn in ("first_value", "last_value"):
-                spec = (out, arg_col, fn, None)
+                if getattr(w, "ignore_nulls", False):
+                    spec = (out, arg_col, fn + "_ig", None)
+                else:
+                    spec = (out, arg_col, fn, None)
             elif fn == "nth_value":
                 k = int(w.args[1].value) if len(w.args) > 1 else 1
                 spec = (out, arg_col, "nth_value", k)
@@ -417,6 +420,8 @@ class Planner:
                         spec = (out, arg_col, running[base], None)
                 elif fn == "count" and (w.star or not w.args):
                     spec = (out, "", "transform_size", None)
+                elif fn == "count" and w.distinct:
+                    spec = (out, arg_col, "transform_nunique", None)
                 else:
                     spec = (out, arg_col,
                             f"transform_{AGG_FUNCS.get(fn, fn)}", None)

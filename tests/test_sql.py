@@ -1135,3 +1135,21 @@ def test_within_group_percentile_listagg():
     assert pg.tolist() == [2.0, 2.0]
     gl = bc.sql("select listagg(s, '|') as l from t").to_pandas()["l"]
     assert gl.tolist() == ["y|x|z"]
+
+
+def test_ignore_nulls_and_full_frames():
+    """FIRST/LAST_VALUE IGNORE NULLS, UNBOUNDED FOLLOWING frames,
+    COUNT(DISTINCT) OVER."""
+    df = pd.DataFrame({"g": ["a", "a", "a", "b"], "o": [1, 2, 3, 1],
+                       "v": [None, 1.0, 2.0, None]})
+    bc = BodoSQLContext({"t": df})
+    f = bc.sql("select first_value(v) ignore nulls over "
+               "(partition by g order by o) as f from t").to_pandas()["f"]
+    assert f.tolist()[:3] == [1.0, 1.0, 1.0] and pd.isna(f[3])
+    l = bc.sql("select last_value(v) over (partition by g order by o rows "
+               "between unbounded preceding and unbounded following) as l "
+               "from t").to_pandas()["l"]
+    assert l.tolist()[:3] == [2.0, 2.0, 2.0]
+    c = bc.sql("select count(distinct g) over () as c from t") \
+        .to_pandas()["c"]
+    assert c.tolist() == [2, 2, 2, 2]
