@@ -1351,6 +1351,67 @@ class Planner:
                 items = tuple(self.expr(e.args[i], scope)
                               for i in range(1, len(e.args), 2))
                 return ex.StructBuild(names, items)
+            if name == "parse_json":
+                import json as _json
+
+                def _pj(v):
+                    try:
+                        return _json.loads(v)
+                    except Exception:
+                        return None
+
+                return ex.UdfMap(self.expr(e.args[0], scope), _pj, "ignore")
+            if name == "regexp_instr":
+                import re as _re
+
+                pat = e.args[1].value
+
+                def _ri(v, _p=pat):
+                    m = _re.search(_p, str(v))
+                    return m.start() + 1 if m else 0
+
+                return ex.UdfMap(self.expr(e.args[0], scope), _ri, "ignore")
+            if name == "getbit":
+                k = _lit_int(e.args[1])
+                shifted = ex.BinOp("rshift", self.expr(e.args[0], scope),
+                                   ex.Const(k))
+                return ex.BinOp("bitand", shifted, ex.Const(1))
+            if name == "atan2":
+                import math as _math
+
+                y = self.expr(e.args[0], scope)
+                xv = _lit_num(e.args[1])
+                return ex.UdfMap(y, lambda v, _x=xv: _math.atan2(v, _x),
+                                 "ignore")
+            if name == "cot":
+                import math as _math
+
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v: 1.0 / _math.tan(v), "ignore")
+            if name == "haversine":
+                import math as _math
+
+                vals = [_lit_num(a) for a in e.args]
+                la1, lo1, la2, lo2 = [_math.radians(v) for v in vals]
+                h = (_math.sin((la2 - la1) / 2) ** 2
+                     + _math.cos(la1) * _math.cos(la2)
+                     * _math.sin((lo2 - lo1) / 2) ** 2)
+                return ex.Const(2 * 6371.0087714 * _math.asin(_math.sqrt(h)))
+            if name == "soundex":
+                return ex.UdfMap(self.expr(e.args[0], scope), _soundex,
+                                 "ignore")
+            if name in ("editdistance", "edit_distance"):
+                other = e.args[1].value
+                return ex.UdfMap(self.expr(e.args[0], scope),
+                                 lambda v, _o=other: _levenshtein(str(v),
+                                                                  _o),
+                                 "ignore")
+            if name == "jarowinkler_similarity":
+                other = e.args[1].value
+                return ex.UdfMap(
+                    self.expr(e.args[0], scope),
+                    lambda v, _o=other: int(round(
+                        _jaro_winkler(str(v), _o) * 100)), "ignore")
             if name == "div0":
                 # a / b, 0 when b = 0 (Snowflake DIV0)
                 a = self.expr(e.args[0], scope)
@@ -2104,3 +2165,73 @@ def _has_window(e) -> bool:
         elif hasattr(v, "__dataclass_fields__") and _has_window(v):
             return True
     return False
+
+
+def _soundex(v):
+    """American Soundex (Snowflake SOUNDEX)."""
+    s = "".join(c for c in str(v).upper() if c.isalpha())
+    if not s:
+        return ""
+    codes = {**{c: "1" for c in "BFPV"}, **{c: "2" for c in "CGJKQSXZ"},
+             **{c: "3" for c in "DT"}, "L": "4",
+             **{c: "5" for c in "MN"}, "R": "6"}
+    out = s[0]
+    prev = codes.get(s[0], "")
+    for c in s[1:]:
+        d = codes.get(c, "")
+        if d and d != prev:
+            out += d
+        if c not in "HW":
+            prev = d
+    return (out + "000")[:4]
+
+
+def _levenshtein(a: str, b: str) -> int:
+    if len(a) < len(b):
+        a, b = b, a
+    prev = list(range(len(b) + 1))
+    for i, ca in enumerate(a, 1):
+        cur = [i]
+        for j, cb in enumerate(b, 1):
+            cur.append(min(prev[j] + 1, cur[j - 1] + 1,
+                           prev[j - 1] + (ca != cb)))
+        prev = cur
+    return prev[-1]
+
+
+def _jaro_winkler(a: str, b: str) -> float:
+    if a == b:
+        return 1.0
+    la, lb = len(a), len(b)
+    if not la or not lb:
+        return 0.0
+    win = max(la, lb) // 2 - 1
+    ma = [False] * la
+    mb = [False] * lb
+    m = 0
+    for i in range(la):
+        lo, hi = max(0, i - win), min(lb, i + win + 1)
+        for j in range(lo, hi):
+            if not mb[j] and a[i] == b[j]:
+                ma[i] = mb[j] = True
+                m += 1
+                break
+    if m == 0:
+        return 0.0
+    t = 0
+    k = 0
+    for i in range(la):
+        if ma[i]:
+            while not mb[k]:
+                k += 1
+            if a[i] != b[k]:
+                t += 1
+            k += 1
+    t //= 2
+    jaro = (m / la + m / lb + (m - t) / m) / 3
+    pref = 0
+    for x, y in zip(a, b):
+        if x != y or pref == 4:
+            break
+        pref += 1
+    return jaro + pref * 0.1 * (1 - jaro)

@@ -1153,3 +1153,33 @@ def test_ignore_nulls_and_full_frames():
     c = bc.sql("select count(distinct g) over () as c from t") \
         .to_pandas()["c"]
     assert c.tolist() == [2, 2, 2, 2]
+
+
+def test_string_distance_and_misc_functions():
+    """SOUNDEX, EDITDISTANCE, JAROWINKLER_SIMILARITY, REGEXP_INSTR,
+    GETBIT, HAVERSINE, COT/ATAN2, PARSE_JSON."""
+    import math
+
+    df = pd.DataFrame({"s": ["a1b2", "Robert", None],
+                       "n": [3, 7, 15], "v": [1.0, -1.0, 0.5]})
+    bc = BodoSQLContext({"t": df})
+    assert bc.sql("select soundex(s) as r from t").to_pandas()["r"] \
+        .tolist()[:2] == ["A100", "R163"]
+    assert bc.sql("select editdistance(s, 'a1b3') as r from t") \
+        .to_pandas()["r"].tolist()[:2] == [1, 5]
+    assert bc.sql("select jarowinkler_similarity(s, 'a1b2') as r from t") \
+        .to_pandas()["r"][0] == 100
+    assert bc.sql("select regexp_instr(s, '[0-9]') as r from t") \
+        .to_pandas()["r"].tolist()[:2] == [2, 0]
+    assert bc.sql("select getbit(n, 1) as r from t") \
+        .to_pandas()["r"].tolist() == [1, 1, 1]
+    hv = bc.sql("select haversine(10, 20, 30, 40) as r from t") \
+        .to_pandas()["r"][0]
+    assert abs(hv - 3040.6) < 1.0
+    row = bc.sql("select cot(v) as c, atan2(v, 2) as a from t limit 1") \
+        .to_pandas().iloc[0]
+    assert abs(row["c"] - 1 / math.tan(1.0)) < 1e-12
+    assert abs(row["a"] - math.atan2(1.0, 2)) < 1e-12
+    pj = bc.sql("select parse_json('{\"a\": 1}') as r from t limit 1") \
+        .to_pandas()["r"]
+    assert pj.iloc[0] == {"a": 1}
