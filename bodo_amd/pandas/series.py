@@ -274,7 +274,7 @@ class BodoSeries:
         plan = pn.Distinct(self._as_projection_plan(), ("v",))
         return api.collect(plan)["v"].to_numpy()
 
-    def value_counts(self, ascending=False, dropna=True):
+    def value_counts(self, normalize=False, ascending=False, dropna=True):
         from ..engine import api
 
         plan = pn.Aggregate(self._as_projection_plan(), ("v",),
@@ -283,8 +283,12 @@ class BodoSeries:
         plan = pn.Sort(plan, ("count", "v"), (ascending, True))
         pdf = api.collect(plan)
         # ties: pandas orders by value; match roughly
-        out = pd.Series(pdf["count"].to_numpy(), index=pdf["v"].to_numpy(),
-                        name="count")
+        vals = pdf["count"].to_numpy()
+        name = "count"
+        if normalize:
+            vals = vals / vals.sum() if vals.sum() else vals.astype(float)
+            name = "proportion"
+        out = pd.Series(vals, index=pdf["v"].to_numpy(), name=name)
         out.index.name = self.name
         return out
 
@@ -549,6 +553,9 @@ class BodoSeries:
 
     def cummax(self):
         return self._cum("cummax")
+
+    def mask(self, cond, other=float("nan")) -> "BodoSeries":
+        return self.where(~cond, other)
 
     def rank(self, method="average", ascending=True,
              pct=False) -> "BodoSeries":

@@ -780,3 +780,12 @@ def test_corrwith_and_dot():
     np.testing.assert_allclose(b[["x", "y"]].corrwith(b["z"]).values,
                                df[["x", "y"]].corrwith(df["z"]).values)
     assert abs(float(b["x"].dot(b["y"])) - df["x"].dot(df["y"])) < 1e-9
+
+
+def test_value_counts_normalize_and_mask():
+    df = pd.DataFrame({"g": ["a", "b", "a", "c"], "x": [1, 2, 3, 4]})
+    b = bpd.from_pandas(df)
+    vc = b["g"].value_counts(normalize=True)
+    assert abs(vc.sum() - 1.0) < 1e-12 and max(vc) == 0.5
+    m = b["x"].mask(b["x"] > 2, 0).to_pandas()
+    assert m.tolist() == df["x"].mask(df["x"] > 2, 0).tolist()
