@@ -877,9 +877,12 @@ def _dt_field_cpu(a: Column, fld: str) -> Column:
     if fld == "last_day":
         out = (idx + pd.offsets.MonthEnd(0)).normalize().asi8
         return _dt_res(bt.timestamp_ns, out, a, fld)
-    if fld == "epoch_second":
+    if fld in ("epoch_second", "epoch"):
         out = idx.asi8 // 10**9
         return _dt_res(bt.int64, out, a, fld)
+    if fld in ("week", "weekofyear", "weekiso"):
+        out = np.asarray(idx.isocalendar().week, dtype=np.int16)
+        return _dt_res(bt.int16, out, a, fld)
     if fld in ("dayname", "monthname"):
         import pyarrow as pa
 

@@ -1183,3 +1183,12 @@ def test_string_distance_and_misc_functions():
     pj = bc.sql("select parse_json('{\"a\": 1}') as r from t limit 1") \
         .to_pandas()["r"]
     assert pj.iloc[0] == {"a": 1}
+
+
+def test_extract_week_epoch():
+    df = pd.DataFrame({"d": pd.to_datetime(["2024-03-05 10:30:45"])})
+    bc = BodoSQLContext({"t": df})
+    assert bc.sql("select extract(week from d) as r from t") \
+        .to_pandas()["r"][0] == 10
+    assert bc.sql("select extract(epoch from d) as r from t") \
+        .to_pandas()["r"][0] == int(df["d"][0].timestamp())
