@@ -1062,3 +1062,22 @@ def test_dist_array_agg():
     assert got["g"] == sorted(exp)
     for g, lst in zip(got["g"], got["lists"]):
         assert lst == exp[g]
+
+
+def _q_global_listagg(bpd, rank, payload):
+    from bodo_amd.sql import BodoSQLContext
+
+    bc = BodoSQLContext({"t": payload["df"]})
+    out = bc.sql("select listagg(s, ',') as l, "
+                 "percentile_cont(0.5) within group (order by v) as p "
+                 "from t").to_pandas()
+    return {"l": out["l"][0], "p": float(out["p"][0])}
+
+
+def test_dist_global_listagg_percentile():
+    """2-rank callable reduces: LISTAGG / PERCENTILE_CONT combine the
+    gathered values in rank order."""
+    df = pd.DataFrame({"s": list("abcdef"), "v": [1.0, 5, 2, 4, 3, 6]})
+    got = run_dist(_q_global_listagg, {"df": df})
+    assert got["l"] == "a,b,c,d,e,f"
+    assert got["p"] == 3.5
