@@ -1570,12 +1570,17 @@ class Planner:
                      ex.Cmp("gt", x, ex.Const(0))),
                     (ex.Const(-1), ex.Const(1)), ex.Const(0))
             if name in ("greatest", "least"):
+                # Snowflake: NULL if ANY argument is NULL (the pairwise
+                # CASE chain alone leaked null-row storage values)
                 op = "gt" if name == "greatest" else "lt"
                 args = [self.expr(a, scope) for a in e.args]
                 out = args[0]
                 for a in args[1:]:
                     out = ex.Case((ex.Cmp(op, out, a),), (out,), a)
-                return out
+                any_null = ex.IsNull(args[0])
+                for a in args[1:]:
+                    any_null = ex.BoolOp("or", any_null, ex.IsNull(a))
+                return ex.Case((any_null,), (ex.Const(None, None),), out)
             if name == "replace":
                 return ex.StrOp(self.expr(e.args[0], scope), "replace",
                                 (e.args[1].value, e.args[2].value),

@@ -1192,3 +1192,14 @@ def test_extract_week_epoch():
         .to_pandas()["r"][0] == 10
     assert bc.sql("select extract(epoch from d) as r from t") \
         .to_pandas()["r"][0] == int(df["d"][0].timestamp())
+
+
+def test_greatest_least_null_semantics():
+    """Snowflake GREATEST/LEAST: NULL when any argument is NULL (the
+    pairwise CASE chain leaked storage values of null rows)."""
+    df = pd.DataFrame({"x": [1, 2, 3], "y": [0.5, None, 2.0]})
+    bc = BodoSQLContext({"t": df})
+    g = bc.sql("select greatest(x, y, 0) as g from t").to_pandas()["g"]
+    assert g[0] == 1.0 and pd.isna(g[1]) and g[2] == 3.0
+    l = bc.sql("select least(x, y) as l from t").to_pandas()["l"]
+    assert l[0] == 0.5 and pd.isna(l[1]) and l[2] == 2.0
