@@ -975,8 +975,16 @@ def cast_column(a: Column, to: DType, safe: bool = False) -> Column:
         # host path for string casts
         ser = a.to_pandas()
         if to.kind == TypeKind.STRING:
-            res = ser.astype(str)
-            return Column.from_arrow(__import__("pyarrow").array(res), a.device)
+            import pyarrow as pa
+
+            if a.dtype.is_integer:
+                # via pylist: a masked int must print "1", not "1.0"/"nan"
+                vals = a.to_arrow().to_pylist()
+                res = pd.Series([None if v is None else str(v)
+                                 for v in vals])
+            else:
+                res = ser.astype(str).where(ser.notna(), None)
+            return Column.from_arrow(pa.Array.from_pandas(res), a.device)
         npv = ser.to_numpy(dtype=bt.numpy_storage_dtype(to))
         return Column(to, torch.from_numpy(npv).to(a.device))
     if to.kind == TypeKind.TIMESTAMP_NS and a.dtype.kind == TypeKind.DATE32:

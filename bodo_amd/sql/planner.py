@@ -1566,9 +1566,10 @@ class Planner:
             if name == "sign":
                 x = self.expr(e.args[0], scope)
                 return ex.Case(
-                    (ex.Cmp("lt", x, ex.Const(0)),
+                    (ex.IsNull(x), ex.Cmp("lt", x, ex.Const(0)),
                      ex.Cmp("gt", x, ex.Const(0))),
-                    (ex.Const(-1), ex.Const(1)), ex.Const(0))
+                    (ex.Const(None, None), ex.Const(-1), ex.Const(1)),
+                    ex.Const(0))
             if name in ("greatest", "least"):
                 # Snowflake: NULL if ANY argument is NULL (the pairwise
                 # CASE chain alone leaked null-row storage values)

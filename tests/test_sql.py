@@ -1203,3 +1203,17 @@ def test_greatest_least_null_semantics():
     assert g[0] == 1.0 and pd.isna(g[1]) and g[2] == 3.0
     l = bc.sql("select least(x, y) as l from t").to_pandas()["l"]
     assert l[0] == 0.5 and pd.isna(l[1]) and l[2] == 2.0
+
+
+def test_null_value_semantics_cast_sign():
+    """Masked-int -> varchar prints '1' not '1.0'/'nan'; SIGN(NULL) is
+    NULL (storage-leak fixes)."""
+    df = pd.DataFrame({"x": pd.array([1, None, 3], dtype="Int64"),
+                       "s": ["a", None, "c"]})
+    bc = BodoSQLContext({"t": df})
+    cv = bc.sql("select cast(x as varchar) as r from t").to_pandas()["r"]
+    assert cv.where(cv.notna(), None).tolist() == ["1", None, "3"]
+    sg = bc.sql("select sign(x) as r from t").to_pandas()["r"]
+    assert sg[0] == 1 and pd.isna(sg[1]) and sg[2] == 1
+    cs = bc.sql("select cast(s as varchar) as r from t").to_pandas()["r"]
+    assert cs.where(cs.notna(), None).tolist() == ["a", None, "c"]
