@@ -985,6 +985,11 @@ def cast_column(a: Column, to: DType, safe: bool = False) -> Column:
             else:
                 res = ser.astype(str).where(ser.notna(), None)
             return Column.from_arrow(pa.Array.from_pandas(res), a.device)
+        if to.kind == TypeKind.TIMESTAMP_NS:
+            import pyarrow as pa
+
+            ts = pd.to_datetime(ser)
+            return Column.from_arrow(pa.Array.from_pandas(ts), a.device)
         npv = ser.to_numpy(dtype=bt.numpy_storage_dtype(to))
         return Column(to, torch.from_numpy(npv).to(a.device))
     if to.kind == TypeKind.TIMESTAMP_NS and a.dtype.kind == TypeKind.DATE32:

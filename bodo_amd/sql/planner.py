@@ -1256,6 +1256,15 @@ class Planner:
             return ex.RoundExpr(self._post_agg_expr(e.args[0], key_map, agg_map,
                                                     scope),
                                 int(e.args[1].value) if len(e.args) > 1 else 0)
+        if isinstance(e, ast.Func) and _FN_ALIASES.get(
+                e.name, e.name) == "coalesce":
+            # IFNULL/NVL/COALESCE over aggregate outputs
+            args = [self._post_agg_expr(a, key_map, agg_map, scope)
+                    for a in e.args]
+            out = args[-1]
+            for a in reversed(args[:-1]):
+                out = ex.Case((ex.IsNull(a),), (out,), a)
+            return out
         if isinstance(e, ast.CastE):
             return ex.Cast(self._post_agg_expr(e.operand, key_map, agg_map,
                                                scope), CAST_TYPES[e.to],
@@ -1945,6 +1954,10 @@ class Planner:
                 return ex.DtField(self.expr(e.args[0], scope), "last_day")
             if name in ("to_date", "try_to_date", "date"):
                 return ex.DtField(self.expr(e.args[0], scope), "floor_day")
+            if name in ("to_timestamp", "try_to_timestamp",
+                        "to_timestamp_ntz"):
+                return ex.Cast(self.expr(e.args[0], scope), bt.timestamp_ns,
+                               name.startswith("try"))
             if name in ("epoch_second", "date_part_epoch_second"):
                 return ex.DtField(self.expr(e.args[0], scope),
                                   "epoch_second")

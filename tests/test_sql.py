@@ -1217,3 +1217,15 @@ def test_null_value_semantics_cast_sign():
     assert sg[0] == 1 and pd.isna(sg[1]) and sg[2] == 1
     cs = bc.sql("select cast(s as varchar) as r from t").to_pandas()["r"]
     assert cs.where(cs.notna(), None).tolist() == ["a", None, "c"]
+
+
+def test_to_timestamp_and_postagg_ifnull():
+    df = pd.DataFrame({"s": ["2024-01-02 03:04:05", None],
+                       "g": ["a", "b"], "v": [1.0, None]})
+    bc = BodoSQLContext({"t": df})
+    out = bc.sql("select to_timestamp(s) as r from t").to_pandas()["r"]
+    assert pd.Timestamp(out[0]) == pd.Timestamp("2024-01-02 03:04:05")
+    assert pd.isna(out[1])
+    r = bc.sql("select g, ifnull(sum(v), 0) as r from t group by g "
+               "order by g").to_pandas()["r"]
+    assert r.tolist() == [1.0, 0.0]
